@@ -1,0 +1,109 @@
+"""Llama 2 / 3.x for MI355X.
+
+The canonical dense model (ref `models/llama/modeling_llama.py`, 1378 LoC):
+decoder layer = RMSNorm -> attention -> residual -> RMSNorm -> gated MLP ->
+residual, with Llama-3 scaled RoPE handled in `modules.rope`
+(ref :805-871) and HF->native state-dict key rewriting
+(ref convert_hf_to_neuron_state_dict :1208-1276).
+
+Also serves Mistral/Qwen2-style architectures through small subclasses.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List
+
+import torch
+import torch.nn as nn
+
+from neuronx_distributed_inference_amd.config import InferenceConfig
+from neuronx_distributed_inference_amd.models.model_base import CausalLMBase, DecoderModel
+from neuronx_distributed_inference_amd.modules.attention.attention_base import AttentionBase
+from neuronx_distributed_inference_amd.modules.mlp import GatedMLP
+from neuronx_distributed_inference_amd.modules.norm import RMSNorm
+
+
+class LlamaInferenceConfig(InferenceConfig):
+    def get_required_attributes(self) -> List[str]:
+        return [
+            "hidden_size",
+            "num_attention_heads",
+            "num_hidden_layers",
+            "num_key_value_heads",
+            "vocab_size",
+            "intermediate_size",
+        ]
+
+    def add_derived_config(self):
+        if not hasattr(self, "rms_norm_eps"):
+            self.rms_norm_eps = 1e-6
+        if not hasattr(self, "rope_theta"):
+            self.rope_theta = 10000.0
+        if not hasattr(self, "num_key_value_heads"):
+            self.num_key_value_heads = getattr(self, "num_attention_heads", None)
+        if not hasattr(self, "tie_word_embeddings"):
+            self.tie_word_embeddings = False
+
+
+class LlamaDecoderLayer(nn.Module):
+    def __init__(self, config: InferenceConfig, layer_idx: int):
+        super().__init__()
+        ec = config.neuron_config
+        self.self_attn = AttentionBase(config, layer_idx)
+        self.mlp = GatedMLP(
+            config.hidden_size,
+            config.intermediate_size,
+            tp_degree=ec.tp_degree,
+            dtype=ec.torch_dtype,
+            hidden_act=getattr(config, "hidden_act", "silu"),
+            bias=bool(getattr(config, "mlp_bias", False)),
+        )
+        self.input_layernorm = RMSNorm(
+            config.hidden_size, eps=config.rms_norm_eps, dtype=ec.torch_dtype
+        )
+        self.post_attention_layernorm = RMSNorm(
+            config.hidden_size, eps=config.rms_norm_eps, dtype=ec.torch_dtype
+        )
+
+    def forward(self, h, cos, sin, kv_mgr, position_ids, phase, attention_mask, seq_ids):
+        residual = h
+        h = self.input_layernorm(h)
+        h = self.self_attn(h, cos, sin, kv_mgr, position_ids, phase, attention_mask, seq_ids)
+        h = residual + h
+        residual = h
+        h = self.post_attention_layernorm(h)
+        h = self.mlp(h)
+        return residual + h
+
+
+class LlamaModel(DecoderModel):
+    def build_layers(self, config: InferenceConfig) -> nn.ModuleList:
+        return nn.ModuleList(
+            LlamaDecoderLayer(config, i) for i in range(config.num_hidden_layers)
+        )
+
+
+class LlamaForCausalLM(CausalLMBase):
+    _model_cls = LlamaModel
+    _config_cls = LlamaInferenceConfig
+
+    @staticmethod
+    def convert_hf_to_native_state_dict(
+        sd: Dict[str, torch.Tensor], config: InferenceConfig
+    ) -> Dict[str, torch.Tensor]:
+        out: Dict[str, torch.Tensor] = {}
+        for k, v in sd.items():
+            nk = k
+            if nk.startswith("model."):
+                nk = nk[len("model."):]
+            # attention/mlp projections keep their names but move under the
+            # gqa/mlp wrappers
+            nk = nk.replace("self_attn.q_proj", "self_attn.qkv_proj.q_proj")
+            nk = nk.replace("self_attn.k_proj", "self_attn.qkv_proj.k_proj")
+            nk = nk.replace("self_attn.v_proj", "self_attn.qkv_proj.v_proj")
+            nk = nk.replace("self_attn.o_proj", "self_attn.o_proj.o_proj")
+            out[nk] = v
+        if "lm_head.weight" not in out and getattr(config, "tie_word_embeddings", False):
+            out["lm_head.weight"] = out["embed_tokens.weight"]
+        dtype = config.neuron_config.torch_dtype
+        return {k: v.to(dtype) for k, v in out.items()}

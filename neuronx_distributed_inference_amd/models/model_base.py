@@ -1,0 +1,255 @@
+"""Model base: the device graph body and the host-side dispatcher.
+
+Analogs of the reference's `models/model_base.py`:
+
+- :class:`DecoderModel` ~ ``NeuronBaseModel`` (:70-1597): embeddings ->
+  decoder layers (attention + MLP with the HIP kernels + RCCL collectives)
+  -> norm -> last-token gather -> lm_head -> on-device sampler. One module
+  instance serves every phase; context-encoding vs token-generation is
+  selected from the input shape exactly like the reference
+  (`_is_context_encoding` :181, `_is_for_speculation` :184). This is the
+  module that hipGraph capture records per (batch, seq) bucket.
+
+- :class:`CausalLMBase` ~ ``NeuronBaseForCausalLM`` (:3024-3960): host-side
+  phase dispatch by ``position_ids.min() == 0`` (:3546), submodel wrappers,
+  ``kv_cache_populated`` / ``reset()`` bookkeeping (:3926).
+"""
+
+from __future__ import annotations
+
+import logging
+import os
+from dataclasses import dataclass
+from typing import Dict, List, Optional
+
+import torch
+import torch.nn as nn
+
+from neuronx_distributed_inference_amd.config import InferenceConfig
+from neuronx_distributed_inference_amd.modules.attention.attention_base import (
+    PHASE_CTE,
+    PHASE_TKG,
+)
+from neuronx_distributed_inference_amd.modules.generation.sampling import (
+    Sampler,
+    prepare_sampling_params,
+)
+from neuronx_distributed_inference_amd.modules.kvcache.kv_cache_manager import KVCacheManager
+from neuronx_distributed_inference_amd.modules.norm import RMSNorm
+from neuronx_distributed_inference_amd.modules.rope import RotaryEmbedding
+from neuronx_distributed_inference_amd.parallel import mappings
+from neuronx_distributed_inference_amd.parallel.layers import (
+    ColumnParallelLinear,
+    ParallelEmbedding,
+)
+
+logger = logging.getLogger("nxdi_amd")
+
+
+@dataclass
+class ModelOutput:
+    """Host-visible output of one forward (HF CausalLMOutputWithPast-shaped,
+    ref model_base.py:3797-3862)."""
+
+    tokens: Optional[torch.Tensor] = None   # (B,) or (B, spec) sampled ids
+    logits: Optional[torch.Tensor] = None   # (B, S, V) when output_logits
+    hidden_states: Optional[torch.Tensor] = None
+
+
+class DecoderModel(nn.Module):
+    """The device graph body. Subclasses implement `build_layers` and
+    (optionally) `build_rotary`."""
+
+    def __init__(self, config: InferenceConfig):
+        super().__init__()
+        ec = config.neuron_config
+        self.config = config
+        self.ec = ec
+        self.vocab_size = config.vocab_size
+        self.hidden_size = config.hidden_size
+        self.num_layers = config.num_hidden_layers
+        self.dtype = ec.torch_dtype
+
+        self.embed_tokens = ParallelEmbedding(
+            config.vocab_size, config.hidden_size, dtype=self.dtype,
+            tp_degree=ec.tp_degree,
+        )
+        self.rotary = self.build_rotary(config)
+        self.layers = self.build_layers(config)
+        self.norm = RMSNorm(
+            config.hidden_size, eps=getattr(config, "rms_norm_eps", 1e-6), dtype=self.dtype
+        )
+        self.lm_head = ColumnParallelLinear(
+            config.hidden_size, config.vocab_size, bias=False, dtype=self.dtype,
+            tp_degree=ec.tp_degree,
+        )
+        head_dim = getattr(config, "head_dim", None) or config.hidden_size // config.num_attention_heads
+        self.kv_mgr = KVCacheManager(
+            num_layers=self.num_layers,
+            num_kv_heads=getattr(config, "num_key_value_heads", config.num_attention_heads),
+            head_dim=head_dim,
+            max_len=ec.max_length,
+            batch_size=ec.kv_cache_batch_size,
+            dtype=ec.torch_dtype,
+            padding_rows=ec.kv_cache_padding_size,
+            tp_degree=ec.tp_degree,
+        )
+        self.sampler = (
+            Sampler(ec.on_device_sampling_config, config.vocab_size)
+            if ec.on_device_sampling_config
+            else None
+        )
+
+    # -- subclass hooks -------------------------------------------------
+    def build_rotary(self, config: InferenceConfig) -> RotaryEmbedding:
+        head_dim = getattr(config, "head_dim", None) or config.hidden_size // config.num_attention_heads
+        return RotaryEmbedding(
+            head_dim,
+            max_position_embeddings=getattr(config, "max_position_embeddings", 4096),
+            base=getattr(config, "rope_theta", 10000.0),
+            scaling=getattr(config, "rope_scaling", None),
+        )
+
+    def build_layers(self, config: InferenceConfig) -> nn.ModuleList:
+        raise NotImplementedError
+
+    # -- phase detection (ref model_base.py:181-186) --------------------
+    @staticmethod
+    def _phase(input_ids: torch.Tensor, position_ids: torch.Tensor) -> str:
+        if input_ids.shape[-1] > 1 and bool(position_ids.min() == 0):
+            return PHASE_CTE
+        return PHASE_TKG
+
+    # -- the graph body -------------------------------------------------
+    def forward(
+        self,
+        input_ids: torch.Tensor,
+        attention_mask: Optional[torch.Tensor],
+        position_ids: torch.Tensor,
+        seq_ids: Optional[torch.Tensor] = None,
+        sampling_params: Optional[torch.Tensor] = None,
+    ):
+        phase = self._phase(input_ids, position_ids)
+        B, S = input_ids.shape
+        h = self.embed_tokens(input_ids)
+        cos, sin = self.rotary.get_cos_sin(position_ids, device=h.device)
+        for layer in self.layers:
+            h = layer(h, cos, sin, self.kv_mgr, position_ids, phase, attention_mask, seq_ids)
+        h = self.norm(h)
+
+        if phase == PHASE_CTE and S > 1:
+            if attention_mask is not None:
+                last = attention_mask.sum(dim=-1).long() - 1
+            else:
+                last = torch.full((B,), S - 1, dtype=torch.long, device=h.device)
+            h_last = h[torch.arange(B, device=h.device), last].unsqueeze(1)
+        else:
+            h_last = h[:, -1:, :]
+
+        logits_shard = self.lm_head(h_last)  # (B, 1|S, V/tp)
+
+        tokens = None
+        logits = None
+        if self.sampler is not None:
+            tokens = self.sampler(logits_shard, sampling_params)
+        if self.sampler is None or self.ec.output_logits:
+            logits = mappings.gather_from_tensor_parallel_region(logits_shard, dim=-1)
+        return tokens, logits
+
+    def reset_cache(self):
+        self.kv_mgr.reset()
+
+
+class CausalLMBase(nn.Module):
+    """Host-side application model: owns the DecoderModel, dispatches
+    prefill vs decode, tracks KV state. Submodel graph capture and bucketing
+    are layered on by `ModelWrapper`/`ApplicationBase`."""
+
+    _model_cls = DecoderModel
+
+    def __init__(self, config: InferenceConfig, model: Optional[DecoderModel] = None):
+        super().__init__()
+        self.config = config
+        self.ec = config.neuron_config
+        self.model = model or self._model_cls(config)
+        self.kv_cache_populated = False
+
+    # ------------------------------------------------------------------
+    @classmethod
+    def from_pretrained(cls, model_path: str, config: InferenceConfig) -> "CausalLMBase":
+        """Instantiate with HF checkpoint weights (full, unsharded — used by
+        the CPU path and as the sharding source)."""
+        obj = cls(config)
+        from neuronx_distributed_inference_amd.modules.checkpoint import load_state_dict
+
+        hf_sd = load_state_dict(model_path)
+        sd = cls.convert_hf_to_native_state_dict(hf_sd, config)
+        from neuronx_distributed_inference_amd.utils.checkpoint_sharding import (
+            shard_state_dict_for_rank,
+        )
+        from neuronx_distributed_inference_amd.parallel import state as ps
+
+        tp = config.neuron_config.tp_degree
+        if tp > 1:
+            sd = shard_state_dict_for_rank(
+                sd, obj.model, ps.get_tensor_model_parallel_rank(), tp
+            )
+        missing, unexpected = obj.model.load_state_dict(sd, strict=False)
+        real_missing = [m for m in missing if not m.startswith("kv_mgr.")]
+        if real_missing:
+            raise RuntimeError(f"missing weights: {real_missing[:10]}")
+        if unexpected:
+            logger.warning("unexpected checkpoint keys: %s", unexpected[:10])
+        return obj
+
+    @staticmethod
+    def convert_hf_to_native_state_dict(sd: Dict[str, torch.Tensor], config) -> Dict[str, torch.Tensor]:
+        """Subclasses rewrite HF key names to native module names
+        (ref convert_hf_to_neuron_state_dict, modeling_llama.py:1208-1276)."""
+        return sd
+
+    # ------------------------------------------------------------------
+    def _is_prefill(self, position_ids: torch.Tensor) -> bool:
+        # ref model_base.py:3546
+        return bool(position_ids.min() == 0) and position_ids.shape[-1] > 1
+
+    def infer_attention_mask(self, position_ids: torch.Tensor) -> torch.Tensor:
+        """Rebuild a padding mask from position ids (ref :3485-3504)."""
+        B, S = position_ids.shape
+        if S == 1:
+            seq_len = int(position_ids.max().item()) + 1
+            ar = torch.arange(seq_len, device=position_ids.device).view(1, -1)
+            return (ar <= position_ids).to(torch.int32)
+        return (position_ids >= 0).to(torch.int32) | (position_ids == 0).to(torch.int32)
+
+    def forward(
+        self,
+        input_ids: torch.Tensor,
+        attention_mask: Optional[torch.Tensor] = None,
+        position_ids: Optional[torch.Tensor] = None,
+        seq_ids: Optional[torch.Tensor] = None,
+        sampling_params: Optional[torch.Tensor] = None,
+    ) -> ModelOutput:
+        B, S = input_ids.shape
+        if position_ids is None:
+            if attention_mask is not None:
+                position_ids = (attention_mask.long().cumsum(-1) - 1).clamp(min=0)
+                position_ids = position_ids[:, -S:]
+            else:
+                position_ids = torch.arange(S, device=input_ids.device).view(1, S).expand(B, S)
+        if sampling_params is None and self.ec.on_device_sampling_config:
+            c = self.ec.on_device_sampling_config
+            sampling_params = prepare_sampling_params(
+                B, c.top_k if c.do_sample else 1, c.top_p, c.temperature,
+                device=input_ids.device,
+            )
+        tokens, logits = self.model(
+            input_ids, attention_mask, position_ids, seq_ids, sampling_params
+        )
+        self.kv_cache_populated = True
+        return ModelOutput(tokens=tokens, logits=logits)
+
+    def reset(self):
+        """Clear KV state (ref model_base.py:3926)."""
+        self.kv_cache_populated = False
+        self.model.reset_cache()

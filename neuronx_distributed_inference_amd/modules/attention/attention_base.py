@@ -1,0 +1,144 @@
+"""The attention module every model subclasses.
+
+MI355X-native analog of the reference's `NeuronAttentionBase`
+(`modules/attention/attention_base.py:134-2488`): one nn.Module serving both
+phases — context encoding (flash prefill kernel, K1-analog) and token
+generation (decode kernel against the KV cache, K2-analog) — selected by the
+active-token count. KV-cache writes go through the in-place HIP scatter
+before attention, so the captured decode graph reads what it just wrote with
+no aliasing machinery.
+
+Design difference vs the reference: no trace boundary means no dual
+CTE/TKG projection sets unless CP != DP meshes demand it; the same
+Column/Row-parallel projections serve both phases.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+from neuronx_distributed_inference_amd import ops
+from neuronx_distributed_inference_amd.modules.attention import utils as attn_utils
+from neuronx_distributed_inference_amd.modules.attention.gqa import (
+    GroupQueryAttentionO,
+    GroupQueryAttentionQKV,
+)
+from neuronx_distributed_inference_amd.modules.rope import RotaryEmbedding
+from neuronx_distributed_inference_amd.parallel import state as ps
+
+PHASE_CTE = "context_encoding"
+PHASE_TKG = "token_generation"
+
+
+class AttentionBase(nn.Module):
+    def __init__(self, config, layer_idx: int = 0, rotary: Optional[RotaryEmbedding] = None):
+        super().__init__()
+        ec = config.neuron_config
+        self.config = config
+        self.layer_idx = layer_idx
+        self.hidden_size = config.hidden_size
+        self.num_heads = config.num_attention_heads
+        self.num_kv_heads = getattr(config, "num_key_value_heads", self.num_heads)
+        self.head_dim = getattr(config, "head_dim", None) or self.hidden_size // self.num_heads
+        self.tp_degree = ec.tp_degree
+        self.torch_dtype = ec.torch_dtype
+        self.sliding_window = getattr(config, "sliding_window", None) or 0
+        self.attn_bias = bool(getattr(config, "attention_bias", False))
+
+        self.qkv_proj = GroupQueryAttentionQKV(
+            self.hidden_size, self.num_heads, self.num_kv_heads, self.head_dim,
+            self.tp_degree, self.torch_dtype, bias=self.attn_bias,
+            fused=ec.fused_qkv,
+        )
+        self.o_proj = GroupQueryAttentionO(
+            self.hidden_size, self.num_heads, self.head_dim, self.tp_degree,
+            self.torch_dtype, bias=False,
+        )
+        self.num_heads_per_rank = self.qkv_proj.num_heads_per_rank
+        self.num_kv_heads_per_rank = self.qkv_proj.num_kv_heads_per_rank
+        self.scale = 1.0 / math.sqrt(self.head_dim)
+        self.rotary = rotary
+        # optional per-head q/k norms (Qwen3 PRE_ROPE qk-norm); models set these
+        self.q_norm: Optional[nn.Module] = None
+        self.k_norm: Optional[nn.Module] = None
+        self.sinks: Optional[nn.Parameter] = None  # GPT-OSS learned sinks
+
+    # ------------------------------------------------------------------
+    def prep_qkv(
+        self, hidden: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor
+    ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+        """Projections + BSHD->BHSD + optional qk-norm + RoPE
+        (ref prep_qkv_tensors attention_base.py:489-563)."""
+        B, S, _ = hidden.shape
+        q, k, v = self.qkv_proj(hidden)
+        q = q.view(B, S, self.num_heads_per_rank, self.head_dim)
+        k = k.view(B, S, self.num_kv_heads_per_rank, self.head_dim)
+        v = v.view(B, S, self.num_kv_heads_per_rank, self.head_dim)
+        if self.q_norm is not None:
+            q = self.q_norm(q)
+        if self.k_norm is not None:
+            k = self.k_norm(k)
+        q = q.transpose(1, 2)  # BHSD
+        k = k.transpose(1, 2)
+        v = v.transpose(1, 2)
+        if cos is not None:
+            q, k = ops.rope_apply(q, k, cos, sin)
+        return q, k, v
+
+    # ------------------------------------------------------------------
+    def forward(
+        self,
+        hidden: torch.Tensor,
+        cos: torch.Tensor,
+        sin: torch.Tensor,
+        kv_mgr,
+        position_ids: torch.Tensor,
+        phase: str,
+        attention_mask: Optional[torch.Tensor] = None,
+        seq_ids: Optional[torch.Tensor] = None,
+        active_mask: Optional[torch.Tensor] = None,
+    ) -> torch.Tensor:
+        B, S, _ = hidden.shape
+        q, k, v = self.prep_qkv(hidden, cos, sin)
+
+        # write this step's K/V into the cache first (in-place scatter)
+        kv_mgr.update_cache(self.layer_idx, k, v, position_ids, seq_ids)
+
+        if phase == PHASE_CTE:
+            out = self._prefill(q, k, v, attention_mask)
+        else:
+            out = self._decode(q, kv_mgr, position_ids, seq_ids, active_mask)
+
+        out = out.transpose(1, 2).reshape(B, S, -1)  # BHSD -> BS(H*D)
+        return self.o_proj(out)
+
+    # ------------------------------------------------------------------
+    def _prefill(self, q, k, v, attention_mask) -> torch.Tensor:
+        """Causal flash attention over the fresh K/V (the cache holds the
+        same values; using the projections directly avoids a re-read)."""
+        mask = None
+        if attention_mask is not None and not bool(attention_mask.all()):
+            mask = attn_utils.padded_causal_mask(attention_mask, q.shape[2])
+            if self.sliding_window:
+                mask = mask & attn_utils.sliding_window_mask(
+                    q.shape[2], k.shape[2], self.sliding_window, device=q.device
+                )
+            return ops.attention_prefill(
+                q, k, v, mask=mask, scale=self.scale, sinks=self.sinks, causal=False
+            )
+        return ops.attention_prefill(
+            q, k, v, scale=self.scale, sinks=self.sinks, causal=True,
+            sliding_window=self.sliding_window,
+        )
+
+    def _decode(self, q, kv_mgr, position_ids, seq_ids, active_mask) -> torch.Tensor:
+        k_cache, v_cache = kv_mgr.get_cache(self.layer_idx)
+        seq_lens = position_ids[:, -1] + 1  # attend to everything written so far
+        return ops.attention_decode(
+            q, k_cache, v_cache, seq_lens, seq_ids=seq_ids, scale=self.scale,
+            sinks=self.sinks, sliding_window=self.sliding_window,
+        )

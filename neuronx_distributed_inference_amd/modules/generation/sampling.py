@@ -1,0 +1,113 @@
+"""On-device sampling.
+
+Analog of the reference's `modules/generation/sampling.py` (Sampler at :241,
+greedy :372-388, multinomial :285-370, per-request params :97-198): sampling
+runs inside the captured graph on vocab-sharded logits — greedy via
+distributed argmax, multinomial via distributed top-`global_topk` then the
+fused HIP top-k/top-p/temperature/inverse-CDF kernel.
+
+Per-request `sampling_params` is a (B, 3) tensor [top_k, top_p, temperature]
+exactly like the reference so serving stacks can drive it per sequence.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+from neuronx_distributed_inference_amd import ops
+from neuronx_distributed_inference_amd.config import OnDeviceSamplingConfig
+from neuronx_distributed_inference_amd.parallel import ops as pops
+from neuronx_distributed_inference_amd.parallel import state as ps
+
+
+def prepare_sampling_params(
+    batch_size: int,
+    top_k=1,
+    top_p: float = 1.0,
+    temperature: float = 1.0,
+    device=None,
+) -> torch.Tensor:
+    """Build the (B,3) sampling-params tensor (ref sampling.py:183-198)."""
+
+    def _t(v, dtype):
+        t = torch.as_tensor(v, dtype=dtype, device=device)
+        if t.dim() == 0:
+            t = t.expand(batch_size).clone()
+        if t.numel() != batch_size:
+            raise ValueError(f"sampling param wrong batch size: {t.numel()} != {batch_size}")
+        return t.float()
+
+    return torch.stack(
+        [_t(top_k, torch.float32), _t(top_p, torch.float32), _t(temperature, torch.float32)],
+        dim=1,
+    )
+
+
+def validate_sampling_params(params: torch.Tensor, config: OnDeviceSamplingConfig):
+    if params.dim() != 2 or params.shape[1] != 3:
+        raise ValueError(f"sampling_params must be (B,3), got {tuple(params.shape)}")
+    if (params[:, 0] > config.global_topk).any():
+        raise ValueError(f"top_k exceeds global_topk={config.global_topk}")
+    if ((params[:, 1] <= 0) | (params[:, 1] > 1.0)).any():
+        raise ValueError("top_p must be in (0, 1]")
+
+
+class Sampler(nn.Module):
+    """Samples token ids from vocab-sharded logits on device."""
+
+    def __init__(self, config: OnDeviceSamplingConfig, vocab_size: int,
+                 deterministic: Optional[bool] = None):
+        super().__init__()
+        self.config = config
+        self.vocab_size = vocab_size
+        self.deterministic = config.deterministic if deterministic is None else deterministic
+        self._gen: Optional[torch.Generator] = None
+
+    def _rand(self, batch: int, device) -> torch.Tensor:
+        if self.deterministic:
+            if self._gen is None or self._gen.device != torch.device(device):
+                self._gen = torch.Generator(device=device)
+                self._gen.manual_seed(0)
+            return torch.rand(batch, device=device, generator=self._gen)
+        return torch.rand(batch, device=device)
+
+    def forward(
+        self, logits: torch.Tensor, sampling_params: Optional[torch.Tensor] = None
+    ) -> torch.Tensor:
+        """logits: (B, S, V/rank) or (B, V/rank) — samples from the last
+        position; returns (B,) token ids (global vocab indices)."""
+        if logits.dim() == 3:
+            logits = logits[:, -1, :]
+        B = logits.shape[0]
+        if sampling_params is None:
+            sampling_params = prepare_sampling_params(
+                B,
+                self.config.top_k if self.config.do_sample else 1,
+                self.config.top_p,
+                self.config.temperature,
+                device=logits.device,
+            )
+        top_k = sampling_params[:, 0]
+        greedy_all = bool((top_k == 1).all())
+        if greedy_all:
+            return pops.distributed_argmax(logits, dim=-1)
+
+        # multinomial: distributed top-global_topk then fused sampling
+        k = min(self.config.global_topk, self.vocab_size)
+        vals, idxs = pops.distributed_topk(logits, k, dim=-1)
+        rand = self._rand(B, logits.device)
+        # `vals` is sorted descending, so the fused kernel's internal top-k is
+        # the identity and its output indexes the candidate list; map back to
+        # global vocab ids through `idxs`.
+        local = ops.topk_topp_sample(
+            vals.float(), top_k, sampling_params[:, 1], sampling_params[:, 2],
+            rand, global_topk=k,
+        )
+        return idxs.gather(1, local.view(-1, 1)).squeeze(1)
+
+
+def create_sampler(config: OnDeviceSamplingConfig, vocab_size: int) -> Sampler:
+    return Sampler(config, vocab_size)

@@ -1,0 +1,53 @@
+"""Gated MLP (SwiGLU / GeGLU) with tensor-parallel projections.
+
+Analog of the reference's kernel-fused MLP path (K5,
+`models/llama/modeling_llama.py:454-737`): gate/up are column-parallel,
+down is row-parallel (all-reduce C1); the activation x multiply is the
+fused HIP `silu_mul` kernel so the intermediate never round-trips HBM twice.
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+
+from neuronx_distributed_inference_amd import ops
+from neuronx_distributed_inference_amd.parallel.layers import (
+    ColumnParallelLinear,
+    RowParallelLinear,
+)
+
+
+class GatedMLP(nn.Module):
+    def __init__(
+        self,
+        hidden_size: int,
+        intermediate_size: int,
+        tp_degree: int,
+        dtype: torch.dtype,
+        hidden_act: str = "silu",
+        bias: bool = False,
+        group=None,
+    ):
+        super().__init__()
+        self.gate_proj = ColumnParallelLinear(
+            hidden_size, intermediate_size, bias=bias, dtype=dtype,
+            tp_degree=tp_degree, group=group,
+        )
+        self.up_proj = ColumnParallelLinear(
+            hidden_size, intermediate_size, bias=bias, dtype=dtype,
+            tp_degree=tp_degree, group=group,
+        )
+        self.down_proj = RowParallelLinear(
+            intermediate_size, hidden_size, bias=bias, dtype=dtype,
+            tp_degree=tp_degree, group=group,
+        )
+        if hidden_act in ("silu", "swish"):
+            self.act_mul = ops.silu_mul
+        elif hidden_act in ("gelu", "gelu_pytorch_tanh", "gelu_new"):
+            self.act_mul = ops.gelu_tanh_mul
+        else:
+            raise ValueError(f"unsupported hidden_act {hidden_act}")
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.down_proj(self.act_mul(self.gate_proj(x), self.up_proj(x)))

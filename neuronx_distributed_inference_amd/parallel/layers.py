@@ -1,0 +1,194 @@
+"""Tensor-parallel layers.
+
+MI355X-native analogs of NxD's ColumnParallelLinear / RowParallelLinear /
+ParallelEmbedding (used throughout the reference, e.g.
+`models/llama/modeling_llama.py:354-387,1112-1133`). GEMMs go through
+torch F.linear (rocBLAS/hipBLASLt on ROCm); the collective seams are
+`parallel.mappings` RCCL calls.
+
+Sharding contract: every parallel parameter carries ``partition_dim`` (0 for
+column-parallel output shards, 1 for row-parallel input shards) and
+``partition_stride``; `utils.checkpoint.shard_state_dict` slices the full
+checkpoint per rank from this metadata — the analog of the reference's
+preshard hooks + `weights/tp{rank}_sharded_checkpoint.safetensors` layout
+(`models/application_base.py:389-399`).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from neuronx_distributed_inference_amd.parallel import mappings, state as ps
+
+
+def mark_parallel(param: torch.Tensor, partition_dim: int, stride: int = 1, num_partitions: Optional[int] = None):
+    param.tensor_model_parallel = True
+    param.partition_dim = partition_dim
+    param.partition_stride = stride
+    if num_partitions is not None:
+        param.num_partitions = num_partitions
+    return param
+
+
+def divide(a: int, b: int) -> int:
+    if a % b != 0:
+        raise ValueError(f"{a} not divisible by tp degree {b}")
+    return a // b
+
+
+class ColumnParallelLinear(nn.Module):
+    """Y = X W^T with W sharded along the output dim across the TP group."""
+
+    def __init__(
+        self,
+        input_size: int,
+        output_size: int,
+        bias: bool = False,
+        gather_output: bool = False,
+        dtype: torch.dtype = torch.float32,
+        tp_degree: Optional[int] = None,
+        group=None,
+        stride: int = 1,
+    ):
+        super().__init__()
+        self.input_size = input_size
+        self.output_size = output_size
+        self.gather_output = gather_output
+        self._group = group
+        self.tp_degree = tp_degree or (
+            ps.get_tensor_model_parallel_size() if group is None else 1
+        )
+        if group is not None:
+            import torch.distributed as dist
+
+            self.tp_degree = dist.get_world_size(group=group) if dist.is_initialized() else 1
+        self.output_size_per_partition = divide(output_size, self.tp_degree)
+        self.weight = nn.Parameter(
+            torch.empty(self.output_size_per_partition, input_size, dtype=dtype)
+        )
+        mark_parallel(self.weight, partition_dim=0, stride=stride, num_partitions=self.tp_degree)
+        if bias:
+            self.bias = nn.Parameter(torch.empty(self.output_size_per_partition, dtype=dtype))
+            mark_parallel(self.bias, partition_dim=0, stride=stride, num_partitions=self.tp_degree)
+        else:
+            self.register_parameter("bias", None)
+
+    @property
+    def group(self):
+        return self._group if self._group is not None else ps.get_tensor_model_parallel_group()
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        y = F.linear(x, self.weight, self.bias)
+        if self.gather_output:
+            y = mappings.gather_from_tensor_parallel_region(y, dim=-1, group=self.group)
+        return y
+
+
+class RowParallelLinear(nn.Module):
+    """Y = X W^T with W sharded along the input dim; output all-reduced
+    (collective C1 — the decode-latency-critical all-reduce)."""
+
+    def __init__(
+        self,
+        input_size: int,
+        output_size: int,
+        bias: bool = False,
+        input_is_parallel: bool = True,
+        dtype: torch.dtype = torch.float32,
+        tp_degree: Optional[int] = None,
+        group=None,
+        reduce_output: bool = True,
+        reduce_dtype: Optional[torch.dtype] = None,
+    ):
+        super().__init__()
+        self.input_size = input_size
+        self.output_size = output_size
+        self.input_is_parallel = input_is_parallel
+        self.reduce_output = reduce_output
+        self.reduce_dtype = reduce_dtype
+        self._group = group
+        self.tp_degree = tp_degree or (
+            ps.get_tensor_model_parallel_size() if group is None else 1
+        )
+        if group is not None:
+            import torch.distributed as dist
+
+            self.tp_degree = dist.get_world_size(group=group) if dist.is_initialized() else 1
+        self.input_size_per_partition = divide(input_size, self.tp_degree)
+        self.weight = nn.Parameter(
+            torch.empty(output_size, self.input_size_per_partition, dtype=dtype)
+        )
+        mark_parallel(self.weight, partition_dim=1, num_partitions=self.tp_degree)
+        if bias:
+            # bias is NOT sharded; applied once after the reduce
+            self.bias = nn.Parameter(torch.empty(output_size, dtype=dtype))
+        else:
+            self.register_parameter("bias", None)
+
+    @property
+    def group(self):
+        return self._group if self._group is not None else ps.get_tensor_model_parallel_group()
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if not self.input_is_parallel:
+            x = mappings.scatter_to_tensor_parallel_region(x, dim=-1, group=self.group)
+        y = F.linear(x, self.weight)
+        if self.reduce_output:
+            y = mappings.reduce_from_tensor_parallel_region(y, group=self.group)
+        if self.bias is not None:
+            y = y + self.bias
+        return y
+
+
+class ParallelEmbedding(nn.Module):
+    """Vocab-sharded embedding: each rank owns rows
+    [rank*V/tp, (rank+1)*V/tp); out-of-shard ids hit a zero row and the
+    partial sums are all-reduced (ref `modeling_llama.py:1112-1133`)."""
+
+    def __init__(
+        self,
+        num_embeddings: int,
+        embedding_dim: int,
+        dtype: torch.dtype = torch.float32,
+        tp_degree: Optional[int] = None,
+        shard_along_embedding: bool = False,
+        pad: bool = True,
+    ):
+        super().__init__()
+        self.num_embeddings = num_embeddings
+        self.embedding_dim = embedding_dim
+        self.tp_degree = tp_degree or ps.get_tensor_model_parallel_size()
+        self.shard_along_embedding = shard_along_embedding
+        if shard_along_embedding:
+            per = divide(embedding_dim, self.tp_degree)
+            self.weight = nn.Parameter(torch.empty(num_embeddings, per, dtype=dtype))
+            mark_parallel(self.weight, partition_dim=1, num_partitions=self.tp_degree)
+            self.vocab_start = 0
+            self.vocab_size_per_partition = num_embeddings
+        else:
+            self.vocab_size_per_partition = divide(num_embeddings, self.tp_degree)
+            self.weight = nn.Parameter(
+                torch.empty(self.vocab_size_per_partition, embedding_dim, dtype=dtype)
+            )
+            mark_parallel(self.weight, partition_dim=0, num_partitions=self.tp_degree)
+            self.vocab_start = None  # resolved at forward from the live rank
+
+    def forward(self, ids: torch.Tensor) -> torch.Tensor:
+        if self.shard_along_embedding:
+            y = F.embedding(ids, self.weight)
+            return mappings.gather_from_tensor_parallel_region(y, dim=-1)
+        tp = ps.get_tensor_model_parallel_size()
+        if tp == 1:
+            return F.embedding(ids, self.weight)
+        rank = ps.get_tensor_model_parallel_rank()
+        start = rank * self.vocab_size_per_partition
+        in_shard = (ids >= start) & (ids < start + self.vocab_size_per_partition)
+        local = torch.where(in_shard, ids - start, torch.zeros_like(ids))
+        y = F.embedding(local, self.weight)
+        y = y * in_shard.unsqueeze(-1).to(y.dtype)
+        return mappings.reduce_from_tensor_parallel_region(y)

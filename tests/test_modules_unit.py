@@ -1,0 +1,193 @@
+"""Tier-1 unit tests: ops reference implementations, bucketing, masks,
+GQA shard math, KV managers, sampling (CPU, no device)."""
+
+import math
+
+import pytest
+import torch
+
+from neuronx_distributed_inference_amd.modules import autobucketing as ab
+from neuronx_distributed_inference_amd.modules.attention import utils as attn_utils
+from neuronx_distributed_inference_amd.modules.attention.gqa import (
+    GQA,
+    get_shardable_head_counts,
+)
+from neuronx_distributed_inference_amd.ops import ref
+
+
+# ---------------- autobucketing -----------------
+def test_generate_buckets_powers_of_two():
+    assert ab.generate_buckets(128, 1024) == [128, 256, 512, 1024]
+    assert ab.generate_buckets(128, 100) == [100]
+    assert ab.generate_buckets(128, 1000) == [128, 256, 512, 1000]
+
+
+def test_get_target_bucket():
+    assert ab.get_target_bucket(5, [128, 256]) == 128
+    assert ab.get_target_bucket(129, [128, 256]) == 256
+    with pytest.raises(ValueError):
+        ab.get_target_bucket(300, [128, 256])
+
+
+# ---------------- masks -----------------
+def test_causal_mask():
+    m = attn_utils.causal_mask(4, 4)[0, 0]
+    assert m.equal(torch.ones(4, 4, dtype=torch.bool).tril())
+
+
+def test_causal_mask_kv_longer():
+    m = attn_utils.causal_mask(2, 6)[0, 0]
+    # queries are positions 4,5
+    assert m[0].tolist() == [True] * 5 + [False]
+    assert m[1].tolist() == [True] * 6
+
+
+def test_sliding_window_mask():
+    m = attn_utils.sliding_window_mask(5, 5, 2)[0, 0]
+    for i in range(5):
+        for j in range(5):
+            assert m[i, j] == (j <= i and j > i - 2)
+
+
+def test_speculation_mask():
+    m = attn_utils.speculation_mask(1, 3, 4)[0, 0]
+    assert m.shape == (3, 7)
+    assert m[:, :4].all()
+    assert m[:, 4:].equal(torch.ones(3, 3, dtype=torch.bool).tril())
+
+
+# ---------------- GQA shard math -----------------
+def test_gqa_shard_strategies():
+    assert get_shardable_head_counts(4, 32, 8) == (32, 8, GQA.SHARD_OVER_TP)
+    assert get_shardable_head_counts(8, 32, 32) == (32, 32, GQA.MHA)
+    heads, kv, strat = get_shardable_head_counts(16, 32, 8)
+    assert (heads, kv, strat) == (32, 16, GQA.REPLICATE_TO_TP_DEGREE)
+    # non-divisible q heads get padded
+    heads, _, _ = get_shardable_head_counts(8, 30, 30)
+    assert heads == 32
+
+
+# ---------------- ops reference numerics -----------------
+def test_rms_norm_matches_manual():
+    x = torch.randn(2, 5, 16)
+    w = torch.randn(16)
+    out, _ = ref.rms_norm(x, w, 1e-6)
+    expect = x * torch.rsqrt(x.pow(2).mean(-1, keepdim=True) + 1e-6) * w
+    torch.testing.assert_close(out, expect, atol=1e-5, rtol=1e-5)
+
+
+def test_rms_norm_fused_residual():
+    x = torch.randn(2, 5, 16)
+    r = torch.randn(2, 5, 16)
+    w = torch.ones(16)
+    out, new_r = ref.rms_norm(x, w, 1e-6, residual=r)
+    torch.testing.assert_close(new_r, x + r)
+
+
+def test_rope_matches_hf():
+    from transformers.models.llama.modeling_llama import apply_rotary_pos_emb
+
+    B, H, S, D = 2, 4, 6, 16
+    q = torch.randn(B, H, S, D)
+    k = torch.randn(B, 2, S, D)
+    inv = 1.0 / (10000.0 ** (torch.arange(0, D, 2).float() / D))
+    pos = torch.arange(S).float()
+    freqs = torch.outer(pos, inv)
+    emb = torch.cat([freqs, freqs], dim=-1)
+    cos, sin = emb.cos(), emb.sin()
+    q2, k2 = ref.rope_apply(q, k, cos, sin)
+    qh, kh = apply_rotary_pos_emb(q, k, cos.unsqueeze(0), sin.unsqueeze(0))
+    torch.testing.assert_close(q2, qh, atol=1e-5, rtol=1e-5)
+    torch.testing.assert_close(k2, kh, atol=1e-5, rtol=1e-5)
+
+
+def test_sdpa_gqa_equivalence():
+    B, Hq, Hkv, S, D = 1, 4, 2, 8, 16
+    q = torch.randn(B, Hq, S, D)
+    k = torch.randn(B, Hkv, S, D)
+    v = torch.randn(B, Hkv, S, D)
+    mask = attn_utils.causal_mask(S, S)
+    out = ref.sdpa(q, k, v, mask=mask)
+    expect = torch.nn.functional.scaled_dot_product_attention(
+        q, k.repeat_interleave(2, 1), v.repeat_interleave(2, 1), is_causal=True
+    )
+    torch.testing.assert_close(out, expect, atol=1e-5, rtol=1e-5)
+
+
+def test_kv_cache_update_with_seq_ids():
+    kc = torch.zeros(4, 2, 10, 8)
+    vc = torch.zeros(4, 2, 10, 8)
+    k = torch.randn(2, 2, 3, 8)
+    v = torch.randn(2, 2, 3, 8)
+    pos = torch.tensor([[0, 1, 2], [4, 5, 6]])
+    sid = torch.tensor([3, 1])
+    ref.kv_cache_update(kc, vc, k, v, pos, sid)
+    torch.testing.assert_close(kc[3, :, 0:3], k[0])
+    torch.testing.assert_close(kc[1, :, 4:7], k[1])
+    assert kc[0].abs().sum() == 0 and kc[2].abs().sum() == 0
+
+
+def test_block_kv_cache_update_and_gather():
+    nb, bs, H, D = 8, 4, 2, 8
+    kc = torch.zeros(nb, bs, H, D)
+    vc = torch.zeros(nb, bs, H, D)
+    n = 6
+    k = torch.randn(n, H, D)
+    v = torch.randn(n, H, D)
+    # sequence occupies blocks 5 then 2
+    slots = torch.tensor([5 * bs + 0, 5 * bs + 1, 5 * bs + 2, 5 * bs + 3, 2 * bs + 0, 2 * bs + 1])
+    ref.block_kv_cache_update(kc, vc, k, v, slots)
+    table = torch.tensor([[5, 2]])
+    kg, vg = ref.paged_gather_kv(kc, vc, table, 6)
+    torch.testing.assert_close(kg[0].transpose(0, 1), k)
+
+
+def test_topk_topp_sample_greedy_and_topp():
+    torch.manual_seed(0)
+    logits = torch.randn(3, 50)
+    out = ref.topk_topp_sample(
+        logits,
+        top_k=torch.tensor([1, 1, 1]),
+        top_p=torch.ones(3),
+        temperature=torch.ones(3),
+        rand=torch.zeros(3),
+    )
+    assert torch.equal(out, logits.argmax(-1))
+    # top_k=5, rand=0 picks the argmax too (first CDF entry)
+    out2 = ref.topk_topp_sample(
+        logits, torch.tensor([5, 5, 5]), torch.full((3,), 0.9), torch.ones(3), torch.zeros(3)
+    )
+    assert torch.equal(out2, logits.argmax(-1))
+    # rand near 1 with top_k=2 picks the 2nd-best
+    out3 = ref.topk_topp_sample(
+        logits, torch.tensor([2, 2, 2]), torch.ones(3), torch.ones(3), torch.full((3,), 0.999)
+    )
+    second = logits.topk(2, -1).indices[:, 1]
+    assert torch.equal(out3, second)
+
+
+# ---------------- sampler module -----------------
+def test_sampler_greedy():
+    from neuronx_distributed_inference_amd.config import OnDeviceSamplingConfig
+    from neuronx_distributed_inference_amd.modules.generation.sampling import Sampler
+
+    s = Sampler(OnDeviceSamplingConfig(top_k=1), vocab_size=64)
+    logits = torch.randn(4, 1, 64)
+    out = s(logits)
+    assert torch.equal(out, logits[:, -1].argmax(-1))
+
+
+def test_sampler_multinomial_topk():
+    from neuronx_distributed_inference_amd.config import OnDeviceSamplingConfig
+    from neuronx_distributed_inference_amd.modules.generation.sampling import (
+        Sampler,
+        prepare_sampling_params,
+    )
+
+    torch.manual_seed(0)
+    s = Sampler(OnDeviceSamplingConfig(top_k=4, do_sample=True, deterministic=True), vocab_size=64)
+    logits = torch.randn(4, 64)
+    params = prepare_sampling_params(4, top_k=4, top_p=0.9, temperature=0.7)
+    out = s(logits, params)
+    top4 = logits.topk(4, -1).indices
+    assert all(out[i] in top4[i] for i in range(4))
