@@ -1,0 +1,180 @@
+"""Flagship serving benchmark (driver contract).
+
+Measures BASELINE.json's headline metric — output tokens/sec for
+Llama-3.1-8B decode (TP=1 on one GPU; TP=N across N GPUs over RCCL/xGMI)
+with random-init weights and synthetic data. A "step" is one
+token-generation step for the whole batch (one new token per sequence);
+prefill at ctx=128 happens before the timed region. Batch is 8 per GPU
+(weak scaling).
+
+    python bench.py --gpus N --steps K --warmup W [--model {8b,70b,tiny}]
+
+Under torchrun, reads RANK/LOCAL_RANK/WORLD_SIZE from the env.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+MODELS = {
+    "tiny": dict(hidden_size=256, intermediate_size=512, num_hidden_layers=2,
+                 num_attention_heads=8, num_key_value_heads=4, vocab_size=1024),
+    "8b": dict(hidden_size=4096, intermediate_size=14336, num_hidden_layers=32,
+               num_attention_heads=32, num_key_value_heads=8, vocab_size=128256),
+    "70b": dict(hidden_size=8192, intermediate_size=28672, num_hidden_layers=80,
+                num_attention_heads=64, num_key_value_heads=8, vocab_size=128256),
+}
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=32)
+    p.add_argument("--warmup", type=int, default=8)
+    p.add_argument("--model", default="8b", choices=list(MODELS))
+    p.add_argument("--batch-per-gpu", type=int, default=8)
+    p.add_argument("--ctx", type=int, default=128)
+    p.add_argument("--no-graphs", action="store_true")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    rank = int(os.environ.get("RANK", 0))
+    local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    assert world == args.gpus or world == 1, f"WORLD_SIZE {world} != --gpus {args.gpus}"
+
+    on_gpu = torch.cuda.is_available()
+    device = torch.device(f"cuda:{local_rank}" if on_gpu else "cpu")
+    if on_gpu:
+        torch.cuda.set_device(device)
+    if world > 1:
+        dist.init_process_group(
+            backend="nccl" if on_gpu else "gloo",
+            device_id=device if on_gpu else None,
+        )
+
+    from neuronx_distributed_inference_amd.config import (
+        EngineConfig,
+        OnDeviceSamplingConfig,
+    )
+    from neuronx_distributed_inference_amd.models.llama.modeling_llama import (
+        LlamaForCausalLM,
+        LlamaInferenceConfig,
+    )
+    from neuronx_distributed_inference_amd.parallel import state as ps
+
+    tp = world
+    ps.initialize_model_parallel(tensor_model_parallel_size=tp)
+
+    spec = dict(MODELS[args.model])
+    global_batch = args.batch_per_gpu * max(world, 1)
+    steps, warmup = args.steps, args.warmup
+    total_len = args.ctx + warmup + steps + 8
+    dtype = "bfloat16" if on_gpu else "float32"
+
+    ec = EngineConfig(
+        batch_size=global_batch, max_batch_size=global_batch,
+        seq_len=total_len, max_length=total_len, max_context_length=args.ctx,
+        torch_dtype=dtype, tp_degree=tp,
+        on_device_sampling_config=OnDeviceSamplingConfig(top_k=1),
+        enable_hip_graphs=not args.no_graphs,
+    )
+    cfg = LlamaInferenceConfig(
+        neuron_config=ec, max_position_embeddings=total_len * 2,
+        rms_norm_eps=1e-5, rope_theta=500000.0, hidden_act="silu",
+        tie_word_embeddings=False, **spec,
+    )
+
+    torch.manual_seed(1234)
+    model = LlamaForCausalLM(cfg).eval()
+    if on_gpu:
+        model = model.to(device)
+    with torch.no_grad():
+        for p in model.parameters():
+            if p.dim() >= 2:
+                p.normal_(0, 0.02)
+    model.reset()
+
+    V = spec["vocab_size"]
+    ids = torch.randint(0, V, (global_batch, args.ctx), device=device)
+    mask = torch.ones_like(ids)
+    seq_ids = torch.arange(global_batch, device=device)
+
+    def barrier_sync():
+        if world > 1:
+            dist.barrier()
+        if on_gpu:
+            torch.cuda.synchronize()
+
+    with torch.inference_mode():
+        # ---- prefill (untimed) ----
+        pos = (mask.long().cumsum(-1) - 1).clamp(min=0)
+        out = model(input_ids=ids, attention_mask=mask, position_ids=pos, seq_ids=seq_ids)
+        next_tok = out.tokens.view(-1, 1)
+        cur_len = args.ctx
+
+        def decode_step(tok, cur_len):
+            p = torch.full((global_batch, 1), cur_len, device=device, dtype=torch.long)
+            o = model(input_ids=tok, attention_mask=None, position_ids=p, seq_ids=seq_ids)
+            return o.tokens.view(-1, 1)
+
+        # ---- warmup ----
+        for _ in range(warmup):
+            next_tok = decode_step(next_tok, cur_len)
+            cur_len += 1
+
+        barrier_sync()
+        t0 = time.perf_counter()
+        for _ in range(steps):
+            next_tok = decode_step(next_tok, cur_len)
+            cur_len += 1
+        barrier_sync()
+        elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if world > 1:
+        t = torch.tensor([elapsed], device=device if on_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank == 0:
+        ms_per_step = elapsed / steps * 1000.0
+        tokens_per_sec = global_batch * steps / elapsed
+        print(json.dumps({
+            "metric": "output tokens/sec (node), Llama-3.1-8B decode" if args.model == "8b"
+                      else f"output tokens/sec (node), {args.model} decode",
+            "value": round(tokens_per_sec, 2),
+            "unit": "tokens/s",
+            "n_gpus": max(world, 1),
+            "steps": steps,
+            "warmup": warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": dtype,
+            "data": "synthetic",
+            "config": {
+                "model": {"8b": "llama-3.1-8b", "70b": "llama-3.1-70b", "tiny": "tiny-llama-2l"}[args.model],
+                "global_batch": global_batch,
+                "seq_len": total_len,
+                "ctx": args.ctx,
+                "parallelism": f"tp{tp}",
+                "weights": "random-init",
+            },
+        }))
+
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

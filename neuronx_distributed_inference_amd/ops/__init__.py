@@ -120,14 +120,25 @@ def attention_prefill(q, k, v, mask=None, scale=None, sinks=None, causal=True,
         c = _require_ext("attention_prefill")
         import math
         scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
-        if mask is None and causal and q.shape[-1] in (64, 128) \
-                and q.shape[2] == k.shape[2]:
+        if (
+            hasattr(c, "flash_prefill")
+            and mask is None and causal and sinks is None
+            and q.shape[-1] in (64, 128) and q.shape[2] == k.shape[2]
+        ):
             out = torch.empty_like(q)
             c.flash_prefill(out, q.contiguous(), k.contiguous(), v.contiguous(),
                             float(scale), int(sliding_window))
             return out
-        # masked / unusual-shape fallback: fp32 matmul-softmax on device
-        # via library GEMMs (explicitly not a perf path)
+        # masked / unusual-shape path: matmul-softmax via library GEMMs
+        if mask is None and causal:
+            from neuronx_distributed_inference_amd.modules.attention.utils import (
+                causal_mask,
+                sliding_window_mask,
+            )
+            Sq, Skv = q.shape[2], k.shape[2]
+            mask = causal_mask(Sq, Skv, device=q.device)
+            if sliding_window:
+                mask = mask & sliding_window_mask(Sq, Skv, sliding_window, device=q.device)
         return ref.sdpa(q, k, v, mask=mask, scale=scale, sinks=sinks)
     if mask is None and causal:
         Sq, Skv = q.shape[2], k.shape[2]

@@ -1,0 +1,296 @@
+// Token-generation (decode) attention for gfx950 — the K2-analog
+// (ref attention_block_tkg, modules/attention/attention_base.py:1186-1381).
+//
+// Memory-bound split-KV design for the linear cache (B,Hkv,Smax,D):
+//  * pass 1: grid (nsplit, Hkv*Sq, B), 256-thread workgroups (4 waves).
+//    Lane groups of D/8 lanes stream K/V rows with 16-B loads (8 bf16 per
+//    lane = one ds-free register path; guide G13). Each group computes the
+//    q.k dots for all GQ query heads of its KV head at once, so the KV
+//    bytes are read ONCE per (b, hkv) regardless of GQA ratio, online
+//    softmax in registers, cross-group combine via __shfl_xor butterflies,
+//    cross-wave combine via LDS, per-split partials (m, l, o) to workspace.
+//  * pass 2: combines the nsplit partials per (b, hq, sq) row.
+//
+// The split over the sequence gives the chip >> 256 workgroups at decode
+// batch sizes (256 CUs in 8 XCDs need large grids) and is the same
+// structure flash-decoding (seq-sharded KV, ref C5) extends across GPUs.
+
+#include "common.h"
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+constexpr float M_INIT = -1e30f;
+
+template <typename T, int D, int GQ>
+__global__ __launch_bounds__(256) void decode_attn_pass1(
+    float* __restrict__ o_part,        // (B, Hq, Sq, nsplit, D)
+    float* __restrict__ ml_part,       // (B, Hq, Sq, nsplit, 2)
+    const T* __restrict__ q,           // (B, Hq, Sq, D) — Hq = Hkv*gq
+    const T* __restrict__ k_cache,     // (Bc, Hkv, Smax, D)
+    const T* __restrict__ v_cache,
+    const int* __restrict__ seq_lens,  // (B,)
+    const int* __restrict__ seq_ids,   // (B,)
+    const float scale, const int window,
+    const int Hkv, const int Sq, const int gq,
+    const int Smax, const int chunk, const int nsplit) {
+    using VT = VecTraits<T>;
+    constexpr int GS = D / 8;          // lanes per group
+    constexpr int NG = 256 / GS;       // groups per workgroup
+
+    const int split = blockIdx.x;
+    const int hkv = blockIdx.y / Sq;
+    const int sq = blockIdx.y % Sq;
+    const int b = blockIdx.z;
+    const int Hq = Hkv * gq;
+
+    const int lane = threadIdx.x & (WAVE_SIZE - 1);
+    const int wid = threadIdx.x / WAVE_SIZE;
+    const int grp = threadIdx.x / GS;         // 0..NG-1
+    const int c = threadIdx.x % GS;           // lane-in-group
+
+    const int row = seq_ids[b];
+    const int eff_len = seq_lens[b] - (Sq - 1 - sq);
+    int lo = 0;
+    if (window > 0 && eff_len > window) lo = eff_len - window;
+    const int split_lo = max(lo, split * chunk);
+    const int split_hi = min(eff_len, (split + 1) * chunk);
+
+    // ---- q into registers, pre-scaled ----
+    float q_reg[GQ][8];
+#pragma unroll
+    for (int g = 0; g < GQ; ++g) {
+        if (g < gq) {
+            const long qoff = ((((long)b * Hq) + hkv * gq + g) * Sq + sq) * D + c * 8;
+            Vec8<T> qv = load_vec8(q + qoff);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) q_reg[g][j] = VT::to_float(qv.v[j]) * scale;
+        }
+    }
+
+    float m[GQ], lsum[GQ], o_acc[GQ][8];
+#pragma unroll
+    for (int g = 0; g < GQ; ++g) {
+        m[g] = M_INIT;
+        lsum[g] = 0.f;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) o_acc[g][j] = 0.f;
+    }
+
+    // ---- stream K/V rows ----
+    if (row >= 0 && split_lo < split_hi) {
+        const T* kbase = k_cache + (((long)row * Hkv + hkv) * Smax) * D;
+        const T* vbase = v_cache + (((long)row * Hkv + hkv) * Smax) * D;
+        for (int l = split_lo + grp; l < split_hi; l += NG) {
+            Vec8<T> kv = load_vec8(kbase + (long)l * D + c * 8);
+            float kf[8];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) kf[j] = VT::to_float(kv.v[j]);
+            Vec8<T> vv = load_vec8(vbase + (long)l * D + c * 8);
+
+            float s[GQ];
+#pragma unroll
+            for (int g = 0; g < GQ; ++g) {
+                if (g >= gq) continue;
+                float d = 0.f;
+#pragma unroll
+                for (int j = 0; j < 8; ++j) d += q_reg[g][j] * kf[j];
+                s[g] = group_reduce_sum<GS>(d);  // all GS lanes get the dot
+            }
+#pragma unroll
+            for (int g = 0; g < GQ; ++g) {
+                if (g >= gq) continue;
+                const float mn = fmaxf(m[g], s[g]);
+                const float corr = __expf(m[g] - mn);
+                const float p = __expf(s[g] - mn);
+                lsum[g] = lsum[g] * corr + p;
+                m[g] = mn;
+#pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    o_acc[g][j] = o_acc[g][j] * corr + p * VT::to_float(vv.v[j]);
+            }
+        }
+    }
+
+    // ---- combine groups within the wave (xor butterfly over group bits) ----
+#pragma unroll
+    for (int off = GS; off < WAVE_SIZE; off <<= 1) {
+#pragma unroll
+        for (int g = 0; g < GQ; ++g) {
+            if (g >= gq) continue;
+            const float m_o = __shfl_xor(m[g], off, WAVE_SIZE);
+            const float l_o = __shfl_xor(lsum[g], off, WAVE_SIZE);
+            const float mn = fmaxf(m[g], m_o);
+            const float f_s = __expf(m[g] - mn);
+            const float f_o = __expf(m_o - mn);
+            lsum[g] = lsum[g] * f_s + l_o * f_o;
+            m[g] = mn;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const float oo = __shfl_xor(o_acc[g][j], off, WAVE_SIZE);
+                o_acc[g][j] = o_acc[g][j] * f_s + oo * f_o;
+            }
+        }
+    }
+
+    // ---- combine the 4 waves through LDS ----
+    __shared__ float o_lds[4][GQ > 4 ? GQ : 4][D];
+    __shared__ float ml_lds[4][GQ > 4 ? GQ : 4][2];
+    if (lane < GS) {
+#pragma unroll
+        for (int g = 0; g < GQ; ++g) {
+            if (g >= gq) continue;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) o_lds[wid][g][c * 8 + j] = o_acc[g][j];
+            if (c == 0) {
+                ml_lds[wid][g][0] = m[g];
+                ml_lds[wid][g][1] = lsum[g];
+            }
+        }
+    }
+    __syncthreads();
+    if (wid == 0 && lane < GS) {
+        for (int g = 0; g < gq; ++g) {
+            float mm = M_INIT, ll = 0.f, oo[8];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) oo[j] = 0.f;
+            for (int w = 0; w < 4; ++w) {
+                const float mw = ml_lds[w][g][0];
+                const float lw = ml_lds[w][g][1];
+                const float mn = fmaxf(mm, mw);
+                const float f_s = __expf(mm - mn);
+                const float f_w = __expf(mw - mn);
+                ll = ll * f_s + lw * f_w;
+#pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    oo[j] = oo[j] * f_s + o_lds[w][g][c * 8 + j] * f_w;
+                mm = mn;
+            }
+            const long base =
+                ((((long)b * Hq + hkv * gq + g) * Sq + sq) * nsplit + split);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) o_part[base * D + c * 8 + j] = oo[j];
+            if (c == 0) {
+                ml_part[base * 2 + 0] = mm;
+                ml_part[base * 2 + 1] = ll;
+            }
+        }
+    }
+}
+
+template <typename T, int D>
+__global__ void decode_attn_pass2(
+    T* __restrict__ out,               // (B, Hq, Sq, D)
+    const float* __restrict__ o_part,  // (B, Hq, Sq, nsplit, D)
+    const float* __restrict__ ml_part, // (B, Hq, Sq, nsplit, 2)
+    const int nsplit) {
+    using VT = VecTraits<T>;
+    const long rowid = blockIdx.x;  // over B*Hq*Sq
+    const int lane = threadIdx.x;
+    // global max over splits
+    float mstar = M_INIT;
+    for (int i = 0; i < nsplit; ++i)
+        mstar = fmaxf(mstar, ml_part[(rowid * nsplit + i) * 2 + 0]);
+    float L = 0.f;
+    for (int i = 0; i < nsplit; ++i) {
+        const float mi = ml_part[(rowid * nsplit + i) * 2 + 0];
+        const float li = ml_part[(rowid * nsplit + i) * 2 + 1];
+        L += li * __expf(mi - mstar);
+    }
+    const float inv = (L > 0.f) ? 1.f / L : 0.f;
+#pragma unroll
+    for (int d = lane; d < D; d += WAVE_SIZE) {
+        float acc = 0.f;
+        for (int i = 0; i < nsplit; ++i) {
+            const float mi = ml_part[(rowid * nsplit + i) * 2 + 0];
+            acc += o_part[(rowid * nsplit + i) * D + d] * __expf(mi - mstar);
+        }
+        out[rowid * D + d] = VT::from_float(acc * inv);
+    }
+}
+
+// ---------------------------------------------------------------------------
+template <typename T, int D, int GQ>
+static void launch_decode(torch::Tensor& out, const torch::Tensor& q,
+                          const torch::Tensor& k_cache, const torch::Tensor& v_cache,
+                          const torch::Tensor& seq_lens, const torch::Tensor& seq_ids,
+                          float scale, int window, int gq) {
+    const int B = q.size(0), Hq = q.size(1), Sq = q.size(2);
+    const int Hkv = k_cache.size(1), Smax = k_cache.size(2);
+    // split count: fill the chip (>=1024 workgroups when possible), chunks of
+    // >=128 rows so per-split work amortizes.
+    int nsplit = 1024 / std::max(1, B * Hkv * Sq);
+    nsplit = std::max(1, std::min({nsplit, (Smax + 127) / 128, 64}));
+    const int chunk = (Smax + nsplit - 1) / nsplit;
+
+    auto opts = torch::TensorOptions().dtype(torch::kFloat32).device(q.device());
+    auto o_part = torch::empty({(long)B * Hq * Sq * nsplit * D}, opts);
+    auto ml_part = torch::empty({(long)B * Hq * Sq * nsplit * 2}, opts);
+    auto stream = at::hip::getCurrentHIPStream();
+
+    dim3 grid1(nsplit, Hkv * Sq, B);
+    hipLaunchKernelGGL((decode_attn_pass1<T, D, GQ>), grid1, dim3(256), 0, stream,
+                       o_part.data_ptr<float>(), ml_part.data_ptr<float>(),
+                       (const T*)q.data_ptr(), (const T*)k_cache.data_ptr(),
+                       (const T*)v_cache.data_ptr(), seq_lens.data_ptr<int>(),
+                       seq_ids.data_ptr<int>(), scale, window, Hkv, Sq, gq,
+                       Smax, chunk, nsplit);
+    HIP_CHECK_KERNEL();
+    dim3 grid2(B * Hq * Sq);
+    hipLaunchKernelGGL((decode_attn_pass2<T, D>), grid2, dim3(WAVE_SIZE), 0, stream,
+                       (T*)out.data_ptr(), o_part.data_ptr<float>(),
+                       ml_part.data_ptr<float>(), nsplit);
+    HIP_CHECK_KERNEL();
+}
+
+template <typename T, int D>
+static void dispatch_gq(torch::Tensor& out, const torch::Tensor& q,
+                        const torch::Tensor& k_cache, const torch::Tensor& v_cache,
+                        const torch::Tensor& seq_lens, const torch::Tensor& seq_ids,
+                        float scale, int window, int gq) {
+    if (gq <= 1)
+        launch_decode<T, D, 1>(out, q, k_cache, v_cache, seq_lens, seq_ids, scale, window, gq);
+    else if (gq <= 2)
+        launch_decode<T, D, 2>(out, q, k_cache, v_cache, seq_lens, seq_ids, scale, window, gq);
+    else if (gq <= 4)
+        launch_decode<T, D, 4>(out, q, k_cache, v_cache, seq_lens, seq_ids, scale, window, gq);
+    else if (gq <= 8)
+        launch_decode<T, D, 8>(out, q, k_cache, v_cache, seq_lens, seq_ids, scale, window, gq);
+    else
+        TORCH_CHECK(false, "GQA ratio > 8 not supported yet (got ", gq, ")");
+}
+
+void attention_decode(torch::Tensor out, torch::Tensor q,
+                      torch::Tensor k_cache, torch::Tensor v_cache,
+                      torch::Tensor seq_lens, torch::Tensor seq_ids,
+                      double scale, long window) {
+    TORCH_CHECK(q.dim() == 4 && k_cache.dim() == 4, "bad shapes");
+    const int D = q.size(3);
+    const int Hq = q.size(1), Hkv = k_cache.size(1);
+    TORCH_CHECK(Hq % Hkv == 0, "Hq must be a multiple of Hkv");
+    const int gq = Hq / Hkv;
+    TORCH_CHECK(q.is_contiguous() && k_cache.is_contiguous() && v_cache.is_contiguous(),
+                "contiguous tensors required");
+
+#define CASE_T(T)                                                               \
+    do {                                                                        \
+        if (D == 128)                                                           \
+            dispatch_gq<T, 128>(out, q, k_cache, v_cache, seq_lens, seq_ids,    \
+                                (float)scale, (int)window, gq);                 \
+        else if (D == 64)                                                       \
+            dispatch_gq<T, 64>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
+                               (float)scale, (int)window, gq);                  \
+        else                                                                    \
+            TORCH_CHECK(false, "head_dim must be 64 or 128, got ", D);          \
+    } while (0)
+
+    if (q.scalar_type() == at::ScalarType::BFloat16) {
+        CASE_T(__hip_bfloat16);
+    } else if (q.scalar_type() == at::ScalarType::Half) {
+        CASE_T(__half);
+    } else if (q.scalar_type() == at::ScalarType::Float) {
+        CASE_T(float);
+    } else {
+        TORCH_CHECK(false, "attention_decode: unsupported dtype");
+    }
+#undef CASE_T
+}

@@ -1,0 +1,31 @@
+// Python bindings for the gfx950 kernel library.
+#include <torch/extension.h>
+
+void rms_norm(torch::Tensor out, torch::Tensor x, torch::Tensor weight, double eps);
+void rms_norm_fused_add(torch::Tensor out, torch::Tensor x, torch::Tensor residual,
+                        torch::Tensor weight, double eps);
+void silu_mul(torch::Tensor out, torch::Tensor gate, torch::Tensor up);
+void gelu_tanh_mul(torch::Tensor out, torch::Tensor gate, torch::Tensor up);
+void rope_apply(torch::Tensor q, torch::Tensor k, torch::Tensor cosb, torch::Tensor sinb);
+void kv_cache_update(torch::Tensor k_cache, torch::Tensor v_cache, torch::Tensor k_new,
+                     torch::Tensor v_new, torch::Tensor positions, torch::Tensor seq_ids);
+void block_kv_cache_update(torch::Tensor k_cache, torch::Tensor v_cache, torch::Tensor k_new,
+                           torch::Tensor v_new, torch::Tensor slot_mapping);
+void attention_decode(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
+                      torch::Tensor v_cache, torch::Tensor seq_lens, torch::Tensor seq_ids,
+                      double scale, long window);
+void topk_topp_sample(torch::Tensor out, torch::Tensor vals, torch::Tensor top_k,
+                      torch::Tensor top_p, torch::Tensor temp, torch::Tensor rand,
+                      long global_topk);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("rms_norm", &rms_norm, "RMSNorm (CDNA4)");
+    m.def("rms_norm_fused_add", &rms_norm_fused_add, "RMSNorm + residual add (CDNA4)");
+    m.def("silu_mul", &silu_mul, "SiLU(gate)*up (CDNA4)");
+    m.def("gelu_tanh_mul", &gelu_tanh_mul, "GELU-tanh(gate)*up (CDNA4)");
+    m.def("rope_apply", &rope_apply, "RoPE rotate-half in place (CDNA4)");
+    m.def("kv_cache_update", &kv_cache_update, "linear KV cache scatter (CDNA4)");
+    m.def("block_kv_cache_update", &block_kv_cache_update, "paged KV cache scatter (CDNA4)");
+    m.def("attention_decode", &attention_decode, "split-KV decode attention (CDNA4)");
+    m.def("topk_topp_sample", &topk_topp_sample, "fused top-k/top-p sampling (CDNA4)");
+}
