@@ -102,6 +102,10 @@ def main():
             if p.dim() >= 2:
                 p.normal_(0, 0.02)
     model.reset()
+    model.enable_submodels()
+    if on_gpu and not args.no_graphs:
+        with torch.inference_mode():
+            model.warmup()  # hipGraph capture for the decode step
 
     V = spec["vocab_size"]
     ids = torch.randint(0, V, (global_batch, args.ctx), device=device)

@@ -114,11 +114,11 @@ class DecoderModel(nn.Module):
         raise NotImplementedError
 
     # -- phase detection (ref model_base.py:181-186) --------------------
+    # Shape-only so no host sync happens inside a captured decode step;
+    # speculation passes `phase` explicitly.
     @staticmethod
     def _phase(input_ids: torch.Tensor, position_ids: torch.Tensor) -> str:
-        if input_ids.shape[-1] > 1 and bool(position_ids.min() == 0):
-            return PHASE_CTE
-        return PHASE_TKG
+        return PHASE_CTE if input_ids.shape[-1] > 1 else PHASE_TKG
 
     # -- the graph body -------------------------------------------------
     def forward(
@@ -128,8 +128,9 @@ class DecoderModel(nn.Module):
         position_ids: torch.Tensor,
         seq_ids: Optional[torch.Tensor] = None,
         sampling_params: Optional[torch.Tensor] = None,
+        phase: Optional[str] = None,
     ):
-        phase = self._phase(input_ids, position_ids)
+        phase = phase or self._phase(input_ids, position_ids)
         B, S = input_ids.shape
         h = self.embed_tokens(input_ids)
         cos, sin = self.rotary.get_cos_sin(position_ids, device=h.device)
@@ -173,6 +174,34 @@ class CausalLMBase(nn.Module):
         self.ec = config.neuron_config
         self.model = model or self._model_cls(config)
         self.kv_cache_populated = False
+        self.context_encoding_model = None   # ModelWrapper (ref :3107)
+        self.token_generation_model = None   # ModelWrapper (ref :3130)
+        self.speculation_model = None
+
+    def enable_submodels(self):
+        """Create the per-phase submodel wrappers (ref enable_context_encoding
+        / enable_token_generation, model_base.py:3107-3190)."""
+        from neuronx_distributed_inference_amd.models.model_wrapper import (
+            CONTEXT_ENCODING_MODEL_TAG,
+            TOKEN_GENERATION_MODEL_TAG,
+            ModelWrapper,
+        )
+
+        self.context_encoding_model = ModelWrapper(
+            self.config, self.model, tag=CONTEXT_ENCODING_MODEL_TAG,
+            batch_size=self.ec.ctx_batch_size,
+        )
+        self.token_generation_model = ModelWrapper(
+            self.config, self.model, tag=TOKEN_GENERATION_MODEL_TAG,
+            batch_size=self.ec.tkg_batch_size, n_active_tokens=1,
+        )
+
+    def warmup(self):
+        """Run/capture every submodel bucket once (ref warmup :349-373).
+        On GPU this is where hipGraph capture happens."""
+        if self.context_encoding_model is None:
+            self.enable_submodels()
+        self.token_generation_model.capture()
 
     # ------------------------------------------------------------------
     @classmethod
@@ -233,8 +262,12 @@ class CausalLMBase(nn.Module):
         B, S = input_ids.shape
         if position_ids is None:
             if attention_mask is not None:
-                position_ids = (attention_mask.long().cumsum(-1) - 1).clamp(min=0)
-                position_ids = position_ids[:, -S:]
+                # padded entries get position -1 so the KV scatter skips them
+                # (a clamped position would overwrite a real cache row)
+                cum = attention_mask.long().cumsum(-1) - 1
+                position_ids = torch.where(
+                    attention_mask.bool(), cum, torch.full_like(cum, -1)
+                )[:, -S:]
             else:
                 position_ids = torch.arange(S, device=input_ids.device).view(1, S).expand(B, S)
         if sampling_params is None and self.ec.on_device_sampling_config:
@@ -243,9 +276,26 @@ class CausalLMBase(nn.Module):
                 B, c.top_k if c.do_sample else 1, c.top_p, c.temperature,
                 device=input_ids.device,
             )
-        tokens, logits = self.model(
-            input_ids, attention_mask, position_ids, seq_ids, sampling_params
-        )
+        # drop an all-ones padding mask on the host so the device path never
+        # syncs per layer
+        if attention_mask is not None and S > 1 and bool(attention_mask.all()):
+            attention_mask = None
+        if seq_ids is None:
+            seq_ids = torch.arange(B, device=input_ids.device)
+        # route through the submodel wrappers (bucketing + hipGraph replay)
+        # when enabled, else call the module directly
+        if S > 1 and self.context_encoding_model is not None:
+            tokens, logits = self.context_encoding_model(
+                input_ids, attention_mask, position_ids, seq_ids, sampling_params
+            )
+        elif S == 1 and self.token_generation_model is not None:
+            tokens, logits = self.token_generation_model(
+                input_ids, attention_mask, position_ids, seq_ids, sampling_params
+            )
+        else:
+            tokens, logits = self.model(
+                input_ids, attention_mask, position_ids, seq_ids, sampling_params
+            )
         self.kv_cache_populated = True
         return ModelOutput(tokens=tokens, logits=logits)
 

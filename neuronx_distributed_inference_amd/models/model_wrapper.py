@@ -1,0 +1,173 @@
+"""ModelWrapper: per-submodel bucketing, padding and hipGraph dispatch.
+
+Analog of the reference's `models/model_wrapper.py` (class :50; pad_inputs
+:730-829; get_target_bucket :831; runtime forward :1315-1440) with the
+neuronx-cc flag assembly replaced by hipGraph capture policy. One wrapper
+per submodel tag; tags match the reference's compiled-dir subfolders
+(`model_wrapper.py:37-42`).
+
+Dispatch: inputs are padded to the smallest (batch, seq) bucket, the bucket's
+captured hipGraph is replayed (token generation), or the module runs eagerly
+(context encoding — prefill is GEMM-bound so launch overhead is amortized;
+it also keeps the dense/padded mask specialization out of the graph store).
+Outputs are sliced back to the true batch.
+"""
+
+from __future__ import annotations
+
+import logging
+from typing import List, Optional
+
+import torch
+
+from neuronx_distributed_inference_amd.modules import autobucketing as ab
+from neuronx_distributed_inference_amd.modules.attention.attention_base import (
+    PHASE_CTE,
+    PHASE_TKG,
+)
+
+logger = logging.getLogger("nxdi_amd")
+
+CONTEXT_ENCODING_MODEL_TAG = "context_encoding_model"
+TOKEN_GENERATION_MODEL_TAG = "token_generation_model"
+SPECULATION_MODEL_TAG = "speculation_model"
+FUSED_SPECULATION_MODEL_TAG = "fused_speculation_model"
+MEDUSA_MODEL_TAG = "medusa_speculation_model"
+VISION_ENCODER_MODEL_TAG = "vision_encoder_model"
+
+
+def pad_dim(t: Optional[torch.Tensor], dim: int, target: int, value) -> Optional[torch.Tensor]:
+    if t is None or t.size(dim) >= target:
+        return t
+    pad_shape = list(t.shape)
+    pad_shape[dim] = target - t.size(dim)
+    pad = torch.full(pad_shape, value, dtype=t.dtype, device=t.device)
+    return torch.cat([t, pad], dim=dim)
+
+
+class ModelWrapper:
+    def __init__(
+        self,
+        config,
+        model,  # DecoderModel
+        tag: str = CONTEXT_ENCODING_MODEL_TAG,
+        buckets: Optional[List[int]] = None,
+        batch_size: Optional[int] = None,
+        n_active_tokens: int = 1,
+    ):
+        self.config = config
+        self.ec = config.neuron_config
+        self.model = model
+        self.tag = tag
+        self.is_cte = tag == CONTEXT_ENCODING_MODEL_TAG
+        self.phase = PHASE_CTE if self.is_cte else PHASE_TKG
+        self.n_active_tokens = n_active_tokens
+        if buckets is None:
+            if self.is_cte:
+                buckets = ab.generate_buckets_for_cte(
+                    self.ec.max_context_length, self.ec.max_length,
+                    self.ec.context_encoding_buckets, self.ec.enable_bucketing,
+                )
+            else:
+                buckets = [self.ec.max_length]
+        self.buckets = sorted(buckets)
+        self.batch_size = batch_size or (
+            self.ec.ctx_batch_size if self.is_cte else self.ec.tkg_batch_size
+        )
+        self._graphs = None  # BucketedGraphModel, created at capture time
+
+    # ------------------------------------------------------------------
+    def _use_graphs(self) -> bool:
+        return (
+            not self.is_cte
+            and self.ec.enable_hip_graphs
+            and torch.cuda.is_available()
+            and next(self.model.parameters()).is_cuda
+        )
+
+    def _example_inputs(self, device):
+        B = self.batch_size
+        S = self.n_active_tokens
+        input_ids = torch.zeros(B, S, dtype=torch.long, device=device)
+        position_ids = torch.zeros(B, S, dtype=torch.long, device=device)
+        seq_ids = torch.arange(B, dtype=torch.long, device=device)
+        sampling_params = None
+        if self.ec.on_device_sampling_config is not None:
+            sampling_params = torch.ones(B, 3, dtype=torch.float32, device=device)
+        return (input_ids, position_ids, seq_ids, sampling_params)
+
+    def capture(self):
+        """Capture the decode graph(s). Called from warmup, post weight load."""
+        if not self._use_graphs():
+            return
+        from neuronx_distributed_inference_amd.runtime.graph_engine import (
+            BucketedGraphModel,
+        )
+
+        device = next(self.model.parameters()).device
+        self.model.rotary.ensure_table(self.ec.max_length, device)
+        self._graphs = BucketedGraphModel()
+        ids, pos, sids, sp = self._example_inputs(device)
+
+        def fn(input_ids, position_ids, seq_ids, sampling_params):
+            return self.model(
+                input_ids, None, position_ids, seq_ids, sampling_params,
+                phase=self.phase,
+            )
+
+        key = (self.batch_size, self.n_active_tokens)
+        self._graphs.capture(key, fn, (ids, pos, sids, sp))
+        logger.info("%s: captured hipGraph %s", self.tag, key)
+
+    # ------------------------------------------------------------------
+    def forward(
+        self,
+        input_ids: torch.Tensor,
+        attention_mask: Optional[torch.Tensor],
+        position_ids: torch.Tensor,
+        seq_ids: torch.Tensor,
+        sampling_params: Optional[torch.Tensor] = None,
+    ):
+        B, S = input_ids.shape
+
+        if self.is_cte:
+            bucket = ab.get_target_bucket(S, self.buckets)
+            input_ids = pad_dim(input_ids, 1, bucket, 0)
+            attention_mask = pad_dim(attention_mask, 1, bucket, 0)
+            position_ids = pad_dim(position_ids, 1, bucket, -1)
+            if attention_mask is None and bucket != S:
+                # dense batch padded to the bucket needs an explicit mask
+                attention_mask = torch.zeros(
+                    B, bucket, dtype=torch.int32, device=input_ids.device
+                )
+                attention_mask[:, :S] = 1
+
+        # batch padding: garbage rows carry seq_id -1 so every cache write is
+        # skipped (the SPMD-uniform trick, ref _forward_with_pad :574-703)
+        Bb = self.batch_size
+        if B < Bb:
+            input_ids = pad_dim(input_ids, 0, Bb, 0)
+            attention_mask = pad_dim(attention_mask, 0, Bb, 0)
+            position_ids = pad_dim(position_ids, 0, Bb, 0 if not self.is_cte else -1)
+            seq_ids = pad_dim(seq_ids, 0, Bb, -1)
+            if sampling_params is not None:
+                sampling_params = pad_dim(sampling_params, 0, Bb, 1.0)
+
+        if (
+            self._graphs is not None
+            and self._graphs.has((Bb, S))
+            and not self.is_cte
+        ):
+            tokens, logits = self._graphs.run((Bb, S), input_ids, position_ids, seq_ids, sampling_params)
+        else:
+            tokens, logits = self.model(
+                input_ids, attention_mask, position_ids, seq_ids, sampling_params,
+                phase=self.phase,
+            )
+
+        if B < Bb:
+            tokens = tokens[:B] if tokens is not None else None
+            logits = logits[:B] if logits is not None else None
+        return tokens, logits
+
+    __call__ = forward

@@ -119,9 +119,13 @@ class AttentionBase(nn.Module):
     # ------------------------------------------------------------------
     def _prefill(self, q, k, v, attention_mask) -> torch.Tensor:
         """Causal flash attention over the fresh K/V (the cache holds the
-        same values; using the projections directly avoids a re-read)."""
+        same values; using the projections directly avoids a re-read).
+
+        attention_mask is None for dense (un-padded) batches — the caller
+        drops an all-ones mask on the host so no per-layer sync happens.
+        """
         mask = None
-        if attention_mask is not None and not bool(attention_mask.all()):
+        if attention_mask is not None:
             mask = attn_utils.padded_causal_mask(attention_mask, q.shape[2])
             if self.sliding_window:
                 mask = mask & attn_utils.sliding_window_mask(

@@ -70,13 +70,27 @@ class RotaryEmbedding(nn.Module):
         self._sin_cached = emb.sin()
         self._cached_len = seqlen
 
+    def ensure_table(self, seqlen: int, device):
+        """Build the cos/sin table up to `seqlen` once (host-side, outside
+        any hipGraph capture)."""
+        seqlen = max(seqlen, self.max_position_embeddings)
+        if (
+            self._cos_cached is None
+            or self._cached_len < seqlen
+            or self._cos_cached.device != torch.device(device)
+        ):
+            self._build_table(seqlen, device, torch.float32)
+
     def get_cos_sin(
         self, position_ids: torch.Tensor, device=None, dtype=torch.float32
     ) -> Tuple[torch.Tensor, torch.Tensor]:
-        """position_ids: (B, S) -> cos/sin (B, S, D) fp32."""
+        """position_ids: (B, S) -> cos/sin (B, S, D) fp32.
+
+        Graph-capture safe: pure tensor indexing, no host sync — the table
+        must already cover the positions (ensure_table)."""
         device = device or position_ids.device
-        need = int(position_ids.max().item()) + 1 if position_ids.numel() else 1
-        if self._cos_cached is None or self._cached_len < need or self._cos_cached.device != device:
-            self._build_table(max(need, self.max_position_embeddings), device, dtype)
+        self.ensure_table(1, device)
         pos = position_ids.long().clamp(min=0)
+        if self._cached_len > 0:
+            pos = pos.clamp(max=self._cached_len - 1)
         return self._cos_cached[pos], self._sin_cached[pos]

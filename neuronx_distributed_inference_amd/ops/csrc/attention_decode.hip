@@ -279,8 +279,14 @@ void attention_decode(torch::Tensor out, torch::Tensor q,
         else if (D == 64)                                                       \
             dispatch_gq<T, 64>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
                                (float)scale, (int)window, gq);                  \
+        else if (D == 32)                                                       \
+            dispatch_gq<T, 32>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
+                               (float)scale, (int)window, gq);                  \
+        else if (D == 256)                                                      \
+            dispatch_gq<T, 256>(out, q, k_cache, v_cache, seq_lens, seq_ids,    \
+                                (float)scale, (int)window, gq);                 \
         else                                                                    \
-            TORCH_CHECK(false, "head_dim must be 64 or 128, got ", D);          \
+            TORCH_CHECK(false, "head_dim must be 32/64/128/256, got ", D);      \
     } while (0)
 
     if (q.scalar_type() == at::ScalarType::BFloat16) {

@@ -90,9 +90,10 @@ def kv_cache_update(
         row = int(seq_ids[b])
         if row < 0 or row >= k_cache.shape[0]:
             continue  # garbage row for SPMD-uniform DP writes
-        pos = positions[b].clamp(0, max_s - 1).long()
-        k_cache[row, :, pos, :] = k_new[b].to(k_cache.dtype)
-        v_cache[row, :, pos, :] = v_new[b].to(v_cache.dtype)
+        valid = (positions[b] >= 0) & (positions[b] < max_s)
+        pos = positions[b][valid].long()
+        k_cache[row, :, pos, :] = k_new[b][:, valid].to(k_cache.dtype)
+        v_cache[row, :, pos, :] = v_new[b][:, valid].to(v_cache.dtype)
 
 
 def block_kv_cache_update(

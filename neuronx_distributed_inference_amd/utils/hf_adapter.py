@@ -76,8 +76,10 @@ class HuggingFaceGenerationAdapter:
         cur_mask = attention_mask
         finished = torch.zeros(B, dtype=torch.bool, device=device)
 
-        # ---- prefill ----
-        position_ids = (cur_mask.long().cumsum(-1) - 1).clamp(min=0)
+        # ---- prefill ---- (padded entries get position -1: skipped by the
+        # KV scatter, masked out of attention)
+        cum = cur_mask.long().cumsum(-1) - 1
+        position_ids = torch.where(cur_mask.bool(), cum, torch.full_like(cum, -1))
         out = model(
             input_ids=input_ids,
             attention_mask=cur_mask,
