@@ -65,6 +65,11 @@ class Sampler(nn.Module):
         self.vocab_size = vocab_size
         self.deterministic = config.deterministic if deterministic is None else deterministic
         self._gen: Optional[torch.Generator] = None
+        # The greedy/multinomial choice must be static (no host sync inside a
+        # captured graph). The multinomial path handles per-row top_k=1
+        # correctly (softmax over one candidate picks the argmax), so dynamic
+        # per-request configs always take it.
+        self.static_greedy = not (config.do_sample or config.dynamic)
 
     def _rand(self, batch: int, device) -> torch.Tensor:
         if self.deterministic:
@@ -91,8 +96,7 @@ class Sampler(nn.Module):
                 device=logits.device,
             )
         top_k = sampling_params[:, 0]
-        greedy_all = bool((top_k == 1).all())
-        if greedy_all:
+        if self.static_greedy:
             return pops.distributed_argmax(logits, dim=-1)
 
         # multinomial: distributed top-global_topk then fused sampling
