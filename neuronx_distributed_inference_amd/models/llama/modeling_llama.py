@@ -65,15 +65,19 @@ class LlamaDecoderLayer(nn.Module):
             config.hidden_size, eps=config.rms_norm_eps, dtype=ec.torch_dtype
         )
 
-    def forward(self, h, cos, sin, kv_mgr, position_ids, phase, attention_mask, seq_ids):
-        residual = h
-        h = self.input_layernorm(h)
+    def forward(self, h, residual, cos, sin, kv_mgr, position_ids, phase,
+                attention_mask, seq_ids):
+        # (h, residual) threading: the residual add is fused into the RMSNorm
+        # kernel (one HBM pass instead of add + norm)
+        if residual is None:
+            residual = h
+            h = self.input_layernorm(h)
+        else:
+            h, residual = self.input_layernorm(h, residual)
         h = self.self_attn(h, cos, sin, kv_mgr, position_ids, phase, attention_mask, seq_ids)
-        h = residual + h
-        residual = h
-        h = self.post_attention_layernorm(h)
+        h, residual = self.post_attention_layernorm(h, residual)
         h = self.mlp(h)
-        return residual + h
+        return h, residual
 
 
 class LlamaModel(DecoderModel):

@@ -134,9 +134,16 @@ class DecoderModel(nn.Module):
         B, S = input_ids.shape
         h = self.embed_tokens(input_ids)
         cos, sin = self.rotary.get_cos_sin(position_ids, device=h.device)
+        residual = None
         for layer in self.layers:
-            h = layer(h, cos, sin, self.kv_mgr, position_ids, phase, attention_mask, seq_ids)
-        h = self.norm(h)
+            h, residual = layer(
+                h, residual, cos, sin, self.kv_mgr, position_ids, phase,
+                attention_mask, seq_ids,
+            )
+        if residual is not None:
+            h, _ = self.norm(h, residual)
+        else:
+            h = self.norm(h)
 
         if phase == PHASE_CTE and S > 1:
             if attention_mask is not None:
