@@ -17,6 +17,8 @@ void attention_decode(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
 void topk_topp_sample(torch::Tensor out, torch::Tensor vals, torch::Tensor top_k,
                       torch::Tensor top_p, torch::Tensor temp, torch::Tensor rand,
                       long global_topk);
+void flash_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
+                   torch::Tensor v, double scale, long window);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("rms_norm", &rms_norm, "RMSNorm (CDNA4)");
@@ -28,4 +30,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("block_kv_cache_update", &block_kv_cache_update, "paged KV cache scatter (CDNA4)");
     m.def("attention_decode", &attention_decode, "split-KV decode attention (CDNA4)");
     m.def("topk_topp_sample", &topk_topp_sample, "fused top-k/top-p sampling (CDNA4)");
+    m.def("flash_prefill", &flash_prefill, "MFMA causal flash prefill (CDNA4)");
 }

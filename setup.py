@@ -23,6 +23,7 @@ sources = [
     os.path.join(CSRC, "kv_update.hip"),
     os.path.join(CSRC, "attention_decode.hip"),
     os.path.join(CSRC, "sampling.hip"),
+    os.path.join(CSRC, "attention_prefill.hip"),
 ]
 
 extra = {

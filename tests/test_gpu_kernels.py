@@ -225,3 +225,55 @@ def test_e2e_tiny_decode_gpu_matches_cpu():
     # bf16 vs fp32: allow small divergence late in the horizon
     agree = (got.cpu() == want).float().mean()
     assert agree > 0.85, f"agreement {agree}: {got.cpu()} vs {want}"
+
+
+@pytest.mark.parametrize("S", [96, 256, 1024])
+@pytest.mark.parametrize("D", [64, 128])
+def test_flash_prefill_shapes(S, D):
+    torch.manual_seed(0)
+    B, Hq, Hkv = 2, 4, 2
+    dev = _dev()
+    q = torch.randn(B, Hq, S, D, dtype=torch.bfloat16, device=dev)
+    k = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16, device=dev)
+    v = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16, device=dev)
+    out = ops.attention_prefill(q, k, v, causal=True)
+    from neuronx_distributed_inference_amd.modules.attention.utils import causal_mask
+
+    want = ref.sdpa(q.float().cpu(), k.float().cpu(), v.float().cpu(),
+                    mask=causal_mask(S, S), scale=1.0 / math.sqrt(D))
+    torch.testing.assert_close(out.float().cpu(), want, atol=3e-2, rtol=3e-2)
+
+
+def test_flash_prefill_sliding_window():
+    torch.manual_seed(0)
+    B, Hq, Hkv, S, D = 1, 2, 1, 512, 128
+    dev = _dev()
+    q = torch.randn(B, Hq, S, D, dtype=torch.bfloat16, device=dev)
+    k = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16, device=dev)
+    v = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16, device=dev)
+    out = ops.attention_prefill(q, k, v, causal=True, sliding_window=128)
+    from neuronx_distributed_inference_amd.modules.attention.utils import (
+        sliding_window_mask,
+    )
+
+    want = ref.sdpa(q.float().cpu(), k.float().cpu(), v.float().cpu(),
+                    mask=sliding_window_mask(S, S, 128), scale=1.0 / math.sqrt(D))
+    torch.testing.assert_close(out.float().cpu(), want, atol=3e-2, rtol=3e-2)
+
+
+def test_flash_prefill_online_softmax_spike():
+    """Force the online-softmax rescale branch (guide rule 26): one huge K row
+    late in the sequence makes the running max jump after many tiles."""
+    torch.manual_seed(0)
+    B, Hq, Hkv, S, D = 1, 2, 2, 512, 128
+    dev = _dev()
+    q = torch.randn(B, Hq, S, D, dtype=torch.bfloat16, device=dev)
+    k = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16, device=dev) * 0.1
+    v = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16, device=dev)
+    k[:, :, 400, :] = q[:, :1, 480, :] * 3.0  # spike aligned with a late query
+    out = ops.attention_prefill(q, k, v, causal=True)
+    from neuronx_distributed_inference_amd.modules.attention.utils import causal_mask
+
+    want = ref.sdpa(q.float().cpu(), k.float().cpu(), v.float().cpu(),
+                    mask=causal_mask(S, S), scale=1.0 / math.sqrt(D))
+    torch.testing.assert_close(out.float().cpu(), want, atol=3e-2, rtol=3e-2)
