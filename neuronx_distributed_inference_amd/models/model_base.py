@@ -76,9 +76,7 @@ class DecoderModel(nn.Module):
         )
         self.rotary = self.build_rotary(config)
         self.layers = self.build_layers(config)
-        self.norm = RMSNorm(
-            config.hidden_size, eps=getattr(config, "rms_norm_eps", 1e-6), dtype=self.dtype
-        )
+        self.norm = self.build_norm(config)
         self.lm_head = ColumnParallelLinear(
             config.hidden_size, config.vocab_size, bias=False, dtype=self.dtype,
             tp_degree=ec.tp_degree,
@@ -112,6 +110,12 @@ class DecoderModel(nn.Module):
 
     def build_layers(self, config: InferenceConfig) -> nn.ModuleList:
         raise NotImplementedError
+
+    def build_norm(self, config: InferenceConfig) -> nn.Module:
+        return RMSNorm(
+            config.hidden_size, eps=getattr(config, "rms_norm_eps", 1e-6),
+            dtype=self.dtype,
+        )
 
     # -- phase detection (ref model_base.py:181-186) --------------------
     # Shape-only so no host sync happens inside a captured decode step;

@@ -48,6 +48,7 @@ class AttentionBase(nn.Module):
         self.torch_dtype = ec.torch_dtype
         self.sliding_window = getattr(config, "sliding_window", None) or 0
         self.attn_bias = bool(getattr(config, "attention_bias", False))
+        self.clip_qkv = getattr(config, "clip_qkv", None)
 
         self.qkv_proj = GroupQueryAttentionQKV(
             self.hidden_size, self.num_heads, self.num_kv_heads, self.head_dim,
@@ -75,6 +76,10 @@ class AttentionBase(nn.Module):
         (ref prep_qkv_tensors attention_base.py:489-563)."""
         B, S, _ = hidden.shape
         q, k, v = self.qkv_proj(hidden)
+        if self.clip_qkv is not None:  # DBRX clamp (ref modeling_dbrx.py)
+            q = q.clamp(-self.clip_qkv, self.clip_qkv)
+            k = k.clamp(-self.clip_qkv, self.clip_qkv)
+            v = v.clamp(-self.clip_qkv, self.clip_qkv)
         q = q.view(B, S, self.num_heads_per_rank, self.head_dim)
         k = k.view(B, S, self.num_kv_heads_per_rank, self.head_dim)
         v = v.view(B, S, self.num_kv_heads_per_rank, self.head_dim)

@@ -1,0 +1,167 @@
+"""Model-hub oracle tests: every model family vs its HF transformers
+implementation on a tiny random-weight config (CPU fp32, exact greedy
+token match — the reference's check_accuracy strategy)."""
+
+import pytest
+import torch
+
+from neuronx_distributed_inference_amd.config import EngineConfig
+from neuronx_distributed_inference_amd.utils.hf_adapter import (
+    GenerationConfig,
+    HuggingFaceGenerationAdapter,
+)
+
+COMMON = dict(vocab_size=128, hidden_size=64, intermediate_size=128,
+              num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+              max_position_embeddings=128, tie_word_embeddings=False)
+
+
+def _engine():
+    return EngineConfig(batch_size=2, max_batch_size=2, seq_len=48, max_length=48,
+                        torch_dtype="float32", tp_degree=1)
+
+
+def _check_match(ours, hf, n_new=10, vocab=128):
+    torch.manual_seed(5)
+    ids = torch.randint(0, vocab, (2, 8))
+    mask = torch.ones_like(ids)
+    got = HuggingFaceGenerationAdapter(ours).generate(
+        ids, attention_mask=mask, generation_config=GenerationConfig(max_new_tokens=n_new))
+    want = hf.generate(ids, attention_mask=mask, max_new_tokens=n_new, do_sample=False)
+    assert torch.equal(got, want), (got.tolist(), want.tolist())
+
+
+@torch.inference_mode()
+def test_mistral():
+    from transformers import MistralConfig, MistralForCausalLM as HF
+
+    from neuronx_distributed_inference_amd.models.mistral import (
+        MistralForCausalLM, MistralInferenceConfig,
+    )
+
+    torch.manual_seed(0)
+    hf_cfg = MistralConfig(**COMMON, rms_norm_eps=1e-5, sliding_window=None)
+    hf = HF(hf_cfg).eval()
+    cfg = MistralInferenceConfig(neuron_config=_engine(), **hf_cfg.to_dict())
+    ours = MistralForCausalLM(cfg).eval()
+    sd = MistralForCausalLM.convert_hf_to_native_state_dict(hf.state_dict(), cfg)
+    missing, unexpected = ours.model.load_state_dict(sd, strict=False)
+    assert not [m for m in missing if not m.startswith("kv_mgr")] and not unexpected
+    _check_match(ours, hf)
+
+
+@torch.inference_mode()
+def test_qwen2():
+    from transformers import Qwen2Config, Qwen2ForCausalLM as HF
+
+    from neuronx_distributed_inference_amd.models.qwen2 import (
+        Qwen2ForCausalLM, Qwen2InferenceConfig,
+    )
+
+    torch.manual_seed(0)
+    hf_cfg = Qwen2Config(**COMMON, rms_norm_eps=1e-5)
+    hf = HF(hf_cfg).eval()
+    cfg = Qwen2InferenceConfig(neuron_config=_engine(), **hf_cfg.to_dict())
+    ours = Qwen2ForCausalLM(cfg).eval()
+    sd = Qwen2ForCausalLM.convert_hf_to_native_state_dict(hf.state_dict(), cfg)
+    missing, unexpected = ours.model.load_state_dict(sd, strict=False)
+    assert not [m for m in missing if not m.startswith("kv_mgr")] and not unexpected
+    _check_match(ours, hf)
+
+
+@torch.inference_mode()
+def test_qwen3():
+    from transformers import Qwen3Config, Qwen3ForCausalLM as HF
+
+    from neuronx_distributed_inference_amd.models.qwen3 import (
+        Qwen3ForCausalLM, Qwen3InferenceConfig,
+    )
+
+    torch.manual_seed(0)
+    hf_cfg = Qwen3Config(**COMMON, rms_norm_eps=1e-5, head_dim=16)
+    hf = HF(hf_cfg).eval()
+    cfg = Qwen3InferenceConfig(neuron_config=_engine(), **hf_cfg.to_dict())
+    ours = Qwen3ForCausalLM(cfg).eval()
+    sd = Qwen3ForCausalLM.convert_hf_to_native_state_dict(hf.state_dict(), cfg)
+    missing, unexpected = ours.model.load_state_dict(sd, strict=False)
+    assert not [m for m in missing if not m.startswith("kv_mgr")] and not unexpected
+    _check_match(ours, hf)
+
+
+@torch.inference_mode()
+def test_mixtral():
+    from transformers import MixtralConfig, MixtralForCausalLM as HF
+
+    from neuronx_distributed_inference_amd.models.mixtral import (
+        MixtralForCausalLM, MixtralInferenceConfig,
+    )
+
+    torch.manual_seed(0)
+    hf_cfg = MixtralConfig(**COMMON, rms_norm_eps=1e-5, num_local_experts=4,
+                           num_experts_per_tok=2)
+    hf = HF(hf_cfg).eval()
+    cfg = MixtralInferenceConfig(neuron_config=_engine(), **hf_cfg.to_dict())
+    ours = MixtralForCausalLM(cfg).eval()
+    sd = MixtralForCausalLM.convert_hf_to_native_state_dict(hf.state_dict(), cfg)
+    missing, unexpected = ours.model.load_state_dict(sd, strict=False)
+    assert not [m for m in missing if not m.startswith("kv_mgr")] and not unexpected
+    _check_match(ours, hf)
+
+
+@torch.inference_mode()
+def test_qwen3_moe():
+    from transformers import Qwen3MoeConfig, Qwen3MoeForCausalLM as HF
+
+    from neuronx_distributed_inference_amd.models.qwen3_moe import (
+        Qwen3MoeForCausalLM, Qwen3MoeInferenceConfig,
+    )
+
+    torch.manual_seed(0)
+    hf_cfg = Qwen3MoeConfig(**COMMON, rms_norm_eps=1e-5, head_dim=16,
+                            num_experts=4, num_experts_per_tok=2,
+                            moe_intermediate_size=64, decoder_sparse_step=1,
+                            norm_topk_prob=True)
+    hf = HF(hf_cfg).eval()
+    cfg = Qwen3MoeInferenceConfig(neuron_config=_engine(), **hf_cfg.to_dict())
+    ours = Qwen3MoeForCausalLM(cfg).eval()
+    sd = Qwen3MoeForCausalLM.convert_hf_to_native_state_dict(hf.state_dict(), cfg)
+    missing, unexpected = ours.model.load_state_dict(sd, strict=False)
+    assert not [m for m in missing if not m.startswith("kv_mgr")] and not unexpected
+    _check_match(ours, hf)
+
+
+@torch.inference_mode()
+def test_dbrx():
+    from transformers import DbrxConfig, DbrxForCausalLM as HF
+
+    from neuronx_distributed_inference_amd.models.dbrx import (
+        DbrxForCausalLM, DbrxInferenceConfig,
+    )
+
+    torch.manual_seed(0)
+    hf_cfg = DbrxConfig(
+        d_model=64, n_heads=4, n_layers=2, max_seq_len=128, vocab_size=128,
+        attn_config={"kv_n_heads": 2, "clip_qkv": 8.0, "rope_theta": 10000.0},
+        ffn_config={"ffn_hidden_size": 96, "moe_num_experts": 4, "moe_top_k": 2},
+        resid_pdrop=0.0, emb_pdrop=0.0,
+    )
+    hf = HF(hf_cfg).eval()
+    d = hf_cfg.to_dict()
+    cfg = DbrxInferenceConfig(neuron_config=_engine(), **d)
+    ours = DbrxForCausalLM(cfg).eval()
+    sd = DbrxForCausalLM.convert_hf_to_native_state_dict(hf.state_dict(), cfg)
+    missing, unexpected = ours.model.load_state_dict(sd, strict=False)
+    assert not [m for m in missing if not m.startswith("kv_mgr")], missing
+    assert not unexpected, unexpected
+    _check_match(ours, hf)
+
+
+def test_registry_resolves_all():
+    from neuronx_distributed_inference_amd.registry import (
+        available_model_types,
+        get_model_classes,
+    )
+
+    for mt in available_model_types():
+        cfg_cls, model_cls = get_model_classes(mt)
+        assert cfg_cls is not None and model_cls is not None
