@@ -123,10 +123,14 @@ class EngineConfig:
 
     def __init__(self, **kwargs):
         # --- batch / sequence geometry (ref config.py:130-180) ---
-        self.batch_size: int = kwargs.pop("batch_size", 1)
-        self.max_batch_size: int = kwargs.pop("max_batch_size", self.batch_size)
-        self.ctx_batch_size: int = kwargs.pop("ctx_batch_size", self.max_batch_size)
-        self.tkg_batch_size: int = kwargs.pop("tkg_batch_size", self.max_batch_size)
+        def _opt(name, default):
+            v = kwargs.pop(name, None)
+            return default if v is None else v
+
+        self.batch_size: int = _opt("batch_size", 1)
+        self.max_batch_size: int = _opt("max_batch_size", self.batch_size)
+        self.ctx_batch_size: int = _opt("ctx_batch_size", self.max_batch_size)
+        self.tkg_batch_size: int = _opt("tkg_batch_size", self.max_batch_size)
         self.seq_len: int = kwargs.pop("seq_len", 2048)
         self.max_context_length: int = kwargs.pop("max_context_length", self.seq_len)
         self.max_length: int = kwargs.pop("max_length", self.seq_len)

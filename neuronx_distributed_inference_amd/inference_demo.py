@@ -1,0 +1,270 @@
+"""`inference_demo` CLI.
+
+Analog of the reference's `inference_demo.py` (:73-803): argparse front-end
+over the compile -> load -> accuracy-check -> generate -> benchmark
+pipeline, with the same flag names where they exist on MI355X. Multi-rank
+runs launch this same entry under torchrun (one process per GPU over RCCL).
+
+    inference_demo --model-type llama --task-type causal-lm run \
+        --model-path /ckpt --compiled-model-path /compiled \
+        --torch-dtype bfloat16 --tp-degree 8 --batch-size 2 \
+        --max-context-length 128 --seq-len 256 \
+        --on-device-sampling --top-k 1 \
+        --prompt "I believe the meaning of life is" \
+        --check-accuracy-mode token-matching --benchmark
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import logging
+import os
+import sys
+from typing import List, Optional
+
+import torch
+
+from neuronx_distributed_inference_amd.config import (
+    EngineConfig,
+    MoEEngineConfig,
+    OnDeviceSamplingConfig,
+)
+from neuronx_distributed_inference_amd.registry import available_model_types, get_model_classes
+
+logger = logging.getLogger("nxdi_amd")
+
+MODEL_TYPES = available_model_types()
+CHECK_ACCURACY_MODES = ["skip-accuracy-check", "token-matching", "logit-matching"]
+
+
+def parse_args(argv=None):
+    p = argparse.ArgumentParser(prog="inference_demo")
+    p.add_argument("--model-type", required=True, choices=MODEL_TYPES)
+    p.add_argument("--task-type", default="causal-lm", choices=["causal-lm"])
+    sub = p.add_subparsers(dest="command", required=True)
+    run = sub.add_parser("run")
+    setup_run_parser(run)
+    return p.parse_args(argv)
+
+
+def setup_run_parser(run: argparse.ArgumentParser):
+    run.add_argument("--model-path", required=True)
+    run.add_argument("--compiled-model-path", required=True)
+    # generation
+    run.add_argument("--prompt", action="append", dest="prompts", default=None)
+    run.add_argument("--prompt-ids", default=None,
+                     help="JSON list-of-lists of token ids (tokenizer-free runs)")
+    run.add_argument("--max-new-tokens", type=int, default=32)
+    run.add_argument("--top-k", type=int, default=1)
+    run.add_argument("--top-p", type=float, default=1.0)
+    run.add_argument("--temperature", type=float, default=1.0)
+    run.add_argument("--do-sample", action="store_true")
+    run.add_argument("--global-topk", type=int, default=256)
+    run.add_argument("--on-device-sampling", action="store_true")
+    run.add_argument("--pad-token-id", type=int, default=0)
+    # engine geometry
+    run.add_argument("--torch-dtype", default="bfloat16")
+    run.add_argument("--batch-size", type=int, default=1)
+    run.add_argument("--max-batch-size", type=int, default=None)
+    run.add_argument("--ctx-batch-size", type=int, default=None)
+    run.add_argument("--tkg-batch-size", type=int, default=None)
+    run.add_argument("--max-context-length", type=int, default=128)
+    run.add_argument("--seq-len", type=int, default=256)
+    run.add_argument("--n-active-tokens", type=int, default=1)
+    run.add_argument("--enable-bucketing", action="store_true", default=True)
+    run.add_argument("--no-bucketing", dest="enable_bucketing", action="store_false")
+    run.add_argument("--context-encoding-buckets", nargs="+", type=int, default=None)
+    run.add_argument("--token-generation-buckets", nargs="+", type=int, default=None)
+    run.add_argument("--continuous-batching", action="store_true")
+    # parallelism
+    run.add_argument("--tp-degree", type=int, default=1)
+    run.add_argument("--pp-degree", type=int, default=1)
+    run.add_argument("--ep-degree", type=int, default=1)
+    run.add_argument("--cp-degree", type=int, default=1)
+    run.add_argument("--attention-dp-degree", type=int, default=1)
+    run.add_argument("--start-rank-id", type=int, default=0)
+    run.add_argument("--local-ranks-size", type=int, default=None)
+    # speculation
+    run.add_argument("--draft-model-path", default=None)
+    run.add_argument("--speculation-length", type=int, default=0)
+    run.add_argument("--enable-fused-speculation", action="store_true")
+    # engine knobs
+    run.add_argument("--no-hip-graphs", dest="enable_hip_graphs", action="store_false",
+                     default=True)
+    run.add_argument("--skip-warmup", action="store_true")
+    run.add_argument("--on-cpu", action="store_true")
+    # pipeline control
+    run.add_argument("--check-accuracy-mode", default="skip-accuracy-check",
+                     choices=CHECK_ACCURACY_MODES)
+    run.add_argument("--divergence-difference-tol", type=float, default=0.001)
+    run.add_argument("--benchmark", action="store_true")
+    run.add_argument("--num-benchmark-runs", type=int, default=5)
+    run.add_argument("--skip-compile", action="store_true")
+    run.add_argument("--compile-only", action="store_true")
+    run.add_argument("--seed", type=int, default=0)
+
+
+def create_engine_config(args, model_type: str) -> EngineConfig:
+    """args -> EngineConfig (ref create_neuron_config, inference_demo.py:436)."""
+    odsc = None
+    if args.on_device_sampling:
+        odsc = OnDeviceSamplingConfig(
+            top_k=args.top_k, top_p=args.top_p, temperature=args.temperature,
+            do_sample=args.do_sample, global_topk=args.global_topk,
+        )
+    cls = MoEEngineConfig if model_type in ("mixtral", "dbrx", "qwen3_moe") else EngineConfig
+    return cls(
+        batch_size=args.batch_size,
+        max_batch_size=args.max_batch_size or args.batch_size,
+        ctx_batch_size=args.ctx_batch_size,
+        tkg_batch_size=args.tkg_batch_size,
+        max_context_length=args.max_context_length,
+        seq_len=args.seq_len,
+        max_length=args.seq_len,
+        n_active_tokens=args.n_active_tokens,
+        torch_dtype=args.torch_dtype if not args.on_cpu else "float32",
+        enable_bucketing=args.enable_bucketing,
+        context_encoding_buckets=args.context_encoding_buckets,
+        token_generation_buckets=args.token_generation_buckets,
+        is_continuous_batching=args.continuous_batching,
+        tp_degree=args.tp_degree,
+        pp_degree=args.pp_degree,
+        ep_degree=args.ep_degree,
+        cp_degree=args.cp_degree,
+        attention_dp_degree=args.attention_dp_degree,
+        start_rank_id=args.start_rank_id,
+        local_ranks_size=args.local_ranks_size,
+        on_device_sampling_config=odsc,
+        speculation_length=args.speculation_length,
+        enable_fused_speculation=args.enable_fused_speculation,
+        enable_hip_graphs=args.enable_hip_graphs,
+        skip_warmup=args.skip_warmup,
+        on_cpu=args.on_cpu,
+    )
+
+
+def load_tokenizer(path: str):
+    try:
+        from transformers import AutoTokenizer
+
+        return AutoTokenizer.from_pretrained(path)
+    except Exception as e:  # noqa: BLE001
+        logger.warning("no tokenizer loaded from %s (%s)", path, e)
+        return None
+
+
+def get_input_ids(args, tokenizer, device):
+    if args.prompt_ids:
+        ids = json.loads(args.prompt_ids)
+        maxlen = max(len(r) for r in ids)
+        input_ids = torch.full((len(ids), maxlen), args.pad_token_id, dtype=torch.long)
+        mask = torch.zeros_like(input_ids)
+        for i, r in enumerate(ids):
+            input_ids[i, : len(r)] = torch.tensor(r)
+            mask[i, : len(r)] = 1
+        return input_ids.to(device), mask.to(device)
+    assert tokenizer is not None, "--prompt requires tokenizer files beside the model"
+    prompts = args.prompts or ["I believe the meaning of life is"]
+    enc = tokenizer(prompts, return_tensors="pt", padding=True)
+    return enc["input_ids"].to(device), enc["attention_mask"].to(device)
+
+
+def run_inference(args, model_type: str) -> int:
+    from neuronx_distributed_inference_amd.models.application_base import (
+        InferenceApplication,
+    )
+    from neuronx_distributed_inference_amd.parallel import state as ps
+    from neuronx_distributed_inference_amd.utils.hf_adapter import (
+        GenerationConfig,
+        HuggingFaceGenerationAdapter,
+    )
+
+    torch.manual_seed(args.seed)
+    config_cls, model_cls = get_model_classes(model_type)
+    local_rank = ps.init_distributed_from_env()
+    ps.initialize_model_parallel(
+        tensor_model_parallel_size=args.tp_degree,
+        pipeline_model_parallel_size=args.pp_degree,
+        expert_model_parallel_size=args.ep_degree,
+        context_parallel_size=args.cp_degree,
+        attention_dp_size=args.attention_dp_degree,
+    )
+    device = torch.device(f"cuda:{local_rank}") if (
+        torch.cuda.is_available() and not args.on_cpu
+    ) else torch.device("cpu")
+
+    engine_config = create_engine_config(args, model_type)
+    config = config_cls.from_hf_config(args.model_path, neuron_config=engine_config)
+    app = InferenceApplication(model_cls, config, model_path=args.model_path)
+
+    rank0 = ps.get_tensor_model_parallel_rank() == 0
+    if not args.skip_compile:
+        if rank0:
+            app.compile(args.compiled_model_path)
+        if torch.distributed.is_initialized():
+            torch.distributed.barrier()
+    if args.compile_only:
+        return 0
+
+    model = app.load(args.compiled_model_path, device=device)
+    tokenizer = load_tokenizer(args.compiled_model_path) or load_tokenizer(args.model_path)
+    input_ids, attention_mask = get_input_ids(args, tokenizer, device)
+
+    gc = GenerationConfig(
+        max_new_tokens=args.max_new_tokens, do_sample=args.do_sample,
+        top_k=args.top_k, top_p=args.top_p, temperature=args.temperature,
+        pad_token_id=args.pad_token_id,
+        eos_token_id=getattr(tokenizer, "eos_token_id", None) if tokenizer else None,
+    )
+
+    if args.check_accuracy_mode != "skip-accuracy-check" and rank0:
+        run_accuracy_check(args, model, config, input_ids, attention_mask, gc)
+
+    out = HuggingFaceGenerationAdapter(model).generate(
+        input_ids, attention_mask=attention_mask, generation_config=gc
+    )
+    if rank0:
+        if tokenizer is not None:
+            for i, row in enumerate(out):
+                print(f"--- generation {i} ---")
+                print(tokenizer.decode(row, skip_special_tokens=True))
+        else:
+            print(out.tolist())
+
+    if args.benchmark and rank0:
+        from neuronx_distributed_inference_amd.utils.benchmark import benchmark_sampling
+
+        report = benchmark_sampling(
+            model, input_ids, attention_mask=attention_mask,
+            n_runs=args.num_benchmark_runs, generation_config=gc,
+            report_path=os.path.join(args.compiled_model_path, "benchmark_report.json"),
+        )
+        print(json.dumps(report, indent=2))
+    return 0
+
+
+def run_accuracy_check(args, model, config, input_ids, attention_mask, gc) -> None:
+    """Token/logit matching vs the HF CPU implementation of the same
+    checkpoint (ref run_accuracy_check, inference_demo.py:710)."""
+    from transformers import AutoModelForCausalLM
+
+    from neuronx_distributed_inference_amd.utils.accuracy import check_accuracy
+
+    hf = AutoModelForCausalLM.from_pretrained(args.model_path, torch_dtype=torch.float32)
+    hf.eval()
+    check_accuracy(
+        model, input_ids, attention_mask=attention_mask, hf_model=hf,
+        generation_config=gc,
+    )
+    print("accuracy check passed (token matching)")
+
+
+def main(argv=None) -> int:
+    logging.basicConfig(level=os.environ.get("NXDI_LOG_LEVEL", "INFO"))
+    args = parse_args(argv)
+    return run_inference(args, args.model_type)
+
+
+if __name__ == "__main__":
+    sys.exit(main())
