@@ -149,12 +149,15 @@ class DecoderModel(nn.Module):
         else:
             h = self.norm(h)
 
+        is_speculation = phase == PHASE_TKG and S > 1
         if phase == PHASE_CTE and S > 1:
             if attention_mask is not None:
                 last = attention_mask.sum(dim=-1).long() - 1
             else:
                 last = torch.full((B,), S - 1, dtype=torch.long, device=h.device)
             h_last = h[torch.arange(B, device=h.device), last].unsqueeze(1)
+        elif is_speculation:
+            h_last = h  # verify pass needs every position's logits
         else:
             h_last = h[:, -1:, :]
 
@@ -163,7 +166,15 @@ class DecoderModel(nn.Module):
         tokens = None
         logits = None
         if self.sampler is not None:
-            tokens = self.sampler(logits_shard, sampling_params)
+            if is_speculation:
+                # greedy per-position target tokens for the accept test
+                from neuronx_distributed_inference_amd.parallel.ops import (
+                    distributed_argmax,
+                )
+
+                tokens = distributed_argmax(logits_shard, dim=-1)  # (B, S)
+            else:
+                tokens = self.sampler(logits_shard, sampling_params)
         if self.sampler is None or self.ec.output_logits:
             logits = mappings.gather_from_tensor_parallel_region(logits_shard, dim=-1)
         return tokens, logits
@@ -206,6 +217,16 @@ class CausalLMBase(nn.Module):
             self.config, self.model, tag=TOKEN_GENERATION_MODEL_TAG,
             batch_size=self.ec.tkg_batch_size, n_active_tokens=1,
         )
+        if self.ec.enable_speculation:
+            from neuronx_distributed_inference_amd.models.model_wrapper import (
+                SPECULATION_MODEL_TAG,
+            )
+
+            self.speculation_model = ModelWrapper(
+                self.config, self.model, tag=SPECULATION_MODEL_TAG,
+                batch_size=self.ec.tkg_batch_size,
+                n_active_tokens=self.ec.speculation_length + 1,
+            )
 
     def warmup(self):
         """Run/capture every submodel bucket once (ref warmup :349-373).
@@ -213,6 +234,8 @@ class CausalLMBase(nn.Module):
         if self.context_encoding_model is None:
             self.enable_submodels()
         self.token_generation_model.capture()
+        if self.speculation_model is not None:
+            self.speculation_model.capture()
 
     # ------------------------------------------------------------------
     @classmethod
@@ -295,7 +318,12 @@ class CausalLMBase(nn.Module):
             seq_ids = torch.arange(B, device=input_ids.device)
         # route through the submodel wrappers (bucketing + hipGraph replay)
         # when enabled, else call the module directly
-        if S > 1 and self.context_encoding_model is not None:
+        is_spec_step = S > 1 and self.kv_cache_populated
+        if is_spec_step and self.speculation_model is not None:
+            tokens, logits = self.speculation_model(
+                input_ids, attention_mask, position_ids, seq_ids, sampling_params
+            )
+        elif S > 1 and not is_spec_step and self.context_encoding_model is not None:
             tokens, logits = self.context_encoding_model(
                 input_ids, attention_mask, position_ids, seq_ids, sampling_params
             )
@@ -304,8 +332,13 @@ class CausalLMBase(nn.Module):
                 input_ids, attention_mask, position_ids, seq_ids, sampling_params
             )
         else:
+            from neuronx_distributed_inference_amd.modules.attention.attention_base import (
+                PHASE_TKG,
+            )
+
             tokens, logits = self.model(
-                input_ids, attention_mask, position_ids, seq_ids, sampling_params
+                input_ids, attention_mask, position_ids, seq_ids, sampling_params,
+                phase=PHASE_TKG if is_spec_step else None,
             )
         self.kv_cache_populated = True
         return ModelOutput(tokens=tokens, logits=logits)
