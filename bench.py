@@ -86,6 +86,7 @@ def main():
         torch_dtype=dtype, tp_degree=tp,
         on_device_sampling_config=OnDeviceSamplingConfig(top_k=1),
         enable_hip_graphs=not args.no_graphs,
+        fused_qkv=True, fused_gate_up=True,
     )
     cfg = LlamaInferenceConfig(
         neuron_config=ec, max_position_embeddings=total_len * 2,

@@ -241,6 +241,7 @@ class EngineConfig:
 
         # --- attention structure ---
         self.fused_qkv: bool = kwargs.pop("fused_qkv", False)
+        self.fused_gate_up: bool = kwargs.pop("fused_gate_up", False)
         self.qkv_kernel_enabled: bool = kwargs.pop("qkv_kernel_enabled", False)
         self.sliding_window: Optional[int] = kwargs.pop("sliding_window", None)
 

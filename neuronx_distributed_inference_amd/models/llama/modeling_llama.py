@@ -23,6 +23,33 @@ from neuronx_distributed_inference_amd.modules.mlp import GatedMLP
 from neuronx_distributed_inference_amd.modules.norm import RMSNorm
 
 
+def fuse_qkv_keys(sd: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
+    """cat q|k|v weights (and biases) into qkv_proj.qkv.* keys."""
+    out = dict(sd)
+    for suffix in ("weight", "bias"):
+        qs = [k for k in sd if k.endswith(f"qkv_proj.q_proj.{suffix}")]
+        for qk in qs:
+            base = qk[: -len(f"q_proj.{suffix}")]
+            kk, vk = f"{base}k_proj.{suffix}", f"{base}v_proj.{suffix}"
+            out[f"{base}qkv.{suffix}"] = torch.cat([sd[qk], sd[kk], sd[vk]], dim=0)
+            for key in (qk, kk, vk):
+                out.pop(key)
+    return out
+
+
+def fuse_gate_up_keys(sd: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
+    out = dict(sd)
+    for suffix in ("weight", "bias"):
+        gs = [k for k in sd if k.endswith(f"mlp.gate_proj.{suffix}")]
+        for gk in gs:
+            base = gk[: -len(f"gate_proj.{suffix}")]
+            uk = f"{base}up_proj.{suffix}"
+            out[f"{base}gate_up_proj.{suffix}"] = torch.cat([sd[gk], sd[uk]], dim=0)
+            out.pop(gk)
+            out.pop(uk)
+    return out
+
+
 class LlamaInferenceConfig(InferenceConfig):
     def get_required_attributes(self) -> List[str]:
         return [
@@ -57,6 +84,7 @@ class LlamaDecoderLayer(nn.Module):
             dtype=ec.torch_dtype,
             hidden_act=getattr(config, "hidden_act", "silu"),
             bias=bool(getattr(config, "mlp_bias", False)),
+            fused_gate_up=ec.fused_gate_up,
         )
         self.input_layernorm = RMSNorm(
             config.hidden_size, eps=config.rms_norm_eps, dtype=ec.torch_dtype
@@ -95,6 +123,7 @@ class LlamaForCausalLM(CausalLMBase):
     def convert_hf_to_native_state_dict(
         sd: Dict[str, torch.Tensor], config: InferenceConfig
     ) -> Dict[str, torch.Tensor]:
+        ec = config.neuron_config
         out: Dict[str, torch.Tensor] = {}
         for k, v in sd.items():
             nk = k
@@ -107,6 +136,10 @@ class LlamaForCausalLM(CausalLMBase):
             nk = nk.replace("self_attn.v_proj", "self_attn.qkv_proj.v_proj")
             nk = nk.replace("self_attn.o_proj", "self_attn.o_proj.o_proj")
             out[nk] = v
+        if ec.fused_qkv:
+            out = fuse_qkv_keys(out)
+        if ec.fused_gate_up:
+            out = fuse_gate_up_keys(out)
         if "lm_head.weight" not in out and getattr(config, "tie_word_embeddings", False):
             out["lm_head.weight"] = out["embed_tokens.weight"]
         dtype = config.neuron_config.torch_dtype

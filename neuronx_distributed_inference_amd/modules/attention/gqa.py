@@ -97,6 +97,47 @@ class GroupQueryAttentionQKV(nn.Module):
         self.num_heads_per_rank = padded_heads // tp_degree
         self.num_kv_heads_per_rank = max(eff_kv_heads // tp_degree, 1)
 
+        hd = head_dim
+        rep = (eff_kv_heads // num_kv_heads
+               if self.strategy == GQA.REPLICATE_TO_TP_DEGREE else 1)
+
+        if fused:
+            # ONE column-parallel GEMM for q|k|v. The checkpoint carries the
+            # full cat([q, k, v]); the preshard hook pads q heads, replicates
+            # kv heads and reorders rows rank-major so a plain dim-0 slice
+            # yields [q_r | k_r | v_r] per rank.
+            out_rows = (padded_heads + 2 * eff_kv_heads) * head_dim
+            self.qkv = ColumnParallelLinear(
+                hidden_size, out_rows, bias=bias, dtype=dtype,
+                tp_degree=tp_degree, group=group,
+            )
+
+            def _fuse_preshard(w, tp, hd=hd, rep=rep,
+                               nq=num_heads, nkv=num_kv_heads,
+                               pq=padded_heads, ekv=eff_kv_heads):
+                q, k, v = torch.split(w, [nq * hd, nkv * hd, nkv * hd], dim=0)
+                q = _pad_head_blocks(q, hd, pq)
+                if rep > 1:
+                    k = _replicate_head_blocks(k, hd, rep)
+                    v = _replicate_head_blocks(v, hd, rep)
+                qpr, kpr = pq * hd // tp, ekv * hd // tp
+                parts = []
+                for r in range(tp):
+                    parts.append(q[r * qpr:(r + 1) * qpr])
+                    parts.append(k[r * kpr:(r + 1) * kpr])
+                    parts.append(v[r * kpr:(r + 1) * kpr])
+                return torch.cat(parts, dim=0)
+
+            self.qkv.weight.preshard_fn = _fuse_preshard
+            if bias:
+                self.qkv.bias.preshard_fn = _fuse_preshard
+            self._split_sizes = (
+                self.num_heads_per_rank * head_dim,
+                self.num_kv_heads_per_rank * head_dim,
+                self.num_kv_heads_per_rank * head_dim,
+            )
+            return
+
         self.q_proj = ColumnParallelLinear(
             hidden_size, padded_heads * head_dim, bias=bias, dtype=dtype,
             tp_degree=tp_degree, group=group,
@@ -112,18 +153,19 @@ class GroupQueryAttentionQKV(nn.Module):
 
         # preshard hooks: transform the FULL checkpoint tensor so dim-0
         # slicing yields the right shard (ref gqa.py:679-954)
-        hd = head_dim
         if padded_heads != num_heads:
             for p in ([self.q_proj.weight] + ([self.q_proj.bias] if bias else [])):
                 p.preshard_fn = lambda w, tp, hd=hd, th=padded_heads: _pad_head_blocks(w, hd, th)
         if self.strategy == GQA.REPLICATE_TO_TP_DEGREE:
-            rep = eff_kv_heads // num_kv_heads
             for proj in (self.k_proj, self.v_proj):
                 params = [proj.weight] + ([proj.bias] if bias else [])
                 for p in params:
                     p.preshard_fn = lambda w, tp, hd=hd, rep=rep: _replicate_head_blocks(w, hd, rep)
 
     def forward(self, x: torch.Tensor):
+        if self.fused:
+            y = self.qkv(x)
+            return torch.split(y, self._split_sizes, dim=-1)
         return self.q_proj(x), self.k_proj(x), self.v_proj(x)
 
 

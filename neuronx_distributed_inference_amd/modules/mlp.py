@@ -28,16 +28,40 @@ class GatedMLP(nn.Module):
         hidden_act: str = "silu",
         bias: bool = False,
         group=None,
+        fused_gate_up: bool = False,
     ):
         super().__init__()
-        self.gate_proj = ColumnParallelLinear(
-            hidden_size, intermediate_size, bias=bias, dtype=dtype,
-            tp_degree=tp_degree, group=group,
-        )
-        self.up_proj = ColumnParallelLinear(
-            hidden_size, intermediate_size, bias=bias, dtype=dtype,
-            tp_degree=tp_degree, group=group,
-        )
+        self.fused = fused_gate_up
+        if fused_gate_up:
+            # one GEMM for gate|up: checkpoint carries cat([gate, up]); the
+            # preshard hook reorders rows rank-major so a dim-0 slice yields
+            # [gate_r | up_r]
+            self.gate_up_proj = ColumnParallelLinear(
+                hidden_size, 2 * intermediate_size, bias=bias, dtype=dtype,
+                tp_degree=tp_degree, group=group,
+            )
+
+            def _fuse_preshard(w, tp, I=intermediate_size):
+                g, u = torch.split(w, [I, I], dim=0)
+                per = I // tp
+                parts = []
+                for r in range(tp):
+                    parts.append(g[r * per:(r + 1) * per])
+                    parts.append(u[r * per:(r + 1) * per])
+                return torch.cat(parts, dim=0)
+
+            self.gate_up_proj.weight.preshard_fn = _fuse_preshard
+            if bias:
+                self.gate_up_proj.bias.preshard_fn = _fuse_preshard
+        else:
+            self.gate_proj = ColumnParallelLinear(
+                hidden_size, intermediate_size, bias=bias, dtype=dtype,
+                tp_degree=tp_degree, group=group,
+            )
+            self.up_proj = ColumnParallelLinear(
+                hidden_size, intermediate_size, bias=bias, dtype=dtype,
+                tp_degree=tp_degree, group=group,
+            )
         self.down_proj = RowParallelLinear(
             intermediate_size, hidden_size, bias=bias, dtype=dtype,
             tp_degree=tp_degree, group=group,
@@ -50,4 +74,8 @@ class GatedMLP(nn.Module):
             raise ValueError(f"unsupported hidden_act {hidden_act}")
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if self.fused:
+            y = self.gate_up_proj(x)
+            g, u = y.chunk(2, dim=-1)
+            return self.down_proj(self.act_mul(g, u))
         return self.down_proj(self.act_mul(self.gate_proj(x), self.up_proj(x)))

@@ -105,3 +105,39 @@ def test_compile_load_roundtrip(tmp_path, model, tiny_llama_config):
     a = adap_a.generate(ids, generation_config=GenerationConfig(max_new_tokens=5))
     b = adap_b.generate(ids, generation_config=GenerationConfig(max_new_tokens=5))
     assert torch.equal(a, b)
+
+
+@torch.inference_mode()
+def test_fused_projections_match_unfused(tiny_llama_config):
+    """fused_qkv + fused_gate_up must be numerically identical to the
+    separate-projection path (weights fused via the converter)."""
+    import copy
+
+    from neuronx_distributed_inference_amd.models.llama.modeling_llama import (
+        fuse_gate_up_keys,
+        fuse_qkv_keys,
+    )
+
+    torch.manual_seed(0)
+    base = LlamaForCausalLM(tiny_llama_config).eval()
+    with torch.no_grad():
+        for p in base.parameters():
+            if p.dim() >= 2:
+                p.normal_(0, 0.1)
+
+    cfg2 = copy.deepcopy(tiny_llama_config)
+    cfg2.neuron_config.fused_qkv = True
+    cfg2.neuron_config.fused_gate_up = True
+    fused = LlamaForCausalLM(cfg2).eval()
+    sd = {k: v for k, v in base.model.state_dict().items() if not k.startswith("kv_mgr.")}
+    sd = fuse_gate_up_keys(fuse_qkv_keys(sd))
+    missing, unexpected = fused.model.load_state_dict(sd, strict=False)
+    assert not [m for m in missing if not m.startswith("kv_mgr")], missing
+    assert not unexpected, unexpected
+
+    ids = torch.randint(0, 128, (2, 10))
+    a = HuggingFaceGenerationAdapter(base).generate(
+        ids, generation_config=GenerationConfig(max_new_tokens=8))
+    b = HuggingFaceGenerationAdapter(fused).generate(
+        ids, generation_config=GenerationConfig(max_new_tokens=8))
+    assert torch.equal(a, b)
