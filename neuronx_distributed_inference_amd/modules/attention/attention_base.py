@@ -108,6 +108,31 @@ class AttentionBase(nn.Module):
         active_mask: Optional[torch.Tensor] = None,
     ) -> torch.Tensor:
         B, S, _ = hidden.shape
+
+        # Fused TKG front end (K2-analog): one GEMM -> one kernel doing
+        # RoPE + cache write + q relayout. Skipped when per-head qk-norms or
+        # clip_qkv need the unfused path.
+        if (
+            phase == PHASE_TKG
+            and self.qkv_proj.fused
+            and self.q_norm is None
+            and self.clip_qkv is None
+            and cos is not None
+        ):
+            y = self.qkv_proj.qkv(hidden)
+            kc, vc = kv_mgr.get_cache(self.layer_idx)
+            import neuronx_distributed_inference_amd.ops as _ops
+
+            q = _ops.qkv_rope_update(
+                y, cos, sin, kc.data if hasattr(kc, "data") else kc,
+                vc.data if hasattr(vc, "data") else vc,
+                position_ids, seq_ids,
+                self.num_heads_per_rank, self.num_kv_heads_per_rank,
+            )
+            out = self._decode(q, kv_mgr, position_ids, seq_ids, active_mask)
+            out = out.transpose(1, 2).reshape(B, S, -1)
+            return self.o_proj(out)
+
         q, k, v = self.prep_qkv(hidden, cos, sin)
 
         # write this step's K/V into the cache first (in-place scatter)

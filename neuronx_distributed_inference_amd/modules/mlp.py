@@ -68,14 +68,14 @@ class GatedMLP(nn.Module):
         )
         if hidden_act in ("silu", "swish"):
             self.act_mul = ops.silu_mul
+            self.act_mul_fused = ops.silu_mul_fused
         elif hidden_act in ("gelu", "gelu_pytorch_tanh", "gelu_new"):
             self.act_mul = ops.gelu_tanh_mul
+            self.act_mul_fused = ops.gelu_tanh_mul_fused
         else:
             raise ValueError(f"unsupported hidden_act {hidden_act}")
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if self.fused:
-            y = self.gate_up_proj(x)
-            g, u = y.chunk(2, dim=-1)
-            return self.down_proj(self.act_mul(g, u))
+            return self.down_proj(self.act_mul_fused(self.gate_up_proj(x)))
         return self.down_proj(self.act_mul(self.gate_proj(x), self.up_proj(x)))

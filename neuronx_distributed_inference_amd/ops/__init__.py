@@ -202,3 +202,56 @@ def topk_topp_sample(logits, top_k, top_p, temperature, rand, global_topk: int =
         )
         return out
     return ref.topk_topp_sample(logits, top_k, top_p, temperature, rand, global_topk)
+
+
+def qkv_rope_update(qkv, cos, sin, k_cache, v_cache, positions, seq_ids, Hq, Hkv):
+    """Fused qkv-GEMM output -> RoPE(q,k) -> cache write(k,v) -> q (B,Hq,S,D).
+
+    qkv: (B, S, (Hq+2Hkv)*D) contiguous; cos/sin (B,S,D) fp32.
+    """
+    B, S, _ = qkv.shape
+    D = k_cache.shape[3]
+    if qkv.is_cuda:
+        c = _require_ext("qkv_rope_update")
+        q_out = torch.empty(B, Hq, S, D, dtype=qkv.dtype, device=qkv.device)
+        if seq_ids is None:
+            seq_ids = torch.arange(B, device=qkv.device, dtype=torch.int32)
+        c.qkv_rope_update(
+            q_out, qkv.contiguous(), cos.contiguous(), sin.contiguous(),
+            k_cache, v_cache, positions.to(torch.int32).contiguous(),
+            seq_ids.to(torch.int32).contiguous(), Hq, Hkv,
+        )
+        return q_out
+    # CPU reference composition
+    q, k, v = torch.split(qkv, [Hq * D, Hkv * D, Hkv * D], dim=-1)
+    q = q.reshape(B, S, Hq, D).transpose(1, 2)
+    k = k.reshape(B, S, Hkv, D).transpose(1, 2)
+    v = v.reshape(B, S, Hkv, D).transpose(1, 2)
+    q, k = ref.rope_apply(q, k, cos, sin)
+    ref.kv_cache_update(k_cache, v_cache, k, v, positions, seq_ids)
+    return q.contiguous()
+
+
+def silu_mul_fused(gate_up):
+    """SiLU(gate)*up on a fused (rows, 2I) [gate|up] tensor."""
+    I = gate_up.shape[-1] // 2
+    if gate_up.is_cuda:
+        c = _require_ext("silu_mul_fused")
+        out = torch.empty(gate_up.shape[:-1] + (I,), dtype=gate_up.dtype,
+                          device=gate_up.device)
+        c.silu_mul_fused(out, gate_up.contiguous())
+        return out
+    g, u = gate_up.chunk(2, dim=-1)
+    return ref.silu_mul(g.contiguous(), u.contiguous())
+
+
+def gelu_tanh_mul_fused(gate_up):
+    I = gate_up.shape[-1] // 2
+    if gate_up.is_cuda:
+        c = _require_ext("gelu_tanh_mul_fused")
+        out = torch.empty(gate_up.shape[:-1] + (I,), dtype=gate_up.dtype,
+                          device=gate_up.device)
+        c.gelu_tanh_mul_fused(out, gate_up.contiguous())
+        return out
+    g, u = gate_up.chunk(2, dim=-1)
+    return ref.gelu_tanh_mul(g.contiguous(), u.contiguous())

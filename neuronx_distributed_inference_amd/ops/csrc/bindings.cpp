@@ -19,6 +19,11 @@ void topk_topp_sample(torch::Tensor out, torch::Tensor vals, torch::Tensor top_k
                       long global_topk);
 void flash_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                    torch::Tensor v, double scale, long window);
+void qkv_rope_update(torch::Tensor q_out, torch::Tensor qkv, torch::Tensor cosb,
+                     torch::Tensor sinb, torch::Tensor k_cache, torch::Tensor v_cache,
+                     torch::Tensor positions, torch::Tensor seq_ids, long Hq, long Hkv);
+void silu_mul_fused(torch::Tensor out, torch::Tensor gate_up);
+void gelu_tanh_mul_fused(torch::Tensor out, torch::Tensor gate_up);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("rms_norm", &rms_norm, "RMSNorm (CDNA4)");
@@ -31,4 +36,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("attention_decode", &attention_decode, "split-KV decode attention (CDNA4)");
     m.def("topk_topp_sample", &topk_topp_sample, "fused top-k/top-p sampling (CDNA4)");
     m.def("flash_prefill", &flash_prefill, "MFMA causal flash prefill (CDNA4)");
+    m.def("qkv_rope_update", &qkv_rope_update, "fused qkv->rope->cache-write (CDNA4)");
+    m.def("silu_mul_fused", &silu_mul_fused, "SiLU on fused gate|up (CDNA4)");
+    m.def("gelu_tanh_mul_fused", &gelu_tanh_mul_fused, "GELU on fused gate|up (CDNA4)");
 }

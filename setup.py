@@ -24,6 +24,7 @@ sources = [
     os.path.join(CSRC, "attention_decode.hip"),
     os.path.join(CSRC, "sampling.hip"),
     os.path.join(CSRC, "attention_prefill.hip"),
+    os.path.join(CSRC, "qkv_fused.hip"),
 ]
 
 extra = {

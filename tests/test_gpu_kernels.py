@@ -277,3 +277,35 @@ def test_flash_prefill_online_softmax_spike():
     want = ref.sdpa(q.float().cpu(), k.float().cpu(), v.float().cpu(),
                     mask=causal_mask(S, S), scale=1.0 / math.sqrt(D))
     torch.testing.assert_close(out.float().cpu(), want, atol=3e-2, rtol=3e-2)
+
+
+def test_qkv_rope_update_fused():
+    torch.manual_seed(0)
+    B, S, Hq, Hkv, D, Bc, Smax = 3, 1, 8, 2, 128, 4, 64
+    dev = _dev()
+    qkv = torch.randn(B, S, (Hq + 2 * Hkv) * D, dtype=torch.bfloat16, device=dev)
+    inv = 1.0 / (10000.0 ** (torch.arange(0, D, 2).float() / D))
+    emb = torch.cat([torch.outer(torch.arange(Smax).float(), inv)] * 2, dim=-1)
+    pos = torch.tensor([[5], [0], [63]], device=dev)
+    cos = emb.cos().to(dev)[pos.long().clamp(min=0)]
+    sin = emb.sin().to(dev)[pos.long().clamp(min=0)]
+    kc = torch.zeros(Bc, Hkv, Smax, D, dtype=torch.bfloat16, device=dev)
+    vc = torch.zeros_like(kc)
+    sid = torch.tensor([1, 0, 3], device=dev)
+    q = ops.qkv_rope_update(qkv, cos, sin, kc, vc, pos, sid, Hq, Hkv)
+
+    kc_ref = torch.zeros(Bc, Hkv, Smax, D)
+    vc_ref = torch.zeros_like(kc_ref)
+    q_ref = ops.qkv_rope_update(qkv.float().cpu(), cos.cpu(), sin.cpu(),
+                                kc_ref, vc_ref, pos.cpu(), sid.cpu(), Hq, Hkv)
+    torch.testing.assert_close(q.float().cpu(), q_ref, atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(kc.float().cpu(), kc_ref, atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(vc.float().cpu(), vc_ref, atol=2e-2, rtol=2e-2)
+
+
+def test_silu_mul_fused_gpu():
+    torch.manual_seed(0)
+    y = torch.randn(4, 3, 512, dtype=torch.bfloat16, device=_dev())
+    out = ops.silu_mul_fused(y)
+    want = ops.silu_mul_fused(y.float().cpu())
+    torch.testing.assert_close(out.float().cpu(), want, atol=2e-2, rtol=2e-2)
