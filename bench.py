@@ -4,7 +4,7 @@ Measures BASELINE.json's headline metric — output tokens/sec for
 Llama-3.1-8B decode (TP=1 on one GPU; TP=N across N GPUs over RCCL/xGMI)
 with random-init weights and synthetic data. A "step" is one
 token-generation step for the whole batch (one new token per sequence);
-prefill at ctx=128 happens before the timed region. Batch is 8 per GPU
+prefill at ctx=128 happens before the timed region. Batch is 64 per GPU
 (weak scaling).
 
     python bench.py --gpus N --steps K --warmup W [--model {8b,70b,tiny}]
@@ -38,7 +38,7 @@ def parse_args():
     p.add_argument("--steps", type=int, default=32)
     p.add_argument("--warmup", type=int, default=8)
     p.add_argument("--model", default="8b", choices=list(MODELS))
-    p.add_argument("--batch-per-gpu", type=int, default=8)
+    p.add_argument("--batch-per-gpu", type=int, default=64)
     p.add_argument("--ctx", type=int, default=128)
     p.add_argument("--no-graphs", action="store_true")
     return p.parse_args()

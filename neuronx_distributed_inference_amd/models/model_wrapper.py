@@ -116,8 +116,15 @@ class ModelWrapper:
             )
 
         key = (self.batch_size, self.n_active_tokens)
-        self._graphs.capture(key, fn, (ids, pos, sids, sp))
-        logger.info("%s: captured hipGraph %s", self.tag, key)
+        try:
+            self._graphs.capture(key, fn, (ids, pos, sids, sp))
+            logger.info("%s: captured hipGraph %s", self.tag, key)
+        except Exception:  # noqa: BLE001
+            # never let a capture failure (e.g. an uncapturable collective on
+            # an untested stack) kill the run — eager decode still works
+            logger.exception("%s: hipGraph capture failed; running eager", self.tag)
+            self._graphs = None
+            torch.cuda.synchronize()
 
     # ------------------------------------------------------------------
     def forward(
