@@ -82,16 +82,33 @@ class DecoderModel(nn.Module):
             tp_degree=ec.tp_degree,
         )
         head_dim = getattr(config, "head_dim", None) or config.hidden_size // config.num_attention_heads
-        self.kv_mgr = KVCacheManager(
-            num_layers=self.num_layers,
-            num_kv_heads=getattr(config, "num_key_value_heads", config.num_attention_heads),
-            head_dim=head_dim,
-            max_len=ec.max_length,
-            batch_size=ec.kv_cache_batch_size,
-            dtype=ec.torch_dtype,
-            padding_rows=ec.kv_cache_padding_size,
-            tp_degree=ec.tp_degree,
-        )
+        n_kv_heads = getattr(config, "num_key_value_heads", config.num_attention_heads)
+        if ec.is_block_kv_layout:
+            from neuronx_distributed_inference_amd.modules.kvcache.block_kv_cache_manager import (
+                BlockKVCacheManager,
+            )
+
+            num_blocks = ec.pa_num_blocks or (
+                (ec.max_length + ec.pa_block_size - 1) // ec.pa_block_size
+                * ec.kv_cache_batch_size
+            )
+            self.kv_mgr = BlockKVCacheManager(
+                num_layers=self.num_layers, num_kv_heads=n_kv_heads,
+                head_dim=head_dim, num_blocks=num_blocks,
+                block_size=ec.pa_block_size, dtype=ec.torch_dtype,
+                tp_degree=ec.tp_degree,
+            )
+        else:
+            self.kv_mgr = KVCacheManager(
+                num_layers=self.num_layers,
+                num_kv_heads=n_kv_heads,
+                head_dim=head_dim,
+                max_len=ec.max_length,
+                batch_size=ec.kv_cache_batch_size,
+                dtype=ec.torch_dtype,
+                padding_rows=ec.kv_cache_padding_size,
+                tp_degree=ec.tp_degree,
+            )
         self.sampler = (
             Sampler(ec.on_device_sampling_config, config.vocab_size)
             if ec.on_device_sampling_config
@@ -133,9 +150,17 @@ class DecoderModel(nn.Module):
         seq_ids: Optional[torch.Tensor] = None,
         sampling_params: Optional[torch.Tensor] = None,
         phase: Optional[str] = None,
+        block_table: Optional[torch.Tensor] = None,
     ):
         phase = phase or self._phase(input_ids, position_ids)
         B, S = input_ids.shape
+        if block_table is not None and getattr(self.kv_mgr, "is_paged", False):
+            from neuronx_distributed_inference_amd.modules.kvcache.block_kv_cache_manager import (
+                generate_slot_mapping,
+            )
+
+            slots = generate_slot_mapping(block_table, position_ids, self.kv_mgr.block_size)
+            self.kv_mgr.set_step_context(block_table, slots)
         h = self.embed_tokens(input_ids)
         cos, sin = self.rotary.get_cos_sin(position_ids, device=h.device)
         residual = None

@@ -118,6 +118,7 @@ class AttentionBase(nn.Module):
             and self.q_norm is None
             and self.clip_qkv is None
             and cos is not None
+            and not getattr(kv_mgr, "is_paged", False)
         ):
             y = self.qkv_proj.qkv(hidden)
             kc, vc = kv_mgr.get_cache(self.layer_idx)
@@ -172,6 +173,11 @@ class AttentionBase(nn.Module):
     def _decode(self, q, kv_mgr, position_ids, seq_ids, active_mask) -> torch.Tensor:
         k_cache, v_cache = kv_mgr.get_cache(self.layer_idx)
         seq_lens = position_ids[:, -1] + 1  # attend to everything written so far
+        if getattr(kv_mgr, "is_paged", False):
+            return ops.attention_decode_paged(
+                q, k_cache.data, v_cache.data, seq_lens, kv_mgr.block_table,
+                scale=self.scale, sliding_window=self.sliding_window,
+            )
         return ops.attention_decode(
             q, k_cache, v_cache, seq_lens, seq_ids=seq_ids, scale=self.scale,
             sinks=self.sinks, sliding_window=self.sliding_window,

@@ -18,6 +18,8 @@ from neuronx_distributed_inference_amd.parallel import state as ps
 
 
 class BlockKVCacheManager(nn.Module):
+    is_paged = True
+
     def __init__(
         self,
         num_layers: int,
@@ -49,6 +51,26 @@ class BlockKVCacheManager(nn.Module):
     def estimate_bytes(self) -> int:
         return sum(k.numel() * k.element_size() * 2 for k in self.k_caches)
 
+    # per-forward step context set by DecoderModel (block table + slots)
+    def set_step_context(self, block_table: torch.Tensor, slot_mapping: torch.Tensor):
+        self._block_table = block_table
+        self._slot_mapping = slot_mapping
+
+    @property
+    def block_table(self):
+        return self._block_table
+
+    def get_cache(self, layer_idx: int, bucket_len=None):
+        return self.k_caches[layer_idx], self.v_caches[layer_idx]
+
+    def update_cache(self, layer_idx: int, k_new, v_new, positions, seq_ids=None):
+        """Same contract as the linear manager: k_new/v_new (B,Hkv,S,D);
+        routes through the slot mapping installed by set_step_context."""
+        B, Hkv, S, D = k_new.shape
+        k_flat = k_new.transpose(1, 2).reshape(B * S, Hkv, D)
+        v_flat = v_new.transpose(1, 2).reshape(B * S, Hkv, D)
+        self.update(layer_idx, k_flat, v_flat, self._slot_mapping)
+
     def update(self, layer_idx: int, k_new: torch.Tensor, v_new: torch.Tensor,
                slot_mapping: torch.Tensor) -> None:
         """k_new/v_new: (N_tokens, Hkv, D); slot_mapping: (N_tokens,) flat
@@ -78,8 +100,13 @@ def generate_slot_mapping(block_table: torch.Tensor, positions: torch.Tensor,
                           block_size: int) -> torch.Tensor:
     """In-graph slot-mapping computation (ref
     `block_kv_cache_manager.py:376-430`): positions (B,S) + per-sequence
-    block table (B, max_blocks) -> flat slots (B*S,)."""
-    blk = positions // block_size
-    off = positions % block_size
-    blocks = torch.gather(block_table, 1, blk.long())
-    return (blocks * block_size + off).flatten()
+    block table (B, max_blocks) -> flat slots (B*S,). Negative positions
+    (padding) map to slot -1 (dropped by the scatter kernel)."""
+    valid = positions >= 0
+    pos = positions.clamp(min=0)
+    blk = (pos // block_size).clamp(max=block_table.shape[1] - 1)
+    off = pos % block_size
+    blocks = torch.gather(block_table.long(), 1, blk.long())
+    slots = blocks * block_size + off
+    slots = torch.where(valid & (blocks >= 0), slots, torch.full_like(slots, -1))
+    return slots.flatten()

@@ -255,3 +255,31 @@ def gelu_tanh_mul_fused(gate_up):
         return out
     g, u = gate_up.chunk(2, dim=-1)
     return ref.gelu_tanh_mul(g.contiguous(), u.contiguous())
+
+
+def attention_decode_paged(q, k_cache, v_cache, seq_lens, block_table,
+                           scale=None, sliding_window: int = 0):
+    """Decode attention against a paged (num_blocks, block_size, Hkv, D)
+    cache via a (B, max_blocks) block table."""
+    import math
+    scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+    if q.is_cuda:
+        c = _require_ext("attention_decode_paged")
+        out = torch.empty_like(q)
+        c.attention_decode_paged(
+            out, q.contiguous(), k_cache, v_cache,
+            seq_lens.to(torch.int32).contiguous(),
+            block_table.to(torch.int32).contiguous(), float(scale),
+            int(sliding_window),
+        )
+        return out
+    # CPU reference: gather contiguous K/V per sequence then mask-attend
+    max_len = int(seq_lens.max())
+    k, v = ref.paged_gather_kv(k_cache, v_cache, block_table, max_len)
+    B, Hq, Sq, D = q.shape
+    ar = torch.arange(max_len).view(1, 1, 1, -1)
+    qpos = seq_lens.view(B, 1, 1, 1) - (Sq - 1) + torch.arange(Sq).view(1, 1, Sq, 1)
+    mask = ar <= (qpos - 1)
+    if sliding_window:
+        mask = mask & (ar > (qpos - 1 - sliding_window))
+    return ref.sdpa(q, k, v, mask=mask, scale=scale)

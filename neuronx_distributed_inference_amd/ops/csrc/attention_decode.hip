@@ -21,7 +21,11 @@
 
 constexpr float M_INIT = -1e30f;
 
-template <typename T, int D, int GQ>
+// PAGED=false: linear cache (Bc, Hkv, Smax, D), row l at seq row seq_ids[b].
+// PAGED=true: paged cache (num_blocks, block_size, Hkv, D) walked through
+// block_table (B, max_blocks); Smax is the per-sequence capacity
+// max_blocks*block_size.
+template <typename T, int D, int GQ, bool PAGED>
 __global__ __launch_bounds__(256) void decode_attn_pass1(
     float* __restrict__ o_part,        // (B, Hq, Sq, nsplit, D)
     float* __restrict__ ml_part,       // (B, Hq, Sq, nsplit, 2)
@@ -29,10 +33,12 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
     const T* __restrict__ k_cache,     // (Bc, Hkv, Smax, D)
     const T* __restrict__ v_cache,
     const int* __restrict__ seq_lens,  // (B,)
-    const int* __restrict__ seq_ids,   // (B,)
+    const int* __restrict__ seq_ids,   // (B,) (linear) / unused (paged)
+    const int* __restrict__ block_table,  // (B, max_blocks) when PAGED
     const float scale, const int window,
     const int Hkv, const int Sq, const int gq,
-    const int Smax, const int chunk, const int nsplit) {
+    const int Smax, const int chunk, const int nsplit,
+    const int block_size, const int max_blocks) {
     using VT = VecTraits<T>;
     constexpr int GS = D / 8;          // lanes per group
     constexpr int NG = 256 / GS;       // groups per workgroup
@@ -77,15 +83,23 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
     }
 
     // ---- stream K/V rows ----
-    if (row >= 0 && split_lo < split_hi) {
-        const T* kbase = k_cache + (((long)row * Hkv + hkv) * Smax) * D;
-        const T* vbase = v_cache + (((long)row * Hkv + hkv) * Smax) * D;
+    if ((PAGED || row >= 0) && split_lo < split_hi) {
+        const T* kbase = PAGED ? k_cache : k_cache + (((long)row * Hkv + hkv) * Smax) * D;
+        const T* vbase = PAGED ? v_cache : v_cache + (((long)row * Hkv + hkv) * Smax) * D;
         for (int l = split_lo + grp; l < split_hi; l += NG) {
-            Vec8<T> kv = load_vec8(kbase + (long)l * D + c * 8);
+            long off;
+            if constexpr (PAGED) {
+                const int blk = block_table[(long)b * max_blocks + l / block_size];
+                if (blk < 0) continue;
+                off = (((long)blk * block_size + l % block_size) * Hkv + hkv) * D + c * 8;
+            } else {
+                off = (long)l * D + c * 8;
+            }
+            Vec8<T> kv = load_vec8(kbase + off);
             float kf[8];
 #pragma unroll
             for (int j = 0; j < 8; ++j) kf[j] = VT::to_float(kv.v[j]);
-            Vec8<T> vv = load_vec8(vbase + (long)l * D + c * 8);
+            Vec8<T> vv = load_vec8(vbase + off);
 
             float s[GQ];
 #pragma unroll
@@ -213,9 +227,14 @@ template <typename T, int D, int GQ>
 static void launch_decode(torch::Tensor& out, const torch::Tensor& q,
                           const torch::Tensor& k_cache, const torch::Tensor& v_cache,
                           const torch::Tensor& seq_lens, const torch::Tensor& seq_ids,
-                          float scale, int window, int gq) {
+                          float scale, int window, int gq,
+                          const torch::Tensor* block_table = nullptr,
+                          int block_size = 0) {
     const int B = q.size(0), Hq = q.size(1), Sq = q.size(2);
-    const int Hkv = k_cache.size(1), Smax = k_cache.size(2);
+    const bool paged = block_table != nullptr;
+    const int Hkv = paged ? k_cache.size(2) : k_cache.size(1);
+    const int max_blocks = paged ? (int)block_table->size(1) : 0;
+    const int Smax = paged ? max_blocks * block_size : (int)k_cache.size(2);
     // split count: fill the chip (>=1024 workgroups when possible), chunks of
     // >=128 rows so per-split work amortizes.
     int nsplit = 1024 / std::max(1, B * Hkv * Sq);
@@ -228,12 +247,22 @@ static void launch_decode(torch::Tensor& out, const torch::Tensor& q,
     auto stream = at::hip::getCurrentHIPStream();
 
     dim3 grid1(nsplit, Hkv * Sq, B);
-    hipLaunchKernelGGL((decode_attn_pass1<T, D, GQ>), grid1, dim3(256), 0, stream,
-                       o_part.data_ptr<float>(), ml_part.data_ptr<float>(),
-                       (const T*)q.data_ptr(), (const T*)k_cache.data_ptr(),
-                       (const T*)v_cache.data_ptr(), seq_lens.data_ptr<int>(),
-                       seq_ids.data_ptr<int>(), scale, window, Hkv, Sq, gq,
-                       Smax, chunk, nsplit);
+    if (paged) {
+        hipLaunchKernelGGL((decode_attn_pass1<T, D, GQ, true>), grid1, dim3(256), 0, stream,
+                           o_part.data_ptr<float>(), ml_part.data_ptr<float>(),
+                           (const T*)q.data_ptr(), (const T*)k_cache.data_ptr(),
+                           (const T*)v_cache.data_ptr(), seq_lens.data_ptr<int>(),
+                           seq_ids.data_ptr<int>(), block_table->data_ptr<int>(),
+                           scale, window, Hkv, Sq, gq, Smax, chunk, nsplit,
+                           block_size, max_blocks);
+    } else {
+        hipLaunchKernelGGL((decode_attn_pass1<T, D, GQ, false>), grid1, dim3(256), 0, stream,
+                           o_part.data_ptr<float>(), ml_part.data_ptr<float>(),
+                           (const T*)q.data_ptr(), (const T*)k_cache.data_ptr(),
+                           (const T*)v_cache.data_ptr(), seq_lens.data_ptr<int>(),
+                           seq_ids.data_ptr<int>(), nullptr,
+                           scale, window, Hkv, Sq, gq, Smax, chunk, nsplit, 0, 0);
+    }
     HIP_CHECK_KERNEL();
     dim3 grid2(B * Hq * Sq);
     hipLaunchKernelGGL((decode_attn_pass2<T, D>), grid2, dim3(WAVE_SIZE), 0, stream,
@@ -246,17 +275,55 @@ template <typename T, int D>
 static void dispatch_gq(torch::Tensor& out, const torch::Tensor& q,
                         const torch::Tensor& k_cache, const torch::Tensor& v_cache,
                         const torch::Tensor& seq_lens, const torch::Tensor& seq_ids,
-                        float scale, int window, int gq) {
+                        float scale, int window, int gq,
+                        const torch::Tensor* bt = nullptr, int bs = 0) {
     if (gq <= 1)
-        launch_decode<T, D, 1>(out, q, k_cache, v_cache, seq_lens, seq_ids, scale, window, gq);
+        launch_decode<T, D, 1>(out, q, k_cache, v_cache, seq_lens, seq_ids, scale, window, gq, bt, bs);
     else if (gq <= 2)
-        launch_decode<T, D, 2>(out, q, k_cache, v_cache, seq_lens, seq_ids, scale, window, gq);
+        launch_decode<T, D, 2>(out, q, k_cache, v_cache, seq_lens, seq_ids, scale, window, gq, bt, bs);
     else if (gq <= 4)
-        launch_decode<T, D, 4>(out, q, k_cache, v_cache, seq_lens, seq_ids, scale, window, gq);
+        launch_decode<T, D, 4>(out, q, k_cache, v_cache, seq_lens, seq_ids, scale, window, gq, bt, bs);
     else if (gq <= 8)
-        launch_decode<T, D, 8>(out, q, k_cache, v_cache, seq_lens, seq_ids, scale, window, gq);
+        launch_decode<T, D, 8>(out, q, k_cache, v_cache, seq_lens, seq_ids, scale, window, gq, bt, bs);
     else
         TORCH_CHECK(false, "GQA ratio > 8 not supported yet (got ", gq, ")");
+}
+
+void attention_decode_paged(torch::Tensor out, torch::Tensor q,
+                            torch::Tensor k_cache, torch::Tensor v_cache,
+                            torch::Tensor seq_lens, torch::Tensor block_table,
+                            double scale, long window) {
+    // paged cache (num_blocks, block_size, Hkv, D)
+    TORCH_CHECK(q.dim() == 4 && k_cache.dim() == 4, "bad shapes");
+    const int D = q.size(3);
+    const int Hq = q.size(1), Hkv = k_cache.size(2);
+    const int bs = k_cache.size(1);
+    TORCH_CHECK(Hq % Hkv == 0, "Hq must be a multiple of Hkv");
+    const int gq = Hq / Hkv;
+    auto seq_ids = seq_lens;  // unused in the paged path
+
+#define CASE_P(T)                                                               \
+    do {                                                                        \
+        if (D == 128)                                                           \
+            dispatch_gq<T, 128>(out, q, k_cache, v_cache, seq_lens, seq_ids,    \
+                                (float)scale, (int)window, gq, &block_table, bs); \
+        else if (D == 64)                                                       \
+            dispatch_gq<T, 64>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
+                               (float)scale, (int)window, gq, &block_table, bs); \
+        else                                                                    \
+            TORCH_CHECK(false, "paged decode: head_dim 64/128 only");           \
+    } while (0)
+
+    if (q.scalar_type() == at::ScalarType::BFloat16) {
+        CASE_P(__hip_bfloat16);
+    } else if (q.scalar_type() == at::ScalarType::Half) {
+        CASE_P(__half);
+    } else if (q.scalar_type() == at::ScalarType::Float) {
+        CASE_P(float);
+    } else {
+        TORCH_CHECK(false, "attention_decode_paged: unsupported dtype");
+    }
+#undef CASE_P
 }
 
 void attention_decode(torch::Tensor out, torch::Tensor q,

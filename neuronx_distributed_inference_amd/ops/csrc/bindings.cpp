@@ -14,6 +14,9 @@ void block_kv_cache_update(torch::Tensor k_cache, torch::Tensor v_cache, torch::
 void attention_decode(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
                       torch::Tensor v_cache, torch::Tensor seq_lens, torch::Tensor seq_ids,
                       double scale, long window);
+void attention_decode_paged(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
+                            torch::Tensor v_cache, torch::Tensor seq_lens,
+                            torch::Tensor block_table, double scale, long window);
 void topk_topp_sample(torch::Tensor out, torch::Tensor vals, torch::Tensor top_k,
                       torch::Tensor top_p, torch::Tensor temp, torch::Tensor rand,
                       long global_topk);
@@ -34,6 +37,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("kv_cache_update", &kv_cache_update, "linear KV cache scatter (CDNA4)");
     m.def("block_kv_cache_update", &block_kv_cache_update, "paged KV cache scatter (CDNA4)");
     m.def("attention_decode", &attention_decode, "split-KV decode attention (CDNA4)");
+    m.def("attention_decode_paged", &attention_decode_paged, "paged split-KV decode attention (CDNA4)");
     m.def("topk_topp_sample", &topk_topp_sample, "fused top-k/top-p sampling (CDNA4)");
     m.def("flash_prefill", &flash_prefill, "MFMA causal flash prefill (CDNA4)");
     m.def("qkv_rope_update", &qkv_rope_update, "fused qkv->rope->cache-write (CDNA4)");

@@ -309,3 +309,22 @@ def test_silu_mul_fused_gpu():
     out = ops.silu_mul_fused(y)
     want = ops.silu_mul_fused(y.float().cpu())
     torch.testing.assert_close(out.float().cpu(), want, atol=2e-2, rtol=2e-2)
+
+
+def test_attention_decode_paged_gpu():
+    torch.manual_seed(0)
+    B, Hkv, gq, D = 2, 2, 4, 128
+    nb, bs = 32, 16
+    dev = _dev()
+    kc = torch.randn(nb, bs, Hkv, D, dtype=torch.bfloat16, device=dev)
+    vc = torch.randn_like(kc)
+    q = torch.randn(B, Hkv * gq, 1, D, dtype=torch.bfloat16, device=dev)
+    bt = torch.stack([
+        torch.arange(0, 16, 2), torch.arange(1, 17, 2),
+    ]).to(dev)
+    seq_lens = torch.tensor([37, 100], device=dev)
+    out = ops.attention_decode_paged(q, kc, vc, seq_lens, bt)
+    want = ops.attention_decode_paged(
+        q.float().cpu(), kc.float().cpu(), vc.float().cpu(), seq_lens.cpu(), bt.cpu()
+    )
+    torch.testing.assert_close(out.float().cpu(), want, atol=3e-2, rtol=3e-2)
