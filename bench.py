@@ -25,6 +25,9 @@ import torch.distributed as dist
 MODELS = {
     "tiny": dict(hidden_size=256, intermediate_size=512, num_hidden_layers=2,
                  num_attention_heads=8, num_key_value_heads=4, vocab_size=1024),
+    "1b": dict(hidden_size=2048, intermediate_size=8192, num_hidden_layers=16,
+               num_attention_heads=32, num_key_value_heads=8, vocab_size=128256,
+               head_dim=64),
     "8b": dict(hidden_size=4096, intermediate_size=14336, num_hidden_layers=32,
                num_attention_heads=32, num_key_value_heads=8, vocab_size=128256),
     "70b": dict(hidden_size=8192, intermediate_size=28672, num_hidden_layers=80,
@@ -41,11 +44,116 @@ def parse_args():
     p.add_argument("--batch-per-gpu", type=int, default=64)
     p.add_argument("--ctx", type=int, default=128)
     p.add_argument("--no-graphs", action="store_true")
+    p.add_argument("--spec", action="store_true",
+                   help="fused speculative decoding: --model target + 1b draft, spec-len 5")
+    p.add_argument("--spec-len", type=int, default=5)
     return p.parse_args()
+
+
+def run_spec_bench(args):
+    """Fused speculation round benchmark (BASELINE config 5: 8B target +
+    1B draft, spec-len 5). Random-init weights make draft/target agreement
+    chance-level, so tokens/s here measures the worst-case (~1 token/round);
+    round latency is the hardware-meaningful number."""
+    import torch
+
+    on_gpu = torch.cuda.is_available()
+    device = torch.device("cuda:0" if on_gpu else "cpu")
+    from neuronx_distributed_inference_amd.config import (
+        EngineConfig, OnDeviceSamplingConfig,
+    )
+    from neuronx_distributed_inference_amd.models.llama.modeling_llama import (
+        LlamaForCausalLM, LlamaInferenceConfig,
+    )
+    from neuronx_distributed_inference_amd.modules.speculation import (
+        FusedSpeculationModel,
+    )
+    from neuronx_distributed_inference_amd.runtime.graph_engine import GraphRunner
+
+    steps, warmup = args.steps, args.warmup
+    B = args.batch_per_gpu
+    total_len = args.ctx + (steps + warmup + 2) * (args.spec_len + 1)
+    dtype = "bfloat16" if on_gpu else "float32"
+
+    def build(spec_name):
+        ec = EngineConfig(
+            batch_size=B, max_batch_size=B, seq_len=total_len, max_length=total_len,
+            torch_dtype=dtype, tp_degree=1,
+            on_device_sampling_config=OnDeviceSamplingConfig(top_k=1),
+            fused_qkv=True, fused_gate_up=True,
+        )
+        cfg = LlamaInferenceConfig(
+            neuron_config=ec, max_position_embeddings=total_len * 2,
+            rms_norm_eps=1e-5, rope_theta=500000.0, hidden_act="silu",
+            tie_word_embeddings=False, **MODELS[spec_name],
+        )
+        torch.manual_seed(7)
+        m = LlamaForCausalLM(cfg).eval()
+        if on_gpu:
+            m = m.to(device)
+        with torch.no_grad():
+            for p in m.parameters():
+                if p.dim() >= 2:
+                    p.normal_(0, 0.02)
+        m.reset()
+        return m
+
+    target = build(args.model)
+    draft = build("1b" if args.model != "tiny" else "tiny")
+    fused = FusedSpeculationModel(target.model, draft.model, args.spec_len)
+
+    ids = torch.randint(0, MODELS[args.model]["vocab_size"], (B, args.ctx), device=device)
+    seq_ids = torch.arange(B, device=device)
+    pos = torch.arange(args.ctx, device=device).view(1, -1).expand(B, -1)
+    with torch.inference_mode():
+        t_out = target(input_ids=ids, position_ids=pos, seq_ids=seq_ids)
+        draft(input_ids=ids, position_ids=pos, seq_ids=seq_ids)
+        tok = t_out.tokens.view(B, 1)
+        lengths = torch.full((B, 1), args.ctx, device=device, dtype=torch.long)
+
+        runner = fused
+        if on_gpu and not args.no_graphs:
+            runner = GraphRunner(fused, (tok, lengths, seq_ids))
+
+        accepted_total = 0
+        def round_(tok, lengths, accepted_total):
+            acc, n = runner(tok, lengths, seq_ids)
+            lengths = lengths + n.view(B, 1) + 1
+            tok = acc.gather(1, n.view(B, 1))
+            return tok, lengths, accepted_total + int(n.sum()) + B
+
+        for _ in range(warmup):
+            tok, lengths, _ = round_(tok, lengths, 0)
+        if on_gpu:
+            torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(steps):
+            tok, lengths, accepted_total = round_(tok, lengths, accepted_total)
+        if on_gpu:
+            torch.cuda.synchronize()
+        elapsed = time.perf_counter() - t0
+
+    print(json.dumps({
+        "metric": "output tokens/sec (node), fused speculation 8B+1B spec-5",
+        "value": round(accepted_total / elapsed, 2),
+        "unit": "tokens/s",
+        "n_gpus": 1, "steps": steps, "warmup": warmup,
+        "ms_per_step": round(elapsed / steps * 1000.0, 3),
+        "higher_is_better": True, "scaling": "weak", "vs_baseline": None,
+        "dtype": dtype, "data": "synthetic",
+        "config": {
+            "model": f"llama-3.1-{args.model} + llama-3.2-1b draft",
+            "global_batch": B, "ctx": args.ctx, "speculation_length": args.spec_len,
+            "parallelism": "tp1", "weights": "random-init",
+            "note": "random weights => chance-level acceptance; ms_per_step is one spec round",
+        },
+    }))
 
 
 def main():
     args = parse_args()
+    if args.spec:
+        return run_spec_bench(args)
     world = int(os.environ.get("WORLD_SIZE", 1))
     rank = int(os.environ.get("RANK", 0))
     local_rank = int(os.environ.get("LOCAL_RANK", 0))

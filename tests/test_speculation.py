@@ -99,3 +99,20 @@ def test_spec_through_submodel_wrappers():
         ids, generation_config=GenerationConfig(max_new_tokens=16))
     n = min(direct.shape[1], wrapped.shape[1])
     assert torch.equal(direct[:, :n], wrapped[:, :n])
+
+
+@torch.inference_mode()
+def test_fused_spec_matches_target_only():
+    from neuronx_distributed_inference_amd.modules.speculation import (
+        FusedSpeculativeAdapter,
+    )
+
+    target = _build(64, 2, seed=1, spec_len=4)
+    draft = _build(32, 1, seed=2)
+    ids = torch.randint(0, 64, (2, 8))
+    want = HuggingFaceGenerationAdapter(_build(64, 2, seed=1)).generate(
+        ids, generation_config=GenerationConfig(max_new_tokens=24))
+    got = FusedSpeculativeAdapter(target, draft, 4, use_graphs=False).generate(
+        ids, generation_config=GenerationConfig(max_new_tokens=24))
+    n = min(got.shape[1], want.shape[1])
+    assert torch.equal(got[:, :n], want[:, :n]), (got.tolist(), want.tolist())
