@@ -113,7 +113,13 @@ def run_spec_bench(args):
 
         runner = fused
         if on_gpu and not args.no_graphs:
-            runner = GraphRunner(fused, (tok, lengths, seq_ids))
+            try:
+                runner = GraphRunner(fused, (tok, lengths, seq_ids))
+            except Exception:  # noqa: BLE001
+                import traceback
+                traceback.print_exc()
+                torch.cuda.synchronize()
+                runner = fused
 
         accepted_total = 0
         def round_(tok, lengths, accepted_total):

@@ -87,6 +87,10 @@ class Sampler(nn.Module):
         if logits.dim() == 3:
             logits = logits[:, -1, :]
         B = logits.shape[0]
+        if self.static_greedy:
+            # no sampling-params tensor needed (and building one would do a
+            # host->device copy, which is illegal inside graph capture)
+            return pops.distributed_argmax(logits, dim=-1)
         if sampling_params is None:
             sampling_params = prepare_sampling_params(
                 B,
@@ -96,8 +100,6 @@ class Sampler(nn.Module):
                 device=logits.device,
             )
         top_k = sampling_params[:, 0]
-        if self.static_greedy:
-            return pops.distributed_argmax(logits, dim=-1)
 
         # multinomial: distributed top-global_topk then fused sampling
         k = min(self.config.global_topk, self.vocab_size)
