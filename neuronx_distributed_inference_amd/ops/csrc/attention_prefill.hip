@@ -43,7 +43,7 @@ template <> struct MfmaTraits<__half> {
 
 constexpr int QBLK = 128;       // q rows per workgroup
 constexpr int WQ = 32;          // q rows per wave
-constexpr int KVBLK = 32;       // keys per LDS tile
+constexpr int KVBLK = 64;       // keys per LDS tile
 constexpr float NEG_INF = -1e30f;
 
 template <typename T, int D>
@@ -62,6 +62,13 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
     constexpr int K_STRIDE = D * 2 + 16;
     constexpr int VT_STRIDE = KVBLK * 2 + 16;
     constexpr int P_STRIDE = KVBLK * 2 + 16;
+    // XOR swizzle on the V^T image in-row offset, keyed by the V^T row (the
+    // head-dim d). The transpose staging writes one b16 per (d, key) from 16
+    // threads whose d differ by 8 at byte stride VT_STRIDE (a multiple of
+    // 16B) — all on one bank without this (measured: SQ_LDS_BANK_CONFLICT
+    // 2.5x SQ_BUSY_CYCLES). The swizzle spreads them over 8 bank slots; the
+    // b128 PV reads stay 16B-aligned and at worst 2-way.
+#define VT_SWZ(row) ((((row) >> 3) & 7) << 4)
 
     __shared__ __attribute__((aligned(16))) char smem[
         KVBLK * K_STRIDE + D * VT_STRIDE + 4 * WQ * P_STRIDE];
@@ -131,17 +138,19 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
             Vec8<T> vv8 = load_vec8(v + kv_base + (long)krow * D + d);
 #pragma unroll
             for (int j = 0; j < 8; ++j)  // transpose: V^T[d+j][row]
-                *reinterpret_cast<T*>(vt_lds + (d + j) * VT_STRIDE + row * 2) = vv8.v[j];
+                *reinterpret_cast<T*>(
+                    vt_lds + (d + j) * VT_STRIDE + ((row * 2) ^ VT_SWZ(d + j))) = vv8.v[j];
         }
         __syncthreads();
 
         if (k0 <= wq_hi) {  // this wave still has un-masked keys in the tile
-            // ---- QK^T: S_tile[32q][32k] as 2x2 fragments ----
-            f32x4 s_frag[2][2];
+            // ---- QK^T: S_tile[32q][KVBLK k] as 2 x KVBLK/16 fragments ----
+            constexpr int KI = KVBLK / 16;
+            f32x4 s_frag[2][KI];
 #pragma unroll
             for (int qi = 0; qi < 2; ++qi)
 #pragma unroll
-                for (int ki = 0; ki < 2; ++ki) {
+                for (int ki = 0; ki < KI; ++ki) {
                     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
                     for (int ks = 0; ks < KS; ++ks) {
@@ -156,6 +165,12 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
 
             // ---- mask + online softmax (C layout: col=lane&15 is the key,
             // row=(lane>>4)*4+reg is the q) ----
+            // Tiles fully below the causal diagonal (and inside the window /
+            // sequence bounds) skip every per-element mask op — that is most
+            // tiles, and the masking VALU was a large share of wave cycles.
+            const bool full_tile =
+                (k0 + KVBLK - 1 <= wq_lo) && (k0 + KVBLK <= S)
+                && (window <= 0 || k0 >= wq_lo + WQ - window);
             float p_new[2][4];  // per-qi per-reg row max of this tile
 #pragma unroll
             for (int qi = 0; qi < 2; ++qi) {
@@ -163,8 +178,16 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
                 for (int r = 0; r < 4; ++r) {
                     const int q_abs = wq_lo + qi * 16 + (lane >> 4) * 4 + r;
                     float mx = NEG_INF;
+                    if (full_tile) {
 #pragma unroll
-                    for (int ki = 0; ki < 2; ++ki) {
+                        for (int ki = 0; ki < KI; ++ki) {
+                            const float sv = s_frag[qi][ki][r] * scale;
+                            s_frag[qi][ki][r] = sv;
+                            mx = fmaxf(mx, sv);
+                        }
+                    } else
+#pragma unroll
+                    for (int ki = 0; ki < KI; ++ki) {
                         const int k_abs = k0 + ki * 16 + (lane & 15);
                         float sv = s_frag[qi][ki][r] * scale;
                         const bool ok = (k_abs <= q_abs) && (k_abs < S) && (q_abs < S)
@@ -187,8 +210,16 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
                     const float m_new = fmaxf(m_run[qi][r], p_new[qi][r]);
                     const float corr = __expf(m_run[qi][r] - m_new);
                     float row_sum = 0.f;
+                    if (full_tile) {
 #pragma unroll
-                    for (int ki = 0; ki < 2; ++ki) {
+                        for (int ki = 0; ki < KI; ++ki) {
+                            const float p = __expf(s_frag[qi][ki][r] - m_new);
+                            s_frag[qi][ki][r] = p;
+                            row_sum += p;
+                        }
+                    } else
+#pragma unroll
+                    for (int ki = 0; ki < KI; ++ki) {
                         const float sv = s_frag[qi][ki][r];
                         const float p = (sv <= NEG_INF) ? 0.f : __expf(sv - m_new);
                         s_frag[qi][ki][r] = p;
@@ -210,7 +241,7 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
 #pragma unroll
             for (int qi = 0; qi < 2; ++qi)
 #pragma unroll
-                for (int ki = 0; ki < 2; ++ki)
+                for (int ki = 0; ki < KI; ++ki)
 #pragma unroll
                     for (int r = 0; r < 4; ++r) {
                         const int qrow = qi * 16 + (lane >> 4) * 4 + r;
@@ -223,16 +254,23 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
             // "memory" clobber keeps the compiler from hoisting the reads.
             asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
-            // ---- PV: O[32q][D] += P[32q][32k] * V[32k][D] ----
+            // ---- PV: O[32q][D] += P[32q][KVBLK] * V[KVBLK][D], in 32-key
+            // chunks (one A fragment per chunk) ----
 #pragma unroll
-            for (int qi = 0; qi < 2; ++qi) {
-                const vec8 pa = *reinterpret_cast<const vec8*>(
-                    p_lds + (qi * 16 + (lane & 15)) * P_STRIDE + ((lane >> 4) * 8) * 2);
+            for (int kc2 = 0; kc2 < KVBLK / 32; ++kc2) {
 #pragma unroll
-                for (int di = 0; di < DI; ++di) {
-                    const vec8 vb = *reinterpret_cast<const vec8*>(
-                        vt_lds + (di * 16 + (lane & 15)) * VT_STRIDE + ((lane >> 4) * 8) * 2);
-                    o_acc[qi][di] = MT::mfma(pa, vb, o_acc[qi][di]);
+                for (int qi = 0; qi < 2; ++qi) {
+                    const vec8 pa = *reinterpret_cast<const vec8*>(
+                        p_lds + (qi * 16 + (lane & 15)) * P_STRIDE +
+                        (kc2 * 32 + (lane >> 4) * 8) * 2);
+#pragma unroll
+                    for (int di = 0; di < DI; ++di) {
+                        const int vrow = di * 16 + (lane & 15);
+                        const vec8 vb = *reinterpret_cast<const vec8*>(
+                            vt_lds + vrow * VT_STRIDE +
+                            (((kc2 * 32 + (lane >> 4) * 8) * 2) ^ VT_SWZ(vrow)));
+                        o_acc[qi][di] = MT::mfma(pa, vb, o_acc[qi][di]);
+                    }
                 }
             }
         }
