@@ -47,7 +47,7 @@ constexpr int KVBLK = 64;       // keys per LDS tile
 constexpr float NEG_INF = -1e30f;
 
 template <typename T, int D>
-__global__ __launch_bounds__(256) void flash_prefill_kernel(
+__global__ __launch_bounds__(256, 2) void flash_prefill_kernel(
     T* __restrict__ out,       // (B, Hq, S, D)
     const T* __restrict__ q,   // (B, Hq, S, D)
     const T* __restrict__ k,   // (B, Hkv, S, D)
@@ -125,21 +125,29 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
     int kv_start = 0;
     if (window > 0) kv_start = max(0, (q_lo - window + 1) / KVBLK * KVBLK);
 
+    // Staging note: occupancy (not staging latency) is the binding
+    // constraint — keep live state minimal so 2 waves/SIMD fit (the
+    // register-prefetch variant measured null because it pushed the kernel
+    // to 329 regs = 1 wave/SIMD).
+    constexpr int NIT = (KVBLK * D) / (256 * 8);
+    const int st_row = (threadIdx.x * 8) / D;        // this thread's key row
+    const int st_d = (threadIdx.x * 8) % D;          // and d offset
+    const int row_step = (256 * 8) / D;
+
     for (int k0 = kv_start; k0 < kv_hi; k0 += KVBLK) {
-        // ---- stage K and V^T (whole workgroup) ----
-        __syncthreads();
-        for (int idx = threadIdx.x * 8; idx < KVBLK * D; idx += 256 * 8) {
-            const int row = idx / D;         // key within tile
-            const int d = idx % D;
+#pragma unroll
+        for (int it = 0; it < NIT; ++it) {
+            const int row = st_row + it * row_step;
             const int krow = min(k0 + row, S - 1);
-            Vec8<T> kv8 = load_vec8(k + kv_base + (long)krow * D + d);
-            *reinterpret_cast<int4*>(k_lds + row * K_STRIDE + d * 2) =
+            Vec8<T> kv8 = load_vec8(k + kv_base + (long)krow * D + st_d);
+            Vec8<T> vv8 = load_vec8(v + kv_base + (long)krow * D + st_d);
+            *reinterpret_cast<int4*>(k_lds + row * K_STRIDE + st_d * 2) =
                 *reinterpret_cast<int4*>(kv8.v);
-            Vec8<T> vv8 = load_vec8(v + kv_base + (long)krow * D + d);
 #pragma unroll
             for (int j = 0; j < 8; ++j)  // transpose: V^T[d+j][row]
                 *reinterpret_cast<T*>(
-                    vt_lds + (d + j) * VT_STRIDE + ((row * 2) ^ VT_SWZ(d + j))) = vv8.v[j];
+                    vt_lds + (st_d + j) * VT_STRIDE + ((row * 2) ^ VT_SWZ(st_d + j))) =
+                    vv8.v[j];
         }
         __syncthreads();
 
@@ -274,6 +282,7 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
                 }
             }
         }
+        __syncthreads();  // compute done before the next write_tile overwrites
     }
 
     // ---- epilogue: normalize and store ----
