@@ -94,7 +94,7 @@ class LlamaDecoderLayer(nn.Module):
         )
 
     def forward(self, h, residual, cos, sin, kv_mgr, position_ids, phase,
-                attention_mask, seq_ids):
+                attention_mask, seq_ids, sequence_parallel=False):
         # (h, residual) threading: the residual add is fused into the RMSNorm
         # kernel (one HBM pass instead of add + norm)
         if residual is None:
@@ -102,9 +102,10 @@ class LlamaDecoderLayer(nn.Module):
             h = self.input_layernorm(h)
         else:
             h, residual = self.input_layernorm(h, residual)
-        h = self.self_attn(h, cos, sin, kv_mgr, position_ids, phase, attention_mask, seq_ids)
+        h = self.self_attn(h, cos, sin, kv_mgr, position_ids, phase,
+                           attention_mask, seq_ids, sequence_parallel=sequence_parallel)
         h, residual = self.post_attention_layernorm(h, residual)
-        h = self.mlp(h)
+        h = self.mlp(h, sequence_parallel=sequence_parallel)
         return h, residual
 
 

@@ -161,18 +161,39 @@ class DecoderModel(nn.Module):
 
             slots = generate_slot_mapping(block_table, position_ids, self.kv_mgr.block_size)
             self.kv_mgr.set_step_context(block_table, slots)
-        h = self.embed_tokens(input_ids)
+        from neuronx_distributed_inference_amd.parallel import state as _ps
+
+        tp = _ps.get_tensor_model_parallel_size()
+        # SP: prefill activations sharded along S (ref §2.8 SP; disabled for
+        # TKG exactly like the reference, model_base.py:3093,3135)
+        sp_active = (
+            self.ec.sequence_parallel_enabled
+            and phase == PHASE_CTE and tp > 1 and S % tp == 0 and S >= tp
+        )
+        h = self.embed_tokens(input_ids, sequence_parallel=sp_active) \
+            if sp_active else self.embed_tokens(input_ids)
         cos, sin = self.rotary.get_cos_sin(position_ids, device=h.device)
         residual = None
         for layer in self.layers:
-            h, residual = layer(
-                h, residual, cos, sin, self.kv_mgr, position_ids, phase,
-                attention_mask, seq_ids,
-            )
+            if sp_active:
+                h, residual = layer(
+                    h, residual, cos, sin, self.kv_mgr, position_ids, phase,
+                    attention_mask, seq_ids, sequence_parallel=True,
+                )
+            else:
+                h, residual = layer(
+                    h, residual, cos, sin, self.kv_mgr, position_ids, phase,
+                    attention_mask, seq_ids,
+                )
         if residual is not None:
             h, _ = self.norm(h, residual)
         else:
             h = self.norm(h)
+        if sp_active:
+            # exit SP before the last-token gather (ref
+            # seq_parallel_logits_slice.py; full gather kept simple here —
+            # prefill-only, once per request)
+            h = mappings.gather_along_dim(h, dim=1)
 
         is_speculation = phase == PHASE_TKG and S > 1
         if phase == PHASE_CTE and S > 1:

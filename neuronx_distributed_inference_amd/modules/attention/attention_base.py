@@ -70,12 +70,13 @@ class AttentionBase(nn.Module):
 
     # ------------------------------------------------------------------
     def prep_qkv(
-        self, hidden: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor
+        self, hidden: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+        sequence_parallel: bool = False,
     ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
         """Projections + BSHD->BHSD + optional qk-norm + RoPE
         (ref prep_qkv_tensors attention_base.py:489-563)."""
-        B, S, _ = hidden.shape
-        q, k, v = self.qkv_proj(hidden)
+        q, k, v = self.qkv_proj(hidden, sequence_parallel=sequence_parallel)
+        B, S = q.shape[0], q.shape[1]
         if self.clip_qkv is not None:  # DBRX clamp (ref modeling_dbrx.py)
             q = q.clamp(-self.clip_qkv, self.clip_qkv)
             k = k.clamp(-self.clip_qkv, self.clip_qkv)
@@ -106,8 +107,11 @@ class AttentionBase(nn.Module):
         attention_mask: Optional[torch.Tensor] = None,
         seq_ids: Optional[torch.Tensor] = None,
         active_mask: Optional[torch.Tensor] = None,
+        sequence_parallel: bool = False,
     ) -> torch.Tensor:
         B, S, _ = hidden.shape
+        if sequence_parallel:
+            S = S * self.tp_degree  # hidden is S-sharded; attention runs full-S
 
         # Fused TKG front end (K2-analog): one GEMM -> one kernel doing
         # RoPE + cache write + q relayout. Skipped when per-head qk-norms or
@@ -134,7 +138,7 @@ class AttentionBase(nn.Module):
             out = out.transpose(1, 2).reshape(B, S, -1)
             return self.o_proj(out)
 
-        q, k, v = self.prep_qkv(hidden, cos, sin)
+        q, k, v = self.prep_qkv(hidden, cos, sin, sequence_parallel=sequence_parallel)
 
         # write this step's K/V into the cache first (in-place scatter)
         kv_mgr.update_cache(self.layer_idx, k, v, position_ids, seq_ids)
@@ -145,7 +149,7 @@ class AttentionBase(nn.Module):
             out = self._decode(q, kv_mgr, position_ids, seq_ids, active_mask)
 
         out = out.transpose(1, 2).reshape(B, S, -1)  # BHSD -> BS(H*D)
-        return self.o_proj(out)
+        return self.o_proj(out, sequence_parallel=sequence_parallel)
 
     # ------------------------------------------------------------------
     def _prefill(self, q, k, v, attention_mask) -> torch.Tensor:

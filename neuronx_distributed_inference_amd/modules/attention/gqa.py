@@ -162,11 +162,14 @@ class GroupQueryAttentionQKV(nn.Module):
                 for p in params:
                     p.preshard_fn = lambda w, tp, hd=hd, rep=rep: _replicate_head_blocks(w, hd, rep)
 
-    def forward(self, x: torch.Tensor):
+    def forward(self, x: torch.Tensor, sequence_parallel: bool = False):
+        sp = sequence_parallel
         if self.fused:
-            y = self.qkv(x)
+            y = self.qkv(x, sequence_parallel=sp)
             return torch.split(y, self._split_sizes, dim=-1)
-        return self.q_proj(x), self.k_proj(x), self.v_proj(x)
+        return (self.q_proj(x, sequence_parallel=sp),
+                self.k_proj(x, sequence_parallel=sp),
+                self.v_proj(x, sequence_parallel=sp))
 
 
 class GroupQueryAttentionO(nn.Module):
@@ -202,5 +205,5 @@ class GroupQueryAttentionO(nn.Module):
 
             self.o_proj.weight.preshard_fn = _pad_in
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return self.o_proj(x)
+    def forward(self, x: torch.Tensor, sequence_parallel: bool = False) -> torch.Tensor:
+        return self.o_proj(x, sequence_parallel=sequence_parallel)

@@ -75,7 +75,11 @@ class GatedMLP(nn.Module):
         else:
             raise ValueError(f"unsupported hidden_act {hidden_act}")
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, sequence_parallel: bool = False) -> torch.Tensor:
+        sp = sequence_parallel
         if self.fused:
-            return self.down_proj(self.act_mul_fused(self.gate_up_proj(x)))
-        return self.down_proj(self.act_mul(self.gate_proj(x), self.up_proj(x)))
+            y = self.gate_up_proj(x, sequence_parallel=sp)
+            return self.down_proj(self.act_mul_fused(y), sequence_parallel=sp)
+        g = self.gate_proj(x, sequence_parallel=sp)
+        u = self.up_proj(x, sequence_parallel=sp)
+        return self.down_proj(self.act_mul(g, u), sequence_parallel=sp)

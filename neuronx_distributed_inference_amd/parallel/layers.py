@@ -54,11 +54,16 @@ class ColumnParallelLinear(nn.Module):
         tp_degree: Optional[int] = None,
         group=None,
         stride: int = 1,
+        sequence_parallel: bool = False,
     ):
         super().__init__()
         self.input_size = input_size
         self.output_size = output_size
         self.gather_output = gather_output
+        # SP: input arrives sharded along the sequence dim (dim 1 of B,S,H);
+        # all-gather it before the GEMM (ref C3, gather_from_sequence_parallel
+        # region before QKV/MLP GEMMs, modeling_llama.py:677-679)
+        self.sequence_parallel = sequence_parallel
         self._group = group
         self.tp_degree = tp_degree or (
             ps.get_tensor_model_parallel_size() if group is None else 1
@@ -82,7 +87,10 @@ class ColumnParallelLinear(nn.Module):
     def group(self):
         return self._group if self._group is not None else ps.get_tensor_model_parallel_group()
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, sequence_parallel: Optional[bool] = None) -> torch.Tensor:
+        sp = self.sequence_parallel if sequence_parallel is None else sequence_parallel
+        if sp:
+            x = mappings.gather_along_dim(x, dim=1, group=self.group)
         y = F.linear(x, self.weight, self.bias)
         if self.gather_output:
             y = mappings.gather_from_tensor_parallel_region(y, dim=-1, group=self.group)
@@ -134,12 +142,17 @@ class RowParallelLinear(nn.Module):
     def group(self):
         return self._group if self._group is not None else ps.get_tensor_model_parallel_group()
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, sequence_parallel: bool = False) -> torch.Tensor:
         if not self.input_is_parallel:
             x = mappings.scatter_to_tensor_parallel_region(x, dim=-1, group=self.group)
         y = F.linear(x, self.weight)
         if self.reduce_output:
-            y = mappings.reduce_from_tensor_parallel_region(y, group=self.group)
+            if sequence_parallel:
+                # SP exit: reduce-scatter along S instead of all-reduce
+                # (ref C2, attention_base.py:1350)
+                y = mappings.reduce_scatter_along_dim(y, dim=1, group=self.group)
+            else:
+                y = mappings.reduce_from_tensor_parallel_region(y, group=self.group)
         if self.bias is not None:
             y = y + self.bias
         return y
@@ -178,7 +191,7 @@ class ParallelEmbedding(nn.Module):
             mark_parallel(self.weight, partition_dim=0, num_partitions=self.tp_degree)
             self.vocab_start = None  # resolved at forward from the live rank
 
-    def forward(self, ids: torch.Tensor) -> torch.Tensor:
+    def forward(self, ids: torch.Tensor, sequence_parallel: bool = False) -> torch.Tensor:
         if self.shard_along_embedding:
             y = F.embedding(ids, self.weight)
             return mappings.gather_from_tensor_parallel_region(y, dim=-1)
@@ -191,4 +204,8 @@ class ParallelEmbedding(nn.Module):
         local = torch.where(in_shard, ids - start, torch.zeros_like(ids))
         y = F.embedding(local, self.weight)
         y = y * in_shard.unsqueeze(-1).to(y.dtype)
+        if sequence_parallel:
+            # SP entry: partial sums reduce-scattered along S (ref C2,
+            # model_base.py:1509-1514)
+            return mappings.reduce_scatter_along_dim(y, dim=1)
         return mappings.reduce_from_tensor_parallel_region(y)

@@ -191,3 +191,59 @@ def _worker_tp2_generation(rank):
 
 def test_tp2_generation_matches_tp1():
     spawn("_worker_tp2_generation")
+
+
+# ---------------------------------------------------------------------------
+def _worker_sp_generation(rank):
+    """Sequence-parallel prefill (TP=2) must match the non-SP TP=2 result."""
+    from neuronx_distributed_inference_amd.config import EngineConfig
+    from neuronx_distributed_inference_amd.models.llama.modeling_llama import (
+        LlamaForCausalLM,
+        LlamaInferenceConfig,
+    )
+    from neuronx_distributed_inference_amd.parallel import state as ps
+    from neuronx_distributed_inference_amd.utils.checkpoint_sharding import (
+        shard_state_dict_for_rank,
+    )
+    from neuronx_distributed_inference_amd.utils.hf_adapter import (
+        GenerationConfig,
+        HuggingFaceGenerationAdapter,
+    )
+
+    def make_cfg(tp, sp):
+        ec = EngineConfig(batch_size=2, max_batch_size=2, seq_len=64, max_length=64,
+                          torch_dtype="float32", tp_degree=tp,
+                          sequence_parallel_enabled=sp)
+        return LlamaInferenceConfig(
+            neuron_config=ec, vocab_size=128, hidden_size=64, intermediate_size=128,
+            num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+            max_position_embeddings=128, rms_norm_eps=1e-5, rope_theta=10000.0,
+            hidden_act="silu", tie_word_embeddings=False)
+
+    ps.destroy_model_parallel()
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(21)
+    full = LlamaForCausalLM(make_cfg(1, False)).eval()
+    with torch.no_grad():
+        for p in full.parameters():
+            if p.dim() >= 2:
+                p.normal_(0, 0.1)
+    ids = torch.randint(0, 128, (2, 8))  # 8 divisible by tp=2 -> SP active
+    want = HuggingFaceGenerationAdapter(full).generate(
+        ids, generation_config=GenerationConfig(max_new_tokens=6))
+
+    ps.destroy_model_parallel()
+    ps.initialize_model_parallel(tensor_model_parallel_size=WORLD)
+    sp_model = LlamaForCausalLM(make_cfg(WORLD, True)).eval()
+    full_sd = {k: v for k, v in full.model.state_dict().items()
+               if not k.startswith("kv_mgr.")}
+    shard = shard_state_dict_for_rank(full_sd, sp_model.model, rank, WORLD)
+    sp_model.model.load_state_dict(shard, strict=False)
+    got = HuggingFaceGenerationAdapter(sp_model).generate(
+        ids, generation_config=GenerationConfig(max_new_tokens=6))
+    assert torch.equal(got, want), (got.tolist(), want.tolist())
+    return True
+
+
+def test_sequence_parallel_generation():
+    spawn("_worker_sp_generation")
