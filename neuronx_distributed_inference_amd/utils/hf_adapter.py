@@ -90,14 +90,21 @@ class HuggingFaceGenerationAdapter:
         next_tokens = self._next_token(out, gc)
         lengths = cur_mask.sum(-1)  # (B,) true lengths
 
+        # eos handling: the `finished.all()` readback is the only per-step
+        # host sync; skip it entirely when no eos is requested (the analog of
+        # the reference's async mode hiding the token readback,
+        # modules/async_execution.py)
+        check_eos = gc.eos_token_id is not None
         while True:
             next_tokens = torch.where(
                 finished, torch.full_like(next_tokens, gc.pad_token_id), next_tokens
             )
             out_ids = torch.cat([out_ids, next_tokens.view(B, 1)], dim=-1)
-            if gc.eos_token_id is not None:
+            if check_eos:
                 finished |= next_tokens.view(-1) == gc.eos_token_id
-            if bool(finished.all()) or out_ids.shape[1] >= max_len:
+                if bool(finished.all()):
+                    break
+            if out_ids.shape[1] >= max_len:
                 break
             position_ids = lengths.view(B, 1).long()
             lengths = lengths + 1
