@@ -83,6 +83,10 @@ class DecoderModel(nn.Module):
         )
         head_dim = getattr(config, "head_dim", None) or config.hidden_size // config.num_attention_heads
         n_kv_heads = getattr(config, "num_key_value_heads", config.num_attention_heads)
+        # FP8 KV-cache quantization (ref kv_cache_manager.py:636-692): cache
+        # stored OCP e4m3, quantized on write and dequantized in the decode
+        # kernel — halves KV HBM traffic and doubles cache capacity
+        cache_dtype = torch.float8_e4m3fn if ec.kv_cache_quant else ec.torch_dtype
         if ec.is_block_kv_layout:
             from neuronx_distributed_inference_amd.modules.kvcache.block_kv_cache_manager import (
                 BlockKVCacheManager,
@@ -95,7 +99,7 @@ class DecoderModel(nn.Module):
             self.kv_mgr = BlockKVCacheManager(
                 num_layers=self.num_layers, num_kv_heads=n_kv_heads,
                 head_dim=head_dim, num_blocks=num_blocks,
-                block_size=ec.pa_block_size, dtype=ec.torch_dtype,
+                block_size=ec.pa_block_size, dtype=cache_dtype,
                 tp_degree=ec.tp_degree,
             )
         else:
@@ -105,7 +109,7 @@ class DecoderModel(nn.Module):
                 head_dim=head_dim,
                 max_len=ec.max_length,
                 batch_size=ec.kv_cache_batch_size,
-                dtype=ec.torch_dtype,
+                dtype=cache_dtype,
                 padding_rows=ec.kv_cache_padding_size,
                 tp_degree=ec.tp_degree,
             )

@@ -25,13 +25,16 @@ constexpr float M_INIT = -1e30f;
 // PAGED=true: paged cache (num_blocks, block_size, Hkv, D) walked through
 // block_table (B, max_blocks); Smax is the per-sequence capacity
 // max_blocks*block_size.
-template <typename T, int D, int GQ, bool PAGED>
+// CT = cache storage type (T itself, or fp8-e4m3 when the KV cache is
+// quantized; the k-scale is folded into the softmax scale by the host and
+// the v-scale applied in pass 2).
+template <typename T, typename CT, int D, int GQ, bool PAGED>
 __global__ __launch_bounds__(256) void decode_attn_pass1(
     float* __restrict__ o_part,        // (B, Hq, Sq, nsplit, D)
     float* __restrict__ ml_part,       // (B, Hq, Sq, nsplit, 2)
     const T* __restrict__ q,           // (B, Hq, Sq, D) — Hq = Hkv*gq
-    const T* __restrict__ k_cache,     // (Bc, Hkv, Smax, D)
-    const T* __restrict__ v_cache,
+    const CT* __restrict__ k_cache,    // (Bc, Hkv, Smax, D)
+    const CT* __restrict__ v_cache,
     const int* __restrict__ seq_lens,  // (B,)
     const int* __restrict__ seq_ids,   // (B,) (linear) / unused (paged)
     const int* __restrict__ block_table,  // (B, max_blocks) when PAGED
@@ -84,8 +87,8 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
 
     // ---- stream K/V rows ----
     if ((PAGED || row >= 0) && split_lo < split_hi) {
-        const T* kbase = PAGED ? k_cache : k_cache + (((long)row * Hkv + hkv) * Smax) * D;
-        const T* vbase = PAGED ? v_cache : v_cache + (((long)row * Hkv + hkv) * Smax) * D;
+        const CT* kbase = PAGED ? k_cache : k_cache + (((long)row * Hkv + hkv) * Smax) * D;
+        const CT* vbase = PAGED ? v_cache : v_cache + (((long)row * Hkv + hkv) * Smax) * D;
         for (int l = split_lo + grp; l < split_hi; l += NG) {
             long off;
             if constexpr (PAGED) {
@@ -95,11 +98,11 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
             } else {
                 off = (long)l * D + c * 8;
             }
-            Vec8<T> kv = load_vec8(kbase + off);
+            Vec8<CT> kv = load_vec8(kbase + off);
             float kf[8];
 #pragma unroll
-            for (int j = 0; j < 8; ++j) kf[j] = VT::to_float(kv.v[j]);
-            Vec8<T> vv = load_vec8(vbase + off);
+            for (int j = 0; j < 8; ++j) kf[j] = VecTraits<CT>::to_float(kv.v[j]);
+            Vec8<CT> vv = load_vec8(vbase + off);
 
             float s[GQ];
 #pragma unroll
@@ -120,7 +123,7 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
                 m[g] = mn;
 #pragma unroll
                 for (int j = 0; j < 8; ++j)
-                    o_acc[g][j] = o_acc[g][j] * corr + p * VT::to_float(vv.v[j]);
+                    o_acc[g][j] = o_acc[g][j] * corr + p * VecTraits<CT>::to_float(vv.v[j]);
             }
         }
     }
@@ -196,7 +199,7 @@ __global__ void decode_attn_pass2(
     T* __restrict__ out,               // (B, Hq, Sq, D)
     const float* __restrict__ o_part,  // (B, Hq, Sq, nsplit, D)
     const float* __restrict__ ml_part, // (B, Hq, Sq, nsplit, 2)
-    const int nsplit) {
+    const int nsplit, const float v_scale) {
     using VT = VecTraits<T>;
     const long rowid = blockIdx.x;  // over B*Hq*Sq
     const int lane = threadIdx.x;
@@ -210,7 +213,7 @@ __global__ void decode_attn_pass2(
         const float li = ml_part[(rowid * nsplit + i) * 2 + 1];
         L += li * __expf(mi - mstar);
     }
-    const float inv = (L > 0.f) ? 1.f / L : 0.f;
+    const float inv = (L > 0.f) ? v_scale / L : 0.f;
 #pragma unroll
     for (int d = lane; d < D; d += WAVE_SIZE) {
         float acc = 0.f;
@@ -223,13 +226,13 @@ __global__ void decode_attn_pass2(
 }
 
 // ---------------------------------------------------------------------------
-template <typename T, int D, int GQ>
+template <typename T, typename CT, int D, int GQ>
 static void launch_decode(torch::Tensor& out, const torch::Tensor& q,
                           const torch::Tensor& k_cache, const torch::Tensor& v_cache,
                           const torch::Tensor& seq_lens, const torch::Tensor& seq_ids,
                           float scale, int window, int gq,
                           const torch::Tensor* block_table = nullptr,
-                          int block_size = 0) {
+                          int block_size = 0, float v_scale = 1.0f) {
     const int B = q.size(0), Hq = q.size(1), Sq = q.size(2);
     const bool paged = block_table != nullptr;
     const int Hkv = paged ? k_cache.size(2) : k_cache.size(1);
@@ -248,18 +251,18 @@ static void launch_decode(torch::Tensor& out, const torch::Tensor& q,
 
     dim3 grid1(nsplit, Hkv * Sq, B);
     if (paged) {
-        hipLaunchKernelGGL((decode_attn_pass1<T, D, GQ, true>), grid1, dim3(256), 0, stream,
+        hipLaunchKernelGGL((decode_attn_pass1<T, CT, D, GQ, true>), grid1, dim3(256), 0, stream,
                            o_part.data_ptr<float>(), ml_part.data_ptr<float>(),
-                           (const T*)q.data_ptr(), (const T*)k_cache.data_ptr(),
-                           (const T*)v_cache.data_ptr(), seq_lens.data_ptr<int>(),
+                           (const T*)q.data_ptr(), (const CT*)k_cache.data_ptr(),
+                           (const CT*)v_cache.data_ptr(), seq_lens.data_ptr<int>(),
                            seq_ids.data_ptr<int>(), block_table->data_ptr<int>(),
                            scale, window, Hkv, Sq, gq, Smax, chunk, nsplit,
                            block_size, max_blocks);
     } else {
-        hipLaunchKernelGGL((decode_attn_pass1<T, D, GQ, false>), grid1, dim3(256), 0, stream,
+        hipLaunchKernelGGL((decode_attn_pass1<T, CT, D, GQ, false>), grid1, dim3(256), 0, stream,
                            o_part.data_ptr<float>(), ml_part.data_ptr<float>(),
-                           (const T*)q.data_ptr(), (const T*)k_cache.data_ptr(),
-                           (const T*)v_cache.data_ptr(), seq_lens.data_ptr<int>(),
+                           (const T*)q.data_ptr(), (const CT*)k_cache.data_ptr(),
+                           (const CT*)v_cache.data_ptr(), seq_lens.data_ptr<int>(),
                            seq_ids.data_ptr<int>(), nullptr,
                            scale, window, Hkv, Sq, gq, Smax, chunk, nsplit, 0, 0);
     }
@@ -267,7 +270,7 @@ static void launch_decode(torch::Tensor& out, const torch::Tensor& q,
     dim3 grid2(B * Hq * Sq);
     hipLaunchKernelGGL((decode_attn_pass2<T, D>), grid2, dim3(WAVE_SIZE), 0, stream,
                        (T*)out.data_ptr(), o_part.data_ptr<float>(),
-                       ml_part.data_ptr<float>(), nsplit);
+                       ml_part.data_ptr<float>(), nsplit, v_scale);
     HIP_CHECK_KERNEL();
 }
 
@@ -276,17 +279,32 @@ static void dispatch_gq(torch::Tensor& out, const torch::Tensor& q,
                         const torch::Tensor& k_cache, const torch::Tensor& v_cache,
                         const torch::Tensor& seq_lens, const torch::Tensor& seq_ids,
                         float scale, int window, int gq,
-                        const torch::Tensor* bt = nullptr, int bs = 0) {
-    if (gq <= 1)
-        launch_decode<T, D, 1>(out, q, k_cache, v_cache, seq_lens, seq_ids, scale, window, gq, bt, bs);
-    else if (gq <= 2)
-        launch_decode<T, D, 2>(out, q, k_cache, v_cache, seq_lens, seq_ids, scale, window, gq, bt, bs);
-    else if (gq <= 4)
-        launch_decode<T, D, 4>(out, q, k_cache, v_cache, seq_lens, seq_ids, scale, window, gq, bt, bs);
-    else if (gq <= 8)
-        launch_decode<T, D, 8>(out, q, k_cache, v_cache, seq_lens, seq_ids, scale, window, gq, bt, bs);
-    else
-        TORCH_CHECK(false, "GQA ratio > 8 not supported yet (got ", gq, ")");
+                        const torch::Tensor* bt = nullptr, int bs = 0,
+                        float v_scale = 1.0f) {
+    const bool fp8 = k_cache.scalar_type() == at::ScalarType::Float8_e4m3fn;
+#define GQ_CASE(CT)                                                                     \
+    do {                                                                                \
+        if (gq <= 1)                                                                    \
+            launch_decode<T, CT, D, 1>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
+                                       scale, window, gq, bt, bs, v_scale);             \
+        else if (gq <= 2)                                                               \
+            launch_decode<T, CT, D, 2>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
+                                       scale, window, gq, bt, bs, v_scale);             \
+        else if (gq <= 4)                                                               \
+            launch_decode<T, CT, D, 4>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
+                                       scale, window, gq, bt, bs, v_scale);             \
+        else if (gq <= 8)                                                               \
+            launch_decode<T, CT, D, 8>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
+                                       scale, window, gq, bt, bs, v_scale);             \
+        else                                                                            \
+            TORCH_CHECK(false, "GQA ratio > 8 not supported yet (got ", gq, ")");       \
+    } while (0)
+    if (fp8) {
+        GQ_CASE(__hip_fp8_e4m3);
+    } else {
+        GQ_CASE(T);
+    }
+#undef GQ_CASE
 }
 
 void attention_decode_paged(torch::Tensor out, torch::Tensor q,

@@ -6,6 +6,7 @@
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 #include <hip/hip_fp16.h>
+#include <hip/hip_fp8.h>
 
 #define WAVE_SIZE 64
 
@@ -29,6 +30,12 @@ template <> struct VecTraits<float> {
     static DEV_INLINE float to_float(scalar_t v) { return v; }
     static DEV_INLINE scalar_t from_float(float f) { return f; }
 };
+// OCP FP8 e4m3fn (gfx950-native format — NOT the MI300X fnuz variant)
+template <> struct VecTraits<__hip_fp8_e4m3> {
+    using scalar_t = __hip_fp8_e4m3;
+    static DEV_INLINE float to_float(scalar_t v) { return (float)v; }
+    static DEV_INLINE scalar_t from_float(float f) { return scalar_t(f); }
+};
 
 // 8-element vector of T loaded as one 16B (bf16/f16) or 32B (f32: two 16B) op.
 template <typename T>
@@ -39,7 +46,9 @@ struct Vec8 {
 template <typename T>
 DEV_INLINE Vec8<T> load_vec8(const T* p) {
     Vec8<T> out;
-    if constexpr (sizeof(T) == 2) {
+    if constexpr (sizeof(T) == 1) {
+        *reinterpret_cast<int2*>(out.v) = *reinterpret_cast<const int2*>(p);
+    } else if constexpr (sizeof(T) == 2) {
         *reinterpret_cast<int4*>(out.v) = *reinterpret_cast<const int4*>(p);
     } else {
         *reinterpret_cast<int4*>(out.v) = *reinterpret_cast<const int4*>(p);
@@ -50,12 +59,24 @@ DEV_INLINE Vec8<T> load_vec8(const T* p) {
 
 template <typename T>
 DEV_INLINE void store_vec8(T* p, const Vec8<T>& x) {
-    if constexpr (sizeof(T) == 2) {
+    if constexpr (sizeof(T) == 1) {
+        *reinterpret_cast<int2*>(p) = *reinterpret_cast<const int2*>(x.v);
+    } else if constexpr (sizeof(T) == 2) {
         *reinterpret_cast<int4*>(p) = *reinterpret_cast<const int4*>(x.v);
     } else {
         *reinterpret_cast<int4*>(p) = *reinterpret_cast<const int4*>(x.v);
         *reinterpret_cast<int4*>(p + 4) = *reinterpret_cast<const int4*>(x.v + 4);
     }
+}
+
+// convert an 8-vector between storage types through fp32
+template <typename DST, typename SRC>
+DEV_INLINE Vec8<DST> convert_vec8(const Vec8<SRC>& x, float scale = 1.0f) {
+    Vec8<DST> out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+        out.v[j] = VecTraits<DST>::from_float(VecTraits<SRC>::to_float(x.v[j]) * scale);
+    return out;
 }
 
 // ------------------------- wave reductions -------------------------

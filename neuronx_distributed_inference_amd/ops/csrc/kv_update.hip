@@ -10,9 +10,9 @@
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
-template <typename T>
+template <typename T, typename CT>
 __global__ void kv_update_kernel(
-    T* __restrict__ k_cache, T* __restrict__ v_cache,
+    CT* __restrict__ k_cache, CT* __restrict__ v_cache,
     const T* __restrict__ k_new, const T* __restrict__ v_new,
     const int* __restrict__ positions,   // (B, Sn)
     const int* __restrict__ seq_ids,     // (B,)
@@ -35,14 +35,14 @@ __global__ void kv_update_kernel(
         if (pos < 0 || pos >= Smax) continue;
         const long src = (((long)b * Hkv + h) * Sn + s) * D + d8 * 8;
         const long dst = (((long)row * Hkv + h) * Smax + pos) * D + d8 * 8;
-        store_vec8(k_cache + dst, load_vec8(k_new + src));
-        store_vec8(v_cache + dst, load_vec8(v_new + src));
+        store_vec8(k_cache + dst, convert_vec8<CT, T>(load_vec8(k_new + src)));
+        store_vec8(v_cache + dst, convert_vec8<CT, T>(load_vec8(v_new + src)));
     }
 }
 
-template <typename T>
+template <typename T, typename CT>
 __global__ void block_kv_update_kernel(
-    T* __restrict__ k_cache, T* __restrict__ v_cache,
+    CT* __restrict__ k_cache, CT* __restrict__ v_cache,
     const T* __restrict__ k_new, const T* __restrict__ v_new,
     const int* __restrict__ slot_mapping,  // (N,)
     const int N, const int Hkv, const int D, const long nslots) {
@@ -58,8 +58,8 @@ __global__ void block_kv_update_kernel(
         if (slot < 0 || slot >= nslots) continue;
         const long src = ((long)n * Hkv + h) * D + d8 * 8;
         const long dst = ((long)slot * Hkv + h) * D + d8 * 8;
-        store_vec8(k_cache + dst, load_vec8(k_new + src));
-        store_vec8(v_cache + dst, load_vec8(v_new + src));
+        store_vec8(k_cache + dst, convert_vec8<CT, T>(load_vec8(k_new + src)));
+        store_vec8(v_cache + dst, convert_vec8<CT, T>(load_vec8(v_new + src)));
     }
 }
 
@@ -86,17 +86,30 @@ void kv_cache_update(torch::Tensor k_cache, torch::Tensor v_cache,
     const int B = k_new.size(0), Hkv = k_new.size(1), Sn = k_new.size(2), D = k_new.size(3);
     const int Bc = k_cache.size(0), Smax = k_cache.size(2);
     TORCH_CHECK(D % 8 == 0, "head_dim must be a multiple of 8");
-    TORCH_CHECK(k_cache.scalar_type() == k_new.scalar_type(), "cache/new dtype mismatch");
+    const bool fp8 = k_cache.scalar_type() == at::ScalarType::Float8_e4m3fn;
+    TORCH_CHECK(fp8 || k_cache.scalar_type() == k_new.scalar_type(),
+                "cache/new dtype mismatch");
     const long total = (long)B * Hkv * Sn * (D / 8);
     long blocks = (total + 255) / 256;
     if (blocks > 2048) blocks = 2048;
     auto stream = at::hip::getCurrentHIPStream();
-    DISPATCH_FLOAT_TYPES_KV(k_cache.scalar_type(), "kv_cache_update", [&] {
-        hipLaunchKernelGGL((kv_update_kernel<scalar_t>), dim3((int)blocks), dim3(256), 0, stream,
-                           (scalar_t*)k_cache.data_ptr(), (scalar_t*)v_cache.data_ptr(),
-                           (const scalar_t*)k_new.data_ptr(), (const scalar_t*)v_new.data_ptr(),
-                           positions.data_ptr<int>(), seq_ids.data_ptr<int>(),
-                           B, Hkv, Sn, D, Bc, Smax);
+    DISPATCH_FLOAT_TYPES_KV(k_new.scalar_type(), "kv_cache_update", [&] {
+        if (fp8) {
+            hipLaunchKernelGGL((kv_update_kernel<scalar_t, __hip_fp8_e4m3>),
+                               dim3((int)blocks), dim3(256), 0, stream,
+                               (__hip_fp8_e4m3*)k_cache.data_ptr(),
+                               (__hip_fp8_e4m3*)v_cache.data_ptr(),
+                               (const scalar_t*)k_new.data_ptr(), (const scalar_t*)v_new.data_ptr(),
+                               positions.data_ptr<int>(), seq_ids.data_ptr<int>(),
+                               B, Hkv, Sn, D, Bc, Smax);
+        } else {
+            hipLaunchKernelGGL((kv_update_kernel<scalar_t, scalar_t>),
+                               dim3((int)blocks), dim3(256), 0, stream,
+                               (scalar_t*)k_cache.data_ptr(), (scalar_t*)v_cache.data_ptr(),
+                               (const scalar_t*)k_new.data_ptr(), (const scalar_t*)v_new.data_ptr(),
+                               positions.data_ptr<int>(), seq_ids.data_ptr<int>(),
+                               B, Hkv, Sn, D, Bc, Smax);
+        }
     });
     HIP_CHECK_KERNEL();
 }
@@ -111,12 +124,22 @@ void block_kv_cache_update(torch::Tensor k_cache, torch::Tensor v_cache,
     long blocks = (total + 255) / 256;
     if (blocks > 2048) blocks = 2048;
     auto stream = at::hip::getCurrentHIPStream();
-    DISPATCH_FLOAT_TYPES_KV(k_cache.scalar_type(), "block_kv_cache_update", [&] {
-        hipLaunchKernelGGL((block_kv_update_kernel<scalar_t>), dim3((int)blocks), dim3(256), 0,
-                           stream,
-                           (scalar_t*)k_cache.data_ptr(), (scalar_t*)v_cache.data_ptr(),
-                           (const scalar_t*)k_new.data_ptr(), (const scalar_t*)v_new.data_ptr(),
-                           slot_mapping.data_ptr<int>(), N, Hkv, D, nslots);
+    const bool fp8 = k_cache.scalar_type() == at::ScalarType::Float8_e4m3fn;
+    DISPATCH_FLOAT_TYPES_KV(k_new.scalar_type(), "block_kv_cache_update", [&] {
+        if (fp8) {
+            hipLaunchKernelGGL((block_kv_update_kernel<scalar_t, __hip_fp8_e4m3>),
+                               dim3((int)blocks), dim3(256), 0, stream,
+                               (__hip_fp8_e4m3*)k_cache.data_ptr(),
+                               (__hip_fp8_e4m3*)v_cache.data_ptr(),
+                               (const scalar_t*)k_new.data_ptr(), (const scalar_t*)v_new.data_ptr(),
+                               slot_mapping.data_ptr<int>(), N, Hkv, D, nslots);
+        } else {
+            hipLaunchKernelGGL((block_kv_update_kernel<scalar_t, scalar_t>),
+                               dim3((int)blocks), dim3(256), 0, stream,
+                               (scalar_t*)k_cache.data_ptr(), (scalar_t*)v_cache.data_ptr(),
+                               (const scalar_t*)k_new.data_ptr(), (const scalar_t*)v_new.data_ptr(),
+                               slot_mapping.data_ptr<int>(), N, Hkv, D, nslots);
+        }
     });
     HIP_CHECK_KERNEL();
 }

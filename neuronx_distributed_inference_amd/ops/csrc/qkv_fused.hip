@@ -12,14 +12,14 @@
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
-template <typename T>
+template <typename T, typename CT>
 __global__ void qkv_rope_update_kernel(
     T* __restrict__ q_out,           // (B, Hq, S, D)
     const T* __restrict__ qkv,       // (B, S, (Hq+2Hkv)*D)
     const float* __restrict__ cosb,  // (B, S, D)
     const float* __restrict__ sinb,
-    T* __restrict__ k_cache,         // (Bc, Hkv, Smax, D)
-    T* __restrict__ v_cache,
+    CT* __restrict__ k_cache,        // (Bc, Hkv, Smax, D)
+    CT* __restrict__ v_cache,
     const int* __restrict__ positions,  // (B, S)
     const int* __restrict__ seq_ids,    // (B,)
     const int B, const int S, const int Hq, const int Hkv,
@@ -41,31 +41,31 @@ __global__ void qkv_rope_update_kernel(
         const float x1f = VT::to_float(src[i]);
         const float x2f = VT::to_float(src[i + half]);
 
-        T o1, o2;
+        float o1f, o2f;
         if (h < Hq + Hkv) {  // rope for q and k
             const float c = cosb[((long)b * S + s) * D + i];
             const float sn = sinb[((long)b * S + s) * D + i];
-            o1 = VT::from_float(x1f * c - x2f * sn);
-            o2 = VT::from_float(x2f * c + x1f * sn);
+            o1f = x1f * c - x2f * sn;
+            o2f = x2f * c + x1f * sn;
         } else {             // v: plain copy
-            o1 = src[i];
-            o2 = src[i + half];
+            o1f = x1f;
+            o2f = x2f;
         }
 
         if (h < Hq) {
             T* dst = q_out + (((long)b * Hq + h) * S + s) * D;
-            dst[i] = o1;
-            dst[i + half] = o2;
+            dst[i] = VT::from_float(o1f);
+            dst[i + half] = VT::from_float(o2f);
         } else {
             const int row = seq_ids[b];
             const int pos = positions[(long)b * S + s];
             if (row < 0 || row >= Bc || pos < 0 || pos >= Smax) continue;
             const bool is_k = h < Hq + Hkv;
             const int hk = is_k ? h - Hq : h - Hq - Hkv;
-            T* cache = is_k ? k_cache : v_cache;
-            T* dst = cache + (((long)row * Hkv + hk) * Smax + pos) * D;
-            dst[i] = o1;
-            dst[i + half] = o2;
+            CT* cache = is_k ? k_cache : v_cache;
+            CT* dst = cache + (((long)row * Hkv + hk) * Smax + pos) * D;
+            dst[i] = VecTraits<CT>::from_float(o1f);
+            dst[i + half] = VecTraits<CT>::from_float(o2f);
         }
     }
 }
@@ -130,14 +130,26 @@ void qkv_rope_update(torch::Tensor q_out, torch::Tensor qkv,
     long blocks = (total + 255) / 256;
     if (blocks > 2048) blocks = 2048;
     auto stream = at::hip::getCurrentHIPStream();
+    const bool fp8 = k_cache.scalar_type() == at::ScalarType::Float8_e4m3fn;
     DISPATCH_FT(qkv.scalar_type(), "qkv_rope_update", [&] {
-        hipLaunchKernelGGL((qkv_rope_update_kernel<scalar_t>), dim3((int)blocks),
-                           dim3(256), 0, stream,
-                           (scalar_t*)q_out.data_ptr(), (const scalar_t*)qkv.data_ptr(),
-                           cosb.data_ptr<float>(), sinb.data_ptr<float>(),
-                           (scalar_t*)k_cache.data_ptr(), (scalar_t*)v_cache.data_ptr(),
-                           positions.data_ptr<int>(), seq_ids.data_ptr<int>(),
-                           B, S, (int)Hq, (int)Hkv, D, Bc, Smax);
+        if (fp8) {
+            hipLaunchKernelGGL((qkv_rope_update_kernel<scalar_t, __hip_fp8_e4m3>),
+                               dim3((int)blocks), dim3(256), 0, stream,
+                               (scalar_t*)q_out.data_ptr(), (const scalar_t*)qkv.data_ptr(),
+                               cosb.data_ptr<float>(), sinb.data_ptr<float>(),
+                               (__hip_fp8_e4m3*)k_cache.data_ptr(),
+                               (__hip_fp8_e4m3*)v_cache.data_ptr(),
+                               positions.data_ptr<int>(), seq_ids.data_ptr<int>(),
+                               B, S, (int)Hq, (int)Hkv, D, Bc, Smax);
+        } else {
+            hipLaunchKernelGGL((qkv_rope_update_kernel<scalar_t, scalar_t>),
+                               dim3((int)blocks), dim3(256), 0, stream,
+                               (scalar_t*)q_out.data_ptr(), (const scalar_t*)qkv.data_ptr(),
+                               cosb.data_ptr<float>(), sinb.data_ptr<float>(),
+                               (scalar_t*)k_cache.data_ptr(), (scalar_t*)v_cache.data_ptr(),
+                               positions.data_ptr<int>(), seq_ids.data_ptr<int>(),
+                               B, S, (int)Hq, (int)Hkv, D, Bc, Smax);
+        }
     });
     HIP_CHECK_KERNEL();
 }

@@ -328,3 +328,38 @@ def test_attention_decode_paged_gpu():
         q.float().cpu(), kc.float().cpu(), vc.float().cpu(), seq_lens.cpu(), bt.cpu()
     )
     torch.testing.assert_close(out.float().cpu(), want, atol=3e-2, rtol=3e-2)
+
+
+def test_attention_decode_fp8_cache():
+    """Decode attention reading an OCP-e4m3 KV cache vs the fp32 reference
+    on the dequantized values."""
+    torch.manual_seed(0)
+    B, Hkv, gq, Smax, D = 2, 2, 4, 256, 128
+    dev = _dev()
+    kc_f = torch.randn(B, Hkv, Smax, D, device=dev) * 2
+    vc_f = torch.randn(B, Hkv, Smax, D, device=dev) * 2
+    kc = kc_f.to(torch.float8_e4m3fn)
+    vc = vc_f.to(torch.float8_e4m3fn)
+    q = torch.randn(B, Hkv * gq, 1, D, dtype=torch.bfloat16, device=dev)
+    seq_lens = torch.tensor([100, 256], device=dev)
+    out = ops.attention_decode(q, kc, vc, seq_lens)
+    want = ops.attention_decode(
+        q.float().cpu(), kc.cpu().float(), vc.cpu().float(), seq_lens.cpu()
+    )
+    torch.testing.assert_close(out.float().cpu(), want, atol=5e-2, rtol=5e-2)
+
+
+def test_kv_update_fp8_cache():
+    torch.manual_seed(0)
+    Bc, Hkv, Smax, D = 2, 2, 32, 128
+    dev = _dev()
+    kc = torch.zeros(Bc, Hkv, Smax, D, device=dev).to(torch.float8_e4m3fn)
+    vc = torch.zeros_like(kc)
+    k = torch.randn(2, Hkv, 3, D, dtype=torch.bfloat16, device=dev)
+    v = torch.randn_like(k)
+    pos = torch.tensor([[0, 1, 2], [5, 6, 7]], device=dev)
+    sid = torch.tensor([0, 1], device=dev)
+    ops.kv_cache_update(kc, vc, k, v, pos, sid)
+    torch.testing.assert_close(
+        kc[0, :, 0:3].cpu().float(), k[0].float().cpu(), atol=0.15, rtol=0.15
+    )

@@ -77,3 +77,41 @@ def test_slot_mapping_negative_positions():
     pos = torch.tensor([[-1, 0, 1, 8, 9]])
     slots = generate_slot_mapping(bt, pos, 8)
     assert slots.tolist() == [-1, 16, 17, 32, 33]
+
+
+@torch.inference_mode()
+def test_fp8_kv_cache_cpu():
+    """FP8-quantized KV cache: generation runs and last-token logits stay
+    close to the bf16-cache model (quantization noise only)."""
+    torch.manual_seed(0)
+    cfg_full = _cfg(False)
+    m_full = LlamaForCausalLM(cfg_full).eval()
+    with torch.no_grad():
+        for p in m_full.parameters():
+            if p.dim() >= 2:
+                p.normal_(0, 0.1)
+    cfg_q = _cfg(False)
+    cfg_q.neuron_config.kv_cache_quant = True
+    m_q = LlamaForCausalLM(cfg_q).eval()
+    m_q.model.load_state_dict(
+        {k: v for k, v in m_full.model.state_dict().items() if not k.startswith("kv_mgr.")},
+        strict=False,
+    )
+    assert m_q.model.kv_mgr.k_caches[0].dtype == torch.float8_e4m3fn
+
+    ids = torch.randint(0, 128, (2, 8))
+    m_full.reset(); m_q.reset()
+    # prefill then one decode step; compare decode logits (which READ cache)
+    m_full.ec.output_logits = True
+    m_q.ec.output_logits = True
+    m_full.model.sampler = None
+    m_q.model.sampler = None
+    o1 = m_full(input_ids=ids)
+    o2 = m_q(input_ids=ids)
+    tok = o1.logits[:, -1].argmax(-1).view(2, 1)
+    p = torch.full((2, 1), 8, dtype=torch.long)
+    d1 = m_full(input_ids=tok, position_ids=p)
+    d2 = m_q(input_ids=tok, position_ids=p)
+    diff = (d1.logits - d2.logits).abs().max()
+    assert diff < 1.0, f"fp8 cache diverged too far: {diff}"
+    assert diff > 0  # actually quantized
