@@ -85,24 +85,39 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
         for (int j = 0; j < 8; ++j) o_acc[g][j] = 0.f;
     }
 
-    // ---- stream K/V rows ----
+    // ---- stream K/V rows (next row's loads issued before computing the
+    // current one so the HBM latency hides under the dot/softmax VALU) ----
     if ((PAGED || row >= 0) && split_lo < split_hi) {
         const CT* kbase = PAGED ? k_cache : k_cache + (((long)row * Hkv + hkv) * Smax) * D;
         const CT* vbase = PAGED ? v_cache : v_cache + (((long)row * Hkv + hkv) * Smax) * D;
-        for (int l = split_lo + grp; l < split_hi; l += NG) {
-            long off;
+        // unconditional prefetch with clamped addresses — a conditional
+        // ("load or keep") makes hipcc branch around the loads and wait
+        // vmcnt(0) per iteration (guide §5 trap 4(c))
+        bool valid_next;
+        auto row_addr = [&](int l, bool& valid) -> long {
             if constexpr (PAGED) {
                 const int blk = block_table[(long)b * max_blocks + l / block_size];
-                if (blk < 0) continue;
-                off = (((long)blk * block_size + l % block_size) * Hkv + hkv) * D + c * 8;
-            } else {
-                off = (long)l * D + c * 8;
+                valid = blk >= 0;
+                return (((long)max(blk, 0) * block_size + l % block_size) * Hkv + hkv) * D + c * 8;
             }
-            Vec8<CT> kv = load_vec8(kbase + off);
+            valid = true;
+            return (long)l * D + c * 8;
+        };
+        long off = row_addr(split_lo + grp, valid_next);
+        Vec8<CT> kv = load_vec8(kbase + off);
+        Vec8<CT> vv = load_vec8(vbase + off);
+        for (int l = split_lo + grp; l < split_hi; l += NG) {
+            const bool valid_cur = valid_next;
+            Vec8<CT> kv_cur = kv, vv_cur = vv;
+            const int lnext = (l + NG < split_hi) ? l + NG : l;
+            off = row_addr(lnext, valid_next);
+            kv = load_vec8(kbase + off);
+            vv = load_vec8(vbase + off);
+            if (PAGED && !valid_cur) continue;
             float kf[8];
 #pragma unroll
-            for (int j = 0; j < 8; ++j) kf[j] = VecTraits<CT>::to_float(kv.v[j]);
-            Vec8<CT> vv = load_vec8(vbase + off);
+            for (int j = 0; j < 8; ++j) kf[j] = VecTraits<CT>::to_float(kv_cur.v[j]);
+            const Vec8<CT>& vv_use = vv_cur;
 
             float s[GQ];
 #pragma unroll
@@ -123,7 +138,7 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
                 m[g] = mn;
 #pragma unroll
                 for (int j = 0; j < 8; ++j)
-                    o_acc[g][j] = o_acc[g][j] * corr + p * VecTraits<CT>::to_float(vv.v[j]);
+                    o_acc[g][j] = o_acc[g][j] * corr + p * VecTraits<CT>::to_float(vv_use.v[j]);
             }
         }
     }
