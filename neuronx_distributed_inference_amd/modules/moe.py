@@ -54,37 +54,88 @@ class MoERouter(nn.Module):
 
 
 class ExpertMLPs(nn.Module):
-    """Stacked gated expert MLPs: w_gate/w_up (E, I/tp, H), w_down (E, H, I/tp)."""
+    """Stacked gated expert MLPs.
+
+    TP x EP sharding (ref hybrid moe_tp/moe_ep,
+    `modules/moe_v2.py:135-161`): the intermediate dim is sharded over
+    moe_tp = tp/ep ranks and the EXPERT dim over ep ranks, so rank r of the
+    TP group holds experts [ (r//moe_tp)*E/ep , ...+E/ep ) with I/moe_tp
+    columns each. Because TP replicates activations, a single all-reduce
+    over the full TP group (done by MoELayer) sums both the I-shards and
+    the expert subsets — the AR dispatch option of the reference's
+    EPDispatchOption enum (attention_base.py:103-113).
+
+    Local shapes: w_gate/w_up (E/ep, I/moe_tp, H), w_down (E/ep, H, I/moe_tp).
+    The preshard hook stacks each global rank's 3D block on dim 0 so the
+    standard dim-0 slicing yields it.
+    """
 
     def __init__(self, num_experts: int, hidden_size: int, intermediate_size: int,
                  tp_degree: int, dtype: torch.dtype, hidden_act: str = "silu",
-                 group=None):
+                 group=None, ep_degree: int = 1):
         super().__init__()
         self.num_experts = num_experts
-        ipp = divide(intermediate_size, tp_degree)
-        self.w_gate = nn.Parameter(torch.empty(num_experts, ipp, hidden_size, dtype=dtype))
-        self.w_up = nn.Parameter(torch.empty(num_experts, ipp, hidden_size, dtype=dtype))
-        self.w_down = nn.Parameter(torch.empty(num_experts, hidden_size, ipp, dtype=dtype))
-        mark_parallel(self.w_gate, partition_dim=1)
-        mark_parallel(self.w_up, partition_dim=1)
-        mark_parallel(self.w_down, partition_dim=2)
+        self.ep_degree = ep_degree
+        moe_tp = divide(tp_degree, ep_degree)
+        self.num_local_experts = divide(num_experts, ep_degree)
+        ipp = divide(intermediate_size, moe_tp)
+        El = self.num_local_experts
+        self.w_gate = nn.Parameter(torch.empty(El, ipp, hidden_size, dtype=dtype))
+        self.w_up = nn.Parameter(torch.empty(El, ipp, hidden_size, dtype=dtype))
+        self.w_down = nn.Parameter(torch.empty(El, hidden_size, ipp, dtype=dtype))
+
+        def _preshard(part_dim):
+            def fn(w, tp, El=El, moe_tp=moe_tp, ipp=ipp):
+                # full w: (E, I, H) or (E, H, I); build per-global-rank 3D
+                # blocks stacked on dim 0 (rank r: ep_idx = r // moe_tp,
+                # tp_idx = r % moe_tp)
+                blocks = []
+                for r in range(tp):
+                    ep_idx, tp_idx = r // moe_tp, r % moe_tp
+                    blk = w[ep_idx * El:(ep_idx + 1) * El]
+                    blk = blk.narrow(part_dim, tp_idx * ipp, ipp)
+                    blocks.append(blk)
+                return torch.cat(blocks, dim=0)
+            return fn
+
+        mark_parallel(self.w_gate, partition_dim=0)
+        mark_parallel(self.w_up, partition_dim=0)
+        mark_parallel(self.w_down, partition_dim=0)
+        self.w_gate.preshard_fn = _preshard(1)
+        self.w_up.preshard_fn = _preshard(1)
+        self.w_down.preshard_fn = _preshard(2)
         self.act_mul = ops.silu_mul if hidden_act in ("silu", "swish") else ops.gelu_tanh_mul
         self._group = group
+        self.expert_offset = 0  # resolved from the live EP rank at forward
+
+    def _local_offset(self) -> int:
+        if self.ep_degree <= 1:
+            return 0
+        from neuronx_distributed_inference_amd.parallel import state as ps
+
+        import torch.distributed as dist
+        if not dist.is_initialized():
+            return 0
+        tp_rank = ps.get_tensor_model_parallel_rank()
+        moe_tp = ps.get_tensor_model_parallel_size() // self.ep_degree
+        return (tp_rank // max(moe_tp, 1)) * self.num_local_experts
 
     def forward(self, x: torch.Tensor, weights: torch.Tensor, idx: torch.Tensor):
-        """x: (T, H); weights/idx: (T, top_k). Returns (T, H) partial (needs
-        TP all-reduce by caller)."""
+        """x: (T, H); weights/idx: (T, top_k) over GLOBAL expert ids.
+        Returns (T, H) partial (summed by the caller's TP all-reduce)."""
         T, H = x.shape
         out = torch.zeros(T, H, dtype=x.dtype, device=x.device)
-        for e in range(self.num_experts):
+        offset = self._local_offset()
+        for e_local in range(self.num_local_experts):
+            e = offset + e_local
             sel = (idx == e)            # (T, top_k)
             rows = sel.any(dim=-1)
             if not bool(rows.any()):
                 continue
             xe = x[rows]
-            ge = F.linear(xe, self.w_gate[e])
-            ue = F.linear(xe, self.w_up[e])
-            he = F.linear(self.act_mul(ge, ue), self.w_down[e])
+            ge = F.linear(xe, self.w_gate[e_local])
+            ue = F.linear(xe, self.w_up[e_local])
+            he = F.linear(self.act_mul(ge, ue), self.w_down[e_local])
             w = (weights * sel).sum(dim=-1)[rows].to(x.dtype).unsqueeze(-1)
             out[rows] += he * w
         return out
@@ -106,7 +157,7 @@ class MoELayer(nn.Module):
         )
         self.experts = ExpertMLPs(
             E, config.hidden_size, inter, ec.tp_degree, ec.torch_dtype,
-            hidden_act=hidden_act,
+            hidden_act=hidden_act, ep_degree=ec.ep_degree,
         )
         self.shared_expert = None  # models with shared experts set this
 
