@@ -41,6 +41,11 @@ _REGISTRY: Dict[str, Tuple[str, str, str]] = {
         "DbrxInferenceConfig",
         "DbrxForCausalLM",
     ),
+    "gemma3": (
+        "neuronx_distributed_inference_amd.models.gemma3.modeling_gemma3",
+        "Gemma3InferenceConfig",
+        "Gemma3ForCausalLM",
+    ),
     "qwen3_moe": (
         "neuronx_distributed_inference_amd.models.qwen3_moe.modeling_qwen3_moe",
         "Qwen3MoeInferenceConfig",

@@ -165,3 +165,29 @@ def test_registry_resolves_all():
     for mt in available_model_types():
         cfg_cls, model_cls = get_model_classes(mt)
         assert cfg_cls is not None and model_cls is not None
+
+
+@torch.inference_mode()
+def test_gemma3():
+    from transformers import Gemma3TextConfig, Gemma3ForCausalLM as HF
+
+    from neuronx_distributed_inference_amd.models.gemma3 import (
+        Gemma3ForCausalLM, Gemma3InferenceConfig,
+    )
+
+    torch.manual_seed(0)
+    hf_cfg = Gemma3TextConfig(
+        vocab_size=128, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, sliding_window=8, sliding_window_pattern=2,
+        max_position_embeddings=128, rms_norm_eps=1e-6, tie_word_embeddings=True,
+    )
+    hf = HF(hf_cfg).eval()
+    d = hf_cfg.to_dict()
+    cfg = Gemma3InferenceConfig(neuron_config=_engine(), **d)
+    ours = Gemma3ForCausalLM(cfg).eval()
+    sd = Gemma3ForCausalLM.convert_hf_to_native_state_dict(hf.state_dict(), cfg)
+    missing, unexpected = ours.model.load_state_dict(sd, strict=False)
+    assert not [m for m in missing if not m.startswith("kv_mgr") and ".rotary." not in m], missing
+    assert not unexpected, unexpected
+    _check_match(ours, hf)
