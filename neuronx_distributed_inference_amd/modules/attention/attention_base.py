@@ -57,7 +57,7 @@ class AttentionBase(nn.Module):
         )
         self.o_proj = GroupQueryAttentionO(
             self.hidden_size, self.num_heads, self.head_dim, self.tp_degree,
-            self.torch_dtype, bias=False,
+            self.torch_dtype, bias=bool(getattr(config, "attention_o_bias", False)),
         )
         self.num_heads_per_rank = self.qkv_proj.num_heads_per_rank
         self.num_kv_heads_per_rank = self.qkv_proj.num_kv_heads_per_rank

@@ -34,12 +34,15 @@ class RotaryEmbedding(nn.Module):
         inv_freq = 1.0 / (
             base ** (torch.arange(0, head_dim, 2, dtype=torch.float32) / head_dim)
         )
+        self.attention_scaling = 1.0
         if scaling:
             rope_type = scaling.get("rope_type", scaling.get("type", "default"))
             if rope_type == "llama3":
                 inv_freq = self._llama3_scale(inv_freq, scaling)
             elif rope_type in ("linear",):
                 inv_freq = inv_freq / float(scaling.get("factor", 1.0))
+            elif rope_type == "yarn":
+                inv_freq, self.attention_scaling = self._yarn(head_dim, base, scaling)
             # "default"/"dynamic": leave as-is (dynamic handled at table build)
         self.register_buffer("inv_freq", inv_freq, persistent=False)
         self._cos_cached: Optional[torch.Tensor] = None
@@ -62,12 +65,52 @@ class RotaryEmbedding(nn.Module):
         is_mid = (wavelen <= low_wl) & (wavelen >= high_wl)
         return torch.where(is_mid, smoothed, out)
 
+    @staticmethod
+    def _yarn(dim: int, base: float, s: dict):
+        """YaRN NTK-by-parts scaling (GPT-OSS, DeepSeek; matches HF
+        _compute_yarn_parameters)."""
+        factor = s.get("factor")
+        orig = s.get("original_max_position_embeddings", 4096)
+        if factor is None:
+            factor = 1.0
+        beta_fast = s.get("beta_fast") or 32
+        beta_slow = s.get("beta_slow") or 1
+        mscale = s.get("mscale")
+        mscale_all_dim = s.get("mscale_all_dim")
+
+        def get_mscale(scale, m=1):
+            return 1.0 if scale <= 1 else 0.1 * m * math.log(scale) + 1.0
+
+        att = s.get("attention_factor")
+        if att is None:
+            if mscale and mscale_all_dim:
+                att = float(get_mscale(factor, mscale) / get_mscale(factor, mscale_all_dim))
+            else:
+                att = get_mscale(factor)
+
+        def corr_dim(rot):
+            return (dim * math.log(orig / (rot * 2 * math.pi))) / (2 * math.log(base))
+
+        low, high = corr_dim(beta_fast), corr_dim(beta_slow)
+        if s.get("truncate", True):
+            low, high = math.floor(low), math.ceil(high)
+        low, high = max(low, 0), min(high, dim - 1)
+        if low == high:
+            high += 0.001
+        pos_freqs = base ** (torch.arange(0, dim, 2, dtype=torch.float32) / dim)
+        extrap = 1.0 / pos_freqs
+        interp = 1.0 / (factor * pos_freqs)
+        ramp = ((torch.arange(dim // 2, dtype=torch.float32) - low) / (high - low)).clamp(0, 1)
+        extrap_factor = 1 - ramp
+        inv_freq = interp * (1 - extrap_factor) + extrap * extrap_factor
+        return inv_freq, float(att)
+
     def _build_table(self, seqlen: int, device, dtype):
         t = torch.arange(seqlen, dtype=torch.float32, device=device)
         freqs = torch.outer(t, self.inv_freq.to(device))
         emb = torch.cat((freqs, freqs), dim=-1)  # (S, D)
-        self._cos_cached = emb.cos()
-        self._sin_cached = emb.sin()
+        self._cos_cached = emb.cos() * self.attention_scaling
+        self._sin_cached = emb.sin() * self.attention_scaling
         self._cached_len = seqlen
 
     def ensure_table(self, seqlen: int, device):

@@ -122,12 +122,13 @@ def attention_prefill(q, k, v, mask=None, scale=None, sinks=None, causal=True,
         scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
         if (
             hasattr(c, "flash_prefill")
-            and mask is None and causal and sinks is None
+            and mask is None and causal
             and q.shape[-1] in (64, 128) and q.shape[2] == k.shape[2]
         ):
             out = torch.empty_like(q)
             c.flash_prefill(out, q.contiguous(), k.contiguous(), v.contiguous(),
-                            float(scale), int(sliding_window))
+                            float(scale), int(sliding_window),
+                            sinks=None if sinks is None else sinks.detach())
             return out
         # masked / unusual-shape path: matmul-softmax via library GEMMs
         if mask is None and causal:
@@ -172,6 +173,7 @@ def attention_decode(q, k_cache, v_cache, seq_lens, seq_ids=None, scale=None,
             out, q.contiguous(), k_cache, v_cache,
             seq_lens.to(torch.int32).contiguous(),
             seq_ids.to(torch.int32).contiguous(), float(scale), int(sliding_window),
+            sinks=None if sinks is None else sinks.detach(),
         )
         return out
     # CPU reference: mask by per-sequence length (+ causal within Sq)

@@ -214,7 +214,9 @@ __global__ void decode_attn_pass2(
     T* __restrict__ out,               // (B, Hq, Sq, D)
     const float* __restrict__ o_part,  // (B, Hq, Sq, nsplit, D)
     const float* __restrict__ ml_part, // (B, Hq, Sq, nsplit, 2)
-    const int nsplit, const float v_scale) {
+    const int nsplit, const float v_scale,
+    const float* __restrict__ sinks,   // (Hq,) learned sink logits or null
+    const int Hq, const int Sq) {
     using VT = VecTraits<T>;
     const long rowid = blockIdx.x;  // over B*Hq*Sq
     const int lane = threadIdx.x;
@@ -223,6 +225,11 @@ __global__ void decode_attn_pass2(
     for (int i = 0; i < nsplit; ++i)
         mstar = fmaxf(mstar, ml_part[(rowid * nsplit + i) * 2 + 0]);
     float L = 0.f;
+    if (sinks != nullptr) {  // sink joins the softmax denominator only
+        const int h = (int)((rowid / Sq) % Hq);
+        mstar = fmaxf(mstar, sinks[h]);
+        L += __expf(sinks[h] - mstar);
+    }
     for (int i = 0; i < nsplit; ++i) {
         const float mi = ml_part[(rowid * nsplit + i) * 2 + 0];
         const float li = ml_part[(rowid * nsplit + i) * 2 + 1];
@@ -247,7 +254,8 @@ static void launch_decode(torch::Tensor& out, const torch::Tensor& q,
                           const torch::Tensor& seq_lens, const torch::Tensor& seq_ids,
                           float scale, int window, int gq,
                           const torch::Tensor* block_table = nullptr,
-                          int block_size = 0, float v_scale = 1.0f) {
+                          int block_size = 0, float v_scale = 1.0f,
+                          const torch::Tensor* sinks = nullptr) {
     const int B = q.size(0), Hq = q.size(1), Sq = q.size(2);
     const bool paged = block_table != nullptr;
     const int Hkv = paged ? k_cache.size(2) : k_cache.size(1);
@@ -285,7 +293,8 @@ static void launch_decode(torch::Tensor& out, const torch::Tensor& q,
     dim3 grid2(B * Hq * Sq);
     hipLaunchKernelGGL((decode_attn_pass2<T, D>), grid2, dim3(WAVE_SIZE), 0, stream,
                        (T*)out.data_ptr(), o_part.data_ptr<float>(),
-                       ml_part.data_ptr<float>(), nsplit, v_scale);
+                       ml_part.data_ptr<float>(), nsplit, v_scale,
+                       sinks ? sinks->data_ptr<float>() : nullptr, Hq, Sq);
     HIP_CHECK_KERNEL();
 }
 
@@ -295,22 +304,22 @@ static void dispatch_gq(torch::Tensor& out, const torch::Tensor& q,
                         const torch::Tensor& seq_lens, const torch::Tensor& seq_ids,
                         float scale, int window, int gq,
                         const torch::Tensor* bt = nullptr, int bs = 0,
-                        float v_scale = 1.0f) {
+                        float v_scale = 1.0f, const torch::Tensor* sinks = nullptr) {
     const bool fp8 = k_cache.scalar_type() == at::ScalarType::Float8_e4m3fn;
 #define GQ_CASE(CT)                                                                     \
     do {                                                                                \
         if (gq <= 1)                                                                    \
             launch_decode<T, CT, D, 1>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
-                                       scale, window, gq, bt, bs, v_scale);             \
+                                       scale, window, gq, bt, bs, v_scale, sinks);      \
         else if (gq <= 2)                                                               \
             launch_decode<T, CT, D, 2>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
-                                       scale, window, gq, bt, bs, v_scale);             \
+                                       scale, window, gq, bt, bs, v_scale, sinks);      \
         else if (gq <= 4)                                                               \
             launch_decode<T, CT, D, 4>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
-                                       scale, window, gq, bt, bs, v_scale);             \
+                                       scale, window, gq, bt, bs, v_scale, sinks);      \
         else if (gq <= 8)                                                               \
             launch_decode<T, CT, D, 8>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
-                                       scale, window, gq, bt, bs, v_scale);             \
+                                       scale, window, gq, bt, bs, v_scale, sinks);      \
         else                                                                            \
             TORCH_CHECK(false, "GQA ratio > 8 not supported yet (got ", gq, ")");       \
     } while (0)
@@ -362,7 +371,8 @@ void attention_decode_paged(torch::Tensor out, torch::Tensor q,
 void attention_decode(torch::Tensor out, torch::Tensor q,
                       torch::Tensor k_cache, torch::Tensor v_cache,
                       torch::Tensor seq_lens, torch::Tensor seq_ids,
-                      double scale, long window) {
+                      double scale, long window,
+                      c10::optional<torch::Tensor> sinks_opt) {
     TORCH_CHECK(q.dim() == 4 && k_cache.dim() == 4, "bad shapes");
     const int D = q.size(3);
     const int Hq = q.size(1), Hkv = k_cache.size(1);
@@ -370,21 +380,31 @@ void attention_decode(torch::Tensor out, torch::Tensor q,
     const int gq = Hq / Hkv;
     TORCH_CHECK(q.is_contiguous() && k_cache.is_contiguous() && v_cache.is_contiguous(),
                 "contiguous tensors required");
+    torch::Tensor sinks_t;
+    const torch::Tensor* sinks = nullptr;
+    if (sinks_opt.has_value()) {
+        sinks_t = sinks_opt->to(torch::kFloat32).contiguous();
+        sinks = &sinks_t;
+    }
 
 #define CASE_T(T)                                                               \
     do {                                                                        \
         if (D == 128)                                                           \
             dispatch_gq<T, 128>(out, q, k_cache, v_cache, seq_lens, seq_ids,    \
-                                (float)scale, (int)window, gq);                 \
+                                (float)scale, (int)window, gq, nullptr, 0,      \
+                                1.0f, sinks);                                   \
         else if (D == 64)                                                       \
             dispatch_gq<T, 64>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
-                               (float)scale, (int)window, gq);                  \
+                               (float)scale, (int)window, gq, nullptr, 0,       \
+                               1.0f, sinks);                                    \
         else if (D == 32)                                                       \
             dispatch_gq<T, 32>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
-                               (float)scale, (int)window, gq);                  \
+                               (float)scale, (int)window, gq, nullptr, 0,       \
+                               1.0f, sinks);                                    \
         else if (D == 256)                                                      \
             dispatch_gq<T, 256>(out, q, k_cache, v_cache, seq_lens, seq_ids,    \
-                                (float)scale, (int)window, gq);                 \
+                                (float)scale, (int)window, gq, nullptr, 0,      \
+                                1.0f, sinks);                                   \
         else                                                                    \
             TORCH_CHECK(false, "head_dim must be 32/64/128/256, got ", D);      \
     } while (0)

@@ -53,7 +53,8 @@ __global__ __launch_bounds__(256, 2) void flash_prefill_kernel(
     const T* __restrict__ k,   // (B, Hkv, S, D)
     const T* __restrict__ v,   // (B, Hkv, S, D)
     const float scale, const int window,
-    const int Hq, const int Hkv, const int S) {
+    const int Hq, const int Hkv, const int S,
+    const float* __restrict__ sinks) {  // (Hq,) learned sink logits or null
     using MT = MfmaTraits<T>;
     using vec8 = typename MT::vec8;
     constexpr int DI = D / 16;       // O col fragments per wave
@@ -292,7 +293,10 @@ __global__ __launch_bounds__(256, 2) void flash_prefill_kernel(
         for (int r = 0; r < 4; ++r) {
             const int q_abs = wq_lo + qi * 16 + (lane >> 4) * 4 + r;
             if (q_abs >= S) continue;
-            const float inv = (l_run[qi][r] > 0.f) ? 1.f / l_run[qi][r] : 0.f;
+            float l_tot = l_run[qi][r];
+            if (sinks != nullptr)  // sink joins the denominator only
+                l_tot += __expf(sinks[hq] - m_run[qi][r]);
+            const float inv = (l_tot > 0.f) ? 1.f / l_tot : 0.f;
 #pragma unroll
             for (int di = 0; di < DI; ++di) {
                 const int dcol = di * 16 + (lane & 15);
@@ -309,13 +313,20 @@ __global__ __launch_bounds__(256, 2) void flash_prefill_kernel(
 // which matches B[k][n] with n=d. Same shape logic for the QK^T B (K tile).
 
 void flash_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
-                   torch::Tensor v, double scale, long window) {
+                   torch::Tensor v, double scale, long window,
+                   c10::optional<torch::Tensor> sinks_opt) {
     const int B = q.size(0), Hq = q.size(1), S = q.size(2), D = q.size(3);
     const int Hkv = k.size(1);
     TORCH_CHECK(Hq % Hkv == 0, "Hq must divide by Hkv");
     TORCH_CHECK(D == 64 || D == 128, "flash_prefill: head_dim 64/128 only");
     dim3 grid((S + QBLK - 1) / QBLK, Hq, B);
     auto stream = at::hip::getCurrentHIPStream();
+    torch::Tensor sinks_t;
+    const float* sinks = nullptr;
+    if (sinks_opt.has_value()) {
+        sinks_t = sinks_opt->to(torch::kFloat32).contiguous();
+        sinks = sinks_t.data_ptr<float>();
+    }
     if (q.scalar_type() == at::ScalarType::BFloat16) {
         if (D == 128)
             hipLaunchKernelGGL((flash_prefill_kernel<__hip_bfloat16, 128>), grid, dim3(256), 0,
@@ -323,25 +334,25 @@ void flash_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                                (const __hip_bfloat16*)q.data_ptr(),
                                (const __hip_bfloat16*)k.data_ptr(),
                                (const __hip_bfloat16*)v.data_ptr(),
-                               (float)scale, (int)window, Hq, Hkv, S);
+                               (float)scale, (int)window, Hq, Hkv, S, sinks);
         else
             hipLaunchKernelGGL((flash_prefill_kernel<__hip_bfloat16, 64>), grid, dim3(256), 0,
                                stream, (__hip_bfloat16*)out.data_ptr(),
                                (const __hip_bfloat16*)q.data_ptr(),
                                (const __hip_bfloat16*)k.data_ptr(),
                                (const __hip_bfloat16*)v.data_ptr(),
-                               (float)scale, (int)window, Hq, Hkv, S);
+                               (float)scale, (int)window, Hq, Hkv, S, sinks);
     } else if (q.scalar_type() == at::ScalarType::Half) {
         if (D == 128)
             hipLaunchKernelGGL((flash_prefill_kernel<__half, 128>), grid, dim3(256), 0,
                                stream, (__half*)out.data_ptr(), (const __half*)q.data_ptr(),
                                (const __half*)k.data_ptr(), (const __half*)v.data_ptr(),
-                               (float)scale, (int)window, Hq, Hkv, S);
+                               (float)scale, (int)window, Hq, Hkv, S, sinks);
         else
             hipLaunchKernelGGL((flash_prefill_kernel<__half, 64>), grid, dim3(256), 0,
                                stream, (__half*)out.data_ptr(), (const __half*)q.data_ptr(),
                                (const __half*)k.data_ptr(), (const __half*)v.data_ptr(),
-                               (float)scale, (int)window, Hq, Hkv, S);
+                               (float)scale, (int)window, Hq, Hkv, S, sinks);
     } else {
         TORCH_CHECK(false, "flash_prefill: bf16/f16 only");
     }

@@ -13,7 +13,7 @@ void block_kv_cache_update(torch::Tensor k_cache, torch::Tensor v_cache, torch::
                            torch::Tensor v_new, torch::Tensor slot_mapping);
 void attention_decode(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
                       torch::Tensor v_cache, torch::Tensor seq_lens, torch::Tensor seq_ids,
-                      double scale, long window);
+                      double scale, long window, c10::optional<torch::Tensor> sinks);
 void attention_decode_paged(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
                             torch::Tensor v_cache, torch::Tensor seq_lens,
                             torch::Tensor block_table, double scale, long window);
@@ -21,7 +21,8 @@ void topk_topp_sample(torch::Tensor out, torch::Tensor vals, torch::Tensor top_k
                       torch::Tensor top_p, torch::Tensor temp, torch::Tensor rand,
                       long global_topk);
 void flash_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
-                   torch::Tensor v, double scale, long window);
+                   torch::Tensor v, double scale, long window,
+                   c10::optional<torch::Tensor> sinks);
 void qkv_rope_update(torch::Tensor q_out, torch::Tensor qkv, torch::Tensor cosb,
                      torch::Tensor sinb, torch::Tensor k_cache, torch::Tensor v_cache,
                      torch::Tensor positions, torch::Tensor seq_ids, long Hq, long Hkv);
@@ -36,10 +37,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("rope_apply", &rope_apply, "RoPE rotate-half in place (CDNA4)");
     m.def("kv_cache_update", &kv_cache_update, "linear KV cache scatter (CDNA4)");
     m.def("block_kv_cache_update", &block_kv_cache_update, "paged KV cache scatter (CDNA4)");
-    m.def("attention_decode", &attention_decode, "split-KV decode attention (CDNA4)");
+    m.def("attention_decode", &attention_decode, "split-KV decode attention (CDNA4)",
+          py::arg("out"), py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
+          py::arg("seq_lens"), py::arg("seq_ids"), py::arg("scale"), py::arg("window"),
+          py::arg("sinks") = py::none());
     m.def("attention_decode_paged", &attention_decode_paged, "paged split-KV decode attention (CDNA4)");
     m.def("topk_topp_sample", &topk_topp_sample, "fused top-k/top-p sampling (CDNA4)");
-    m.def("flash_prefill", &flash_prefill, "MFMA causal flash prefill (CDNA4)");
+    m.def("flash_prefill", &flash_prefill, "MFMA causal flash prefill (CDNA4)",
+          py::arg("out"), py::arg("q"), py::arg("k"), py::arg("v"), py::arg("scale"),
+          py::arg("window"), py::arg("sinks") = py::none());
     m.def("qkv_rope_update", &qkv_rope_update, "fused qkv->rope->cache-write (CDNA4)");
     m.def("silu_mul_fused", &silu_mul_fused, "SiLU on fused gate|up (CDNA4)");
     m.def("gelu_tanh_mul_fused", &gelu_tanh_mul_fused, "GELU on fused gate|up (CDNA4)");

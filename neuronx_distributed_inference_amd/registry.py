@@ -41,6 +41,11 @@ _REGISTRY: Dict[str, Tuple[str, str, str]] = {
         "DbrxInferenceConfig",
         "DbrxForCausalLM",
     ),
+    "gpt_oss": (
+        "neuronx_distributed_inference_amd.models.gpt_oss.modeling_gpt_oss",
+        "GptOssInferenceConfig",
+        "GptOssForCausalLM",
+    ),
     "gemma3": (
         "neuronx_distributed_inference_amd.models.gemma3.modeling_gemma3",
         "Gemma3InferenceConfig",

@@ -191,3 +191,29 @@ def test_gemma3():
     assert not [m for m in missing if not m.startswith("kv_mgr") and ".rotary." not in m], missing
     assert not unexpected, unexpected
     _check_match(ours, hf)
+
+
+@torch.inference_mode()
+def test_gpt_oss():
+    from transformers import GptOssConfig, GptOssForCausalLM as HF
+
+    from neuronx_distributed_inference_amd.models.gpt_oss import (
+        GptOssForCausalLM, GptOssInferenceConfig,
+    )
+
+    torch.manual_seed(0)
+    hf_cfg = GptOssConfig(
+        vocab_size=128, hidden_size=64, intermediate_size=64,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, num_local_experts=4, num_experts_per_tok=2,
+        sliding_window=8, max_position_embeddings=128,
+    )
+    hf = HF(hf_cfg).eval()
+    d = hf_cfg.to_dict()
+    cfg = GptOssInferenceConfig(neuron_config=_engine(), **d)
+    ours = GptOssForCausalLM(cfg).eval()
+    sd = GptOssForCausalLM.convert_hf_to_native_state_dict(hf.state_dict(), cfg)
+    missing, unexpected = ours.model.load_state_dict(sd, strict=False)
+    assert not [m for m in missing if not m.startswith("kv_mgr")], missing
+    assert not unexpected, unexpected
+    _check_match(ours, hf)
