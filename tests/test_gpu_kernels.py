@@ -363,3 +363,37 @@ def test_kv_update_fp8_cache():
     torch.testing.assert_close(
         kc[0, :, 0:3].cpu().float(), k[0].float().cpu(), atol=0.15, rtol=0.15
     )
+
+
+def test_attention_decode_sinks_gpu():
+    torch.manual_seed(0)
+    B, Hkv, gq, Smax, D = 2, 2, 4, 128, 128
+    dev = _dev()
+    kc = torch.randn(B, Hkv, Smax, D, dtype=torch.bfloat16, device=dev)
+    vc = torch.randn_like(kc)
+    q = torch.randn(B, Hkv * gq, 1, D, dtype=torch.bfloat16, device=dev)
+    sinks = torch.randn(Hkv * gq, device=dev)
+    seq_lens = torch.tensor([64, 128], device=dev)
+    out = ops.attention_decode(q, kc, vc, seq_lens, sinks=sinks)
+    want = ops.attention_decode(
+        q.float().cpu(), kc.float().cpu(), vc.float().cpu(), seq_lens.cpu(),
+        sinks=sinks.cpu(),
+    )
+    torch.testing.assert_close(out.float().cpu(), want, atol=3e-2, rtol=3e-2)
+
+
+def test_flash_prefill_sinks_gpu():
+    torch.manual_seed(0)
+    B, Hq, Hkv, S, D = 1, 4, 2, 256, 128
+    dev = _dev()
+    q = torch.randn(B, Hq, S, D, dtype=torch.bfloat16, device=dev)
+    k = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16, device=dev)
+    v = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16, device=dev)
+    sinks = torch.randn(Hq, device=dev)
+    out = ops.attention_prefill(q, k, v, causal=True, sinks=sinks)
+    from neuronx_distributed_inference_amd.modules.attention.utils import causal_mask
+
+    want = ref.sdpa(q.float().cpu(), k.float().cpu(), v.float().cpu(),
+                    mask=causal_mask(S, S), scale=1.0 / math.sqrt(D),
+                    sinks=sinks.cpu())
+    torch.testing.assert_close(out.float().cpu(), want, atol=3e-2, rtol=3e-2)
