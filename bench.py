@@ -44,6 +44,8 @@ def parse_args():
     p.add_argument("--batch-per-gpu", type=int, default=64)
     p.add_argument("--ctx", type=int, default=128)
     p.add_argument("--no-graphs", action="store_true")
+    p.add_argument("--fp8", action="store_true",
+                   help="fp8 (e4m3) weight quantization via hipBLASLt scaled-mm")
     p.add_argument("--spec", action="store_true",
                    help="fused speculative decoding: --model target + 1b draft, spec-len 5")
     p.add_argument("--spec-len", type=int, default=5)
@@ -217,6 +219,12 @@ def main():
             if p.dim() >= 2:
                 p.normal_(0, 0.02)
     model.reset()
+    if args.fp8:
+        from neuronx_distributed_inference_amd.parallel.layers import quantize_model_fp8
+
+        nq = quantize_model_fp8(model.model)
+        if rank == 0:
+            print(f"# quantized {nq} linears to fp8", flush=True)
     model.enable_submodels()
     if on_gpu and not args.no_graphs:
         with torch.inference_mode():
@@ -279,7 +287,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": dtype,
+            "dtype": ("fp8-weights+bf16" if args.fp8 else dtype),
             "data": "synthetic",
             "config": {
                 "model": {"8b": "llama-3.1-8b", "70b": "llama-3.1-70b", "tiny": "tiny-llama-2l"}[args.model],

@@ -41,6 +41,61 @@ def divide(a: int, b: int) -> int:
     return a // b
 
 
+# ---------------------------------------------------------------------------
+# FP8 (OCP e4m3) weight quantization — the reference's quantized-checkpoint
+# path (ref application_base.py:746-799, quantized_mlp_kernel flags). Weights
+# stored fp8 with a per-output-row scale; on GPU the GEMM runs through
+# torch._scaled_mm (hipBLASLt fp8 MFMA) with dynamic per-tensor activation
+# quantization; on CPU it dequantizes (the oracle path).
+
+FP8_MAX = 448.0
+
+
+def quantize_linear_fp8(mod: nn.Module):
+    """In-place: weight -> fp8 + per-row scale; forward switches to fp8."""
+    w = mod.weight.data
+    scale = w.float().abs().amax(dim=1, keepdim=True).clamp(min=1e-8) / FP8_MAX
+    q = (w.float() / scale).clamp(-FP8_MAX, FP8_MAX).to(torch.float8_e4m3fn)
+    mod.weight = nn.Parameter(q, requires_grad=False)
+    mod.register_buffer("weight_scale", scale.float())
+    mod.fp8 = True
+
+
+def fp8_linear(x: torch.Tensor, w_fp8: torch.Tensor, w_scale: torch.Tensor, bias):
+    """y = x @ dequant(w)^T (+bias). GPU: torch._scaled_mm with dynamic
+    per-tensor activation scale; CPU: dequantized matmul."""
+    if x.is_cuda:
+        shape = x.shape[:-1] + (w_fp8.shape[0],)
+        x2 = x.reshape(-1, x.shape[-1])
+        x_scale = x2.float().abs().amax().clamp(min=1e-8) / FP8_MAX
+        x_q = (x2.float() / x_scale).clamp(-FP8_MAX, FP8_MAX).to(torch.float8_e4m3fn)
+        y = torch._scaled_mm(
+            x_q, w_fp8.t(), scale_a=x_scale.view(1, 1),
+            scale_b=w_scale.view(1, -1), bias=None, out_dtype=x.dtype,
+        )
+        if bias is not None:
+            y = y + bias
+        return y.reshape(shape)
+    w = w_fp8.float() * w_scale
+    y = F.linear(x.float(), w)
+    if bias is not None:
+        y = y + bias.float()
+    return y.to(x.dtype)
+
+
+def quantize_model_fp8(model: nn.Module, skip_names=("lm_head",)):
+    """Quantize every Column/RowParallelLinear except `skip_names`
+    substrings (ref convert()/quantize flow, model_wrapper.py:1455-1627)."""
+    n = 0
+    for name, mod in model.named_modules():
+        if isinstance(mod, (ColumnParallelLinear, RowParallelLinear)):
+            if any(sk in name for sk in skip_names):
+                continue
+            mod.quantize_fp8()
+            n += 1
+    return n
+
+
 class ColumnParallelLinear(nn.Module):
     """Y = X W^T with W sharded along the output dim across the TP group."""
 
@@ -87,11 +142,17 @@ class ColumnParallelLinear(nn.Module):
     def group(self):
         return self._group if self._group is not None else ps.get_tensor_model_parallel_group()
 
+    def quantize_fp8(self):
+        quantize_linear_fp8(self)
+
     def forward(self, x: torch.Tensor, sequence_parallel: Optional[bool] = None) -> torch.Tensor:
         sp = self.sequence_parallel if sequence_parallel is None else sequence_parallel
         if sp:
             x = mappings.gather_along_dim(x, dim=1, group=self.group)
-        y = F.linear(x, self.weight, self.bias)
+        if getattr(self, "fp8", False):
+            y = fp8_linear(x, self.weight, self.weight_scale, self.bias)
+        else:
+            y = F.linear(x, self.weight, self.bias)
         if self.gather_output:
             y = mappings.gather_from_tensor_parallel_region(y, dim=-1, group=self.group)
         return y
@@ -142,10 +203,16 @@ class RowParallelLinear(nn.Module):
     def group(self):
         return self._group if self._group is not None else ps.get_tensor_model_parallel_group()
 
+    def quantize_fp8(self):
+        quantize_linear_fp8(self)
+
     def forward(self, x: torch.Tensor, sequence_parallel: bool = False) -> torch.Tensor:
         if not self.input_is_parallel:
             x = mappings.scatter_to_tensor_parallel_region(x, dim=-1, group=self.group)
-        y = F.linear(x, self.weight)
+        if getattr(self, "fp8", False):
+            y = fp8_linear(x, self.weight, self.weight_scale, None)
+        else:
+            y = F.linear(x, self.weight)
         if self.reduce_output:
             if sequence_parallel:
                 # SP exit: reduce-scatter along S instead of all-reduce
