@@ -65,17 +65,21 @@ def fp8_linear(x: torch.Tensor, w_fp8: torch.Tensor, w_scale: torch.Tensor, bias
     """y = x @ dequant(w)^T (+bias). GPU: torch._scaled_mm with dynamic
     per-tensor activation scale; CPU: dequantized matmul."""
     if x.is_cuda:
+        # hipBLASLt scaled-mm takes per-tensor scales on this stack; apply
+        # the per-row weight scale on the (small M) output instead
         shape = x.shape[:-1] + (w_fp8.shape[0],)
         x2 = x.reshape(-1, x.shape[-1])
         x_scale = x2.float().abs().amax().clamp(min=1e-8) / FP8_MAX
         x_q = (x2.float() / x_scale).clamp(-FP8_MAX, FP8_MAX).to(torch.float8_e4m3fn)
+        one = torch.ones((1, 1), device=x.device, dtype=torch.float32)
         y = torch._scaled_mm(
-            x_q, w_fp8.t(), scale_a=x_scale.view(1, 1),
-            scale_b=w_scale.view(1, -1), bias=None, out_dtype=x.dtype,
+            x_q, w_fp8.t(), scale_a=x_scale.view(1, 1), scale_b=one,
+            bias=None, out_dtype=torch.float32,
         )
+        y = y * w_scale.view(1, -1)
         if bias is not None:
-            y = y + bias
-        return y.reshape(shape)
+            y = y + bias.float()
+        return y.to(x.dtype).reshape(shape)
     w = w_fp8.float() * w_scale
     y = F.linear(x.float(), w)
     if bias is not None:
