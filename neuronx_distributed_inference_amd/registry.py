@@ -46,6 +46,11 @@ _REGISTRY: Dict[str, Tuple[str, str, str]] = {
         "GptOssInferenceConfig",
         "GptOssForCausalLM",
     ),
+    "deepseek_v3": (
+        "neuronx_distributed_inference_amd.models.deepseek.modeling_deepseek",
+        "DeepseekV3InferenceConfig",
+        "DeepseekV3ForCausalLM",
+    ),
     "gemma3": (
         "neuronx_distributed_inference_amd.models.gemma3.modeling_gemma3",
         "Gemma3InferenceConfig",

@@ -217,3 +217,33 @@ def test_gpt_oss():
     assert not [m for m in missing if not m.startswith("kv_mgr")], missing
     assert not unexpected, unexpected
     _check_match(ours, hf)
+
+
+@torch.inference_mode()
+def test_deepseek_v3():
+    from transformers import DeepseekV3Config, DeepseekV3ForCausalLM as HF
+
+    from neuronx_distributed_inference_amd.models.deepseek import (
+        DeepseekV3ForCausalLM, DeepseekV3InferenceConfig,
+    )
+
+    torch.manual_seed(0)
+    hf_cfg = DeepseekV3Config(
+        vocab_size=128, hidden_size=64, intermediate_size=128,
+        moe_intermediate_size=32, num_hidden_layers=3,
+        num_attention_heads=4, num_key_value_heads=4,
+        n_routed_experts=4, num_experts_per_tok=2, n_shared_experts=1,
+        n_group=2, topk_group=1, routed_scaling_factor=2.5,
+        q_lora_rank=32, kv_lora_rank=16, qk_rope_head_dim=8,
+        qk_nope_head_dim=16, v_head_dim=16, first_k_dense_replace=1,
+        max_position_embeddings=128,
+    )
+    hf = HF(hf_cfg).eval()
+    d = hf_cfg.to_dict()
+    cfg = DeepseekV3InferenceConfig(neuron_config=_engine(), **d)
+    ours = DeepseekV3ForCausalLM(cfg).eval()
+    sd = DeepseekV3ForCausalLM.convert_hf_to_native_state_dict(hf.state_dict(), cfg)
+    missing, unexpected = ours.model.load_state_dict(sd, strict=False)
+    assert not [m for m in missing if not m.startswith("kv_mgr")], missing
+    assert not unexpected, unexpected
+    _check_match(ours, hf)
