@@ -211,6 +211,8 @@ class DecoderModel(nn.Module):
         else:
             h_last = h[:, -1:, :]
 
+        # kept for medusa/eagle heads (graph-safe: a static tensor each capture)
+        self.last_hidden = h_last
         logits_shard = self.lm_head(h_last)  # (B, 1|S, V/tp)
 
         tokens = None

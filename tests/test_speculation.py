@@ -116,3 +116,32 @@ def test_fused_spec_matches_target_only():
         ids, generation_config=GenerationConfig(max_new_tokens=24))
     n = min(got.shape[1], want.shape[1])
     assert torch.equal(got[:, :n], want[:, :n]), (got.tolist(), want.tolist())
+
+
+@torch.inference_mode()
+def test_medusa_matches_target_only():
+    """Chain-medusa (random heads => mostly rejected) must still equal the
+    target-only greedy stream."""
+    from neuronx_distributed_inference_amd.modules.medusa import (
+        MedusaGenerationAdapter,
+        MedusaHeads,
+    )
+
+    target = _build(64, 2, seed=1, spec_len=3)
+    torch.manual_seed(9)
+    heads = MedusaHeads(64, 64, num_heads=3, dtype=torch.float32)
+    with torch.no_grad():
+        for p in heads.parameters():
+            p.normal_(0, 0.05)
+    ids = torch.randint(0, 64, (2, 8))
+    want = HuggingFaceGenerationAdapter(_build(64, 2, seed=1)).generate(
+        ids, generation_config=GenerationConfig(max_new_tokens=20))
+    got = MedusaGenerationAdapter(target, heads).generate(
+        ids, generation_config=GenerationConfig(max_new_tokens=20))
+    for i in range(2):
+        row = got[i]
+        n = int((row != 0).float().flip(0).argmax())
+        n = row.shape[0] - n
+        n = min(n, want.shape[1])
+        assert n >= ids.shape[1] + 8
+        assert torch.equal(row[:n], want[i, :n]), (row.tolist(), want[i].tolist())
