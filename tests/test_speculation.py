@@ -145,3 +145,32 @@ def test_medusa_matches_target_only():
         n = min(n, want.shape[1])
         assert n >= ids.shape[1] + 8
         assert torch.equal(row[:n], want[i, :n]), (row.tolist(), want[i].tolist())
+
+
+@torch.inference_mode()
+def test_eagle_matches_target_only():
+    """EAGLE chain (random draft => mostly rejected) must equal the
+    target-only greedy stream."""
+    from neuronx_distributed_inference_amd.modules.eagle import (
+        EagleDraftModel,
+        EagleGenerationAdapter,
+    )
+
+    target = _build(64, 2, seed=1, spec_len=3)
+    draft_lm = _build(64, 1, seed=7)
+    torch.manual_seed(8)
+    draft = EagleDraftModel(draft_lm.model, hidden_size=64, dtype=torch.float32)
+    with torch.no_grad():
+        draft.fc.weight.normal_(0, 0.05)
+    ids = torch.randint(0, 64, (2, 8))
+    want = HuggingFaceGenerationAdapter(_build(64, 2, seed=1)).generate(
+        ids, generation_config=GenerationConfig(max_new_tokens=20))
+    got = EagleGenerationAdapter(target, draft, speculation_length=3).generate(
+        ids, generation_config=GenerationConfig(max_new_tokens=20))
+    for i in range(2):
+        row = got[i]
+        n = int((row != 0).float().flip(0).argmax())
+        n = row.shape[0] - n
+        n = min(n, want.shape[1])
+        assert n >= ids.shape[1] + 8
+        assert torch.equal(row[:n], want[i, :n]), (row.tolist(), want[i].tolist())
