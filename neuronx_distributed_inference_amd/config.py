@@ -230,6 +230,9 @@ class EngineConfig:
         self.mlp_kernel_enabled: bool = kwargs.pop("mlp_kernel_enabled", True)
         self.sampling_kernel_enabled: bool = kwargs.pop("sampling_kernel_enabled", True)
         self.kv_update_kernel_enabled: bool = kwargs.pop("kv_update_kernel_enabled", True)
+        self.windowed_context_encoding_size: int = kwargs.pop(
+            "windowed_context_encoding_size", 0
+        )
         self.logical_nc_config: int = kwargs.pop("logical_nc_config", 1)  # accepted, unused
         self.async_mode: bool = kwargs.pop("async_mode", False)
         self.on_cpu: bool = kwargs.pop("on_cpu", False)

@@ -370,6 +370,19 @@ class CausalLMBase(nn.Module):
             seq_ids = torch.arange(B, device=input_ids.device)
         # route through the submodel wrappers (bucketing + hipGraph replay)
         # when enabled, else call the module directly
+        # windowed context encoding (ref model_base.py:878-933): prefill in
+        # window-sized chunks to cap the attention working set; each chunk
+        # runs the multi-token prefix path (causal within the chunk, full
+        # prior from the cache), which is also the chunked-prefill compute.
+        wce = getattr(self.ec, "windowed_context_encoding_size", 0)
+        if (
+            S > 1 and not self.kv_cache_populated and wce and S > wce
+            and attention_mask is not None
+        ):
+            return self._windowed_context_encoding(
+                input_ids, attention_mask, position_ids, seq_ids, sampling_params, wce
+            )
+
         is_spec_step = S > 1 and self.kv_cache_populated
         if is_spec_step and self.speculation_model is not None:
             tokens, logits = self.speculation_model(
@@ -394,6 +407,37 @@ class CausalLMBase(nn.Module):
             )
         self.kv_cache_populated = True
         return ModelOutput(tokens=tokens, logits=logits)
+
+    def _windowed_context_encoding(
+        self, input_ids, attention_mask, position_ids, seq_ids, sampling_params,
+        window: int,
+    ) -> "ModelOutput":
+        """Prefill in window-sized chunks (ref windowed CTE,
+        model_base.py:878-933). Every chunk runs the multi-token prefix path
+        (greedy per-position tokens); the per-row next token is gathered at
+        each row's true last index."""
+        from neuronx_distributed_inference_amd.modules.attention.attention_base import (
+            PHASE_TKG,
+        )
+
+        B, S = input_ids.shape
+        chunks = []
+        start = 0
+        while start < S:
+            end = min(start + window, S)
+            t, l = self.model(
+                input_ids[:, start:end], None, position_ids[:, start:end],
+                seq_ids, sampling_params, phase=PHASE_TKG,
+            )
+            if t is None:  # no on-device sampler configured: greedy off logits
+                t = l.argmax(dim=-1)
+            chunks.append(t.view(B, end - start))
+            start = end
+        all_tok = torch.cat(chunks, dim=1)              # (B, S) greedy per pos
+        last = (attention_mask.sum(dim=-1).long() - 1).clamp(min=0)
+        tokens = all_tok.gather(1, last.view(B, 1)).view(B)
+        self.kv_cache_populated = True
+        return ModelOutput(tokens=tokens, logits=None)
 
     def reset(self):
         """Clear KV state (ref model_base.py:3926)."""

@@ -141,3 +141,33 @@ def test_fused_projections_match_unfused(tiny_llama_config):
     b = HuggingFaceGenerationAdapter(fused).generate(
         ids, generation_config=GenerationConfig(max_new_tokens=8))
     assert torch.equal(a, b)
+
+
+@torch.inference_mode()
+def test_windowed_context_encoding_matches_direct(tiny_llama_config):
+    """Prefill chunked through the windowed CTE path must generate the same
+    tokens as one-shot prefill (incl. a ragged row)."""
+    import copy
+
+    torch.manual_seed(0)
+    base = LlamaForCausalLM(tiny_llama_config).eval()
+    with torch.no_grad():
+        for p in base.parameters():
+            if p.dim() >= 2:
+                p.normal_(0, 0.1)
+
+    cfg2 = copy.deepcopy(tiny_llama_config)
+    cfg2.neuron_config.windowed_context_encoding_size = 5
+    wce = LlamaForCausalLM(cfg2).eval()
+    wce.model.load_state_dict(
+        {k: v for k, v in base.model.state_dict().items() if not k.startswith("kv_mgr.")},
+        strict=False,
+    )
+    ids = torch.randint(0, 128, (2, 13))
+    mask = torch.ones_like(ids)
+    mask[1, 9:] = 0  # ragged row
+    a = HuggingFaceGenerationAdapter(base).generate(
+        ids, attention_mask=mask, generation_config=GenerationConfig(max_new_tokens=8))
+    b = HuggingFaceGenerationAdapter(wce).generate(
+        ids, attention_mask=mask, generation_config=GenerationConfig(max_new_tokens=8))
+    assert torch.equal(a, b), (a.tolist(), b.tolist())
