@@ -168,7 +168,7 @@ class DeepseekV3Attention(nn.Module):
             out = ref_ops.sdpa(q_states, k, v, mask=mask, scale=self.scaling)
         else:
             kc, _ = kv_mgr.get_cache(self.layer_idx)
-            seq_lens = position_ids[:, -1] + 1
+            seq_lens = position_ids.max(dim=-1).values + 1
             L = int(seq_lens.max())
             lat = kc[:B, :, :L, :].to(h.dtype)              # (B,1,L,rank+rope)
             kv_n, k_r = torch.split(lat, [self.kv_lora_rank, self.rope_dim], dim=-1)

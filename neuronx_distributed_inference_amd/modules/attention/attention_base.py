@@ -176,7 +176,8 @@ class AttentionBase(nn.Module):
 
     def _decode(self, q, kv_mgr, position_ids, seq_ids, active_mask) -> torch.Tensor:
         k_cache, v_cache = kv_mgr.get_cache(self.layer_idx)
-        seq_lens = position_ids[:, -1] + 1  # attend to everything written so far
+        # row max (not last column): ragged chunks carry -1 pad positions
+        seq_lens = position_ids.max(dim=-1).values + 1
         if getattr(kv_mgr, "is_paged", False):
             return ops.attention_decode_paged(
                 q, k_cache.data, v_cache.data, seq_lens, kv_mgr.block_table,
