@@ -168,15 +168,15 @@ class DeepseekV3Attention(nn.Module):
             out = ref_ops.sdpa(q_states, k, v, mask=mask, scale=self.scaling)
         else:
             kc, _ = kv_mgr.get_cache(self.layer_idx)
-            seq_lens = position_ids.max(dim=-1).values + 1
-            L = int(seq_lens.max())
+            L = int(position_ids.max()) + 1
             lat = kc[:B, :, :L, :].to(h.dtype)              # (B,1,L,rank+rope)
             kv_n, k_r = torch.split(lat, [self.kv_lora_rank, self.rope_dim], dim=-1)
             k, v = self._expand(kv_n, k_r)
             ar = torch.arange(L, device=h.device).view(1, 1, 1, L)
-            qpos = seq_lens.view(B, 1, 1, 1) - (S - 1) + torch.arange(S, device=h.device).view(1, 1, S, 1)
-            mask = ar <= (qpos - 1)
+            qpos = position_ids.view(B, 1, S, 1)   # per-query positions
+            mask = ar <= qpos
             out = ref_ops.sdpa(q_states, k, v, mask=mask, scale=self.scaling)
+            out = torch.nan_to_num(out)
 
         out = out.transpose(1, 2).reshape(B, S, -1)
         return self.o_proj(out)

@@ -176,14 +176,13 @@ class AttentionBase(nn.Module):
 
     def _decode(self, q, kv_mgr, position_ids, seq_ids, active_mask) -> torch.Tensor:
         k_cache, v_cache = kv_mgr.get_cache(self.layer_idx)
-        # row max (not last column): ragged chunks carry -1 pad positions
-        seq_lens = position_ids.max(dim=-1).values + 1
+        # per-query positions (B,Sq): exact for ragged chunks (-1 pads)
         if getattr(kv_mgr, "is_paged", False):
             return ops.attention_decode_paged(
-                q, k_cache.data, v_cache.data, seq_lens, kv_mgr.block_table,
+                q, k_cache.data, v_cache.data, position_ids, kv_mgr.block_table,
                 scale=self.scale, sliding_window=self.sliding_window,
             )
         return ops.attention_decode(
-            q, k_cache, v_cache, seq_lens, seq_ids=seq_ids, scale=self.scale,
+            q, k_cache, v_cache, position_ids, seq_ids=seq_ids, scale=self.scale,
             sinks=self.sinks, sliding_window=self.sliding_window,
         )

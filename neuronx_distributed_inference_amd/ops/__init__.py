@@ -158,38 +158,45 @@ def attention_decode(q, k_cache, v_cache, seq_lens, seq_ids=None, scale=None,
     """Token-generation attention against the linear KV cache.
 
     q: (B,Hq,Sq,D) with small Sq (1 or spec_len); k_cache/v_cache:
-    (B_cache,Hkv,Smax,D); seq_lens: (B,) valid prior length per sequence
-    (the new token's KV must already be written). K2-analog decode kernel.
+    (B_cache,Hkv,Smax,D). `seq_lens` is either (B,) valid lengths (queries
+    assumed to be the last Sq positions) or (B,Sq) per-query absolute
+    POSITIONS (ragged chunks use -1 for pads -> zero output rows).
+    K2-analog decode kernel.
     """
     import math
     scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+    B, Hq, Sq, D = q.shape
+    if seq_lens.dim() == 1:
+        positions = seq_lens.view(B, 1) - Sq + torch.arange(
+            Sq, device=seq_lens.device
+        ).view(1, Sq)
+    else:
+        positions = seq_lens
     if q.is_cuda:
         c = _require_ext("attention_decode")
         out = torch.empty_like(q)
-        B = q.shape[0]
         if seq_ids is None:
             seq_ids = torch.arange(B, device=q.device, dtype=torch.int32)
         c.attention_decode(
             out, q.contiguous(), k_cache, v_cache,
-            seq_lens.to(torch.int32).contiguous(),
+            positions.to(torch.int32).contiguous(),
             seq_ids.to(torch.int32).contiguous(), float(scale), int(sliding_window),
             sinks=None if sinks is None else sinks.detach(),
         )
         return out
-    # CPU reference: mask by per-sequence length (+ causal within Sq)
-    B, Hq, Sq, D = q.shape
+    # CPU reference: mask each query by its own position
     if seq_ids is None:
         seq_ids = torch.arange(B, device=q.device)
     k = k_cache[seq_ids.long()]
     v = v_cache[seq_ids.long()]
     Smax = k.shape[2]
     ar = torch.arange(Smax, device=q.device).view(1, 1, 1, Smax)
-    # query token i (of Sq) attends to positions < seq_lens[b] - (Sq-1-i) + 1
-    qpos = seq_lens.view(B, 1, 1, 1) - (Sq - 1) + torch.arange(Sq, device=q.device).view(1, 1, Sq, 1)
-    mask = ar <= (qpos - 1)
+    qpos = positions.view(B, 1, Sq, 1)
+    mask = ar <= qpos
     if sliding_window:
-        mask = mask & (ar > (qpos - 1 - sliding_window))
-    return ref.sdpa(q, k, v, mask=mask, scale=scale, sinks=sinks)
+        mask = mask & (ar > (qpos - sliding_window))
+    out = ref.sdpa(q, k, v, mask=mask, scale=scale, sinks=sinks)
+    return torch.nan_to_num(out)  # pad queries (pos<0) mask everything
 
 
 def topk_topp_sample(logits, top_k, top_p, temperature, rand, global_topk: int = 256):
@@ -262,26 +269,34 @@ def gelu_tanh_mul_fused(gate_up):
 def attention_decode_paged(q, k_cache, v_cache, seq_lens, block_table,
                            scale=None, sliding_window: int = 0):
     """Decode attention against a paged (num_blocks, block_size, Hkv, D)
-    cache via a (B, max_blocks) block table."""
+    cache via a (B, max_blocks) block table. `seq_lens`: (B,) lengths or
+    (B,Sq) per-query positions."""
     import math
     scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+    B, Hq, Sq, D = q.shape
+    if seq_lens.dim() == 1:
+        positions = seq_lens.view(B, 1) - Sq + torch.arange(
+            Sq, device=seq_lens.device
+        ).view(1, Sq)
+    else:
+        positions = seq_lens
     if q.is_cuda:
         c = _require_ext("attention_decode_paged")
         out = torch.empty_like(q)
         c.attention_decode_paged(
             out, q.contiguous(), k_cache, v_cache,
-            seq_lens.to(torch.int32).contiguous(),
+            positions.to(torch.int32).contiguous(),
             block_table.to(torch.int32).contiguous(), float(scale),
             int(sliding_window),
         )
         return out
     # CPU reference: gather contiguous K/V per sequence then mask-attend
-    max_len = int(seq_lens.max())
+    max_len = int(positions.max()) + 1
     k, v = ref.paged_gather_kv(k_cache, v_cache, block_table, max_len)
-    B, Hq, Sq, D = q.shape
     ar = torch.arange(max_len).view(1, 1, 1, -1)
-    qpos = seq_lens.view(B, 1, 1, 1) - (Sq - 1) + torch.arange(Sq).view(1, 1, Sq, 1)
-    mask = ar <= (qpos - 1)
+    qpos = positions.view(B, 1, Sq, 1)
+    mask = ar <= qpos
     if sliding_window:
-        mask = mask & (ar > (qpos - 1 - sliding_window))
-    return ref.sdpa(q, k, v, mask=mask, scale=scale)
+        mask = mask & (ar > (qpos - sliding_window))
+    out = ref.sdpa(q, k, v, mask=mask, scale=scale)
+    return torch.nan_to_num(out)

@@ -35,7 +35,7 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
     const T* __restrict__ q,           // (B, Hq, Sq, D) — Hq = Hkv*gq
     const CT* __restrict__ k_cache,    // (Bc, Hkv, Smax, D)
     const CT* __restrict__ v_cache,
-    const int* __restrict__ seq_lens,  // (B,)
+    const int* __restrict__ positions, // (B, Sq) per-query absolute position
     const int* __restrict__ seq_ids,   // (B,) (linear) / unused (paged)
     const int* __restrict__ block_table,  // (B, max_blocks) when PAGED
     const float scale, const int window,
@@ -58,7 +58,8 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
     const int c = threadIdx.x % GS;           // lane-in-group
 
     const int row = seq_ids[b];
-    const int eff_len = seq_lens[b] - (Sq - 1 - sq);
+    // query at absolute position p attends cache rows [0, p]
+    const int eff_len = positions[(long)b * Sq + sq] + 1;
     int lo = 0;
     if (window > 0 && eff_len > window) lo = eff_len - window;
     const int split_lo = max(lo, split * chunk);
@@ -335,8 +336,11 @@ void attention_decode_paged(torch::Tensor out, torch::Tensor q,
                             torch::Tensor k_cache, torch::Tensor v_cache,
                             torch::Tensor seq_lens, torch::Tensor block_table,
                             double scale, long window) {
-    // paged cache (num_blocks, block_size, Hkv, D)
+    // paged cache (num_blocks, block_size, Hkv, D); `seq_lens` carries
+    // per-query POSITIONS, shape (B, Sq)
     TORCH_CHECK(q.dim() == 4 && k_cache.dim() == 4, "bad shapes");
+    TORCH_CHECK(seq_lens.numel() == q.size(0) * q.size(2),
+                "positions must be (B, Sq)");
     const int D = q.size(3);
     const int Hq = q.size(1), Hkv = k_cache.size(2);
     const int bs = k_cache.size(1);
@@ -373,7 +377,10 @@ void attention_decode(torch::Tensor out, torch::Tensor q,
                       torch::Tensor seq_lens, torch::Tensor seq_ids,
                       double scale, long window,
                       c10::optional<torch::Tensor> sinks_opt) {
+    // `seq_lens` carries per-query POSITIONS, shape (B, Sq)
     TORCH_CHECK(q.dim() == 4 && k_cache.dim() == 4, "bad shapes");
+    TORCH_CHECK(seq_lens.numel() == q.size(0) * q.size(2),
+                "positions must be (B, Sq)");
     const int D = q.size(3);
     const int Hq = q.size(1), Hkv = k_cache.size(1);
     TORCH_CHECK(Hq % Hkv == 0, "Hq must be a multiple of Hkv");
