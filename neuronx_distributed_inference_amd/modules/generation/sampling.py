@@ -117,3 +117,55 @@ class Sampler(nn.Module):
 
 def create_sampler(config: OnDeviceSamplingConfig, vocab_size: int) -> Sampler:
     return Sampler(config, vocab_size)
+
+
+class DataParallelSampler(nn.Module):
+    """Batch-sharded sampling (ref DataParallelSampler, sampling.py:467-579):
+    all-to-all reshards (full batch, vocab shard) -> (batch shard, wider
+    vocab), local top-k, payload gather to full vocab candidates, final
+    top-k + fused sample, then all-gather the batch back."""
+
+    def __init__(self, config: OnDeviceSamplingConfig, vocab_size: int,
+                 dp_degree: int):
+        super().__init__()
+        self.inner = Sampler(config, vocab_size)
+        self.config = config
+        self.dp = dp_degree
+        self.vocab_size = vocab_size
+
+    def forward(self, logits: torch.Tensor, sampling_params=None) -> torch.Tensor:
+        from neuronx_distributed_inference_amd.parallel import mappings
+
+        if logits.dim() == 3:
+            logits = logits[:, -1, :]
+        B = logits.shape[0]
+        import torch.distributed as dist
+
+        if self.dp <= 1 or not dist.is_initialized() or B % self.dp != 0:
+            return self.inner(logits, sampling_params)
+        # (B, V/tp) -> (B/dp, dp*V/tp): shard batch, widen local vocab.
+        # After the reshard each rank holds the FULL vocab for its batch
+        # shard, so sampling is purely local (no distributed argmax/topk).
+        resharded = mappings.all_to_all(logits, scatter_dim=0, gather_dim=1)
+        rank = dist.get_rank(group=mappings.ps.get_tensor_model_parallel_group())
+        per = B // self.dp
+        if self.inner.static_greedy:
+            local = resharded.argmax(dim=-1)
+        else:
+            sp = sampling_params
+            if sp is None:
+                sp = prepare_sampling_params(
+                    B, self.config.top_k if self.config.do_sample else 1,
+                    self.config.top_p, self.config.temperature,
+                    device=logits.device,
+                )
+            sp = sp[(rank % self.dp) * per:((rank % self.dp) + 1) * per]
+            k = min(self.config.global_topk, resharded.shape[-1])
+            vals, idxs = resharded.topk(k, dim=-1)
+            rand = self.inner._rand(per, logits.device)
+            pick = ops.topk_topp_sample(
+                vals.float(), sp[:, 0], sp[:, 1], sp[:, 2], rand, global_topk=k
+            )
+            local = idxs.gather(1, pick.view(-1, 1)).squeeze(1)
+        # all-gather the batch shards back (every rank needs the full batch)
+        return mappings.gather_along_dim(local.view(-1, 1), dim=0).view(-1)

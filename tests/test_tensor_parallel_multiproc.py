@@ -305,3 +305,31 @@ def _worker_ep_mixtral(rank):
 
 def test_expert_parallel_mixtral():
     spawn("_worker_ep_mixtral")
+
+
+# ---------------------------------------------------------------------------
+def _worker_dp_sampler(rank):
+    """DataParallelSampler (batch<->vocab all-to-all) matches the plain
+    distributed sampler's greedy result."""
+    from neuronx_distributed_inference_amd.config import OnDeviceSamplingConfig
+    from neuronx_distributed_inference_amd.modules.generation.sampling import (
+        DataParallelSampler,
+        Sampler,
+    )
+    from neuronx_distributed_inference_amd.parallel import state as ps
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=WORLD)
+    torch.manual_seed(17)
+    V, B = 40, 4
+    full = torch.randn(B, V)
+    shard = full[:, rank * (V // WORLD):(rank + 1) * (V // WORLD)]
+    cfg = OnDeviceSamplingConfig(top_k=1)
+    want = Sampler(cfg, V)(shard)
+    got = DataParallelSampler(cfg, V, dp_degree=WORLD)(shard)
+    assert torch.equal(got, full.argmax(-1)), (got, full.argmax(-1))
+    assert torch.equal(want, got)
+    return True
+
+
+def test_data_parallel_sampler():
+    spawn("_worker_dp_sampler")
