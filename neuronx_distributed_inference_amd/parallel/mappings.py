@@ -114,6 +114,14 @@ def all_to_all(x: torch.Tensor, scatter_dim: int, gather_dim: int, group=None) -
     world = _size(group)
     if world == 1:
         return x
+    if dist.get_backend(group) == "gloo":
+        # gloo has no alltoall: emulate via all-gather + local select (CPU
+        # test path only; RCCL uses the native collective)
+        rank = _rank(group)
+        gathered = [torch.empty_like(x) for _ in range(world)]
+        dist.all_gather(gathered, x.contiguous(), group=group)
+        picks = [g.chunk(world, dim=scatter_dim)[rank] for g in gathered]
+        return torch.cat(picks, dim=gather_dim)
     inputs = [t.contiguous() for t in x.chunk(world, dim=scatter_dim)]
     outputs = [torch.empty_like(inputs[0]) for _ in range(world)]
     dist.all_to_all(outputs, inputs, group=group)
