@@ -175,3 +175,102 @@ def wrap_model_with_lora(model: nn.Module, cfg: LoraServingConfig,
 
 def lora_layers(model: nn.Module) -> Dict[str, MultiLoraLinear]:
     return {n: m for n, m in model.named_modules() if isinstance(m, MultiLoraLinear)}
+
+
+class AdapterCache:
+    """Dynamic multi-LoRA: CPU-resident adapters with LRU slot assignment;
+    requested adapters are swapped into live device slots between requests
+    (ref dynamic multi-LoRA, lora_model.py:263-682,
+    application_base.py:401-419)."""
+
+    def __init__(self, model: nn.Module, cfg: LoraServingConfig, state: LoraState,
+                 tp_rank: int = 0, tp_degree: int = 1):
+        self.layers = lora_layers(model)
+        self.cfg = cfg
+        self.state = state
+        self.tp_rank = tp_rank
+        self.tp_degree = tp_degree
+        self.cpu_store: Dict[str, Dict[str, tuple]] = {}
+        # slot 0 is the permanent zero adapter ("no LoRA")
+        self.slot_of: Dict[str, int] = {}
+        self.lru: List[str] = []
+        self.free_slots = list(range(1, cfg.max_loras))
+
+    def register_adapter(self, name: str, weights: Dict[str, tuple]):
+        """weights: {layer_name: (A (r,in_full), B (out_full,r))} on CPU."""
+        self.cpu_store[name] = weights
+
+    def load_adapter_dir(self, name: str, path: str):
+        """Read a PEFT-style safetensors adapter: keys
+        `...<target>.lora_A.weight` / `...<target>.lora_B.weight`."""
+        from safetensors.torch import load_file
+        import glob as _glob
+        import os as _os
+
+        files = _glob.glob(_os.path.join(path, "*.safetensors"))
+        sd = {}
+        for f in files:
+            sd.update(load_file(f))
+        weights: Dict[str, tuple] = {}
+        for k, v in sd.items():
+            if ".lora_A." in k:
+                base = k.split(".lora_A.")[0]
+                bkey = k.replace(".lora_A.", ".lora_B.")
+                if bkey in sd:
+                    weights[base] = (v, sd[bkey])
+        self.register_adapter(name, weights)
+
+    def _match_layer(self, adapter_layer_name: str):
+        for lname, layer in self.layers.items():
+            if adapter_layer_name.endswith(lname) or lname.endswith(adapter_layer_name) \
+                    or adapter_layer_name.split(".")[-1] == lname.split(".")[-2]:
+                yield lname, layer
+
+    def ensure_loaded(self, names: List[str]) -> Dict[str, int]:
+        """Make every named adapter resident; returns name -> slot. Evicts
+        least-recently-used residents when slots run out."""
+        out = {}
+        for name in names:
+            if name in self.slot_of:
+                self.lru.remove(name)
+                self.lru.append(name)
+                out[name] = self.slot_of[name]
+                continue
+            if name not in self.cpu_store:
+                raise KeyError(f"unknown adapter {name!r}")
+            if self.free_slots:
+                slot = self.free_slots.pop(0)
+            else:
+                evict = self.lru.pop(0)
+                slot = self.slot_of.pop(evict)
+            self._install(name, slot)
+            self.slot_of[name] = slot
+            self.lru.append(name)
+            out[name] = slot
+        return out
+
+    def _install(self, name: str, slot: int):
+        weights = self.cpu_store[name]
+        for lname, layer in self.layers.items():
+            hit = None
+            for aname, ab in weights.items():
+                if lname.endswith(aname) or aname.split(".")[-1] == lname.split(".")[-2] \
+                        or aname.split(".")[-1] == lname.split(".")[-1]:
+                    hit = ab
+                    break
+            with torch.no_grad():
+                if hit is None:
+                    layer.lora_A[slot].zero_()
+                    layer.lora_B[slot].zero_()
+                else:
+                    a, b = hit
+                    layer.load_adapter(slot, a, b, self.tp_rank, self.tp_degree)
+
+    def set_request_adapters(self, names: List[str]):
+        """Map a batch's adapter names ('' = none) to slots and install the
+        per-row adapter_ids buffer."""
+        uniq = [n for n in dict.fromkeys(names) if n]
+        mapping = self.ensure_loaded(uniq)
+        ids = torch.tensor([mapping.get(n, 0) for n in names], dtype=torch.long)
+        self.state.set(ids)
+        return ids

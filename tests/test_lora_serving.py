@@ -84,3 +84,30 @@ def test_single_layer_delta_math():
     got = wrapped(x)
     want = torch.nn.functional.linear(x, base.weight) + (x @ a.T @ b.T) * (4.0 / 4)
     torch.testing.assert_close(got, want, atol=1e-5, rtol=1e-5)
+
+
+@torch.inference_mode()
+def test_adapter_cache_lru(model):
+    from neuronx_distributed_inference_amd.modules.lora_serving import AdapterCache
+
+    cfg = LoraServingConfig(max_loras=3, max_lora_rank=4)  # 2 dynamic slots
+    state = wrap_model_with_lora(model.model, cfg, max_batch=2)
+    cache = AdapterCache(model.model, cfg, state)
+    layers = lora_layers(model.model)
+    any_layer = next(iter(layers))
+    torch.manual_seed(0)
+
+    def mk():
+        layer = layers[any_layer]
+        return {any_layer: (torch.randn(4, layer.lora_A.shape[2]),
+                            torch.randn(layer.lora_B.shape[1], 4))}
+
+    for nm in ("a", "b", "c"):
+        cache.register_adapter(nm, mk())
+    m1 = cache.ensure_loaded(["a", "b"])
+    assert set(m1.values()) == {1, 2}
+    m2 = cache.ensure_loaded(["c"])       # evicts "a" (LRU)
+    assert m2["c"] == m1["a"]
+    assert "a" not in cache.slot_of
+    ids = cache.set_request_adapters(["b", ""])
+    assert ids[1] == 0  # '' -> zero adapter slot
