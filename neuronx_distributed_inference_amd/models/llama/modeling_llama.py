@@ -27,7 +27,8 @@ def fuse_qkv_keys(sd: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
     """cat q|k|v weights (and biases) into qkv_proj.qkv.* keys."""
     out = dict(sd)
     for suffix in ("weight", "bias"):
-        qs = [k for k in sd if k.endswith(f"qkv_proj.q_proj.{suffix}")]
+        qs = [k for k in sd if k.endswith(f"qkv_proj.q_proj.{suffix}")
+              and "cte_qkv_proj" not in k]
         for qk in qs:
             base = qk[: -len(f"q_proj.{suffix}")]
             kk, vk = f"{base}k_proj.{suffix}", f"{base}v_proj.{suffix}"
@@ -137,6 +138,14 @@ class LlamaForCausalLM(CausalLMBase):
             nk = nk.replace("self_attn.v_proj", "self_attn.qkv_proj.v_proj")
             nk = nk.replace("self_attn.o_proj", "self_attn.o_proj.o_proj")
             out[nk] = v
+        if ec.cp_degree > 1:
+            # duplicate the attention projections for the CP prefill set
+            # (same full weights, sharded over the prefill-TP subgroup)
+            for k in list(out):
+                if ".self_attn.qkv_proj." in k:
+                    out[k.replace(".qkv_proj.", ".cte_qkv_proj.")] = out[k]
+                elif ".self_attn.o_proj.o_proj." in k:
+                    out[k.replace(".o_proj.o_proj.", ".cte_o_proj.o_proj.")] = out[k]
         if ec.fused_qkv:
             out = fuse_qkv_keys(out)
         if ec.fused_gate_up:

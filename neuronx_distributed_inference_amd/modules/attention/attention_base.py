@@ -61,6 +61,25 @@ class AttentionBase(nn.Module):
         )
         self.num_heads_per_rank = self.qkv_proj.num_heads_per_rank
         self.num_kv_heads_per_rank = self.qkv_proj.num_kv_heads_per_rank
+
+        # Context parallelism (ref §2.8 CP + dual projection sets,
+        # attention_base.py:326-421): prefill attention runs in tp/cp-sized
+        # TP subgroups over a SECOND projection set sharded over the prefill
+        # group; each CP rank computes its S/cp query slice against the full
+        # K/V (activations are TP-replicated, so no KV all-gather is needed)
+        # and the outputs are all-gathered along S over the CP group.
+        self.cp_degree = ec.cp_degree
+        if self.cp_degree > 1:
+            tpc = self.tp_degree // self.cp_degree
+            self.cte_qkv_proj = GroupQueryAttentionQKV(
+                self.hidden_size, self.num_heads, self.num_kv_heads, self.head_dim,
+                tpc, self.torch_dtype, bias=self.attn_bias, fused=False,
+                group=ps.get_prefill_tp_group,
+            )
+            self.cte_o_proj = GroupQueryAttentionO(
+                self.hidden_size, self.num_heads, self.head_dim, tpc,
+                self.torch_dtype, bias=False, group=ps.get_prefill_tp_group,
+            )
         self.scale = 1.0 / math.sqrt(self.head_dim)
         self.rotary = rotary
         # optional per-head q/k norms (Qwen3 PRE_ROPE qk-norm); models set these
@@ -138,6 +157,13 @@ class AttentionBase(nn.Module):
             out = out.transpose(1, 2).reshape(B, S, -1)
             return self.o_proj(out)
 
+        if (
+            phase == PHASE_CTE and self.cp_degree > 1
+            and S % self.cp_degree == 0 and not sequence_parallel
+        ):
+            return self._cp_prefill(hidden, cos, sin, kv_mgr, position_ids,
+                                    attention_mask, seq_ids)
+
         q, k, v = self.prep_qkv(hidden, cos, sin, sequence_parallel=sequence_parallel)
 
         # write this step's K/V into the cache first (in-place scatter)
@@ -152,6 +178,50 @@ class AttentionBase(nn.Module):
         return self.o_proj(out, sequence_parallel=sequence_parallel)
 
     # ------------------------------------------------------------------
+    def _cp_prefill(self, hidden, cos, sin, kv_mgr, position_ids,
+                    attention_mask, seq_ids) -> torch.Tensor:
+        from neuronx_distributed_inference_amd.parallel import mappings
+
+        B, S, _ = hidden.shape
+        cp = self.cp_degree
+        cp_rank = ps.get_context_parallel_rank()
+
+        # decode-sharded K/V into the cache (standard projections; the q
+        # output of this pass is discarded — the ref instead reorders kv
+        # heads between meshes, attention/utils.py:455-623)
+        _, k_full, v_full = self.prep_qkv(hidden, cos, sin)
+        kv_mgr.update_cache(self.layer_idx, k_full, v_full, position_ids, seq_ids)
+
+        # CTE-sharded q/k/v over the prefill-TP subgroup
+        qc, kc, vc = self.cte_qkv_proj(hidden)
+        hc = self.cte_qkv_proj.num_heads_per_rank
+        hkvc = self.cte_qkv_proj.num_kv_heads_per_rank
+        qc = qc.view(B, S, hc, self.head_dim).transpose(1, 2)
+        kc = kc.view(B, S, hkvc, self.head_dim).transpose(1, 2)
+        vc = vc.view(B, S, hkvc, self.head_dim).transpose(1, 2)
+        if cos is not None:
+            from neuronx_distributed_inference_amd import ops as _ops
+
+            qc, kc = _ops.rope_apply(qc, kc, cos, sin)
+
+        # this rank's S/cp query slice vs the full K/V, offset-causal mask
+        ssl = S // cp
+        c0 = cp_rank * ssl
+        q_slice = qc[:, :, c0:c0 + ssl].contiguous()
+        ar = torch.arange(S, device=hidden.device).view(1, 1, 1, S)
+        qpos = (c0 + torch.arange(ssl, device=hidden.device)).view(1, 1, ssl, 1)
+        mask = ar <= qpos
+        if attention_mask is not None:
+            mask = mask & attention_mask.view(B, 1, 1, S).bool()
+        if self.sliding_window:
+            mask = mask & (ar > qpos - self.sliding_window)
+        out = ops.attention_prefill(q_slice, kc, vc, mask=mask, scale=self.scale,
+                                    sinks=self.sinks, causal=False)
+        out = out.transpose(1, 2).reshape(B, ssl, -1)
+        out = self.cte_o_proj(out)                   # reduce over prefill group
+        return mappings.gather_along_dim(out, dim=1,
+                                         group=ps.get_context_parallel_group())
+
     def _prefill(self, q, k, v, attention_mask) -> torch.Tensor:
         """Causal flash attention over the fresh K/V (the cache holds the
         same values; using the projections directly avoids a re-read).

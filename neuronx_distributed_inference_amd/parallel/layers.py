@@ -127,7 +127,7 @@ class ColumnParallelLinear(nn.Module):
         self.tp_degree = tp_degree or (
             ps.get_tensor_model_parallel_size() if group is None else 1
         )
-        if group is not None:
+        if group is not None and not callable(group):
             import torch.distributed as dist
 
             self.tp_degree = dist.get_world_size(group=group) if dist.is_initialized() else 1
@@ -144,6 +144,8 @@ class ColumnParallelLinear(nn.Module):
 
     @property
     def group(self):
+        if callable(self._group):
+            return self._group()
         return self._group if self._group is not None else ps.get_tensor_model_parallel_group()
 
     def quantize_fp8(self):
@@ -188,7 +190,7 @@ class RowParallelLinear(nn.Module):
         self.tp_degree = tp_degree or (
             ps.get_tensor_model_parallel_size() if group is None else 1
         )
-        if group is not None:
+        if group is not None and not callable(group):
             import torch.distributed as dist
 
             self.tp_degree = dist.get_world_size(group=group) if dist.is_initialized() else 1
@@ -205,6 +207,8 @@ class RowParallelLinear(nn.Module):
 
     @property
     def group(self):
+        if callable(self._group):
+            return self._group()
         return self._group if self._group is not None else ps.get_tensor_model_parallel_group()
 
     def quantize_fp8(self):

@@ -25,6 +25,10 @@ def _param_meta(model: nn.Module) -> Dict[str, dict]:
             meta[name] = {
                 "dim": p.partition_dim,
                 "preshard_fn": getattr(p, "preshard_fn", None),
+                # a param may shard over a SUB-group of the TP world (e.g.
+                # the CP prefill projections over tp/cp contiguous ranks):
+                # shard index = rank % num_partitions
+                "num_partitions": getattr(p, "num_partitions", None),
             }
     return meta
 
@@ -41,14 +45,15 @@ def shard_state_dict_for_rank(
             out[name] = tensor
             continue
         t = tensor
+        parts = m["num_partitions"] or tp_degree
         if m["preshard_fn"] is not None:
-            t = m["preshard_fn"](t, tp_degree)
+            t = m["preshard_fn"](t, parts)
         dim = m["dim"]
         size = t.shape[dim]
-        if size % tp_degree != 0:
-            raise ValueError(f"{name}: dim {dim} size {size} not divisible by tp {tp_degree}")
-        shard = size // tp_degree
-        out[name] = t.narrow(dim, rank * shard, shard).contiguous()
+        if size % parts != 0:
+            raise ValueError(f"{name}: dim {dim} size {size} not divisible by {parts}")
+        shard = size // parts
+        out[name] = t.narrow(dim, (rank % parts) * shard, shard).contiguous()
     return out
 
 
