@@ -28,7 +28,7 @@ def fuse_qkv_keys(sd: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
     out = dict(sd)
     for suffix in ("weight", "bias"):
         qs = [k for k in sd if k.endswith(f"qkv_proj.q_proj.{suffix}")
-              and "cte_qkv_proj" not in k]
+              and "cte_qkv_proj" not in k and "tkg_qkv_proj" not in k]
         for qk in qs:
             base = qk[: -len(f"q_proj.{suffix}")]
             kk, vk = f"{base}k_proj.{suffix}", f"{base}v_proj.{suffix}"
@@ -146,6 +146,12 @@ class LlamaForCausalLM(CausalLMBase):
                     out[k.replace(".qkv_proj.", ".cte_qkv_proj.")] = out[k]
                 elif ".self_attn.o_proj.o_proj." in k:
                     out[k.replace(".o_proj.o_proj.", ".cte_o_proj.o_proj.")] = out[k]
+        if ec.attention_dp_degree > 1:
+            for k in list(out):
+                if ".self_attn.qkv_proj." in k and "cte_qkv_proj" not in k:
+                    out[k.replace(".qkv_proj.", ".tkg_qkv_proj.")] = out[k]
+                elif ".self_attn.o_proj.o_proj." in k:
+                    out[k.replace(".o_proj.o_proj.", ".tkg_o_proj.o_proj.")] = out[k]
         if ec.fused_qkv:
             out = fuse_qkv_keys(out)
         if ec.fused_gate_up:

@@ -87,6 +87,10 @@ class DecoderModel(nn.Module):
         # stored OCP e4m3, quantized on write and dequantized in the decode
         # kernel — halves KV HBM traffic and doubles cache capacity
         cache_dtype = torch.float8_e4m3fn if ec.kv_cache_quant else ec.torch_dtype
+        # attention-DP decode runs at reduced TP (tp/dp), so each rank's
+        # cache holds correspondingly MORE kv heads (and B/dp rows)
+        cache_tp = ec.tp_degree // ec.attention_dp_degree \
+            if ec.attention_dp_degree > 1 else ec.tp_degree
         if ec.is_block_kv_layout:
             from neuronx_distributed_inference_amd.modules.kvcache.block_kv_cache_manager import (
                 BlockKVCacheManager,
@@ -100,7 +104,7 @@ class DecoderModel(nn.Module):
                 num_layers=self.num_layers, num_kv_heads=n_kv_heads,
                 head_dim=head_dim, num_blocks=num_blocks,
                 block_size=ec.pa_block_size, dtype=cache_dtype,
-                tp_degree=ec.tp_degree,
+                tp_degree=cache_tp,
             )
         else:
             self.kv_mgr = KVCacheManager(
@@ -111,7 +115,7 @@ class DecoderModel(nn.Module):
                 batch_size=ec.kv_cache_batch_size,
                 dtype=cache_dtype,
                 padding_rows=ec.kv_cache_padding_size,
-                tp_degree=ec.tp_degree,
+                tp_degree=cache_tp,
             )
         self.sampler = (
             Sampler(ec.on_device_sampling_config, config.vocab_size)

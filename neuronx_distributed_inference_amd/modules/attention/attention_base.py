@@ -68,6 +68,25 @@ class AttentionBase(nn.Module):
         # group; each CP rank computes its S/cp query slice against the full
         # K/V (activations are TP-replicated, so no KV all-gather is needed)
         # and the outputs are all-gathered along S over the CP group.
+        # Attention data parallelism for decode (ref §2.8 DP +
+        # DataParallelKVCacheManager): TKG attention + KV cache sharded by
+        # batch across DP subgroups, each running reduced-TP (tp/dp)
+        # projections; out-of-shard rows write to nowhere (seq_id -1) and the
+        # outputs are all-gathered along batch over the DP group.
+        self.dp_degree = ec.attention_dp_degree
+        if self.dp_degree > 1:
+            tpd = self.tp_degree // self.dp_degree
+            self.tkg_qkv_proj = GroupQueryAttentionQKV(
+                self.hidden_size, self.num_heads, self.num_kv_heads, self.head_dim,
+                tpd, self.torch_dtype, bias=self.attn_bias, fused=False,
+                group=ps.get_attention_dp_decode_tp_group,
+            )
+            self.tkg_o_proj = GroupQueryAttentionO(
+                self.hidden_size, self.num_heads, self.head_dim, tpd,
+                self.torch_dtype, bias=False,
+                group=ps.get_attention_dp_decode_tp_group,
+            )
+
         self.cp_degree = ec.cp_degree
         if self.cp_degree > 1:
             tpc = self.tp_degree // self.cp_degree
@@ -163,11 +182,18 @@ class AttentionBase(nn.Module):
         ):
             return self._cp_prefill(hidden, cos, sin, kv_mgr, position_ids,
                                     attention_mask, seq_ids)
+        if phase == PHASE_TKG and self.dp_degree > 1 and B % self.dp_degree == 0:
+            return self._dp_decode(hidden, cos, sin, kv_mgr, position_ids, seq_ids)
 
         q, k, v = self.prep_qkv(hidden, cos, sin, sequence_parallel=sequence_parallel)
 
         # write this step's K/V into the cache first (in-place scatter)
-        kv_mgr.update_cache(self.layer_idx, k, v, position_ids, seq_ids)
+        if self.dp_degree > 1 and phase == PHASE_CTE:
+            # attention-DP: the cache lives in DECODE layout (tp/dp sharding,
+            # B/dp local rows) — write it from the TKG projection set
+            self._dp_write_cte_cache(hidden, cos, sin, kv_mgr, position_ids, seq_ids)
+        else:
+            kv_mgr.update_cache(self.layer_idx, k, v, position_ids, seq_ids)
 
         if phase == PHASE_CTE:
             out = self._prefill(q, k, v, attention_mask)
@@ -221,6 +247,65 @@ class AttentionBase(nn.Module):
         out = self.cte_o_proj(out)                   # reduce over prefill group
         return mappings.gather_along_dim(out, dim=1,
                                          group=ps.get_context_parallel_group())
+
+    def _dp_local_rows(self, seq_ids, B, device):
+        dp_rank = ps.get_attention_dp_rank()
+        per = B // self.dp_degree
+        if seq_ids is None:
+            seq_ids = torch.arange(B, device=device)
+        local = seq_ids - dp_rank * per
+        return torch.where((local >= 0) & (local < per), local,
+                           torch.full_like(local, -1))
+
+    def _dp_write_cte_cache(self, hidden, cos, sin, kv_mgr, position_ids, seq_ids):
+        B, S, _ = hidden.shape
+        _, k, v = self.tkg_qkv_proj(hidden)
+        hkvc = self.tkg_qkv_proj.num_kv_heads_per_rank
+        k = k.view(B, S, hkvc, self.head_dim).transpose(1, 2)
+        v = v.view(B, S, hkvc, self.head_dim).transpose(1, 2)
+        if cos is not None:
+            # rope applies to k only; reuse the paired op with a dummy q
+            dummy_q = torch.empty_like(k)
+            _, k = ops.rope_apply(dummy_q, k, cos, sin)
+        local_rows = self._dp_local_rows(seq_ids, B, hidden.device)
+        kv_mgr.update_cache(self.layer_idx, k, v, position_ids, local_rows)
+
+    def _dp_decode(self, hidden, cos, sin, kv_mgr, position_ids, seq_ids):
+        from neuronx_distributed_inference_amd.parallel import mappings
+
+        B, S, _ = hidden.shape
+        dp = self.dp_degree
+        dp_rank = ps.get_attention_dp_rank()
+        per = B // dp
+        if seq_ids is None:
+            seq_ids = torch.arange(B, device=hidden.device)
+
+        q, k, v = self.tkg_qkv_proj(hidden)
+        hc = self.tkg_qkv_proj.num_heads_per_rank
+        hkvc = self.tkg_qkv_proj.num_kv_heads_per_rank
+        q = q.view(B, S, hc, self.head_dim).transpose(1, 2)
+        k = k.view(B, S, hkvc, self.head_dim).transpose(1, 2)
+        v = v.view(B, S, hkvc, self.head_dim).transpose(1, 2)
+        if cos is not None:
+            q, k = ops.rope_apply(q, k, cos, sin)
+
+        # cache rows are LOCAL to this DP shard (kv_cache_batch_size = B/dp);
+        # out-of-shard rows map to -1 and the scatter kernel skips them
+        local_rows = self._dp_local_rows(seq_ids, B, hidden.device)
+        kv_mgr.update_cache(self.layer_idx, k, v, position_ids, local_rows)
+
+        lo = dp_rank * per
+        q_shard = q[lo:lo + per].contiguous()
+        k_cache, v_cache = kv_mgr.get_cache(self.layer_idx)
+        out = ops.attention_decode(
+            q_shard, k_cache, v_cache, position_ids[lo:lo + per],
+            seq_ids=local_rows[lo:lo + per], scale=self.scale, sinks=self.sinks,
+            sliding_window=self.sliding_window,
+        )
+        out = out.transpose(1, 2).reshape(per, S, -1)
+        out = self.tkg_o_proj(out)                  # reduce over decode-TP
+        # all-gather the batch shards over the DP group (ref AG_AR option)
+        return mappings.gather_along_dim(out, dim=0, group=ps.get_attention_dp_group())
 
     def _prefill(self, q, k, v, attention_mask) -> torch.Tensor:
         """Causal flash attention over the fresh K/V (the cache holds the

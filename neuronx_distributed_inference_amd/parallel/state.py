@@ -219,6 +219,10 @@ def get_attention_dp_group():
     return _STATE.dp_group
 
 
+def get_attention_dp_decode_tp_group():
+    return _STATE.decode_tp_group
+
+
 def get_attention_dp_size() -> int:
     return _group_size(_STATE.dp_group) if _STATE.dp_group is not None else (
         _STATE.attention_dp_degree if dist.is_initialized() and _STATE.initialized else 1

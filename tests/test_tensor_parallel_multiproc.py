@@ -397,3 +397,68 @@ def _worker_cp_generation(rank):
 
 def test_context_parallel_generation():
     spawn("_worker_cp_generation")
+
+
+# ---------------------------------------------------------------------------
+def _worker_attention_dp(rank):
+    """Attention-DP decode (tp=2, dp=2 -> decode_tp=1; batch sharded, KV
+    cache B/dp per rank) matches the single-rank result."""
+    from neuronx_distributed_inference_amd.config import EngineConfig
+    from neuronx_distributed_inference_amd.models.llama.modeling_llama import (
+        LlamaForCausalLM,
+        LlamaInferenceConfig,
+    )
+    from neuronx_distributed_inference_amd.parallel import state as ps
+    from neuronx_distributed_inference_amd.utils.checkpoint_sharding import (
+        shard_state_dict_for_rank,
+    )
+    from neuronx_distributed_inference_amd.utils.hf_adapter import (
+        GenerationConfig,
+        HuggingFaceGenerationAdapter,
+    )
+
+    def make_cfg(tp, dp):
+        ec = EngineConfig(batch_size=2, max_batch_size=2, seq_len=64, max_length=64,
+                          torch_dtype="float32", tp_degree=tp,
+                          attention_dp_degree=dp)
+        return LlamaInferenceConfig(
+            neuron_config=ec, vocab_size=128, hidden_size=64, intermediate_size=128,
+            num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+            max_position_embeddings=128, rms_norm_eps=1e-5, rope_theta=10000.0,
+            hidden_act="silu", tie_word_embeddings=False)
+
+    ps.destroy_model_parallel()
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(51)
+    full = LlamaForCausalLM(make_cfg(1, 1)).eval()
+    with torch.no_grad():
+        for p in full.parameters():
+            if p.dim() >= 2:
+                p.normal_(0, 0.1)
+    ids = torch.randint(0, 128, (2, 8))
+    want = HuggingFaceGenerationAdapter(full).generate(
+        ids, generation_config=GenerationConfig(max_new_tokens=6))
+
+    ps.destroy_model_parallel()
+    ps.initialize_model_parallel(tensor_model_parallel_size=WORLD,
+                                 attention_dp_size=WORLD)
+    dp_model = LlamaForCausalLM(make_cfg(WORLD, WORLD)).eval()
+    assert dp_model.model.kv_mgr.k_caches[0].shape[0] == 1  # B/dp rows
+    full_sd = {k: v for k, v in full.model.state_dict().items()
+               if not k.startswith("kv_mgr.")}
+    for k in list(full_sd):
+        if ".self_attn.qkv_proj." in k:
+            full_sd[k.replace(".qkv_proj.", ".tkg_qkv_proj.")] = full_sd[k]
+        elif ".self_attn.o_proj.o_proj." in k:
+            full_sd[k.replace(".o_proj.o_proj.", ".tkg_o_proj.o_proj.")] = full_sd[k]
+    shard = shard_state_dict_for_rank(full_sd, dp_model.model, rank, WORLD)
+    missing, unexpected = dp_model.model.load_state_dict(shard, strict=False)
+    assert not [m for m in missing if not m.startswith("kv_mgr")], missing
+    got = HuggingFaceGenerationAdapter(dp_model).generate(
+        ids, generation_config=GenerationConfig(max_new_tokens=6))
+    assert torch.equal(got, want), (got.tolist(), want.tolist())
+    return True
+
+
+def test_attention_dp_generation():
+    spawn("_worker_attention_dp")
