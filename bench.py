@@ -248,21 +248,22 @@ def main():
         next_tok = out.tokens.view(-1, 1)
         cur_len = args.ctx
 
-        def decode_step(tok, cur_len):
-            p = torch.full((global_batch, 1), cur_len, device=device, dtype=torch.long)
-            o = model(input_ids=tok, attention_mask=None, position_ids=p, seq_ids=seq_ids)
+        # device-resident position counter: no per-step host->device copy
+        pos_t = torch.full((global_batch, 1), cur_len, device=device, dtype=torch.long)
+
+        def decode_step(tok):
+            o = model(input_ids=tok, attention_mask=None, position_ids=pos_t, seq_ids=seq_ids)
+            pos_t.add_(1)
             return o.tokens.view(-1, 1)
 
         # ---- warmup ----
         for _ in range(warmup):
-            next_tok = decode_step(next_tok, cur_len)
-            cur_len += 1
+            next_tok = decode_step(next_tok)
 
         barrier_sync()
         t0 = time.perf_counter()
         for _ in range(steps):
-            next_tok = decode_step(next_tok, cur_len)
-            cur_len += 1
+            next_tok = decode_step(next_tok)
         barrier_sync()
         elapsed = time.perf_counter() - t0
 

@@ -361,11 +361,17 @@ class CausalLMBase(nn.Module):
             else:
                 position_ids = torch.arange(S, device=input_ids.device).view(1, S).expand(B, S)
         if sampling_params is None and self.ec.on_device_sampling_config:
-            c = self.ec.on_device_sampling_config
-            sampling_params = prepare_sampling_params(
-                B, c.top_k if c.do_sample else 1, c.top_p, c.temperature,
-                device=input_ids.device,
-            )
+            # memoized: building this tensor does host->device copies, which
+            # would otherwise run every decode step
+            cached = getattr(self, "_default_sampling_params", None)
+            if cached is None or cached.shape[0] != B or cached.device != input_ids.device:
+                c = self.ec.on_device_sampling_config
+                cached = prepare_sampling_params(
+                    B, c.top_k if c.do_sample else 1, c.top_p, c.temperature,
+                    device=input_ids.device,
+                )
+                self._default_sampling_params = cached
+            sampling_params = cached
         # drop an all-ones padding mask on the host so the device path never
         # syncs per layer
         if attention_mask is not None and S > 1 and bool(attention_mask.all()):
