@@ -181,6 +181,11 @@ class DecoderModel(nn.Module):
         h = self.embed_tokens(input_ids, sequence_parallel=sp_active) \
             if sp_active else self.embed_tokens(input_ids)
         cos, sin = self.rotary.get_cos_sin(position_ids, device=h.device)
+        # int32 once here: the KV/attention kernels take int32 and a per-layer
+        # .to() would launch 4 conversion kernels per layer per step
+        position_ids = position_ids.to(torch.int32)
+        if seq_ids is not None:
+            seq_ids = seq_ids.to(torch.int32)
         residual = None
         for layer in self.layers:
             if sp_active:
