@@ -92,6 +92,17 @@ def setup_run_parser(run: argparse.ArgumentParser):
     # engine knobs
     run.add_argument("--no-hip-graphs", dest="enable_hip_graphs", action="store_false",
                      default=True)
+    run.add_argument("--kv-cache-quant", action="store_true",
+                     help="store the KV cache in fp8 (OCP e4m3)")
+    run.add_argument("--quantized", action="store_true",
+                     help="fp8 weight quantization (per-row scales)")
+    run.add_argument("--fused-qkv", action="store_true")
+    run.add_argument("--fused-gate-up", action="store_true")
+    run.add_argument("--sequence-parallel-enabled", action="store_true")
+    run.add_argument("--windowed-context-encoding-size", type=int, default=0)
+    run.add_argument("--is-block-kv-layout", action="store_true")
+    run.add_argument("--pa-num-blocks", type=int, default=None)
+    run.add_argument("--pa-block-size", type=int, default=128)
     run.add_argument("--skip-warmup", action="store_true")
     run.add_argument("--on-cpu", action="store_true")
     # pipeline control
@@ -141,6 +152,15 @@ def create_engine_config(args, model_type: str) -> EngineConfig:
         enable_hip_graphs=args.enable_hip_graphs,
         skip_warmup=args.skip_warmup,
         on_cpu=args.on_cpu,
+        kv_cache_quant=args.kv_cache_quant,
+        quantized=args.quantized,
+        fused_qkv=args.fused_qkv,
+        fused_gate_up=args.fused_gate_up,
+        sequence_parallel_enabled=args.sequence_parallel_enabled,
+        windowed_context_encoding_size=args.windowed_context_encoding_size,
+        is_block_kv_layout=args.is_block_kv_layout,
+        pa_num_blocks=args.pa_num_blocks,
+        pa_block_size=args.pa_block_size,
     )
 
 

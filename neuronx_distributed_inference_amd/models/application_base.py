@@ -94,6 +94,13 @@ class InferenceApplication:
         if real_missing:
             raise RuntimeError(f"missing weights: {real_missing[:8]}")
         model = model.to(device).eval()
+        if ec.quantized:
+            from neuronx_distributed_inference_amd.parallel.layers import (
+                quantize_model_fp8,
+            )
+
+            n = quantize_model_fp8(model.model)
+            logger.info("quantized %d linears to fp8", n)
         model.enable_submodels()
         if not skip_warmup and not ec.skip_warmup:
             model.warmup()
