@@ -44,6 +44,8 @@ def parse_args():
     p.add_argument("--batch-per-gpu", type=int, default=64)
     p.add_argument("--ctx", type=int, default=128)
     p.add_argument("--no-graphs", action="store_true")
+    p.add_argument("--e2e", action="store_true",
+                   help="also run the generate()-level benchmark (p50/p99 + throughput)")
     p.add_argument("--fp8", action="store_true",
                    help="fp8 (e4m3) weight quantization via hipBLASLt scaled-mm")
     p.add_argument("--spec", action="store_true",
@@ -299,6 +301,17 @@ def main():
                 "weights": "random-init",
             },
         }))
+
+    if args.e2e and rank == 0 and world == 1:
+        from neuronx_distributed_inference_amd.utils.benchmark import benchmark_sampling
+        from neuronx_distributed_inference_amd.utils.hf_adapter import GenerationConfig
+
+        report = benchmark_sampling(
+            model, ids, attention_mask=mask, n_runs=3,
+            generation_config=GenerationConfig(max_new_tokens=min(steps, 32)),
+            report_path=None,
+        )
+        print(json.dumps({"e2e_report": report}))
 
     if world > 1:
         dist.destroy_process_group()
