@@ -213,9 +213,13 @@ def main():
     )
 
     torch.manual_seed(1234)
-    model = LlamaForCausalLM(cfg).eval()
     if on_gpu:
-        model = model.to(device)
+        # construct directly on the GPU: a 70B bf16 model is 140 GB and must
+        # never round-trip host RAM / PCIe
+        with torch.device(device):
+            model = LlamaForCausalLM(cfg).eval()
+    else:
+        model = LlamaForCausalLM(cfg).eval()
     with torch.no_grad():
         for p in model.parameters():
             if p.dim() >= 2:
