@@ -176,6 +176,43 @@ class AttentionBase(nn.Module):
             out = out.transpose(1, 2).reshape(B, S, -1)
             return self.o_proj(out)
 
+        # Fused CTE front end: one GEMM -> qkv_rope_update (RoPE + cache
+        # scatter + q relayout) -> flash attention reading K/V straight from
+        # the cache rows just written. Saves 3 transpose-copies + rope + kv
+        # kernels per layer vs the unfused path. Right-padded batches are
+        # handled without a mask: pad positions (-1) were skipped by the
+        # scatter, so causal-by-row-index only ever sees the real prefix.
+        if (
+            phase == PHASE_CTE
+            and self.qkv_proj.fused
+            and self.q_norm is None
+            and self.clip_qkv is None
+            and cos is not None
+            and seq_ids is not None
+            and not sequence_parallel
+            and self.cp_degree == 1
+            and self.dp_degree == 1
+            and not getattr(kv_mgr, "is_paged", False)
+            and (not hidden.is_cuda or self.head_dim in (64, 128))
+        ):
+            kc, vc = kv_mgr.get_cache(self.layer_idx)
+            kc = kc.data if hasattr(kc, "data") else kc
+            vc = vc.data if hasattr(vc, "data") else vc
+            if kc.dtype == hidden.dtype:  # fp8 cache: unfused CTE path below
+                import neuronx_distributed_inference_amd.ops as _ops
+
+                y = self.qkv_proj.qkv(hidden)
+                q = _ops.qkv_rope_update(
+                    y, cos, sin, kc, vc, position_ids, seq_ids,
+                    self.num_heads_per_rank, self.num_kv_heads_per_rank,
+                )
+                out = _ops.attention_prefill_cached(
+                    q, kc, vc, seq_ids, scale=self.scale, sinks=self.sinks,
+                    sliding_window=self.sliding_window,
+                )
+                out = out.transpose(1, 2).reshape(B, S, -1)
+                return self.o_proj(out)
+
         if (
             phase == PHASE_CTE and self.cp_degree > 1
             and S % self.cp_degree == 0 and not sequence_parallel

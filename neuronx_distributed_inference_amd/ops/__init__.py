@@ -153,6 +153,33 @@ def attention_prefill(q, k, v, mask=None, scale=None, sinks=None, causal=True,
     return ref.sdpa(q, k, v, mask=mask, scale=scale, sinks=sinks)
 
 
+def attention_prefill_cached(q, k_cache, v_cache, seq_ids, scale=None,
+                             sinks=None, sliding_window: int = 0):
+    """Context-encoding attention reading K/V straight from the cache rows
+    that `qkv_rope_update` just wrote (no fresh K/V tensors, no copies).
+
+    q: (B,Hq,S,D); k_cache/v_cache: (Bc,Hkv,Smax,D); seq_ids: (B,) cache
+    rows. Causal by *row index* — valid for right-padded batches because
+    pad positions (-1) were skipped by the cache scatter, so cache rows
+    0..i are always the real prefix for any real query at row i. Fuses the
+    reference's CTE attention + cache update (SURVEY §2 K1+K5)."""
+    import math
+    scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+    if q.is_cuda:
+        c = _require_ext("attention_prefill_cached")
+        out = torch.empty_like(q)
+        c.flash_prefill_cached(out, q.contiguous(), k_cache, v_cache,
+                               seq_ids, float(scale), int(sliding_window),
+                               sinks=None if sinks is None else sinks.detach())
+        return out
+    B, Hq, S, D = q.shape
+    rows = seq_ids.long().clamp_min(0)
+    k = k_cache[rows, :, :S].to(q.dtype)
+    v = v_cache[rows, :, :S].to(q.dtype)
+    return attention_prefill(q, k, v, scale=scale, sinks=sinks,
+                             causal=True, sliding_window=sliding_window)
+
+
 def attention_decode(q, k_cache, v_cache, seq_lens, seq_ids=None, scale=None,
                      sinks=None, sliding_window: int = 0):
     """Token-generation attention against the linear KV cache.

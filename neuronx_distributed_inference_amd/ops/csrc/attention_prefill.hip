@@ -54,7 +54,9 @@ __global__ __launch_bounds__(256, 2) void flash_prefill_kernel(
     const T* __restrict__ v,   // (B, Hkv, S, D)
     const float scale, const int window,
     const int Hq, const int Hkv, const int S,
-    const float* __restrict__ sinks) {  // (Hq,) learned sink logits or null
+    const float* __restrict__ sinks,   // (Hq,) learned sink logits or null
+    const int* __restrict__ seq_ids,   // non-null: K/V live in the KV cache
+    const int Smax) {                  // cache row capacity (= S when direct)
     using MT = MfmaTraits<T>;
     using vec8 = typename MT::vec8;
     constexpr int DI = D / 16;       // O col fragments per wave
@@ -90,7 +92,8 @@ __global__ __launch_bounds__(256, 2) void flash_prefill_kernel(
     const int wq_hi = min(wq_lo + WQ, S) - 1;   // last valid q row (abs)
 
     const long q_base = (((long)b * Hq + hq) * S) * D;
-    const long kv_base = (((long)b * Hkv + hkv) * S) * D;
+    const long kv_row = (seq_ids != nullptr) ? seq_ids[b] : b;
+    const long kv_base = ((kv_row * Hkv + hkv) * (long)Smax) * D;
 
     // ---- Q fragments into registers (A layout: row = lane&15, k-elems
     // (lane>>4)*8 .. +8, per (qi, ks)) ----
@@ -312,11 +315,12 @@ __global__ __launch_bounds__(256, 2) void flash_prefill_kernel(
 // loads V^T[n_dim][k] rows, giving lane {d = di*16 + lane&15, k = (lane>>4)*8+j}
 // which matches B[k][n] with n=d. Same shape logic for the QK^T B (K tile).
 
-void flash_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
-                   torch::Tensor v, double scale, long window,
-                   c10::optional<torch::Tensor> sinks_opt) {
+static void flash_prefill_impl(torch::Tensor& out, torch::Tensor& q,
+                               const torch::Tensor& k, const torch::Tensor& v,
+                               double scale, long window,
+                               c10::optional<torch::Tensor>& sinks_opt,
+                               const int* seq_ids_ptr, int Smax_arg, int Hkv) {
     const int B = q.size(0), Hq = q.size(1), S = q.size(2), D = q.size(3);
-    const int Hkv = k.size(1);
     TORCH_CHECK(Hq % Hkv == 0, "Hq must divide by Hkv");
     TORCH_CHECK(D == 64 || D == 128, "flash_prefill: head_dim 64/128 only");
     dim3 grid((S + QBLK - 1) / QBLK, Hq, B);
@@ -334,27 +338,50 @@ void flash_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                                (const __hip_bfloat16*)q.data_ptr(),
                                (const __hip_bfloat16*)k.data_ptr(),
                                (const __hip_bfloat16*)v.data_ptr(),
-                               (float)scale, (int)window, Hq, Hkv, S, sinks);
+                               (float)scale, (int)window, Hq, Hkv, S, sinks,
+                               seq_ids_ptr, Smax_arg);
         else
             hipLaunchKernelGGL((flash_prefill_kernel<__hip_bfloat16, 64>), grid, dim3(256), 0,
                                stream, (__hip_bfloat16*)out.data_ptr(),
                                (const __hip_bfloat16*)q.data_ptr(),
                                (const __hip_bfloat16*)k.data_ptr(),
                                (const __hip_bfloat16*)v.data_ptr(),
-                               (float)scale, (int)window, Hq, Hkv, S, sinks);
+                               (float)scale, (int)window, Hq, Hkv, S, sinks,
+                               seq_ids_ptr, Smax_arg);
     } else if (q.scalar_type() == at::ScalarType::Half) {
         if (D == 128)
             hipLaunchKernelGGL((flash_prefill_kernel<__half, 128>), grid, dim3(256), 0,
                                stream, (__half*)out.data_ptr(), (const __half*)q.data_ptr(),
                                (const __half*)k.data_ptr(), (const __half*)v.data_ptr(),
-                               (float)scale, (int)window, Hq, Hkv, S, sinks);
+                               (float)scale, (int)window, Hq, Hkv, S, sinks,
+                               seq_ids_ptr, Smax_arg);
         else
             hipLaunchKernelGGL((flash_prefill_kernel<__half, 64>), grid, dim3(256), 0,
                                stream, (__half*)out.data_ptr(), (const __half*)q.data_ptr(),
                                (const __half*)k.data_ptr(), (const __half*)v.data_ptr(),
-                               (float)scale, (int)window, Hq, Hkv, S, sinks);
+                               (float)scale, (int)window, Hq, Hkv, S, sinks,
+                               seq_ids_ptr, Smax_arg);
     } else {
         TORCH_CHECK(false, "flash_prefill: bf16/f16 only");
     }
     HIP_CHECK_KERNEL();
+}
+
+void flash_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
+                   torch::Tensor v, double scale, long window,
+                   c10::optional<torch::Tensor> sinks_opt) {
+    flash_prefill_impl(out, q, k, v, scale, window, sinks_opt, nullptr,
+                       (int)q.size(2), (int)k.size(1));
+}
+
+void flash_prefill_cached(torch::Tensor out, torch::Tensor q,
+                          torch::Tensor k_cache, torch::Tensor v_cache,
+                          torch::Tensor seq_ids, double scale, long window,
+                          c10::optional<torch::Tensor> sinks_opt) {
+    // K/V read from the (Bc, Hkv, Smax, D) cache rows just written —
+    // prefill straight off the cache, no fresh K/V tensors
+    auto sid = seq_ids.to(torch::kInt32).contiguous();
+    flash_prefill_impl(out, q, k_cache, v_cache, scale, window, sinks_opt,
+                       sid.data_ptr<int>(), (int)k_cache.size(2),
+                       (int)k_cache.size(1));
 }

@@ -23,6 +23,9 @@ void topk_topp_sample(torch::Tensor out, torch::Tensor vals, torch::Tensor top_k
 void flash_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                    torch::Tensor v, double scale, long window,
                    c10::optional<torch::Tensor> sinks);
+void flash_prefill_cached(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
+                          torch::Tensor v_cache, torch::Tensor seq_ids, double scale,
+                          long window, c10::optional<torch::Tensor> sinks);
 void qkv_rope_update(torch::Tensor q_out, torch::Tensor qkv, torch::Tensor cosb,
                      torch::Tensor sinb, torch::Tensor k_cache, torch::Tensor v_cache,
                      torch::Tensor positions, torch::Tensor seq_ids, long Hq, long Hkv);
@@ -46,6 +49,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("flash_prefill", &flash_prefill, "MFMA causal flash prefill (CDNA4)",
           py::arg("out"), py::arg("q"), py::arg("k"), py::arg("v"), py::arg("scale"),
           py::arg("window"), py::arg("sinks") = py::none());
+    m.def("flash_prefill_cached", &flash_prefill_cached,
+          "MFMA causal flash prefill reading K/V from the cache (CDNA4)",
+          py::arg("out"), py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
+          py::arg("seq_ids"), py::arg("scale"), py::arg("window"),
+          py::arg("sinks") = py::none());
     m.def("qkv_rope_update", &qkv_rope_update, "fused qkv->rope->cache-write (CDNA4)");
     m.def("silu_mul_fused", &silu_mul_fused, "SiLU on fused gate|up (CDNA4)");
     m.def("gelu_tanh_mul_fused", &gelu_tanh_mul_fused, "GELU on fused gate|up (CDNA4)");

@@ -244,6 +244,55 @@ def test_flash_prefill_shapes(S, D):
     torch.testing.assert_close(out.float().cpu(), want, atol=3e-2, rtol=3e-2)
 
 
+@pytest.mark.parametrize("S", [96, 512])
+@pytest.mark.parametrize("D", [64, 128])
+def test_flash_prefill_cached_matches_direct(S, D):
+    """Cache-resident flash (CTE fused path) vs the plain flash kernel:
+    K/V scattered into a larger cache with shuffled seq_ids, Smax > S."""
+    torch.manual_seed(0)
+    B, Hq, Hkv, Bc, Smax = 2, 4, 2, 4, S + 192
+    dev = _dev()
+    q = torch.randn(B, Hq, S, D, dtype=torch.bfloat16, device=dev)
+    k = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16, device=dev)
+    v = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16, device=dev)
+    kc = torch.randn(Bc, Hkv, Smax, D, dtype=torch.bfloat16, device=dev)
+    vc = torch.randn(Bc, Hkv, Smax, D, dtype=torch.bfloat16, device=dev)
+    seq_ids = torch.tensor([3, 1], dtype=torch.int32, device=dev)
+    pos = torch.arange(S, dtype=torch.int32, device=dev).expand(B, S).contiguous()
+    ops.kv_cache_update(kc, vc, k, v, pos, seq_ids)
+    out = ops.attention_prefill_cached(q, kc, vc, seq_ids)
+    want = ops.attention_prefill(q, k, v, causal=True)
+    torch.testing.assert_close(out.float().cpu(), want.float().cpu(),
+                               atol=3e-2, rtol=3e-2)
+
+
+def test_flash_prefill_cached_ragged_pad_skip():
+    """Ragged right-padded batch: pad positions -1 are skipped by the
+    scatter; cached-flash output at real rows must match a per-row dense
+    reference; stale cache rows beyond each length must not leak in."""
+    torch.manual_seed(1)
+    B, Hq, Hkv, S, D, Bc, Smax = 2, 4, 2, 160, 128, 3, 256
+    dev = _dev()
+    lens = [97, 160]
+    q = torch.randn(B, Hq, S, D, dtype=torch.bfloat16, device=dev)
+    k = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16, device=dev)
+    v = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16, device=dev)
+    kc = torch.full((Bc, Hkv, Smax, D), 7.0, dtype=torch.bfloat16, device=dev)
+    vc = torch.full((Bc, Hkv, Smax, D), 7.0, dtype=torch.bfloat16, device=dev)
+    pos = torch.stack([
+        torch.where(torch.arange(S) < L, torch.arange(S), torch.tensor(-1))
+        for L in lens
+    ]).to(dev, torch.int32)
+    seq_ids = torch.tensor([1, 0], dtype=torch.int32, device=dev)
+    ops.kv_cache_update(kc, vc, k, v, pos, seq_ids)
+    out = ops.attention_prefill_cached(q, kc, vc, seq_ids)
+    for b, L in enumerate(lens):
+        want = ops.attention_prefill(q[b:b + 1, :, :L], k[b:b + 1, :, :L],
+                                     v[b:b + 1, :, :L], causal=True)
+        torch.testing.assert_close(out[b:b + 1, :, :L].float().cpu(),
+                                   want.float().cpu(), atol=3e-2, rtol=3e-2)
+
+
 def test_flash_prefill_sliding_window():
     torch.manual_seed(0)
     B, Hq, Hkv, S, D = 1, 2, 1, 512, 128

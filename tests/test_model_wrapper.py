@@ -144,6 +144,44 @@ def test_fused_projections_match_unfused(tiny_llama_config):
 
 
 @torch.inference_mode()
+def test_fused_cte_ragged_batch_matches_unfused(tiny_llama_config):
+    """The fused CTE path (qkv_rope_update + cache-resident flash) must
+    match the unfused path on a RAGGED right-padded batch — exercises the
+    pad-position=-1 skip invariant end to end."""
+    import copy
+
+    from neuronx_distributed_inference_amd.models.llama.modeling_llama import (
+        fuse_gate_up_keys,
+        fuse_qkv_keys,
+    )
+
+    torch.manual_seed(1)
+    base = LlamaForCausalLM(tiny_llama_config).eval()
+    with torch.no_grad():
+        for p in base.parameters():
+            if p.dim() >= 2:
+                p.normal_(0, 0.1)
+
+    cfg2 = copy.deepcopy(tiny_llama_config)
+    cfg2.neuron_config.fused_qkv = True
+    fused = LlamaForCausalLM(cfg2).eval()
+    sd = {k: v for k, v in base.model.state_dict().items() if not k.startswith("kv_mgr.")}
+    sd = fuse_qkv_keys(sd)
+    fused.model.load_state_dict(sd, strict=False)
+
+    ids = torch.randint(1, 128, (2, 12))
+    mask = torch.ones(2, 12, dtype=torch.long)
+    ids[0, 7:] = 0
+    mask[0, 7:] = 0   # row 0 has length 7, row 1 full: ragged batch
+    gc = GenerationConfig(max_new_tokens=6, pad_token_id=0)
+    a = HuggingFaceGenerationAdapter(base).generate(
+        ids, attention_mask=mask, generation_config=gc)
+    b = HuggingFaceGenerationAdapter(fused).generate(
+        ids, attention_mask=mask, generation_config=gc)
+    assert torch.equal(a, b)
+
+
+@torch.inference_mode()
 def test_windowed_context_encoding_matches_direct(tiny_llama_config):
     """Prefill chunked through the windowed CTE path must generate the same
     tokens as one-shot prefill (incl. a ragged row)."""
