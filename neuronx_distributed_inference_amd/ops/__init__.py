@@ -153,6 +153,48 @@ def attention_prefill(q, k, v, mask=None, scale=None, sinks=None, causal=True,
     return ref.sdpa(q, k, v, mask=mask, scale=scale, sinks=sinks)
 
 
+_lt_state = {"enabled": None, "bad_shapes": set()}
+
+
+def lt_linear_enabled() -> bool:
+    """True when the autotuned hipBLASLt GEMM path is usable here.
+
+    Disable with NXDI_DISABLE_HIPBLASLT=1 (falls back to torch F.linear's
+    rocBLAS dispatch)."""
+    if _lt_state["enabled"] is None:
+        if os.environ.get("NXDI_DISABLE_HIPBLASLT") == "1" or not torch.cuda.is_available():
+            _lt_state["enabled"] = False
+        else:
+            try:
+                c = _require_ext("lt_linear")
+                _lt_state["enabled"] = bool(c.lt_available())
+            except Exception:
+                _lt_state["enabled"] = False
+    return _lt_state["enabled"]
+
+
+def lt_linear(x: torch.Tensor, weight: torch.Tensor,
+              bias: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """y = x @ weight^T + bias via hipBLASLt with a per-shape TIMED
+    autotune on first sight (the analog of the reference compiler's AOT
+    GEMM tile selection). Replay is a plain stream enqueue — hipGraph-safe.
+    Shapes hipblaslt can't serve fall back to F.linear permanently."""
+    shape = x.shape
+    x2 = x.reshape(-1, shape[-1])
+    key = (x2.shape[0], weight.shape[0], weight.shape[1], bias is not None)
+    if key in _lt_state["bad_shapes"]:
+        return torch.nn.functional.linear(x, weight, bias)
+    c = _require_ext("lt_linear")
+    y = torch.empty(x2.shape[0], weight.shape[0], dtype=x.dtype, device=x.device)
+    try:
+        c.lt_linear(y, x2.contiguous(), weight,
+                    None if bias is None else bias.contiguous())
+    except RuntimeError:
+        _lt_state["bad_shapes"].add(key)
+        return torch.nn.functional.linear(x, weight, bias)
+    return y.view(*shape[:-1], weight.shape[0])
+
+
 def attention_prefill_cached(q, k_cache, v_cache, seq_ids, scale=None,
                              sinks=None, sliding_window: int = 0):
     """Context-encoding attention reading K/V straight from the cache rows

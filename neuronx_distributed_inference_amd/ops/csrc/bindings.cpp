@@ -30,6 +30,12 @@ void qkv_rope_update(torch::Tensor q_out, torch::Tensor qkv, torch::Tensor cosb,
                      torch::Tensor sinb, torch::Tensor k_cache, torch::Tensor v_cache,
                      torch::Tensor positions, torch::Tensor seq_ids, long Hq, long Hkv);
 void silu_mul_fused(torch::Tensor out, torch::Tensor gate_up);
+bool lt_available();
+void lt_linear(torch::Tensor y, torch::Tensor x, torch::Tensor w,
+               c10::optional<torch::Tensor> bias);
+int lt_tune(int64_t M, int64_t N, int64_t K, bool has_bias,
+            torch::Tensor sx, torch::Tensor sw, torch::Tensor sy,
+            c10::optional<torch::Tensor> bias);
 void gelu_tanh_mul_fused(torch::Tensor out, torch::Tensor gate_up);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -55,6 +61,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("seq_ids"), py::arg("scale"), py::arg("window"),
           py::arg("sinks") = py::none());
     m.def("qkv_rope_update", &qkv_rope_update, "fused qkv->rope->cache-write (CDNA4)");
+    m.def("lt_available", &lt_available, "hipblaslt usable on this device");
+    m.def("lt_linear", &lt_linear, "autotuned hipBLASLt linear (graph-safe)",
+          py::arg("y"), py::arg("x"), py::arg("w"), py::arg("bias") = py::none());
+    m.def("lt_tune", &lt_tune, "pre-tune one GEMM shape",
+          py::arg("M"), py::arg("N"), py::arg("K"), py::arg("has_bias"),
+          py::arg("sx"), py::arg("sw"), py::arg("sy"), py::arg("bias") = py::none());
     m.def("silu_mul_fused", &silu_mul_fused, "SiLU on fused gate|up (CDNA4)");
     m.def("gelu_tanh_mul_fused", &gelu_tanh_mul_fused, "GELU on fused gate|up (CDNA4)");
 }

@@ -100,6 +100,18 @@ def quantize_model_fp8(model: nn.Module, skip_names=("lm_head",)):
     return n
 
 
+def _linear(x: torch.Tensor, weight: torch.Tensor,
+            bias: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Library GEMM dispatch: autotuned hipBLASLt on GPU (bf16/f16),
+    torch F.linear (rocBLAS) otherwise. See ops.lt_linear."""
+    if x.is_cuda and x.dtype in (torch.bfloat16, torch.float16):
+        import neuronx_distributed_inference_amd.ops as _ops
+
+        if _ops.lt_linear_enabled():
+            return _ops.lt_linear(x, weight, bias)
+    return F.linear(x, weight, bias)
+
+
 class ColumnParallelLinear(nn.Module):
     """Y = X W^T with W sharded along the output dim across the TP group."""
 
@@ -158,7 +170,7 @@ class ColumnParallelLinear(nn.Module):
         if getattr(self, "fp8", False):
             y = fp8_linear(x, self.weight, self.weight_scale, self.bias)
         else:
-            y = F.linear(x, self.weight, self.bias)
+            y = _linear(x, self.weight, self.bias)
         if self.gather_output:
             y = mappings.gather_from_tensor_parallel_region(y, dim=-1, group=self.group)
         return y
@@ -220,7 +232,7 @@ class RowParallelLinear(nn.Module):
         if getattr(self, "fp8", False):
             y = fp8_linear(x, self.weight, self.weight_scale, None)
         else:
-            y = F.linear(x, self.weight)
+            y = _linear(x, self.weight)
         if self.reduce_output:
             if sequence_parallel:
                 # SP exit: reduce-scatter along S instead of all-reduce

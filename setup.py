@@ -25,6 +25,7 @@ sources = [
     os.path.join(CSRC, "sampling.hip"),
     os.path.join(CSRC, "attention_prefill.hip"),
     os.path.join(CSRC, "qkv_fused.hip"),
+    os.path.join(CSRC, "lt_gemm.cpp"),
 ]
 
 extra = {
@@ -46,6 +47,7 @@ setup(
             name="neuronx_distributed_inference_amd.ops._nxdi_C",
             sources=sources,
             extra_compile_args=extra,
+            libraries=["hipblaslt"],
         )
     ],
     cmdclass={"build_ext": BuildExtension},

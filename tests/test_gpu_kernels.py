@@ -446,3 +446,22 @@ def test_flash_prefill_sinks_gpu():
                     mask=causal_mask(S, S), scale=1.0 / math.sqrt(D),
                     sinks=sinks.cpu())
     torch.testing.assert_close(out.float().cpu(), want, atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize("M", [1, 64, 512])
+def test_lt_linear_matches_torch(M):
+    """Autotuned hipBLASLt linear vs F.linear (same math, library GEMM)."""
+    if not ops.lt_linear_enabled():
+        pytest.skip("hipblaslt unavailable")
+    torch.manual_seed(0)
+    dev = _dev()
+    N, K = 1024, 512
+    x = torch.randn(M, K, dtype=torch.bfloat16, device=dev)
+    w = torch.randn(N, K, dtype=torch.bfloat16, device=dev) * 0.05
+    b = torch.randn(N, dtype=torch.bfloat16, device=dev)
+    got = ops.lt_linear(x, w, b)
+    want = torch.nn.functional.linear(x, w, b)
+    torch.testing.assert_close(got.float(), want.float(), atol=2e-2, rtol=2e-2)
+    got2 = ops.lt_linear(x.view(1, M, K), w)  # 3D input, no bias
+    want2 = torch.nn.functional.linear(x.view(1, M, K), w)
+    torch.testing.assert_close(got2.float(), want2.float(), atol=2e-2, rtol=2e-2)
