@@ -105,11 +105,19 @@ class AttentionBase(nn.Module):
         self.q_norm: Optional[nn.Module] = None
         self.k_norm: Optional[nn.Module] = None
         self.sinks: Optional[nn.Parameter] = None  # GPT-OSS learned sinks
+        # Llama4 hooks (ref llama4 text attention): NoPE layers, complex-pair
+        # (interleaved) rope, weightless L2 qk-norm AFTER rope, and the
+        # log-temperature query scaling of NoPE layers
+        self.use_rope = True
+        self.rope_interleaved = False
+        self.qk_norm_post_rope: Optional[nn.Module] = None
+        self.q_scale_fn = None  # fn(position_ids (B,S)) -> (B,S) scale
 
     # ------------------------------------------------------------------
     def prep_qkv(
         self, hidden: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
         sequence_parallel: bool = False,
+        position_ids: Optional[torch.Tensor] = None,
     ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
         """Projections + BSHD->BHSD + optional qk-norm + RoPE
         (ref prep_qkv_tensors attention_base.py:489-563)."""
@@ -129,8 +137,20 @@ class AttentionBase(nn.Module):
         q = q.transpose(1, 2)  # BHSD
         k = k.transpose(1, 2)
         v = v.transpose(1, 2)
-        if cos is not None:
-            q, k = ops.rope_apply(q, k, cos, sin)
+        if cos is not None and self.use_rope:
+            if self.rope_interleaved:
+                from neuronx_distributed_inference_amd.modules.rope import (
+                    rope_apply_interleaved,
+                )
+
+                q, k = rope_apply_interleaved(q, k, cos, sin)
+            else:
+                q, k = ops.rope_apply(q, k, cos, sin)
+        if self.qk_norm_post_rope is not None:
+            q = self.qk_norm_post_rope(q)
+            k = self.qk_norm_post_rope(k)
+        if self.q_scale_fn is not None and position_ids is not None:
+            q = (q * self.q_scale_fn(position_ids).unsqueeze(1).unsqueeze(-1)).to(q.dtype)
         return q, k, v
 
     # ------------------------------------------------------------------
@@ -150,6 +170,8 @@ class AttentionBase(nn.Module):
         B, S, _ = hidden.shape
         if sequence_parallel:
             S = S * self.tp_degree  # hidden is S-sharded; attention runs full-S
+        if not self.use_rope:  # NoPE layer (Llama4 global-attention layers)
+            cos, sin = None, None
 
         # Fused TKG front end (K2-analog): one GEMM -> one kernel doing
         # RoPE + cache write + q relayout. Skipped when per-head qk-norms or
@@ -161,6 +183,9 @@ class AttentionBase(nn.Module):
             and self.clip_qkv is None
             and cos is not None
             and not getattr(kv_mgr, "is_paged", False)
+            and not self.rope_interleaved
+            and self.qk_norm_post_rope is None
+            and self.q_scale_fn is None
         ):
             y = self.qkv_proj.qkv(hidden)
             kc, vc = kv_mgr.get_cache(self.layer_idx)
@@ -189,6 +214,9 @@ class AttentionBase(nn.Module):
             and self.clip_qkv is None
             and cos is not None
             and seq_ids is not None
+            and not self.rope_interleaved
+            and self.qk_norm_post_rope is None
+            and self.q_scale_fn is None
             and not sequence_parallel
             and self.cp_degree == 1
             and self.dp_degree == 1
@@ -222,7 +250,8 @@ class AttentionBase(nn.Module):
         if phase == PHASE_TKG and self.dp_degree > 1 and B % self.dp_degree == 0:
             return self._dp_decode(hidden, cos, sin, kv_mgr, position_ids, seq_ids)
 
-        q, k, v = self.prep_qkv(hidden, cos, sin, sequence_parallel=sequence_parallel)
+        q, k, v = self.prep_qkv(hidden, cos, sin, sequence_parallel=sequence_parallel,
+                                position_ids=position_ids)
 
         # write this step's K/V into the cache first (in-place scatter)
         if self.dp_degree > 1 and phase == PHASE_CTE:
@@ -252,7 +281,7 @@ class AttentionBase(nn.Module):
         # decode-sharded K/V into the cache (standard projections; the q
         # output of this pass is discarded — the ref instead reorders kv
         # heads between meshes, attention/utils.py:455-623)
-        _, k_full, v_full = self.prep_qkv(hidden, cos, sin)
+        _, k_full, v_full = self.prep_qkv(hidden, cos, sin, position_ids=position_ids)
         kv_mgr.update_cache(self.layer_idx, k_full, v_full, position_ids, seq_ids)
 
         # CTE-sharded q/k/v over the prefill-TP subgroup
@@ -354,9 +383,13 @@ class AttentionBase(nn.Module):
         mask = None
         if attention_mask is not None:
             mask = attn_utils.padded_causal_mask(attention_mask, q.shape[2])
-            if self.sliding_window:
+            if self.sliding_window > 0:
                 mask = mask & attn_utils.sliding_window_mask(
                     q.shape[2], k.shape[2], self.sliding_window, device=q.device
+                )
+            elif self.sliding_window < 0:  # chunked attention (Llama4)
+                mask = mask & attn_utils.chunked_attention_mask(
+                    q.shape[2], k.shape[2], -self.sliding_window, device=q.device
                 )
             return ops.attention_prefill(
                 q, k, v, mask=mask, scale=self.scale, sinks=self.sinks, causal=False

@@ -23,6 +23,16 @@ def sliding_window_mask(sq: int, skv: int, window: int, device=None) -> torch.Te
     return m.view(1, 1, sq, skv)
 
 
+def chunked_attention_mask(sq: int, skv: int, chunk: int, device=None) -> torch.Tensor:
+    """Llama4 chunked attention: query at position i attends causally only
+    within its own chunk [(i//chunk)*chunk, i] (HF Llama4 attention_chunk_size;
+    ref llama4 chunked local layers). Queries are the LAST sq of skv."""
+    qpos = torch.arange(skv - sq, skv, device=device).view(-1, 1)
+    kpos = torch.arange(skv, device=device).view(1, -1)
+    m = (kpos <= qpos) & (kpos >= (qpos // chunk) * chunk)
+    return m.view(1, 1, sq, skv)
+
+
 def padded_causal_mask(attention_mask: torch.Tensor, sq: int) -> torch.Tensor:
     """Combine a (B, Skv) padding mask with causality -> (B,1,Sq,Skv)."""
     skv = attention_mask.shape[-1]

@@ -32,6 +32,7 @@ class MoERouter(nn.Module):
 
     def __init__(self, hidden_size: int, num_experts: int, top_k: int,
                  normalize: bool = True, pre_softmax: bool = False,
+                 act: str = "softmax",
                  dtype: torch.dtype = torch.float32):
         super().__init__()
         self.weight = nn.Parameter(torch.empty(num_experts, hidden_size, dtype=dtype))
@@ -39,9 +40,13 @@ class MoERouter(nn.Module):
         self.num_experts = num_experts
         self.normalize = normalize
         self.pre_softmax = pre_softmax
+        self.act = act  # "softmax" | "sigmoid" (Llama4: sigmoid of top-k logits)
 
     def forward(self, x: torch.Tensor):
         logits = F.linear(x.float(), self.weight.float())  # (T, E)
+        if self.act == "sigmoid":
+            vals, idx = logits.topk(self.top_k, dim=-1)
+            return torch.sigmoid(vals), idx
         if self.pre_softmax:
             probs = torch.softmax(logits, dim=-1)
             vals, idx = probs.topk(self.top_k, dim=-1)
@@ -72,10 +77,13 @@ class ExpertMLPs(nn.Module):
 
     def __init__(self, num_experts: int, hidden_size: int, intermediate_size: int,
                  tp_degree: int, dtype: torch.dtype, hidden_act: str = "silu",
-                 group=None, ep_degree: int = 1):
+                 group=None, ep_degree: int = 1, scale_input: bool = False):
         super().__init__()
         self.num_experts = num_experts
         self.ep_degree = ep_degree
+        # Llama4 quirk: the router gate scales the expert INPUT (y = E(g*x)),
+        # not the output (HF Llama4TextMoe routed_in scaling)
+        self.scale_input = scale_input
         moe_tp = divide(tp_degree, ep_degree)
         self.num_local_experts = divide(num_experts, ep_degree)
         ipp = divide(intermediate_size, moe_tp)
@@ -133,11 +141,13 @@ class ExpertMLPs(nn.Module):
             if not bool(rows.any()):
                 continue
             xe = x[rows]
+            w = (weights * sel).sum(dim=-1)[rows].to(x.dtype).unsqueeze(-1)
+            if self.scale_input:
+                xe = xe * w
             ge = F.linear(xe, self.w_gate[e_local])
             ue = F.linear(xe, self.w_up[e_local])
             he = F.linear(self.act_mul(ge, ue), self.w_down[e_local])
-            w = (weights * sel).sum(dim=-1)[rows].to(x.dtype).unsqueeze(-1)
-            out[rows] += he * w
+            out[rows] += he if self.scale_input else he * w
         return out
 
 
@@ -145,7 +155,8 @@ class MoELayer(nn.Module):
     """Router + experts (+ optional shared expert), TP all-reduce at the end."""
 
     def __init__(self, config, hidden_act: str = "silu",
-                 normalize_top_k: bool = True, pre_softmax: bool = False):
+                 normalize_top_k: bool = True, pre_softmax: bool = False,
+                 router_act: str = "softmax", scale_input: bool = False):
         super().__init__()
         ec = config.neuron_config
         E = getattr(config, "num_local_experts", None) or getattr(config, "num_experts")
@@ -153,11 +164,11 @@ class MoELayer(nn.Module):
         inter = getattr(config, "moe_intermediate_size", None) or config.intermediate_size
         self.router = MoERouter(
             config.hidden_size, E, top_k, normalize=normalize_top_k,
-            pre_softmax=pre_softmax,
+            pre_softmax=pre_softmax, act=router_act,
         )
         self.experts = ExpertMLPs(
             E, config.hidden_size, inter, ec.tp_degree, ec.torch_dtype,
-            hidden_act=hidden_act, ep_degree=ec.ep_degree,
+            hidden_act=hidden_act, ep_degree=ec.ep_degree, scale_input=scale_input,
         )
         self.shared_expert = None  # models with shared experts set this
 

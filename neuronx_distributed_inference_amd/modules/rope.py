@@ -137,3 +137,19 @@ class RotaryEmbedding(nn.Module):
         if self._cached_len > 0:
             pos = pos.clamp(max=self._cached_len - 1)
         return self._cos_cached[pos], self._sin_cached[pos]
+
+
+def rope_apply_interleaved(q: torch.Tensor, k: torch.Tensor,
+                           cos: torch.Tensor, sin: torch.Tensor):
+    """Complex-pair (interleaved) RoPE, pairing (2i, 2i+1) lanes — the HF
+    Llama4 `apply_rotary_emb` convention (view_as_complex), output kept
+    interleaved. cos/sin are the (B,S,D) duplicated-half tables produced by
+    RotaryEmbedding.get_cos_sin; pair i's angle sits in the first half."""
+    d = cos.shape[-1] // 2
+    c = cos[..., :d].unsqueeze(1)  # (B,1,S,d)
+    s = sin[..., :d].unsqueeze(1)
+    q1, q2 = q[..., 0::2].float(), q[..., 1::2].float()
+    k1, k2 = k[..., 0::2].float(), k[..., 1::2].float()
+    q_out = torch.stack([q1 * c - q2 * s, q2 * c + q1 * s], dim=-1).flatten(-2)
+    k_out = torch.stack([k1 * c - k2 * s, k2 * c + k1 * s], dim=-1).flatten(-2)
+    return q_out.to(q.dtype), k_out.to(k.dtype)

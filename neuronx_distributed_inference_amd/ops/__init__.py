@@ -134,16 +134,25 @@ def attention_prefill(q, k, v, mask=None, scale=None, sinks=None, causal=True,
         if mask is None and causal:
             from neuronx_distributed_inference_amd.modules.attention.utils import (
                 causal_mask,
+                chunked_attention_mask,
                 sliding_window_mask,
             )
             Sq, Skv = q.shape[2], k.shape[2]
             mask = causal_mask(Sq, Skv, device=q.device)
-            if sliding_window:
+            if sliding_window > 0:
                 mask = mask & sliding_window_mask(Sq, Skv, sliding_window, device=q.device)
+            elif sliding_window < 0:
+                mask = mask & chunked_attention_mask(Sq, Skv, -sliding_window, device=q.device)
         return ref.sdpa(q, k, v, mask=mask, scale=scale, sinks=sinks)
     if mask is None and causal:
         Sq, Skv = q.shape[2], k.shape[2]
         m = torch.ones(Sq, Skv, dtype=torch.bool, device=q.device).tril(Skv - Sq)
+        if sliding_window < 0:
+            from neuronx_distributed_inference_amd.modules.attention.utils import (
+                chunked_attention_mask,
+            )
+            m = m & chunked_attention_mask(Sq, Skv, -sliding_window).view(Sq, Skv)
+            sliding_window = 0
         if sliding_window:
             m = m & ~m.cumsum(-1).le(0)  # placeholder; window applied below
             m = torch.ones(Sq, Skv, dtype=torch.bool).tril(Skv - Sq)
@@ -262,8 +271,11 @@ def attention_decode(q, k_cache, v_cache, seq_lens, seq_ids=None, scale=None,
     ar = torch.arange(Smax, device=q.device).view(1, 1, 1, Smax)
     qpos = positions.view(B, 1, Sq, 1)
     mask = ar <= qpos
-    if sliding_window:
+    if sliding_window > 0:
         mask = mask & (ar > (qpos - sliding_window))
+    elif sliding_window < 0:  # chunked attention: own chunk only (Llama4)
+        C = -sliding_window
+        mask = mask & (ar >= (qpos.clamp(min=0) // C) * C)
     out = ref.sdpa(q, k, v, mask=mask, scale=scale, sinks=sinks)
     return torch.nan_to_num(out)  # pad queries (pos<0) mask everything
 
@@ -365,7 +377,10 @@ def attention_decode_paged(q, k_cache, v_cache, seq_lens, block_table,
     ar = torch.arange(max_len).view(1, 1, 1, -1)
     qpos = positions.view(B, 1, Sq, 1)
     mask = ar <= qpos
-    if sliding_window:
+    if sliding_window > 0:
         mask = mask & (ar > (qpos - sliding_window))
+    elif sliding_window < 0:  # chunked attention: own chunk only (Llama4)
+        C = -sliding_window
+        mask = mask & (ar >= (qpos.clamp(min=0) // C) * C)
     out = ref.sdpa(q, k, v, mask=mask, scale=scale)
     return torch.nan_to_num(out)

@@ -62,6 +62,9 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
     const int eff_len = positions[(long)b * Sq + sq] + 1;
     int lo = 0;
     if (window > 0 && eff_len > window) lo = eff_len - window;
+    // window < 0: Llama4 CHUNKED attention with chunk size -window — the
+    // query attends only rows in its own chunk [(p/C)*C, p]
+    else if (window < 0) lo = ((eff_len - 1) / (-window)) * (-window);
     const int split_lo = max(lo, split * chunk);
     const int split_hi = min(eff_len, (split + 1) * chunk);
 

@@ -128,6 +128,9 @@ __global__ __launch_bounds__(256, 2) void flash_prefill_kernel(
     const int kv_hi = min(S, q_lo + QBLK);  // causal: no keys beyond block end
     int kv_start = 0;
     if (window > 0) kv_start = max(0, (q_lo - window + 1) / KVBLK * KVBLK);
+    // window < 0: chunked attention (chunk C = -window); the block's first
+    // query bounds the earliest reachable row at its chunk start
+    else if (window < 0) kv_start = (q_lo / -window) * -window / KVBLK * KVBLK;
 
     // Staging note: occupancy (not staging latency) is the binding
     // constraint — keep live state minimal so 2 waves/SIMD fit (the
@@ -182,7 +185,8 @@ __global__ __launch_bounds__(256, 2) void flash_prefill_kernel(
             // tiles, and the masking VALU was a large share of wave cycles.
             const bool full_tile =
                 (k0 + KVBLK - 1 <= wq_lo) && (k0 + KVBLK <= S)
-                && (window <= 0 || k0 >= wq_lo + WQ - window);
+                && (window <= 0 || k0 >= wq_lo + WQ - window)
+                && (window >= 0 || k0 >= ((wq_lo + WQ - 1) / -window) * -window);
             float p_new[2][4];  // per-qi per-reg row max of this tile
 #pragma unroll
             for (int qi = 0; qi < 2; ++qi) {
@@ -203,7 +207,8 @@ __global__ __launch_bounds__(256, 2) void flash_prefill_kernel(
                         const int k_abs = k0 + ki * 16 + (lane & 15);
                         float sv = s_frag[qi][ki][r] * scale;
                         const bool ok = (k_abs <= q_abs) && (k_abs < S) && (q_abs < S)
-                            && (window <= 0 || k_abs > q_abs - window);
+                            && (window <= 0 || k_abs > q_abs - window)
+                            && (window >= 0 || k_abs >= (q_abs / -window) * -window);
                         sv = ok ? sv : NEG_INF;
                         s_frag[qi][ki][r] = sv;
                         mx = fmaxf(mx, sv);

@@ -56,6 +56,11 @@ _REGISTRY: Dict[str, Tuple[str, str, str]] = {
         "Gemma3InferenceConfig",
         "Gemma3ForCausalLM",
     ),
+    "llama4": (
+        "neuronx_distributed_inference_amd.models.llama4.modeling_llama4",
+        "Llama4InferenceConfig",
+        "Llama4ForCausalLM",
+    ),
     "qwen3_moe": (
         "neuronx_distributed_inference_amd.models.qwen3_moe.modeling_qwen3_moe",
         "Qwen3MoeInferenceConfig",

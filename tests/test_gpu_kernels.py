@@ -465,3 +465,41 @@ def test_lt_linear_matches_torch(M):
     got2 = ops.lt_linear(x.view(1, M, K), w)  # 3D input, no bias
     want2 = torch.nn.functional.linear(x.view(1, M, K), w)
     torch.testing.assert_close(got2.float(), want2.float(), atol=2e-2, rtol=2e-2)
+
+
+def test_chunked_attention_prefill_kernel():
+    """Negative sliding_window = Llama4 chunked attention in the flash
+    kernel: must match the chunked-mask fp32 reference."""
+    torch.manual_seed(0)
+    B, Hq, Hkv, S, D, C = 2, 4, 2, 512, 128, 96
+    dev = _dev()
+    q = torch.randn(B, Hq, S, D, dtype=torch.bfloat16, device=dev)
+    k = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16, device=dev)
+    v = torch.randn(B, Hkv, S, D, dtype=torch.bfloat16, device=dev)
+    out = ops.attention_prefill(q, k, v, causal=True, sliding_window=-C)
+    from neuronx_distributed_inference_amd.modules.attention.utils import (
+        chunked_attention_mask,
+    )
+
+    want = ref.sdpa(q.float().cpu(), k.float().cpu(), v.float().cpu(),
+                    mask=chunked_attention_mask(S, S, C),
+                    scale=1.0 / math.sqrt(D))
+    torch.testing.assert_close(out.float().cpu(), want, atol=3e-2, rtol=3e-2)
+
+
+def test_chunked_attention_decode_kernel():
+    """Chunked decode: queries at positions crossing chunk boundaries must
+    only see their own chunk."""
+    torch.manual_seed(0)
+    B, Hq, Hkv, Smax, D, C = 2, 4, 2, 256, 128, 32
+    dev = _dev()
+    kc = torch.randn(B, Hkv, Smax, D, dtype=torch.bfloat16, device=dev)
+    vc = torch.randn(B, Hkv, Smax, D, dtype=torch.bfloat16, device=dev)
+    q = torch.randn(B, Hq, 1, D, dtype=torch.bfloat16, device=dev)
+    for plen in (31, 32, 33, 97, 200):
+        lens = torch.tensor([plen, plen - 7], dtype=torch.int32, device=dev)
+        out = ops.attention_decode(q, kc, vc, lens, sliding_window=-C)
+        want = ops.attention_decode(q.cpu(), kc.float().cpu(), vc.float().cpu(),
+                                    lens.cpu(), sliding_window=-C)
+        torch.testing.assert_close(out.float().cpu(), want.float(),
+                                   atol=3e-2, rtol=3e-2)

@@ -247,3 +247,36 @@ def test_deepseek_v3():
     assert not [m for m in missing if not m.startswith("kv_mgr")], missing
     assert not unexpected, unexpected
     _check_match(ours, hf)
+
+
+@torch.inference_mode()
+def test_llama4():
+    """Llama4 text: interleaved rope, NoPE + temperature tuning, chunked
+    local attention, sigmoid top-1 input-scaled MoE + shared expert."""
+    from transformers.models.llama4 import Llama4TextConfig
+    from transformers.models.llama4.modeling_llama4 import Llama4ForCausalLM as HF
+
+    from neuronx_distributed_inference_amd.models.llama4 import (
+        Llama4ForCausalLM, Llama4InferenceConfig,
+    )
+
+    torch.manual_seed(0)
+    hf_cfg = Llama4TextConfig(
+        vocab_size=128, hidden_size=64, intermediate_size=128,
+        intermediate_size_mlp=160, num_hidden_layers=4,
+        num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, max_position_embeddings=128, tie_word_embeddings=False,
+        num_local_experts=4, num_experts_per_tok=1,
+        interleave_moe_layer_step=2, no_rope_layer_interval=2,
+        attention_chunk_size=8, use_qk_norm=True,
+        attn_temperature_tuning=True, floor_scale=8, attn_scale=0.1,
+        rms_norm_eps=1e-5,
+    )
+    hf = HF(hf_cfg).eval()
+    cfg = Llama4InferenceConfig(neuron_config=_engine(), **hf_cfg.to_dict())
+    ours = Llama4ForCausalLM(cfg).eval()
+    sd = Llama4ForCausalLM.convert_hf_to_native_state_dict(hf.state_dict(), cfg)
+    missing, unexpected = ours.model.load_state_dict(sd, strict=False)
+    assert not [m for m in missing if not m.startswith("kv_mgr")], missing
+    assert not unexpected, unexpected
+    _check_match(ours, hf, n_new=12)
