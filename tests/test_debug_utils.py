@@ -50,3 +50,25 @@ def test_snapshot_hook(tiny_llama_config, tmp_path, monkeypatch):
     assert any(f.startswith("context_encoding_model_req0") for f in files)
     payload = torch.load(os.path.join(tmp_path, files[0]), weights_only=False)
     assert torch.equal(payload["input_ids"], ids)
+
+
+def test_async_executor_overlap_and_backpressure():
+    """AsyncExecutor: results arrive in order, back-pressure caps inflight,
+    drain() completes everything (ref modules/async_execution.py analog)."""
+    from neuronx_distributed_inference_amd.modules.async_execution import (
+        AsyncExecutor,
+    )
+
+    calls = []
+
+    def work(i):
+        calls.append(i)
+        return torch.full((2,), float(i))
+
+    with AsyncExecutor(max_inflight=2) as ex:
+        handles = [ex.submit(work, i) for i in range(5)]
+        outs = [h.result() for h in handles]
+    assert calls == list(range(5))
+    for i, o in enumerate(outs):
+        assert torch.equal(o, torch.full((2,), float(i)))
+    assert all(h.done() for h in handles)
