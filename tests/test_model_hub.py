@@ -280,3 +280,42 @@ def test_llama4():
     assert not [m for m in missing if not m.startswith("kv_mgr")], missing
     assert not unexpected, unexpected
     _check_match(ours, hf, n_new=12)
+
+
+@torch.inference_mode()
+def test_whisper():
+    """Whisper encoder-decoder: greedy tokens must match a stepwise
+    teacher-forced HF forward (cache-free oracle)."""
+    from transformers import WhisperConfig
+    from transformers import WhisperForConditionalGeneration as HF
+
+    from neuronx_distributed_inference_amd.models.whisper import (
+        WhisperForConditionalGeneration, WhisperInferenceConfig,
+    )
+
+    torch.manual_seed(0)
+    hf_cfg = WhisperConfig(
+        vocab_size=100, num_mel_bins=8, d_model=32, encoder_layers=2,
+        decoder_layers=2, encoder_attention_heads=4, decoder_attention_heads=4,
+        encoder_ffn_dim=64, decoder_ffn_dim=64, max_source_positions=24,
+        max_target_positions=40, decoder_start_token_id=1, eos_token_id=2,
+        pad_token_id=0, bos_token_id=None,
+    )
+    hf = HF(hf_cfg).eval()
+    cfg = WhisperInferenceConfig(neuron_config=_engine(), **hf_cfg.to_dict())
+    ours = WhisperForConditionalGeneration(cfg).eval()
+    sd = WhisperForConditionalGeneration.convert_hf_to_native_state_dict(
+        hf.state_dict(), cfg)
+    missing, unexpected = ours.load_state_dict(sd, strict=False)
+    assert not missing, missing
+    assert not unexpected, unexpected
+
+    feats = torch.randn(2, 8, 48)
+    n_new = 10
+    # oracle: stepwise cache-free HF greedy
+    ids = torch.full((2, 1), 1, dtype=torch.long)
+    for _ in range(n_new):
+        lg = hf(input_features=feats, decoder_input_ids=ids).logits
+        ids = torch.cat([ids, lg[:, -1].argmax(-1, keepdim=True)], dim=1)
+    got = ours.generate(feats, max_new_tokens=n_new)
+    assert torch.equal(got[:, : ids.shape[1]], ids), (got.tolist(), ids.tolist())
