@@ -319,3 +319,67 @@ def test_whisper():
         ids = torch.cat([ids, lg[:, -1].argmax(-1, keepdim=True)], dim=1)
     got = ours.generate(feats, max_new_tokens=n_new)
     assert torch.equal(got[:, : ids.shape[1]], ids), (got.tolist(), ids.tolist())
+
+
+@torch.inference_mode()
+def test_mllama():
+    """Mllama (Llama-3.2 Vision): tiled ViT + gated cross-attention text
+    model; greedy tokens vs stepwise cache-free HF forward with an image."""
+    from transformers import MllamaConfig
+    from transformers import MllamaForConditionalGeneration as HF
+    from transformers.models.mllama import MllamaTextConfig, MllamaVisionConfig
+
+    from neuronx_distributed_inference_amd.models.mllama import (
+        MllamaForConditionalGeneration, MllamaInferenceConfig,
+    )
+
+    torch.manual_seed(0)
+    tc = MllamaTextConfig(
+        vocab_size=120, hidden_size=32, intermediate_size=64,
+        num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2,
+        cross_attention_layers=[1, 3], rope_theta=10000.0,
+        max_position_embeddings=64, bos_token_id=3, eos_token_id=2,
+        pad_token_id=0, rope_scaling=None, tie_word_embeddings=False)
+    vc = MllamaVisionConfig(
+        hidden_size=32, intermediate_size=64, num_hidden_layers=3,
+        num_global_layers=1, attention_heads=4, image_size=32, patch_size=16,
+        max_num_tiles=2, supported_aspect_ratios=[[1, 1], [1, 2], [2, 1]],
+        intermediate_layers_indices=[1, 2], vision_output_dim=96, num_channels=3)
+    hf_cfg = MllamaConfig(vision_config=vc.to_dict(), text_config=tc.to_dict())
+    hf = HF(hf_cfg).eval()
+
+    ec = EngineConfig(batch_size=1, max_batch_size=1, seq_len=48, max_length=48,
+                      torch_dtype="float32", tp_degree=1)
+    cfg = MllamaInferenceConfig(neuron_config=ec,
+                                text_config=tc.to_dict(),
+                                vision_config=vc.to_dict())
+    ours = MllamaForConditionalGeneration(cfg).eval()
+    sd = MllamaForConditionalGeneration.convert_hf_to_native_state_dict(
+        hf.state_dict(), cfg)
+    missing, unexpected = ours.load_state_dict(sd, strict=False)
+    real_missing = [m for m in missing if ".kv_mgr." not in m and not m.startswith("kv_mgr")]
+    assert not real_missing, real_missing
+    assert not unexpected, unexpected
+
+    px = torch.randn(1, 1, 2, 3, 32, 32)
+    ar_ids = torch.tensor([[2]])           # aspect ratio [1,2] -> id 2
+    ar_mask = torch.tensor([[[1, 1]]])
+    ids = torch.randint(4, 120, (1, 6))
+    n_new = 8
+    ref_ids = ids.clone()
+    for _ in range(n_new):
+        lg = hf(input_ids=ref_ids, pixel_values=px, aspect_ratio_ids=ar_ids,
+                aspect_ratio_mask=ar_mask, use_cache=False).logits
+        ref_ids = torch.cat([ref_ids, lg[:, -1].argmax(-1, keepdim=True)], dim=1)
+    got = ours.generate(ids, pixel_values=px, aspect_ratio_ids=ar_ids,
+                        aspect_ratio_mask=ar_mask, max_new_tokens=n_new)
+    assert torch.equal(got[:, : ref_ids.shape[1]], ref_ids), (
+        got.tolist(), ref_ids.tolist())
+
+    # text-only path must also match (cross layers skipped)
+    ref_ids = ids.clone()
+    for _ in range(4):
+        lg = hf(input_ids=ref_ids, use_cache=False).logits
+        ref_ids = torch.cat([ref_ids, lg[:, -1].argmax(-1, keepdim=True)], dim=1)
+    got = ours.generate(ids, max_new_tokens=4)
+    assert torch.equal(got[:, : ref_ids.shape[1]], ref_ids)
