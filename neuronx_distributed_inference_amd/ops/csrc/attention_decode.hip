@@ -220,7 +220,9 @@ __global__ void decode_attn_pass2(
     const float* __restrict__ ml_part, // (B, Hq, Sq, nsplit, 2)
     const int nsplit, const float v_scale,
     const float* __restrict__ sinks,   // (Hq,) learned sink logits or null
-    const int Hq, const int Sq) {
+    const int Hq, const int Sq,
+    float* __restrict__ lse_out) {     // (B,Hq,Sq) log-sum-exp, or null —
+                                       // the cross-rank flash-decoding handle
     using VT = VecTraits<T>;
     const long rowid = blockIdx.x;  // over B*Hq*Sq
     const int lane = threadIdx.x;
@@ -240,6 +242,8 @@ __global__ void decode_attn_pass2(
         L += li * __expf(mi - mstar);
     }
     const float inv = (L > 0.f) ? v_scale / L : 0.f;
+    if (lse_out != nullptr && lane == 0)
+        lse_out[rowid] = (L > 0.f) ? mstar + __logf(L) : -INFINITY;
 #pragma unroll
     for (int d = lane; d < D; d += WAVE_SIZE) {
         float acc = 0.f;
@@ -259,7 +263,8 @@ static void launch_decode(torch::Tensor& out, const torch::Tensor& q,
                           float scale, int window, int gq,
                           const torch::Tensor* block_table = nullptr,
                           int block_size = 0, float v_scale = 1.0f,
-                          const torch::Tensor* sinks = nullptr) {
+                          const torch::Tensor* sinks = nullptr,
+                          torch::Tensor* lse_out = nullptr) {
     const int B = q.size(0), Hq = q.size(1), Sq = q.size(2);
     const bool paged = block_table != nullptr;
     const int Hkv = paged ? k_cache.size(2) : k_cache.size(1);
@@ -298,7 +303,8 @@ static void launch_decode(torch::Tensor& out, const torch::Tensor& q,
     hipLaunchKernelGGL((decode_attn_pass2<T, D>), grid2, dim3(WAVE_SIZE), 0, stream,
                        (T*)out.data_ptr(), o_part.data_ptr<float>(),
                        ml_part.data_ptr<float>(), nsplit, v_scale,
-                       sinks ? sinks->data_ptr<float>() : nullptr, Hq, Sq);
+                       sinks ? sinks->data_ptr<float>() : nullptr, Hq, Sq,
+                       lse_out ? lse_out->data_ptr<float>() : nullptr);
     HIP_CHECK_KERNEL();
 }
 

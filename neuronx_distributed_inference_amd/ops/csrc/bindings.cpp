@@ -30,6 +30,8 @@ void qkv_rope_update(torch::Tensor q_out, torch::Tensor qkv, torch::Tensor cosb,
                      torch::Tensor sinb, torch::Tensor k_cache, torch::Tensor v_cache,
                      torch::Tensor positions, torch::Tensor seq_ids, long Hq, long Hkv);
 void silu_mul_fused(torch::Tensor out, torch::Tensor gate_up);
+void skinny_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
+                 c10::optional<torch::Tensor> bias);
 bool lt_available();
 void lt_linear(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                c10::optional<torch::Tensor> bias);
@@ -61,6 +63,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("seq_ids"), py::arg("scale"), py::arg("window"),
           py::arg("sinks") = py::none());
     m.def("qkv_rope_update", &qkv_rope_update, "fused qkv->rope->cache-write (CDNA4)");
+    m.def("skinny_gemm", &skinny_gemm, "skinny-M weight-streaming GEMM (CDNA4)",
+          py::arg("y"), py::arg("x"), py::arg("w"), py::arg("bias") = py::none());
     m.def("lt_available", &lt_available, "hipblaslt usable on this device");
     m.def("lt_linear", &lt_linear, "autotuned hipBLASLt linear (graph-safe)",
           py::arg("y"), py::arg("x"), py::arg("w"), py::arg("bias") = py::none());

@@ -503,3 +503,26 @@ def test_chunked_attention_decode_kernel():
                                     lens.cpu(), sliding_window=-C)
         torch.testing.assert_close(out.float().cpu(), want.float(),
                                    atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize("M", [16, 100, 128])
+def test_skinny_gemm_matches_torch(M):
+    """Experimental skinny GEMM numerics vs F.linear (incl. padded lda and
+    non-multiple-of-16 M)."""
+    from neuronx_distributed_inference_amd.ops import _nxdi_C as C
+
+    torch.manual_seed(0)
+    dev = _dev()
+    N, K = 512, 1024
+    x = torch.randn(M, K, dtype=torch.bfloat16, device=dev) * 0.1
+    w = torch.randn(N, K, dtype=torch.bfloat16, device=dev) * 0.1
+    b = torch.randn(N, dtype=torch.bfloat16, device=dev)
+    y = torch.empty(M, N, dtype=torch.bfloat16, device=dev)
+    C.skinny_gemm(y, x, w, b)
+    want = torch.nn.functional.linear(x, w, b)
+    torch.testing.assert_close(y.float(), want.float(), atol=2e-2, rtol=2e-2)
+    wp = torch.empty(N, K + 64, dtype=torch.bfloat16, device=dev)[:, :K]
+    wp.copy_(w)
+    C.skinny_gemm(y, x, wp)
+    want = torch.nn.functional.linear(x, w)
+    torch.testing.assert_close(y.float(), want.float(), atol=2e-2, rtol=2e-2)
