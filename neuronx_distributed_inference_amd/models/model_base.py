@@ -38,6 +38,7 @@ from neuronx_distributed_inference_amd.modules.kvcache.kv_cache_manager import K
 from neuronx_distributed_inference_amd.modules.norm import RMSNorm
 from neuronx_distributed_inference_amd.modules.rope import RotaryEmbedding
 from neuronx_distributed_inference_amd.parallel import mappings
+from neuronx_distributed_inference_amd.parallel import state as ps
 from neuronx_distributed_inference_amd.parallel.layers import (
     ColumnParallelLinear,
     ParallelEmbedding,
@@ -107,11 +108,15 @@ class DecoderModel(nn.Module):
                 tp_degree=cache_tp,
             )
         else:
+            # flash decoding: the decode cache is S-sharded over the
+            # kv-shard group (strided rows) — each rank holds 1/fd of S
+            fd = ps.get_kv_shard_degree() if getattr(
+                ec, "flash_decoding_enabled", False) else 1
             self.kv_mgr = KVCacheManager(
                 num_layers=self.num_layers,
                 num_kv_heads=n_kv_heads,
                 head_dim=head_dim,
-                max_len=ec.max_length,
+                max_len=(ec.max_length + fd - 1) // fd,
                 batch_size=ec.kv_cache_batch_size,
                 dtype=cache_dtype,
                 padding_rows=ec.kv_cache_padding_size,

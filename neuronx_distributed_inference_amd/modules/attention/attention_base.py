@@ -110,6 +110,12 @@ class AttentionBase(nn.Module):
         # log-temperature query scaling of NoPE layers
         self.use_rope = True
         self.rope_interleaved = False
+        # flash decoding: decode KV S-sharded over the
+        # kv-shard group; requires no sinks and plain causal masking
+        self.kv_shard_degree = (
+            ps.get_kv_shard_degree()
+            if getattr(ec, "flash_decoding_enabled", False) else 1
+        )
         self.qk_norm_post_rope: Optional[nn.Module] = None
         self.q_scale_fn = None  # fn(position_ids (B,S)) -> (B,S) scale
 
@@ -191,10 +197,18 @@ class AttentionBase(nn.Module):
             kc, vc = kv_mgr.get_cache(self.layer_idx)
             import neuronx_distributed_inference_amd.ops as _ops
 
+            wpos = position_ids
+            if self.kv_shard_degree > 1:
+                from neuronx_distributed_inference_amd.modules.attention import (
+                    flash_decoding as _fdm,
+                )
+
+                wpos = _fdm.shard_write_positions(
+                    position_ids, self.kv_shard_degree, ps.get_kv_shard_rank())
             q = _ops.qkv_rope_update(
                 y, cos, sin, kc.data if hasattr(kc, "data") else kc,
                 vc.data if hasattr(vc, "data") else vc,
-                position_ids, seq_ids,
+                wpos, seq_ids,
                 self.num_heads_per_rank, self.num_kv_heads_per_rank,
             )
             out = self._decode(q, kv_mgr, position_ids, seq_ids, active_mask)
@@ -221,6 +235,7 @@ class AttentionBase(nn.Module):
             and self.cp_degree == 1
             and self.dp_degree == 1
             and not getattr(kv_mgr, "is_paged", False)
+            and self.kv_shard_degree == 1
             and (not hidden.is_cuda or self.head_dim in (64, 128))
         ):
             kc, vc = kv_mgr.get_cache(self.layer_idx)
@@ -259,7 +274,15 @@ class AttentionBase(nn.Module):
             # B/dp local rows) — write it from the TKG projection set
             self._dp_write_cte_cache(hidden, cos, sin, kv_mgr, position_ids, seq_ids)
         else:
-            kv_mgr.update_cache(self.layer_idx, k, v, position_ids, seq_ids)
+            wpos = position_ids
+            if self.kv_shard_degree > 1:
+                from neuronx_distributed_inference_amd.modules.attention import (
+                    flash_decoding as _fdm,
+                )
+
+                wpos = _fdm.shard_write_positions(
+                    position_ids, self.kv_shard_degree, ps.get_kv_shard_rank())
+            kv_mgr.update_cache(self.layer_idx, k, v, wpos, seq_ids)
 
         if phase == PHASE_CTE:
             out = self._prefill(q, k, v, attention_mask)
@@ -407,6 +430,17 @@ class AttentionBase(nn.Module):
                 q, k_cache.data, v_cache.data, position_ids, kv_mgr.block_table,
                 scale=self.scale, sliding_window=self.sliding_window,
             )
+        if self.kv_shard_degree > 1 and self.sinks is None and self.sliding_window == 0:
+            from neuronx_distributed_inference_amd.modules.attention import (
+                flash_decoding as _fdm,
+            )
+
+            B, _, Sq, _ = q.shape
+            pos = position_ids if position_ids.dim() == 2 else (
+                position_ids.view(B, 1) - Sq + torch.arange(
+                    Sq, device=q.device).view(1, Sq))
+            return _fdm.flash_decoding_attention(
+                q, k_cache, v_cache, pos, seq_ids, self.scale)
         return ops.attention_decode(
             q, k_cache, v_cache, position_ids, seq_ids=seq_ids, scale=self.scale,
             sinks=self.sinks, sliding_window=self.sliding_window,

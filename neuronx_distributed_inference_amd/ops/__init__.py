@@ -280,6 +280,67 @@ def attention_decode(q, k_cache, v_cache, seq_lens, seq_ids=None, scale=None,
     return torch.nan_to_num(out)  # pad queries (pos<0) mask everything
 
 
+def attention_decode_lse(q, k_cache, v_cache, seq_lens, seq_ids=None,
+                         scale=None, sliding_window: int = 0):
+    """attention_decode that ALSO returns the per-query log-sum-exp
+    (fp32 (B,Hq,Sq)) — the cross-rank flash-decoding combine handle
+    (SURVEY §2.8 long-context mechanism 3). Empty query ranges get
+    lse=-inf / out=0 so they drop out of the combine."""
+    import math
+    scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+    B, Hq, Sq, D = q.shape
+    if seq_lens.dim() == 1:
+        positions = seq_lens.view(B, 1) - Sq + torch.arange(
+            Sq, device=seq_lens.device).view(1, Sq)
+    else:
+        positions = seq_lens
+    if q.is_cuda:
+        c = _require_ext("attention_decode")
+        out = torch.empty_like(q)
+        lse = torch.empty(B, Hq, Sq, dtype=torch.float32, device=q.device)
+        if seq_ids is None:
+            seq_ids = torch.arange(B, device=q.device, dtype=torch.int32)
+        c.attention_decode(
+            out, q.contiguous(), k_cache, v_cache,
+            positions.to(torch.int32).contiguous(),
+            seq_ids.to(torch.int32).contiguous(), float(scale),
+            int(sliding_window), lse=lse,
+        )
+        return out, lse
+    # CPU reference
+    if seq_ids is None:
+        seq_ids = torch.arange(B, device=q.device)
+    k = k_cache[seq_ids.long()].float()
+    v = v_cache[seq_ids.long()].float()
+    Smax = k.shape[2]
+    rep = Hq // k.shape[1]
+    if rep > 1:
+        k = k.repeat_interleave(rep, dim=1)
+        v = v.repeat_interleave(rep, dim=1)
+    ar = torch.arange(Smax, device=q.device).view(1, 1, 1, Smax)
+    qpos = positions.view(B, 1, Sq, 1)
+    mask = ar <= qpos
+    if sliding_window > 0:
+        mask = mask & (ar > (qpos - sliding_window))
+    scores = torch.einsum("bhsd,bhtd->bhst", q.float(), k) * scale
+    scores = scores.masked_fill(~mask, float("-inf"))
+    lse = torch.logsumexp(scores, dim=-1)  # -inf where no valid rows
+    out = torch.einsum("bhst,bhtd->bhsd", torch.softmax(scores, dim=-1), v)
+    return torch.nan_to_num(out).to(q.dtype), lse
+
+
+def combine_lse_partials(outs, lses):
+    """Merge flash-decoding partials: outs[i] (B,H,S,D) softmax-normalized
+    over shard i's rows, lses[i] (B,H,S). Standard LSE-weighted merge."""
+    lse = torch.stack(lses, dim=0)                       # (P,B,H,S)
+    m = lse.max(dim=0, keepdim=True).values
+    w = torch.exp(lse - m)                               # -inf -> 0
+    w = torch.nan_to_num(w)
+    num = sum(wi.unsqueeze(-1) * oi.float() for wi, oi in zip(w, outs))
+    den = w.sum(dim=0).clamp_min(1e-20).unsqueeze(-1)
+    return (num / den).to(outs[0].dtype)
+
+
 def topk_topp_sample(logits, top_k, top_p, temperature, rand, global_topk: int = 256):
     if logits.is_cuda:
         c = _require_ext("topk_topp_sample")

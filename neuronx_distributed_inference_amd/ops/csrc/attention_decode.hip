@@ -314,22 +314,27 @@ static void dispatch_gq(torch::Tensor& out, const torch::Tensor& q,
                         const torch::Tensor& seq_lens, const torch::Tensor& seq_ids,
                         float scale, int window, int gq,
                         const torch::Tensor* bt = nullptr, int bs = 0,
-                        float v_scale = 1.0f, const torch::Tensor* sinks = nullptr) {
+                        float v_scale = 1.0f, const torch::Tensor* sinks = nullptr,
+                        torch::Tensor* lse_out = nullptr) {
     const bool fp8 = k_cache.scalar_type() == at::ScalarType::Float8_e4m3fn;
 #define GQ_CASE(CT)                                                                     \
     do {                                                                                \
         if (gq <= 1)                                                                    \
             launch_decode<T, CT, D, 1>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
-                                       scale, window, gq, bt, bs, v_scale, sinks);      \
+                                       scale, window, gq, bt, bs, v_scale, sinks,       \
+                                       lse_out);                                        \
         else if (gq <= 2)                                                               \
             launch_decode<T, CT, D, 2>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
-                                       scale, window, gq, bt, bs, v_scale, sinks);      \
+                                       scale, window, gq, bt, bs, v_scale, sinks,       \
+                                       lse_out);                                        \
         else if (gq <= 4)                                                               \
             launch_decode<T, CT, D, 4>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
-                                       scale, window, gq, bt, bs, v_scale, sinks);      \
+                                       scale, window, gq, bt, bs, v_scale, sinks,       \
+                                       lse_out);                                        \
         else if (gq <= 8)                                                               \
             launch_decode<T, CT, D, 8>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
-                                       scale, window, gq, bt, bs, v_scale, sinks);      \
+                                       scale, window, gq, bt, bs, v_scale, sinks,       \
+                                       lse_out);                                        \
         else                                                                            \
             TORCH_CHECK(false, "GQA ratio > 8 not supported yet (got ", gq, ")");       \
     } while (0)
@@ -385,7 +390,10 @@ void attention_decode(torch::Tensor out, torch::Tensor q,
                       torch::Tensor k_cache, torch::Tensor v_cache,
                       torch::Tensor seq_lens, torch::Tensor seq_ids,
                       double scale, long window,
-                      c10::optional<torch::Tensor> sinks_opt) {
+                      c10::optional<torch::Tensor> sinks_opt,
+                      c10::optional<torch::Tensor> lse_opt) {
+    // lse_opt: fp32 (B, Hq, Sq), filled with per-query log-sum-exp — the
+    // cross-rank flash-decoding combine handle
     // `seq_lens` carries per-query POSITIONS, shape (B, Sq)
     TORCH_CHECK(q.dim() == 4 && k_cache.dim() == 4, "bad shapes");
     TORCH_CHECK(seq_lens.numel() == q.size(0) * q.size(2),
@@ -402,25 +410,34 @@ void attention_decode(torch::Tensor out, torch::Tensor q,
         sinks_t = sinks_opt->to(torch::kFloat32).contiguous();
         sinks = &sinks_t;
     }
+    torch::Tensor* lse_ptr = nullptr;
+    torch::Tensor lse_t;
+    if (lse_opt.has_value()) {
+        lse_t = *lse_opt;
+        TORCH_CHECK(lse_t.scalar_type() == torch::kFloat32 &&
+                    lse_t.numel() == (long)q.size(0) * Hq * q.size(2),
+                    "lse must be fp32 (B,Hq,Sq)");
+        lse_ptr = &lse_t;
+    }
 
 #define CASE_T(T)                                                               \
     do {                                                                        \
         if (D == 128)                                                           \
             dispatch_gq<T, 128>(out, q, k_cache, v_cache, seq_lens, seq_ids,    \
                                 (float)scale, (int)window, gq, nullptr, 0,      \
-                                1.0f, sinks);                                   \
+                                1.0f, sinks, lse_ptr);                          \
         else if (D == 64)                                                       \
             dispatch_gq<T, 64>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
                                (float)scale, (int)window, gq, nullptr, 0,       \
-                               1.0f, sinks);                                    \
+                               1.0f, sinks, lse_ptr);                           \
         else if (D == 32)                                                       \
             dispatch_gq<T, 32>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
                                (float)scale, (int)window, gq, nullptr, 0,       \
-                               1.0f, sinks);                                    \
+                               1.0f, sinks, lse_ptr);                           \
         else if (D == 256)                                                      \
             dispatch_gq<T, 256>(out, q, k_cache, v_cache, seq_lens, seq_ids,    \
                                 (float)scale, (int)window, gq, nullptr, 0,      \
-                                1.0f, sinks);                                   \
+                                1.0f, sinks, lse_ptr);                          \
         else                                                                    \
             TORCH_CHECK(false, "head_dim must be 32/64/128/256, got ", D);      \
     } while (0)

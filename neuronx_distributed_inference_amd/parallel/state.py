@@ -47,6 +47,8 @@ class _ParallelState:
     dp_group: Optional[object] = None           # attention data-parallel (decode)
     decode_tp_group: Optional[object] = None
     draft_tp_group: Optional[object] = None     # speculative draft sub-communicator
+    kv_shard_degree: int = 1
+    kv_shard_group: Optional[object] = None     # flash-decoding S-shard group
     tp_mesh: List[List[int]] = field(default_factory=list)
 
 
@@ -72,6 +74,7 @@ def initialize_model_parallel(
     context_parallel_size: int = 1,
     attention_dp_size: int = 1,
     draft_tp_size: Optional[int] = None,
+    kv_shard_size: int = 1,
 ) -> None:
     """Create all communicators. Must be called identically on every rank."""
     global _STATE
@@ -86,6 +89,7 @@ def initialize_model_parallel(
         ep_degree=ep,
         cp_degree=context_parallel_size,
         attention_dp_degree=attention_dp_size,
+        kv_shard_degree=kv_shard_size,
     )
 
     if world == 1 or not dist.is_initialized():
@@ -148,6 +152,18 @@ def initialize_model_parallel(
                 g = dist.new_group(sub)
                 if rank in sub:
                     st.dp_group = g
+
+    # Flash decoding (SURVEY §2.8 long-context mechanism 3): the ranks that
+    # hold the SAME replicated KV head (contiguous windows of size
+    # kv_shard_size inside each TP group) shard the decode KV cache along S
+    # and LSE-combine their partial attentions.
+    if kv_shard_size > 1:
+        for mesh in st.tp_mesh:
+            for i in range(tp // kv_shard_size):
+                sub = mesh[i * kv_shard_size : (i + 1) * kv_shard_size]
+                g = dist.new_group(sub)
+                if rank in sub:
+                    st.kv_shard_group = g
 
     # Speculative-draft group: draft model at smaller TP
     # (ref application_base.py:61-66, utils/distributed.py:30-36).
@@ -271,3 +287,17 @@ def init_distributed_from_env(backend: Optional[str] = None) -> int:
     else:
         dist.init_process_group(backend="gloo")
     return local_rank
+
+
+def get_kv_shard_group():
+    return _STATE.kv_shard_group
+
+
+def get_kv_shard_degree() -> int:
+    return _STATE.kv_shard_degree
+
+
+def get_kv_shard_rank() -> int:
+    if _STATE.kv_shard_group is None or not dist.is_initialized():
+        return 0
+    return dist.get_rank(group=_STATE.kv_shard_group)

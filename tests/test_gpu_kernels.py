@@ -526,3 +526,23 @@ def test_skinny_gemm_matches_torch(M):
     C.skinny_gemm(y, x, wp)
     want = torch.nn.functional.linear(x, w)
     torch.testing.assert_close(y.float(), want.float(), atol=2e-2, rtol=2e-2)
+
+
+def test_attention_decode_lse_matches_cpu():
+    """lse output of the decode kernel vs the fp32 logsumexp reference
+    (flash-decoding combine handle), incl. empty (-1) query rows."""
+    torch.manual_seed(0)
+    B, Hq, Hkv, Smax, D = 2, 4, 2, 96, 64
+    dev = _dev()
+    kc = torch.randn(B, Hkv, Smax, D, dtype=torch.bfloat16, device=dev)
+    vc = torch.randn(B, Hkv, Smax, D, dtype=torch.bfloat16, device=dev)
+    q = torch.randn(B, Hq, 1, D, dtype=torch.bfloat16, device=dev)
+    pos = torch.tensor([[57], [-1]], dtype=torch.int32, device=dev)
+    out, lse = ops.attention_decode_lse(q, kc, vc, pos)
+    want_o, want_l = ops.attention_decode_lse(
+        q.float().cpu(), kc.float().cpu(), vc.float().cpu(), pos.cpu())
+    torch.testing.assert_close(out.float().cpu(), want_o.float(),
+                               atol=3e-2, rtol=3e-2)
+    assert torch.isinf(lse[1]).all() and (lse[1] < 0).all()
+    torch.testing.assert_close(lse[0].cpu(), want_l[0], atol=2e-2, rtol=2e-2)
+    assert (out[1].float() == 0).all()

@@ -462,3 +462,110 @@ def _worker_attention_dp(rank):
 
 def test_attention_dp_generation():
     spawn("_worker_attention_dp")
+
+
+# ---------------------------------------------------------------------------
+def _worker_flash_decoding(rank):
+    """Flash decoding op-level: strided S-sharded cache + LSE combine must
+    match single-rank full-cache decode."""
+    import torch
+
+    from neuronx_distributed_inference_amd import ops
+    from neuronx_distributed_inference_amd.modules.attention import flash_decoding as fd
+    from neuronx_distributed_inference_amd.parallel import state as ps
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=2, kv_shard_size=2)
+    torch.manual_seed(7)
+    B, Hq, Hkv, Smax, D, Sq = 2, 4, 1, 64, 32, 1
+    fdg = 2
+    # identical full K/V on both ranks (replicated KV head)
+    k_full = torch.randn(B, Hkv, Smax, D)
+    v_full = torch.randn(B, Hkv, Smax, D)
+    # local strided shard caches
+    kc = torch.zeros(B, Hkv, Smax // fdg, D)
+    vc = torch.zeros(B, Hkv, Smax // fdg, D)
+    pos_all = torch.arange(Smax).view(1, Smax).expand(B, Smax).contiguous()
+    wpos = fd.shard_write_positions(pos_all, fdg, rank)
+    seq_ids = torch.arange(B, dtype=torch.int32)
+    ops.kv_cache_update(kc, vc, k_full, v_full, wpos, seq_ids)
+
+    # ragged current positions per row
+    positions = torch.tensor([[37], [52]])
+    # local q heads per rank: rank owns heads [rank*2, rank*2+2)
+    torch.manual_seed(11)
+    q_full = torch.randn(B, Hq, Sq, D)
+    q_local = q_full[:, rank * 2:(rank + 1) * 2].contiguous()
+    out = fd.flash_decoding_attention(
+        q_local, kc, vc, positions, seq_ids, scale=D ** -0.5)
+
+    want_full = ops.attention_decode(
+        q_full.float(), k_full.float(), v_full.float(), positions,
+        seq_ids=seq_ids, scale=D ** -0.5)
+    want = want_full[:, rank * 2:(rank + 1) * 2]
+    torch.testing.assert_close(out.float(), want.float(), atol=2e-3, rtol=2e-3)
+    ps.destroy_model_parallel()
+    return True
+
+
+def test_flash_decoding_sharded_kv():
+    spawn("_worker_flash_decoding")
+
+
+# ---------------------------------------------------------------------------
+def _worker_flash_decoding_e2e(rank):
+    """Full tiny-Llama generation with flash decoding (TP=2, KV head
+    replicated, decode cache S-sharded) matches the single-rank result."""
+    from neuronx_distributed_inference_amd.config import EngineConfig
+    from neuronx_distributed_inference_amd.models.llama.modeling_llama import (
+        LlamaForCausalLM,
+        LlamaInferenceConfig,
+    )
+    from neuronx_distributed_inference_amd.parallel import state as ps
+    from neuronx_distributed_inference_amd.utils.checkpoint_sharding import (
+        shard_state_dict_for_rank,
+    )
+    from neuronx_distributed_inference_amd.utils.hf_adapter import (
+        GenerationConfig,
+        HuggingFaceGenerationAdapter,
+    )
+
+    def make_cfg(tp, flash):
+        ec = EngineConfig(batch_size=2, max_batch_size=2, seq_len=64, max_length=64,
+                          torch_dtype="float32", tp_degree=tp,
+                          flash_decoding_enabled=flash)
+        return LlamaInferenceConfig(
+            neuron_config=ec, vocab_size=128, hidden_size=64, intermediate_size=128,
+            num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=1,
+            max_position_embeddings=128, rms_norm_eps=1e-5, rope_theta=10000.0,
+            hidden_act="silu", tie_word_embeddings=False)
+
+    ps.destroy_model_parallel()
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(23)
+    full = LlamaForCausalLM(make_cfg(1, False)).eval()
+    with torch.no_grad():
+        for p in full.parameters():
+            if p.dim() >= 2:
+                p.normal_(0, 0.1)
+    ids = torch.randint(0, 128, (2, 9))
+    want = HuggingFaceGenerationAdapter(full).generate(
+        ids, generation_config=GenerationConfig(max_new_tokens=8))
+
+    ps.destroy_model_parallel()
+    ps.initialize_model_parallel(tensor_model_parallel_size=WORLD,
+                                 kv_shard_size=WORLD)
+    tp_model = LlamaForCausalLM(make_cfg(WORLD, True)).eval()
+    # S-sharded cache is half length
+    assert tp_model.model.kv_mgr.max_len == 32
+    full_sd = {k: v for k, v in full.model.state_dict().items()
+               if not k.startswith("kv_mgr.")}
+    shard = shard_state_dict_for_rank(full_sd, tp_model.model, rank, WORLD)
+    tp_model.model.load_state_dict(shard, strict=False)
+    got = HuggingFaceGenerationAdapter(tp_model).generate(
+        ids, generation_config=GenerationConfig(max_new_tokens=8))
+    assert torch.equal(got, want), (got.tolist(), want.tolist())
+    return True
+
+
+def test_flash_decoding_e2e_generation():
+    spawn("_worker_flash_decoding_e2e")
