@@ -185,6 +185,13 @@ class DecoderModel(nn.Module):
         )
         h = self.embed_tokens(input_ids, sequence_parallel=sp_active) \
             if sp_active else self.embed_tokens(input_ids)
+        # multimodal splice (llava/pixtral pattern): projected vision
+        # features replace the image-placeholder token embeddings. The mask
+        # derives from input_ids, so bucket padding is handled for free.
+        mm = getattr(self, "multimodal_ctx", None)
+        if mm and mm.get("features") is not None and phase == PHASE_CTE and not sp_active:
+            mask = (input_ids == mm["token_id"]).unsqueeze(-1)
+            h = h.masked_scatter(mask, mm["features"].to(h.dtype).reshape(-1, h.shape[-1]))
         cos, sin = self.rotary.get_cos_sin(position_ids, device=h.device)
         # int32 once here: the KV/attention kernels take int32 and a per-layer
         # .to() would launch 4 conversion kernels per layer per step

@@ -383,3 +383,59 @@ def test_mllama():
         ref_ids = torch.cat([ref_ids, lg[:, -1].argmax(-1, keepdim=True)], dim=1)
     got = ours.generate(ids, max_new_tokens=4)
     assert torch.equal(got[:, : ref_ids.shape[1]], ref_ids)
+
+
+@torch.inference_mode()
+def test_pixtral():
+    """Pixtral/llava: 2D-rope block-diagonal vision encoder + inline image
+    splice into the Mistral text model; greedy vs stepwise HF."""
+    from transformers import LlavaConfig
+    from transformers import LlavaForConditionalGeneration as HF
+    from transformers.models.mistral import MistralConfig
+    from transformers.models.pixtral import PixtralVisionConfig
+
+    from neuronx_distributed_inference_amd.models.pixtral import (
+        PixtralForConditionalGeneration, PixtralInferenceConfig,
+    )
+
+    torch.manual_seed(0)
+    vc = PixtralVisionConfig(
+        hidden_size=32, intermediate_size=64, num_hidden_layers=2,
+        num_attention_heads=4, image_size=32, patch_size=8, num_channels=3,
+        rope_theta=10000.0)
+    tc = MistralConfig(
+        vocab_size=140, hidden_size=48, intermediate_size=96,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, rms_norm_eps=1e-5, sliding_window=None,
+        tie_word_embeddings=False)
+    hf_cfg = LlavaConfig(vision_config=vc.to_dict(), text_config=tc.to_dict(),
+                         image_token_id=139, projector_hidden_act="gelu",
+                         vision_feature_layer=-1,
+                         vision_feature_select_strategy="full")
+    hf = HF(hf_cfg).eval()
+
+    ec = EngineConfig(batch_size=1, max_batch_size=1, seq_len=64, max_length=64,
+                      torch_dtype="float32", tp_degree=1)
+    cfg = PixtralInferenceConfig(
+        neuron_config=ec, text_config=tc.to_dict(), vision_config=vc.to_dict(),
+        image_token_id=139, projector_hidden_act="gelu",
+        multimodal_projector_bias=True)
+    ours = PixtralForConditionalGeneration(cfg).eval()
+    sd = PixtralForConditionalGeneration.convert_hf_to_native_state_dict(
+        hf.state_dict(), cfg)
+    missing, unexpected = ours.load_state_dict(sd, strict=False)
+    real_missing = [m for m in missing if ".kv_mgr." not in m]
+    assert not real_missing, real_missing
+    assert not unexpected, unexpected
+
+    px = torch.randn(1, 3, 32, 32)           # 4x4 = 16 patches
+    ids = torch.randint(0, 139, (1, 22))
+    ids[0, 2:18] = 139                        # 16 image placeholders
+    n_new = 8
+    ref_ids = ids.clone()
+    for _ in range(n_new):
+        lg = hf(input_ids=ref_ids, pixel_values=px, use_cache=False).logits
+        ref_ids = torch.cat([ref_ids, lg[:, -1].argmax(-1, keepdim=True)], dim=1)
+    got = ours.generate(ids, pixel_values=px, max_new_tokens=n_new)
+    assert torch.equal(got[:, : ref_ids.shape[1]], ref_ids), (
+        got.tolist(), ref_ids.tolist())
