@@ -192,7 +192,9 @@ class DecoderModel(nn.Module):
         if mm and mm.get("features") is not None and phase == PHASE_CTE and not sp_active:
             mask = (input_ids == mm["token_id"]).unsqueeze(-1)
             h = h.masked_scatter(mask, mm["features"].to(h.dtype).reshape(-1, h.shape[-1]))
-        cos, sin = self.rotary.get_cos_sin(position_ids, device=h.device)
+        # models with non-1D rope (qwen2-vl m-rope) install a custom hook
+        cos_sin_fn = getattr(self, "_cos_sin_fn", None) or self.rotary.get_cos_sin
+        cos, sin = cos_sin_fn(position_ids, device=h.device)
         # int32 once here: the KV/attention kernels take int32 and a per-layer
         # .to() would launch 4 conversion kernels per layer per step
         position_ids = position_ids.to(torch.int32)
