@@ -201,7 +201,8 @@ class DecoderModel(nn.Module):
         if seq_ids is not None:
             seq_ids = seq_ids.to(torch.int32)
         residual = None
-        for layer in self.layers:
+        post_fns = getattr(self, "_layer_post_fns", None)
+        for li, layer in enumerate(self.layers):
             if sp_active:
                 h, residual = layer(
                     h, residual, cos, sin, self.kv_mgr, position_ids, phase,
@@ -212,6 +213,11 @@ class DecoderModel(nn.Module):
                     h, residual, cos, sin, self.kv_mgr, position_ids, phase,
                     attention_mask, seq_ids,
                 )
+            # per-layer post hook (qwen3-vl deepstack feature injection);
+            # adding to h is equivalent to adding to the logical hidden
+            # (h + deferred residual)
+            if post_fns and li in post_fns:
+                h = post_fns[li](h, input_ids, phase)
         if residual is not None:
             h, _ = self.norm(h, residual)
         else:

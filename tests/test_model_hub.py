@@ -505,3 +505,60 @@ def test_qwen2_vl():
         ref_ids = torch.cat([ref_ids, lg[:, -1].argmax(-1, keepdim=True)], dim=1)
     got = ours.generate(ids2, max_new_tokens=4)
     assert torch.equal(got[:, : ref_ids.shape[1]], ref_ids)
+
+
+@torch.inference_mode()
+def test_qwen3_vl():
+    """Qwen3-VL: DeepStack ViT (learned interpolated pos grid) + Qwen3 text
+    with interleaved m-rope; greedy vs stepwise HF."""
+    from transformers import Qwen3VLForConditionalGeneration as HF
+    from transformers.models.qwen3_vl import Qwen3VLConfig
+
+    from neuronx_distributed_inference_amd.models.qwen3_vl import (
+        Qwen3VLForConditionalGeneration, Qwen3VLInferenceConfig,
+    )
+
+    torch.manual_seed(0)
+    vc = dict(depth=3, hidden_size=32, intermediate_size=64, num_heads=4,
+              in_channels=3, patch_size=4, temporal_patch_size=2,
+              spatial_merge_size=2, out_hidden_size=48,
+              num_position_embeddings=16, deepstack_visual_indexes=[0, 1],
+              hidden_act="gelu_pytorch_tanh")
+    tc = dict(vocab_size=160, hidden_size=48, intermediate_size=96,
+              num_hidden_layers=3, num_attention_heads=4,
+              num_key_value_heads=2, head_dim=12, max_position_embeddings=128,
+              rms_norm_eps=1e-5, rope_theta=10000.0, tie_word_embeddings=False,
+              rope_scaling={"rope_type": "default", "mrope_section": [2, 2, 2]})
+    hf_cfg = Qwen3VLConfig(vision_config=vc, text_config=tc,
+                           image_token_id=159, vision_start_token_id=158,
+                           vision_end_token_id=157, video_token_id=156)
+    hf = HF(hf_cfg).eval()
+
+    ec = EngineConfig(batch_size=1, max_batch_size=1, seq_len=64, max_length=64,
+                      torch_dtype="float32", tp_degree=1)
+    cfg = Qwen3VLInferenceConfig(
+        neuron_config=ec, text_config=dict(tc), vision_config=dict(vc),
+        image_token_id=159)
+    ours = Qwen3VLForConditionalGeneration(cfg).eval()
+    sd = Qwen3VLForConditionalGeneration.convert_hf_to_native_state_dict(
+        hf.state_dict(), cfg)
+    missing, unexpected = ours.load_state_dict(sd, strict=False)
+    real_missing = [m for m in missing if ".kv_mgr." not in m]
+    assert not real_missing, real_missing
+    assert not unexpected, unexpected
+
+    grid = torch.tensor([[1, 4, 4]])
+    px = torch.randn(16, 3 * 2 * 4 * 4)
+    ids = torch.randint(0, 156, (1, 14))
+    ids[0, 3:7] = 159
+    n_new = 8
+    ref_ids = ids.clone()
+    for _ in range(n_new):
+        lg = hf(input_ids=ref_ids, pixel_values=px, image_grid_thw=grid,
+                mm_token_type_ids=(ref_ids == 159).int(),
+                use_cache=False).logits
+        ref_ids = torch.cat([ref_ids, lg[:, -1].argmax(-1, keepdim=True)], dim=1)
+    got = ours.generate(ids, pixel_values=px, image_grid_thw=grid,
+                        max_new_tokens=n_new)
+    assert torch.equal(got[:, : ref_ids.shape[1]], ref_ids), (
+        got.tolist(), ref_ids.tolist())
