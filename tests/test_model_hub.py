@@ -562,3 +562,60 @@ def test_qwen3_vl():
                         max_new_tokens=n_new)
     assert torch.equal(got[:, : ref_ids.shape[1]], ref_ids), (
         got.tolist(), ref_ids.tolist())
+
+
+@torch.inference_mode()
+def test_llama4_vision():
+    """Llama4 conditional generation: unfold-conv ViT + complex 2D rope +
+    pixel-shuffle adapter + inline splice; greedy vs stepwise HF."""
+    from transformers import Llama4Config
+    from transformers.models.llama4.modeling_llama4 import (
+        Llama4ForConditionalGeneration as HF,
+    )
+
+    from neuronx_distributed_inference_amd.models.llama4.modeling_llama4 import (
+        Llama4ForConditionalGeneration, Llama4VisionInferenceConfig,
+    )
+
+    torch.manual_seed(0)
+    # HF invariant: intermediate_size == hidden_size / pixel_shuffle_ratio^2
+    vc = dict(hidden_size=32, intermediate_size=128, num_hidden_layers=2,
+              num_attention_heads=4, image_size=16, patch_size=4,
+              num_channels=3, pixel_shuffle_ratio=0.5,
+              projector_input_dim=128, projector_output_dim=128,
+              vision_output_dim=128, rope_theta=10000)
+    tc = dict(vocab_size=150, hidden_size=48, intermediate_size=96,
+              intermediate_size_mlp=96, num_hidden_layers=2,
+              num_attention_heads=4, num_key_value_heads=2, head_dim=12,
+              max_position_embeddings=128, num_local_experts=2,
+              num_experts_per_tok=1, interleave_moe_layer_step=2,
+              no_rope_layer_interval=2, attention_chunk_size=16,
+              use_qk_norm=True, attn_temperature_tuning=True, floor_scale=16,
+              attn_scale=0.1, rms_norm_eps=1e-5, tie_word_embeddings=False)
+    hf_cfg = Llama4Config(vision_config=vc, text_config=tc, image_token_id=149)
+    hf = HF(hf_cfg).eval()
+
+    ec = EngineConfig(batch_size=1, max_batch_size=1, seq_len=64, max_length=64,
+                      torch_dtype="float32", tp_degree=1)
+    cfg = Llama4VisionInferenceConfig(
+        neuron_config=ec, text_config=dict(tc), vision_config=dict(vc),
+        image_token_id=149)
+    ours = Llama4ForConditionalGeneration(cfg).eval()
+    sd = Llama4ForConditionalGeneration.convert_hf_to_native_state_dict(
+        hf.state_dict(), cfg)
+    missing, unexpected = ours.load_state_dict(sd, strict=False)
+    real_missing = [m for m in missing if ".kv_mgr." not in m]
+    assert not real_missing, real_missing
+    assert not unexpected, unexpected
+
+    px = torch.randn(1, 3, 16, 16)   # 4x4 patches -> shuffle 0.5 -> 4 tokens
+    ids = torch.randint(0, 149, (1, 12))
+    ids[0, 2:6] = 149
+    n_new = 6
+    ref_ids = ids.clone()
+    for _ in range(n_new):
+        lg = hf(input_ids=ref_ids, pixel_values=px, use_cache=False).logits
+        ref_ids = torch.cat([ref_ids, lg[:, -1].argmax(-1, keepdim=True)], dim=1)
+    got = ours.generate(ids, pixel_values=px, max_new_tokens=n_new)
+    assert torch.equal(got[:, : ref_ids.shape[1]], ref_ids), (
+        got.tolist(), ref_ids.tolist())
