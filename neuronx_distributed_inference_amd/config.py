@@ -179,6 +179,9 @@ class EngineConfig:
             "kv_cache_batch_size", self.max_batch_size
         )
         self.kv_cache_padding_size: int = kwargs.pop("kv_cache_padding_size", 0)
+        # compact sliding-window KV: windowed layers allocate only `window`
+        # rows (ring buffer) instead of max_length (ref per-layer KV sizes)
+        self.compact_swa_kv: bool = kwargs.pop("compact_swa_kv", False)
         self.is_block_kv_layout: bool = kwargs.pop("is_block_kv_layout", False)
         self.pa_num_blocks: Optional[int] = kwargs.pop("pa_num_blocks", None)
         self.pa_block_size: int = kwargs.pop("pa_block_size", 128)

@@ -112,6 +112,18 @@ class DecoderModel(nn.Module):
             # kv-shard group (strided rows) — each rank holds 1/fd of S
             fd = ps.get_kv_shard_degree() if getattr(
                 ec, "flash_decoding_enabled", False) else 1
+            layer_sizes = None
+            if ec.compact_swa_kv and fd == 1:
+                sw = getattr(config, "sliding_window", None) or 0
+                lt = getattr(config, "layer_types", None)
+                if sw and sw < ec.max_length:
+                    if lt is not None:
+                        layer_sizes = [
+                            sw if t == "sliding_attention" else ec.max_length
+                            for t in lt
+                        ]
+                    else:
+                        layer_sizes = [sw] * self.num_layers
             self.kv_mgr = KVCacheManager(
                 num_layers=self.num_layers,
                 num_kv_heads=n_kv_heads,
@@ -121,6 +133,7 @@ class DecoderModel(nn.Module):
                 dtype=cache_dtype,
                 padding_rows=ec.kv_cache_padding_size,
                 tp_degree=cache_tp,
+                layer_cache_sizes=layer_sizes,
             )
         self.sampler = (
             Sampler(ec.on_device_sampling_config, config.vocab_size)

@@ -205,6 +205,8 @@ class AttentionBase(nn.Module):
 
                 wpos = _fdm.shard_write_positions(
                     position_ids, self.kv_shard_degree, ps.get_kv_shard_rank())
+            elif getattr(kv_mgr, "ring_layers", None) and kv_mgr.is_ring(self.layer_idx):
+                wpos = kv_mgr.wrap_positions(self.layer_idx, position_ids)
             q = _ops.qkv_rope_update(
                 y, cos, sin, kc.data if hasattr(kc, "data") else kc,
                 vc.data if hasattr(vc, "data") else vc,
@@ -236,6 +238,8 @@ class AttentionBase(nn.Module):
             and self.dp_degree == 1
             and not getattr(kv_mgr, "is_paged", False)
             and self.kv_shard_degree == 1
+            and not (getattr(kv_mgr, "ring_layers", None)
+                     and kv_mgr.is_ring(self.layer_idx))
             and (not hidden.is_cuda or self.head_dim in (64, 128))
         ):
             kc, vc = kv_mgr.get_cache(self.layer_idx)
@@ -441,7 +445,8 @@ class AttentionBase(nn.Module):
                     Sq, device=q.device).view(1, Sq))
             return _fdm.flash_decoding_attention(
                 q, k_cache, v_cache, pos, seq_ids, self.scale)
+        ring = bool(getattr(kv_mgr, "ring_layers", None)) and kv_mgr.is_ring(self.layer_idx)
         return ops.attention_decode(
             q, k_cache, v_cache, position_ids, seq_ids=seq_ids, scale=self.scale,
-            sinks=self.sinks, sliding_window=self.sliding_window,
+            sinks=self.sinks, sliding_window=self.sliding_window, ring=ring,
         )

@@ -59,6 +59,10 @@ class KVCacheManager(nn.Module):
             size = min(sliding_window, max_len) if sliding_window else max_len
             layer_cache_sizes = [size] * num_layers
         self.layer_cache_sizes = layer_cache_sizes
+        # ring layers: compact sliding-window cache (size < max_len) —
+        # positions wrap modulo the layer size (ref gpt_oss per-layer KV
+        # sizes)
+        self.ring_layers = [sz < max_len for sz in layer_cache_sizes]
 
         self.k_caches = nn.ParameterList()
         self.v_caches = nn.ParameterList()
@@ -96,7 +100,17 @@ class KVCacheManager(nn.Module):
         `seq_ids` maps batch lines to cache rows (continuous batching)."""
         kc = self.k_caches[layer_idx].data
         vc = self.v_caches[layer_idx].data
+        if self.ring_layers[layer_idx]:
+            positions = self.wrap_positions(layer_idx, positions)
         ops.kv_cache_update(kc, vc, k_new, v_new, positions, seq_ids)
+
+    def is_ring(self, layer_idx: int) -> bool:
+        return self.ring_layers[layer_idx]
+
+    def wrap_positions(self, layer_idx: int, positions: torch.Tensor) -> torch.Tensor:
+        sz = self.layer_cache_sizes[layer_idx]
+        return torch.where(positions >= 0, positions % sz,
+                           torch.full_like(positions, -1))
 
     def reset(self):
         for k in self.k_caches:

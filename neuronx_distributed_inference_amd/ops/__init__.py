@@ -232,7 +232,7 @@ def attention_prefill_cached(q, k_cache, v_cache, seq_ids, scale=None,
 
 
 def attention_decode(q, k_cache, v_cache, seq_lens, seq_ids=None, scale=None,
-                     sinks=None, sliding_window: int = 0):
+                     sinks=None, sliding_window: int = 0, ring: bool = False):
     """Token-generation attention against the linear KV cache.
 
     q: (B,Hq,Sq,D) with small Sq (1 or spec_len); k_cache/v_cache:
@@ -259,7 +259,7 @@ def attention_decode(q, k_cache, v_cache, seq_lens, seq_ids=None, scale=None,
             out, q.contiguous(), k_cache, v_cache,
             positions.to(torch.int32).contiguous(),
             seq_ids.to(torch.int32).contiguous(), float(scale), int(sliding_window),
-            sinks=None if sinks is None else sinks.detach(),
+            sinks=None if sinks is None else sinks.detach(), ring=int(ring),
         )
         return out
     # CPU reference: mask each query by its own position
@@ -270,6 +270,16 @@ def attention_decode(q, k_cache, v_cache, seq_lens, seq_ids=None, scale=None,
     Smax = k.shape[2]
     ar = torch.arange(Smax, device=q.device).view(1, 1, 1, Smax)
     qpos = positions.view(B, 1, Sq, 1)
+    if ring:
+        # compact sliding-window cache: slot s holds its newest position
+        # == s (mod Smax); mask by TRUE position per query
+        pmax = positions.amax(dim=-1).view(B, 1, 1, 1)
+        p_s = pmax - ((pmax - ar) % Smax)
+        mask = (p_s >= 0) & (p_s <= qpos)
+        if sliding_window > 0:
+            mask = mask & (p_s > (qpos - sliding_window))
+        out = ref.sdpa(q, k, v, mask=mask, scale=scale, sinks=sinks)
+        return torch.nan_to_num(out)
     mask = ar <= qpos
     if sliding_window > 0:
         mask = mask & (ar > (qpos - sliding_window))

@@ -38,7 +38,7 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
     const int* __restrict__ positions, // (B, Sq) per-query absolute position
     const int* __restrict__ seq_ids,   // (B,) (linear) / unused (paged)
     const int* __restrict__ block_table,  // (B, max_blocks) when PAGED
-    const float scale, const int window,
+    const float scale, const int window, const int ring,
     const int Hkv, const int Sq, const int gq,
     const int Smax, const int chunk, const int nsplit,
     const int block_size, const int max_blocks) {
@@ -61,12 +61,22 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
     // query at absolute position p attends cache rows [0, p]
     const int eff_len = positions[(long)b * Sq + sq] + 1;
     int lo = 0;
-    if (window > 0 && eff_len > window) lo = eff_len - window;
+    int span_hi = eff_len;
+    int pmax = 0;
+    if (ring) {
+        // RING cache (compact sliding-window layers): slot s holds the
+        // newest position == s (mod Smax); all written slots are scanned
+        // and masked by true position below
+        pmax = positions[(long)b * Sq + Sq - 1];
+        span_hi = min(pmax + 1, Smax);
+    } else if (window > 0 && eff_len > window) {
+        lo = eff_len - window;
+    }
     // window < 0: Llama4 CHUNKED attention with chunk size -window — the
     // query attends only rows in its own chunk [(p/C)*C, p]
     else if (window < 0) lo = ((eff_len - 1) / (-window)) * (-window);
     const int split_lo = max(lo, split * chunk);
-    const int split_hi = min(eff_len, (split + 1) * chunk);
+    const int split_hi = min(span_hi, (split + 1) * chunk);
 
     // ---- q into registers, pre-scaled ----
     float q_reg[GQ][8];
@@ -118,6 +128,12 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
             kv = load_vec8(kbase + off);
             vv = load_vec8(vbase + off);
             if (PAGED && !valid_cur) continue;
+            if (ring) {
+                const int p_s = pmax - ((pmax - l) % Smax);  // slot's true pos
+                if (p_s >= eff_len ||
+                    (window > 0 && p_s <= eff_len - 1 - window))
+                    continue;
+            }
             float kf[8];
 #pragma unroll
             for (int j = 0; j < 8; ++j) kf[j] = VecTraits<CT>::to_float(kv_cur.v[j]);
@@ -264,7 +280,8 @@ static void launch_decode(torch::Tensor& out, const torch::Tensor& q,
                           const torch::Tensor* block_table = nullptr,
                           int block_size = 0, float v_scale = 1.0f,
                           const torch::Tensor* sinks = nullptr,
-                          torch::Tensor* lse_out = nullptr) {
+                          torch::Tensor* lse_out = nullptr,
+                          int ring = 0) {
     const int B = q.size(0), Hq = q.size(1), Sq = q.size(2);
     const bool paged = block_table != nullptr;
     const int Hkv = paged ? k_cache.size(2) : k_cache.size(1);
@@ -288,7 +305,7 @@ static void launch_decode(torch::Tensor& out, const torch::Tensor& q,
                            (const T*)q.data_ptr(), (const CT*)k_cache.data_ptr(),
                            (const CT*)v_cache.data_ptr(), seq_lens.data_ptr<int>(),
                            seq_ids.data_ptr<int>(), block_table->data_ptr<int>(),
-                           scale, window, Hkv, Sq, gq, Smax, chunk, nsplit,
+                           scale, window, 0, Hkv, Sq, gq, Smax, chunk, nsplit,
                            block_size, max_blocks);
     } else {
         hipLaunchKernelGGL((decode_attn_pass1<T, CT, D, GQ, false>), grid1, dim3(256), 0, stream,
@@ -296,7 +313,7 @@ static void launch_decode(torch::Tensor& out, const torch::Tensor& q,
                            (const T*)q.data_ptr(), (const CT*)k_cache.data_ptr(),
                            (const CT*)v_cache.data_ptr(), seq_lens.data_ptr<int>(),
                            seq_ids.data_ptr<int>(), nullptr,
-                           scale, window, Hkv, Sq, gq, Smax, chunk, nsplit, 0, 0);
+                           scale, window, ring, Hkv, Sq, gq, Smax, chunk, nsplit, 0, 0);
     }
     HIP_CHECK_KERNEL();
     dim3 grid2(B * Hq * Sq);
@@ -315,26 +332,26 @@ static void dispatch_gq(torch::Tensor& out, const torch::Tensor& q,
                         float scale, int window, int gq,
                         const torch::Tensor* bt = nullptr, int bs = 0,
                         float v_scale = 1.0f, const torch::Tensor* sinks = nullptr,
-                        torch::Tensor* lse_out = nullptr) {
+                        torch::Tensor* lse_out = nullptr, int ring = 0) {
     const bool fp8 = k_cache.scalar_type() == at::ScalarType::Float8_e4m3fn;
 #define GQ_CASE(CT)                                                                     \
     do {                                                                                \
         if (gq <= 1)                                                                    \
             launch_decode<T, CT, D, 1>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
                                        scale, window, gq, bt, bs, v_scale, sinks,       \
-                                       lse_out);                                        \
+                                       lse_out, ring);                                  \
         else if (gq <= 2)                                                               \
             launch_decode<T, CT, D, 2>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
                                        scale, window, gq, bt, bs, v_scale, sinks,       \
-                                       lse_out);                                        \
+                                       lse_out, ring);                                  \
         else if (gq <= 4)                                                               \
             launch_decode<T, CT, D, 4>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
                                        scale, window, gq, bt, bs, v_scale, sinks,       \
-                                       lse_out);                                        \
+                                       lse_out, ring);                                  \
         else if (gq <= 8)                                                               \
             launch_decode<T, CT, D, 8>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
                                        scale, window, gq, bt, bs, v_scale, sinks,       \
-                                       lse_out);                                        \
+                                       lse_out, ring);                                  \
         else                                                                            \
             TORCH_CHECK(false, "GQA ratio > 8 not supported yet (got ", gq, ")");       \
     } while (0)
@@ -391,7 +408,8 @@ void attention_decode(torch::Tensor out, torch::Tensor q,
                       torch::Tensor seq_lens, torch::Tensor seq_ids,
                       double scale, long window,
                       c10::optional<torch::Tensor> sinks_opt,
-                      c10::optional<torch::Tensor> lse_opt) {
+                      c10::optional<torch::Tensor> lse_opt,
+                      long ring) {
     // lse_opt: fp32 (B, Hq, Sq), filled with per-query log-sum-exp — the
     // cross-rank flash-decoding combine handle
     // `seq_lens` carries per-query POSITIONS, shape (B, Sq)
@@ -425,19 +443,19 @@ void attention_decode(torch::Tensor out, torch::Tensor q,
         if (D == 128)                                                           \
             dispatch_gq<T, 128>(out, q, k_cache, v_cache, seq_lens, seq_ids,    \
                                 (float)scale, (int)window, gq, nullptr, 0,      \
-                                1.0f, sinks, lse_ptr);                          \
+                                1.0f, sinks, lse_ptr, (int)ring);               \
         else if (D == 64)                                                       \
             dispatch_gq<T, 64>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
                                (float)scale, (int)window, gq, nullptr, 0,       \
-                               1.0f, sinks, lse_ptr);                           \
+                               1.0f, sinks, lse_ptr, (int)ring);                \
         else if (D == 32)                                                       \
             dispatch_gq<T, 32>(out, q, k_cache, v_cache, seq_lens, seq_ids,     \
                                (float)scale, (int)window, gq, nullptr, 0,       \
-                               1.0f, sinks, lse_ptr);                           \
+                               1.0f, sinks, lse_ptr, (int)ring);                \
         else if (D == 256)                                                      \
             dispatch_gq<T, 256>(out, q, k_cache, v_cache, seq_lens, seq_ids,    \
                                 (float)scale, (int)window, gq, nullptr, 0,      \
-                                1.0f, sinks, lse_ptr);                          \
+                                1.0f, sinks, lse_ptr, (int)ring);               \
         else                                                                    \
             TORCH_CHECK(false, "head_dim must be 32/64/128/256, got ", D);      \
     } while (0)

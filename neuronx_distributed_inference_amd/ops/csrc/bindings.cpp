@@ -14,7 +14,7 @@ void block_kv_cache_update(torch::Tensor k_cache, torch::Tensor v_cache, torch::
 void attention_decode(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
                       torch::Tensor v_cache, torch::Tensor seq_lens, torch::Tensor seq_ids,
                       double scale, long window, c10::optional<torch::Tensor> sinks,
-                      c10::optional<torch::Tensor> lse);
+                      c10::optional<torch::Tensor> lse, long ring);
 void attention_decode_paged(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
                             torch::Tensor v_cache, torch::Tensor seq_lens,
                             torch::Tensor block_table, double scale, long window);
@@ -52,7 +52,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("attention_decode", &attention_decode, "split-KV decode attention (CDNA4)",
           py::arg("out"), py::arg("q"), py::arg("k_cache"), py::arg("v_cache"),
           py::arg("seq_lens"), py::arg("seq_ids"), py::arg("scale"), py::arg("window"),
-          py::arg("sinks") = py::none(), py::arg("lse") = py::none());
+          py::arg("sinks") = py::none(), py::arg("lse") = py::none(),
+          py::arg("ring") = 0);
     m.def("attention_decode_paged", &attention_decode_paged, "paged split-KV decode attention (CDNA4)");
     m.def("topk_topp_sample", &topk_topp_sample, "fused top-k/top-p sampling (CDNA4)");
     m.def("flash_prefill", &flash_prefill, "MFMA causal flash prefill (CDNA4)",

@@ -546,3 +546,24 @@ def test_attention_decode_lse_matches_cpu():
     assert torch.isinf(lse[1]).all() and (lse[1] < 0).all()
     torch.testing.assert_close(lse[0].cpu(), want_l[0], atol=2e-2, rtol=2e-2)
     assert (out[1].float() == 0).all()
+
+
+def test_ring_cache_decode_kernel():
+    """Ring (compact sliding-window) decode vs the CPU true-position
+    reference, incl. multi-query (speculation) shapes."""
+    torch.manual_seed(0)
+    B, Hq, Hkv, W, D = 2, 4, 2, 32, 64
+    dev = _dev()
+    win = W
+    kc = torch.randn(B, Hkv, W, D, dtype=torch.bfloat16, device=dev)
+    vc = torch.randn(B, Hkv, W, D, dtype=torch.bfloat16, device=dev)
+    for Sq, plen in ((1, 20), (1, 57), (3, 45)):
+        q = torch.randn(B, Hq, Sq, D, dtype=torch.bfloat16, device=dev)
+        pos = (torch.tensor([plen, plen - 5]).view(B, 1)
+               + torch.arange(Sq).view(1, Sq)).to(dev, torch.int32)
+        out = ops.attention_decode(q, kc, vc, pos, sliding_window=win, ring=True)
+        want = ops.attention_decode(q.float().cpu(), kc.float().cpu(),
+                                    vc.float().cpu(), pos.cpu(),
+                                    sliding_window=win, ring=True)
+        torch.testing.assert_close(out.float().cpu(), want.float(),
+                                   atol=3e-2, rtol=3e-2, msg=f"Sq={Sq} p={plen}")

@@ -191,3 +191,50 @@ def test_sampler_multinomial_topk():
     out = s(logits, params)
     top4 = logits.topk(4, -1).indices
     assert all(out[i] in top4[i] for i in range(4))
+
+
+@torch.inference_mode()
+def test_compact_swa_ring_cache_matches_full():
+    """compact_swa_kv: mistral with a tiny sliding window generates the
+    same tokens with a ring cache (window rows) as with the full cache —
+    incl. prompts and generations well beyond the window."""
+    import copy
+
+    from neuronx_distributed_inference_amd.config import EngineConfig
+    from neuronx_distributed_inference_amd.models.mistral.modeling_mistral import (
+        MistralForCausalLM,
+        MistralInferenceConfig,
+    )
+    from neuronx_distributed_inference_amd.utils.hf_adapter import (
+        GenerationConfig,
+        HuggingFaceGenerationAdapter,
+    )
+
+    def make(compact):
+        ec = EngineConfig(batch_size=2, max_batch_size=2, seq_len=64,
+                          max_length=64, torch_dtype="float32", tp_degree=1,
+                          compact_swa_kv=compact)
+        return MistralInferenceConfig(
+            neuron_config=ec, vocab_size=128, hidden_size=64,
+            intermediate_size=128, num_hidden_layers=2, num_attention_heads=4,
+            num_key_value_heads=2, max_position_embeddings=128,
+            rms_norm_eps=1e-5, sliding_window=8, tie_word_embeddings=False)
+
+    torch.manual_seed(3)
+    full = MistralForCausalLM(make(False)).eval()
+    with torch.no_grad():
+        for p in full.parameters():
+            if p.dim() >= 2:
+                p.normal_(0, 0.1)
+    ring = MistralForCausalLM(make(True)).eval()
+    sd = {k: v for k, v in full.model.state_dict().items()
+          if not k.startswith("kv_mgr.")}
+    ring.model.load_state_dict(sd, strict=False)
+    assert ring.model.kv_mgr.layer_cache_sizes == [8, 8]
+    assert all(ring.model.kv_mgr.ring_layers)
+
+    ids = torch.randint(0, 128, (2, 20))   # prompt longer than the window
+    gc = GenerationConfig(max_new_tokens=12)
+    want = HuggingFaceGenerationAdapter(full).generate(ids, generation_config=gc)
+    got = HuggingFaceGenerationAdapter(ring).generate(ids, generation_config=gc)
+    assert torch.equal(got, want), (got.tolist(), want.tolist())
