@@ -167,16 +167,34 @@ class DeepseekV3Attention(nn.Module):
                 mask = causal_mask(Sq, Sq, device=h.device)
             out = ref_ops.sdpa(q_states, k, v, mask=mask, scale=self.scaling)
         else:
+            # ABSORBED decode (ref modeling_deepseek.py:142-266 wkv_b trick):
+            # fold W_uk into q so attention runs directly over the latent
+            # cache — score = (W_uk^T q_nope)·c + q_rope·k_rope, out =
+            # W_uv (p·c). No per-step K/V expansion, no host sync; pure
+            # batched library GEMMs over (rank+rope)-dim rows, graph-safe.
             kc, _ = kv_mgr.get_cache(self.layer_idx)
-            L = int(position_ids.max()) + 1
-            lat = kc[:B, :, :L, :].to(h.dtype)              # (B,1,L,rank+rope)
-            kv_n, k_r = torch.split(lat, [self.kv_lora_rank, self.rope_dim], dim=-1)
-            k, v = self._expand(kv_n, k_r)
-            ar = torch.arange(L, device=h.device).view(1, 1, 1, L)
-            qpos = position_ids.view(B, 1, S, 1)   # per-query positions
-            mask = ar <= qpos
-            out = ref_ops.sdpa(q_states, k, v, mask=mask, scale=self.scaling)
-            out = torch.nan_to_num(out)
+            if seq_ids is not None:
+                lat = kc[seq_ids.long()].to(h.dtype)
+            else:
+                lat = kc[:B].to(h.dtype)                    # (B,1,Smax,D)
+            lat = lat.squeeze(1)                            # (B,Smax,D)
+            w = self.kv_b_proj.weight.view(
+                self.heads_per_rank, self.nope_dim + self.v_dim, self.kv_lora_rank)
+            w_uk = w[:, : self.nope_dim]                    # (hpr, nope, rank)
+            w_uv = w[:, self.nope_dim:]                     # (hpr, v, rank)
+            q_lat = torch.einsum("bhsd,hdr->bhsr", q_pass, w_uk)
+            q_abs = torch.cat([q_lat, q_rot], dim=-1)       # (B,hpr,S,rank+rope)
+            scores = torch.einsum("bhsr,btr->bhst", q_abs.float(), lat.float())
+            Smax = lat.shape[1]
+            ar = torch.arange(Smax, device=h.device).view(1, 1, 1, Smax)
+            qpos = position_ids.view(B, 1, S, 1)            # per-query positions
+            scores = (scores * self.scaling).masked_fill(ar > qpos, float("-inf"))
+            p = torch.softmax(scores, dim=-1)
+            p = torch.nan_to_num(p)                          # pad queries
+            o_lat = torch.einsum("bhst,btr->bhsr", p,
+                                 lat[..., : self.kv_lora_rank].float())
+            out = torch.einsum("bhsr,hvr->bhsv", o_lat,
+                               w_uv.float()).to(h.dtype)
 
         out = out.transpose(1, 2).reshape(B, S, -1)
         return self.o_proj(out)
