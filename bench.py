@@ -32,6 +32,11 @@ MODELS = {
                num_attention_heads=32, num_key_value_heads=8, vocab_size=128256),
     "70b": dict(hidden_size=8192, intermediate_size=28672, num_hidden_layers=80,
                 num_attention_heads=64, num_key_value_heads=8, vocab_size=128256),
+    # Mixtral-8x7B geometry (MoE): dense-bmm expert decode path
+    "mixtral": dict(hidden_size=4096, intermediate_size=14336,
+                    num_hidden_layers=32, num_attention_heads=32,
+                    num_key_value_heads=8, vocab_size=32000,
+                    num_local_experts=8, num_experts_per_tok=2),
 }
 
 
@@ -187,6 +192,10 @@ def main():
         LlamaForCausalLM,
         LlamaInferenceConfig,
     )
+    from neuronx_distributed_inference_amd.models.mixtral.modeling_mixtral import (
+        MixtralForCausalLM,
+        MixtralInferenceConfig,
+    )
     from neuronx_distributed_inference_amd.parallel import state as ps
 
     tp = world
@@ -206,7 +215,11 @@ def main():
         enable_hip_graphs=not args.no_graphs,
         fused_qkv=True, fused_gate_up=True,
     )
-    cfg = LlamaInferenceConfig(
+    model_cls, cfg_cls = (
+        (MixtralForCausalLM, MixtralInferenceConfig)
+        if args.model == "mixtral" else (LlamaForCausalLM, LlamaInferenceConfig)
+    )
+    cfg = cfg_cls(
         neuron_config=ec, max_position_embeddings=total_len * 2,
         rms_norm_eps=1e-5, rope_theta=500000.0, hidden_act="silu",
         tie_word_embeddings=False, **spec,
@@ -217,9 +230,9 @@ def main():
         # construct directly on the GPU: a 70B bf16 model is 140 GB and must
         # never round-trip host RAM / PCIe
         with torch.device(device):
-            model = LlamaForCausalLM(cfg).eval()
+            model = model_cls(cfg).eval()
     else:
-        model = LlamaForCausalLM(cfg).eval()
+        model = model_cls(cfg).eval()
     with torch.no_grad():
         for p in model.parameters():
             if p.dim() >= 2:
@@ -297,7 +310,9 @@ def main():
             "dtype": ("fp8-weights+bf16" if args.fp8 else dtype),
             "data": "synthetic",
             "config": {
-                "model": {"8b": "llama-3.1-8b", "70b": "llama-3.1-70b", "tiny": "tiny-llama-2l"}[args.model],
+                "model": {"8b": "llama-3.1-8b", "70b": "llama-3.1-70b",
+                          "tiny": "tiny-llama-2l", "1b": "llama-3.2-1b",
+                          "mixtral": "mixtral-8x7b"}[args.model],
                 "global_batch": global_batch,
                 "seq_len": total_len,
                 "ctx": args.ctx,

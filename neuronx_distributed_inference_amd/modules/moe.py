@@ -128,10 +128,36 @@ class ExpertMLPs(nn.Module):
         moe_tp = ps.get_tensor_model_parallel_size() // self.ep_degree
         return (tp_rank // max(moe_tp, 1)) * self.num_local_experts
 
+    def _forward_dense(self, x: torch.Tensor, weights: torch.Tensor,
+                       idx: torch.Tensor) -> torch.Tensor:
+        T, H = x.shape
+        El = self.num_local_experts
+        gates = torch.zeros(T, El, dtype=x.dtype, device=x.device)
+        gates = gates.scatter(1, idx, weights.to(x.dtype))
+        gte = gates.t().unsqueeze(-1)                      # (E, T, 1)
+        if self.scale_input:
+            xe = x.unsqueeze(0) * gte                      # Llama4: gate input
+        else:
+            xe = x.unsqueeze(0).expand(El, T, H)
+        g = torch.bmm(xe, self.w_gate.transpose(1, 2))
+        u = torch.bmm(xe, self.w_up.transpose(1, 2))
+        y = torch.bmm(self.act_mul(g, u), self.w_down.transpose(1, 2))
+        if not self.scale_input:
+            y = y * gte
+        return y.sum(dim=0)
+
     def forward(self, x: torch.Tensor, weights: torch.Tensor, idx: torch.Tensor):
         """x: (T, H); weights/idx: (T, top_k) over GLOBAL expert ids.
         Returns (T, H) partial (summed by the caller's TP all-reduce)."""
         T, H = x.shape
+        # decode-sized batches: DENSE batched-GEMM path. Every expert's
+        # weights must stream from HBM for any realistic decode batch
+        # anyway, so the dense bmm has the SAME memory traffic as the
+        # gather loop while being free of host syncs and dynamic shapes —
+        # i.e. hipGraph-capturable (the reference gets this from the
+        # traced NEFF; SURVEY K12 grouped-GEMM analog).
+        if self.ep_degree == 1 and T <= 512:
+            return self._forward_dense(x, weights, idx)
         out = torch.zeros(T, H, dtype=x.dtype, device=x.device)
         offset = self._local_offset()
         for e_local in range(self.num_local_experts):

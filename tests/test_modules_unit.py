@@ -238,3 +238,37 @@ def test_compact_swa_ring_cache_matches_full():
     want = HuggingFaceGenerationAdapter(full).generate(ids, generation_config=gc)
     got = HuggingFaceGenerationAdapter(ring).generate(ids, generation_config=gc)
     assert torch.equal(got, want), (got.tolist(), want.tolist())
+
+
+@torch.inference_mode()
+def test_moe_dense_matches_gather_path():
+    """Dense bmm expert path (graph-safe decode) vs the gather loop."""
+    from neuronx_distributed_inference_amd.modules.moe import ExpertMLPs
+
+    torch.manual_seed(0)
+    for scale_input in (False, True):
+        m = ExpertMLPs(4, 16, 32, tp_degree=1, dtype=torch.float32,
+                       scale_input=scale_input)
+        for p in m.parameters():
+            p.normal_(0, 0.2)
+        x = torch.randn(6, 16)
+        w = torch.rand(6, 2)
+        idx = torch.stack([torch.randperm(4)[:2] for _ in range(6)])
+        dense = m._forward_dense(x, w, idx)
+        T = x.shape[0]
+        out = torch.zeros(T, 16)
+        for e in range(4):
+            sel = (idx == e)
+            rows = sel.any(dim=-1)
+            if not rows.any():
+                continue
+            xe = x[rows]
+            ww = (w * sel).sum(dim=-1)[rows].unsqueeze(-1)
+            if scale_input:
+                xe = xe * ww
+            import torch.nn.functional as F
+            ge = F.linear(xe, m.w_gate[e])
+            ue = F.linear(xe, m.w_up[e])
+            he = F.linear(m.act_mul(ge, ue), m.w_down[e])
+            out[rows] += he if scale_input else he * ww
+        torch.testing.assert_close(dense, out, atol=1e-5, rtol=1e-5)
