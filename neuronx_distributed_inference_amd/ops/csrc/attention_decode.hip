@@ -28,6 +28,37 @@ constexpr float M_INIT = -1e30f;
 // CT = cache storage type (T itself, or fp8-e4m3 when the KV cache is
 // quantized; the k-scale is folded into the softmax scale by the host and
 // the v-scale applied in pass 2).
+// 2-way packed dot (v_dot2): halves the QK VALU work and the q-register
+// footprint — the GQ=8 (70B TP1) case was occupancy-bound at 178 VGPRs.
+typedef __bf16 dot2_bf16x2 __attribute__((ext_vector_type(2)));
+typedef _Float16 dot2_f16x2 __attribute__((ext_vector_type(2)));
+template <typename T> struct Dot2 {
+    static constexpr bool ok = false;
+    using v2 = float;
+    static __device__ __forceinline__ float dot(v2, v2, float c) { return c; }
+    static __device__ __forceinline__ v2 pack(float, float) { return 0.f; }
+};
+template <> struct Dot2<__hip_bfloat16> {
+    static constexpr bool ok = true;
+    using v2 = dot2_bf16x2;
+    static __device__ __forceinline__ float dot(v2 a, v2 b, float c) {
+        return __builtin_amdgcn_fdot2_f32_bf16(a, b, c, false);
+    }
+    static __device__ __forceinline__ v2 pack(float a, float b) {
+        v2 r; r[0] = (__bf16)a; r[1] = (__bf16)b; return r;
+    }
+};
+template <> struct Dot2<__half> {
+    static constexpr bool ok = true;
+    using v2 = dot2_f16x2;
+    static __device__ __forceinline__ float dot(v2 a, v2 b, float c) {
+        return __builtin_amdgcn_fdot2(a, b, c, false);
+    }
+    static __device__ __forceinline__ v2 pack(float a, float b) {
+        v2 r; r[0] = (_Float16)a; r[1] = (_Float16)b; return r;
+    }
+};
+
 template <typename T, typename CT, int D, int GQ, bool PAGED>
 __global__ __launch_bounds__(256) void decode_attn_pass1(
     float* __restrict__ o_part,        // (B, Hq, Sq, nsplit, D)
@@ -79,14 +110,25 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
     const int split_hi = min(span_hi, (split + 1) * chunk);
 
     // ---- q into registers, pre-scaled ----
-    float q_reg[GQ][8];
+    constexpr bool PK = Dot2<T>::ok && std::is_same<T, CT>::value;
+    using D2 = Dot2<T>;
+    typename D2::v2 q_pk[PK ? GQ : 1][4];
+    float q_reg[PK ? 1 : GQ][8];
 #pragma unroll
     for (int g = 0; g < GQ; ++g) {
         if (g < gq) {
             const long qoff = ((((long)b * Hq) + hkv * gq + g) * Sq + sq) * D + c * 8;
             Vec8<T> qv = load_vec8(q + qoff);
+            if constexpr (PK) {
 #pragma unroll
-            for (int j = 0; j < 8; ++j) q_reg[g][j] = VT::to_float(qv.v[j]) * scale;
+                for (int j = 0; j < 4; ++j)
+                    q_pk[g][j] = D2::pack(VT::to_float(qv.v[2 * j]) * scale,
+                                          VT::to_float(qv.v[2 * j + 1]) * scale);
+            } else {
+#pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    q_reg[g][j] = VT::to_float(qv.v[j]) * scale;
+            }
         }
     }
 
@@ -134,19 +176,33 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
                     (window > 0 && p_s <= eff_len - 1 - window))
                     continue;
             }
-            float kf[8];
-#pragma unroll
-            for (int j = 0; j < 8; ++j) kf[j] = VecTraits<CT>::to_float(kv_cur.v[j]);
             const Vec8<CT>& vv_use = vv_cur;
 
             float s[GQ];
+            if constexpr (PK) {
+                const typename D2::v2* kp =
+                    reinterpret_cast<const typename D2::v2*>(kv_cur.v);
 #pragma unroll
-            for (int g = 0; g < GQ; ++g) {
-                if (g >= gq) continue;
-                float d = 0.f;
+                for (int g = 0; g < GQ; ++g) {
+                    if (g >= gq) continue;
+                    float d = 0.f;
 #pragma unroll
-                for (int j = 0; j < 8; ++j) d += q_reg[g][j] * kf[j];
-                s[g] = group_reduce_sum<GS>(d);  // all GS lanes get the dot
+                    for (int j = 0; j < 4; ++j) d = D2::dot(q_pk[g][j], kp[j], d);
+                    s[g] = group_reduce_sum<GS>(d);
+                }
+            } else {
+                float kf[8];
+#pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    kf[j] = VecTraits<CT>::to_float(kv_cur.v[j]);
+#pragma unroll
+                for (int g = 0; g < GQ; ++g) {
+                    if (g >= gq) continue;
+                    float d = 0.f;
+#pragma unroll
+                    for (int j = 0; j < 8; ++j) d += q_reg[g][j] * kf[j];
+                    s[g] = group_reduce_sum<GS>(d);  // all GS lanes get the dot
+                }
             }
 #pragma unroll
             for (int g = 0; g < GQ; ++g) {
