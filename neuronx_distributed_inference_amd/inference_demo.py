@@ -202,6 +202,11 @@ def run_inference(args, model_type: str) -> int:
 
     torch.manual_seed(args.seed)
     config_cls, model_cls = get_model_classes(model_type)
+    if not hasattr(model_cls, "_model_cls"):
+        print(f"model type {model_type!r} is a conditional-generation family "
+              "(speech / vision); drive it programmatically via "
+              f"{model_cls.__module__} — the CLI run path covers causal LMs.")
+        return 2
     local_rank = ps.init_distributed_from_env()
     ps.initialize_model_parallel(
         tensor_model_parallel_size=args.tp_degree,

@@ -66,6 +66,38 @@ _REGISTRY: Dict[str, Tuple[str, str, str]] = {
         "Qwen3MoeInferenceConfig",
         "Qwen3MoeForCausalLM",
     ),
+    # conditional-generation families (multimodal / speech): resolved for
+    # programmatic use; the CLI's causal-lm run path does not drive these
+    "whisper": (
+        "neuronx_distributed_inference_amd.models.whisper.modeling_whisper",
+        "WhisperInferenceConfig",
+        "WhisperForConditionalGeneration",
+    ),
+    "mllama": (
+        "neuronx_distributed_inference_amd.models.mllama.modeling_mllama",
+        "MllamaInferenceConfig",
+        "MllamaForConditionalGeneration",
+    ),
+    "pixtral": (
+        "neuronx_distributed_inference_amd.models.pixtral.modeling_pixtral",
+        "PixtralInferenceConfig",
+        "PixtralForConditionalGeneration",
+    ),
+    "qwen2_vl": (
+        "neuronx_distributed_inference_amd.models.qwen2_vl.modeling_qwen2_vl",
+        "Qwen2VLInferenceConfig",
+        "Qwen2VLForConditionalGeneration",
+    ),
+    "qwen3_vl": (
+        "neuronx_distributed_inference_amd.models.qwen3_vl.modeling_qwen3_vl",
+        "Qwen3VLInferenceConfig",
+        "Qwen3VLForConditionalGeneration",
+    ),
+    "llama4_vision": (
+        "neuronx_distributed_inference_amd.models.llama4.modeling_llama4",
+        "Llama4VisionInferenceConfig",
+        "Llama4ForConditionalGeneration",
+    ),
 }
 
 
