@@ -149,7 +149,6 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
         // unconditional prefetch with clamped addresses — a conditional
         // ("load or keep") makes hipcc branch around the loads and wait
         // vmcnt(0) per iteration (guide §5 trap 4(c))
-        bool valid_next;
         auto row_addr = [&](int l, bool& valid) -> long {
             if constexpr (PAGED) {
                 const int blk = block_table[(long)b * max_blocks + l / block_size];
@@ -159,13 +158,24 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
             valid = true;
             return (long)l * D + c * 8;
         };
-        long off = row_addr(split_lo + grp, valid_next);
+        // per-group CONTIGUOUS row runs: a group streams rows
+        // [g_lo, g_hi) back to back, so each group is one sequential HBM
+        // stream (the strided l += NG layout read 256B chunks at 4 KB
+        // stride — DRAM-page/channel hostile, same pathology measured on
+        // the skinny GEMM). Online softmax is order-free, so the group->row
+        // assignment is irrelevant to the math.
+        const int span = split_hi - split_lo;
+        const int rpg = (span + NG - 1) / NG;
+        const int g_lo = split_lo + grp * rpg;
+        const int g_hi = min(g_lo + rpg, split_hi);
+        bool valid_next;
+        long off = row_addr(g_lo, valid_next);
         Vec8<CT> kv = load_vec8(kbase + off);
         Vec8<CT> vv = load_vec8(vbase + off);
-        for (int l = split_lo + grp; l < split_hi; l += NG) {
+        for (int l = g_lo; l < g_hi; ++l) {
             const bool valid_cur = valid_next;
             Vec8<CT> kv_cur = kv, vv_cur = vv;
-            const int lnext = (l + NG < split_hi) ? l + NG : l;
+            const int lnext = (l + 1 < g_hi) ? l + 1 : l;
             off = row_addr(lnext, valid_next);
             kv = load_vec8(kbase + off);
             vv = load_vec8(vbase + off);
