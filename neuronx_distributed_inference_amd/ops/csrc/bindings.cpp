@@ -33,6 +33,8 @@ void qkv_rope_update(torch::Tensor q_out, torch::Tensor qkv, torch::Tensor cosb,
 void silu_mul_fused(torch::Tensor out, torch::Tensor gate_up);
 void skinny_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                  c10::optional<torch::Tensor> bias);
+void skinny_gemm2(torch::Tensor y, torch::Tensor y32, torch::Tensor x,
+                  torch::Tensor w, c10::optional<torch::Tensor> bias, long ksplit);
 bool lt_available();
 void lt_linear(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                c10::optional<torch::Tensor> bias);
@@ -67,6 +69,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("qkv_rope_update", &qkv_rope_update, "fused qkv->rope->cache-write (CDNA4)");
     m.def("skinny_gemm", &skinny_gemm, "skinny-M weight-streaming GEMM (CDNA4)",
           py::arg("y"), py::arg("x"), py::arg("w"), py::arg("bias") = py::none());
+    m.def("skinny_gemm2", &skinny_gemm2, "blocked skinny-M GEMM, K-split (CDNA4)",
+          py::arg("y"), py::arg("y32"), py::arg("x"), py::arg("w"),
+          py::arg("bias") = py::none(), py::arg("ksplit") = 1);
     m.def("lt_available", &lt_available, "hipblaslt usable on this device");
     m.def("lt_linear", &lt_linear, "autotuned hipBLASLt linear (graph-safe)",
           py::arg("y"), py::arg("x"), py::arg("w"), py::arg("bias") = py::none());

@@ -26,6 +26,7 @@ sources = [
     os.path.join(CSRC, "attention_prefill.hip"),
     os.path.join(CSRC, "qkv_fused.hip"),
     os.path.join(CSRC, "skinny_gemm.hip"),
+    os.path.join(CSRC, "skinny_gemm2.hip"),
     os.path.join(CSRC, "lt_gemm.cpp"),
 ]
 
