@@ -1,0 +1,153 @@
+"""Continuous-batching serving engine.
+
+The reference defers serving orchestration to vLLM and ships the model-side
+hooks (continuous batching seq_ids, block KV, chunked prefill — SURVEY
+§2.4/§2.6); this engine closes the loop natively: a request queue, slot
+allocation over the KV cache's seq_id rows, and a step loop that
+interleaves one prefill (optionally chunked through the windowed-CTE path)
+with full-batch decode steps.
+
+Scheduling policy (deliberately simple, latency-fair):
+- if a slot is free and a request is queued: run that request's context
+  encoding (chunked when `chunk_size` is set — each chunk is one step, so
+  decode latency of running requests is bounded by one chunk);
+- otherwise: one token-generation step for every active slot, batched and
+  padded to the engine batch (seq_id -1 pads; the captured decode graph
+  serves every step regardless of how many slots are live).
+
+All state (tokens, positions) stays on the model's device between steps.
+"""
+
+from __future__ import annotations
+
+import collections
+from dataclasses import dataclass, field
+from typing import Deque, Dict, List, Optional
+
+import torch
+
+
+@dataclass
+class Request:
+    req_id: int
+    prompt: torch.Tensor            # (S,) input ids
+    max_new_tokens: int = 32
+    output: List[int] = field(default_factory=list)
+    slot: int = -1
+    pos: int = 0                    # next position to fill
+    prefill_done: int = 0           # prompt tokens already encoded
+    last_token: Optional[int] = None
+    done: bool = False
+
+
+class ServingEngine:
+    def __init__(self, causal_lm, max_slots: Optional[int] = None,
+                 chunk_size: Optional[int] = None,
+                 eos_token_id: Optional[int] = None):
+        self.model = causal_lm
+        ec = causal_lm.config.neuron_config
+        self.batch = ec.batch_size
+        self.max_slots = max_slots or ec.batch_size
+        self.chunk_size = chunk_size
+        self.eos = eos_token_id
+        self.device = next(causal_lm.parameters()).device
+        self.free_slots = list(range(self.max_slots))[::-1]
+        self.active: Dict[int, Request] = {}    # slot -> request
+        self.queue: Deque[Request] = collections.deque()
+        self.finished: Dict[int, Request] = {}
+        self._next_id = 0
+
+    # ------------------------------------------------------------------
+    def add_request(self, prompt_ids: torch.Tensor,
+                    max_new_tokens: int = 32) -> int:
+        rid = self._next_id
+        self._next_id += 1
+        self.queue.append(Request(rid, prompt_ids.view(-1).cpu(),
+                                  max_new_tokens))
+        return rid
+
+    def has_work(self) -> bool:
+        return bool(self.queue or self.active)
+
+    # ------------------------------------------------------------------
+    def _prefill_chunk(self, req: Request):
+        S = req.prompt.shape[0]
+        lo = req.prefill_done
+        hi = S if self.chunk_size is None else min(S, lo + self.chunk_size)
+        ids = req.prompt[lo:hi].view(1, -1).to(self.device)
+        positions = torch.arange(lo, hi, device=self.device).view(1, -1)
+        seq_ids = torch.tensor([req.slot], device=self.device)
+        out = self.model(ids, None, positions, seq_ids,
+                         self.model._default_sampling_params(1)
+                         if hasattr(self.model, "_default_sampling_params")
+                         else None)
+        req.prefill_done = hi
+        if hi == S:
+            lg = out.logits
+            tok = int(lg[0, -1].argmax()) if lg.dim() == 3 \
+                else int(lg.view(1, -1)[0, -1])
+            req.last_token = tok
+            req.pos = S
+            req.output.append(tok)
+            self._maybe_finish(req, tok)
+
+    def _decode_step(self):
+        slots = sorted(self.active)
+        B = self.batch
+        ids = torch.zeros(B, 1, dtype=torch.long, device=self.device)
+        pos = torch.full((B, 1), -1, dtype=torch.long, device=self.device)
+        seq_ids = torch.full((B,), -1, dtype=torch.long, device=self.device)
+        for i, slot in enumerate(slots):
+            r = self.active[slot]
+            ids[i, 0] = r.last_token
+            pos[i, 0] = r.pos
+            seq_ids[i] = slot
+        out = self.model(ids, None, pos, seq_ids,
+                         self.model._default_sampling_params(B)
+                         if hasattr(self.model, "_default_sampling_params")
+                         else None)
+        toks = out.logits
+        if toks.dim() > 2:
+            toks = toks.argmax(-1)
+        toks = toks.view(B, -1)[:, -1].tolist()
+        for i, slot in enumerate(slots):
+            r = self.active[slot]
+            tok = int(toks[i])
+            r.last_token = tok
+            r.pos += 1
+            r.output.append(tok)
+            self._maybe_finish(r, tok)
+
+    def _maybe_finish(self, req: Request, tok: int):
+        if len(req.output) >= req.max_new_tokens or (
+            self.eos is not None and tok == self.eos
+        ):
+            req.done = True
+            self.active.pop(req.slot)
+            self.free_slots.append(req.slot)
+            self.finished[req.req_id] = req
+
+    # ------------------------------------------------------------------
+    def step(self):
+        """One scheduling step: admit + prefill one chunk, or decode."""
+        # admit
+        if self.queue and self.free_slots:
+            req = self.queue.popleft()
+            req.slot = self.free_slots.pop()
+            self.active[req.slot] = req
+        # a request mid-prefill takes priority (its chunks must finish
+        # before it can decode); pick the first such
+        for slot in sorted(self.active):
+            r = self.active[slot]
+            if r.prefill_done < r.prompt.shape[0]:
+                self._prefill_chunk(r)
+                return
+        if self.active:
+            self._decode_step()
+
+    def run(self, max_steps: int = 100000) -> Dict[int, List[int]]:
+        steps = 0
+        while self.has_work() and steps < max_steps:
+            self.step()
+            steps += 1
+        return {rid: r.output for rid, r in self.finished.items()}

@@ -1,0 +1,60 @@
+"""Continuous-batching serving engine: mixed prefill/decode interleaving
+over KV slots must reproduce per-request reference generations."""
+
+import pytest
+import torch
+
+from neuronx_distributed_inference_amd.config import EngineConfig
+from neuronx_distributed_inference_amd.models.llama.modeling_llama import (
+    LlamaForCausalLM,
+    LlamaInferenceConfig,
+)
+from neuronx_distributed_inference_amd.serving import ServingEngine
+from neuronx_distributed_inference_amd.utils.hf_adapter import (
+    GenerationConfig,
+    HuggingFaceGenerationAdapter,
+)
+
+
+def _model(batch):
+    ec = EngineConfig(batch_size=batch, max_batch_size=batch, seq_len=96,
+                      max_length=96, torch_dtype="float32", tp_degree=1)
+    cfg = LlamaInferenceConfig(
+        neuron_config=ec, vocab_size=128, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, rms_norm_eps=1e-5, rope_theta=10000.0,
+        tie_word_embeddings=False)
+    torch.manual_seed(17)
+    m = LlamaForCausalLM(cfg).eval()
+    with torch.no_grad():
+        for p in m.parameters():
+            if p.dim() >= 2:
+                p.normal_(0, 0.1)
+    return m
+
+
+@torch.inference_mode()
+@pytest.mark.parametrize("chunk", [None, 8])
+def test_serving_engine_matches_reference(chunk):
+    m = _model(batch=3)
+    prompts = [
+        torch.randint(0, 128, (11,)),
+        torch.randint(0, 128, (19,)),
+        torch.randint(0, 128, (7,)),
+        torch.randint(0, 128, (23,)),  # queued until a slot frees
+        torch.randint(0, 128, (13,)),
+    ]
+    wants = []
+    for p in prompts:
+        m.reset()
+        out = HuggingFaceGenerationAdapter(m).generate(
+            p.view(1, -1), generation_config=GenerationConfig(max_new_tokens=10))
+        wants.append(out[0, p.shape[0]:].tolist())
+
+    m.reset()
+    eng = ServingEngine(m, chunk_size=chunk)
+    ids = [eng.add_request(p, max_new_tokens=10) for p in prompts]
+    results = eng.run()
+    assert set(results) == set(ids)
+    for rid, want in zip(ids, wants):
+        assert results[rid] == want, (rid, results[rid], want)
