@@ -93,6 +93,11 @@ _REGISTRY: Dict[str, Tuple[str, str, str]] = {
         "Qwen3VLInferenceConfig",
         "Qwen3VLForConditionalGeneration",
     ),
+    "flux": (
+        "neuronx_distributed_inference_amd.models.flux.modeling_flux",
+        "FluxInferenceConfig",
+        "FluxTransformer2DModel",
+    ),
     "llama4_vision": (
         "neuronx_distributed_inference_amd.models.llama4.modeling_llama4",
         "Llama4VisionInferenceConfig",

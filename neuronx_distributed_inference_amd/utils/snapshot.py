@@ -71,3 +71,20 @@ def attach_snapshot_hook(causal_lm) -> Optional[SnapshotHook]:
 
     causal_lm.forward = wrapped
     return hook
+
+
+def reconstruct_kv_cache(shards, strategy: str = "auto"):
+    """Reassemble per-rank KV cache shards into the full cache for
+    debugging (ref utils/kv_cache_reconstruct_utils.py:57-218).
+
+    shards: list over TP ranks of (L, B, Hkv_local, S, D)-shaped tensors
+    (or a single layer's (B, Hkv_local, S, D)). Head-sharded caches concat
+    along the head dim; replicated caches (Hkv < tp, each rank holds the
+    same head) verify equality and return shard 0.
+    """
+    import torch
+
+    first = shards[0]
+    if all(torch.equal(s, first) for s in shards[1:]):
+        return first.clone()   # replicated heads
+    return torch.cat(shards, dim=-3)

@@ -72,3 +72,17 @@ def test_async_executor_overlap_and_backpressure():
     for i, o in enumerate(outs):
         assert torch.equal(o, torch.full((2,), float(i)))
     assert all(h.done() for h in handles)
+
+
+def test_reconstruct_kv_cache():
+    from neuronx_distributed_inference_amd.utils.snapshot import (
+        reconstruct_kv_cache,
+    )
+
+    full = torch.randn(2, 4, 16, 8)    # (B, Hkv=4, S, D)
+    shards = [full[:, :2], full[:, 2:]]
+    out = reconstruct_kv_cache(shards)
+    assert torch.equal(out, full)
+    rep = [full[:, :1].clone(), full[:, :1].clone()]
+    out = reconstruct_kv_cache(rep)
+    assert torch.equal(out, full[:, :1])
