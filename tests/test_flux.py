@@ -121,3 +121,55 @@ def test_pipeline_deterministic_denoise():
     assert a.shape == (1, 16, 16)
     assert torch.isfinite(a).all()
     torch.testing.assert_close(a, b)
+
+
+@torch.inference_mode()
+def test_clip_text_encoder_matches_hf():
+    from transformers import CLIPTextConfig, CLIPTextModel
+
+    from neuronx_distributed_inference_amd.models.flux.text_encoders import (
+        ClipTextEncoder,
+    )
+
+    torch.manual_seed(0)
+    cfg = CLIPTextConfig(vocab_size=100, hidden_size=32, intermediate_size=64,
+                         num_hidden_layers=2, num_attention_heads=4,
+                         max_position_embeddings=24, eos_token_id=2,
+                         bos_token_id=1, pad_token_id=0, hidden_act="quick_gelu")
+    hf = CLIPTextModel(cfg).eval()
+    ours = ClipTextEncoder(100, 32, 4, 64, 2, 24, eos_token_id=2).eval()
+    sd = ClipTextEncoder.convert_hf_to_native_state_dict(hf.state_dict())
+    missing, unexpected = ours.load_state_dict(sd, strict=False)
+    assert not missing, missing
+    ids = torch.randint(3, 100, (2, 10))
+    ids[:, -1] = 2
+    want = hf(input_ids=ids)
+    h, pooled = ours(ids)
+    torch.testing.assert_close(h, want.last_hidden_state, atol=1e-4, rtol=1e-4)
+    torch.testing.assert_close(pooled, want.pooler_output, atol=1e-4, rtol=1e-4)
+
+
+@torch.inference_mode()
+def test_t5_encoder_matches_hf():
+    from transformers import T5Config, T5EncoderModel
+
+    from neuronx_distributed_inference_amd.models.flux.text_encoders import (
+        T5TextEncoder,
+    )
+
+    torch.manual_seed(0)
+    cfg = T5Config(vocab_size=120, d_model=32, d_kv=8, d_ff=64,
+                   num_layers=2, num_heads=4, relative_attention_num_buckets=8,
+                   relative_attention_max_distance=20,
+                   feed_forward_proj="gated-gelu", dense_act_fn="gelu_new",
+                   is_gated_act=True, tie_word_embeddings=False)
+    hf = T5EncoderModel(cfg).eval()
+    ours = T5TextEncoder(120, 32, 4, 8, 64, 2, num_buckets=8,
+                         max_distance=20).eval()
+    sd = T5TextEncoder.convert_hf_to_native_state_dict(hf.state_dict())
+    missing, unexpected = ours.load_state_dict(sd, strict=False)
+    assert not missing, missing
+    ids = torch.randint(0, 120, (2, 12))
+    want = hf(input_ids=ids).last_hidden_state
+    got = ours(ids)
+    torch.testing.assert_close(got, want, atol=1e-4, rtol=1e-4)
