@@ -46,7 +46,7 @@ def parse_args():
     p.add_argument("--steps", type=int, default=32)
     p.add_argument("--warmup", type=int, default=8)
     p.add_argument("--model", default="8b", choices=list(MODELS))
-    p.add_argument("--batch-per-gpu", type=int, default=128)
+    p.add_argument("--batch-per-gpu", type=int, default=256)
     p.add_argument("--ctx", type=int, default=128)
     p.add_argument("--no-graphs", action="store_true")
     p.add_argument("--e2e", action="store_true",

@@ -15,7 +15,6 @@ reference (application_base.py:47,292-316,389-399):
 
 from __future__ import annotations
 
-import glob
 import logging
 import os
 import shutil

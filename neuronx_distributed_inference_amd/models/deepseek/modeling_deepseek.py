@@ -23,7 +23,6 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from neuronx_distributed_inference_amd import ops as dev_ops
 from neuronx_distributed_inference_amd.config import InferenceConfig
 from neuronx_distributed_inference_amd.models.llama.modeling_llama import LlamaForCausalLM
 from neuronx_distributed_inference_amd.models.model_base import DecoderModel

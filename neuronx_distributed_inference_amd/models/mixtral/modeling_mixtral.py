@@ -12,7 +12,6 @@ from neuronx_distributed_inference_amd.config import InferenceConfig
 from neuronx_distributed_inference_amd.models.llama.modeling_llama import (
     LlamaForCausalLM,
     LlamaInferenceConfig,
-    LlamaModel,
 )
 from neuronx_distributed_inference_amd.models.model_base import DecoderModel
 from neuronx_distributed_inference_amd.modules.attention.attention_base import AttentionBase

@@ -18,7 +18,6 @@ Analogs of the reference's `models/model_base.py`:
 from __future__ import annotations
 
 import logging
-import os
 from dataclasses import dataclass
 from typing import Dict, List, Optional
 

@@ -19,7 +19,6 @@ import torch
 import torch.nn as nn
 
 from neuronx_distributed_inference_amd.modules.attention.attention_base import (
-    PHASE_CTE,
     PHASE_TKG,
 )
 from neuronx_distributed_inference_amd.parallel.ops import distributed_argmax

@@ -16,7 +16,6 @@ preshard hooks + `weights/tp{rank}_sharded_checkpoint.safetensors` layout
 
 from __future__ import annotations
 
-import math
 from typing import Optional
 
 import torch
