@@ -77,10 +77,7 @@ class ServingEngine:
         ids = req.prompt[lo:hi].view(1, -1).to(self.device)
         positions = torch.arange(lo, hi, device=self.device).view(1, -1)
         seq_ids = torch.tensor([req.slot], device=self.device)
-        out = self.model(ids, None, positions, seq_ids,
-                         self.model._default_sampling_params(1)
-                         if hasattr(self.model, "_default_sampling_params")
-                         else None)
+        out = self.model(ids, None, positions, seq_ids, None)
         req.prefill_done = hi
         if hi == S:
             lg = out.logits
@@ -102,10 +99,7 @@ class ServingEngine:
             ids[i, 0] = r.last_token
             pos[i, 0] = r.pos
             seq_ids[i] = slot
-        out = self.model(ids, None, pos, seq_ids,
-                         self.model._default_sampling_params(B)
-                         if hasattr(self.model, "_default_sampling_params")
-                         else None)
+        out = self.model(ids, None, pos, seq_ids, None)
         toks = out.logits
         if toks.dim() > 2:
             toks = toks.argmax(-1)
