@@ -27,15 +27,17 @@ __global__ void qkv_rope_update_kernel(
     using VT = VecTraits<T>;
     const int half = D / 2;
     const int Ht = Hq + 2 * Hkv;
-    const long total = (long)B * S * Ht * half;
-    const long stride = (long)gridDim.x * blockDim.x;
-    for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total; idx += stride) {
+    // direct 2D grid: blockIdx.x = (b, s) row, blockIdx.y covers Ht*half —
+    // the grid-stride formulation spent its cycles on per-iteration 64-bit
+    // div/mod (measured 45 GB/s at prefill scale)
+    {
+        const int bs = blockIdx.x;
+        const int b = bs / S;
+        const int s = bs % S;
+        const int idx = blockIdx.y * blockDim.x + threadIdx.x;
+        if (idx >= Ht * half) return;
+        const int h = idx / half;
         const int i = idx % half;
-        long rest = idx / half;
-        const int h = rest % Ht;
-        rest /= Ht;
-        const int s = rest % S;
-        const int b = rest / S;
 
         const T* src = qkv + (((long)b * S + s) * Ht + h) * D;
         const float x1f = VT::to_float(src[i]);
@@ -59,7 +61,7 @@ __global__ void qkv_rope_update_kernel(
         } else {
             const int row = seq_ids[b];
             const int pos = positions[(long)b * S + s];
-            if (row < 0 || row >= Bc || pos < 0 || pos >= Smax) continue;
+            if (row < 0 || row >= Bc || pos < 0 || pos >= Smax) return;
             const bool is_k = h < Hq + Hkv;
             const int hk = is_k ? h - Hq : h - Hq - Hkv;
             CT* cache = is_k ? k_cache : v_cache;
@@ -76,11 +78,12 @@ __global__ void act_mul_fused_kernel(
     const T* __restrict__ gate_up,   // (rows, 2I): [gate | up]
     const long rows, const int I) {
     using VT = VecTraits<T>;
-    const long total = rows * (I / 8);
-    const long stride = (long)gridDim.x * blockDim.x;
-    for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total; idx += stride) {
-        const int c8 = idx % (I / 8);
-        const long r = idx / (I / 8);
+    // blockIdx.x = row, blockIdx.y covers I/8 (see qkv kernel note on the
+    // div/mod cost of the grid-stride form)
+    {
+        const long r = blockIdx.x;
+        const int c8 = blockIdx.y * blockDim.x + threadIdx.x;
+        if (c8 >= I / 8) return;
         Vec8<T> g = load_vec8(gate_up + r * 2 * I + c8 * 8);
         Vec8<T> u = load_vec8(gate_up + r * 2 * I + I + c8 * 8);
 #pragma unroll
@@ -126,15 +129,14 @@ void qkv_rope_update(torch::Tensor q_out, torch::Tensor qkv,
     const int Bc = k_cache.size(0), Smax = k_cache.size(2);
     TORCH_CHECK(qkv.size(2) == (Hq + 2 * Hkv) * D, "qkv width mismatch");
     TORCH_CHECK(qkv.is_contiguous(), "qkv must be contiguous");
-    const long total = (long)B * S * (Hq + 2 * Hkv) * (D / 2);
-    long blocks = (total + 255) / 256;
-    if (blocks > 2048) blocks = 2048;
+    const int ht_half = (int)(Hq + 2 * Hkv) * (D / 2);
+    dim3 blocks((unsigned)(B * S), (unsigned)((ht_half + 255) / 256));
     auto stream = at::hip::getCurrentHIPStream();
     const bool fp8 = k_cache.scalar_type() == at::ScalarType::Float8_e4m3fn;
     DISPATCH_FT(qkv.scalar_type(), "qkv_rope_update", [&] {
         if (fp8) {
             hipLaunchKernelGGL((qkv_rope_update_kernel<scalar_t, __hip_fp8_e4m3>),
-                               dim3((int)blocks), dim3(256), 0, stream,
+                               blocks, dim3(256), 0, stream,
                                (scalar_t*)q_out.data_ptr(), (const scalar_t*)qkv.data_ptr(),
                                cosb.data_ptr<float>(), sinb.data_ptr<float>(),
                                (__hip_fp8_e4m3*)k_cache.data_ptr(),
@@ -143,7 +145,7 @@ void qkv_rope_update(torch::Tensor q_out, torch::Tensor qkv,
                                B, S, (int)Hq, (int)Hkv, D, Bc, Smax);
         } else {
             hipLaunchKernelGGL((qkv_rope_update_kernel<scalar_t, scalar_t>),
-                               dim3((int)blocks), dim3(256), 0, stream,
+                               blocks, dim3(256), 0, stream,
                                (scalar_t*)q_out.data_ptr(), (const scalar_t*)qkv.data_ptr(),
                                cosb.data_ptr<float>(), sinb.data_ptr<float>(),
                                (scalar_t*)k_cache.data_ptr(), (scalar_t*)v_cache.data_ptr(),
@@ -159,12 +161,10 @@ void silu_mul_fused(torch::Tensor out, torch::Tensor gate_up) {
     const long rows = gate_up.numel() / I2;
     const int I = I2 / 2;
     TORCH_CHECK(I % 8 == 0, "intermediate size must be a multiple of 8");
-    const long total = rows * (I / 8);
-    long blocks = (total + 255) / 256;
-    if (blocks > 2048) blocks = 2048;
+    dim3 blocks((unsigned)rows, (unsigned)((I / 8 + 255) / 256));
     auto stream = at::hip::getCurrentHIPStream();
     DISPATCH_FT(gate_up.scalar_type(), "silu_mul_fused", [&] {
-        hipLaunchKernelGGL((act_mul_fused_kernel<scalar_t, false>), dim3((int)blocks),
+        hipLaunchKernelGGL((act_mul_fused_kernel<scalar_t, false>), blocks,
                            dim3(256), 0, stream,
                            (scalar_t*)out.data_ptr(), (const scalar_t*)gate_up.data_ptr(),
                            rows, I);
@@ -177,12 +177,10 @@ void gelu_tanh_mul_fused(torch::Tensor out, torch::Tensor gate_up) {
     const long rows = gate_up.numel() / I2;
     const int I = I2 / 2;
     TORCH_CHECK(I % 8 == 0, "intermediate size must be a multiple of 8");
-    const long total = rows * (I / 8);
-    long blocks = (total + 255) / 256;
-    if (blocks > 2048) blocks = 2048;
+    dim3 blocks((unsigned)rows, (unsigned)((I / 8 + 255) / 256));
     auto stream = at::hip::getCurrentHIPStream();
     DISPATCH_FT(gate_up.scalar_type(), "gelu_tanh_mul_fused", [&] {
-        hipLaunchKernelGGL((act_mul_fused_kernel<scalar_t, true>), dim3((int)blocks),
+        hipLaunchKernelGGL((act_mul_fused_kernel<scalar_t, true>), blocks,
                            dim3(256), 0, stream,
                            (scalar_t*)out.data_ptr(), (const scalar_t*)gate_up.data_ptr(),
                            rows, I);
