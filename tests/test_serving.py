@@ -58,3 +58,28 @@ def test_serving_engine_matches_reference(chunk):
     assert set(results) == set(ids)
     for rid, want in zip(ids, wants):
         assert results[rid] == want, (rid, results[rid], want)
+
+
+@pytest.mark.gpu
+@torch.inference_mode()
+def test_serving_engine_gpu():
+    """Serving engine on GPU (bf16, HIP kernels): chunked-prefill
+    interleaving must match per-request generation on the same device."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    m = _model(batch=3)
+    m = m.bfloat16().to("cuda")
+    prompts = [torch.randint(0, 128, (n,)) for n in (11, 19, 7, 23)]
+    wants = []
+    for p in prompts:
+        m.reset()
+        out = HuggingFaceGenerationAdapter(m).generate(
+            p.view(1, -1).cuda(),
+            generation_config=GenerationConfig(max_new_tokens=8))
+        wants.append(out[0, p.shape[0]:].cpu().tolist())
+    m.reset()
+    eng = ServingEngine(m, chunk_size=8)
+    ids = [eng.add_request(p, max_new_tokens=8) for p in prompts]
+    results = eng.run()
+    for rid, want in zip(ids, wants):
+        assert results[rid] == want, (rid, results[rid], want)
