@@ -67,8 +67,21 @@ def test_serving_engine_gpu():
     interleaving must match per-request generation on the same device."""
     if not torch.cuda.is_available():
         pytest.skip("no GPU")
-    m = _model(batch=3)
-    m = m.bfloat16().to("cuda")
+    # head_dim >= 32 for the HIP decode kernel
+    ec = EngineConfig(batch_size=3, max_batch_size=3, seq_len=96,
+                      max_length=96, torch_dtype="bfloat16", tp_degree=1)
+    cfg = LlamaInferenceConfig(
+        neuron_config=ec, vocab_size=128, hidden_size=128, intermediate_size=256,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, rms_norm_eps=1e-5, rope_theta=10000.0,
+        tie_word_embeddings=False)
+    torch.manual_seed(17)
+    m = LlamaForCausalLM(cfg).eval()
+    with torch.no_grad():
+        for p in m.parameters():
+            if p.dim() >= 2:
+                p.normal_(0, 0.1)
+    m = m.to("cuda")
     prompts = [torch.randint(0, 128, (n,)) for n in (11, 19, 7, 23)]
     wants = []
     for p in prompts:
