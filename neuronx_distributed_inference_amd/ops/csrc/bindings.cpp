@@ -35,6 +35,10 @@ void skinny_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                  c10::optional<torch::Tensor> bias);
 void skinny_gemm2(torch::Tensor y, torch::Tensor y32, torch::Tensor x,
                   torch::Tensor w, c10::optional<torch::Tensor> bias, long ksplit);
+void fp8_quant(torch::Tensor x, torch::Tensor out, torch::Tensor scale,
+               torch::Tensor amax_ws);
+void fp8_epilogue(torch::Tensor out, torch::Tensor y32, torch::Tensor w_scale,
+                  c10::optional<torch::Tensor> bias);
 bool lt_available();
 void lt_linear(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                c10::optional<torch::Tensor> bias);
@@ -72,6 +76,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("skinny_gemm2", &skinny_gemm2, "blocked skinny-M GEMM, K-split (CDNA4)",
           py::arg("y"), py::arg("y32"), py::arg("x"), py::arg("w"),
           py::arg("bias") = py::none(), py::arg("ksplit") = 1);
+    m.def("fp8_quant", &fp8_quant, "dynamic per-tensor fp8 quant (CDNA4)");
+    m.def("fp8_epilogue", &fp8_epilogue, "scaled-mm row-scale epilogue (CDNA4)",
+          py::arg("out"), py::arg("y32"), py::arg("w_scale"),
+          py::arg("bias") = py::none());
     m.def("lt_available", &lt_available, "hipblaslt usable on this device");
     m.def("lt_linear", &lt_linear, "autotuned hipBLASLt linear (graph-safe)",
           py::arg("y"), py::arg("x"), py::arg("w"), py::arg("bias") = py::none());

@@ -60,25 +60,44 @@ def quantize_linear_fp8(mod: nn.Module):
     mod.fp8 = True
 
 
+_FP8_ONES = {}
+
+
+def _fp8_one(device):
+    t = _FP8_ONES.get(device)
+    if t is None:
+        t = torch.ones((1, 1), device=device, dtype=torch.float32)
+        _FP8_ONES[device] = t
+    return t
+
+
 def fp8_linear(x: torch.Tensor, w_fp8: torch.Tensor, w_scale: torch.Tensor, bias):
     """y = x @ dequant(w)^T (+bias). GPU: torch._scaled_mm with dynamic
     per-tensor activation scale; CPU: dequantized matmul."""
     if x.is_cuda:
-        # hipBLASLt scaled-mm takes per-tensor scales on this stack; apply
-        # the per-row weight scale on the (small M) output instead
+        # hipBLASLt scaled-mm takes per-tensor scales on this stack; the
+        # per-row weight scale rides in the HIP epilogue kernel. Dynamic
+        # activation quant runs as two HIP kernels (amax atomic + cast) —
+        # no fp32 materialization of x, no host sync (K6-analog,
+        # ops/csrc/fp8_quant.hip).
+        import neuronx_distributed_inference_amd.ops as _ops
+
+        c = _ops._require_ext("fp8_quant")
         shape = x.shape[:-1] + (w_fp8.shape[0],)
-        x2 = x.reshape(-1, x.shape[-1])
-        x_scale = x2.float().abs().amax().clamp(min=1e-8) / FP8_MAX
-        x_q = (x2.float() / x_scale).clamp(-FP8_MAX, FP8_MAX).to(torch.float8_e4m3fn)
-        one = torch.ones((1, 1), device=x.device, dtype=torch.float32)
-        y = torch._scaled_mm(
-            x_q, w_fp8.t(), scale_a=x_scale.view(1, 1), scale_b=one,
+        x2 = x.reshape(-1, x.shape[-1]).contiguous()
+        x_q = torch.empty_like(x2, dtype=torch.float8_e4m3fn)
+        scale = torch.empty(1, 1, dtype=torch.float32, device=x.device)
+        amax = torch.empty(128, dtype=torch.float32, device=x.device)
+        c.fp8_quant(x2, x_q, scale, amax)
+        one = _fp8_one(x.device)
+        y32 = torch._scaled_mm(
+            x_q, w_fp8.t(), scale_a=scale, scale_b=one,
             bias=None, out_dtype=torch.float32,
         )
-        y = y * w_scale.view(1, -1)
-        if bias is not None:
-            y = y + bias.float()
-        return y.to(x.dtype).reshape(shape)
+        out = torch.empty(shape, dtype=x.dtype, device=x.device)
+        c.fp8_epilogue(out, y32, w_scale.reshape(-1).contiguous(),
+                       None if bias is None else bias)
+        return out
     w = w_fp8.float() * w_scale
     y = F.linear(x.float(), w)
     if bias is not None:

@@ -25,6 +25,7 @@ sources = [
     os.path.join(CSRC, "sampling.hip"),
     os.path.join(CSRC, "attention_prefill.hip"),
     os.path.join(CSRC, "qkv_fused.hip"),
+    os.path.join(CSRC, "fp8_quant.hip"),
     os.path.join(CSRC, "skinny_gemm.hip"),
     os.path.join(CSRC, "skinny_gemm2.hip"),
     os.path.join(CSRC, "lt_gemm.cpp"),

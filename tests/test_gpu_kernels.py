@@ -651,3 +651,28 @@ def test_gpu_model_variants_match_cpu():
         ids.to(dev), generation_config=gc)
     agree = (got.cpu() == want).float().mean()
     assert agree > 0.8, (agree, got.cpu().tolist(), want.tolist())
+
+
+def test_fp8_quant_kernels_match_torch():
+    """fp8_quant (amax+cast) and fp8_epilogue vs the torch formulation."""
+    from neuronx_distributed_inference_amd.ops import _nxdi_C as C
+
+    torch.manual_seed(0)
+    dev = _dev()
+    x = torch.randn(64, 512, dtype=torch.bfloat16, device=dev) * 3
+    x_q = torch.empty_like(x, dtype=torch.float8_e4m3fn)
+    scale = torch.empty(1, 1, dtype=torch.float32, device=dev)
+    amax = torch.empty(128, dtype=torch.float32, device=dev)
+    C.fp8_quant(x, x_q, scale, amax)
+    want_scale = x.float().abs().amax().clamp(min=1e-8) / 448.0
+    torch.testing.assert_close(scale.view(()), want_scale, atol=1e-6, rtol=1e-5)
+    want_q = (x.float() / want_scale).clamp(-448, 448).to(torch.float8_e4m3fn)
+    torch.testing.assert_close(x_q.float(), want_q.float())
+    # epilogue
+    y32 = torch.randn(64, 128, dtype=torch.float32, device=dev)
+    ws = torch.rand(128, dtype=torch.float32, device=dev) + 0.5
+    b = torch.randn(128, dtype=torch.bfloat16, device=dev)
+    out = torch.empty(64, 128, dtype=torch.bfloat16, device=dev)
+    C.fp8_epilogue(out, y32, ws, b)
+    want = (y32 * ws + b.float()).to(torch.bfloat16)
+    torch.testing.assert_close(out.float(), want.float(), atol=2e-2, rtol=2e-2)
