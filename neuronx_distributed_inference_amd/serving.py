@@ -32,6 +32,9 @@ class Request:
     req_id: int
     prompt: torch.Tensor            # (S,) input ids
     max_new_tokens: int = 32
+    top_k: int = 1
+    top_p: float = 1.0
+    temperature: float = 1.0
     output: List[int] = field(default_factory=list)
     slot: int = -1
     pos: int = 0                    # next position to fill
@@ -59,12 +62,26 @@ class ServingEngine:
 
     # ------------------------------------------------------------------
     def add_request(self, prompt_ids: torch.Tensor,
-                    max_new_tokens: int = 32) -> int:
+                    max_new_tokens: int = 32, top_k: int = 1,
+                    top_p: float = 1.0, temperature: float = 1.0) -> int:
+        # per-request on-device sampling params ride in the (B,3)
+        # sampling_params rows (ref per-request sampling_params)
         rid = self._next_id
         self._next_id += 1
         self.queue.append(Request(rid, prompt_ids.view(-1).cpu(),
-                                  max_new_tokens))
+                                  max_new_tokens, top_k, top_p, temperature))
         return rid
+
+    def _params_for(self, reqs) -> Optional[torch.Tensor]:
+        if self.model.config.neuron_config.on_device_sampling_config is None:
+            return None
+        B = self.batch
+        p = torch.ones(B, 3, dtype=torch.float32, device=self.device)
+        for i, r in enumerate(reqs):
+            p[i, 0] = r.top_k
+            p[i, 1] = r.top_p
+            p[i, 2] = r.temperature
+        return p
 
     def has_work(self) -> bool:
         return bool(self.queue or self.active)
@@ -77,10 +94,12 @@ class ServingEngine:
         ids = req.prompt[lo:hi].view(1, -1).to(self.device)
         positions = torch.arange(lo, hi, device=self.device).view(1, -1)
         seq_ids = torch.tensor([req.slot], device=self.device)
-        out = self.model(ids, None, positions, seq_ids, None)
+        sp = self._params_for([req])
+        out = self.model(ids, None, positions, seq_ids,
+                         None if sp is None else sp[:1])
         req.prefill_done = hi
         if hi == S:
-            lg = out.logits
+            lg = out.logits if out.logits is not None else out.tokens
             tok = int(lg[0, -1].argmax()) if lg.dim() == 3 \
                 else int(lg.view(1, -1)[0, -1])
             req.last_token = tok
@@ -99,8 +118,9 @@ class ServingEngine:
             ids[i, 0] = r.last_token
             pos[i, 0] = r.pos
             seq_ids[i] = slot
-        out = self.model(ids, None, pos, seq_ids, None)
-        toks = out.logits
+        out = self.model(ids, None, pos, seq_ids,
+                         self._params_for([self.active[s] for s in slots]))
+        toks = out.logits if out.logits is not None else out.tokens
         if toks.dim() > 2:
             toks = toks.argmax(-1)
         toks = toks.view(B, -1)[:, -1].tolist()

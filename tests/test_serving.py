@@ -96,3 +96,38 @@ def test_serving_engine_gpu():
     results = eng.run()
     for rid, want in zip(ids, wants):
         assert results[rid] == want, (rid, results[rid], want)
+
+
+@torch.inference_mode()
+def test_serving_per_request_sampling_params():
+    """Requests carry their own (top_k, top_p, temperature); greedy rows
+    must still match the reference while a high-temperature row samples."""
+    from neuronx_distributed_inference_amd.config import OnDeviceSamplingConfig
+
+    ec = EngineConfig(batch_size=2, max_batch_size=2, seq_len=64, max_length=64,
+                      torch_dtype="float32", tp_degree=1,
+                      on_device_sampling_config=OnDeviceSamplingConfig(
+                          top_k=1, do_sample=True))
+    cfg = LlamaInferenceConfig(
+        neuron_config=ec, vocab_size=128, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, rms_norm_eps=1e-5, rope_theta=10000.0,
+        tie_word_embeddings=False)
+    torch.manual_seed(17)
+    m = LlamaForCausalLM(cfg).eval()
+    with torch.no_grad():
+        for p in m.parameters():
+            if p.dim() >= 2:
+                p.normal_(0, 0.1)
+    prompt = torch.randint(0, 128, (9,))
+    m.reset()
+    want = HuggingFaceGenerationAdapter(m).generate(
+        prompt.view(1, -1), generation_config=GenerationConfig(max_new_tokens=6))
+    m.reset()
+    eng = ServingEngine(m)
+    greedy = eng.add_request(prompt, max_new_tokens=6, top_k=1)
+    sampled = eng.add_request(torch.randint(0, 128, (7,)), max_new_tokens=6,
+                              top_k=20, temperature=1.5)
+    results = eng.run()
+    assert results[greedy] == want[0, 9:].tolist()
+    assert len(results[sampled]) == 6
