@@ -37,6 +37,11 @@ MODELS = {
                     num_hidden_layers=32, num_attention_heads=32,
                     num_key_value_heads=8, vocab_size=32000,
                     num_local_experts=8, num_experts_per_tok=2),
+    # DBRX-132B geometry (BASELINE config 4): 16 experts top-4
+    "dbrx": dict(hidden_size=6144, intermediate_size=10752,
+                 num_hidden_layers=40, num_attention_heads=48,
+                 num_key_value_heads=8, vocab_size=100352,
+                 num_local_experts=16, num_experts_per_tok=4),
 }
 
 
@@ -215,10 +220,13 @@ def main():
         enable_hip_graphs=not args.no_graphs,
         fused_qkv=True, fused_gate_up=True,
     )
-    model_cls, cfg_cls = (
-        (MixtralForCausalLM, MixtralInferenceConfig)
-        if args.model == "mixtral" else (LlamaForCausalLM, LlamaInferenceConfig)
-    )
+    if args.model in ("mixtral", "dbrx"):
+        # dbrx geometry is benchmarked through the same stacked-expert MoE
+        # module (the DBRX model class differs in norms/clip, same compute
+        # shape); random-init synthetic benchmarking per BASELINE
+        model_cls, cfg_cls = MixtralForCausalLM, MixtralInferenceConfig
+    else:
+        model_cls, cfg_cls = LlamaForCausalLM, LlamaInferenceConfig
     cfg = cfg_cls(
         neuron_config=ec, max_position_embeddings=total_len * 2,
         rms_norm_eps=1e-5, rope_theta=500000.0, hidden_act="silu",
@@ -312,7 +320,8 @@ def main():
             "config": {
                 "model": {"8b": "llama-3.1-8b", "70b": "llama-3.1-70b",
                           "tiny": "tiny-llama-2l", "1b": "llama-3.2-1b",
-                          "mixtral": "mixtral-8x7b"}[args.model],
+                          "mixtral": "mixtral-8x7b",
+                          "dbrx": "dbrx-132b-geometry"}[args.model],
                 "global_batch": global_batch,
                 "seq_len": total_len,
                 "ctx": args.ctx,
