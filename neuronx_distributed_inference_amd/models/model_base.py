@@ -385,6 +385,8 @@ class CausalLMBase(nn.Module):
         position_ids: Optional[torch.Tensor] = None,
         seq_ids: Optional[torch.Tensor] = None,
         sampling_params: Optional[torch.Tensor] = None,
+        block_table: Optional[torch.Tensor] = None,
+        phase: Optional[str] = None,
     ) -> ModelOutput:
         B, S = input_ids.shape
         if position_ids is None:
@@ -430,7 +432,17 @@ class CausalLMBase(nn.Module):
                 input_ids, attention_mask, position_ids, seq_ids, sampling_params, wce
             )
 
-        is_spec_step = S > 1 and self.kv_cache_populated
+        is_spec_step = (S > 1 and self.kv_cache_populated) or \
+            (phase == "token_generation" and S > 1)
+        if block_table is not None:
+            # paged requests bypass the submodel wrappers (per-request block
+            # tables change shape semantics; serving drives this path)
+            tokens, logits = self.model(
+                input_ids, attention_mask, position_ids, seq_ids,
+                sampling_params, phase=phase, block_table=block_table,
+            )
+            self.kv_cache_populated = True
+            return ModelOutput(tokens=tokens, logits=logits)
         if is_spec_step and self.speculation_model is not None:
             tokens, logits = self.speculation_model(
                 input_ids, attention_mask, position_ids, seq_ids, sampling_params

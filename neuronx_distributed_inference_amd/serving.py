@@ -1,4 +1,5 @@
-"""Continuous-batching serving engine.
+"""Continuous-batching serving engine (linear or paged KV, with
+prefix caching on the paged path).
 
 The reference defers serving orchestration to vLLM and ships the model-side
 hooks (continuous batching seq_ids, block KV, chunked prefill — SURVEY
@@ -41,6 +42,8 @@ class Request:
     prefill_done: int = 0           # prompt tokens already encoded
     last_token: Optional[int] = None
     done: bool = False
+    blocks: List[int] = field(default_factory=list)  # paged: block chain
+    n_cached: int = 0               # prompt tokens served from prefix cache
 
 
 class ServingEngine:
@@ -54,6 +57,18 @@ class ServingEngine:
         self.chunk_size = chunk_size
         self.eos = eos_token_id
         self.device = next(causal_lm.parameters()).device
+        self.kv_mgr = causal_lm.model.kv_mgr
+        self.paged = bool(getattr(self.kv_mgr, "is_paged", False))
+        if self.paged:
+            from neuronx_distributed_inference_amd.modules.kvcache.prefix_cache import (
+                BlockAllocator,
+                PrefixCache,
+            )
+
+            self.block_size = self.kv_mgr.block_size
+            self.allocator = BlockAllocator(self.kv_mgr.num_blocks)
+            self.prefix_cache = PrefixCache(self.allocator, self.block_size)
+            self.max_blocks = (ec.max_length + self.block_size - 1) // self.block_size
         self.free_slots = list(range(self.max_slots))[::-1]
         self.active: Dict[int, Request] = {}    # slot -> request
         self.queue: Deque[Request] = collections.deque()
@@ -87,16 +102,53 @@ class ServingEngine:
         return bool(self.queue or self.active)
 
     # ------------------------------------------------------------------
+    def _ensure_blocks(self, req: Request, upto_pos: int):
+        """Grow the request's block chain to cover position `upto_pos`,
+        evicting cache-only blocks under pressure."""
+        need = (upto_pos + self.block_size) // self.block_size - len(req.blocks)
+        if need > 0:
+            if self.allocator.available() < need:
+                self.prefix_cache.evict(want_blocks=need)
+            req.blocks.extend(self.allocator.alloc(need))
+
+    def _block_table(self, reqs) -> torch.Tensor:
+        bt = torch.full((len(reqs), self.max_blocks), -1, dtype=torch.long,
+                        device=self.device)
+        for i, r in enumerate(reqs):
+            if r.blocks:
+                bt[i, : len(r.blocks)] = torch.tensor(r.blocks,
+                                                      device=self.device)
+        return bt
+
     def _prefill_chunk(self, req: Request):
         S = req.prompt.shape[0]
+        if self.paged and req.prefill_done == 0 and not req.blocks:
+            # prefix-cache hit: shared FULL blocks skip their prefill
+            req.blocks = self.prefix_cache.match(req.prompt.tolist())
+            req.n_cached = len(req.blocks) * self.block_size
+            # never skip the whole prompt: the last token must run to
+            # produce logits
+            if req.n_cached >= S:
+                drop = req.blocks.pop()
+                self.allocator.deref([drop])
+                req.n_cached -= self.block_size
+            req.prefill_done = req.n_cached
         lo = req.prefill_done
         hi = S if self.chunk_size is None else min(S, lo + self.chunk_size)
         ids = req.prompt[lo:hi].view(1, -1).to(self.device)
         positions = torch.arange(lo, hi, device=self.device).view(1, -1)
         seq_ids = torch.tensor([req.slot], device=self.device)
         sp = self._params_for([req])
+        kw = {}
+        if self.paged:
+            self._ensure_blocks(req, hi - 1)
+            kw["block_table"] = self._block_table([req])
+            if lo > 0:
+                # continuation chunk over prior (possibly shared) KV must
+                # run the multi-token TKG path, like windowed CTE
+                kw["phase"] = "token_generation"
         out = self.model(ids, None, positions, seq_ids,
-                         None if sp is None else sp[:1])
+                         None if sp is None else sp[:1], **kw)
         req.prefill_done = hi
         if hi == S:
             lg = out.logits if out.logits is not None else out.tokens
@@ -118,8 +170,16 @@ class ServingEngine:
             ids[i, 0] = r.last_token
             pos[i, 0] = r.pos
             seq_ids[i] = slot
-        out = self.model(ids, None, pos, seq_ids,
-                         self._params_for([self.active[s] for s in slots]))
+        reqs = [self.active[s] for s in slots]
+        kw = {}
+        if self.paged:
+            for r in reqs:
+                self._ensure_blocks(r, r.pos)
+            full = torch.full((B, self.max_blocks), -1, dtype=torch.long,
+                              device=self.device)
+            full[: len(reqs)] = self._block_table(reqs)
+            kw["block_table"] = full
+        out = self.model(ids, None, pos, seq_ids, self._params_for(reqs), **kw)
         toks = out.logits if out.logits is not None else out.tokens
         if toks.dim() > 2:
             toks = toks.argmax(-1)
@@ -140,6 +200,14 @@ class ServingEngine:
             self.active.pop(req.slot)
             self.free_slots.append(req.slot)
             self.finished[req.req_id] = req
+            if self.paged and req.blocks:
+                # register the prompt's FULL blocks for future prefix hits,
+                # then drop this request's references
+                n_prompt_blocks = req.prompt.shape[0] // self.block_size
+                self.prefix_cache.insert(req.prompt.tolist(),
+                                         req.blocks[:n_prompt_blocks])
+                self.allocator.deref(req.blocks)
+                req.blocks = []
 
     # ------------------------------------------------------------------
     def step(self):

@@ -131,3 +131,59 @@ def test_serving_per_request_sampling_params():
     results = eng.run()
     assert results[greedy] == want[0, 9:].tolist()
     assert len(results[sampled]) == 6
+
+
+@torch.inference_mode()
+def test_serving_paged_with_prefix_caching():
+    """Paged serving: a later request sharing a prompt prefix reuses the
+    cached blocks (tail-only prefill) and still generates exactly the
+    per-request reference."""
+    ec = EngineConfig(
+        batch_size=2, max_batch_size=2, seq_len=64, max_length=64,
+        torch_dtype="float32", tp_degree=1,
+        is_block_kv_layout=True, pa_block_size=8, pa_num_blocks=40)
+    cfg = LlamaInferenceConfig(
+        neuron_config=ec, vocab_size=128, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, rms_norm_eps=1e-5, rope_theta=10000.0,
+        tie_word_embeddings=False)
+    torch.manual_seed(31)
+    m = LlamaForCausalLM(cfg).eval()
+    with torch.no_grad():
+        for p in m.parameters():
+            if p.dim() >= 2:
+                p.normal_(0, 0.1)
+
+    shared = torch.randint(0, 128, (16,))           # 2 full blocks
+    prompts = [
+        torch.cat([shared, torch.randint(0, 128, (5,))]),
+        torch.randint(0, 128, (11,)),
+        torch.cat([shared, torch.randint(0, 128, (3,))]),  # prefix hit
+    ]
+    # references on a LINEAR-cache twin
+    ec2 = EngineConfig(batch_size=2, max_batch_size=2, seq_len=64, max_length=64,
+                       torch_dtype="float32", tp_degree=1)
+    cfg2 = LlamaInferenceConfig(
+        neuron_config=ec2, vocab_size=128, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, rms_norm_eps=1e-5, rope_theta=10000.0,
+        tie_word_embeddings=False)
+    lin = LlamaForCausalLM(cfg2).eval()
+    lin.model.load_state_dict(
+        {k: v for k, v in m.model.state_dict().items()
+         if not k.startswith("kv_mgr.")}, strict=False)
+    wants = []
+    for p in prompts:
+        lin.reset()
+        out = HuggingFaceGenerationAdapter(lin).generate(
+            p.view(1, -1), generation_config=GenerationConfig(max_new_tokens=6))
+        wants.append(out[0, p.shape[0]:].tolist())
+
+    eng = ServingEngine(m, chunk_size=8)
+    assert eng.paged
+    ids = [eng.add_request(p, max_new_tokens=6) for p in prompts]
+    results = eng.run()
+    for rid, want in zip(ids, wants):
+        assert results[rid] == want, (rid, results[rid], want)
+    # the third request must have hit the 2-block shared prefix
+    assert eng.finished[ids[2]].n_cached == 16
