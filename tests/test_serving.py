@@ -187,3 +187,48 @@ def test_serving_paged_with_prefix_caching():
         assert results[rid] == want, (rid, results[rid], want)
     # the third request must have hit the 2-block shared prefix
     assert eng.finished[ids[2]].n_cached == 16
+
+
+@pytest.mark.gpu
+@torch.inference_mode()
+def test_serving_paged_prefix_caching_gpu():
+    """Paged serving + prefix caching on GPU (HIP paged decode kernel)."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    ec = EngineConfig(
+        batch_size=2, max_batch_size=2, seq_len=64, max_length=64,
+        torch_dtype="bfloat16", tp_degree=1,
+        is_block_kv_layout=True, pa_block_size=8, pa_num_blocks=40)
+    # head_dim 64 for the paged HIP decode kernel
+    cfg = LlamaInferenceConfig(
+        neuron_config=ec, vocab_size=128, hidden_size=256, intermediate_size=512,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, rms_norm_eps=1e-5, rope_theta=10000.0,
+        tie_word_embeddings=False)
+    torch.manual_seed(31)
+    m = LlamaForCausalLM(cfg).eval()
+    with torch.no_grad():
+        for p in m.parameters():
+            if p.dim() >= 2:
+                p.normal_(0, 0.1)
+    m = m.to("cuda")
+    shared = torch.randint(0, 128, (16,))
+    prompts = [
+        torch.cat([shared, torch.randint(0, 128, (5,))]),
+        torch.cat([shared, torch.randint(0, 128, (3,))]),
+    ]
+    wants = []
+    for p in prompts:
+        # per-request reference on fresh blocks, same engine machinery
+        e = ServingEngine(m)
+        rid = e.add_request(p, max_new_tokens=6)
+        wants.append(e.run()[rid])
+    # one engine, sequential requests: the second hits the first's prefix
+    eng = ServingEngine(m)
+    r0 = eng.add_request(prompts[0], max_new_tokens=6)
+    eng.run()
+    r1 = eng.add_request(prompts[1], max_new_tokens=6)
+    eng.run()
+    assert eng.finished[r0].output == wants[0]
+    assert eng.finished[r1].output == wants[1]
+    assert eng.finished[r1].n_cached == 16
