@@ -49,7 +49,9 @@ class Request:
 class ServingEngine:
     def __init__(self, causal_lm, max_slots: Optional[int] = None,
                  chunk_size: Optional[int] = None,
-                 eos_token_id: Optional[int] = None):
+                 eos_token_id: Optional[int] = None,
+                 capture_decode: Optional[bool] = None,
+                 on_token=None):
         self.model = causal_lm
         ec = causal_lm.config.neuron_config
         self.batch = ec.batch_size
@@ -74,6 +76,14 @@ class ServingEngine:
         self.queue: Deque[Request] = collections.deque()
         self.finished: Dict[int, Request] = {}
         self._next_id = 0
+        self.on_token = on_token        # callback(req_id, token) per token
+        # hipGraph-captured decode step (paged serving would otherwise run
+        # its full-batch decode eagerly; linear serving uses the model's
+        # own captured wrappers)
+        if capture_decode is None:
+            capture_decode = self.paged and self.device.type == "cuda"
+        self._capture_decode = capture_decode and self.device.type == "cuda"
+        self._decode_graph = None
 
     # ------------------------------------------------------------------
     def add_request(self, prompt_ids: torch.Tensor,
@@ -157,6 +167,8 @@ class ServingEngine:
             req.last_token = tok
             req.pos = S
             req.output.append(tok)
+            if self.on_token is not None:
+                self.on_token(req.req_id, tok)
             self._maybe_finish(req, tok)
 
     def _decode_step(self):
@@ -179,8 +191,12 @@ class ServingEngine:
                               device=self.device)
             full[: len(reqs)] = self._block_table(reqs)
             kw["block_table"] = full
-        out = self.model(ids, None, pos, seq_ids, self._params_for(reqs), **kw)
-        toks = out.logits if out.logits is not None else out.tokens
+        sp = self._params_for(reqs)
+        if self._capture_decode and self.paged:
+            toks = self._run_decode_graph(ids, pos, seq_ids, sp, kw["block_table"])
+        else:
+            out = self.model(ids, None, pos, seq_ids, sp, **kw)
+            toks = out.logits if out.logits is not None else out.tokens
         if toks.dim() > 2:
             toks = toks.argmax(-1)
         toks = toks.view(B, -1)[:, -1].tolist()
@@ -190,7 +206,35 @@ class ServingEngine:
             r.last_token = tok
             r.pos += 1
             r.output.append(tok)
+            if self.on_token is not None:
+                self.on_token(r.req_id, tok)
             self._maybe_finish(r, tok)
+
+    def _run_decode_graph(self, ids, pos, seq_ids, sp, block_table):
+        from neuronx_distributed_inference_amd.runtime.graph_engine import (
+            GraphRunner,
+        )
+
+        dm = self.model.model
+
+        if self._decode_graph is None:
+            if sp is None:
+                def fn(i, p, s, bt):
+                    t, lg = dm(i, None, p, s, None,
+                               phase="token_generation", block_table=bt)
+                    return t if t is not None else lg
+                ex = (ids, pos, seq_ids, block_table)
+            else:
+                def fn(i, p, s, smp, bt):
+                    t, lg = dm(i, None, p, s, smp,
+                               phase="token_generation", block_table=bt)
+                    return t if t is not None else lg
+                ex = (ids, pos, seq_ids, sp, block_table)
+            self._decode_graph = GraphRunner(fn, ex)
+            self._decode_graph_has_sp = sp is not None
+        args = (ids, pos, seq_ids) + ((sp,) if self._decode_graph_has_sp else ()) \
+            + (block_table,)
+        return self._decode_graph(*args).clone()
 
     def _maybe_finish(self, req: Request, tok: int):
         if len(req.output) >= req.max_new_tokens or (

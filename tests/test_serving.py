@@ -34,6 +34,16 @@ def _model(batch):
 
 
 @torch.inference_mode()
+def test_serving_streaming_callback():
+    m = _model(batch=2)
+    seen = []
+    eng = ServingEngine(m, on_token=lambda rid, tok: seen.append((rid, tok)))
+    rid = eng.add_request(torch.randint(0, 128, (6,)), max_new_tokens=4)
+    results = eng.run()
+    assert [t for r, t in seen if r == rid] == results[rid]
+
+
+@torch.inference_mode()
 @pytest.mark.parametrize("chunk", [None, 8])
 def test_serving_engine_matches_reference(chunk):
     m = _model(batch=3)
