@@ -36,6 +36,7 @@ class Request:
     top_k: int = 1
     top_p: float = 1.0
     temperature: float = 1.0
+    adapter_id: int = 0             # multi-LoRA slot (0 = no adapter)
     output: List[int] = field(default_factory=list)
     slot: int = -1
     pos: int = 0                    # next position to fill
@@ -51,7 +52,7 @@ class ServingEngine:
                  chunk_size: Optional[int] = None,
                  eos_token_id: Optional[int] = None,
                  capture_decode: Optional[bool] = None,
-                 on_token=None):
+                 on_token=None, lora_state=None):
         self.model = causal_lm
         ec = causal_lm.config.neuron_config
         self.batch = ec.batch_size
@@ -77,6 +78,7 @@ class ServingEngine:
         self.finished: Dict[int, Request] = {}
         self._next_id = 0
         self.on_token = on_token        # callback(req_id, token) per token
+        self.lora_state = lora_state    # multi-LoRA: per-request adapter rows
         # hipGraph-captured decode step (paged serving would otherwise run
         # its full-batch decode eagerly; linear serving uses the model's
         # own captured wrappers)
@@ -88,14 +90,25 @@ class ServingEngine:
     # ------------------------------------------------------------------
     def add_request(self, prompt_ids: torch.Tensor,
                     max_new_tokens: int = 32, top_k: int = 1,
-                    top_p: float = 1.0, temperature: float = 1.0) -> int:
+                    top_p: float = 1.0, temperature: float = 1.0,
+                    adapter_id: int = 0) -> int:
         # per-request on-device sampling params ride in the (B,3)
-        # sampling_params rows (ref per-request sampling_params)
+        # sampling_params rows; per-request LoRA adapters in the LoraState
+        # adapter_ids rows (ref per-request sampling_params / adapter_ids)
         rid = self._next_id
         self._next_id += 1
         self.queue.append(Request(rid, prompt_ids.view(-1).cpu(),
-                                  max_new_tokens, top_k, top_p, temperature))
+                                  max_new_tokens, top_k, top_p, temperature,
+                                  adapter_id))
         return rid
+
+    def _set_adapters(self, reqs):
+        if self.lora_state is None:
+            return
+        ids = torch.zeros(self.batch, dtype=torch.long)
+        for i, r in enumerate(reqs):
+            ids[i] = r.adapter_id
+        self.lora_state.set(ids)
 
     def _params_for(self, reqs) -> Optional[torch.Tensor]:
         if self.model.config.neuron_config.on_device_sampling_config is None:
@@ -149,6 +162,7 @@ class ServingEngine:
         positions = torch.arange(lo, hi, device=self.device).view(1, -1)
         seq_ids = torch.tensor([req.slot], device=self.device)
         sp = self._params_for([req])
+        self._set_adapters([req])
         kw = {}
         if self.paged:
             self._ensure_blocks(req, hi - 1)
@@ -183,6 +197,7 @@ class ServingEngine:
             pos[i, 0] = r.pos
             seq_ids[i] = slot
         reqs = [self.active[s] for s in slots]
+        self._set_adapters(reqs)
         kw = {}
         if self.paged:
             for r in reqs:
