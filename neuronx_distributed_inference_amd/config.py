@@ -386,6 +386,13 @@ class InferenceConfig:
                 out[k] = v.to_dict() if v else None
             elif isinstance(v, torch.dtype):
                 out[k] = dtype_name(v)
+            elif isinstance(v, InferenceConfig):
+                # nested config (multimodal text_config etc.): serialize as
+                # a dict; the shared engine config is rebuilt on load
+                nested = v.to_dict()
+                nested.pop("neuron_config", None)
+                nested.pop("_config_class", None)
+                out[k] = nested
             else:
                 try:
                     json.dumps(v)

@@ -272,3 +272,43 @@ def test_moe_dense_matches_gather_path():
             he = F.linear(m.act_mul(ge, ue), m.w_down[e])
             out[rows] += he if scale_input else he * ww
         torch.testing.assert_close(dense, out, atol=1e-5, rtol=1e-5)
+
+
+def test_composite_config_json_roundtrip():
+    """Multimodal composite configs (nested text/vision configs) must
+    survive save()/load() — nested InferenceConfig objects serialize as
+    dicts and are reconstructed by add_derived_config."""
+    import tempfile
+
+    from neuronx_distributed_inference_amd.config import (
+        EngineConfig,
+        InferenceConfig,
+    )
+    from neuronx_distributed_inference_amd.models.mllama import (
+        MllamaInferenceConfig,
+    )
+
+    ec = EngineConfig(batch_size=1, max_batch_size=1, seq_len=48, max_length=48,
+                      torch_dtype="float32", tp_degree=1)
+    tc = dict(vocab_size=120, hidden_size=32, intermediate_size=64,
+              num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2,
+              cross_attention_layers=[1, 3], rope_theta=10000.0,
+              max_position_embeddings=64, rms_norm_eps=1e-5,
+              tie_word_embeddings=False)
+    vc = dict(hidden_size=32, intermediate_size=64, num_hidden_layers=3,
+              num_global_layers=1, attention_heads=4, image_size=32,
+              patch_size=16, max_num_tiles=2,
+              supported_aspect_ratios=[[1, 1], [1, 2]],
+              intermediate_layers_indices=[1, 2], vision_output_dim=96)
+    cfg = MllamaInferenceConfig(neuron_config=ec, text_config=tc,
+                                vision_config=vc)
+    with tempfile.TemporaryDirectory() as d:
+        cfg.save(d)
+        back = MllamaInferenceConfig.load(d)
+        assert back.text_config.hidden_size == 32
+        assert back.text_config.cross_attention_layers == [1, 3]
+        assert back.vision_config["max_num_tiles"] == 2
+        assert back.hidden_size == 32
+        # generic load reconstructs the concrete class by dotted path
+        back2 = InferenceConfig.load(d)
+        assert type(back2).__name__ == "MllamaInferenceConfig"
