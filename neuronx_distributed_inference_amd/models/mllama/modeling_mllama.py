@@ -280,14 +280,25 @@ class _MllamaTextCrossAttention(nn.Module):
                                      self.head_dim).transpose(1, 2).contiguous()
         return k, v
 
-    def forward(self, x: torch.Tensor, k: torch.Tensor, v: torch.Tensor):
+    def forward(self, x: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                mask: Optional[torch.Tensor] = None):
         B, S, _ = x.shape
         q = self.q_proj(x).view(B, S, self.heads_per_rank, self.head_dim)
         q = self.q_norm(q).transpose(1, 2)
-        # every text token attends the full vision sequence: constant
-        # full-length per-query positions through the HIP decode kernel
-        pos = torch.full((B, S), k.shape[2] - 1, device=x.device, dtype=torch.int32)
-        a = ops.attention_decode(q, k, v, pos, scale=self.scale)
+        if mask is not None:
+            # per-token tile mask (additive, (B,1,S,Sv)): eager masked path —
+            # Sv is a few thousand vision tokens, so this is tiny at decode
+            g = self.heads_per_rank // self.kv_heads_per_rank
+            kk = k.repeat_interleave(g, dim=1) if g > 1 else k
+            vv = v.repeat_interleave(g, dim=1) if g > 1 else v
+            s = (q @ kk.transpose(-1, -2)) * self.scale + mask
+            a = s.float().softmax(-1).to(q.dtype) @ vv
+        else:
+            # all-attend: constant full-length per-query positions through
+            # the HIP decode kernel
+            pos = torch.full((B, S), k.shape[2] - 1, device=x.device,
+                             dtype=torch.int32)
+            a = ops.attention_decode(q, k, v, pos, scale=self.scale)
         return self.o_proj(a.transpose(1, 2).reshape(B, S, -1))
 
 

@@ -16,16 +16,16 @@ P3=$(pick SQ_BUSY_CYCLES; pick SQ_WAVE_CYCLES; pick SQ_INSTS_VALU)
 echo "P1=[$P1] P2=[$P2] P3=[$P3]" | tee "$O/picked.txt"
 
 for m in skinny skinny2 lib lt; do
-  timeout 150 rocprofv3 --kernel-trace --stats -d "$O/stats_$m" -o s \
+  timeout 150 rocprofv3 --output-format csv --kernel-trace --stats -d "$O/stats_$m" -o s \
     -- python "$R/scripts/prof_skinny_tcc.py" $m > "$O/stats_$m.log" 2>&1 || true
 done
 for m in skinny2 lib; do
-  [ -n "$P1" ] && timeout 150 rocprofv3 --pmc $P1 -d "$O/p1_$m" -o p \
+  [ -n "$P1" ] && timeout 150 rocprofv3 --output-format csv --pmc $P1 -d "$O/p1_$m" -o p \
     -- python "$R/scripts/prof_skinny_tcc.py" $m > "$O/p1_$m.log" 2>&1 || true
-  [ -n "$P2" ] && timeout 150 rocprofv3 --pmc $P2 -d "$O/p2_$m" -o p \
+  [ -n "$P2" ] && timeout 150 rocprofv3 --output-format csv --pmc $P2 -d "$O/p2_$m" -o p \
     -- python "$R/scripts/prof_skinny_tcc.py" $m > "$O/p2_$m.log" 2>&1 || true
 done
-[ -n "$P3" ] && timeout 150 rocprofv3 --pmc $P3 -d "$O/p3_skinny2" -o p \
+[ -n "$P3" ] && timeout 150 rocprofv3 --output-format csv --pmc $P3 -d "$O/p3_skinny2" -o p \
   -- python "$R/scripts/prof_skinny_tcc.py" skinny2 > "$O/p3_skinny2.log" 2>&1 || true
 
 # compact: keep only csv summaries, drop anything big
