@@ -16,9 +16,13 @@ stack). Architecture matched against HF `modeling_mllama.py`:
   full-length positions, like Whisper's cross attention).
 - text-only requests skip the cross layers entirely (HF behavior).
 
-Simplification vs HF: `cross_attention_mask` is not yet supported — every
-text row attends every vision token of its own batch row (the common
-single-image chat case); per-token tile masks are a GAPS.md item.
+`cross_attention_mask` (per-token tile masks) is supported with HF-exact
+semantics: the (B, T, n_images, n_tiles) 0/1 mask expands to an additive
+vision-token mask, fully-masked rows fall back to uniform attention, and
+the full-row mask zeroes those rows' cross-MLP contribution; generated
+tokens past the prompt reuse the last mask row (HF generation repeats
+it per new token). Without a mask every text row attends every vision
+token via the HIP decode kernel (the common single-image chat case).
 """
 
 from __future__ import annotations
@@ -246,6 +250,24 @@ class MllamaVisionModel(nn.Module):
 
 # ----------------------------------------------------------------------
 # text side
+def _prepare_cross_attention_mask(cross_attention_mask: torch.Tensor,
+                                  num_vision_tokens: int, dtype: torch.dtype):
+    """(B, T, n_images, n_tiles) 0/1 tile mask -> additive (B,1,T,Sv) mask +
+    (B,1,T,1) full-row mask, HF semantics (transformers mllama
+    `_prepare_cross_attention_mask`): masked entries get finfo.min, and rows
+    whose EVERY vision token is masked are flipped back to all-zero
+    (uniform attention, no NaN) while the row mask later zeroes their MLP
+    contribution."""
+    B, T = cross_attention_mask.shape[:2]
+    m = cross_attention_mask.repeat_interleave(num_vision_tokens, dim=3)
+    m = m.view(B, T, -1).unsqueeze(1).to(dtype)
+    inv = 1.0 - m
+    neg = torch.finfo(dtype).min
+    m = inv.masked_fill(inv.bool(), neg)
+    row = (m != neg).any(dim=-1).type_as(m)[..., None]
+    return m * row, row
+
+
 class _MllamaTextCrossAttention(nn.Module):
     """The projection/norm block, named to mirror the HF checkpoint
     (layers.N.cross_attn.{q,k,v,o}_proj / q_norm / k_norm)."""
@@ -337,10 +359,25 @@ class MllamaCrossAttentionLayer(nn.Module):
         if phase == "context_encoding" or self._kv is None:
             self._kv = self.cross_attn.project_kv(states)
         k, v = self._kv
-        a = self.cross_attn(x, k, v)
+        mask, rowq = self.ctx.get("mask"), None
+        if mask is not None:
+            # gather this step's rows by text position; positions past the
+            # prompt reuse the LAST row (HF generation repeats it per new
+            # token; ref mllama _update_model_kwargs cross_attention_mask cat)
+            B, S = position_ids.shape
+            Sv, T = mask.shape[-1], mask.shape[2]
+            pos = position_ids.clamp(min=0, max=T - 1).long()
+            mask = mask.gather(2, pos[:, None, :, None].expand(B, 1, S, Sv))
+            rowq = self.ctx["row_mask"].gather(2, pos[:, None, :, None])
+        a = self.cross_attn(x, k, v, mask)
         h = self.cross_attn_attn_gate.tanh() * a
         x, residual = self.post_attention_layernorm(h, residual)
-        h = self.cross_attn_mlp_gate.tanh() * self.mlp(x)
+        out = self.mlp(x)
+        if rowq is not None:
+            # rows whose every vision token is masked contribute no MLP
+            # update either (HF full_text_row_masked_out_mask semantics)
+            out = rowq[:, 0] * out
+        h = self.cross_attn_mlp_gate.tanh() * out
         return h, residual
 
 
@@ -400,6 +437,7 @@ class MllamaForConditionalGeneration(nn.Module):
     @torch.inference_mode()
     def generate(self, input_ids: torch.Tensor, pixel_values=None,
                  aspect_ratio_ids=None, aspect_ratio_mask=None,
+                 cross_attention_mask=None,
                  attention_mask=None, max_new_tokens: int = 16, **kw):
         from neuronx_distributed_inference_amd.utils.hf_adapter import (
             GenerationConfig,
@@ -412,9 +450,14 @@ class MllamaForConditionalGeneration(nn.Module):
                 layer._kv = None
         if pixel_values is not None:
             states = self.encode_images(pixel_values, aspect_ratio_ids, aspect_ratio_mask)
-            self.language_model.model.cross_ctx["states"] = states.to(
-                self.language_model.model.embed_tokens.weight.dtype
-            )
+            dtype = self.language_model.model.embed_tokens.weight.dtype
+            self.language_model.model.cross_ctx["states"] = states.to(dtype)
+            if cross_attention_mask is not None:
+                m, row = _prepare_cross_attention_mask(
+                    cross_attention_mask, self.vision_model.num_patches, dtype
+                )
+                self.language_model.model.cross_ctx["mask"] = m
+                self.language_model.model.cross_ctx["row_mask"] = row
         adapter = HuggingFaceGenerationAdapter(self.language_model)
         return adapter.generate(
             input_ids, attention_mask=attention_mask,

@@ -24,8 +24,9 @@ for i in range(40):
     w = ws[i % 10]
     if mode == "skinny":
         C.skinny_gemm(y, x, w, None)
-    elif mode == "skinny2":
-        C.skinny_gemm2(y, y32, x, w, None, 8)
+    elif mode.startswith("skinny2"):
+        ks = int(mode[7:] or 8)
+        C.skinny_gemm2(y, y32, x, w, None, ks)
     elif mode == "lt":
         C.lt_linear(y, x, w, None)
     else:

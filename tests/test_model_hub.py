@@ -384,6 +384,24 @@ def test_mllama():
     got = ours.generate(ids, max_new_tokens=4)
     assert torch.equal(got[:, : ref_ids.shape[1]], ref_ids)
 
+    # per-token tile masks (cross_attention_mask), incl. a fully-masked row
+    # and a partially-masked tile; HF repeats the last row per new token
+    xm = torch.ones(1, 6, 1, 2, dtype=torch.long)
+    xm[0, 0] = 0          # row 0: no vision at all (full-row masked out)
+    xm[0, 2, 0, 1] = 0    # row 2: second tile hidden
+    ref_ids, ref_xm = ids.clone(), xm.clone()
+    for _ in range(6):
+        lg = hf(input_ids=ref_ids, pixel_values=px, aspect_ratio_ids=ar_ids,
+                aspect_ratio_mask=ar_mask, cross_attention_mask=ref_xm,
+                use_cache=False).logits
+        ref_ids = torch.cat([ref_ids, lg[:, -1].argmax(-1, keepdim=True)], dim=1)
+        ref_xm = torch.cat([ref_xm, ref_xm[:, -1:]], dim=1)
+    got = ours.generate(ids, pixel_values=px, aspect_ratio_ids=ar_ids,
+                        aspect_ratio_mask=ar_mask, cross_attention_mask=xm,
+                        max_new_tokens=6)
+    assert torch.equal(got[:, : ref_ids.shape[1]], ref_ids), (
+        got.tolist(), ref_ids.tolist())
+
 
 @torch.inference_mode()
 def test_pixtral():
