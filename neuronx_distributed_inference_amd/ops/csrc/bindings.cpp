@@ -35,6 +35,8 @@ void skinny_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                  c10::optional<torch::Tensor> bias);
 void skinny_gemm2(torch::Tensor y, torch::Tensor y32, torch::Tensor x,
                   torch::Tensor w, c10::optional<torch::Tensor> bias, long ksplit);
+void skinny_gemm3(torch::Tensor y, torch::Tensor x, torch::Tensor w,
+                  c10::optional<torch::Tensor> bias);
 void fp8_quant(torch::Tensor x, torch::Tensor out, torch::Tensor scale,
                torch::Tensor amax_ws);
 void fp8_epilogue(torch::Tensor out, torch::Tensor y32, torch::Tensor w_scale,
@@ -72,6 +74,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("sinks") = py::none());
     m.def("qkv_rope_update", &qkv_rope_update, "fused qkv->rope->cache-write (CDNA4)");
     m.def("skinny_gemm", &skinny_gemm, "skinny-M weight-streaming GEMM (CDNA4)",
+          py::arg("y"), py::arg("x"), py::arg("w"), py::arg("bias") = py::none());
+    m.def("skinny_gemm3", &skinny_gemm3,
+          "skinny-M GEMM v3: full-K register accum, depth-3 prefetch (CDNA4)",
           py::arg("y"), py::arg("x"), py::arg("w"), py::arg("bias") = py::none());
     m.def("skinny_gemm2", &skinny_gemm2, "blocked skinny-M GEMM, K-split (CDNA4)",
           py::arg("y"), py::arg("y32"), py::arg("x"), py::arg("w"),

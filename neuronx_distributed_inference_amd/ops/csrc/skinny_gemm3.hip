@@ -1,0 +1,188 @@
+// Skinny-M weight-streaming GEMM v3: y = x @ W^T (+bias), M <= 128.
+//
+// Built from the round-1 TCC counter evidence
+// (profiles/r01_skinny_tcc_analysis.md): v2's plateau was (a) atomic
+// K-split partials adding 44x the library's EA write traffic and (b) a
+// depth-1 prefetch pipeline (one LDS chunk in flight) leaving each WG's
+// stream latency-bound at ~4.4 GB/s. v3 removes both:
+//
+// - NO K-split: each WG register-accumulates its FULL K span and writes
+//   bf16 y exactly once (no fp32 workspace, no convert kernel, no RMW).
+//   Grid depth comes from the N split x M split instead:
+//   (N/32) x ceil(M/64) = 384 WGs for the qkv shape (M=128, N=6144).
+// - Depth-3 register prefetch: three staged chunks in flight over a
+//   single LDS buffer (~9 outstanding b128 loads per lane vs v2's 3).
+// - Weight loads are nontemporal (single-use stream; keeps L2 for x,
+//   which every WG re-reads and which stays XCD-L2-resident at 1 MB).
+//
+// Same LDS pitch/fragment math as v2 (9-slot b128 pitch, conflict-free
+// mfma_16x16x32 fragment reads). bf16/f16; N % 32 == 0, K % 64 == 0.
+//
+// Reference parity: this replaces the reference's fused decode-GEMM
+// tiles; the engine wires it only if it beats the
+// autotuned hipBLASLt path on the captured shape.
+
+#include "common.h"
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+namespace {
+
+typedef __bf16 sk3_bf16x8 __attribute__((ext_vector_type(8)));
+typedef _Float16 sk3_f16x8 __attribute__((ext_vector_type(8)));
+typedef float sk3_f32x4 __attribute__((ext_vector_type(4)));
+typedef int sk3_i32x4 __attribute__((ext_vector_type(4)));
+
+template <typename T> struct Sk3Mfma;
+template <> struct Sk3Mfma<__hip_bfloat16> {
+    using vec8 = sk3_bf16x8;
+    static __device__ __forceinline__ sk3_f32x4 mfma(vec8 a, vec8 b, sk3_f32x4 c) {
+        return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+    }
+    static __device__ __forceinline__ float to_f(__hip_bfloat16 v) {
+        return __bfloat162float(v);
+    }
+    static __device__ __forceinline__ __hip_bfloat16 from_f(float v) {
+        return __float2bfloat16(v);
+    }
+};
+template <> struct Sk3Mfma<__half> {
+    using vec8 = sk3_f16x8;
+    static __device__ __forceinline__ sk3_f32x4 mfma(vec8 a, vec8 b, sk3_f32x4 c) {
+        return __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b, c, 0, 0, 0);
+    }
+    static __device__ __forceinline__ float to_f(__half v) { return __half2float(v); }
+    static __device__ __forceinline__ __half from_f(float v) { return __float2half(v); }
+};
+
+constexpr int KC3 = 64;                // K elems per LDS chunk
+constexpr int PITCH3 = KC3 + 8;        // 9 b128 slots: conflict-free frags
+constexpr int NT3 = 32;                // W rows per WG
+constexpr int MT3 = 64;                // x rows per WG (4 waves x 16)
+
+// 4 waves each own one 16-row m-tile and loop the 2 n-tiles.
+template <typename T>
+__launch_bounds__(256, 4)
+__global__ void skinny_gemm3_kernel(
+    T* __restrict__ y,            // (M, N)
+    const T* __restrict__ x,      // (M, K)
+    const T* __restrict__ w,      // (N, K)
+    const T* __restrict__ bias,   // (N,) or nullptr
+    const int M, const int N, const int K) {
+    using MF = Sk3Mfma<T>;
+    using vec8 = typename MF::vec8;
+
+    const int n_base = blockIdx.x * NT3;
+    const int m_base = blockIdx.y * MT3;
+    const int wave = threadIdx.x >> 6;
+    const int lane = threadIdx.x & 63;
+    const int col = lane & 15;
+    const int kseg = lane >> 4;
+
+    __shared__ T wt[NT3 * PITCH3];
+    __shared__ T xt[MT3 * PITCH3];
+
+    // staging maps (one chunk = W 32x64 = 4 KB + x 64x64 = 8 KB):
+    // W: 8 threads per 128 B row -> 1 x b128 per thread
+    const int ww_row = threadIdx.x >> 3;
+    const int ww_off = (threadIdx.x & 7) * 8;   // elems
+    // x: 4 threads per 128 B row -> 2 x b128 per thread
+    const int xx_row = threadIdx.x >> 2;
+    const int xx_off = (threadIdx.x & 3) * 16;  // elems
+
+    struct Stage { sk3_i32x4 wr; sk3_i32x4 xr[2]; };
+    auto load_stage = [&](int k0, Stage& s) {
+        s.wr = __builtin_nontemporal_load(reinterpret_cast<const sk3_i32x4*>(
+            w + (long)(n_base + ww_row) * K + k0 + ww_off));
+        const int xrow = min(m_base + xx_row, M - 1);
+#pragma unroll
+        for (int p = 0; p < 2; ++p)
+            s.xr[p] = *reinterpret_cast<const sk3_i32x4*>(
+                x + (long)xrow * K + k0 + xx_off + p * 8);
+    };
+    auto commit_stage = [&](const Stage& s) {
+        *reinterpret_cast<sk3_i32x4*>(wt + ww_row * PITCH3 + ww_off) = s.wr;
+#pragma unroll
+        for (int p = 0; p < 2; ++p)
+            *reinterpret_cast<sk3_i32x4*>(xt + xx_row * PITCH3 + xx_off + p * 8) = s.xr[p];
+    };
+
+    sk3_f32x4 acc[2] = {sk3_f32x4{0.f, 0.f, 0.f, 0.f}, sk3_f32x4{0.f, 0.f, 0.f, 0.f}};
+
+    const int nchunks = K / KC3;
+    Stage st[3];
+    load_stage(0, st[0]);
+    if (nchunks > 1) load_stage(KC3, st[1]);
+    if (nchunks > 2) load_stage(2 * KC3, st[2]);
+    commit_stage(st[0]);
+
+    for (int i = 0; i < nchunks; ++i) {
+        __syncthreads();                 // chunk i visible in LDS
+#pragma unroll
+        for (int ks = 0; ks < KC3 / 32; ++ks) {
+            const vec8 afrag = *reinterpret_cast<const vec8*>(
+                xt + (wave * 16 + col) * PITCH3 + ks * 32 + kseg * 8);
+#pragma unroll
+            for (int n = 0; n < 2; ++n) {
+                const vec8 bfrag = *reinterpret_cast<const vec8*>(
+                    wt + (n * 16 + col) * PITCH3 + ks * 32 + kseg * 8);
+                acc[n] = MF::mfma(afrag, bfrag, acc[n]);
+            }
+        }
+        __syncthreads();                 // LDS reads done before restage
+        if (i + 1 < nchunks) {
+            commit_stage(st[(i + 1) % 3]);        // chunk i+1 -> LDS
+            if (i + 3 < nchunks) load_stage((i + 3) * KC3, st[i % 3]);
+        }
+    }
+
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int m = m_base + wave * 16 + kseg * 4 + r;
+            if (m >= M) continue;
+            const int nn = n_base + n * 16 + col;
+            float v = acc[n][r];
+            if (bias != nullptr) v += MF::to_f(bias[nn]);
+            y[(long)m * N + nn] = MF::from_f(v);
+        }
+    }
+}
+
+template <typename T>
+void launch_sk3(torch::Tensor& y, const torch::Tensor& x, const torch::Tensor& w,
+                const torch::Tensor* bias) {
+    const int M = x.size(0), K = x.size(1), N = w.size(0);
+    auto stream = at::hip::getCurrentHIPStream();
+    dim3 grid(N / NT3, (M + MT3 - 1) / MT3);
+    hipLaunchKernelGGL((skinny_gemm3_kernel<T>), grid, dim3(256), 0, stream,
+                       (T*)y.data_ptr(), (const T*)x.data_ptr(),
+                       (const T*)w.data_ptr(),
+                       bias ? (const T*)bias->data_ptr() : nullptr, M, N, K);
+    HIP_CHECK_KERNEL();
+}
+
+}  // namespace
+
+void skinny_gemm3(torch::Tensor y, torch::Tensor x, torch::Tensor w,
+                  c10::optional<torch::Tensor> bias_opt) {
+    const int M = x.size(0), K = x.size(1), N = w.size(0);
+    TORCH_CHECK(M <= 128 && N % NT3 == 0 && K % KC3 == 0 && K >= 3 * KC3,
+                "skinny_gemm3: M<=128, N%32, K%64, K>=192");
+    TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && y.is_contiguous());
+    torch::Tensor bias_t;
+    const torch::Tensor* bias = nullptr;
+    if (bias_opt.has_value()) {
+        bias_t = bias_opt->contiguous();
+        bias = &bias_t;
+    }
+    if (x.scalar_type() == at::ScalarType::BFloat16) {
+        launch_sk3<__hip_bfloat16>(y, x, w, bias);
+    } else if (x.scalar_type() == at::ScalarType::Half) {
+        launch_sk3<__half>(y, x, w, bias);
+    } else {
+        TORCH_CHECK(false, "skinny_gemm3: bf16/f16 only");
+    }
+}

@@ -27,6 +27,8 @@ for i in range(40):
     elif mode.startswith("skinny2"):
         ks = int(mode[7:] or 8)
         C.skinny_gemm2(y, y32, x, w, None, ks)
+    elif mode == "skinny3":
+        C.skinny_gemm3(y, x, w, None)
     elif mode == "lt":
         C.lt_linear(y, x, w, None)
     else:
