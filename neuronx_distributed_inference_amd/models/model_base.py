@@ -201,9 +201,15 @@ class DecoderModel(nn.Module):
         # features replace the image-placeholder token embeddings. The mask
         # derives from input_ids, so bucket padding is handled for free.
         mm = getattr(self, "multimodal_ctx", None)
-        if mm and mm.get("features") is not None and phase == PHASE_CTE and not sp_active:
-            mask = (input_ids == mm["token_id"]).unsqueeze(-1)
-            h = h.masked_scatter(mask, mm["features"].to(h.dtype).reshape(-1, h.shape[-1]))
+        if mm and phase == PHASE_CTE and not sp_active:
+            # one splice per modality: [(placeholder_token_id, features), ...]
+            # (single image/videos: qwen2-vl passes both in one prefill)
+            splices = mm.get("splices")
+            if splices is None and mm.get("features") is not None:
+                splices = [(mm["token_id"], mm["features"])]
+            for tok, feats in splices or []:
+                mask = (input_ids == tok).unsqueeze(-1)
+                h = h.masked_scatter(mask, feats.to(h.dtype).reshape(-1, h.shape[-1]))
         # models with non-1D rope (qwen2-vl m-rope) install a custom hook
         cos_sin_fn = getattr(self, "_cos_sin_fn", None) or self.rotary.get_cos_sin
         cos, sin = cos_sin_fn(position_ids, device=h.device)

@@ -189,32 +189,43 @@ class Qwen2VLForConditionalGeneration(nn.Module):
 
     # ------------------------------------------------------------------
     def _compute_rope_index(self, input_ids: torch.Tensor,
-                            grid_thw: Optional[torch.Tensor]):
-        """3D (t,h,w) positions for a mixed text+image prefill and the
-        decode delta (HF get_rope_index behavior, images only)."""
+                            grid_thw: Optional[torch.Tensor],
+                            video_grid_thw: Optional[torch.Tensor] = None):
+        """3D (t,h,w) positions for a mixed text+image/video prefill and
+        the decode delta (HF get_rope_index behavior). Video segments
+        advance the temporal axis per frame; the next text position jumps
+        past max(t, lh, lw)."""
         B, S = input_ids.shape
         image_tok = self.config.image_token_id
+        video_tok = getattr(self.config, "video_token_id", None)
         pos3 = torch.zeros(3, B, S, dtype=torch.long)
         deltas = torch.zeros(B, dtype=torch.long)
         for b in range(B):
             ids = input_ids[b].tolist()
-            img_i = 0
+            img_i = vid_i = 0
             cur = 0  # next text position
             i = 0
             m = self.visual.spatial_merge_size
             while i < S:
-                if ids[i] == image_tok:
-                    t, h, w = grid_thw[img_i].tolist()
+                is_img = ids[i] == image_tok
+                is_vid = video_tok is not None and ids[i] == video_tok
+                if is_img or is_vid:
+                    if is_img:
+                        t, h, w = grid_thw[img_i].tolist()
+                        img_i += 1
+                    else:
+                        t, h, w = video_grid_thw[vid_i].tolist()
+                        vid_i += 1
                     lh, lw = h // m, w // m
                     n = t * lh * lw
+                    ti = torch.arange(t).view(-1, 1).expand(t, lh * lw).reshape(-1)
                     r = torch.arange(lh).view(-1, 1).expand(lh, lw).reshape(-1)
                     c = torch.arange(lw).view(1, -1).expand(lh, lw).reshape(-1)
-                    pos3[0, b, i:i + n] = cur
+                    pos3[0, b, i:i + n] = cur + ti
                     pos3[1, b, i:i + n] = cur + r.repeat(t)
                     pos3[2, b, i:i + n] = cur + c.repeat(t)
-                    cur = cur + max(lh, lw)
+                    cur = cur + max(t, lh, lw)
                     i += n
-                    img_i += 1
                 else:
                     pos3[:, b, i] = cur
                     cur += 1
@@ -254,7 +265,8 @@ class Qwen2VLForConditionalGeneration(nn.Module):
     # ------------------------------------------------------------------
     @torch.inference_mode()
     def generate(self, input_ids: torch.Tensor, pixel_values=None,
-                 image_grid_thw=None, attention_mask=None,
+                 image_grid_thw=None, pixel_values_videos=None,
+                 video_grid_thw=None, attention_mask=None,
                  max_new_tokens: int = 16, **kw):
         from neuronx_distributed_inference_amd.utils.hf_adapter import (
             GenerationConfig,
@@ -264,12 +276,17 @@ class Qwen2VLForConditionalGeneration(nn.Module):
         dm = self.language_model.model
         dm.multimodal_ctx = {}
         self._mrope_ctx = {}
+        splices = []
         if pixel_values is not None:
-            feats = self.visual(pixel_values, image_grid_thw)
-            dm.multimodal_ctx = {
-                "features": feats, "token_id": self.config.image_token_id,
-            }
-            pos3, deltas = self._compute_rope_index(input_ids, image_grid_thw)
+            splices.append((self.config.image_token_id,
+                            self.visual(pixel_values, image_grid_thw)))
+        if pixel_values_videos is not None:
+            splices.append((self.config.video_token_id,
+                            self.visual(pixel_values_videos, video_grid_thw)))
+        if splices:
+            dm.multimodal_ctx = {"splices": splices}
+            pos3, deltas = self._compute_rope_index(
+                input_ids, image_grid_thw, video_grid_thw)
             self._mrope_ctx = {"pos3d": pos3, "delta": deltas}
         adapter = HuggingFaceGenerationAdapter(self.language_model)
         return adapter.generate(

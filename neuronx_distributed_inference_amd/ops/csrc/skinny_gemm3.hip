@@ -111,13 +111,18 @@ __global__ void skinny_gemm3_kernel(
     sk3_f32x4 acc[2] = {sk3_f32x4{0.f, 0.f, 0.f, 0.f}, sk3_f32x4{0.f, 0.f, 0.f, 0.f}};
 
     const int nchunks = K / KC3;
-    Stage st[3];
-    load_stage(0, st[0]);
-    if (nchunks > 1) load_stage(KC3, st[1]);
-    if (nchunks > 2) load_stage(2 * KC3, st[2]);
-    commit_stage(st[0]);
+    // three NAMED stages, modulo schedule unrolled x3: dynamic indexing of
+    // a Stage array would demote it to scratch (160 B/lane measured — every
+    // staged byte took a global->scratch->LDS round trip)
+    Stage st0, st1, st2;
+    load_stage(0, st0);
+    load_stage(KC3, st1);
+    load_stage(2 * KC3, st2);
+    commit_stage(st0);
 
-    for (int i = 0; i < nchunks; ++i) {
+    auto iter_body = [&](int i, Stage& reuse, Stage& next) {
+        // consume chunk i from LDS; commit chunk i+1 (in `next`); refill
+        // `reuse` (held chunk i, already in LDS) with chunk i+3
         __syncthreads();                 // chunk i visible in LDS
 #pragma unroll
         for (int ks = 0; ks < KC3 / 32; ++ks) {
@@ -132,9 +137,14 @@ __global__ void skinny_gemm3_kernel(
         }
         __syncthreads();                 // LDS reads done before restage
         if (i + 1 < nchunks) {
-            commit_stage(st[(i + 1) % 3]);        // chunk i+1 -> LDS
-            if (i + 3 < nchunks) load_stage((i + 3) * KC3, st[i % 3]);
+            commit_stage(next);          // chunk i+1 -> LDS
+            if (i + 3 < nchunks) load_stage((i + 3) * KC3, reuse);
         }
+    };
+    for (int i = 0; i < nchunks; i += 3) {
+        iter_body(i, st0, st1);
+        if (i + 1 < nchunks) iter_body(i + 1, st1, st2);
+        if (i + 2 < nchunks) iter_body(i + 2, st2, st0);
     }
 
 #pragma unroll
