@@ -111,18 +111,23 @@ __global__ void skinny_gemm3_kernel(
     sk3_f32x4 acc[2] = {sk3_f32x4{0.f, 0.f, 0.f, 0.f}, sk3_f32x4{0.f, 0.f, 0.f, 0.f}};
 
     const int nchunks = K / KC3;
-    // three NAMED stages, modulo schedule unrolled x3: dynamic indexing of
-    // a Stage array would demote it to scratch (160 B/lane measured — every
-    // staged byte took a global->scratch->LDS round trip)
-    Stage st0, st1, st2;
+    // six NAMED stages (~18 b128 loads in flight per lane: the chunk's
+    // 1400-cycle HBM latency needs ~5 iterations of cover), modulo
+    // schedule unrolled x6: dynamic indexing of a Stage array would
+    // demote it to scratch (160 B/lane measured at Q=3 — every staged
+    // byte took a global->scratch->LDS round trip)
+    Stage st0, st1, st2, st3, st4, st5;
     load_stage(0, st0);
     load_stage(KC3, st1);
     load_stage(2 * KC3, st2);
+    load_stage(3 * KC3, st3);
+    load_stage(4 * KC3, st4);
+    load_stage(5 * KC3, st5);
     commit_stage(st0);
 
     auto iter_body = [&](int i, Stage& reuse, Stage& next) {
         // consume chunk i from LDS; commit chunk i+1 (in `next`); refill
-        // `reuse` (held chunk i, already in LDS) with chunk i+3
+        // `reuse` (held chunk i, already in LDS) with chunk i+6
         __syncthreads();                 // chunk i visible in LDS
 #pragma unroll
         for (int ks = 0; ks < KC3 / 32; ++ks) {
@@ -138,13 +143,16 @@ __global__ void skinny_gemm3_kernel(
         __syncthreads();                 // LDS reads done before restage
         if (i + 1 < nchunks) {
             commit_stage(next);          // chunk i+1 -> LDS
-            if (i + 3 < nchunks) load_stage((i + 3) * KC3, reuse);
+            if (i + 6 < nchunks) load_stage((i + 6) * KC3, reuse);
         }
     };
-    for (int i = 0; i < nchunks; i += 3) {
+    for (int i = 0; i < nchunks; i += 6) {
         iter_body(i, st0, st1);
         if (i + 1 < nchunks) iter_body(i + 1, st1, st2);
-        if (i + 2 < nchunks) iter_body(i + 2, st2, st0);
+        if (i + 2 < nchunks) iter_body(i + 2, st2, st3);
+        if (i + 3 < nchunks) iter_body(i + 3, st3, st4);
+        if (i + 4 < nchunks) iter_body(i + 4, st4, st5);
+        if (i + 5 < nchunks) iter_body(i + 5, st5, st0);
     }
 
 #pragma unroll
@@ -179,8 +187,8 @@ void launch_sk3(torch::Tensor& y, const torch::Tensor& x, const torch::Tensor& w
 void skinny_gemm3(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                   c10::optional<torch::Tensor> bias_opt) {
     const int M = x.size(0), K = x.size(1), N = w.size(0);
-    TORCH_CHECK(M <= 128 && N % NT3 == 0 && K % KC3 == 0 && K >= 3 * KC3,
-                "skinny_gemm3: M<=128, N%32, K%64, K>=192");
+    TORCH_CHECK(M <= 128 && N % NT3 == 0 && K % KC3 == 0 && K >= 6 * KC3,
+                "skinny_gemm3: M<=128, N%32, K%64, K>=384");
     TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && y.is_contiguous());
     torch::Tensor bias_t;
     const torch::Tensor* bias = nullptr;

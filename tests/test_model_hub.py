@@ -489,7 +489,7 @@ def test_qwen2_vl():
                       torch_dtype="float32", tp_degree=1)
     cfg = Qwen2VLInferenceConfig(
         neuron_config=ec, text_config=dict(tc), vision_config=dict(vc),
-        image_token_id=159)
+        image_token_id=159, video_token_id=156)
     ours = Qwen2VLForConditionalGeneration(cfg).eval()
     sd = Qwen2VLForConditionalGeneration.convert_hf_to_native_state_dict(
         hf.state_dict(), cfg)
@@ -523,6 +523,31 @@ def test_qwen2_vl():
         ref_ids = torch.cat([ref_ids, lg[:, -1].argmax(-1, keepdim=True)], dim=1)
     got = ours.generate(ids2, max_new_tokens=4)
     assert torch.equal(got[:, : ref_ids.shape[1]], ref_ids)
+
+    # video (t=2): temporal m-rope axis advances per frame; 2*4*4 patches
+    # -> 8 merged video placeholders, plus an image in the same prompt
+    vgrid = torch.tensor([[2, 4, 4]])
+    vpx = torch.randn(32, 3 * 2 * 4 * 4)
+    ids3 = torch.randint(0, 156, (1, 20))
+    ids3[0, 2:10] = 156                   # 8 video placeholders
+    ids3[0, 12:16] = 159                  # 4 image placeholders
+    mm_tt = torch.zeros_like(ids3)
+    mm_tt[ids3 == 159] = 1
+    mm_tt[ids3 == 156] = 2
+    ref_ids = ids3.clone()
+    for _ in range(6):
+        lg = hf(input_ids=ref_ids, pixel_values=px, image_grid_thw=grid,
+                pixel_values_videos=vpx, video_grid_thw=vgrid,
+                mm_token_type_ids=torch.cat(
+                    [mm_tt, torch.zeros(1, ref_ids.shape[1] - 20,
+                                        dtype=mm_tt.dtype)], dim=1),
+                use_cache=False).logits
+        ref_ids = torch.cat([ref_ids, lg[:, -1].argmax(-1, keepdim=True)], dim=1)
+    got = ours.generate(ids3, pixel_values=px, image_grid_thw=grid,
+                        pixel_values_videos=vpx, video_grid_thw=vgrid,
+                        max_new_tokens=6)
+    assert torch.equal(got[:, : ref_ids.shape[1]], ref_ids), (
+        got.tolist(), ref_ids.tolist())
 
 
 @torch.inference_mode()
