@@ -221,29 +221,48 @@ class Qwen3VLForConditionalGeneration(nn.Module):
 
     def _make_deepstack_fn(self, li: int):
         def fn(h, input_ids, phase):
+            # [(placeholder_token_id, per-level features), ...] — one entry
+            # per modality (image / video), injected at its own positions
             ds = self._mrope_ctx.get("deepstack")
             if ds is None or phase != "context_encoding":
                 return h
-            mask = (input_ids == self.config.image_token_id).unsqueeze(-1)
-            add = torch.zeros_like(h).masked_scatter(
-                mask, ds[li].to(h.dtype).reshape(-1, h.shape[-1]))
-            return h + add
+            for tok, levels in ds:
+                mask = (input_ids == tok).unsqueeze(-1)
+                add = torch.zeros_like(h).masked_scatter(
+                    mask, levels[li].to(h.dtype).reshape(-1, h.shape[-1]))
+                h = h + add
+            return h
         return fn
 
-    # -- rope index (images only): text +1; vision block advances by
-    #    max(lh, lw) with (t=cur, h=cur+row, w=cur+col)
-    def _compute_rope_index(self, input_ids, grid_thw):
+    # -- rope index: text +1; vision block advances by max(lh, lw) with
+    #    (t=cur, h=cur+row, w=cur+col). Videos are split into per-frame
+    #    (1, h, w) grids (HF qwen3-vl: timestamp text tokens separate the
+    #    frames, so each frame is its own vision segment).
+    def _compute_rope_index(self, input_ids, grid_thw,
+                            video_grid_thw=None):
         B, S = input_ids.shape
         image_tok = self.config.image_token_id
+        video_tok = getattr(self.config, "video_token_id", None)
+        frames = []
+        if video_grid_thw is not None:
+            for t, h, w in video_grid_thw.tolist():
+                frames.extend([[1, h, w]] * t)
         pos3 = torch.zeros(3, B, S, dtype=torch.long)
         deltas = torch.zeros(B, dtype=torch.long)
         m = self.visual.spatial_merge_size
         for b in range(B):
             ids = input_ids[b].tolist()
             img_i, cur, i = 0, 0, 0
+            vid_i = 0
             while i < S:
-                if ids[i] == image_tok:
-                    t, h, w = grid_thw[img_i].tolist()
+                is_vid = video_tok is not None and ids[i] == video_tok
+                if ids[i] == image_tok or is_vid:
+                    if is_vid:
+                        t, h, w = frames[vid_i]
+                        vid_i += 1
+                    else:
+                        t, h, w = grid_thw[img_i].tolist()
+                        img_i += 1
                     lh, lw = h // m, w // m
                     n = t * lh * lw
                     r = torch.arange(lh).view(-1, 1).expand(lh, lw).reshape(-1)
@@ -253,7 +272,6 @@ class Qwen3VLForConditionalGeneration(nn.Module):
                     pos3[2, b, i:i + n] = cur + c.repeat(t)
                     cur += max(lh, lw)
                     i += n
-                    img_i += 1
                 else:
                     pos3[:, b, i] = cur
                     cur += 1
@@ -292,7 +310,8 @@ class Qwen3VLForConditionalGeneration(nn.Module):
     # ------------------------------------------------------------------
     @torch.inference_mode()
     def generate(self, input_ids: torch.Tensor, pixel_values=None,
-                 image_grid_thw=None, attention_mask=None,
+                 image_grid_thw=None, pixel_values_videos=None,
+                 video_grid_thw=None, attention_mask=None,
                  max_new_tokens: int = 16, **kw):
         from neuronx_distributed_inference_amd.utils.hf_adapter import (
             GenerationConfig,
@@ -302,14 +321,22 @@ class Qwen3VLForConditionalGeneration(nn.Module):
         dm = self.language_model.model
         dm.multimodal_ctx = {}
         self._mrope_ctx = {}
+        splices, ds = [], []
         if pixel_values is not None:
             feats, deepstack = self.visual(pixel_values, image_grid_thw)
-            dm.multimodal_ctx = {
-                "features": feats, "token_id": self.config.image_token_id,
-            }
-            pos3, deltas = self._compute_rope_index(input_ids, image_grid_thw)
+            splices.append((self.config.image_token_id, feats))
+            ds.append((self.config.image_token_id, deepstack))
+        if pixel_values_videos is not None:
+            vfeats, vdeepstack = self.visual(pixel_values_videos,
+                                             video_grid_thw)
+            splices.append((self.config.video_token_id, vfeats))
+            ds.append((self.config.video_token_id, vdeepstack))
+        if splices:
+            dm.multimodal_ctx = {"splices": splices}
+            pos3, deltas = self._compute_rope_index(
+                input_ids, image_grid_thw, video_grid_thw)
             self._mrope_ctx = {"pos3d": pos3, "delta": deltas,
-                               "deepstack": deepstack}
+                               "deepstack": ds}
         adapter = HuggingFaceGenerationAdapter(self.language_model)
         return adapter.generate(
             input_ids, attention_mask=attention_mask,

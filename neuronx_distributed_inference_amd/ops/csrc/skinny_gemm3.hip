@@ -91,43 +91,56 @@ __global__ void skinny_gemm3_kernel(
     const int xx_row = threadIdx.x >> 2;
     const int xx_off = (threadIdx.x & 3) * 16;  // elems
 
-    struct Stage { sk3_i32x4 wr; sk3_i32x4 xr[2]; };
-    auto load_stage = [&](int k0, Stage& s) {
-        s.wr = __builtin_nontemporal_load(reinterpret_cast<const sk3_i32x4*>(
+    // SPLIT prefetch queues (the single-queue version collapsed: 6 fused
+    // stages x 12 VGPRs + temps exceeded the 128-VGPR budget and the
+    // allocator reused stage registers, shrinking the pipeline to depth
+    // ~1.5 — the ISA showed `s_waitcnt vmcnt(2)`/`vmcnt(0)` at every
+    // commit). W is the HBM stream and needs the deep queue (8 stages x
+    // 4 VGPRs); x is XCD-L2-resident (192 WGs re-read the same 1 MB) so
+    // a short 4-stage queue covers its latency.
+    auto load_w = [&](int k0, sk3_i32x4& r) {
+        r = __builtin_nontemporal_load(reinterpret_cast<const sk3_i32x4*>(
             w + (long)(n_base + ww_row) * K + k0 + ww_off));
+    };
+    auto load_x = [&](int k0, sk3_i32x4* r) {
         const int xrow = min(m_base + xx_row, M - 1);
 #pragma unroll
         for (int p = 0; p < 2; ++p)
-            s.xr[p] = *reinterpret_cast<const sk3_i32x4*>(
+            r[p] = *reinterpret_cast<const sk3_i32x4*>(
                 x + (long)xrow * K + k0 + xx_off + p * 8);
     };
-    auto commit_stage = [&](const Stage& s) {
-        *reinterpret_cast<sk3_i32x4*>(wt + ww_row * PITCH3 + ww_off) = s.wr;
+    auto commit_w = [&](const sk3_i32x4& r) {
+        *reinterpret_cast<sk3_i32x4*>(wt + ww_row * PITCH3 + ww_off) = r;
+    };
+    auto commit_x = [&](const sk3_i32x4* r) {
 #pragma unroll
         for (int p = 0; p < 2; ++p)
-            *reinterpret_cast<sk3_i32x4*>(xt + xx_row * PITCH3 + xx_off + p * 8) = s.xr[p];
+            *reinterpret_cast<sk3_i32x4*>(xt + xx_row * PITCH3 + xx_off + p * 8) = r[p];
     };
 
     sk3_f32x4 acc[2] = {sk3_f32x4{0.f, 0.f, 0.f, 0.f}, sk3_f32x4{0.f, 0.f, 0.f, 0.f}};
 
     const int nchunks = K / KC3;
-    // six NAMED stages (~18 b128 loads in flight per lane: the chunk's
-    // 1400-cycle HBM latency needs ~5 iterations of cover), modulo
-    // schedule unrolled x6: dynamic indexing of a Stage array would
-    // demote it to scratch (160 B/lane measured at Q=3 — every staged
-    // byte took a global->scratch->LDS round trip)
-    Stage st0, st1, st2, st3, st4, st5;
-    load_stage(0, st0);
-    load_stage(KC3, st1);
-    load_stage(2 * KC3, st2);
-    load_stage(3 * KC3, st3);
-    load_stage(4 * KC3, st4);
-    load_stage(5 * KC3, st5);
-    commit_stage(st0);
+    // NAMED stages, modulo schedule unrolled x8: dynamic indexing of a
+    // stage array would demote it to scratch (160 B/lane measured)
+    sk3_i32x4 w0, w1, w2, w3, w4, w5, w6, w7;
+    sk3_i32x4 x0[2], x1[2], x2[2], x3[2];
+    load_w(0, w0); load_w(KC3, w1); load_w(2 * KC3, w2); load_w(3 * KC3, w3);
+    load_w(4 * KC3, w4); load_w(5 * KC3, w5); load_w(6 * KC3, w6);
+    load_w(7 * KC3, w7);
+    load_x(0, x0); load_x(KC3, x1); load_x(2 * KC3, x2); load_x(3 * KC3, x3);
+    commit_w(w0);
+    commit_x(x0);
 
-    auto iter_body = [&](int i, Stage& reuse, Stage& next) {
-        // consume chunk i from LDS; commit chunk i+1 (in `next`); refill
-        // `reuse` (held chunk i, already in LDS) with chunk i+6
+    auto iter_body = [&](int i, sk3_i32x4& w_reuse, sk3_i32x4& w_next,
+                         sk3_i32x4* x_reuse, sk3_i32x4* x_next, bool steady) {
+        // consume chunk i from LDS; commit chunk i+1; refill the stages
+        // that held chunk i with chunks i+8 (W) and i+4 (x). In the
+        // steady state everything is UNCONDITIONAL (loads clamped): a
+        // load behind a branch breaks the compiler's static outstanding-
+        // load count and every later ds_write gets `s_waitcnt vmcnt(0)`,
+        // draining the whole queue (measured: no gain from depth 3->6
+        // until the branches went away).
         __syncthreads();                 // chunk i visible in LDS
 #pragma unroll
         for (int ks = 0; ks < KC3 / 32; ++ks) {
@@ -141,18 +154,37 @@ __global__ void skinny_gemm3_kernel(
             }
         }
         __syncthreads();                 // LDS reads done before restage
-        if (i + 1 < nchunks) {
-            commit_stage(next);          // chunk i+1 -> LDS
-            if (i + 6 < nchunks) load_stage((i + 6) * KC3, reuse);
+        if (steady) {
+            commit_w(w_next);            // chunk i+1 -> LDS
+            commit_x(x_next);
+            load_w(min(i + 8, nchunks - 1) * KC3, w_reuse);
+            load_x(min(i + 4, nchunks - 1) * KC3, x_reuse);
+        } else if (i + 1 < nchunks) {
+            commit_w(w_next);
+            commit_x(x_next);
         }
     };
-    for (int i = 0; i < nchunks; i += 6) {
-        iter_body(i, st0, st1);
-        if (i + 1 < nchunks) iter_body(i + 1, st1, st2);
-        if (i + 2 < nchunks) iter_body(i + 2, st2, st3);
-        if (i + 3 < nchunks) iter_body(i + 3, st3, st4);
-        if (i + 4 < nchunks) iter_body(i + 4, st4, st5);
-        if (i + 5 < nchunks) iter_body(i + 5, st5, st0);
+    int i = 0;
+    for (; i + 8 < nchunks; i += 8) {
+        iter_body(i + 0, w0, w1, x0, x1, true);
+        iter_body(i + 1, w1, w2, x1, x2, true);
+        iter_body(i + 2, w2, w3, x2, x3, true);
+        iter_body(i + 3, w3, w4, x3, x0, true);
+        iter_body(i + 4, w4, w5, x0, x1, true);
+        iter_body(i + 5, w5, w6, x1, x2, true);
+        iter_body(i + 6, w6, w7, x2, x3, true);
+        iter_body(i + 7, w7, w0, x3, x0, true);
+    }
+    // tail: every remaining chunk is already loaded; commit + consume only
+    {
+        iter_body(i + 0, w0, w1, x0, x1, false);
+        if (i + 1 < nchunks) iter_body(i + 1, w1, w2, x1, x2, false);
+        if (i + 2 < nchunks) iter_body(i + 2, w2, w3, x2, x3, false);
+        if (i + 3 < nchunks) iter_body(i + 3, w3, w4, x3, x0, false);
+        if (i + 4 < nchunks) iter_body(i + 4, w4, w5, x0, x1, false);
+        if (i + 5 < nchunks) iter_body(i + 5, w5, w6, x1, x2, false);
+        if (i + 6 < nchunks) iter_body(i + 6, w6, w7, x2, x3, false);
+        if (i + 7 < nchunks) iter_body(i + 7, w7, w0, x3, x0, false);
     }
 
 #pragma unroll
@@ -187,8 +219,8 @@ void launch_sk3(torch::Tensor& y, const torch::Tensor& x, const torch::Tensor& w
 void skinny_gemm3(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                   c10::optional<torch::Tensor> bias_opt) {
     const int M = x.size(0), K = x.size(1), N = w.size(0);
-    TORCH_CHECK(M <= 128 && N % NT3 == 0 && K % KC3 == 0 && K >= 6 * KC3,
-                "skinny_gemm3: M<=128, N%32, K%64, K>=384");
+    TORCH_CHECK(M <= 128 && N % NT3 == 0 && K % KC3 == 0 && K >= 8 * KC3,
+                "skinny_gemm3: M<=128, N%32, K%64, K>=512");
     TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && y.is_contiguous());
     torch::Tensor bias_t;
     const torch::Tensor* bias = nullptr;

@@ -581,7 +581,7 @@ def test_qwen3_vl():
                       torch_dtype="float32", tp_degree=1)
     cfg = Qwen3VLInferenceConfig(
         neuron_config=ec, text_config=dict(tc), vision_config=dict(vc),
-        image_token_id=159)
+        image_token_id=159, video_token_id=156)
     ours = Qwen3VLForConditionalGeneration(cfg).eval()
     sd = Qwen3VLForConditionalGeneration.convert_hf_to_native_state_dict(
         hf.state_dict(), cfg)
@@ -603,6 +603,30 @@ def test_qwen3_vl():
         ref_ids = torch.cat([ref_ids, lg[:, -1].argmax(-1, keepdim=True)], dim=1)
     got = ours.generate(ids, pixel_values=px, image_grid_thw=grid,
                         max_new_tokens=n_new)
+    assert torch.equal(got[:, : ref_ids.shape[1]], ref_ids), (
+        got.tolist(), ref_ids.tolist())
+
+    # video: HF splits each video into per-frame (1,h,w) grids (timestamp
+    # text tokens separate the frame runs); DeepStack injects at video
+    # positions too
+    vgrid = torch.tensor([[2, 4, 4]])
+    vpx = torch.randn(32, 3 * 2 * 4 * 4)
+    ids3 = torch.randint(0, 156, (1, 16))
+    ids3[0, 2:6] = 156                    # frame 1 (4 merged tokens)
+    ids3[0, 7:11] = 156                   # frame 2 (timestamp text between)
+    mm_tt = torch.zeros_like(ids3)
+    mm_tt[ids3 == 156] = 2
+    ref_ids = ids3.clone()
+    for _ in range(6):
+        lg = hf(input_ids=ref_ids, pixel_values_videos=vpx,
+                video_grid_thw=vgrid,
+                mm_token_type_ids=torch.cat(
+                    [mm_tt, torch.zeros(1, ref_ids.shape[1] - 16,
+                                        dtype=mm_tt.dtype)], dim=1),
+                use_cache=False).logits
+        ref_ids = torch.cat([ref_ids, lg[:, -1].argmax(-1, keepdim=True)], dim=1)
+    got = ours.generate(ids3, pixel_values_videos=vpx, video_grid_thw=vgrid,
+                        max_new_tokens=6)
     assert torch.equal(got[:, : ref_ids.shape[1]], ref_ids), (
         got.tolist(), ref_ids.tolist())
 
