@@ -323,12 +323,28 @@ class AttentionBase(nn.Module):
 
             qc, kc = _ops.rope_apply(qc, kc, cos, sin)
 
-        # this rank's S/cp query slice vs the full K/V, offset-causal mask
+        # this rank's S/cp query rows vs the full K/V, offset-causal mask.
+        # ZIGZAG assignment when S % 2cp == 0 (rank r takes chunks r and
+        # 2cp-1-r of 2cp): causal work per rank is constant, where
+        # contiguous slices leave the first rank nearly idle (the ref's
+        # strided CP query assignment, attention/utils.py CP paths).
         ssl = S // cp
-        c0 = cp_rank * ssl
-        q_slice = qc[:, :, c0:c0 + ssl].contiguous()
-        ar = torch.arange(S, device=hidden.device).view(1, 1, 1, S)
-        qpos = (c0 + torch.arange(ssl, device=hidden.device)).view(1, 1, ssl, 1)
+        zig = S % (2 * cp) == 0
+        dev = hidden.device
+        if zig:
+            half = S // (2 * cp)
+            c0a, c0b = cp_rank * half, (2 * cp - 1 - cp_rank) * half
+            q_slice = torch.cat(
+                [qc[:, :, c0a:c0a + half], qc[:, :, c0b:c0b + half]], dim=2
+            ).contiguous()
+            qpos = torch.cat([c0a + torch.arange(half, device=dev),
+                              c0b + torch.arange(half, device=dev)])
+            qpos = qpos.view(1, 1, ssl, 1)
+        else:
+            c0 = cp_rank * ssl
+            q_slice = qc[:, :, c0:c0 + ssl].contiguous()
+            qpos = (c0 + torch.arange(ssl, device=dev)).view(1, 1, ssl, 1)
+        ar = torch.arange(S, device=dev).view(1, 1, 1, S)
         mask = ar <= qpos
         if attention_mask is not None:
             mask = mask & attention_mask.view(B, 1, 1, S).bool()
@@ -338,8 +354,18 @@ class AttentionBase(nn.Module):
                                     sinks=self.sinks, causal=False)
         out = out.transpose(1, 2).reshape(B, ssl, -1)
         out = self.cte_o_proj(out)                   # reduce over prefill group
-        return mappings.gather_along_dim(out, dim=1,
-                                         group=ps.get_context_parallel_group())
+        out = mappings.gather_along_dim(out, dim=1,
+                                        group=ps.get_context_parallel_group())
+        if zig:
+            # gathered order is [r0:(chunk 0, chunk 2cp-1), r1:(1, 2cp-2)...];
+            # permute back to natural chunk order
+            half = S // (2 * cp)
+            chunks = out.view(B, 2 * cp, half, -1)
+            order = [min(c, 2 * cp - 1 - c) * 2 + (0 if c < cp else 1)
+                     for c in range(2 * cp)]
+            idx = torch.tensor(order, device=out.device)
+            out = chunks.index_select(1, idx).reshape(B, S, -1)
+        return out
 
     def _dp_local_rows(self, seq_ids, B, device):
         dp_rank = ps.get_attention_dp_rank()

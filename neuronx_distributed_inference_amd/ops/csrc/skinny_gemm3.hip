@@ -162,6 +162,10 @@ __global__ void skinny_gemm3_kernel(
         } else if (i + 1 < nchunks) {
             commit_w(w_next);
             commit_x(x_next);
+            // the x queue is only 4 deep vs the 8-body tail: the first
+            // tail bodies must still refill it (W's 8-deep queue is fully
+            // loaded by the steady loop)
+            if (i + 4 < nchunks) load_x((i + 4) * KC3, x_reuse);
         }
     };
     int i = 0;
