@@ -159,6 +159,11 @@ __global__ void skinny_gemm3_kernel(
             commit_x(x_next);
             load_w(min(i + 8, nchunks - 1) * KC3, w_reuse);
             load_x(min(i + 4, nchunks - 1) * KC3, x_reuse);
+            // pin the schedule: without this the scheduler sinks stage
+            // loads toward their commits (ISA showed vmcnt(4-12) where
+            // the 7-body pipeline implies vmcnt(~21)), collapsing the
+            // latency cover
+            __builtin_amdgcn_sched_barrier(0);
         } else if (i + 1 < nchunks) {
             commit_w(w_next);
             commit_x(x_next);
