@@ -81,32 +81,41 @@ __global__ void skinny_gemm3_kernel(
     const int kseg = lane >> 4;
 
     __shared__ T wt[NT3 * PITCH3];
+    __shared__ T xt[MT3 * PITCH3];
 
-    // staging map (one chunk = W 32x64 = 4 KB):
+    // staging maps (one chunk = W 32x64 = 4 KB + x 64x64 = 8 KB):
     // W: 8 threads per 128 B row -> 1 x b128 per thread
     const int ww_row = threadIdx.x >> 3;
     const int ww_off = (threadIdx.x & 7) * 8;   // elems
+    // x: 4 threads per 128 B row -> 2 x b128 per thread
+    const int xx_row = threadIdx.x >> 2;
+    const int xx_off = (threadIdx.x & 3) * 16;  // elems
 
-    // W is the HBM stream: coalesced staging through LDS with a deep
-    // (8-stage x 4 VGPR) register queue. x never touches LDS: each
-    // wave's A-fragment is a per-lane 16 B slice of the XCD-L2-resident
-    // x (1 MB, re-read by all 192 N-column WGs), loaded straight into
-    // the MFMA operand registers two bodies ahead. This halves the
-    // LDS-array traffic (x was 2/3 of the staged bytes) and frees the
-    // x stage registers for the W queue.
+    // SPLIT prefetch queues (the single-queue version collapsed: 6 fused
+    // stages x 12 VGPRs + temps exceeded the 128-VGPR budget and the
+    // allocator reused stage registers, shrinking the pipeline to depth
+    // ~1.5 — the ISA showed `s_waitcnt vmcnt(2)`/`vmcnt(0)` at every
+    // commit). W is the HBM stream and needs the deep queue (8 stages x
+    // 4 VGPRs); x is XCD-L2-resident (192 WGs re-read the same 1 MB) so
+    // a short 4-stage queue covers its latency.
     auto load_w = [&](int k0, sk3_i32x4& r) {
         r = __builtin_nontemporal_load(reinterpret_cast<const sk3_i32x4*>(
             w + (long)(n_base + ww_row) * K + k0 + ww_off));
     };
-    const int arow = min(m_base + wave * 16 + col, M - 1);
-    auto load_a = [&](int k0, sk3_i32x4* r) {
+    auto load_x = [&](int k0, sk3_i32x4* r) {
+        const int xrow = min(m_base + xx_row, M - 1);
 #pragma unroll
-        for (int ks = 0; ks < KC3 / 32; ++ks)
-            r[ks] = *reinterpret_cast<const sk3_i32x4*>(
-                x + (long)arow * K + k0 + ks * 32 + kseg * 8);
+        for (int p = 0; p < 2; ++p)
+            r[p] = *reinterpret_cast<const sk3_i32x4*>(
+                x + (long)xrow * K + k0 + xx_off + p * 8);
     };
     auto commit_w = [&](const sk3_i32x4& r) {
         *reinterpret_cast<sk3_i32x4*>(wt + ww_row * PITCH3 + ww_off) = r;
+    };
+    auto commit_x = [&](const sk3_i32x4* r) {
+#pragma unroll
+        for (int p = 0; p < 2; ++p)
+            *reinterpret_cast<sk3_i32x4*>(xt + xx_row * PITCH3 + xx_off + p * 8) = r[p];
     };
 
     sk3_f32x4 acc[2] = {sk3_f32x4{0.f, 0.f, 0.f, 0.f}, sk3_f32x4{0.f, 0.f, 0.f, 0.f}};
@@ -115,43 +124,41 @@ __global__ void skinny_gemm3_kernel(
     // NAMED stages, modulo schedule unrolled x8: dynamic indexing of a
     // stage array would demote it to scratch (160 B/lane measured)
     sk3_i32x4 w0, w1, w2, w3, w4, w5, w6, w7;
-    sk3_i32x4 a0[2], a1[2];
+    sk3_i32x4 x0[2], x1[2], x2[2], x3[2];
     load_w(0, w0); load_w(KC3, w1); load_w(2 * KC3, w2); load_w(3 * KC3, w3);
     load_w(4 * KC3, w4); load_w(5 * KC3, w5); load_w(6 * KC3, w6);
     load_w(7 * KC3, w7);
-    load_a(0, a0); load_a(KC3, a1);
+    load_x(0, x0); load_x(KC3, x1); load_x(2 * KC3, x2); load_x(3 * KC3, x3);
     commit_w(w0);
+    commit_x(x0);
 
     auto iter_body = [&](int i, sk3_i32x4& w_reuse, sk3_i32x4& w_next,
-                         sk3_i32x4* a_cur, sk3_i32x4* a_other, bool steady) {
-        // consume chunk i (W from LDS, A from registers); commit W chunk
-        // i+1; refill the W stage that held chunk i with chunk i+8 and
-        // the A register pair with chunk i+2. In the steady state
-        // everything is UNCONDITIONAL (loads clamped): a load behind a
-        // branch breaks the compiler's static outstanding-load count and
-        // every later ds_write gets `s_waitcnt vmcnt(0)`, draining the
-        // whole queue (measured: no gain from depth 3->6 until the
-        // branches went away).
-        __syncthreads();                 // W chunk i visible in LDS
-        vec8 af[KC3 / 32];
-#pragma unroll
-        for (int ks = 0; ks < KC3 / 32; ++ks)
-            af[ks] = *reinterpret_cast<const vec8*>(&a_cur[ks]);
-        if (steady) load_a(min(i + 2, nchunks - 1) * KC3, a_cur);
-        else if (i + 2 < nchunks) load_a((i + 2) * KC3, a_cur);
+                         sk3_i32x4* x_reuse, sk3_i32x4* x_next, bool steady) {
+        // consume chunk i from LDS; commit chunk i+1; refill the stages
+        // that held chunk i with chunks i+8 (W) and i+4 (x). In the
+        // steady state everything is UNCONDITIONAL (loads clamped): a
+        // load behind a branch breaks the compiler's static outstanding-
+        // load count and every later ds_write gets `s_waitcnt vmcnt(0)`,
+        // draining the whole queue (measured: no gain from depth 3->6
+        // until the branches went away).
+        __syncthreads();                 // chunk i visible in LDS
 #pragma unroll
         for (int ks = 0; ks < KC3 / 32; ++ks) {
+            const vec8 afrag = *reinterpret_cast<const vec8*>(
+                xt + (wave * 16 + col) * PITCH3 + ks * 32 + kseg * 8);
 #pragma unroll
             for (int n = 0; n < 2; ++n) {
                 const vec8 bfrag = *reinterpret_cast<const vec8*>(
                     wt + (n * 16 + col) * PITCH3 + ks * 32 + kseg * 8);
-                acc[n] = MF::mfma(af[ks], bfrag, acc[n]);
+                acc[n] = MF::mfma(afrag, bfrag, acc[n]);
             }
         }
         __syncthreads();                 // LDS reads done before restage
         if (steady) {
-            commit_w(w_next);            // W chunk i+1 -> LDS
+            commit_w(w_next);            // chunk i+1 -> LDS
+            commit_x(x_next);
             load_w(min(i + 8, nchunks - 1) * KC3, w_reuse);
+            load_x(min(i + 4, nchunks - 1) * KC3, x_reuse);
             // pin the schedule: without this the scheduler sinks stage
             // loads toward their commits (ISA showed vmcnt(4-12) where
             // the 7-body pipeline implies vmcnt(~21)), collapsing the
@@ -159,30 +166,34 @@ __global__ void skinny_gemm3_kernel(
             __builtin_amdgcn_sched_barrier(0);
         } else if (i + 1 < nchunks) {
             commit_w(w_next);
+            commit_x(x_next);
+            // the x queue is only 4 deep vs the 8-body tail: the first
+            // tail bodies must still refill it (W's 8-deep queue is fully
+            // loaded by the steady loop)
+            if (i + 4 < nchunks) load_x((i + 4) * KC3, x_reuse);
         }
-        (void)a_other;
     };
     int i = 0;
     for (; i + 8 < nchunks; i += 8) {
-        iter_body(i + 0, w0, w1, a0, a1, true);
-        iter_body(i + 1, w1, w2, a1, a0, true);
-        iter_body(i + 2, w2, w3, a0, a1, true);
-        iter_body(i + 3, w3, w4, a1, a0, true);
-        iter_body(i + 4, w4, w5, a0, a1, true);
-        iter_body(i + 5, w5, w6, a1, a0, true);
-        iter_body(i + 6, w6, w7, a0, a1, true);
-        iter_body(i + 7, w7, w0, a1, a0, true);
+        iter_body(i + 0, w0, w1, x0, x1, true);
+        iter_body(i + 1, w1, w2, x1, x2, true);
+        iter_body(i + 2, w2, w3, x2, x3, true);
+        iter_body(i + 3, w3, w4, x3, x0, true);
+        iter_body(i + 4, w4, w5, x0, x1, true);
+        iter_body(i + 5, w5, w6, x1, x2, true);
+        iter_body(i + 6, w6, w7, x2, x3, true);
+        iter_body(i + 7, w7, w0, x3, x0, true);
     }
-    // tail: W chunks are all loaded; A pairs keep cycling with guards
+    // tail: every remaining chunk is already loaded; commit + consume only
     {
-        iter_body(i + 0, w0, w1, a0, a1, false);
-        if (i + 1 < nchunks) iter_body(i + 1, w1, w2, a1, a0, false);
-        if (i + 2 < nchunks) iter_body(i + 2, w2, w3, a0, a1, false);
-        if (i + 3 < nchunks) iter_body(i + 3, w3, w4, a1, a0, false);
-        if (i + 4 < nchunks) iter_body(i + 4, w4, w5, a0, a1, false);
-        if (i + 5 < nchunks) iter_body(i + 5, w5, w6, a1, a0, false);
-        if (i + 6 < nchunks) iter_body(i + 6, w6, w7, a0, a1, false);
-        if (i + 7 < nchunks) iter_body(i + 7, w7, w0, a1, a0, false);
+        iter_body(i + 0, w0, w1, x0, x1, false);
+        if (i + 1 < nchunks) iter_body(i + 1, w1, w2, x1, x2, false);
+        if (i + 2 < nchunks) iter_body(i + 2, w2, w3, x2, x3, false);
+        if (i + 3 < nchunks) iter_body(i + 3, w3, w4, x3, x0, false);
+        if (i + 4 < nchunks) iter_body(i + 4, w4, w5, x0, x1, false);
+        if (i + 5 < nchunks) iter_body(i + 5, w5, w6, x1, x2, false);
+        if (i + 6 < nchunks) iter_body(i + 6, w6, w7, x2, x3, false);
+        if (i + 7 < nchunks) iter_body(i + 7, w7, w0, x3, x0, false);
     }
 
 #pragma unroll
