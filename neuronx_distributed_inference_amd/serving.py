@@ -52,7 +52,7 @@ class ServingEngine:
                  chunk_size: Optional[int] = None,
                  eos_token_id: Optional[int] = None,
                  capture_decode: Optional[bool] = None,
-                 on_token=None, lora_state=None):
+                 on_token=None, lora_state=None, bucket_chunks: bool = True):
         self.model = causal_lm
         ec = causal_lm.config.neuron_config
         self.batch = ec.batch_size
@@ -79,6 +79,13 @@ class ServingEngine:
         self._next_id = 0
         self.on_token = on_token        # callback(req_id, token) per token
         self.lora_state = lora_state    # multi-LoRA: per-request adapter rows
+        # pad prefill chunks up a power-of-2 ladder (pad rows carry
+        # position -1, so the pad-skip invariant keeps them out of the KV
+        # cache): bounds the set of [active, prefix] shapes the engine
+        # presents, the analog of the reference's 2-D prefix buckets
+        # (ref model_wrapper.py:923) — and what makes tail-prefill
+        # autotune/capture shapes finite
+        self.bucket_chunks = bucket_chunks
         # hipGraph-captured decode step (paged serving would otherwise run
         # its full-batch decode eagerly; linear serving uses the model's
         # own captured wrappers)
@@ -160,6 +167,24 @@ class ServingEngine:
         hi = S if self.chunk_size is None else min(S, lo + self.chunk_size)
         ids = req.prompt[lo:hi].view(1, -1).to(self.device)
         positions = torch.arange(lo, hi, device=self.device).view(1, -1)
+        L = hi - lo
+        # only INTERMEDIATE chunks are padded: their logits are discarded
+        # (just the pad-skipped KV writes matter). The final chunk must
+        # stay ragged — the CTE last-token gather reads the last row
+        if self.bucket_chunks and hi < S:
+            Lp = 1
+            while Lp < L:
+                Lp *= 2
+            if self.chunk_size is not None:
+                Lp = min(Lp, self.chunk_size)
+            if Lp > L:
+                pad = Lp - L
+                ids = torch.cat(
+                    [ids, torch.zeros(1, pad, dtype=ids.dtype,
+                                      device=self.device)], dim=1)
+                positions = torch.cat(
+                    [positions, torch.full((1, pad), -1, dtype=positions.dtype,
+                                           device=self.device)], dim=1)
         seq_ids = torch.tensor([req.slot], device=self.device)
         sp = self._params_for([req])
         self._set_adapters([req])
