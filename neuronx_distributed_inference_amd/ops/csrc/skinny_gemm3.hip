@@ -58,10 +58,12 @@ template <> struct Sk3Mfma<__half> {
 
 constexpr int KC3 = 64;                // K elems per LDS chunk
 constexpr int PITCH3 = KC3 + 8;        // 9 b128 slots: conflict-free frags
-constexpr int NT3 = 32;                // W rows per WG
+constexpr int NT3 = 64;                // W rows per WG (x logical traffic
+                                       // scales as N/NT3: PMC showed the
+                                       // x re-read was the 1.56x over-fetch)
 constexpr int MT3 = 64;                // x rows per WG (4 waves x 16)
 
-// 4 waves each own one 16-row m-tile and loop the 2 n-tiles.
+// 4 waves each own one 16-row m-tile and loop the 4 n-tiles.
 template <typename T>
 __launch_bounds__(256, 4)
 __global__ void skinny_gemm3_kernel(
@@ -98,9 +100,11 @@ __global__ void skinny_gemm3_kernel(
     // commit). W is the HBM stream and needs the deep queue (8 stages x
     // 4 VGPRs); x is XCD-L2-resident (192 WGs re-read the same 1 MB) so
     // a short 4-stage queue covers its latency.
-    auto load_w = [&](int k0, sk3_i32x4& r) {
-        r = __builtin_nontemporal_load(reinterpret_cast<const sk3_i32x4*>(
-            w + (long)(n_base + ww_row) * K + k0 + ww_off));
+    auto load_w = [&](int k0, sk3_i32x4* r) {
+#pragma unroll
+        for (int p = 0; p < 2; ++p)
+            r[p] = __builtin_nontemporal_load(reinterpret_cast<const sk3_i32x4*>(
+                w + (long)(n_base + ww_row) * K + k0 + ww_off + p * 8));
     };
     auto load_x = [&](int k0, sk3_i32x4* r) {
         const int xrow = min(m_base + xx_row, M - 1);
@@ -109,8 +113,10 @@ __global__ void skinny_gemm3_kernel(
             r[p] = *reinterpret_cast<const sk3_i32x4*>(
                 x + (long)xrow * K + k0 + xx_off + p * 8);
     };
-    auto commit_w = [&](const sk3_i32x4& r) {
-        *reinterpret_cast<sk3_i32x4*>(wt + ww_row * PITCH3 + ww_off) = r;
+    auto commit_w = [&](const sk3_i32x4* r) {
+#pragma unroll
+        for (int p = 0; p < 2; ++p)
+            *reinterpret_cast<sk3_i32x4*>(wt + ww_row * PITCH3 + ww_off + p * 8) = r[p];
     };
     auto commit_x = [&](const sk3_i32x4* r) {
 #pragma unroll
@@ -118,24 +124,26 @@ __global__ void skinny_gemm3_kernel(
             *reinterpret_cast<sk3_i32x4*>(xt + xx_row * PITCH3 + xx_off + p * 8) = r[p];
     };
 
-    sk3_f32x4 acc[2] = {sk3_f32x4{0.f, 0.f, 0.f, 0.f}, sk3_f32x4{0.f, 0.f, 0.f, 0.f}};
+    sk3_f32x4 acc[4];
+#pragma unroll
+    for (int n = 0; n < 4; ++n) acc[n] = sk3_f32x4{0.f, 0.f, 0.f, 0.f};
 
     const int nchunks = K / KC3;
-    // NAMED stages, modulo schedule unrolled x8: dynamic indexing of a
-    // stage array would demote it to scratch (160 B/lane measured)
-    sk3_i32x4 w0, w1, w2, w3, w4, w5, w6, w7;
-    sk3_i32x4 x0[2], x1[2], x2[2], x3[2];
+    // NAMED stages, modulo schedule unrolled x6 (W queue 6 x 2 b128,
+    // x queue 3 x 2 b128): dynamic indexing of a stage array would
+    // demote it to scratch (160 B/lane measured)
+    sk3_i32x4 w0[2], w1[2], w2[2], w3[2], w4[2], w5[2];
+    sk3_i32x4 x0[2], x1[2], x2[2];
     load_w(0, w0); load_w(KC3, w1); load_w(2 * KC3, w2); load_w(3 * KC3, w3);
-    load_w(4 * KC3, w4); load_w(5 * KC3, w5); load_w(6 * KC3, w6);
-    load_w(7 * KC3, w7);
-    load_x(0, x0); load_x(KC3, x1); load_x(2 * KC3, x2); load_x(3 * KC3, x3);
+    load_w(4 * KC3, w4); load_w(5 * KC3, w5);
+    load_x(0, x0); load_x(KC3, x1); load_x(2 * KC3, x2);
     commit_w(w0);
     commit_x(x0);
 
-    auto iter_body = [&](int i, sk3_i32x4& w_reuse, sk3_i32x4& w_next,
+    auto iter_body = [&](int i, sk3_i32x4* w_reuse, sk3_i32x4* w_next,
                          sk3_i32x4* x_reuse, sk3_i32x4* x_next, bool steady) {
         // consume chunk i from LDS; commit chunk i+1; refill the stages
-        // that held chunk i with chunks i+8 (W) and i+4 (x). In the
+        // that held chunk i with chunks i+6 (W) and i+3 (x). In the
         // steady state everything is UNCONDITIONAL (loads clamped): a
         // load behind a branch breaks the compiler's static outstanding-
         // load count and every later ds_write gets `s_waitcnt vmcnt(0)`,
@@ -147,7 +155,7 @@ __global__ void skinny_gemm3_kernel(
             const vec8 afrag = *reinterpret_cast<const vec8*>(
                 xt + (wave * 16 + col) * PITCH3 + ks * 32 + kseg * 8);
 #pragma unroll
-            for (int n = 0; n < 2; ++n) {
+            for (int n = 0; n < 4; ++n) {
                 const vec8 bfrag = *reinterpret_cast<const vec8*>(
                     wt + (n * 16 + col) * PITCH3 + ks * 32 + kseg * 8);
                 acc[n] = MF::mfma(afrag, bfrag, acc[n]);
@@ -157,47 +165,41 @@ __global__ void skinny_gemm3_kernel(
         if (steady) {
             commit_w(w_next);            // chunk i+1 -> LDS
             commit_x(x_next);
-            load_w(min(i + 8, nchunks - 1) * KC3, w_reuse);
-            load_x(min(i + 4, nchunks - 1) * KC3, x_reuse);
+            load_w(min(i + 6, nchunks - 1) * KC3, w_reuse);
+            load_x(min(i + 3, nchunks - 1) * KC3, x_reuse);
             // pin the schedule: without this the scheduler sinks stage
-            // loads toward their commits (ISA showed vmcnt(4-12) where
-            // the 7-body pipeline implies vmcnt(~21)), collapsing the
-            // latency cover
+            // loads toward their commits, collapsing the latency cover
             __builtin_amdgcn_sched_barrier(0);
         } else if (i + 1 < nchunks) {
             commit_w(w_next);
             commit_x(x_next);
-            // the x queue is only 4 deep vs the 8-body tail: the first
-            // tail bodies must still refill it (W's 8-deep queue is fully
+            // the x queue is only 3 deep vs the 6-body tail: the first
+            // tail bodies must still refill it (W's 6-deep queue is fully
             // loaded by the steady loop)
-            if (i + 4 < nchunks) load_x((i + 4) * KC3, x_reuse);
+            if (i + 3 < nchunks) load_x((i + 3) * KC3, x_reuse);
         }
     };
     int i = 0;
-    for (; i + 8 < nchunks; i += 8) {
+    for (; i + 6 < nchunks; i += 6) {
         iter_body(i + 0, w0, w1, x0, x1, true);
         iter_body(i + 1, w1, w2, x1, x2, true);
-        iter_body(i + 2, w2, w3, x2, x3, true);
-        iter_body(i + 3, w3, w4, x3, x0, true);
-        iter_body(i + 4, w4, w5, x0, x1, true);
-        iter_body(i + 5, w5, w6, x1, x2, true);
-        iter_body(i + 6, w6, w7, x2, x3, true);
-        iter_body(i + 7, w7, w0, x3, x0, true);
+        iter_body(i + 2, w2, w3, x2, x0, true);
+        iter_body(i + 3, w3, w4, x0, x1, true);
+        iter_body(i + 4, w4, w5, x1, x2, true);
+        iter_body(i + 5, w5, w0, x2, x0, true);
     }
-    // tail: every remaining chunk is already loaded; commit + consume only
+    // tail: every remaining W chunk is already loaded; x refills guarded
     {
         iter_body(i + 0, w0, w1, x0, x1, false);
         if (i + 1 < nchunks) iter_body(i + 1, w1, w2, x1, x2, false);
-        if (i + 2 < nchunks) iter_body(i + 2, w2, w3, x2, x3, false);
-        if (i + 3 < nchunks) iter_body(i + 3, w3, w4, x3, x0, false);
-        if (i + 4 < nchunks) iter_body(i + 4, w4, w5, x0, x1, false);
-        if (i + 5 < nchunks) iter_body(i + 5, w5, w6, x1, x2, false);
-        if (i + 6 < nchunks) iter_body(i + 6, w6, w7, x2, x3, false);
-        if (i + 7 < nchunks) iter_body(i + 7, w7, w0, x3, x0, false);
+        if (i + 2 < nchunks) iter_body(i + 2, w2, w3, x2, x0, false);
+        if (i + 3 < nchunks) iter_body(i + 3, w3, w4, x0, x1, false);
+        if (i + 4 < nchunks) iter_body(i + 4, w4, w5, x1, x2, false);
+        if (i + 5 < nchunks) iter_body(i + 5, w5, w0, x2, x0, false);
     }
 
 #pragma unroll
-    for (int n = 0; n < 2; ++n) {
+    for (int n = 0; n < 4; ++n) {
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
             const int m = m_base + wave * 16 + kseg * 4 + r;
@@ -228,8 +230,8 @@ void launch_sk3(torch::Tensor& y, const torch::Tensor& x, const torch::Tensor& w
 void skinny_gemm3(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                   c10::optional<torch::Tensor> bias_opt) {
     const int M = x.size(0), K = x.size(1), N = w.size(0);
-    TORCH_CHECK(M <= 128 && N % NT3 == 0 && K % KC3 == 0 && K >= 8 * KC3,
-                "skinny_gemm3: M<=128, N%32, K%64, K>=512");
+    TORCH_CHECK(M <= 1024 && N % NT3 == 0 && K % KC3 == 0 && K >= 6 * KC3,
+                "skinny_gemm3: M<=1024, N%64, K%64, K>=384");
     TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && y.is_contiguous());
     torch::Tensor bias_t;
     const torch::Tensor* bias = nullptr;
