@@ -87,8 +87,8 @@ __global__ void skinny_gemm3_kernel(
 
     // staging maps (one chunk = W 32x64 = 4 KB + x 64x64 = 8 KB):
     // W: 8 threads per 128 B row -> 1 x b128 per thread
-    const int ww_row = threadIdx.x >> 3;
-    const int ww_off = (threadIdx.x & 7) * 8;   // elems
+    const int ww_row = threadIdx.x >> 2;
+    const int ww_off = (threadIdx.x & 3) * 16;  // elems
     // x: 4 threads per 128 B row -> 2 x b128 per thread
     const int xx_row = threadIdx.x >> 2;
     const int xx_off = (threadIdx.x & 3) * 16;  // elems
