@@ -22,6 +22,7 @@ All state (tokens, positions) stays on the model's device between steps.
 from __future__ import annotations
 
 import collections
+import time
 from dataclasses import dataclass, field
 from typing import Deque, Dict, List, Optional
 
@@ -45,6 +46,9 @@ class Request:
     done: bool = False
     blocks: List[int] = field(default_factory=list)  # paged: block chain
     n_cached: int = 0               # prompt tokens served from prefix cache
+    t_submit: float = 0.0           # metrics timestamps (monotonic)
+    t_first_token: float = 0.0
+    t_finish: float = 0.0
 
 
 class ServingEngine:
@@ -107,6 +111,7 @@ class ServingEngine:
         self.queue.append(Request(rid, prompt_ids.view(-1).cpu(),
                                   max_new_tokens, top_k, top_p, temperature,
                                   adapter_id))
+        self.queue[-1].t_submit = time.monotonic()
         return rid
 
     def _set_adapters(self, reqs):
@@ -206,6 +211,7 @@ class ServingEngine:
             req.last_token = tok
             req.pos = S
             req.output.append(tok)
+            req.t_first_token = time.monotonic()
             if self.on_token is not None:
                 self.on_token(req.req_id, tok)
             self._maybe_finish(req, tok)
@@ -281,6 +287,7 @@ class ServingEngine:
             self.eos is not None and tok == self.eos
         ):
             req.done = True
+            req.t_finish = time.monotonic()
             self.active.pop(req.slot)
             self.free_slots.append(req.slot)
             self.finished[req.req_id] = req
@@ -317,3 +324,35 @@ class ServingEngine:
             self.step()
             steps += 1
         return {rid: r.output for rid, r in self.finished.items()}
+
+    def metrics(self) -> Dict[str, float]:
+        """Serving metrics over finished requests (ref utils/benchmark.py
+        p50/p90/p99 latency report): TTFT = submit -> first token, TPOT =
+        mean inter-token time after the first, plus prefix-cache stats."""
+        def pct(vals, q):
+            if not vals:
+                return 0.0
+            vals = sorted(vals)
+            return vals[min(len(vals) - 1, int(q * len(vals)))]
+
+        ttft, tpot, cached, prompt = [], [], 0, 0
+        for r in self.finished.values():
+            if r.t_first_token and r.t_submit:
+                ttft.append(r.t_first_token - r.t_submit)
+            if r.t_finish and r.t_first_token and len(r.output) > 1:
+                tpot.append((r.t_finish - r.t_first_token)
+                            / (len(r.output) - 1))
+            cached += r.n_cached
+            prompt += int(r.prompt.shape[0])
+        return {
+            "requests": float(len(self.finished)),
+            "ttft_p50_s": pct(ttft, 0.50),
+            "ttft_p90_s": pct(ttft, 0.90),
+            "ttft_p99_s": pct(ttft, 0.99),
+            "tpot_p50_s": pct(tpot, 0.50),
+            "tpot_p90_s": pct(tpot, 0.90),
+            "output_tokens": float(sum(len(r.output)
+                                       for r in self.finished.values())),
+            "prefix_cached_tokens": float(cached),
+            "prefix_hit_rate": cached / prompt if prompt else 0.0,
+        }

@@ -86,3 +86,52 @@ def test_reconstruct_kv_cache():
     rep = [full[:, :1].clone(), full[:, :1].clone()]
     out = reconstruct_kv_cache(rep)
     assert torch.equal(out, full[:, :1])
+
+
+def test_kv_cache_reconstruction():
+    """Cross-shard KV reconstruction (ref utils/kv_cache_reconstruct_utils.py):
+    TP head split (incl. GQA replication), attention-DP batch split, and
+    flash-decoding strided sequence shards all invert back to the full
+    cache."""
+    import torch
+
+    from neuronx_distributed_inference_amd.utils.kv_reconstruct import (
+        merge_kv_shard_seq,
+        reconstruct_kv_cache,
+    )
+
+    torch.manual_seed(0)
+    B, H, S, D = 2, 4, 12, 8
+    full = torch.randn(B, H, S, D)
+
+    # TP=4: one head per rank
+    shards = [full[:, h:h + 1] for h in range(4)]
+    got = reconstruct_kv_cache(shards, num_kv_heads=H, tp_degree=4)
+    assert torch.equal(got, full)
+
+    # TP=8 > H: each head replicated twice
+    shards = [full[:, h // 2:h // 2 + 1] for h in range(8)]
+    got = reconstruct_kv_cache(shards, num_kv_heads=H, tp_degree=8)
+    assert torch.equal(got, full)
+
+    # flash-decoding fd=2 within tp=8 (4 head groups x 2 seq shards)
+    fd = 2
+    seq_shards = [full[:, :, r::fd] for r in range(fd)]
+    assert torch.equal(merge_kv_shard_seq(seq_shards, S), full)
+    shards = []
+    for h in range(4):
+        for r in range(fd):
+            shards.append(full[:, h:h + 1, r::fd])
+    got = reconstruct_kv_cache(shards, num_kv_heads=H, tp_degree=8,
+                               kv_shard_degree=fd, total_len=S)
+    assert torch.equal(got, full)
+
+    # attention-DP=2 over tp=2 (batch rows split across dp blocks)
+    shards = []
+    for d in range(2):
+        rows = full[d * (B // 2):(d + 1) * (B // 2)]
+        for h in range(2):
+            shards.append(rows[:, h * 2:(h + 1) * 2])
+    got = reconstruct_kv_cache(shards, num_kv_heads=H, tp_degree=2,
+                               dp_degree=2)
+    assert torch.equal(got, full)

@@ -281,3 +281,20 @@ def test_serving_per_request_lora_adapters():
     results = eng.run()
     assert results[r0] == wants[0]
     assert results[r1] == wants[1]
+
+
+@torch.inference_mode()
+def test_serving_metrics():
+    """metrics(): TTFT/TPOT percentiles + prefix-hit stats over finished
+    requests (ref utils/benchmark.py latency report)."""
+    m = _model(batch=2)
+    eng = ServingEngine(m)
+    eng.add_request(torch.randint(0, 128, (4,)), max_new_tokens=4)
+    eng.add_request(torch.randint(0, 128, (3,)), max_new_tokens=3)
+    eng.run()
+    m = eng.metrics()
+    assert m["requests"] == 2.0
+    assert m["output_tokens"] == 7.0
+    assert m["ttft_p50_s"] >= 0 and m["ttft_p99_s"] >= m["ttft_p50_s"] - 1e-9
+    assert m["tpot_p50_s"] >= 0
+    assert 0.0 <= m["prefix_hit_rate"] <= 1.0
