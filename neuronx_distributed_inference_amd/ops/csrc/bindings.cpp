@@ -37,6 +37,8 @@ void skinny_gemm2(torch::Tensor y, torch::Tensor y32, torch::Tensor x,
                   torch::Tensor w, c10::optional<torch::Tensor> bias, long ksplit);
 void skinny_gemm3(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                   c10::optional<torch::Tensor> bias);
+void skinny_gemm4(torch::Tensor y, torch::Tensor x, torch::Tensor w,
+                  c10::optional<torch::Tensor> bias);
 void fp8_quant(torch::Tensor x, torch::Tensor out, torch::Tensor scale,
                torch::Tensor amax_ws);
 void fp8_epilogue(torch::Tensor out, torch::Tensor y32, torch::Tensor w_scale,
@@ -74,6 +76,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("sinks") = py::none());
     m.def("qkv_rope_update", &qkv_rope_update, "fused qkv->rope->cache-write (CDNA4)");
     m.def("skinny_gemm", &skinny_gemm, "skinny-M weight-streaming GEMM (CDNA4)",
+          py::arg("y"), py::arg("x"), py::arg("w"), py::arg("bias") = py::none());
+    m.def("skinny_gemm4", &skinny_gemm4,
+          "skinny-M GEMM v4: asm loads + hand-counted vmcnt pipeline (CDNA4)",
           py::arg("y"), py::arg("x"), py::arg("w"), py::arg("bias") = py::none());
     m.def("skinny_gemm3", &skinny_gemm3,
           "skinny-M GEMM v3: full-K register accum, depth-3 prefetch (CDNA4)",

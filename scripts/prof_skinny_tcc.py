@@ -29,6 +29,8 @@ for i in range(40):
         C.skinny_gemm2(y, y32, x, w, None, ks)
     elif mode == "skinny3":
         C.skinny_gemm3(y, x, w, None)
+    elif mode == "skinny4":
+        C.skinny_gemm4(y, x, w, None)
     elif mode == "lt":
         C.lt_linear(y, x, w, None)
     else:
