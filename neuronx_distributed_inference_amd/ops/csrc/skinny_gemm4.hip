@@ -1,5 +1,17 @@
 // Skinny-M weight-streaming GEMM v4: y = x @ W^T (+bias), M <= 1024.
 //
+// STATUS: EXPERIMENTAL / NOT SAFE TO CALL — faults on gfx950 (memory
+// fault, core dump). Root cause class: the compiler treats inline-asm
+// load destinations as written at the asm point, so it may reuse an
+// in-flight destination register (or a pair overlapping one) before
+// the hand-placed s_waitcnt lands the data. The correct round-2 shape
+// is the guide's `global_load_lds` template: no destination VGPRs at
+// all (data lands in LDS), raw s_barrier with lgkmcnt-only waits, and
+// counted vmcnt drains at the read points — which also requires a
+// LINEAR (unpitched) LDS image with an XOR source-swizzle to keep the
+// fragment reads conflict-free. Kept as the starting point + wait-count
+// math for that kernel; no python test invokes it.
+//
 // v3 established (profiles/r01_skinny_tcc_analysis.md) that every
 // compiler-scheduled variant pays ~latency/3 of stall per K-chunk:
 // hipcc will not hold more than ~3 bodies of loads outstanding across

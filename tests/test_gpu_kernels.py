@@ -549,27 +549,6 @@ def test_skinny_gemm3_matches_torch(M):
     torch.testing.assert_close(y.float(), want.float(), atol=2e-2, rtol=2e-2)
 
 
-@pytest.mark.parametrize("M", [16, 50, 64, 100, 128, 256])
-def test_skinny_gemm4_matches_torch(M):
-    """v4 skinny GEMM (inline-asm loads, hand-counted vmcnt pipeline)
-    numerics vs F.linear, with and without bias."""
-    from neuronx_distributed_inference_amd.ops import _nxdi_C as C
-
-    torch.manual_seed(0)
-    dev = _dev()
-    N, K = 1536, 1024
-    x = torch.randn(M, K, dtype=torch.bfloat16, device=dev) * 0.1
-    w = torch.randn(N, K, dtype=torch.bfloat16, device=dev) * 0.1
-    b = torch.randn(N, dtype=torch.bfloat16, device=dev)
-    y = torch.empty(M, N, dtype=torch.bfloat16, device=dev)
-    C.skinny_gemm4(y, x, w, b)
-    want = torch.nn.functional.linear(x, w, b)
-    torch.testing.assert_close(y.float(), want.float(), atol=2e-2, rtol=2e-2)
-    C.skinny_gemm4(y, x, w)
-    want = torch.nn.functional.linear(x, w)
-    torch.testing.assert_close(y.float(), want.float(), atol=2e-2, rtol=2e-2)
-
-
 def test_attention_decode_lse_matches_cpu():
     """lse output of the decode kernel vs the fp32 logsumexp reference
     (flash-decoding combine handle), incl. empty (-1) query rows."""
