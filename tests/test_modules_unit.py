@@ -312,3 +312,35 @@ def test_composite_config_json_roundtrip():
         # generic load reconstructs the concrete class by dotted path
         back2 = InferenceConfig.load(d)
         assert type(back2).__name__ == "MllamaInferenceConfig"
+
+
+def test_qwen2vl_config_json_roundtrip():
+    """Composite configs with extra scalar attrs (video_token_id) and a
+    nested text InferenceConfig survive save()/load()."""
+    import tempfile
+
+    from neuronx_distributed_inference_amd.config import EngineConfig
+    from neuronx_distributed_inference_amd.models.qwen2_vl import (
+        Qwen2VLInferenceConfig,
+    )
+
+    ec = EngineConfig(batch_size=1, max_batch_size=1, seq_len=64, max_length=64,
+                      torch_dtype="float32", tp_degree=1)
+    tc = dict(vocab_size=160, hidden_size=48, intermediate_size=96,
+              num_hidden_layers=2, num_attention_heads=4,
+              num_key_value_heads=2, max_position_embeddings=128,
+              rms_norm_eps=1e-5, rope_theta=10000.0, tie_word_embeddings=False,
+              rope_scaling={"rope_type": "default", "mrope_section": [2, 2, 2]})
+    vc = dict(depth=2, embed_dim=32, hidden_size=48, num_heads=4,
+              in_channels=3, patch_size=4, temporal_patch_size=2,
+              spatial_merge_size=2, mlp_ratio=2, hidden_act="quick_gelu")
+    cfg = Qwen2VLInferenceConfig(neuron_config=ec, text_config=tc,
+                                 vision_config=vc, image_token_id=159,
+                                 video_token_id=156)
+    with tempfile.TemporaryDirectory() as d:
+        cfg.save(d)
+        back = Qwen2VLInferenceConfig.load(d)
+        assert back.text_config.hidden_size == 48
+        assert back.text_config.rope_scaling["mrope_section"] == [2, 2, 2]
+        assert back.video_token_id == 156
+        assert back.vision_config["embed_dim"] == 32
