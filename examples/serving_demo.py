@@ -14,4 +14,5 @@ def serve(model, tokenizer, prompts):
         tokenizer(p, return_tensors="pt").input_ids[0], max_new_tokens=128)
         for p in prompts]
     results = eng.run()
+    print(eng.metrics())  # TTFT/TPOT percentiles, prefix-hit rate
     return {rid: tokenizer.decode(results[rid]) for rid in ids}

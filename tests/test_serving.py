@@ -298,3 +298,21 @@ def test_serving_metrics():
     assert m["ttft_p50_s"] >= 0 and m["ttft_p99_s"] >= m["ttft_p50_s"] - 1e-9
     assert m["tpot_p50_s"] >= 0
     assert 0.0 <= m["prefix_hit_rate"] <= 1.0
+
+
+@torch.inference_mode()
+def test_serving_bucketed_chunks_match_ragged():
+    """Intermediate-chunk padding (bucket_chunks) must not change any
+    generated token vs exact ragged chunking."""
+    m = _model(batch=2)
+    prompts = [torch.randint(0, 128, (n,)) for n in (13, 21)]
+    m.reset()
+    eng = ServingEngine(m, chunk_size=8, bucket_chunks=False)
+    ids = [eng.add_request(p, max_new_tokens=6) for p in prompts]
+    ragged = eng.run()
+    m.reset()
+    eng2 = ServingEngine(m, chunk_size=8, bucket_chunks=True)
+    ids2 = [eng2.add_request(p, max_new_tokens=6) for p in prompts]
+    bucketed = eng2.run()
+    for a, b in zip(ids, ids2):
+        assert ragged[a] == bucketed[b], (ragged[a], bucketed[b])
