@@ -429,11 +429,10 @@ class CausalLMBase(nn.Module):
         # window-sized chunks to cap the attention working set; each chunk
         # runs the multi-token prefix path (causal within the chunk, full
         # prior from the cache), which is also the chunked-prefill compute.
+        # attention_mask=None (the common dense/unpadded batch, incl. one
+        # dropped above) is fine: the last-token gather falls back to S-1.
         wce = getattr(self.ec, "windowed_context_encoding_size", 0)
-        if (
-            S > 1 and not self.kv_cache_populated and wce and S > wce
-            and attention_mask is not None
-        ):
+        if S > 1 and not self.kv_cache_populated and wce and S > wce:
             return self._windowed_context_encoding(
                 input_ids, attention_mask, position_ids, seq_ids, sampling_params, wce
             )
@@ -499,7 +498,10 @@ class CausalLMBase(nn.Module):
             chunks.append(t.view(B, end - start))
             start = end
         all_tok = torch.cat(chunks, dim=1)              # (B, S) greedy per pos
-        last = (attention_mask.sum(dim=-1).long() - 1).clamp(min=0)
+        if attention_mask is not None:
+            last = (attention_mask.sum(dim=-1).long() - 1).clamp(min=0)
+        else:  # dense/unpadded batch: every row's last token is at S-1
+            last = torch.full((B,), S - 1, dtype=torch.long, device=all_tok.device)
         tokens = all_tok.gather(1, last.view(B, 1)).view(B)
         self.kv_cache_populated = True
         return ModelOutput(tokens=tokens, logits=None)

@@ -97,8 +97,12 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
     if (ring) {
         // RING cache (compact sliding-window layers): slot s holds the
         // newest position == s (mod Smax); all written slots are scanned
-        // and masked by true position below
-        pmax = positions[(long)b * Sq + Sq - 1];
+        // and masked by true position below. pmax = max over VALID (>=0)
+        // positions of the row — ragged chunked-prefill batches carry
+        // trailing -1 pads (CPU ref ops/__init__.py uses positions.amax).
+        pmax = -1;
+        for (int s = 0; s < Sq; ++s)
+            pmax = max(pmax, positions[(long)b * Sq + s]);
         span_hi = min(pmax + 1, Smax);
     } else if (window > 0 && eff_len > window) {
         lo = eff_len - window;
@@ -169,7 +173,11 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
         const int g_lo = split_lo + grp * rpg;
         const int g_hi = min(g_lo + rpg, split_hi);
         bool valid_next;
-        long off = row_addr(g_lo, valid_next);
+        // clamp the initial prefetch row: late lane-groups with an empty
+        // range (g_lo >= g_hi) would otherwise load past the valid span
+        // (linear: past the tensor end on the last row/head; paged: index
+        // block_table at l/block_size >= max_blocks)
+        long off = row_addr(min(g_lo, split_hi - 1), valid_next);
         Vec8<CT> kv = load_vec8(kbase + off);
         Vec8<CT> vv = load_vec8(vbase + off);
         for (int l = g_lo; l < g_hi; ++l) {

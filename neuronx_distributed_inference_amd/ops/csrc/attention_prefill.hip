@@ -92,7 +92,10 @@ __global__ __launch_bounds__(256, 2) void flash_prefill_kernel(
     const int wq_hi = min(wq_lo + WQ, S) - 1;   // last valid q row (abs)
 
     const long q_base = (((long)b * Hq + hq) * S) * D;
-    const long kv_row = (seq_ids != nullptr) ? seq_ids[b] : b;
+    // seq_ids[b] can be -1 (ModelWrapper pads CTE batches with seq_id -1);
+    // clamp to row 0 so the cached path never reads before the allocation —
+    // pad-row outputs are discarded by the host anyway.
+    const long kv_row = (seq_ids != nullptr) ? max((long)seq_ids[b], 0L) : b;
     const long kv_base = ((kv_row * Hkv + hkv) * (long)Smax) * D;
 
     // ---- Q fragments into registers (A layout: row = lane&15, k-elems

@@ -108,6 +108,18 @@ class ServingEngine:
         # adapter_ids rows (ref per-request sampling_params / adapter_ids)
         rid = self._next_id
         self._next_id += 1
+        # a static-greedy sampler (no on-device sampling config, or one with
+        # do_sample=False and dynamic=False) ignores the per-request (B,3)
+        # sampling_params rows — refuse silently-greedy sampling requests
+        if top_k != 1 or top_p != 1.0 or temperature != 1.0:
+            sc = self.model.config.neuron_config.on_device_sampling_config
+            if sc is None or (not getattr(sc, "do_sample", False)
+                              and not getattr(sc, "dynamic", False)):
+                raise ValueError(
+                    "request asks for sampling (top_k=%s top_p=%s temperature=%s) "
+                    "but the engine's sampler is static-greedy; configure "
+                    "OnDeviceSamplingConfig(do_sample=True) or dynamic=True"
+                    % (top_k, top_p, temperature))
         self.queue.append(Request(rid, prompt_ids.view(-1).cpu(),
                                   max_new_tokens, top_k, top_p, temperature,
                                   adapter_id))

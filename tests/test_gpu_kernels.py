@@ -588,6 +588,20 @@ def test_ring_cache_decode_kernel():
                                     sliding_window=win, ring=True)
         torch.testing.assert_close(out.float().cpu(), want.float(),
                                    atol=3e-2, rtol=3e-2, msg=f"Sq={Sq} p={plen}")
+    # ragged chunk: trailing -1 pads must not zero the valid queries
+    # (pmax = max over VALID positions, matching the CPU amax semantics)
+    Sq = 4
+    q = torch.randn(B, Hq, Sq, D, dtype=torch.bfloat16, device=dev)
+    pos = torch.tensor([[45, 46, -1, -1], [38, -1, -1, -1]],
+                       dtype=torch.int32, device=dev)
+    out = ops.attention_decode(q, kc, vc, pos, sliding_window=win, ring=True)
+    want = ops.attention_decode(q.float().cpu(), kc.float().cpu(),
+                                vc.float().cpu(), pos.cpu(),
+                                sliding_window=win, ring=True)
+    torch.testing.assert_close(out.float().cpu(), want.float(),
+                               atol=3e-2, rtol=3e-2)
+    assert out[0, :, 0].abs().sum() > 0  # valid rows nonzero
+    assert (out[0, :, 2:].float() == 0).all()  # pad rows zero
 
 
 @torch.inference_mode()

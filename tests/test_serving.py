@@ -316,3 +316,20 @@ def test_serving_bucketed_chunks_match_ragged():
     bucketed = eng2.run()
     for a, b in zip(ids, ids2):
         assert ragged[a] == bucketed[b], (ragged[a], bucketed[b])
+
+
+@torch.inference_mode()
+def test_serving_rejects_sampling_on_static_greedy():
+    """A request asking for sampling while the engine's sampler is
+    static-greedy must be rejected loudly, not silently decoded greedy
+    (ADVICE r01)."""
+    m = _model(batch=2)  # no on_device_sampling_config -> static greedy
+    eng = ServingEngine(m)
+    with pytest.raises(ValueError, match="static-greedy"):
+        eng.add_request(torch.randint(0, 128, (6,)), max_new_tokens=2, top_k=5)
+    with pytest.raises(ValueError, match="static-greedy"):
+        eng.add_request(torch.randint(0, 128, (6,)), max_new_tokens=2,
+                        temperature=0.7)
+    # greedy requests still fine
+    rid = eng.add_request(torch.randint(0, 128, (6,)), max_new_tokens=2)
+    assert len(eng.run()[rid]) == 2
