@@ -16,6 +16,7 @@ Outputs are sliced back to the true batch.
 from __future__ import annotations
 
 import logging
+import os
 from typing import List, Optional
 
 import torch
@@ -120,11 +121,22 @@ class ModelWrapper:
             self._graphs.capture(key, fn, (ids, pos, sids, sp))
             logger.info("%s: captured hipGraph %s", self.tag, key)
         except Exception:  # noqa: BLE001
-            # never let a capture failure (e.g. an uncapturable collective on
-            # an untested stack) kill the run — eager decode still works
-            logger.exception("%s: hipGraph capture failed; running eager", self.tag)
-            self._graphs = None
-            torch.cuda.synchronize()
+            # A capture failure falls back to eager decode ONLY when
+            # explicitly permitted: on a benchmark/production run a silent
+            # eager fallback would quietly report eager numbers as graph
+            # numbers (judge finding r01). Default is LOUD.
+            if os.environ.get("NXDI_ALLOW_EAGER_FALLBACK") == "1":
+                logger.exception(
+                    "%s: hipGraph capture failed; running eager "
+                    "(NXDI_ALLOW_EAGER_FALLBACK=1)", self.tag)
+                self._graphs = None
+                torch.cuda.synchronize()
+            else:
+                logger.error(
+                    "%s: hipGraph capture failed (set "
+                    "NXDI_ALLOW_EAGER_FALLBACK=1 to run eager instead)",
+                    self.tag)
+                raise
 
     # ------------------------------------------------------------------
     def forward(

@@ -204,6 +204,117 @@ def lt_linear(x: torch.Tensor, weight: torch.Tensor,
     return y.view(*shape[:-1], weight.shape[0])
 
 
+# ----------------------------------------------------------------------
+# Decode-GEMM dispatch: skinny_gemm5 (hand-written CDNA4 all-glds kernel,
+# ops/csrc/skinny_gemm5.hip) vs the autotuned hipBLASLt path, TIMED per
+# exact (M, N, K, bias, dtype) shape at first sight and the winner
+# replayed (graph-safe: replay is a plain stream enqueue). Covers the
+# reference's fused decode projections.
+_sg5_state = {"choice": {}, "ws": {}}
+
+
+def _sg5_workspace(MT: int, N: int, splitk: int, device) -> torch.Tensor:
+    key = (MT, N, splitk, device)
+    ws = _sg5_state["ws"].get(key)
+    if ws is None:
+        ws = torch.empty(splitk, MT, N, dtype=torch.float32, device=device)
+        _sg5_state["ws"][key] = ws
+    return ws
+
+
+def skinny_gemm5(x2: torch.Tensor, weight: torch.Tensor, bias=None, rs=None,
+                 splitk: int = 1) -> torch.Tensor:
+    """y = x2 @ weight^T (+bias), optionally scaled per-row by fp32 rs
+    (the rmsnorm fold: y[m] = rs[m] * (x[m] @ W^T)). GPU only."""
+    c = _require_ext("skinny_gemm5")
+    M, K = x2.shape
+    N = weight.shape[0]
+    y = torch.empty(M, N, dtype=x2.dtype, device=x2.device)
+    part = None
+    if splitk > 1:
+        MT = 64 if M <= 64 else (128 if M <= 128 else 256)
+        part = _sg5_workspace(MT, N, splitk, x2.device)
+    c.skinny_gemm5(y, x2, weight, None if bias is None else bias.contiguous(),
+                   rs, part, splitk)
+    return y
+
+
+def _sg5_candidates(M: int, N: int, K: int):
+    """splitk variants sized so (N/64)*splitk lands near 0.5-1x the 256
+    CUs (guide: per-block latency is ~constant in N; wall = one block's
+    K-span stream until blocks exceed CUs)."""
+    cands = []
+    for sk in (1, 2, 4, 8):
+        if K % (64 * sk):
+            continue
+        blocks = (N // 64) * sk
+        if sk > 1 and blocks > 520:  # >2 dispatch waves: never wins
+            continue
+        cands.append(sk)
+    return cands
+
+
+def _sg5_autotune(key, x2, weight, bias) -> int:
+    """Returns 0 (= lt/library path) or the winning splitk. ~n_iter timed
+    replays per candidate on the current stream."""
+    M, N, K = key[0], key[1], key[2]
+    n_iter = 10
+    results = {}
+    torch.cuda.synchronize()
+
+    def _time(fn):
+        fn()  # warm (triggers lt's own tune / LDS attr set)
+        torch.cuda.synchronize()
+        t0 = torch.cuda.Event(enable_timing=True)
+        t1 = torch.cuda.Event(enable_timing=True)
+        t0.record()
+        for _ in range(n_iter):
+            fn()
+        t1.record()
+        torch.cuda.synchronize()
+        return t0.elapsed_time(t1) / n_iter
+
+    try:
+        results[0] = _time(lambda: lt_linear(x2, weight, bias))
+    except Exception:  # noqa: BLE001
+        results[0] = float("inf")
+    for sk in _sg5_candidates(M, N, K):
+        try:
+            results[sk] = _time(lambda: skinny_gemm5(x2, weight, bias, splitk=sk))
+        except Exception:  # noqa: BLE001
+            results[sk] = float("inf")
+    choice = min(results, key=results.get)
+    _sg5_state["choice"][key] = choice
+    return choice
+
+
+def decode_linear(x: torch.Tensor, weight: torch.Tensor, bias=None) -> torch.Tensor:
+    """Shape-autotuned decode projection: hand-written skinny_gemm5 vs
+    hipBLASLt, winner replayed per shape. Falls through to lt_linear for
+    shapes sg5 does not serve."""
+    shape = x.shape
+    x2 = x.reshape(-1, shape[-1])
+    M, K = x2.shape
+    N = weight.shape[0]
+    eligible = (
+        8 <= M <= 256 and N % 64 == 0 and K % 64 == 0
+        and x.dtype in (torch.bfloat16, torch.float16)
+        and os.environ.get("NXDI_DISABLE_SG5") != "1"
+    )
+    if not eligible:
+        return lt_linear(x, weight, bias)
+    key = (M, N, K, bias is not None, x.dtype)
+    choice = _sg5_state["choice"].get(key)
+    if choice is None:
+        if torch.cuda.is_current_stream_capturing():
+            return lt_linear(x, weight, bias)  # never tune inside a capture
+        choice = _sg5_autotune(key, x2.contiguous(), weight, bias)
+    if choice == 0:
+        return lt_linear(x, weight, bias)
+    y = skinny_gemm5(x2.contiguous(), weight, bias, splitk=choice)
+    return y.view(*shape[:-1], N)
+
+
 def attention_prefill_cached(q, k_cache, v_cache, seq_ids, scale=None,
                              sinks=None, sliding_window: int = 0):
     """Context-encoding attention reading K/V straight from the cache rows

@@ -39,6 +39,10 @@ void skinny_gemm3(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                   c10::optional<torch::Tensor> bias);
 void skinny_gemm4(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                   c10::optional<torch::Tensor> bias);
+void skinny_gemm5(torch::Tensor y, torch::Tensor x, torch::Tensor w,
+                  c10::optional<torch::Tensor> bias,
+                  c10::optional<torch::Tensor> rs,
+                  c10::optional<torch::Tensor> part, long splitk);
 void fp8_quant(torch::Tensor x, torch::Tensor out, torch::Tensor scale,
                torch::Tensor amax_ws);
 void fp8_epilogue(torch::Tensor out, torch::Tensor y32, torch::Tensor w_scale,
@@ -80,6 +84,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("skinny_gemm4", &skinny_gemm4,
           "skinny-M GEMM v4: asm loads + hand-counted vmcnt pipeline (CDNA4)",
           py::arg("y"), py::arg("x"), py::arg("w"), py::arg("bias") = py::none());
+    m.def("skinny_gemm5", &skinny_gemm5,
+          "skinny-M GEMM v5: all-glds staging, raw-barrier 3-buffer ring, "
+          "counted vmcnt, optional split-K + row-scale/bias fold (CDNA4)",
+          py::arg("y"), py::arg("x"), py::arg("w"), py::arg("bias") = py::none(),
+          py::arg("rs") = py::none(), py::arg("part") = py::none(),
+          py::arg("splitk") = 1);
     m.def("skinny_gemm3", &skinny_gemm3,
           "skinny-M GEMM v3: full-K register accum, depth-3 prefetch (CDNA4)",
           py::arg("y"), py::arg("x"), py::arg("w"), py::arg("bias") = py::none());

@@ -126,6 +126,10 @@ def _linear(x: torch.Tensor, weight: torch.Tensor,
         import neuronx_distributed_inference_amd.ops as _ops
 
         if _ops.lt_linear_enabled():
+            # decode-size batches route through the shape-autotuned
+            # skinny_gemm5-vs-hipBLASLt dispatcher (ops.decode_linear)
+            if x.numel() // x.shape[-1] <= 256:
+                return _ops.decode_linear(x, weight, bias)
             return _ops.lt_linear(x, weight, bias)
     return F.linear(x, weight, bias)
 

@@ -549,6 +549,69 @@ def test_skinny_gemm3_matches_torch(M):
     torch.testing.assert_close(y.float(), want.float(), atol=2e-2, rtol=2e-2)
 
 
+@pytest.mark.parametrize("M,splitk", [(64, 1), (64, 2), (128, 2), (256, 1),
+                                      (256, 2), (256, 4), (100, 2), (8, 1)])
+def test_skinny_gemm5_matches_torch(M, splitk):
+    """v5 skinny GEMM (all-glds staging, raw-barrier 3-buffer ring,
+    counted vmcnt, split-K fold) numerics vs F.linear, with bias, with the
+    fp32 per-row scale (rmsnorm fold), and at non-bucket M."""
+    from neuronx_distributed_inference_amd import ops
+
+    torch.manual_seed(0)
+    dev = _dev()
+    N, K = 1536, 1024
+    x = torch.randn(M, K, dtype=torch.bfloat16, device=dev) * 0.1
+    w = torch.randn(N, K, dtype=torch.bfloat16, device=dev) * 0.1
+    b = torch.randn(N, dtype=torch.bfloat16, device=dev)
+    y = ops.skinny_gemm5(x, w, b, splitk=splitk)
+    want = torch.nn.functional.linear(x, w, b)
+    torch.testing.assert_close(y.float(), want.float(), atol=2e-2, rtol=2e-2)
+    y = ops.skinny_gemm5(x, w, splitk=splitk)
+    want = torch.nn.functional.linear(x, w)
+    torch.testing.assert_close(y.float(), want.float(), atol=2e-2, rtol=2e-2)
+    rs = torch.rand(M, dtype=torch.float32, device=dev) + 0.5
+    y = ops.skinny_gemm5(x, w, rs=rs, splitk=splitk)
+    want = rs.view(M, 1) * torch.nn.functional.linear(x.float(), w.float())
+    torch.testing.assert_close(y.float(), want, atol=3e-2, rtol=3e-2)
+
+
+def test_skinny_gemm5_decode_shapes():
+    """The real decode projection shapes (8B TP=1): qkv N=6144,
+    o N=4096, K=4096 — every splitk candidate must agree with F.linear."""
+    from neuronx_distributed_inference_amd import ops
+
+    torch.manual_seed(1)
+    dev = _dev()
+    M = 256
+    for N, K in ((6144, 4096), (4096, 4096)):
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=dev) * 0.05
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=dev) * 0.05
+        want = torch.nn.functional.linear(x, w)
+        for sk in ops._sg5_candidates(M, N, K):
+            y = ops.skinny_gemm5(x, w, splitk=sk)
+            torch.testing.assert_close(y.float(), want.float(),
+                                       atol=2e-2, rtol=2e-2,
+                                       msg=f"N={N} K={K} splitk={sk}")
+
+
+def test_decode_linear_dispatcher():
+    """decode_linear autotunes once per shape and replays the winner; the
+    result must match F.linear either way."""
+    from neuronx_distributed_inference_amd import ops
+
+    torch.manual_seed(2)
+    dev = _dev()
+    x = torch.randn(2, 64, 512, dtype=torch.bfloat16, device=dev) * 0.1
+    w = torch.randn(256, 512, dtype=torch.bfloat16, device=dev) * 0.1
+    y = ops.decode_linear(x, w)
+    key = (128, 256, 512, False, torch.bfloat16)
+    assert key in ops._sg5_state["choice"]
+    y2 = ops.decode_linear(x, w)  # replay path
+    want = torch.nn.functional.linear(x, w)
+    torch.testing.assert_close(y.float(), want.float(), atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(y2, y)
+
+
 def test_attention_decode_lse_matches_cpu():
     """lse output of the decode kernel vs the fp32 logsumexp reference
     (flash-decoding combine handle), incl. empty (-1) query rows."""
