@@ -223,7 +223,7 @@ def _sg5_workspace(MT: int, N: int, splitk: int, device) -> torch.Tensor:
 
 
 def skinny_gemm5(x2: torch.Tensor, weight: torch.Tensor, bias=None, rs=None,
-                 splitk: int = 1) -> torch.Tensor:
+                 splitk: int = 1, mode: int = 0) -> torch.Tensor:
     """y = x2 @ weight^T (+bias), optionally scaled per-row by fp32 rs
     (the rmsnorm fold: y[m] = rs[m] * (x[m] @ W^T)). GPU only."""
     c = _require_ext("skinny_gemm5")
@@ -235,7 +235,7 @@ def skinny_gemm5(x2: torch.Tensor, weight: torch.Tensor, bias=None, rs=None,
         MT = 64 if M <= 64 else (128 if M <= 128 else 256)
         part = _sg5_workspace(MT, N, splitk, x2.device)
     c.skinny_gemm5(y, x2, weight, None if bias is None else bias.contiguous(),
-                   rs, part, splitk)
+                   rs, part, splitk, mode)
     return y
 
 

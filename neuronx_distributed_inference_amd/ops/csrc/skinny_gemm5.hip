@@ -93,9 +93,25 @@ DEV_INLINE void vm_wait() {
     asm volatile("s_waitcnt vmcnt(%c0)" ::"i"(N) : "memory");
 }
 
+// register-staged fallback piece: same image/swizzle as glds_rows128 but
+// via an ordinary 16B load + ds_write (debug MODE 2: isolates glds
+// mechanics from the addressing/swizzle math)
+DEV_INLINE void regstage_rows128(const char* gsrc_rowbase, long grow_stride,
+                                 char* lds_dst_base) {
+    const int lane = threadIdx.x & 63;
+    const int row = lane >> 3;
+    const int slot = (lane & 7) ^ row;
+    const char* src = gsrc_rowbase + (long)row * grow_stride + slot * 16;
+    int4 v = *reinterpret_cast<const int4*>(src);
+    *reinterpret_cast<int4*>(lds_dst_base + lane * 16) = v;
+}
+
 // MT: padded M tile (64/128/256). 8 waves; each owns 32 rows x
 // (BN*8*32/MT/64) n-fragments.
-template <typename T, int MT>
+// MODE 0: production (3-buffer glds ring, raw barrier, counted vmcnt)
+// MODE 1: debug — serial single-buffer glds (vmcnt(0) + __syncthreads)
+// MODE 2: debug — serial single-buffer register staging (no glds)
+template <typename T, int MT, int MODE = 0>
 __global__ __launch_bounds__(512, 2) void sk5_kernel(
     T* __restrict__ y,           // (M, N) written when splitk == 1
     float* __restrict__ part,    // (splitk, MT, N) fp32, when splitk > 1
@@ -209,20 +225,40 @@ __global__ __launch_bounds__(512, 2) void sk5_kernel(
         if ((t) + 2 < nchunks) STAGE((t) + 2, b2);                           \
     } while (0)
 
-    // ---- prologue: chunks 0,1 in flight ----
-    STAGE(0, 0);
-    if (nchunks > 1) STAGE(1, 1);
+    if constexpr (MODE == 0) {
+        // ---- prologue: chunks 0,1 in flight ----
+        STAGE(0, 0);
+        if (nchunks > 1) STAGE(1, 1);
 
-    int t = 0;
-    while (t + 3 <= nchunks) {
-        BODY(t + 0, 0, 2);
-        BODY(t + 1, 1, 0);
-        BODY(t + 2, 2, 1);
-        t += 3;
+        int t = 0;
+        while (t + 3 <= nchunks) {
+            BODY(t + 0, 0, 2);
+            BODY(t + 1, 1, 0);
+            BODY(t + 2, 2, 1);
+            t += 3;
+        }
+        if (t < nchunks) BODY(t, 0, 2);
+        if (t + 1 < nchunks) BODY(t + 1, 1, 0);
+        // (nchunks % 3 == 2 leaves buffers 0,1 consumed; == 1 leaves 0)
+    } else {
+        // debug: fully serial, single buffer
+        for (int t = 0; t < nchunks; ++t) {
+            if constexpr (MODE == 1) {
+                STAGE(t, 0);
+                vm_wait<0>();
+            } else {
+                const long koff = (long)t * 128;
+#pragma unroll
+                for (int i = 0; i < NXG; ++i)
+                    regstage_rows128(xsrc[i] + koff, (long)K * sizeof(T),
+                                     xdst[0] + i * 1024);
+                regstage_rows128(wsrc + koff, (long)K * sizeof(T), wdst[0]);
+            }
+            __syncthreads();
+            FRAGS_AND_MFMA(0);
+            __syncthreads();
+        }
     }
-    if (t < nchunks) BODY(t, 0, 2);
-    if (t + 1 < nchunks) BODY(t + 1, 1, 0);
-    // (nchunks % 3 == 2 leaves buffers 0,1 consumed; == 1 leaves 0)
 
 #undef BODY
 #undef FRAGS_AND_MFMA
@@ -331,7 +367,7 @@ __global__ void sk5_reduce_kernel(T* __restrict__ y,
     }
 }
 
-template <typename T, int MT>
+template <typename T, int MT, int MODE = 0>
 void launch_sk5(torch::Tensor& y, const torch::Tensor& x, const torch::Tensor& w,
                 const c10::optional<torch::Tensor>& bias,
                 const c10::optional<torch::Tensor>& rs,
@@ -341,7 +377,7 @@ void launch_sk5(torch::Tensor& y, const torch::Tensor& x, const torch::Tensor& w
     constexpr int LDS = NBUF * CHUNK_B;
     static bool attr_set = false;
     if (!attr_set) {
-        (void)hipFuncSetAttribute((const void*)sk5_kernel<T, MT>,
+        (void)hipFuncSetAttribute((const void*)sk5_kernel<T, MT, MODE>,
                                   hipFuncAttributeMaxDynamicSharedMemorySize, LDS);
         attr_set = true;
     }
@@ -352,7 +388,7 @@ void launch_sk5(torch::Tensor& y, const torch::Tensor& x, const torch::Tensor& w
     TORCH_CHECK(splitk == 1 || part_p != nullptr,
                 "splitk > 1 needs a (splitk, MT, N) fp32 workspace");
     dim3 grid((N / BN) * splitk);
-    hipLaunchKernelGGL((sk5_kernel<T, MT>), grid, dim3(512), LDS, stream,
+    hipLaunchKernelGGL((sk5_kernel<T, MT, MODE>), grid, dim3(512), LDS, stream,
                        (T*)y.data_ptr(), part_p, (const T*)x.data_ptr(),
                        (const T*)w.data_ptr(), rs_p, bias_p, M, N, K, splitk);
     HIP_CHECK_KERNEL();
@@ -372,7 +408,7 @@ void launch_sk5(torch::Tensor& y, const torch::Tensor& x, const torch::Tensor& w
 void skinny_gemm5(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                   c10::optional<torch::Tensor> bias,
                   c10::optional<torch::Tensor> rs,
-                  c10::optional<torch::Tensor> part, long splitk) {
+                  c10::optional<torch::Tensor> part, long splitk, long mode) {
     TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && y.dim() == 2, "2-D tensors");
     TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && y.is_contiguous(),
                 "contiguous tensors required");
@@ -384,14 +420,23 @@ void skinny_gemm5(torch::Tensor y, torch::Tensor x, torch::Tensor w,
     const int MT = M <= 64 ? 64 : (M <= 128 ? 128 : 256);
     TORCH_CHECK(M >= 8, "M >= 8 required");
 
+#define SK5_MODE(T, MTV)                                                      \
+    do {                                                                      \
+        if (mode == 1)                                                        \
+            launch_sk5<T, MTV, 1>(y, x, w, bias, rs, part, (int)splitk);      \
+        else if (mode == 2)                                                   \
+            launch_sk5<T, MTV, 2>(y, x, w, bias, rs, part, (int)splitk);      \
+        else                                                                  \
+            launch_sk5<T, MTV, 0>(y, x, w, bias, rs, part, (int)splitk);      \
+    } while (0)
 #define SK5_T(T)                                                              \
     do {                                                                      \
         if (MT == 64)                                                         \
-            launch_sk5<T, 64>(y, x, w, bias, rs, part, (int)splitk);          \
+            SK5_MODE(T, 64);                                                  \
         else if (MT == 128)                                                   \
-            launch_sk5<T, 128>(y, x, w, bias, rs, part, (int)splitk);         \
+            SK5_MODE(T, 128);                                                 \
         else                                                                  \
-            launch_sk5<T, 256>(y, x, w, bias, rs, part, (int)splitk);         \
+            SK5_MODE(T, 256);                                                 \
     } while (0)
 
     if (x.scalar_type() == at::ScalarType::BFloat16) {
@@ -402,4 +447,5 @@ void skinny_gemm5(torch::Tensor y, torch::Tensor x, torch::Tensor w,
         TORCH_CHECK(false, "skinny_gemm5: bf16/f16 only");
     }
 #undef SK5_T
+#undef SK5_MODE
 }

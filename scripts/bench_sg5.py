@@ -8,10 +8,14 @@ Usage: python scripts/bench_sg5.py [--m 256] [--iters 40]
 
 import argparse
 import json
+import os
+import sys
 
-import torch
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
-from neuronx_distributed_inference_amd import ops
+import torch  # noqa: E402
+
+from neuronx_distributed_inference_amd import ops  # noqa: E402
 
 
 def bench_shape(M, N, K, iters=40, ncopies=10):
