@@ -262,6 +262,20 @@ class DecoderModel(nn.Module):
         self.last_hidden = h_last
         logits_shard = self.lm_head(h_last)  # (B, 1|S, V/tp)
 
+        # rejection-sampling speculation reads the per-position top-Kg
+        # distribution of every draft/verify pass (ref draft_probs /
+        # target_probs flow, model_base.py:1678-1759); stored as an
+        # attribute so FusedSpeculationModel can read it inside the same
+        # graph capture (same pattern as last_hidden)
+        if getattr(self, "spec_capture_topk", 0):
+            from neuronx_distributed_inference_amd.parallel.ops import (
+                distributed_topk,
+            )
+
+            self.spec_topk = distributed_topk(
+                logits_shard, int(self.spec_capture_topk), dim=-1
+            )
+
         tokens = None
         logits = None
         if self.sampler is not None:

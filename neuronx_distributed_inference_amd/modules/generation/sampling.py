@@ -169,3 +169,140 @@ class DataParallelSampler(nn.Module):
             local = idxs.gather(1, pick.view(-1, 1)).squeeze(1)
         # all-gather the batch shards back (every rank needs the full batch)
         return mappings.gather_along_dim(local.view(-1, 1), dim=0).view(-1)
+
+
+# ----------------------------------------------------------------------
+# Rejection sampling for speculative decoding (ref model_base.py:1678-1759:
+# `_adjust_target_probs`, `_speculative_mask`, `_speculative_token_selection`
+# and the on-device rand accept draw at :1706). All tensor ops, graph-safe.
+
+
+def filtered_probs(vals: torch.Tensor, sampling_params: torch.Tensor) -> torch.Tensor:
+    """Probabilities after temperature/top-k/top-p over sorted-descending
+    top-`Kg` logits `vals` (B, S, Kg). `sampling_params` (B,3) =
+    [top_k, top_p, temperature]. Mirrors the fused sampling kernel's
+    filtering: temperature -> top-k mask (vals sorted, so a position
+    cutoff) -> softmax -> top-p nucleus (keep tokens whose PRECEDING
+    cumulative mass < top_p; the first is always kept) -> renormalize."""
+    B, S, Kg = vals.shape
+    tk = sampling_params[:, 0].view(B, 1, 1)
+    tp = sampling_params[:, 1].view(B, 1, 1)
+    temp = sampling_params[:, 2].view(B, 1, 1).clamp(min=1e-6)
+    x = vals.float() / temp
+    pos = torch.arange(Kg, device=vals.device).view(1, 1, Kg)
+    x = x.masked_fill(pos >= tk, float("-inf"))
+    probs = torch.softmax(x, dim=-1)
+    cum = probs.cumsum(-1)
+    keep = (cum - probs) < tp
+    probs = probs * keep
+    return probs / probs.sum(-1, keepdim=True).clamp(min=1e-30)
+
+
+def _select_value(from_idx: torch.Tensor, from_val: torch.Tensor,
+                  ids: torch.Tensor) -> torch.Tensor:
+    """value of token `ids` (B,k) under the (indices, values) distribution
+    (B,k,Kg); 0 where absent (ref `_select_from`, model_base.py:1664)."""
+    eq = from_idx == ids.unsqueeze(-1)
+    return (from_val * eq).sum(-1)
+
+
+def _map_probs(from_idx: torch.Tensor, from_val: torch.Tensor,
+               to_idx: torch.Tensor) -> torch.Tensor:
+    """re-express (from_idx, from_val) probs on the `to_idx` support:
+    out[b,j,i] = from_val at token to_idx[b,j,i] (0 if absent)."""
+    eq = to_idx.unsqueeze(-1) == from_idx.unsqueeze(-2)  # (B,k,Kt,Kf)
+    return (from_val.unsqueeze(-2) * eq).sum(-1)
+
+
+def sample_from_probs(probs: torch.Tensor, rand: torch.Tensor) -> torch.Tensor:
+    """Inverse-CDF sample: probs (..., Kg) need not be normalized; rand
+    (...) in [0,1). Returns LOCAL candidate indices (...)."""
+    cum = probs.cumsum(-1)
+    total = cum[..., -1:]
+    return (cum < rand.unsqueeze(-1) * total).sum(-1).clamp(max=probs.shape[-1] - 1)
+
+
+class RejectionSampler(nn.Module):
+    """Speculative accept/resample with true rejection sampling so spec
+    decode works under top-k/top-p sampling (ref model_base.py:1678-1759).
+
+    Contract: draft token d_j was sampled from q_j; accept d_j iff
+    rand < min(1, p_j(d_j)/q_j(d_j)); at the first rejection resample from
+    normalize(clamp(p_j - q_j, 0)); if all k accepted, sample the bonus
+    from p_k. Output distribution == target distribution regardless of
+    the draft (the speculative-sampling theorem), which the CPU oracle
+    test checks empirically."""
+
+    def __init__(self, config: OnDeviceSamplingConfig, vocab_size: int,
+                 deterministic: Optional[bool] = None):
+        super().__init__()
+        self.config = config
+        self.k_g = min(config.global_topk, vocab_size)
+        self.deterministic = (
+            config.deterministic if deterministic is None else deterministic
+        )
+        self._gen: Optional[torch.Generator] = None
+
+    def _rand(self, *shape, device) -> torch.Tensor:
+        if self.deterministic:
+            if self._gen is None or self._gen.device != torch.device(device):
+                self._gen = torch.Generator(device=device)
+                self._gen.manual_seed(0)
+            return torch.rand(*shape, device=device, generator=self._gen)
+        return torch.rand(*shape, device=device)
+
+    def dist(self, logits_shard: torch.Tensor,
+             sampling_params: torch.Tensor):
+        """(B,S,V/tp) vocab-sharded logits -> filtered probs + GLOBAL
+        vocab indices (B,S,Kg), sorted descending."""
+        vals, idxs = pops.distributed_topk(logits_shard, self.k_g, dim=-1)
+        return filtered_probs(vals, sampling_params), idxs
+
+    def sample(self, probs: torch.Tensor, idxs: torch.Tensor,
+               rand: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """Sample token ids (B,S) from filtered probs/indices (B,S,Kg)."""
+        if rand is None:
+            rand = self._rand(*probs.shape[:-1], device=probs.device)
+        local = sample_from_probs(probs, rand)
+        return idxs.gather(-1, local.unsqueeze(-1)).squeeze(-1)
+
+    def accept(self, draft_ids, q_probs, q_idx, p_probs, p_idx,
+               pad_id: int = -1):
+        """draft_ids (B,k); q_* (B,k,Kg) draft dists; p_* (B,k+1,Kg)
+        target dists. Returns (tokens (B,k+1) `pad_id`-padded after the
+        first rejection's replacement, n_accept (B,))."""
+        B, k = draft_ids.shape
+        device = draft_ids.device
+        p_head, p_head_idx = p_probs[:, :k], p_idx[:, :k]
+        p_d = _select_value(p_head_idx, p_head, draft_ids)
+        q_d = _select_value(q_idx, q_probs, draft_ids)
+        ratio = (p_d / q_d.clamp(min=1e-30)).clamp(max=1.0)
+        rand_acc = self._rand(B, k, device=device)
+        acc = (rand_acc < ratio).long()
+        acc_prefix = acc.cumprod(dim=1)                      # contiguous prefix
+        n_accept = acc_prefix.sum(dim=1)                     # (B,)
+
+        # adjusted residual distribution on the target support
+        q_on_p = _map_probs(q_idx, q_probs, p_head_idx)      # (B,k,Kg)
+        adj = (p_head - q_on_p).clamp(min=0)
+        s = adj.sum(-1, keepdim=True)
+        is_zero = s < 1e-30
+        # ref: where the residual vanishes (q covers p exactly) fall back
+        # to a uniform over the support (ref sets all-ones)
+        adj = torch.where(is_zero, torch.ones_like(adj), adj / s.clamp(min=1e-30))
+        adj = torch.cat([adj, p_probs[:, k:]], dim=1)        # (B,k+1,Kg)
+        rand_res = self._rand(B, k + 1, device=device)
+        local = sample_from_probs(adj, rand_res)
+        res_ids = p_idx.gather(-1, local.unsqueeze(-1)).squeeze(-1)  # (B,k+1)
+
+        idxpos = torch.arange(k + 1, device=device).view(1, -1)
+        acc_mask = torch.cat(
+            [acc_prefix.bool(),
+             torch.zeros(B, 1, dtype=torch.bool, device=device)], dim=1)
+        draft_ext = torch.cat(
+            [draft_ids, torch.zeros(B, 1, dtype=draft_ids.dtype, device=device)],
+            dim=1)
+        tokens = torch.where(acc_mask, draft_ext, res_ids)
+        keep = idxpos <= n_accept.view(B, 1)
+        tokens = torch.where(keep, tokens, torch.full_like(tokens, pad_id))
+        return tokens, n_accept

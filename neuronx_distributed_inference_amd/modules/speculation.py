@@ -124,14 +124,30 @@ class FusedSpeculationModel(torch.nn.Module):
 
     ACCEPT_PAD = -1
 
-    def __init__(self, target, draft, speculation_length: int):
+    def __init__(self, target, draft, speculation_length: int,
+                 rejection_config=None):
+        """`rejection_config`: an OnDeviceSamplingConfig -> true rejection
+        sampling (accept d_j iff rand < p(d_j)/q(d_j), resample from the
+        residual; ref model_base.py:1678-1759), so speculation works with
+        top-k/top-p sampling. None -> exact greedy token match."""
         super().__init__()
         self.target = target
         self.draft = draft
         self.k = speculation_length
+        self.rejection = None
+        if rejection_config is not None:
+            from neuronx_distributed_inference_amd.modules.generation.sampling import (
+                RejectionSampler,
+            )
+
+            vocab = getattr(target, "vocab_size", None) or target.config.vocab_size
+            self.rejection = RejectionSampler(rejection_config, vocab)
+            target.spec_capture_topk = self.rejection.k_g
+            draft.spec_capture_topk = self.rejection.k_g
 
     def forward(self, input_ids: torch.Tensor, position_ids: torch.Tensor,
-                seq_ids: torch.Tensor):
+                seq_ids: torch.Tensor,
+                sampling_params: Optional[torch.Tensor] = None):
         """input_ids (B,1): last committed (unfed) token; position_ids (B,1)
         its position. Returns (accepted_tokens (B,k+1) padded with -1,
         n_accept (B,))."""
@@ -142,6 +158,9 @@ class FusedSpeculationModel(torch.nn.Module):
         B = input_ids.shape[0]
         k = self.k
         device = input_ids.device
+        if self.rejection is not None:
+            return self._forward_rejection(
+                input_ids, position_ids, seq_ids, sampling_params)
         d_in = input_ids
         cand = []
         for i in range(k):
@@ -170,6 +189,64 @@ class FusedSpeculationModel(torch.nn.Module):
                         torch.full_like(cand_ext, self.ACCEPT_PAD)),
         )
         return accepted, n_accept
+
+    def _forward_rejection(self, input_ids, position_ids, seq_ids,
+                           sampling_params):
+        """Sampled draft + rejection accept + residual resample, all
+        tensor-side (one hipGraph covers the round)."""
+        from neuronx_distributed_inference_amd.modules.attention.attention_base import (
+            PHASE_TKG,
+        )
+        from neuronx_distributed_inference_amd.modules.generation.sampling import (
+            prepare_sampling_params,
+        )
+
+        B = input_ids.shape[0]
+        k = self.k
+        device = input_ids.device
+        rj = self.rejection
+        if sampling_params is None:
+            c = rj.config
+            sampling_params = prepare_sampling_params(
+                B, c.top_k if c.do_sample else 1, c.top_p, c.temperature,
+                device=device)
+
+        d_in = input_ids
+        cand = []
+        q_probs, q_idx = [], []
+        for i in range(k):
+            # the draft model's own sampler output is ignored: the round
+            # samples d_j itself from the SAME filtered distribution it
+            # uses for q(d_j), so accept ratios are exact by construction
+            self.draft(d_in, None, position_ids + i, seq_ids,
+                       sampling_params, phase=PHASE_TKG)
+            vals, idxs = self.draft.spec_topk           # (B,1,Kg)
+            from neuronx_distributed_inference_amd.modules.generation.sampling import (
+                filtered_probs,
+            )
+
+            qp = filtered_probs(vals, sampling_params)
+            d_tok = rj.sample(qp, idxs)                 # (B,1)
+            d_in = d_tok.view(B, 1)
+            cand.append(d_in)
+            q_probs.append(qp)
+            q_idx.append(idxs)
+        cand = torch.cat(cand, dim=1)                   # (B,k)
+        q_probs = torch.cat(q_probs, dim=1)             # (B,k,Kg)
+        q_idx = torch.cat(q_idx, dim=1)
+
+        verify_in = torch.cat([input_ids, cand], dim=1)
+        verify_pos = position_ids + torch.arange(k + 1, device=device).view(1, -1)
+        self.target(verify_in, None, verify_pos, seq_ids,
+                    sampling_params, phase=PHASE_TKG)
+        t_vals, t_idx = self.target.spec_topk           # (B,k+1,Kg)
+        from neuronx_distributed_inference_amd.modules.generation.sampling import (
+            filtered_probs,
+        )
+
+        p_probs = filtered_probs(t_vals, sampling_params)
+        return rj.accept(cand, q_probs, q_idx, p_probs, t_idx,
+                         pad_id=self.ACCEPT_PAD)
 
 
 class FusedSpeculativeAdapter:

@@ -215,10 +215,16 @@ __global__ __launch_bounds__(512, 2) void sk5_kernel(
         }                                                                    \
     } while (0)
 
-    // body: chunk t computed from buffer b, chunk t+2 staged into b2
+    // body: chunk t computed from buffer b, chunk t+2 staged into b2.
+    // Steady state has chunks t and t+1 in flight at the wait point, so
+    // vmcnt(G) leaves t+1 flying and guarantees t landed. At the FINAL
+    // chunk nothing extra is in flight — vmcnt(G) would wait for nothing
+    // and the reads would race the last chunk's own DMA (one corrupted
+    // chunk per block, nondeterministic; found by the mode-1/2 bisect).
 #define BODY(t, b, b2)                                                       \
     do {                                                                     \
-        vm_wait<G>();                                                        \
+        if ((t) + 1 < nchunks) vm_wait<G>();                                 \
+        else vm_wait<0>();                                                   \
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                   \
         __builtin_amdgcn_s_barrier();                                        \
         FRAGS_AND_MFMA(b);                                                   \
