@@ -213,44 +213,56 @@ def lt_linear(x: torch.Tensor, weight: torch.Tensor,
 _sg5_state = {"choice": {}, "ws": {}}
 
 
-def _sg5_workspace(MT: int, N: int, splitk: int, device) -> torch.Tensor:
-    key = (MT, N, splitk, device)
+def _sg5_workspace(rows: int, N: int, splitk: int, device) -> torch.Tensor:
+    key = (rows, N, splitk, device)
     ws = _sg5_state["ws"].get(key)
     if ws is None:
-        ws = torch.empty(splitk, MT, N, dtype=torch.float32, device=device)
+        ws = torch.empty(splitk, rows, N, dtype=torch.float32, device=device)
         _sg5_state["ws"][key] = ws
     return ws
 
 
 def skinny_gemm5(x2: torch.Tensor, weight: torch.Tensor, bias=None, rs=None,
-                 splitk: int = 1, mode: int = 0) -> torch.Tensor:
+                 splitk: int = 1, mode: int = 0, mt: int = 0) -> torch.Tensor:
     """y = x2 @ weight^T (+bias), optionally scaled per-row by fp32 rs
-    (the rmsnorm fold: y[m] = rs[m] * (x[m] @ W^T)). GPU only."""
+    (the rmsnorm fold: y[m] = rs[m] * (x[m] @ W^T)). GPU only.
+
+    `mt`: row-block tile (0 = one block over all rows; 64/128 split M
+    across blocks for >1 block/CU co-residency)."""
     c = _require_ext("skinny_gemm5")
     M, K = x2.shape
     N = weight.shape[0]
     y = torch.empty(M, N, dtype=x2.dtype, device=x2.device)
     part = None
     if splitk > 1:
-        MT = 64 if M <= 64 else (128 if M <= 128 else 256)
-        part = _sg5_workspace(MT, N, splitk, x2.device)
+        MT = mt if mt else (64 if M <= 64 else (128 if M <= 128 else 256))
+        rows = ((M + MT - 1) // MT) * MT
+        part = _sg5_workspace(rows, N, splitk, x2.device)
     c.skinny_gemm5(y, x2, weight, None if bias is None else bias.contiguous(),
-                   rs, part, splitk, mode)
+                   rs, part, splitk, mode, mt)
     return y
 
 
 def _sg5_candidates(M: int, N: int, K: int):
-    """splitk variants sized so (N/64)*splitk lands near 0.5-1x the 256
-    CUs (guide: per-block latency is ~constant in N; wall = one block's
-    K-span stream until blocks exceed CUs)."""
+    """(mt, splitk) variants: splitk sized so total blocks land near
+    0.5-2x the 256 CUs (guide: per-block latency ~constant in N); row
+    sub-tiles (mt < M) trade extra W traffic for >1 block/CU
+    co-residency (block-level DMA/compute overlap)."""
     cands = []
-    for sk in (1, 2, 4, 8):
-        if K % (64 * sk):
-            continue
-        blocks = (N // 64) * sk
-        if sk > 1 and blocks > 520:  # >2 dispatch waves: never wins
-            continue
-        cands.append(sk)
+    mts = [0]
+    if M > 128:
+        mts.append(128)
+    if M > 64:
+        mts.append(64)
+    for mt in mts:
+        mb = 1 if mt == 0 else (M + mt - 1) // mt
+        for sk in (1, 2, 4, 8):
+            if K % (64 * sk):
+                continue
+            blocks = (N // 64) * sk * mb
+            if blocks > 560 or (sk > 1 and blocks > 520):
+                continue
+            cands.append((mt, sk))
     return cands
 
 
@@ -278,11 +290,12 @@ def _sg5_autotune(key, x2, weight, bias) -> int:
         results[0] = _time(lambda: lt_linear(x2, weight, bias))
     except Exception:  # noqa: BLE001
         results[0] = float("inf")
-    for sk in _sg5_candidates(M, N, K):
+    for mt, sk in _sg5_candidates(M, N, K):
         try:
-            results[sk] = _time(lambda: skinny_gemm5(x2, weight, bias, splitk=sk))
+            results[(mt, sk)] = _time(
+                lambda: skinny_gemm5(x2, weight, bias, splitk=sk, mt=mt))
         except Exception:  # noqa: BLE001
-            results[sk] = float("inf")
+            results[(mt, sk)] = float("inf")
     choice = min(results, key=results.get)
     _sg5_state["choice"][key] = choice
     return choice
@@ -311,7 +324,8 @@ def decode_linear(x: torch.Tensor, weight: torch.Tensor, bias=None) -> torch.Ten
         choice = _sg5_autotune(key, x2.contiguous(), weight, bias)
     if choice == 0:
         return lt_linear(x, weight, bias)
-    y = skinny_gemm5(x2.contiguous(), weight, bias, splitk=choice)
+    mt, sk = choice
+    y = skinny_gemm5(x2.contiguous(), weight, bias, splitk=sk, mt=mt)
     return y.view(*shape[:-1], N)
 
 

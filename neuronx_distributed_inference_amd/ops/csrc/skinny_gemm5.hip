@@ -74,13 +74,17 @@ constexpr int NBUF = 3;   // LDS chunk ring
 // one glds: stage 1 KB = 8 rows x 128 B, lane-linear dest, XOR-swizzled
 // per-lane source (slot ^= row&7, row = lane/8 within the piece).
 // aux=2 => nontemporal (weights); aux=0 plain (x).
+// `maxrow`: clamp the PER-LANE row offset (rows past it re-read maxrow —
+// ragged M; the clamp must be per lane, not on the piece base, so image
+// row r keeps global row r for every r < M).
 template <int AUX>
 DEV_INLINE void glds_rows128(const char* gsrc_rowbase, long grow_stride,
-                             char* lds_dst_base) {
+                             char* lds_dst_base, int maxrow = 7) {
     const int lane = threadIdx.x & 63;
     const int row = lane >> 3;              // 0..7
     const int slot = (lane & 7) ^ row;      // XOR source-swizzle (rule 21)
-    const char* src = gsrc_rowbase + (long)row * grow_stride + slot * 16;
+    const int reff = max(0, min(row, maxrow));
+    const char* src = gsrc_rowbase + (long)reff * grow_stride + slot * 16;
     char* dst = lds_dst_base;               // wave-uniform; HW adds lane*16
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) unsigned int*)src,
@@ -97,11 +101,12 @@ DEV_INLINE void vm_wait() {
 // via an ordinary 16B load + ds_write (debug MODE 2: isolates glds
 // mechanics from the addressing/swizzle math)
 DEV_INLINE void regstage_rows128(const char* gsrc_rowbase, long grow_stride,
-                                 char* lds_dst_base) {
+                                 char* lds_dst_base, int maxrow = 7) {
     const int lane = threadIdx.x & 63;
     const int row = lane >> 3;
     const int slot = (lane & 7) ^ row;
-    const char* src = gsrc_rowbase + (long)row * grow_stride + slot * 16;
+    const int reff = max(0, min(row, maxrow));
+    const char* src = gsrc_rowbase + (long)reff * grow_stride + slot * 16;
     int4 v = *reinterpret_cast<const int4*>(src);
     *reinterpret_cast<int4*>(lds_dst_base + lane * 16) = v;
 }
@@ -136,6 +141,7 @@ __global__ __launch_bounds__(512, 2) void sk5_kernel(
     const int tile = blockIdx.x / splitk;
     const int slice = blockIdx.x % splitk;
     const int n0 = tile * BN;
+    const int m0blk = blockIdx.y * MT;   // this block's row-block base
     const int kspan = K / splitk;
     const int k_lo = slice * kspan;
     const int nchunks = kspan / BK;
@@ -148,16 +154,17 @@ __global__ __launch_bounds__(512, 2) void sk5_kernel(
     const int wn0 = wn * (BN / WN);      // first col (within BN)
 
     // ---- staging source bases (advance by 128 B per chunk) ----
-    // x piece for glds i: rows wid*XROWS_W + i*8 .. +8 (clamped to M-1 so
-    // M < MT reads stay in bounds; junk rows are discarded at epilogue)
+    // x piece for glds i: rows m0blk + wid*XROWS_W + i*8 .. +8; rows past
+    // M-1 re-read row M-1 per lane (junk rows discarded at epilogue)
     const char* xsrc[NXG];
+    int xmaxrow[NXG];
 #pragma unroll
     for (int i = 0; i < NXG; ++i) {
-        const int r0 = wid * XROWS_W + i * 8;
-        // per-lane row clamp happens inside glds_rows128 via stride math:
-        // clamp the whole 8-row piece base; rows past M-1 re-read row M-1
+        const int r0 = m0blk + wid * XROWS_W + i * 8;
+        const int base = min(r0, M - 1);
         xsrc[i] = reinterpret_cast<const char*>(x) +
-                  ((long)min(r0, max(M - 8, 0)) * K + k_lo) * sizeof(T);
+                  ((long)base * K + k_lo) * sizeof(T);
+        xmaxrow[i] = max(M - 1 - base, 0);
     }
     const char* wsrc = reinterpret_cast<const char*>(w) +
                        ((long)(n0 + wid * 8) * K + k_lo) * sizeof(T);
@@ -176,7 +183,7 @@ __global__ __launch_bounds__(512, 2) void sk5_kernel(
         const long koff = (long)(t) * 128;                                   \
         _Pragma("unroll") for (int i = 0; i < NXG; ++i)                      \
             glds_rows128<0>(xsrc[i] + koff, (long)K * sizeof(T),             \
-                            xdst[b] + i * 1024);                             \
+                            xdst[b] + i * 1024, xmaxrow[i]);                 \
         glds_rows128<2>(wsrc + koff, (long)K * sizeof(T), wdst[b]);          \
     } while (0)
 
@@ -257,7 +264,7 @@ __global__ __launch_bounds__(512, 2) void sk5_kernel(
 #pragma unroll
                 for (int i = 0; i < NXG; ++i)
                     regstage_rows128(xsrc[i] + koff, (long)K * sizeof(T),
-                                     xdst[0] + i * 1024);
+                                     xdst[0] + i * 1024, xmaxrow[i]);
                 regstage_rows128(wsrc + koff, (long)K * sizeof(T), wdst[0]);
             }
             __syncthreads();
@@ -276,7 +283,8 @@ __global__ __launch_bounds__(512, 2) void sk5_kernel(
     // acc[qi][nj] reg r -> y[m = wm0 + qi*16 + (lane>>4)*4 + r]
     //                       [n = n0 + wn0 + nj*16 + (lane&15)]
     if (splitk > 1) {
-        float* p = part + (long)slice * MT * N;
+        const int mt_tot = gridDim.y * MT;  // padded workspace rows
+        float* p = part + (long)slice * mt_tot * N;
 #pragma unroll
         for (int qi = 0; qi < 2; ++qi)
 #pragma unroll
@@ -284,7 +292,7 @@ __global__ __launch_bounds__(512, 2) void sk5_kernel(
                 const int n = n0 + wn0 + nj * 16 + lrow;
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
-                    const int m = wm0 + qi * 16 + lslot * 4 + r;
+                    const int m = m0blk + wm0 + qi * 16 + lslot * 4 + r;
                     p[(long)m * N + n] = acc[qi][nj][r];
                 }
             }
@@ -302,7 +310,8 @@ __global__ __launch_bounds__(512, 2) void sk5_kernel(
             for (int qi = 0; qi < 2; ++qi)
 #pragma unroll
                 for (int r = 0; r < 4; ++r)
-                    rsv[qi][r] = rs[min(wm0 + qi * 16 + lslot * 4 + r, M - 1)];
+                    rsv[qi][r] =
+                        rs[min(m0blk + wm0 + qi * 16 + lslot * 4 + r, M - 1)];
         }
         float bv[NFRAG];
 #pragma unroll
@@ -319,7 +328,7 @@ __global__ __launch_bounds__(512, 2) void sk5_kernel(
                 const int n = n0 + wn0 + nj * 16 + lrow;
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
-                    const int m = wm0 + qi * 16 + lslot * 4 + r;
+                    const int m = m0blk + wm0 + qi * 16 + lslot * 4 + r;
                     if (m < M) {
                         y[(long)m * N + n] = VecTraits<T>::from_float(
                             acc[qi][nj][r] * rsv[qi][r] + bv[nj]);
@@ -379,6 +388,7 @@ void launch_sk5(torch::Tensor& y, const torch::Tensor& x, const torch::Tensor& w
                 const c10::optional<torch::Tensor>& rs,
                 const c10::optional<torch::Tensor>& part, int splitk) {
     const int M = x.size(0), K = x.size(1), N = w.size(0);
+    const int mblocks = (M + MT - 1) / MT;
     constexpr int CHUNK_B = MT * 128 + BN * 128;
     constexpr int LDS = NBUF * CHUNK_B;
     static bool attr_set = false;
@@ -392,8 +402,11 @@ void launch_sk5(torch::Tensor& y, const torch::Tensor& x, const torch::Tensor& w
     const float* rs_p = rs.has_value() ? rs->data_ptr<float>() : nullptr;
     float* part_p = part.has_value() ? part->data_ptr<float>() : nullptr;
     TORCH_CHECK(splitk == 1 || part_p != nullptr,
-                "splitk > 1 needs a (splitk, MT, N) fp32 workspace");
-    dim3 grid((N / BN) * splitk);
+                "splitk > 1 needs a (splitk, mblocks*MT, N) fp32 workspace");
+    TORCH_CHECK(splitk == 1 ||
+                    part->numel() >= (long)splitk * mblocks * MT * N,
+                "split-K workspace too small");
+    dim3 grid((N / BN) * splitk, mblocks);
     hipLaunchKernelGGL((sk5_kernel<T, MT, MODE>), grid, dim3(512), LDS, stream,
                        (T*)y.data_ptr(), part_p, (const T*)x.data_ptr(),
                        (const T*)w.data_ptr(), rs_p, bias_p, M, N, K, splitk);
@@ -404,7 +417,7 @@ void launch_sk5(torch::Tensor& y, const torch::Tensor& x, const torch::Tensor& w
         const int blocks = (int)std::min<long>((total + threads - 1) / threads, 2048);
         hipLaunchKernelGGL((sk5_reduce_kernel<T>), dim3(blocks), dim3(threads), 0,
                            stream, (T*)y.data_ptr(), part_p, rs_p, bias_p,
-                           M, N, MT, splitk);
+                           M, N, mblocks * MT, splitk);
         HIP_CHECK_KERNEL();
     }
 }
@@ -414,7 +427,8 @@ void launch_sk5(torch::Tensor& y, const torch::Tensor& x, const torch::Tensor& w
 void skinny_gemm5(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                   c10::optional<torch::Tensor> bias,
                   c10::optional<torch::Tensor> rs,
-                  c10::optional<torch::Tensor> part, long splitk, long mode) {
+                  c10::optional<torch::Tensor> part, long splitk, long mode,
+                  long mt) {
     TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && y.dim() == 2, "2-D tensors");
     TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && y.is_contiguous(),
                 "contiguous tensors required");
@@ -423,7 +437,12 @@ void skinny_gemm5(torch::Tensor y, torch::Tensor x, torch::Tensor w,
     TORCH_CHECK(N % BN == 0, "N must be a multiple of 64");
     TORCH_CHECK(K % (BK * splitk) == 0, "K must be a multiple of 64*splitk");
     TORCH_CHECK(M <= 256, "skinny_gemm5 is for decode batches (M <= 256)");
-    const int MT = M <= 64 ? 64 : (M <= 128 ? 128 : 256);
+    // mt: row-block tile. 0 = auto (one block over all rows). Smaller
+    // tiles give >1 block/CU co-residency (block-level DMA/compute
+    // overlap) at the cost of re-reading W per row-block.
+    const int MT = mt > 0 ? (int)mt
+                          : (M <= 64 ? 64 : (M <= 128 ? 128 : 256));
+    TORCH_CHECK(MT == 64 || MT == 128 || MT == 256, "mt must be 64/128/256");
     TORCH_CHECK(M >= 8, "M >= 8 required");
 
 #define SK5_MODE(T, MTV)                                                      \

@@ -42,11 +42,12 @@ def bench_shape(M, N, K, iters=40, ncopies=10):
 
     res = {}
     res["lt"] = timed(lambda w: ops.lt_linear(x, w))
-    for sk in ops._sg5_candidates(M, N, K):
-        y = ops.skinny_gemm5(x, ws[0], splitk=sk)
+    for mt, sk in ops._sg5_candidates(M, N, K):
+        y = ops.skinny_gemm5(x, ws[0], splitk=sk, mt=mt)
         err = (y.float() - want).abs().max().item()
-        res[f"sg5_k{sk}"] = timed(lambda w, s=sk: ops.skinny_gemm5(x, w, splitk=s))
-        res[f"sg5_k{sk}_err"] = round(err, 5)
+        res[f"sg5_m{mt}k{sk}"] = timed(
+            lambda w, s=sk, m=mt: ops.skinny_gemm5(x, w, splitk=s, mt=m))
+        res[f"sg5_m{mt}k{sk}_err"] = round(err, 5)
     gb = N * K * 2 / 1e9
     best = min(v for k, v in res.items() if not k.endswith("_err"))
     res["w_mb"] = round(gb * 1000, 1)
@@ -75,7 +76,7 @@ def main():
             r = bench_shape(M, N, K, iters=args.iters)
             out[f"{name}_M{M}"] = r
             lt = r["lt"]
-            best_k = min((k for k in r if k.startswith("sg5_k") and not
+            best_k = min((k for k in r if k.startswith("sg5_") and not
                           k.endswith("_err")), key=r.get, default=None)
             line = f"{name:18s} M={M:3d} N={N:6d} K={K:6d}  lt={lt:7.1f}us"
             if best_k:

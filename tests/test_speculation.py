@@ -174,3 +174,163 @@ def test_eagle_matches_target_only():
         n = min(n, want.shape[1])
         assert n >= ids.shape[1] + 8
         assert torch.equal(row[:n], want[i, :n]), (row.tolist(), want[i].tolist())
+
+
+# ---------------------------------------------------------------------------
+# Rejection sampling (ref model_base.py:1678-1759)
+
+
+@torch.inference_mode()
+def test_filtered_probs_topk_topp():
+    from neuronx_distributed_inference_amd.modules.generation.sampling import (
+        filtered_probs,
+        prepare_sampling_params,
+    )
+
+    vals = torch.tensor([[[3.0, 2.0, 1.0, 0.0]]])
+    # top_k=2: only first two survive
+    p = filtered_probs(vals, prepare_sampling_params(1, top_k=2))
+    assert p[0, 0, 2] == 0 and p[0, 0, 3] == 0
+    want = torch.softmax(torch.tensor([3.0, 2.0]), -1)
+    torch.testing.assert_close(p[0, 0, :2], want)
+    # top_p tiny: only the first token survives (always kept)
+    p = filtered_probs(vals, prepare_sampling_params(1, top_k=4, top_p=0.01))
+    assert p[0, 0, 0] == 1.0 and p[0, 0, 1:].sum() == 0
+    # temperature ~0 concentrates on argmax
+    p = filtered_probs(vals, prepare_sampling_params(1, top_k=4, temperature=1e-6))
+    assert p[0, 0, 0] > 0.999
+
+
+@torch.inference_mode()
+def test_rejection_accept_reduces_to_greedy_match():
+    """With top_k=1 params, q and p are point masses: accept iff draft
+    token == target argmax — exactly the greedy token-match semantics."""
+    from neuronx_distributed_inference_amd.modules.generation.sampling import (
+        RejectionSampler,
+    )
+
+    rj = RejectionSampler(OnDeviceSamplingConfig(top_k=1, global_topk=4), 64,
+                          deterministic=True)
+    B, k, Kg = 2, 3, 4
+    # q: point mass on its first index; p likewise
+    q_idx = torch.tensor([[[5, 1, 2, 3], [7, 1, 2, 3], [9, 1, 2, 3]],
+                          [[4, 1, 2, 3], [6, 1, 2, 3], [8, 1, 2, 3]]])
+    q_probs = torch.zeros(B, k, Kg)
+    q_probs[:, :, 0] = 1.0
+    # target agrees on first two positions for row 0; disagrees at 0 for row 1
+    p_idx = torch.tensor([[[5, 0, 2, 3], [7, 0, 2, 3], [11, 0, 2, 3], [13, 0, 2, 3]],
+                          [[5, 0, 2, 3], [6, 0, 2, 3], [8, 0, 2, 3], [10, 0, 2, 3]]])
+    p_probs = torch.zeros(B, k + 1, Kg)
+    p_probs[:, :, 0] = 1.0
+    draft_ids = q_idx[:, :, 0]                       # q's point mass
+    tokens, n_acc = rj.accept(draft_ids, q_probs, q_idx, p_probs, p_idx)
+    assert n_acc.tolist() == [2, 0]
+    # row 0: accepted 5,7 then resample at j=2 -> p point mass = 11
+    assert tokens[0].tolist() == [5, 7, 11, -1]
+    # row 1: rejected at 0 -> replacement = p's point mass at 0 = 5
+    assert tokens[1].tolist() == [5, -1, -1, -1]
+
+
+@torch.inference_mode()
+def test_rejection_output_distribution_matches_target():
+    """The speculative-sampling theorem: the emitted token at each position
+    is distributed per the TARGET distribution regardless of the draft.
+    Empirical check on a toy 8-token vocab."""
+    from neuronx_distributed_inference_amd.modules.generation.sampling import (
+        RejectionSampler,
+    )
+
+    torch.manual_seed(0)
+    Kg, V = 8, 8
+    rj = RejectionSampler(
+        OnDeviceSamplingConfig(top_k=8, global_topk=Kg, do_sample=True), V)
+    # draft and target dists over tokens 0..7, deliberately different
+    q = torch.tensor([0.4, 0.3, 0.1, 0.05, 0.05, 0.05, 0.03, 0.02])
+    p = torch.tensor([0.1, 0.1, 0.3, 0.3, 0.05, 0.05, 0.05, 0.05])
+    N = 20000
+    B, k = N, 1
+    idx = torch.arange(V).view(1, 1, V)
+    q_idx = idx.expand(B, k, V).contiguous()
+    q_probs = q.view(1, 1, V).expand(B, k, V).contiguous()
+    p_idx = idx.expand(B, k + 1, V).contiguous()
+    p_probs = p.view(1, 1, V).expand(B, k + 1, V).contiguous()
+    draft_ids = rj.sample(q_probs[:, :1], q_idx[:, :1])      # (B,1)
+    tokens, n_acc = rj.accept(draft_ids, q_probs, q_idx, p_probs, p_idx)
+    first = tokens[:, 0]                                     # always emitted
+    emp = torch.bincount(first, minlength=V).float() / N
+    assert (emp - p).abs().max() < 0.02, (emp.tolist(), p.tolist())
+    # acceptance rate should equal sum_t min(p_t, q_t)
+    acc_rate = (n_acc.float().mean()).item()
+    want_rate = torch.minimum(p, q).sum().item()
+    assert abs(acc_rate - want_rate) < 0.02, (acc_rate, want_rate)
+
+
+@torch.inference_mode()
+def test_fused_spec_rejection_topk1_matches_greedy():
+    """Fused spec with rejection sampling at top_k=1 must produce the
+    same accept/token behavior as the greedy token-match fused path."""
+    from neuronx_distributed_inference_amd.modules.speculation import (
+        FusedSpeculationModel,
+    )
+
+    target = _build(64, 2, seed=1, spec_len=4)
+    draft = _build(32, 1, seed=2)
+    ids = torch.randint(0, 64, (2, 8))
+    # prefill both
+    pos = torch.arange(8).view(1, 8).expand(2, 8)
+    sid = torch.arange(2)
+    t_out = target(input_ids=ids, position_ids=pos, seq_ids=sid)
+    draft(input_ids=ids, position_ids=pos, seq_ids=sid)
+    next_tok = t_out.tokens.view(2, 1)
+
+    greedy = FusedSpeculationModel(target.model, draft.model, 4)
+    acc_g, n_g = greedy(next_tok, torch.full((2, 1), 8), sid)
+
+    # fresh models (the fused call advanced the caches)
+    target2 = _build(64, 2, seed=1, spec_len=4)
+    draft2 = _build(32, 1, seed=2)
+    t_out2 = target2(input_ids=ids, position_ids=pos, seq_ids=sid)
+    draft2(input_ids=ids, position_ids=pos, seq_ids=sid)
+    rej = FusedSpeculationModel(
+        target2.model, draft2.model, 4,
+        rejection_config=OnDeviceSamplingConfig(top_k=1, global_topk=8))
+    acc_r, n_r = rej(t_out2.tokens.view(2, 1), torch.full((2, 1), 8), sid)
+    assert torch.equal(n_g, n_r), (n_g, n_r)
+    assert torch.equal(acc_g, acc_r), (acc_g, acc_r)
+
+
+@torch.inference_mode()
+def test_fused_spec_rejection_sampled_runs():
+    """Sampled rejection spec produces valid tokens with acceptance > 0
+    across rounds ('acceptance >0 under sampling')."""
+    from neuronx_distributed_inference_amd.modules.speculation import (
+        FusedSpeculationModel,
+    )
+
+    target = _build(64, 2, seed=1, spec_len=4)
+    draft = _build(64, 2, seed=1)  # identical draft -> high acceptance
+    ids = torch.randint(0, 64, (2, 8))
+    pos = torch.arange(8).view(1, 8).expand(2, 8)
+    sid = torch.arange(2)
+    t_out = target(input_ids=ids, position_ids=pos, seq_ids=sid)
+    draft(input_ids=ids, position_ids=pos, seq_ids=sid)
+    rej = FusedSpeculationModel(
+        target.model, draft.model, 4,
+        rejection_config=OnDeviceSamplingConfig(
+            top_k=8, top_p=0.9, temperature=0.8, do_sample=True, global_topk=16))
+    total_acc = 0
+    nt = t_out.tokens.view(2, 1)
+    lengths = torch.full((2, 1), 8)
+    for _ in range(4):
+        acc, n = rej(nt, lengths, sid)
+        assert ((acc >= -1) & (acc < 64)).all()
+        # emitted tokens are contiguous then -1-padded
+        for b in range(2):
+            row = acc[b]
+            nb = int(n[b])
+            assert (row[:nb + 1] >= 0).all()
+            assert (row[nb + 1:] == -1).all()
+        total_acc += int(n.sum())
+        nt = acc.gather(1, n.view(2, 1))
+        lengths = lengths + n.view(2, 1) + 1
+    assert total_acc > 0
