@@ -299,6 +299,7 @@ class FusedSpeculativeAdapter:
             attention_mask = torch.ones_like(input_ids)
         max_len = gc.max_length or min(self.target.ec.max_length, S0 + gc.max_new_tokens)
         max_len = min(max_len, self.target.ec.max_length - self.k - 2)
+        k = self.k
 
         self.target.reset()
         self.draft.reset()
@@ -312,39 +313,60 @@ class FusedSpeculativeAdapter:
         if self._graph is None:
             self._maybe_capture(B, device)
 
-        lengths = attention_mask.sum(-1).long()
+        # device-side state: output buffer, per-row lengths, finished flags
+        pad = gc.pad_token_id if gc.pad_token_id is not None else 0
+        out_buf = torch.full((B, max_len + k + 1), pad,
+                             dtype=torch.long, device=device)
+        lengths = attention_mask.sum(-1).long()             # committed length
+        am = attention_mask.bool()
+        # left-align the prompt rows into the buffer
+        src_pos = (cum.clamp(min=0))
+        out_buf.scatter_(1, torch.where(am, src_pos, torch.full_like(src_pos, max_len + k)),
+                         torch.where(am, input_ids, torch.full_like(input_ids, pad)))
+        out_buf[:, max_len + k] = pad
         next_tok = t_out.tokens.view(B)
-        out_rows = [input_ids[i][attention_mask[i].bool()].tolist() for i in range(B)]
-        for i in range(B):
-            out_rows[i].append(int(next_tok[i]))
+        out_buf.scatter_(1, lengths.view(B, 1), next_tok.view(B, 1))
+        lengths = lengths + 1
         finished = torch.zeros(B, dtype=torch.bool, device=device)
+        if gc.eos_token_id is not None:
+            finished |= next_tok == gc.eos_token_id
 
-        while True:
-            if gc.eos_token_id is not None:
-                finished |= next_tok == gc.eos_token_id
-            if bool(finished.all()) or max(len(r) for r in out_rows) >= max_len:
-                break
+        idxpos = torch.arange(k + 1, device=device).view(1, -1)
+        rounds = 0
+        max_rounds = max_len  # worst case: 1 token per round
+        while rounds < max_rounds:
             fn = self._graph if self._graph is not None else self.fused
-            accepted, n_accept = fn(next_tok.view(B, 1), lengths.view(B, 1), seq_ids)
-            acc = accepted.cpu()
-            nacc = n_accept.cpu()
-            for i in range(B):
-                if finished[i]:
-                    continue
-                row = out_rows[i]
-                for j in range(int(nacc[i]) + 1):
-                    tok = int(acc[i, j])
-                    if len(row) >= max_len:
-                        break
-                    row.append(tok)
-                    if gc.eos_token_id is not None and tok == gc.eos_token_id:
-                        finished[i] = True
-                        break
-            lengths = lengths + n_accept + 1
-            next_tok = torch.tensor([r[-1] for r in out_rows], device=device, dtype=torch.long)
+            # next_tok sits at position lengths-1 (committed but not yet
+            # fed); the fused round feeds it there and drafts from there
+            accepted, n_accept = fn(next_tok.view(B, 1),
+                                    (lengths - 1).view(B, 1), seq_ids)
+            # --- device-side emission (no per-row python) ---
+            n_eff = torch.where(finished, torch.full_like(n_accept, -1), n_accept)
+            span = idxpos <= n_eff.view(B, 1)               # candidate slots
+            if gc.eos_token_id is not None:
+                is_eos = span & (accepted == gc.eos_token_id)
+                # first eos position (k+1 if none)
+                eos_pos = torch.where(
+                    is_eos.any(1),
+                    is_eos.float().argmax(1).long(),
+                    torch.full((B,), k + 1, dtype=torch.long, device=device))
+            else:
+                eos_pos = torch.full((B,), k + 1, dtype=torch.long, device=device)
+            emit = torch.minimum(n_eff + 1, eos_pos + 1)
+            emit = torch.minimum(emit, (max_len - lengths).clamp(min=0))
+            valid = idxpos < emit.view(B, 1)
+            tgt = lengths.view(B, 1) + idxpos               # write positions
+            tgt = torch.where(valid, tgt, torch.full_like(tgt, max_len + k))
+            out_buf.scatter_(1, tgt, torch.where(valid, accepted,
+                                                 torch.full_like(accepted, pad)))
+            out_buf[:, max_len + k] = pad
+            lengths = lengths + emit
+            finished = finished | (eos_pos <= n_eff) | (lengths >= max_len)
+            next_tok = out_buf.gather(1, (lengths - 1).clamp(min=0).view(B, 1)).view(B)
+            rounds += 1
+            # ONE host sync every SYNC_EVERY rounds
+            if rounds % self.SYNC_EVERY == 0 and bool(finished.all()):
+                break
 
-        longest = max(len(r) for r in out_rows)
-        out = torch.full((B, longest), gc.pad_token_id, dtype=torch.long, device=device)
-        for i, r in enumerate(out_rows):
-            out[i, : len(r)] = torch.tensor(r, device=device)
-        return out
+        longest = int(lengths.max())
+        return out_buf[:, :longest]

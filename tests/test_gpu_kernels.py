@@ -587,11 +587,11 @@ def test_skinny_gemm5_decode_shapes():
         x = torch.randn(M, K, dtype=torch.bfloat16, device=dev) * 0.05
         w = torch.randn(N, K, dtype=torch.bfloat16, device=dev) * 0.05
         want = torch.nn.functional.linear(x, w)
-        for sk in ops._sg5_candidates(M, N, K):
-            y = ops.skinny_gemm5(x, w, splitk=sk)
+        for mt, sk in ops._sg5_candidates(M, N, K):
+            y = ops.skinny_gemm5(x, w, splitk=sk, mt=mt)
             torch.testing.assert_close(y.float(), want.float(),
                                        atol=2e-2, rtol=2e-2,
-                                       msg=f"N={N} K={K} splitk={sk}")
+                                       msg=f"N={N} K={K} mt={mt} splitk={sk}")
 
 
 def test_decode_linear_dispatcher():
