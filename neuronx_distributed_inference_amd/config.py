@@ -98,12 +98,18 @@ class FusedSpecConfig:
     draft_model_path: str = ""
     draft_config: Optional["InferenceConfig"] = None
     speculation_length: int = 5
+    # True rejection sampling in the fused round (accept d_j iff rand <
+    # p(d_j)/q(d_j), resample from the residual; ref model_base.py:1678-
+    # 1759) so speculation composes with top-k/top-p sampling. False =
+    # exact greedy token match.
+    rejection_sampling: bool = False
 
     def to_dict(self):
         return {
             "draft_model_path": self.draft_model_path,
             "draft_config": self.draft_config.to_dict() if self.draft_config else None,
             "speculation_length": self.speculation_length,
+            "rejection_sampling": self.rejection_sampling,
         }
 
     @classmethod
@@ -113,6 +119,7 @@ class FusedSpecConfig:
             draft_model_path=d.get("draft_model_path", ""),
             draft_config=InferenceConfig.from_dict(dc) if dc else None,
             speculation_length=d.get("speculation_length", 5),
+            rejection_sampling=d.get("rejection_sampling", False),
         )
 
 

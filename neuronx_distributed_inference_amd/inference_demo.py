@@ -87,8 +87,14 @@ def setup_run_parser(run: argparse.ArgumentParser):
     run.add_argument("--local-ranks-size", type=int, default=None)
     # speculation
     run.add_argument("--draft-model-path", default=None)
+    run.add_argument("--draft-model-type", default=None,
+                     help="model family of the draft (defaults to the target's)")
     run.add_argument("--speculation-length", type=int, default=0)
     run.add_argument("--enable-fused-speculation", action="store_true")
+    run.add_argument("--spec-rejection-sampling", action="store_true",
+                     help="true rejection sampling in the fused round "
+                          "(spec decode under top-k/top-p sampling)")
+    run.add_argument("--medusa-speculation-length", type=int, default=0)
     # engine knobs
     run.add_argument("--no-hip-graphs", dest="enable_hip_graphs", action="store_false",
                      default=True)
@@ -246,9 +252,49 @@ def run_inference(args, model_type: str) -> int:
     if args.check_accuracy_mode != "skip-accuracy-check" and rank0:
         run_accuracy_check(args, model, config, input_ids, attention_mask, gc)
 
-    out = HuggingFaceGenerationAdapter(model).generate(
-        input_ids, attention_mask=attention_mask, generation_config=gc
-    )
+    if args.draft_model_path and args.speculation_length > 0:
+        # two-model speculation (ref speculation_model submodel /
+        # NeuronFusedSpecModel; draft class resolved like the reference's
+        # dotted-path reconstruction, config.py:956-1038)
+        from neuronx_distributed_inference_amd.modules.speculation import (
+            FusedSpeculativeAdapter,
+            SpeculativeGenerationAdapter,
+        )
+
+        d_cfg_cls, d_model_cls = get_model_classes(
+            args.draft_model_type or model_type)
+        draft_ec = create_engine_config(args, args.draft_model_type or model_type)
+        draft_ec.speculation_length = 0
+        draft_cfg = d_cfg_cls.from_hf_config(
+            args.draft_model_path, neuron_config=draft_ec)
+        draft_app = InferenceApplication(
+            d_model_cls, draft_cfg, model_path=args.draft_model_path)
+        draft_dir = os.path.join(args.compiled_model_path, "draft")
+        if not args.skip_compile and rank0:
+            draft_app.compile(draft_dir)
+        if torch.distributed.is_initialized():
+            torch.distributed.barrier()
+        draft = draft_app.load(draft_dir, device=device)
+        if args.enable_fused_speculation:
+            rej = None
+            if args.spec_rejection_sampling:
+                rej = engine_config.on_device_sampling_config or \
+                    OnDeviceSamplingConfig(
+                        top_k=args.top_k, top_p=args.top_p,
+                        temperature=args.temperature,
+                        do_sample=args.do_sample,
+                        global_topk=args.global_topk)
+            adapter = FusedSpeculativeAdapter(
+                model, draft, args.speculation_length, rejection_config=rej)
+        else:
+            adapter = SpeculativeGenerationAdapter(
+                model, draft, args.speculation_length)
+        out = adapter.generate(
+            input_ids, attention_mask=attention_mask, generation_config=gc)
+    else:
+        out = HuggingFaceGenerationAdapter(model).generate(
+            input_ids, attention_mask=attention_mask, generation_config=gc
+        )
     if rank0:
         if tokenizer is not None:
             for i, row in enumerate(out):
