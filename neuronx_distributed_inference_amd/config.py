@@ -163,6 +163,15 @@ class EngineConfig:
         self.token_generation_buckets: Optional[List[int]] = kwargs.pop(
             "token_generation_buckets", None
         )
+        # decode BATCH sub-buckets: serving at partial occupancy replays
+        # the smallest captured graph covering the active batch (ref 2-D
+        # batch x seq buckets, autobucketing.py:203-223). None = auto
+        # ladder; [] / enable_batch_bucketing=False = single full-batch
+        # graph (the r01 behavior).
+        self.token_generation_batch_buckets: Optional[List[int]] = kwargs.pop(
+            "token_generation_batch_buckets", None
+        )
+        self.enable_batch_bucketing: bool = kwargs.pop("enable_batch_bucketing", True)
         self.bucket_n_active_tokens: bool = kwargs.pop("bucket_n_active_tokens", False)
 
         # --- parallelism degrees (ref config.py:360-390) ---

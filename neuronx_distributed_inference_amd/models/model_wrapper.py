@@ -86,8 +86,8 @@ class ModelWrapper:
             and next(self.model.parameters()).is_cuda
         )
 
-    def _example_inputs(self, device):
-        B = self.batch_size
+    def _example_inputs(self, device, batch: Optional[int] = None):
+        B = batch or self.batch_size
         S = self.n_active_tokens
         input_ids = torch.zeros(B, S, dtype=torch.long, device=device)
         position_ids = torch.zeros(B, S, dtype=torch.long, device=device)
@@ -108,7 +108,6 @@ class ModelWrapper:
         device = next(self.model.parameters()).device
         self.model.rotary.ensure_table(self.ec.max_length, device)
         self._graphs = BucketedGraphModel()
-        ids, pos, sids, sp = self._example_inputs(device)
 
         def fn(input_ids, position_ids, seq_ids, sampling_params):
             return self.model(
@@ -116,10 +115,20 @@ class ModelWrapper:
                 phase=self.phase,
             )
 
-        key = (self.batch_size, self.n_active_tokens)
+        # decode BATCH sub-bucket ladder (ref 2-D batch x seq buckets,
+        # autobucketing.py:203-223): capture one graph per rung, largest
+        # first so the shared pool is sized by the biggest capture
+        self.batch_buckets = ab.generate_batch_buckets(
+            self.batch_size,
+            getattr(self.ec, "token_generation_batch_buckets", None),
+            getattr(self.ec, "enable_batch_bucketing", True),
+        )
         try:
-            self._graphs.capture(key, fn, (ids, pos, sids, sp))
-            logger.info("%s: captured hipGraph %s", self.tag, key)
+            for bb in sorted(self.batch_buckets, reverse=True):
+                ids, pos, sids, sp = self._example_inputs(device, batch=bb)
+                key = (bb, self.n_active_tokens)
+                self._graphs.capture(key, fn, (ids, pos, sids, sp))
+                logger.info("%s: captured hipGraph %s", self.tag, key)
         except Exception:  # noqa: BLE001
             # A capture failure falls back to eager decode ONLY when
             # explicitly permitted: on a benchmark/production run a silent
@@ -162,8 +171,18 @@ class ModelWrapper:
                 attention_mask[:, :S] = 1
 
         # batch padding: garbage rows carry seq_id -1 so every cache write is
-        # skipped (the SPMD-uniform trick, ref _forward_with_pad :574-703)
+        # skipped (the SPMD-uniform trick, ref _forward_with_pad :574-703).
+        # With a captured batch ladder, pad only to the smallest rung
+        # covering B (partial occupancy pays sub-bucket cost, not
+        # full-batch cost).
         Bb = self.batch_size
+        if not self.is_cte and self._graphs is not None:
+            bb = getattr(self, "batch_buckets", None)
+            if bb:
+                for cand in bb:
+                    if cand >= B:
+                        Bb = cand
+                        break
         if B < Bb:
             input_ids = pad_dim(input_ids, 0, Bb, 0)
             attention_mask = pad_dim(attention_mask, 0, Bb, 0)

@@ -64,6 +64,28 @@ def generate_buckets_for_fused_spec(max_length: int, enable: bool = True) -> Lis
     return generate_buckets_for_tkg(max_length, enable=enable)
 
 
+def generate_batch_buckets(
+    max_batch: int,
+    buckets: Optional[List[int]] = None,
+    enable: bool = True,
+    max_graphs: int = 4,
+) -> List[int]:
+    """Decode BATCH sub-buckets (ref batch x seq 2-D buckets,
+    autobucketing.py:203-223): a serving engine at partial occupancy
+    replays the smallest captured graph covering its active batch instead
+    of always paying full-batch cost. Powers-of-2 ladder, capped at
+    `max_graphs` rungs ending at max_batch."""
+    if buckets:
+        out = sorted(set(int(b) for b in buckets))
+        if out[-1] != max_batch:
+            out.append(max_batch)
+        return out
+    if not enable or max_batch <= 1:
+        return [max_batch]
+    ladder = generate_buckets(1, max_batch)
+    return ladder[-max_graphs:]
+
+
 def get_target_bucket(length: int, buckets: List[int], strategy: str = "first_fit") -> int:
     """Smallest bucket >= length (ref model_wrapper.get_target_bucket :831)."""
     for b in sorted(buckets):

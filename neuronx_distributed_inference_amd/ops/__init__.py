@@ -267,35 +267,45 @@ def _sg5_candidates(M: int, N: int, K: int):
 
 
 def _sg5_autotune(key, x2, weight, bias) -> int:
-    """Returns 0 (= lt/library path) or the winning splitk. ~n_iter timed
-    replays per candidate on the current stream."""
+    """Returns 0 (= lt/library path) or the winning (mt, splitk). Timed
+    over ROTATING weight clones so the stream is cache-cold like the real
+    decode loop (the 256 MB L3 would otherwise hold the whole weight and
+    flip the verdict); x stays hot, which is also realistic."""
     M, N, K = key[0], key[1], key[2]
-    n_iter = 10
+    n_iter = 12
+    # rotate enough copies that successive calls miss L3 (256 MB)
+    wbytes = weight.numel() * weight.element_size()
+    ncopy = max(2, min(8, int(320e6 // max(wbytes, 1)) + 1))
+    try:
+        wcopies = [weight] + [weight.clone() for _ in range(ncopy - 1)]
+    except torch.cuda.OutOfMemoryError:  # pragma: no cover
+        wcopies = [weight]
     results = {}
     torch.cuda.synchronize()
 
     def _time(fn):
-        fn()  # warm (triggers lt's own tune / LDS attr set)
+        fn(wcopies[0])  # warm (triggers lt's own tune / LDS attr set)
         torch.cuda.synchronize()
         t0 = torch.cuda.Event(enable_timing=True)
         t1 = torch.cuda.Event(enable_timing=True)
         t0.record()
-        for _ in range(n_iter):
-            fn()
+        for i in range(n_iter):
+            fn(wcopies[i % len(wcopies)])
         t1.record()
         torch.cuda.synchronize()
         return t0.elapsed_time(t1) / n_iter
 
     try:
-        results[0] = _time(lambda: lt_linear(x2, weight, bias))
+        results[0] = _time(lambda w: lt_linear(x2, w, bias))
     except Exception:  # noqa: BLE001
         results[0] = float("inf")
     for mt, sk in _sg5_candidates(M, N, K):
         try:
             results[(mt, sk)] = _time(
-                lambda: skinny_gemm5(x2, weight, bias, splitk=sk, mt=mt))
+                lambda w, s=sk, m=mt: skinny_gemm5(x2, w, bias, splitk=s, mt=m))
         except Exception:  # noqa: BLE001
             results[(mt, sk)] = float("inf")
+    del wcopies
     choice = min(results, key=results.get)
     _sg5_state["choice"][key] = choice
     return choice

@@ -227,6 +227,56 @@ def test_e2e_tiny_decode_gpu_matches_cpu():
     assert agree > 0.85, f"agreement {agree}: {got.cpu()} vs {want}"
 
 
+def test_decode_batch_bucket_ladder_gpu():
+    """Decode at partial batch replays a SUB-BUCKET graph (smallest rung
+    covering B) and must produce the same tokens as the full-batch path
+    (ref 2-D batch buckets, autobucketing.py:203-223)."""
+    import neuronx_distributed_inference_amd.models.llama.modeling_llama as ml
+    from neuronx_distributed_inference_amd.config import (
+        EngineConfig,
+        OnDeviceSamplingConfig,
+    )
+    from neuronx_distributed_inference_amd.utils.hf_adapter import (
+        GenerationConfig,
+        HuggingFaceGenerationAdapter,
+    )
+
+    def build(batch_buckets):
+        torch.manual_seed(3)
+        ec = EngineConfig(
+            batch_size=8, max_batch_size=8, seq_len=64, max_length=64,
+            torch_dtype="bfloat16", tp_degree=1,
+            on_device_sampling_config=OnDeviceSamplingConfig(top_k=1),
+            token_generation_batch_buckets=batch_buckets,
+            enable_batch_bucketing=batch_buckets is not None,
+        )
+        cfg = ml.LlamaInferenceConfig(
+            neuron_config=ec, vocab_size=128, hidden_size=128,
+            intermediate_size=256, num_hidden_layers=2,
+            num_attention_heads=4, num_key_value_heads=2,
+            max_position_embeddings=128, rms_norm_eps=1e-5,
+            rope_theta=10000.0, hidden_act="silu", tie_word_embeddings=False)
+        m = ml.LlamaForCausalLM(cfg).eval().to(_dev())
+        with torch.no_grad():
+            for p in m.parameters():
+                if p.dim() >= 2:
+                    p.normal_(0, 0.1)
+        return m
+
+    ids = torch.randint(0, 128, (3, 8), generator=torch.Generator().manual_seed(5))
+    ladder = build([2, 4, 8])
+    got = HuggingFaceGenerationAdapter(ladder).generate(
+        ids.to(_dev()), generation_config=GenerationConfig(max_new_tokens=8))
+    # B=3 must have used the 4-rung (not the full batch 8)
+    tkg = ladder.token_generation_model
+    assert tkg._graphs is not None and (4, 1) in tkg._graphs.runners
+    single = build(None)
+    single.config.neuron_config.enable_batch_bucketing = False
+    want = HuggingFaceGenerationAdapter(single).generate(
+        ids.to(_dev()), generation_config=GenerationConfig(max_new_tokens=8))
+    assert torch.equal(got.cpu(), want.cpu())
+
+
 @pytest.mark.parametrize("S", [96, 256, 1024])
 @pytest.mark.parametrize("D", [64, 128])
 def test_flash_prefill_shapes(S, D):
