@@ -56,7 +56,8 @@ class ServingEngine:
                  chunk_size: Optional[int] = None,
                  eos_token_id: Optional[int] = None,
                  capture_decode: Optional[bool] = None,
-                 on_token=None, lora_state=None, bucket_chunks: bool = True):
+                 on_token=None, lora_state=None, bucket_chunks: bool = True,
+                 async_depth: Optional[int] = None):
         self.model = causal_lm
         ec = causal_lm.config.neuron_config
         self.batch = ec.batch_size
@@ -97,6 +98,23 @@ class ServingEngine:
             capture_decode = self.paged and self.device.type == "cuda"
         self._capture_decode = capture_decode and self.device.type == "cuda"
         self._decode_graph = None
+        # async double-buffered decode (ref modules/async_execution.py +
+        # _get_model_outputs_async, model_base.py:3737): the decode step's
+        # token feedback stays ON DEVICE (self._slot_tok), host readback
+        # of step i's tokens happens after step i+1 is launched, so the
+        # per-token host<->device round trip hides under device execution.
+        from neuronx_distributed_inference_amd.modules.async_execution import (
+            AsyncExecutor,
+        )
+
+        if async_depth is None:
+            async_depth = 2 if self.device.type == "cuda" else 1
+        self.async_depth = max(1, async_depth)
+        self._async = AsyncExecutor(max_inflight=self.async_depth,
+                                    device=self.device)
+        self._slot_tok = torch.zeros(self.max_slots, dtype=torch.long,
+                                     device=self.device)
+        self._inflight: Deque = collections.deque()  # (slots, AsyncResult)
 
     # ------------------------------------------------------------------
     def add_request(self, prompt_ids: torch.Tensor,
@@ -146,7 +164,7 @@ class ServingEngine:
         return p
 
     def has_work(self) -> bool:
-        return bool(self.queue or self.active)
+        return bool(self.queue or self.active or self._inflight)
 
     # ------------------------------------------------------------------
     def _ensure_blocks(self, req: Request, upto_pos: int):
@@ -221,6 +239,7 @@ class ServingEngine:
             tok = int(lg[0, -1].argmax()) if lg.dim() == 3 \
                 else int(lg.view(1, -1)[0, -1])
             req.last_token = tok
+            self._slot_tok[req.slot] = tok
             req.pos = S
             req.output.append(tok)
             req.t_first_token = time.monotonic()
@@ -229,16 +248,25 @@ class ServingEngine:
             self._maybe_finish(req, tok)
 
     def _decode_step(self):
+        """Launch one batched decode step. Token feedback for the NEXT
+        step is read from the device-side self._slot_tok (no host sync);
+        host bookkeeping of this step's tokens is deferred until after
+        the next launch (async double-buffering — ref
+        modules/async_execution.py, model_base.py:3737)."""
         slots = sorted(self.active)
         B = self.batch
-        ids = torch.zeros(B, 1, dtype=torch.long, device=self.device)
         pos = torch.full((B, 1), -1, dtype=torch.long, device=self.device)
         seq_ids = torch.full((B,), -1, dtype=torch.long, device=self.device)
         for i, slot in enumerate(slots):
             r = self.active[slot]
-            ids[i, 0] = r.last_token
             pos[i, 0] = r.pos
             seq_ids[i] = slot
+        slot_idx = torch.tensor(slots, dtype=torch.long, device=self.device)
+        ids = self._slot_tok.index_select(0, slot_idx).view(-1, 1)
+        if ids.shape[0] < B:
+            ids = torch.cat([ids, torch.zeros(B - ids.shape[0], 1,
+                                              dtype=torch.long,
+                                              device=self.device)], dim=0)
         reqs = [self.active[s] for s in slots]
         self._set_adapters(reqs)
         kw = {}
@@ -250,23 +278,43 @@ class ServingEngine:
             full[: len(reqs)] = self._block_table(reqs)
             kw["block_table"] = full
         sp = self._params_for(reqs)
-        if self._capture_decode and self.paged:
-            toks = self._run_decode_graph(ids, pos, seq_ids, sp, kw["block_table"])
-        else:
-            out = self.model(ids, None, pos, seq_ids, sp, **kw)
-            toks = out.logits if out.logits is not None else out.tokens
-        if toks.dim() > 2:
-            toks = toks.argmax(-1)
-        toks = toks.view(B, -1)[:, -1].tolist()
-        for i, slot in enumerate(slots):
-            r = self.active[slot]
-            tok = int(toks[i])
-            r.last_token = tok
-            r.pos += 1
-            r.output.append(tok)
-            if self.on_token is not None:
-                self.on_token(r.req_id, tok)
-            self._maybe_finish(r, tok)
+
+        def run():
+            if self._capture_decode and self.paged:
+                toks = self._run_decode_graph(ids, pos, seq_ids, sp,
+                                              kw["block_table"])
+            else:
+                out = self.model(ids, None, pos, seq_ids, sp, **kw)
+                toks = out.logits if out.logits is not None else out.tokens
+            if toks.dim() > 2:
+                toks = toks.argmax(-1)
+            toks = toks.view(B, -1)[:, -1].clone()
+            # device-side feedback for the next step
+            self._slot_tok.scatter_(0, slot_idx, toks[: len(slots)])
+            return toks
+
+        res = self._async.submit(run)
+        for r in reqs:
+            r.pos += 1  # launch-time advance (host-known)
+        self._inflight.append((slots, res))
+        self._drain_inflight(keep=self.async_depth - 1)
+
+    def _drain_inflight(self, keep: int = 0):
+        """Process deferred decode results (host bookkeeping: outputs,
+        callbacks, finish checks) down to `keep` outstanding."""
+        while len(self._inflight) > keep:
+            slots, res = self._inflight.popleft()
+            toks = res.result().tolist()
+            for i, slot in enumerate(slots):
+                r = self.active.get(slot)
+                if r is None or r.done:
+                    continue  # finished meanwhile; extra token discarded
+                tok = int(toks[i])
+                r.last_token = tok
+                r.output.append(tok)
+                if self.on_token is not None:
+                    self.on_token(r.req_id, tok)
+                self._maybe_finish(r, tok)
 
     def _run_decode_graph(self, ids, pos, seq_ids, sp, block_table):
         from neuronx_distributed_inference_amd.runtime.graph_engine import (
@@ -315,7 +363,10 @@ class ServingEngine:
     # ------------------------------------------------------------------
     def step(self):
         """One scheduling step: admit + prefill one chunk, or decode."""
-        # admit
+        # admit (deferred decode results must land first: a finished slot
+        # may be freed and re-assigned, and pending tokens reference it)
+        if self.queue:
+            self._drain_inflight(0)
         if self.queue and self.free_slots:
             req = self.queue.popleft()
             req.slot = self.free_slots.pop()
@@ -325,6 +376,7 @@ class ServingEngine:
         for slot in sorted(self.active):
             r = self.active[slot]
             if r.prefill_done < r.prompt.shape[0]:
+                self._drain_inflight(0)
                 self._prefill_chunk(r)
                 return
         if self.active:
@@ -335,6 +387,7 @@ class ServingEngine:
         while self.has_work() and steps < max_steps:
             self.step()
             steps += 1
+        self._drain_inflight(0)
         return {rid: r.output for rid, r in self.finished.items()}
 
     def metrics(self) -> Dict[str, float]:

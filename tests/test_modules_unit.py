@@ -344,3 +344,22 @@ def test_qwen2vl_config_json_roundtrip():
         assert back.text_config.rope_scaling["mrope_section"] == [2, 2, 2]
         assert back.video_token_id == 156
         assert back.vision_config["embed_dim"] == 32
+
+
+def test_async_executor_inflight_cap_and_order():
+    """AsyncExecutor (ref modules/async_execution.py): back-pressure at
+    max_inflight, results in submission order, drain completes all."""
+    import torch
+
+    from neuronx_distributed_inference_amd.modules.async_execution import (
+        AsyncExecutor,
+    )
+
+    done = []
+    ex = AsyncExecutor(max_inflight=2, device=torch.device("cpu"))
+    results = [ex.submit(lambda i=i: (done.append(i), i * 10)[1])
+               for i in range(5)]
+    ex.drain()
+    assert [r.result() for r in results] == [0, 10, 20, 30, 40]
+    assert done == list(range(5))
+    assert all(r.done() for r in results)
