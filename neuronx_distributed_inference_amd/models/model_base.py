@@ -240,14 +240,31 @@ class DecoderModel(nn.Module):
             h, _ = self.norm(h, residual)
         else:
             h = self.norm(h)
-        if sp_active:
-            # exit SP before the last-token gather (ref
-            # seq_parallel_logits_slice.py; full gather kept simple here —
-            # prefill-only, once per request)
+        is_speculation = phase == PHASE_TKG and S > 1
+        h_last = None
+        if sp_active and phase == PHASE_CTE and S > 1:
+            # SP exit: move ONE token per row instead of all-gathering the
+            # (B, S, H) activations (ref seq_parallel_logits_slice.py:9-70
+            # — index math + masking + reduce). Each rank contributes its
+            # own shard's last-token rows; the sum-reduce assembles them.
+            if attention_mask is not None:
+                last = attention_mask.sum(dim=-1).long() - 1
+            else:
+                last = torch.full((B,), S - 1, dtype=torch.long, device=h.device)
+            Sl = h.shape[1]                      # local shard length
+            rank = mappings.ps.get_tensor_model_parallel_rank()
+            owner = last // Sl
+            local = (last % Sl).view(B, 1, 1).expand(B, 1, h.shape[-1])
+            mine = (owner == rank).view(B, 1, 1)
+            contrib = torch.where(mine, h.gather(1, local), torch.zeros_like(h[:, :1]))
+            h_last = mappings.reduce_from_tensor_parallel_region(contrib.contiguous())
+        elif sp_active:
+            # non-CTE SP (disabled for TKG like the reference) — full exit
             h = mappings.gather_along_dim(h, dim=1)
 
-        is_speculation = phase == PHASE_TKG and S > 1
-        if phase == PHASE_CTE and S > 1:
+        if h_last is not None:
+            pass
+        elif phase == PHASE_CTE and S > 1:
             if attention_mask is not None:
                 last = attention_mask.sum(dim=-1).long() - 1
             else:

@@ -34,6 +34,17 @@ def _init_world1():
     dist.init_process_group("nccl", rank=0, world_size=1)
 
 
+@pytest.fixture(scope="module", autouse=True)
+def _teardown_world():
+    """Destroy the RCCL communicator before interpreter exit — RCCL's
+    atexit teardown after a graph-captured communicator can dump core,
+    which would fail the whole GPU test session."""
+    yield
+    if dist.is_initialized():
+        torch.cuda.synchronize()
+        dist.destroy_process_group()
+
+
 def test_allreduce_inside_hipgraph():
     _init_world1()
     dev = torch.device("cuda:0")
