@@ -146,10 +146,68 @@ class ExpertMLPs(nn.Module):
             y = y * gte
         return y.sum(dim=0)
 
+    def _forward_grouped(self, x: torch.Tensor, weights: torch.Tensor,
+                         idx: torch.Tensor) -> torch.Tensor:
+        """Grouped expert GEMM: sort assignments by expert,
+        pad each segment to 64-row tiles, one kernel launch per
+        projection for ALL experts. Static shapes + tensor-only table
+        construction -> hipGraph-capturable; no top_k/E redundant FLOPs
+        (the dense-bmm path is compute-bound at decode batch >=128).
+        Non-local assignments (EP) route to a sentinel segment whose
+        tiles the kernel skips; their contribution is masked out."""
+        T, H = x.shape
+        El = self.num_local_experts
+        offset = self._local_offset()
+        top_k = idx.shape[1]
+        flat = (idx - offset).view(-1)
+        local = (flat >= 0) & (flat < El)
+        flat_routed = torch.where(local, flat, torch.full_like(flat, El))
+        order, dest, expert_ids, Mpad = ops.moe_grouped_tables(
+            flat_routed, El + 1)
+        # sentinel segment (id El) never computes
+        expert_ids = torch.where(expert_ids >= El,
+                                 torch.full_like(expert_ids, -1), expert_ids)
+        tok = order // top_k
+        gatew = weights.view(-1)[order].to(x.dtype)
+        lmask = local[order]
+        x_rows = x.index_select(0, tok)
+        if self.scale_input:
+            x_rows = x_rows * gatew.unsqueeze(-1)
+        x_sorted = torch.zeros(Mpad, H, dtype=x.dtype, device=x.device)
+        x_sorted.index_copy_(0, dest, x_rows)
+        g = ops.moe_grouped_gemm(x_sorted, self.w_gate, expert_ids)
+        u = ops.moe_grouped_gemm(x_sorted, self.w_up, expert_ids)
+        h = self.act_mul(g, u)
+        y_sorted = ops.moe_grouped_gemm(h, self.w_down, expert_ids)
+        contrib = y_sorted.index_select(0, dest)
+        if not self.scale_input:
+            contrib = contrib * gatew.unsqueeze(-1)
+        # skipped tiles hold uninitialized memory: mask, don't multiply
+        contrib = torch.where(lmask.unsqueeze(-1), contrib,
+                              torch.zeros_like(contrib))
+        out = torch.zeros(T, H, dtype=x.dtype, device=x.device)
+        out.index_add_(0, tok, contrib)
+        return out
+
     def forward(self, x: torch.Tensor, weights: torch.Tensor, idx: torch.Tensor):
         """x: (T, H); weights/idx: (T, top_k) over GLOBAL expert ids.
         Returns (T, H) partial (summed by the caller's TP all-reduce)."""
         T, H = x.shape
+        # GPU: grouped expert GEMM once the batch is big enough that the
+        # dense bmm's top_k/E redundant FLOPs exceed its launch savings
+        # (measured crossover ~batch 64; NXDI_MOE_GROUPED=0/1 overrides)
+        import os as _os
+
+        grouped_env = _os.environ.get("NXDI_MOE_GROUPED")
+        use_grouped = (
+            x.is_cuda and ops.extension_available()
+            and x.dtype in (torch.bfloat16, torch.float16)
+            and self.w_gate.shape[2] % 64 == 0 and self.w_gate.shape[1] % 64 == 0
+            and H % 64 == 0
+            and (grouped_env == "1" or (grouped_env != "0" and T >= 64))
+        )
+        if use_grouped:
+            return self._forward_grouped(x, weights, idx)
         # decode-sized batches: DENSE batched-GEMM path. Every expert's
         # weights must stream from HBM for any realistic decode batch
         # anyway, so the dense bmm has the SAME memory traffic as the

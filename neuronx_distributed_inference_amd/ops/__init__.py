@@ -590,3 +590,66 @@ def attention_decode_paged(q, k_cache, v_cache, seq_lens, block_table,
         mask = mask & (ar >= (qpos.clamp(min=0) // C) * C)
     out = ref.sdpa(q, k, v, mask=mask, scale=scale)
     return torch.nan_to_num(out)
+
+
+# ----------------------------------------------------------------------
+# Grouped MoE expert GEMM: tokens sorted by expert, padded
+# per expert to 64-row tiles; one kernel launch serves every expert's
+# segment. All table construction is tensor ops with STATIC shapes, so
+# the whole MoE layer is hipGraph-capturable.
+
+MOE_TILE = 64
+
+
+def moe_grouped_tables(idx_flat: torch.Tensor, num_experts: int):
+    """idx_flat: (T*top_k,) local expert id per assignment. Returns
+    (order, dest, expert_ids, Mpad):
+      order      (Ttk,)  argsort by expert (stable)
+      dest       (Ttk,)  destination row of each sorted assignment in the
+                         padded sorted space
+      expert_ids (Mpad/64,) int32 expert per row tile, -1 past the end
+      Mpad       static worst-case padded rows (multiple of 64)
+    """
+    E = num_experts
+    Ttk = idx_flat.numel()
+    dev = idx_flat.device
+    MT = MOE_TILE
+    Mpad = ((Ttk + E * (MT - 1)) + MT - 1) // MT * MT
+    order = torch.argsort(idx_flat, stable=True)
+    sorted_e = idx_flat[order]
+    counts = torch.bincount(idx_flat, minlength=E)
+    padded = (counts + MT - 1) // MT * MT
+    ends_pad = padded.cumsum(0)                 # per-expert padded end
+    starts_pad = ends_pad - padded
+    starts_raw = counts.cumsum(0) - counts
+    within = torch.arange(Ttk, device=dev) - starts_raw[sorted_e]
+    dest = starts_pad[sorted_e] + within
+    tile_start = torch.arange(Mpad // MT, device=dev) * MT
+    eid = torch.searchsorted(ends_pad, tile_start, right=True)
+    total_pad = ends_pad[-1] if E > 0 else torch.zeros((), device=dev)
+    expert_ids = torch.where(tile_start < total_pad, eid,
+                             torch.full_like(eid, -1)).to(torch.int32)
+    return order, dest, expert_ids, Mpad
+
+
+def moe_grouped_gemm(x_sorted: torch.Tensor, w: torch.Tensor,
+                     expert_ids: torch.Tensor) -> torch.Tensor:
+    """y_sorted[m] = x_sorted[m] @ w[expert_of(m)]^T. x_sorted (Mpad, K),
+    w (E, N, K), expert_ids (Mpad/64,) int32. GPU: one grouped HIP
+    kernel; CPU reference: per-tile loop."""
+    Mpad, K = x_sorted.shape
+    N = w.shape[1]
+    if x_sorted.is_cuda:
+        c = _require_ext("moe_grouped_gemm")
+        y = torch.empty(Mpad, N, dtype=x_sorted.dtype, device=x_sorted.device)
+        c.moe_grouped_gemm(y, x_sorted.contiguous(), w.contiguous(),
+                           expert_ids.contiguous())
+        return y
+    y = torch.zeros(Mpad, N, dtype=x_sorted.dtype)
+    for t in range(Mpad // MOE_TILE):
+        e = int(expert_ids[t])
+        if e < 0:
+            continue
+        rows = slice(t * MOE_TILE, (t + 1) * MOE_TILE)
+        y[rows] = x_sorted[rows] @ w[e].t().to(x_sorted.dtype)
+    return y

@@ -47,6 +47,8 @@ void fp8_quant(torch::Tensor x, torch::Tensor out, torch::Tensor scale,
                torch::Tensor amax_ws);
 void fp8_epilogue(torch::Tensor out, torch::Tensor y32, torch::Tensor w_scale,
                   c10::optional<torch::Tensor> bias);
+void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
+                      torch::Tensor expert_ids);
 bool lt_available();
 void lt_linear(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                c10::optional<torch::Tensor> bias);
@@ -100,6 +102,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("fp8_epilogue", &fp8_epilogue, "scaled-mm row-scale epilogue (CDNA4)",
           py::arg("out"), py::arg("y32"), py::arg("w_scale"),
           py::arg("bias") = py::none());
+    m.def("moe_grouped_gemm", &moe_grouped_gemm,
+          "grouped expert GEMM over sorted padded token segments (CDNA4)");
     m.def("lt_available", &lt_available, "hipblaslt usable on this device");
     m.def("lt_linear", &lt_linear, "autotuned hipBLASLt linear (graph-safe)",
           py::arg("y"), py::arg("x"), py::arg("w"), py::arg("bias") = py::none());

@@ -1,0 +1,234 @@
+// Grouped expert GEMM for MoE (ref NxD blockwise-matmul MoE
+// kernels via modules/moe_v2.py:104-161). One launch computes
+//
+//   y_sorted[m, n] = x_sorted[m, :] @ W[expert_of(m), n, :]^T
+//
+// over token rows SORTED by expert and padded per expert to MT-row
+// tiles. Replaces the dense-bmm decode path's top_k/E redundant FLOPs
+// (Mixtral batch 256: dense is compute-bound at ~2x the weight-stream
+// floor) and the prefill gather loop's E x 3 launches + host syncs.
+//
+// Graph-safe contract: grid is STATIC for a given (T*top_k, N) — the
+// host sizes it at the worst-case padded row count; blocks whose row
+// tile has expert_id < 0 (beyond this step's real segments) write
+// nothing. The per-step segment tables (expert id per row tile, sorted
+// token order) are built with torch ops on device, so the whole MoE
+// layer captures into one hipGraph.
+//
+// Structure = skinny_gemm5's verified pipeline (all-glds staging, 3-LDS
+// buffer ring, ONE raw s_barrier per 64-deep K chunk, hand-counted
+// vmcnt, XOR source-swizzle on both [rows][128B] images): the only
+// deltas are the expert-indexed W base and the row-tile grid dim.
+
+#include "common.h"
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+namespace {
+
+typedef __bf16 mg_bf16x8 __attribute__((ext_vector_type(8)));
+typedef _Float16 mg_f16x8 __attribute__((ext_vector_type(8)));
+typedef float mg_f32x4 __attribute__((ext_vector_type(4)));
+
+template <typename T> struct MgMfma;
+template <> struct MgMfma<__hip_bfloat16> {
+    using vec8 = mg_bf16x8;
+    static DEV_INLINE mg_f32x4 mfma(vec8 a, vec8 b, mg_f32x4 c) {
+        return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+    }
+};
+template <> struct MgMfma<__half> {
+    using vec8 = mg_f16x8;
+    static DEV_INLINE mg_f32x4 mfma(vec8 a, vec8 b, mg_f32x4 c) {
+        return __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b, c, 0, 0, 0);
+    }
+};
+
+constexpr int MG_BK = 64;
+constexpr int MG_BN = 64;
+constexpr int MG_NBUF = 3;
+
+template <int AUX>
+DEV_INLINE void mg_glds_rows128(const char* gsrc_rowbase, long grow_stride,
+                                char* lds_dst_base, int maxrow = 7) {
+    const int lane = threadIdx.x & 63;
+    const int row = lane >> 3;
+    const int slot = (lane & 7) ^ row;
+    const int reff = max(0, min(row, maxrow));
+    const char* src = gsrc_rowbase + (long)reff * grow_stride + slot * 16;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src,
+        (__attribute__((address_space(3))) unsigned int*)lds_dst_base, 16, 0, AUX);
+}
+
+template <int N>
+DEV_INLINE void mg_vm_wait() {
+    asm volatile("s_waitcnt vmcnt(%c0)" ::"i"(N) : "memory");
+}
+
+// MT=64 row tiles (8 waves as 2(M) x 4(N): wave = 32 rows x 16 cols)
+template <typename T>
+__global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
+    T* __restrict__ y,              // (Mpad, N) sorted-space output
+    const T* __restrict__ x,        // (Mpad, K) sorted-space input
+    const T* __restrict__ w,        // (E, N, K) stacked expert weights
+    const int* __restrict__ expert_ids,  // (n_row_tiles,) expert or -1
+    const int N, const int K) {
+    using MF = MgMfma<T>;
+    using vec8 = typename MF::vec8;
+    constexpr int MT = 64;
+    constexpr int NXG = 1;          // x glds per wave per chunk (8 rows)
+    constexpr int G = NXG + 1;
+    constexpr int XBYTES = MT * 128;
+    constexpr int CHUNK_B = XBYTES + MG_BN * 128;
+
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+
+    const int tile = blockIdx.x;    // n tile
+    const int rt = blockIdx.y;      // row tile
+    const int e = expert_ids[rt];
+    if (e < 0) return;              // beyond this step's padded rows
+    const int n0 = tile * MG_BN;
+    const int m0 = rt * MT;
+    const int nchunks = K / MG_BK;
+
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int wm0 = (wid >> 2) * 32;        // 2 waves along M
+    const int wn0 = (wid & 3) * 16;         // 4 waves along N
+    const int lrow = lane & 15;
+    const int lslot = lane >> 4;
+
+    const char* xsrc = reinterpret_cast<const char*>(x) +
+                       ((long)(m0 + wid * 8) * K) * sizeof(T);
+    const char* wsrc = reinterpret_cast<const char*>(w) +
+                       (((long)e * N + n0 + wid * 8) * K) * sizeof(T);
+
+    char* xdst[MG_NBUF];
+    char* wdst[MG_NBUF];
+#pragma unroll
+    for (int b = 0; b < MG_NBUF; ++b) {
+        xdst[b] = smem + b * CHUNK_B + wid * 1024;
+        wdst[b] = smem + b * CHUNK_B + XBYTES + wid * 1024;
+    }
+
+#define MG_STAGE(t, b)                                                       \
+    do {                                                                     \
+        const long koff = (long)(t) * 128;                                   \
+        mg_glds_rows128<0>(xsrc + koff, (long)K * sizeof(T), xdst[b]);       \
+        mg_glds_rows128<2>(wsrc + koff, (long)K * sizeof(T), wdst[b]);       \
+    } while (0)
+
+    mg_f32x4 acc[2];
+    acc[0] = {0.f, 0.f, 0.f, 0.f};
+    acc[1] = {0.f, 0.f, 0.f, 0.f};
+
+#define MG_FRAGS(b)                                                          \
+    do {                                                                     \
+        const char* xb = smem + (b) * CHUNK_B;                               \
+        const char* wb = xb + XBYTES;                                        \
+        _Pragma("unroll") for (int ks = 0; ks < 2; ++ks) {                   \
+            vec8 a0, a1, bb;                                                 \
+            {                                                                \
+                const int m = wm0 + lrow;                                    \
+                const int s = (ks * 4 + lslot) ^ (m & 7);                    \
+                a0 = *reinterpret_cast<const vec8*>(xb + m * 128 + s * 16);  \
+            }                                                                \
+            {                                                                \
+                const int m = wm0 + 16 + lrow;                               \
+                const int s = (ks * 4 + lslot) ^ (m & 7);                    \
+                a1 = *reinterpret_cast<const vec8*>(xb + m * 128 + s * 16);  \
+            }                                                                \
+            {                                                                \
+                const int n = wn0 + lrow;                                    \
+                const int s = (ks * 4 + lslot) ^ (n & 7);                    \
+                bb = *reinterpret_cast<const vec8*>(wb + n * 128 + s * 16);  \
+            }                                                                \
+            acc[0] = MF::mfma(a0, bb, acc[0]);                               \
+            acc[1] = MF::mfma(a1, bb, acc[1]);                               \
+        }                                                                    \
+    } while (0)
+
+#define MG_BODY(t, b, b2)                                                    \
+    do {                                                                     \
+        if ((t) + 1 < nchunks) mg_vm_wait<G>();                              \
+        else mg_vm_wait<0>();                                                \
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                   \
+        __builtin_amdgcn_s_barrier();                                        \
+        MG_FRAGS(b);                                                         \
+        if ((t) + 2 < nchunks) MG_STAGE((t) + 2, b2);                        \
+    } while (0)
+
+    MG_STAGE(0, 0);
+    if (nchunks > 1) MG_STAGE(1, 1);
+    int t = 0;
+    while (t + 3 <= nchunks) {
+        MG_BODY(t + 0, 0, 2);
+        MG_BODY(t + 1, 1, 0);
+        MG_BODY(t + 2, 2, 1);
+        t += 3;
+    }
+    if (t < nchunks) MG_BODY(t, 0, 2);
+    if (t + 1 < nchunks) MG_BODY(t + 1, 1, 0);
+#undef MG_BODY
+#undef MG_FRAGS
+#undef MG_STAGE
+    mg_vm_wait<0>();
+
+    // epilogue: y[m0 + wm0 + qi*16 + lslot*4 + r][n0 + wn0 + lrow]
+#pragma unroll
+    for (int qi = 0; qi < 2; ++qi) {
+        const int n = n0 + wn0 + lrow;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int m = m0 + wm0 + qi * 16 + lslot * 4 + r;
+            y[(long)m * N + n] = VecTraits<T>::from_float(acc[qi][r]);
+        }
+    }
+}
+
+}  // namespace
+
+void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
+                      torch::Tensor expert_ids) {
+    TORCH_CHECK(y.dim() == 2 && x.dim() == 2 && w.dim() == 3, "shapes");
+    TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && y.is_contiguous() &&
+                expert_ids.is_contiguous(), "contiguous tensors required");
+    const int Mpad = x.size(0), K = x.size(1);
+    const int N = w.size(1);
+    TORCH_CHECK(w.size(2) == K && y.size(0) == Mpad && y.size(1) == N, "shape");
+    TORCH_CHECK(N % MG_BN == 0, "N must be a multiple of 64");
+    TORCH_CHECK(K % MG_BK == 0, "K must be a multiple of 64");
+    TORCH_CHECK(Mpad % 64 == 0, "padded rows must be a multiple of 64");
+    TORCH_CHECK(expert_ids.scalar_type() == at::ScalarType::Int &&
+                expert_ids.numel() == Mpad / 64, "expert_ids must be int (Mpad/64,)");
+
+    constexpr int LDS = MG_NBUF * (64 * 128 + MG_BN * 128);
+    auto stream = at::hip::getCurrentHIPStream();
+    dim3 grid(N / MG_BN, Mpad / 64);
+
+#define MG_T(T)                                                               \
+    do {                                                                      \
+        static bool attr_set = false;                                         \
+        if (!attr_set) {                                                      \
+            (void)hipFuncSetAttribute((const void*)moe_grouped_kernel<T>,     \
+                                      hipFuncAttributeMaxDynamicSharedMemorySize, \
+                                      LDS);                                   \
+            attr_set = true;                                                  \
+        }                                                                     \
+        hipLaunchKernelGGL((moe_grouped_kernel<T>), grid, dim3(512), LDS,     \
+                           stream, (T*)y.data_ptr(), (const T*)x.data_ptr(),  \
+                           (const T*)w.data_ptr(), expert_ids.data_ptr<int>(),\
+                           N, K);                                             \
+    } while (0)
+
+    if (x.scalar_type() == at::ScalarType::BFloat16) {
+        MG_T(__hip_bfloat16);
+    } else if (x.scalar_type() == at::ScalarType::Half) {
+        MG_T(__half);
+    } else {
+        TORCH_CHECK(false, "moe_grouped_gemm: bf16/f16 only");
+    }
+#undef MG_T
+    HIP_CHECK_KERNEL();
+}

@@ -31,6 +31,7 @@ sources = [
     os.path.join(CSRC, "skinny_gemm3.hip"),
     os.path.join(CSRC, "skinny_gemm4.hip"),
     os.path.join(CSRC, "skinny_gemm5.hip"),
+    os.path.join(CSRC, "moe_grouped.hip"),
     os.path.join(CSRC, "lt_gemm.cpp"),
 ]
 

@@ -363,3 +363,71 @@ def test_async_executor_inflight_cap_and_order():
     assert [r.result() for r in results] == [0, 10, 20, 30, 40]
     assert done == list(range(5))
     assert all(r.done() for r in results)
+
+
+def test_moe_grouped_tables():
+    import torch
+
+    from neuronx_distributed_inference_amd import ops
+
+    torch.manual_seed(0)
+    E, Ttk = 5, 37
+    idx = torch.randint(0, E, (Ttk,))
+    order, dest, expert_ids, Mpad = ops.moe_grouped_tables(idx, E)
+    assert Mpad % 64 == 0 and dest.max() < Mpad
+    # each sorted assignment's destination tile belongs to its expert
+    sorted_e = idx[order]
+    tiles = (dest // ops.MOE_TILE).long()
+    assert (expert_ids[tiles].long() == sorted_e).all()
+    # destinations are unique
+    assert dest.unique().numel() == Ttk
+    # tiles past the padded end are -1
+    total_pad = ((torch.bincount(idx, minlength=E) + 63) // 64 * 64).sum()
+    assert (expert_ids[int(total_pad) // 64:] == -1).all()
+
+
+def test_moe_grouped_matches_dense():
+    """Grouped expert path (CPU reference kernel) must equal the dense
+    bmm path, including gate scaling and input-scaled (Llama4) mode."""
+    import torch
+
+    from neuronx_distributed_inference_amd.modules.moe import ExpertMLPs
+
+    torch.manual_seed(1)
+    for scale_input in (False, True):
+        m = ExpertMLPs(num_experts=4, hidden_size=64, intermediate_size=128,
+                       tp_degree=1, dtype=torch.float32,
+                       scale_input=scale_input)
+        with torch.no_grad():
+            for p in m.parameters():
+                p.normal_(0, 0.1)
+        T, top_k = 9, 2
+        x = torch.randn(T, 64) * 0.3
+        logits = torch.randn(T, 4)
+        w, idx = torch.softmax(logits, -1).topk(top_k, dim=-1)
+        want = m._forward_dense(x, w, idx)
+        got = m._forward_grouped(x, w, idx)
+        torch.testing.assert_close(got, want, atol=1e-4, rtol=1e-4)
+
+
+def test_moe_grouped_masks_nonlocal_experts():
+    """EP: assignments to experts outside the local range contribute
+    zero (sentinel segment skipped + masked combine)."""
+    import torch
+
+    from neuronx_distributed_inference_amd.modules.moe import ExpertMLPs
+
+    torch.manual_seed(2)
+    m = ExpertMLPs(num_experts=2, hidden_size=64, intermediate_size=128,
+                   tp_degree=1, dtype=torch.float32)
+    with torch.no_grad():
+        for p in m.parameters():
+            p.normal_(0, 0.1)
+    T = 5
+    x = torch.randn(T, 64) * 0.3
+    # top_k = 2: one local (0/1), one out-of-range (7)
+    idx = torch.stack([torch.randint(0, 2, (T,)), torch.full((T,), 7)], dim=1)
+    w = torch.full((T, 2), 0.5)
+    got = m._forward_grouped(x, w, idx)
+    want = m._forward_grouped(x, w[:, :1], idx[:, :1])
+    torch.testing.assert_close(got, want, atol=1e-5, rtol=1e-5)
