@@ -211,6 +211,9 @@ class EngineConfig:
                 self.chunked_prefill_config
             )
         self.kv_cache_quant: bool = kwargs.pop("kv_cache_quant", False)
+        # GPT-OSS MXFP4 MoE weights kept packed (e2m1 + e8m0), dequantized
+        # in-kernel (ref models/gpt_oss/mx_layout_transform.py)
+        self.moe_mxfp4: bool = kwargs.pop("moe_mxfp4", False)
         self.flash_decoding_enabled: bool = kwargs.pop("flash_decoding_enabled", False)
         self.num_cores_per_group: int = kwargs.pop("num_cores_per_group", 1)
 

@@ -102,6 +102,9 @@ def setup_run_parser(run: argparse.ArgumentParser):
                      help="store the KV cache in fp8 (OCP e4m3)")
     run.add_argument("--quantized", action="store_true",
                      help="fp8 weight quantization (per-row scales)")
+    run.add_argument("--moe-mxfp4", action="store_true",
+                     help="keep GPT-OSS MoE weights in MXFP4 packing "
+                          "(in-kernel dequant)")
     run.add_argument("--fused-qkv", action="store_true")
     run.add_argument("--fused-gate-up", action="store_true")
     run.add_argument("--sequence-parallel-enabled", action="store_true")
@@ -159,6 +162,7 @@ def create_engine_config(args, model_type: str) -> EngineConfig:
         skip_warmup=args.skip_warmup,
         on_cpu=args.on_cpu,
         kv_cache_quant=args.kv_cache_quant,
+        moe_mxfp4=getattr(args, "moe_mxfp4", False),
         quantized=args.quantized,
         fused_qkv=args.fused_qkv,
         fused_gate_up=args.fused_gate_up,

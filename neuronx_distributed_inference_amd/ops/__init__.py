@@ -653,3 +653,78 @@ def moe_grouped_gemm(x_sorted: torch.Tensor, w: torch.Tensor,
         rows = slice(t * MOE_TILE, (t + 1) * MOE_TILE)
         y[rows] = x_sorted[rows] @ w[e].t().to(x_sorted.dtype)
     return y
+
+
+# ----------------------------------------------------------------------
+# OCP MXFP4 (ref models/gpt_oss/mx_layout_transform.py):
+# e2m1 nibbles packed 2/byte + one e8m0 scale per 32 elements.
+
+_MX4_VALS = torch.tensor(
+    [0.0, 0.5, 1.0, 1.5, 2.0, 3.0, 4.0, 6.0,
+     -0.0, -0.5, -1.0, -1.5, -2.0, -3.0, -4.0, -6.0])
+
+
+def mx4_dequant(wq: torch.Tensor, ws: torch.Tensor) -> torch.Tensor:
+    """wq (..., K/2) u8 packed e2m1 (low nibble first), ws (..., K/32) u8
+    e8m0 -> float (..., K)."""
+    lo = (wq & 0xF).long()
+    hi = (wq >> 4).long()
+    vals = torch.stack([_MX4_VALS.to(wq.device)[lo],
+                        _MX4_VALS.to(wq.device)[hi]], dim=-1).flatten(-2)
+    scale = torch.pow(2.0, ws.float() - 127.0)
+    K = vals.shape[-1]
+    return (vals.view(*vals.shape[:-1], K // 32, 32)
+            * scale.unsqueeze(-1)).view(*vals.shape[:-1], K)
+
+
+def mx4_quantize(w: torch.Tensor):
+    """float (..., K) -> (wq (..., K/2) u8, ws (..., K/32) u8). Per-32
+    e8m0 scale = 2^(floor(log2(amax)) - 2) so amax maps into e2m1 range;
+    round-to-nearest onto the e2m1 grid."""
+    K = w.shape[-1]
+    assert K % 32 == 0
+    g = w.float().view(*w.shape[:-1], K // 32, 32)
+    amax = g.abs().amax(-1).clamp(min=1e-30)
+    e = torch.floor(torch.log2(amax)) - 2.0
+    e = e.clamp(min=-127.0, max=127.0)
+    scale = torch.pow(2.0, e)
+    ws = (e + 127.0).to(torch.uint8)
+    v = (g / scale.unsqueeze(-1)).clamp(min=-6.0, max=6.0)
+    grid = _MX4_VALS[:8]
+    # nearest positive magnitude, then sign
+    d = (v.abs().unsqueeze(-1) - grid.view(1, 1, 1, -1) if v.dim() == 3
+         else (v.abs().unsqueeze(-1) - grid.view(*([1] * v.dim()), -1))).abs()
+    mag = d.argmin(-1)
+    nib = torch.where(v < 0, mag + 8, mag).to(torch.uint8)
+    nib = nib.view(*w.shape[:-1], K)
+    wq = (nib[..., 0::2] | (nib[..., 1::2] << 4)).contiguous()
+    return wq, ws.contiguous()
+
+
+def moe_grouped_gemm_mx4(x_sorted: torch.Tensor, wq: torch.Tensor,
+                         ws: torch.Tensor, expert_ids: torch.Tensor,
+                         bias=None) -> torch.Tensor:
+    """Grouped expert GEMM with MXFP4-packed weights (E, N, K/2)+(E, N,
+    K/32), dequantized in-kernel during LDS staging (~3.8x less weight
+    traffic than bf16). CPU: dequant reference."""
+    Mpad, K = x_sorted.shape
+    N = wq.shape[1]
+    if x_sorted.is_cuda:
+        c = _require_ext("moe_grouped_gemm_mx4")
+        y = torch.empty(Mpad, N, dtype=x_sorted.dtype, device=x_sorted.device)
+        c.moe_grouped_gemm_mx4(y, x_sorted.contiguous(), wq.contiguous(),
+                               ws.contiguous(), expert_ids.contiguous(),
+                               None if bias is None else bias.contiguous())
+        return y
+    y = torch.zeros(Mpad, N, dtype=x_sorted.dtype)
+    for t in range(Mpad // MOE_TILE):
+        e = int(expert_ids[t])
+        if e < 0:
+            continue
+        rows = slice(t * MOE_TILE, (t + 1) * MOE_TILE)
+        wd = mx4_dequant(wq[e], ws[e]).to(x_sorted.dtype)
+        r = x_sorted[rows] @ wd.t()
+        if bias is not None:
+            r = r + bias[e].to(r.dtype)
+        y[rows] = r
+    return y

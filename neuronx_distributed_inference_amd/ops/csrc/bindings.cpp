@@ -48,7 +48,11 @@ void fp8_quant(torch::Tensor x, torch::Tensor out, torch::Tensor scale,
 void fp8_epilogue(torch::Tensor out, torch::Tensor y32, torch::Tensor w_scale,
                   c10::optional<torch::Tensor> bias);
 void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
-                      torch::Tensor expert_ids);
+                      torch::Tensor expert_ids,
+                      c10::optional<torch::Tensor> bias);
+void moe_grouped_gemm_mx4(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
+                          torch::Tensor ws, torch::Tensor expert_ids,
+                          c10::optional<torch::Tensor> bias);
 bool lt_available();
 void lt_linear(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                c10::optional<torch::Tensor> bias);
@@ -103,7 +107,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("out"), py::arg("y32"), py::arg("w_scale"),
           py::arg("bias") = py::none());
     m.def("moe_grouped_gemm", &moe_grouped_gemm,
-          "grouped expert GEMM over sorted padded token segments (CDNA4)");
+          "grouped expert GEMM over sorted padded token segments (CDNA4)",
+          py::arg("y"), py::arg("x"), py::arg("w"), py::arg("expert_ids"),
+          py::arg("bias") = py::none());
+    m.def("moe_grouped_gemm_mx4", &moe_grouped_gemm_mx4,
+          "grouped expert GEMM, MXFP4 packed weights dequantized in-kernel (CDNA4)",
+          py::arg("y"), py::arg("x"), py::arg("wq"), py::arg("ws"),
+          py::arg("expert_ids"), py::arg("bias") = py::none());
     m.def("lt_available", &lt_available, "hipblaslt usable on this device");
     m.def("lt_linear", &lt_linear, "autotuned hipBLASLt linear (graph-safe)",
           py::arg("y"), py::arg("x"), py::arg("w"), py::arg("bias") = py::none());

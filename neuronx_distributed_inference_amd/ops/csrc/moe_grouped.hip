@@ -73,6 +73,7 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
     const T* __restrict__ x,        // (Mpad, K) sorted-space input
     const T* __restrict__ w,        // (E, N, K) stacked expert weights
     const int* __restrict__ expert_ids,  // (n_row_tiles,) expert or -1
+    const T* __restrict__ bias,     // (E, N) per-expert bias or null
     const int N, const int K) {
     using MF = MgMfma<T>;
     using vec8 = typename MF::vec8;
@@ -176,13 +177,180 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
     mg_vm_wait<0>();
 
     // epilogue: y[m0 + wm0 + qi*16 + lslot*4 + r][n0 + wn0 + lrow]
+    const int n_out = n0 + wn0 + lrow;
+    const float bv = bias ? VecTraits<T>::to_float(bias[(long)e * N + n_out]) : 0.f;
 #pragma unroll
     for (int qi = 0; qi < 2; ++qi) {
-        const int n = n0 + wn0 + lrow;
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
             const int m = m0 + wm0 + qi * 16 + lslot * 4 + r;
-            y[(long)m * N + n] = VecTraits<T>::from_float(acc[qi][r]);
+            y[(long)m * N + n_out] = VecTraits<T>::from_float(acc[qi][r] + bv);
+        }
+    }
+}
+
+
+// ---------------------------------------------------------------------------
+// MXFP4 variant (ref models/gpt_oss/mx_layout_transform.py):
+// expert weights stay PACKED in HBM as OCP MXFP4 — e2m1 nibbles
+// (E, N, K/2) u8 + e8m0 block scales (E, N, K/32) u8, 4.25 bits/weight —
+// and are dequantized to the bf16 LDS image during staging (register
+// stage + ds_write; x keeps glds). Weight traffic drops ~3.8x vs bf16,
+// which is the decode-side win on this memory-bound path; the fp4-MFMA
+// compute path (mfma_scale_*_f8f6f4) is a further step, not needed to
+// beat the bf16 stream.
+
+// decode 8 e2m1 nibbles (packed u32) * e8m0 scale into bf16x8.
+// magnitude table x2 packed as nibbles: [0,1,2,3,4,6,8,12] (= 2*|v|)
+template <typename T>
+DEV_INLINE void mx4_dequant8(unsigned p, float half_scale, T* out) {
+    constexpr unsigned MAG2 = 0xC8643210u;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+        const unsigned nib = (p >> (4 * j)) & 0xF;
+        float f = (float)((MAG2 >> ((nib & 7) * 4)) & 0xF);
+        f = (nib & 8) ? -f : f;
+        out[j] = VecTraits<T>::from_float(f * half_scale);
+    }
+}
+
+template <typename T>
+__global__ __launch_bounds__(512, 2) void moe_grouped_mx4_kernel(
+    T* __restrict__ y,                   // (Mpad, N)
+    const T* __restrict__ x,             // (Mpad, K)
+    const unsigned char* __restrict__ wq,  // (E, N, K/2) packed e2m1
+    const unsigned char* __restrict__ ws,  // (E, N, K/32) e8m0 scales
+    const int* __restrict__ expert_ids,  // (n_row_tiles,)
+    const T* __restrict__ bias,          // (E, N) or null
+    const int N, const int K) {
+    using MF = MgMfma<T>;
+    using vec8 = typename MF::vec8;
+    constexpr int MT = 64;
+    constexpr int G = 1;                 // glds per wave per chunk (x only)
+    constexpr int XBYTES = MT * 128;
+    constexpr int CHUNK_B = XBYTES + MG_BN * 128;
+
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+
+    const int tile = blockIdx.x;
+    const int rt = blockIdx.y;
+    const int e = expert_ids[rt];
+    if (e < 0) return;
+    const int n0 = tile * MG_BN;
+    const int m0 = rt * MT;
+    const int nchunks = K / MG_BK;
+
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int wm0 = (wid >> 2) * 32;
+    const int wn0 = (wid & 3) * 16;
+    const int lrow = lane & 15;
+    const int lslot = lane >> 4;
+
+    const char* xsrc = reinterpret_cast<const char*>(x) +
+                       ((long)(m0 + wid * 8) * K) * sizeof(T);
+    // this lane stages W row (n0 + wid*8 + lane/8), k-slot lane%8
+    const int wrow_l = wid * 8 + (lane >> 3);
+    const int wslot = lane & 7;          // 8 bf16 = 4 packed bytes
+    const unsigned char* wq_row = wq + ((long)e * N + n0 + wrow_l) * (K / 2);
+    const unsigned char* ws_row = ws + ((long)e * N + n0 + wrow_l) * (K / 32);
+
+    char* xdst[MG_NBUF];
+    char* wdst[MG_NBUF];
+#pragma unroll
+    for (int b = 0; b < MG_NBUF; ++b) {
+        xdst[b] = smem + b * CHUNK_B + wid * 1024;
+        wdst[b] = smem + b * CHUNK_B + XBYTES;
+    }
+
+    // x: glds with counted vmcnt; W: ordinary loads + in-register
+    // dequant + swizzled ds_write (the ds_write slot XOR matches the
+    // read-side XOR; note hipcc waits vmcnt(0) for the W loads while a
+    // glds is in flight — the known mixed-kind cost, acceptable v1)
+#define MX_STAGE_X(t, b)                                                     \
+    mg_glds_rows128<0>(xsrc + (long)(t) * 128, (long)K * sizeof(T), xdst[b])
+#define MX_STAGE_W(t, b)                                                     \
+    do {                                                                     \
+        const int kc = (t) * MG_BK;                                          \
+        const unsigned pk = *reinterpret_cast<const unsigned*>(              \
+            wq_row + (kc + wslot * 8) / 2);                                  \
+        const float sc = exp2f((int)ws_row[(kc + wslot * 8) / 32] - 127);    \
+        T tmp[8];                                                            \
+        mx4_dequant8<T>(pk, 0.5f * sc, tmp);                                 \
+        const int sl = wslot ^ (lane >> 3);                                  \
+        *reinterpret_cast<int4*>(wdst[b] + wrow_l * 128 + sl * 16) =         \
+            *reinterpret_cast<int4*>(tmp);                                   \
+    } while (0)
+
+    mg_f32x4 acc[2];
+    acc[0] = {0.f, 0.f, 0.f, 0.f};
+    acc[1] = {0.f, 0.f, 0.f, 0.f};
+
+#define MX_FRAGS(b)                                                          \
+    do {                                                                     \
+        const char* xb = smem + (b) * CHUNK_B;                               \
+        const char* wb = xb + XBYTES;                                        \
+        _Pragma("unroll") for (int ks = 0; ks < 2; ++ks) {                   \
+            vec8 a0, a1, bb;                                                 \
+            {                                                                \
+                const int m = wm0 + lrow;                                    \
+                const int sfrag = (ks * 4 + lslot) ^ (m & 7);                \
+                a0 = *reinterpret_cast<const vec8*>(xb + m * 128 + sfrag * 16); \
+            }                                                                \
+            {                                                                \
+                const int m = wm0 + 16 + lrow;                               \
+                const int sfrag = (ks * 4 + lslot) ^ (m & 7);                \
+                a1 = *reinterpret_cast<const vec8*>(xb + m * 128 + sfrag * 16); \
+            }                                                                \
+            {                                                                \
+                const int n = wn0 + lrow;                                    \
+                const int sfrag = (ks * 4 + lslot) ^ (n & 7);                \
+                bb = *reinterpret_cast<const vec8*>(wb + n * 128 + sfrag * 16); \
+            }                                                                \
+            acc[0] = MF::mfma(a0, bb, acc[0]);                               \
+            acc[1] = MF::mfma(a1, bb, acc[1]);                               \
+        }                                                                    \
+    } while (0)
+
+#define MX_BODY(t, b, b2)                                                    \
+    do {                                                                     \
+        if ((t) + 1 < nchunks) mg_vm_wait<G>();                              \
+        else mg_vm_wait<0>();                                                \
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                   \
+        __builtin_amdgcn_s_barrier();                                        \
+        MX_FRAGS(b);                                                         \
+        if ((t) + 2 < nchunks) {                                             \
+            MX_STAGE_X((t) + 2, b2);                                         \
+            MX_STAGE_W((t) + 2, b2);                                         \
+        }                                                                    \
+    } while (0)
+
+    MX_STAGE_X(0, 0);
+    MX_STAGE_W(0, 0);
+    if (nchunks > 1) { MX_STAGE_X(1, 1); MX_STAGE_W(1, 1); }
+    int t = 0;
+    while (t + 3 <= nchunks) {
+        MX_BODY(t + 0, 0, 2);
+        MX_BODY(t + 1, 1, 0);
+        MX_BODY(t + 2, 2, 1);
+        t += 3;
+    }
+    if (t < nchunks) MX_BODY(t, 0, 2);
+    if (t + 1 < nchunks) MX_BODY(t + 1, 1, 0);
+#undef MX_BODY
+#undef MX_FRAGS
+#undef MX_STAGE_W
+#undef MX_STAGE_X
+    mg_vm_wait<0>();
+
+    const int n_out = n0 + wn0 + lrow;
+    const float bv = bias ? VecTraits<T>::to_float(bias[(long)e * N + n_out]) : 0.f;
+#pragma unroll
+    for (int qi = 0; qi < 2; ++qi) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int m = m0 + wm0 + qi * 16 + lslot * 4 + r;
+            y[(long)m * N + n_out] = VecTraits<T>::from_float(acc[qi][r] + bv);
         }
     }
 }
@@ -190,7 +358,8 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
 }  // namespace
 
 void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
-                      torch::Tensor expert_ids) {
+                      torch::Tensor expert_ids,
+                      c10::optional<torch::Tensor> bias) {
     TORCH_CHECK(y.dim() == 2 && x.dim() == 2 && w.dim() == 3, "shapes");
     TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && y.is_contiguous() &&
                 expert_ids.is_contiguous(), "contiguous tensors required");
@@ -219,6 +388,7 @@ void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
         hipLaunchKernelGGL((moe_grouped_kernel<T>), grid, dim3(512), LDS,     \
                            stream, (T*)y.data_ptr(), (const T*)x.data_ptr(),  \
                            (const T*)w.data_ptr(), expert_ids.data_ptr<int>(),\
+                           bias ? (const T*)bias->data_ptr() : nullptr,       \
                            N, K);                                             \
     } while (0)
 
@@ -230,5 +400,54 @@ void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
         TORCH_CHECK(false, "moe_grouped_gemm: bf16/f16 only");
     }
 #undef MG_T
+    HIP_CHECK_KERNEL();
+}
+
+void moe_grouped_gemm_mx4(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
+                          torch::Tensor ws, torch::Tensor expert_ids,
+                          c10::optional<torch::Tensor> bias) {
+    TORCH_CHECK(y.dim() == 2 && x.dim() == 2 && wq.dim() == 3 && ws.dim() == 3,
+                "shapes");
+    TORCH_CHECK(x.is_contiguous() && wq.is_contiguous() && ws.is_contiguous() &&
+                y.is_contiguous() && expert_ids.is_contiguous(),
+                "contiguous tensors required");
+    const int Mpad = x.size(0), K = x.size(1);
+    const int N = wq.size(1);
+    TORCH_CHECK(wq.size(2) == K / 2 && ws.size(2) == K / 32, "packed shape");
+    TORCH_CHECK(y.size(0) == Mpad && y.size(1) == N, "shape");
+    TORCH_CHECK(N % MG_BN == 0 && K % MG_BK == 0 && Mpad % 64 == 0, "align");
+    TORCH_CHECK(wq.scalar_type() == at::ScalarType::Byte &&
+                ws.scalar_type() == at::ScalarType::Byte, "u8 packed");
+
+    constexpr int LDS = MG_NBUF * (64 * 128 + MG_BN * 128);
+    auto stream = at::hip::getCurrentHIPStream();
+    dim3 grid(N / MG_BN, Mpad / 64);
+
+#define MGX_T(T)                                                              \
+    do {                                                                      \
+        static bool attr_set = false;                                         \
+        if (!attr_set) {                                                      \
+            (void)hipFuncSetAttribute((const void*)moe_grouped_mx4_kernel<T>, \
+                                      hipFuncAttributeMaxDynamicSharedMemorySize, \
+                                      LDS);                                   \
+            attr_set = true;                                                  \
+        }                                                                     \
+        hipLaunchKernelGGL((moe_grouped_mx4_kernel<T>), grid, dim3(512), LDS, \
+                           stream, (T*)y.data_ptr(), (const T*)x.data_ptr(),  \
+                           (const unsigned char*)wq.data_ptr(),               \
+                           (const unsigned char*)ws.data_ptr(),               \
+                           expert_ids.data_ptr<int>(),                        \
+                           bias.has_value() ? (const T*)bias->data_ptr() : nullptr, \
+                           N, K);                                             \
+    } while (0)
+
+    if (x.scalar_type() == at::ScalarType::BFloat16) {
+        MGX_T(__hip_bfloat16);
+    } else if (x.scalar_type() == at::ScalarType::Half) {
+        MGX_T(__half);
+    } else {
+        TORCH_CHECK(false, "moe_grouped_gemm_mx4: bf16/f16 only");
+    }
+#undef MGX_T
     HIP_CHECK_KERNEL();
 }

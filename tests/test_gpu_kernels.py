@@ -882,3 +882,77 @@ def test_moe_grouped_matches_dense_gpu():
     tg = t(lambda: m._forward_grouped(x, w, idx))
     td = t(lambda: m._forward_dense(x, w, idx))
     print(f"\nMoE T={T} grouped={tg:.0f}us dense={td:.0f}us ratio={tg/td:.2f}")
+
+
+def test_moe_grouped_mx4_kernel_gpu():
+    """MXFP4 grouped kernel (in-LDS dequant) vs the CPU dequant reference
+    on the same packed weights, with per-expert bias."""
+    torch.manual_seed(0)
+    dev = _dev()
+    E, N, K = 4, 256, 512
+    w = torch.randn(E, N, K) * 0.2
+    wq, ws = ops.mx4_quantize(w)
+    bias = torch.randn(E, N, dtype=torch.bfloat16) * 0.1
+    idx = torch.randint(0, E, (90,))
+    order, dest, expert_ids, Mpad = ops.moe_grouped_tables(idx, E)
+    x_rows = torch.randn(90, K, dtype=torch.bfloat16) * 0.1
+    x_sorted = torch.zeros(Mpad, K, dtype=torch.bfloat16)
+    x_sorted.index_copy_(0, dest, x_rows)
+    want = ops.moe_grouped_gemm_mx4(x_sorted, wq, ws, expert_ids, bias=bias)
+    got = ops.moe_grouped_gemm_mx4(
+        x_sorted.to(dev), wq.to(dev), ws.to(dev), expert_ids.to(dev),
+        bias=bias.to(dev))
+    torch.testing.assert_close(got.index_select(0, dest.to(dev)).float().cpu(),
+                               want.index_select(0, dest).float(),
+                               atol=3e-2, rtol=3e-2)
+
+
+def test_gpt_oss_mxfp4_gpu_matches_dequant():
+    """GPT-OSS expert layer on GPU: MXFP4 grouped path vs the bf16 dense
+    weights produced by dequantizing the same packing."""
+    from neuronx_distributed_inference_amd.config import EngineConfig
+    from neuronx_distributed_inference_amd.models.gpt_oss.modeling_gpt_oss import (
+        GptOssExperts,
+        GptOssInferenceConfig,
+    )
+
+    torch.manual_seed(1)
+    dev = _dev()
+    E, H, I = 4, 256, 256
+
+    def mkcfg(mx):
+        ec = EngineConfig(batch_size=2, seq_len=32, max_length=32,
+                          torch_dtype="bfloat16", tp_degree=1, moe_mxfp4=mx)
+        return GptOssInferenceConfig(
+            neuron_config=ec, vocab_size=64, hidden_size=H,
+            intermediate_size=I, num_hidden_layers=1, num_attention_heads=4,
+            num_key_value_heads=2, num_local_experts=E, num_experts_per_tok=2,
+            max_position_embeddings=64, rms_norm_eps=1e-5, rope_theta=10000.0,
+            layer_types=["full_attention"], tie_word_embeddings=False)
+
+    mx = GptOssExperts(mkcfg(True))
+    w_gu = torch.randn(E, 2 * I, H) * 0.2
+    w_dn = torch.randn(E, H, I) * 0.2
+    with torch.no_grad():
+        q, s = ops.mx4_quantize(w_gu)
+        mx.gate_up_q.copy_(q)
+        mx.gate_up_s.copy_(s)
+        q, s = ops.mx4_quantize(w_dn)
+        mx.down_q.copy_(q)
+        mx.down_s.copy_(s)
+        mx.gate_up_proj_bias.normal_(0, 0.05)
+        mx.down_proj_bias.normal_(0, 0.05)
+    dense = GptOssExperts(mkcfg(False))
+    with torch.no_grad():
+        dense.gate_up_proj.copy_(ops.mx4_dequant(mx.gate_up_q, mx.gate_up_s))
+        dense.down_proj.copy_(ops.mx4_dequant(mx.down_q, mx.down_s))
+        dense.gate_up_proj_bias.copy_(mx.gate_up_proj_bias)
+        dense.down_proj_bias.copy_(mx.down_proj_bias)
+    mx, dense = mx.to(dev), dense.to(dev)
+    T = 64
+    x = (torch.randn(T, H) * 0.3).to(dev, torch.bfloat16)
+    wts = torch.softmax(torch.randn(T, 2, device=dev), dim=-1)
+    idx = torch.randint(0, E, (T, 2), device=dev)
+    got = mx(x, wts, idx)
+    want = dense(x, wts, idx)
+    torch.testing.assert_close(got.float(), want.float(), atol=4e-2, rtol=4e-2)

@@ -431,3 +431,67 @@ def test_moe_grouped_masks_nonlocal_experts():
     got = m._forward_grouped(x, w, idx)
     want = m._forward_grouped(x, w[:, :1], idx[:, :1])
     torch.testing.assert_close(got, want, atol=1e-5, rtol=1e-5)
+
+
+def test_gpt_oss_mxfp4_experts_match_dequantized():
+    """GPT-OSS experts with MXFP4-packed weights (grouped path, CPU
+    reference dequant) must match the dense path on the SAME dequantized
+    values — the packing is exact given identical dequant."""
+    import torch
+
+    from neuronx_distributed_inference_amd import ops
+    from neuronx_distributed_inference_amd.config import EngineConfig
+    from neuronx_distributed_inference_amd.models.gpt_oss.modeling_gpt_oss import (
+        GptOssExperts,
+        GptOssInferenceConfig,
+    )
+
+    torch.manual_seed(0)
+    E, H, I = 4, 64, 64
+    ec = EngineConfig(batch_size=2, seq_len=32, max_length=32,
+                      torch_dtype="float32", tp_degree=1, moe_mxfp4=True)
+    cfg = GptOssInferenceConfig(
+        neuron_config=ec, vocab_size=64, hidden_size=H, intermediate_size=I,
+        num_hidden_layers=1, num_attention_heads=4, num_key_value_heads=2,
+        num_local_experts=E, num_experts_per_tok=2, max_position_embeddings=64,
+        rms_norm_eps=1e-5, rope_theta=10000.0, layer_types=["full_attention"],
+        tie_word_embeddings=False)
+    mx = GptOssExperts(cfg)
+    assert mx.mxfp4
+    w_gu = torch.randn(E, 2 * I, H) * 0.2
+    w_dn = torch.randn(E, H, I) * 0.2
+    with torch.no_grad():
+        q, s = ops.mx4_quantize(w_gu)
+        mx.gate_up_q.copy_(q)
+        mx.gate_up_s.copy_(s)
+        q, s = ops.mx4_quantize(w_dn)
+        mx.down_q.copy_(q)
+        mx.down_s.copy_(s)
+        mx.gate_up_proj_bias.normal_(0, 0.05)
+        mx.down_proj_bias.normal_(0, 0.05)
+
+    ec2 = EngineConfig(batch_size=2, seq_len=32, max_length=32,
+                       torch_dtype="float32", tp_degree=1)
+    cfg2 = GptOssInferenceConfig(
+        neuron_config=ec2, vocab_size=64, hidden_size=H, intermediate_size=I,
+        num_hidden_layers=1, num_attention_heads=4, num_key_value_heads=2,
+        num_local_experts=E, num_experts_per_tok=2, max_position_embeddings=64,
+        rms_norm_eps=1e-5, rope_theta=10000.0, layer_types=["full_attention"],
+        tie_word_embeddings=False)
+    dense = GptOssExperts(cfg2)
+    with torch.no_grad():
+        dense.gate_up_proj.copy_(ops.mx4_dequant(mx.gate_up_q, mx.gate_up_s))
+        dense.down_proj.copy_(ops.mx4_dequant(mx.down_q, mx.down_s))
+        dense.gate_up_proj_bias.copy_(mx.gate_up_proj_bias)
+        dense.down_proj_bias.copy_(mx.down_proj_bias)
+
+    T = 7
+    x = torch.randn(T, H) * 0.3
+    wts = torch.softmax(torch.randn(T, 2), dim=-1)
+    idx = torch.randint(0, E, (T, 2))
+    got = mx(x, wts, idx)
+    want = dense(x, wts, idx)
+    torch.testing.assert_close(got, want, atol=1e-4, rtol=1e-4)
+    # and the grouped CPU path agrees with the per-expert loop
+    got_g = mx._forward_grouped(x, wts, idx)
+    torch.testing.assert_close(got_g, want, atol=1e-4, rtol=1e-4)
