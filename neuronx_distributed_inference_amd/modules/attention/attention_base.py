@@ -307,9 +307,20 @@ class AttentionBase(nn.Module):
 
         # decode-sharded K/V into the cache (standard projections; the q
         # output of this pass is discarded — the ref instead reorders kv
-        # heads between meshes, attention/utils.py:455-623)
+        # heads between meshes, attention/utils.py:455-623). With
+        # flash decoding the decode cache is S-SHARDED: map positions to
+        # this rank's span exactly like the non-CP write path (the
+        # CP+flash-decode seam bug the combined gloo test caught).
         _, k_full, v_full = self.prep_qkv(hidden, cos, sin, position_ids=position_ids)
-        kv_mgr.update_cache(self.layer_idx, k_full, v_full, position_ids, seq_ids)
+        wpos = position_ids
+        if self.kv_shard_degree > 1:
+            from neuronx_distributed_inference_amd.modules.attention import (
+                flash_decoding as _fdm,
+            )
+
+            wpos = _fdm.shard_write_positions(
+                position_ids, self.kv_shard_degree, ps.get_kv_shard_rank())
+        kv_mgr.update_cache(self.layer_idx, k_full, v_full, wpos, seq_ids)
 
         # CTE-sharded q/k/v over the prefill-TP subgroup
         qc, kc, vc = self.cte_qkv_proj(hidden)

@@ -569,3 +569,71 @@ def _worker_flash_decoding_e2e(rank):
 
 def test_flash_decoding_e2e_generation():
     spawn("_worker_flash_decoding_e2e")
+
+
+# ---------------------------------------------------------------------------
+def _worker_cp_flash_decoding_e2e(rank):
+    """COMBINED seam test: context-parallel prefill
+    (tp=2, cp=2) writing into an S-sharded flash-decoding cache
+    (kv_shard=2) — the CP query reorder and the cross-rank LSE-combined
+    decode must compose to the single-rank result."""
+    from neuronx_distributed_inference_amd.config import EngineConfig
+    from neuronx_distributed_inference_amd.models.llama.modeling_llama import (
+        LlamaForCausalLM,
+        LlamaInferenceConfig,
+    )
+    from neuronx_distributed_inference_amd.parallel import state as ps
+    from neuronx_distributed_inference_amd.utils.checkpoint_sharding import (
+        shard_state_dict_for_rank,
+    )
+    from neuronx_distributed_inference_amd.utils.hf_adapter import (
+        GenerationConfig,
+        HuggingFaceGenerationAdapter,
+    )
+
+    def make_cfg(tp, cp, flash):
+        ec = EngineConfig(batch_size=2, max_batch_size=2, seq_len=64,
+                          max_length=64, torch_dtype="float32", tp_degree=tp,
+                          cp_degree=cp, flash_decoding_enabled=flash)
+        return LlamaInferenceConfig(
+            neuron_config=ec, vocab_size=128, hidden_size=64,
+            intermediate_size=128, num_hidden_layers=2,
+            num_attention_heads=4, num_key_value_heads=1,
+            max_position_embeddings=128, rms_norm_eps=1e-5,
+            rope_theta=10000.0, hidden_act="silu", tie_word_embeddings=False)
+
+    ps.destroy_model_parallel()
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(29)
+    full = LlamaForCausalLM(make_cfg(1, 1, False)).eval()
+    with torch.no_grad():
+        for p in full.parameters():
+            if p.dim() >= 2:
+                p.normal_(0, 0.1)
+    ids = torch.randint(0, 128, (2, 8))  # S divisible by cp
+    want = HuggingFaceGenerationAdapter(full).generate(
+        ids, generation_config=GenerationConfig(max_new_tokens=8))
+
+    ps.destroy_model_parallel()
+    ps.initialize_model_parallel(tensor_model_parallel_size=WORLD,
+                                 context_parallel_size=WORLD,
+                                 kv_shard_size=WORLD)
+    model = LlamaForCausalLM(make_cfg(WORLD, WORLD, True)).eval()
+    assert model.model.kv_mgr.max_len == 32  # S-sharded cache
+    full_sd = {k: v for k, v in full.model.state_dict().items()
+               if not k.startswith("kv_mgr.")}
+    for k in list(full_sd):
+        if ".self_attn.qkv_proj." in k:
+            full_sd[k.replace(".qkv_proj.", ".cte_qkv_proj.")] = full_sd[k]
+        elif ".self_attn.o_proj.o_proj." in k:
+            full_sd[k.replace(".o_proj.o_proj.", ".cte_o_proj.o_proj.")] = full_sd[k]
+    shard = shard_state_dict_for_rank(full_sd, model.model, rank, WORLD)
+    model.model.load_state_dict(shard, strict=False)
+    got = HuggingFaceGenerationAdapter(model).generate(
+        ids, generation_config=GenerationConfig(max_new_tokens=8))
+    assert torch.equal(got, want), (got.tolist(), want.tolist())
+    return True
+
+
+def test_cp_plus_flash_decoding_e2e():
+    spawn("_worker_cp_flash_decoding_e2e")
