@@ -178,6 +178,11 @@ class AttentionBase(nn.Module):
             S = S * self.tp_degree  # hidden is S-sharded; attention runs full-S
         if not self.use_rope:  # NoPE layer (Llama4 global-attention layers)
             cos, sin = None, None
+        # a 4-D mask in TKG is an EXPLICIT attention mask (tree-structured
+        # speculation); 2-D masks stay padding masks
+        if (active_mask is None and attention_mask is not None
+                and attention_mask.dim() == 4 and phase == PHASE_TKG):
+            active_mask = attention_mask
 
         # Fused TKG front end (K2-analog): one GEMM -> one kernel doing
         # RoPE + cache write + q relayout. Skipped when per-head qk-norms or
@@ -465,6 +470,23 @@ class AttentionBase(nn.Module):
 
     def _decode(self, q, kv_mgr, position_ids, seq_ids, active_mask) -> torch.Tensor:
         k_cache, v_cache = kv_mgr.get_cache(self.layer_idx)
+        if active_mask is not None and active_mask.dim() == 4:
+            # EXPLICIT-mask decode (tree-structured speculation: a node
+            # attends the prior plus its ancestor chain only — position
+            # causality is wrong for sibling nodes; ref tree TKG
+            # model_base.py:2094 + token_tree masks). Runs masked
+            # attention against the cache prefix the mask covers.
+            kc = k_cache.data if hasattr(k_cache, "data") else k_cache
+            vc = v_cache.data if hasattr(v_cache, "data") else v_cache
+            B = q.shape[0]
+            rows = (seq_ids.long().clamp_min(0) if seq_ids is not None
+                    else torch.arange(B, device=q.device))
+            Sm = active_mask.shape[-1]
+            k = kc[rows][:, :, :Sm].to(q.dtype)
+            v = vc[rows][:, :, :Sm].to(q.dtype)
+            return ops.attention_prefill(q, k, v, mask=active_mask,
+                                         scale=self.scale, sinks=self.sinks,
+                                         causal=False)
         # per-query positions (B,Sq): exact for ragged chunks (-1 pads)
         if getattr(kv_mgr, "is_paged", False):
             return ops.attention_decode_paged(

@@ -112,6 +112,25 @@ class KVCacheManager(nn.Module):
         return torch.where(positions >= 0, positions % sz,
                            torch.full_like(positions, -1))
 
+    def compact_rows(self, seq_ids: torch.Tensor, src: torch.Tensor,
+                     dst: torch.Tensor):
+        """Move cache rows src[b, j] -> dst[b, j] for every layer (tree
+        speculation compaction: accepted tree nodes live at scattered
+        rows L+node_idx and must become the contiguous committed prefix —
+        ref medusa/tree KV gather-scatter, kv_cache_manager.py:265-321).
+        Advanced indexing reads the full source before writing, so
+        overlapping src/dst are safe."""
+        if src.numel() == 0:
+            return
+        rows = seq_ids.long().clamp_min(0).view(-1, 1)
+        src = src.long()
+        dst = dst.long()
+        for li in range(self.num_layers):
+            kc = self.k_caches[li].data
+            vc = self.v_caches[li].data
+            kc[rows, :, dst] = kc[rows, :, src]
+            vc[rows, :, dst] = vc[rows, :, src]
+
     def reset(self):
         for k in self.k_caches:
             k.data.zero_()
