@@ -30,6 +30,7 @@ import torch.nn.functional as F
 from neuronx_distributed_inference_amd.parallel import mappings
 from neuronx_distributed_inference_amd.parallel.layers import (
     ColumnParallelLinear,
+    ParallelEmbedding,
     RowParallelLinear,
 )
 
@@ -157,6 +158,97 @@ class MultiLoraLinear(nn.Module):
                 self.lora_B[slot, :, :r].copy_(b.to(self.lora_B.dtype))
 
 
+
+
+class MultiLoraEmbedding(nn.Module):
+    """Embedding with stacked per-adapter low-rank deltas (ref
+    MultiLoraEmbedding, lora_layer.py:10-239): delta = (one_hot(ids) @ A^T)
+    @ B^T computed as A-row gather + B projection. Wraps ParallelEmbedding
+    (vocab-sharded base; LoRA factors replicated — adapter deltas are
+    dense over the hidden dim, so the delta adds AFTER the base's
+    all-reduce)."""
+
+    def __init__(self, base: nn.Module, cfg: LoraServingConfig, state: LoraState):
+        super().__init__()
+        self.base = base
+        self.cfg = cfg
+        self.state = state
+        r = cfg.max_lora_rank
+        vocab, hidden = base.num_embeddings, base.embedding_dim
+        dtype = base.weight.dtype
+        # A: (L, vocab, r) gathered by token id; B: (L, hidden, r)
+        self.lora_A = nn.Parameter(
+            torch.zeros(cfg.max_loras, vocab, r, dtype=dtype), requires_grad=False)
+        self.lora_B = nn.Parameter(
+            torch.zeros(cfg.max_loras, hidden, r, dtype=dtype), requires_grad=False)
+        self.scaling = cfg.lora_alpha / max(r, 1)
+
+    def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
+        y = self.base(input_ids)
+        ids = self.state.adapter_ids[: input_ids.shape[0]]
+        A = self.lora_A[ids]                       # (B, V, r)
+        h = A.gather(1, input_ids.unsqueeze(-1).expand(
+            -1, -1, A.shape[-1]))                  # (B, S, r)
+        Bm = self.lora_B[ids]                      # (B, H, r)
+        d = torch.einsum("bsr,bhr->bsh", h.float(), Bm.float())
+        return y + (d * self.scaling).to(y.dtype)
+
+    def load_adapter(self, slot: int, a: torch.Tensor, b: torch.Tensor,
+                     tp_rank: int = 0, tp_degree: int = 1):
+        """a: (r, vocab) (HF embedding lora_A is (r, num_embeddings)),
+        b: (hidden, r)."""
+        r = a.shape[0]
+        with torch.no_grad():
+            self.lora_A[slot].zero_()
+            self.lora_B[slot].zero_()
+            self.lora_A[slot, :, :r].copy_(a.t().to(self.lora_A.dtype))
+            self.lora_B[slot, :, :r].copy_(b.to(self.lora_B.dtype))
+
+
+class MultiLoraConv2d(nn.Module):
+    """Conv2d with stacked per-adapter low-rank deltas (vision towers;
+    ref MultiLoraConv2d, lora_layer.py:10-239). delta(x) = conv(x, B) of
+    conv(x, A) where A: (r, Cin, kh, kw) with the base stride/padding and
+    B: (Cout, r, 1, 1) — grouped per request via a batched loop over the
+    distinct adapters in the batch (vision prefill is once per request)."""
+
+    def __init__(self, base: nn.Conv2d, cfg: LoraServingConfig, state: LoraState):
+        super().__init__()
+        self.base = base
+        self.cfg = cfg
+        self.state = state
+        r = cfg.max_lora_rank
+        dtype = base.weight.dtype
+        self.lora_A = nn.Parameter(
+            torch.zeros(cfg.max_loras, r, base.in_channels,
+                        *base.kernel_size, dtype=dtype), requires_grad=False)
+        self.lora_B = nn.Parameter(
+            torch.zeros(cfg.max_loras, base.out_channels, r, 1, 1, dtype=dtype),
+            requires_grad=False)
+        self.scaling = cfg.lora_alpha / max(r, 1)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        y = self.base(x)
+        ids = self.state.adapter_ids[: x.shape[0]]
+        for slot in ids.unique().tolist():
+            rows = ids == slot
+            h = F.conv2d(x[rows], self.lora_A[slot], stride=self.base.stride,
+                         padding=self.base.padding, dilation=self.base.dilation)
+            d = F.conv2d(h, self.lora_B[slot])
+            y[rows] = y[rows] + (d * self.scaling).to(y.dtype)
+        return y
+
+    def load_adapter(self, slot: int, a: torch.Tensor, b: torch.Tensor,
+                     tp_rank: int = 0, tp_degree: int = 1):
+        r = a.shape[0]
+        with torch.no_grad():
+            self.lora_A[slot].zero_()
+            self.lora_B[slot].zero_()
+            self.lora_A[slot, :r].copy_(a.to(self.lora_A.dtype))
+            self.lora_B[slot, :, :r].copy_(b.view(*b.shape[:2], 1, 1)
+                                           .to(self.lora_B.dtype))
+
+
 def wrap_model_with_lora(model: nn.Module, cfg: LoraServingConfig,
                          max_batch: int) -> LoraState:
     """Swap every target (parallel) linear for a MultiLoraLinear
@@ -170,11 +262,16 @@ def wrap_model_with_lora(model: nn.Module, cfg: LoraServingConfig,
                 setattr(module, child_name, MultiLoraLinear(child, cfg, state, "column"))
             elif isinstance(child, RowParallelLinear):
                 setattr(module, child_name, MultiLoraLinear(child, cfg, state, "row"))
+            elif isinstance(child, ParallelEmbedding):
+                setattr(module, child_name, MultiLoraEmbedding(child, cfg, state))
+            elif isinstance(child, nn.Conv2d):
+                setattr(module, child_name, MultiLoraConv2d(child, cfg, state))
     return state
 
 
-def lora_layers(model: nn.Module) -> Dict[str, MultiLoraLinear]:
-    return {n: m for n, m in model.named_modules() if isinstance(m, MultiLoraLinear)}
+def lora_layers(model: nn.Module) -> Dict[str, nn.Module]:
+    return {n: m for n, m in model.named_modules()
+            if isinstance(m, (MultiLoraLinear, MultiLoraEmbedding, MultiLoraConv2d))}
 
 
 class AdapterCache:

@@ -111,3 +111,55 @@ def test_adapter_cache_lru(model):
     assert "a" not in cache.slot_of
     ids = cache.set_request_adapters(["b", ""])
     assert ids[1] == 0  # '' -> zero adapter slot
+
+
+def test_multi_lora_embedding_and_conv2d():
+    """Embedding/Conv2d LoRA variants (ref lora_layer.py:10-239): delta
+    math vs explicit dense computation, per-request adapter selection."""
+    import torch
+    import torch.nn.functional as F
+
+    from neuronx_distributed_inference_amd.modules.lora_serving import (
+        LoraServingConfig,
+        LoraState,
+        MultiLoraConv2d,
+        MultiLoraEmbedding,
+    )
+    from neuronx_distributed_inference_amd.parallel.layers import ParallelEmbedding
+
+    torch.manual_seed(0)
+    cfg = LoraServingConfig(max_loras=3, max_lora_rank=4, lora_alpha=8)
+    state = LoraState(max_batch=2)
+
+    emb = ParallelEmbedding(32, 16, dtype=torch.float32, tp_degree=1)
+    with torch.no_grad():
+        emb.weight.normal_(0, 0.1)
+    ml = MultiLoraEmbedding(emb, cfg, state)
+    a = torch.randn(4, 32) * 0.1   # (r, vocab)
+    b = torch.randn(16, 4) * 0.1   # (hidden, r)
+    ml.load_adapter(1, a, b)
+    ids = torch.randint(0, 32, (2, 5))
+    state.set(torch.tensor([1, 0]))
+    y = ml(ids)
+    base = emb(ids)
+    onehot = F.one_hot(ids[0], 32).float()
+    want0 = base[0] + (onehot @ a.t() @ b.t()) * (8 / 4)
+    torch.testing.assert_close(y[0], want0, atol=1e-5, rtol=1e-5)
+    torch.testing.assert_close(y[1], base[1])  # slot 0 = empty adapter
+
+    conv = torch.nn.Conv2d(3, 8, 4, stride=4)
+    with torch.no_grad():
+        conv.weight.normal_(0, 0.1)
+        conv.bias.zero_()
+    mc = MultiLoraConv2d(conv, cfg, state)
+    ca = torch.randn(4, 3, 4, 4) * 0.1       # (r, Cin, kh, kw)
+    cb = torch.randn(8, 4) * 0.1             # (Cout, r)
+    mc.load_adapter(1, ca, cb)
+    x = torch.randn(2, 3, 8, 8)
+    state.set(torch.tensor([1, 0]))
+    y = mc(x)
+    base = conv(x)
+    h = F.conv2d(x[:1], ca, stride=4)
+    want0 = base[0] + (F.conv2d(h, cb.view(8, 4, 1, 1)) * (8 / 4))[0]
+    torch.testing.assert_close(y[0], want0, atol=1e-5, rtol=1e-5)
+    torch.testing.assert_close(y[1], base[1])
