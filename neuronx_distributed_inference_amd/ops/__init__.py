@@ -307,6 +307,12 @@ def _sg5_autotune(key, x2, weight, bias) -> int:
             results[(mt, sk)] = float("inf")
     del wcopies
     choice = min(results, key=results.get)
+    # microbench rankings transfer to in-graph steady state with ~±20%
+    # error (DVFS + cache state differ): ship the hand-written kernel
+    # only on a CLEAR win, else keep the library path (measured: near
+    # parity flips either way run to run)
+    if choice != 0 and results[choice] > 0.90 * results.get(0, float("inf")):
+        choice = 0
     _sg5_state["choice"][key] = choice
     return choice
 
