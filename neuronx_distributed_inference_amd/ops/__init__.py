@@ -311,7 +311,7 @@ def _sg5_autotune(key, x2, weight, bias) -> int:
     # error (DVFS + cache state differ): ship the hand-written kernel
     # only on a CLEAR win, else keep the library path (measured: near
     # parity flips either way run to run)
-    if choice != 0 and results[choice] > 0.90 * results.get(0, float("inf")):
+    if choice != 0 and results[choice] > 0.95 * results.get(0, float("inf")):
         choice = 0
     _sg5_state["choice"][key] = choice
     return choice
