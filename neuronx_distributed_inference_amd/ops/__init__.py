@@ -661,17 +661,19 @@ def moe_grouped_tables(idx_flat: torch.Tensor, num_experts: int):
 
 
 def moe_grouped_gemm(x_sorted: torch.Tensor, w: torch.Tensor,
-                     expert_ids: torch.Tensor) -> torch.Tensor:
-    """y_sorted[m] = x_sorted[m] @ w[expert_of(m)]^T. x_sorted (Mpad, K),
-    w (E, N, K), expert_ids (Mpad/64,) int32. GPU: one grouped HIP
-    kernel; CPU reference: per-tile loop."""
+                     expert_ids: torch.Tensor, bias=None) -> torch.Tensor:
+    """y_sorted[m] = x_sorted[m] @ w[expert_of(m)]^T (+ bias[expert]).
+    x_sorted (Mpad, K), w (E, N, K), expert_ids (Mpad/64,) int32, bias
+    (E, N) or None. GPU: one grouped HIP kernel; CPU reference:
+    per-tile loop."""
     Mpad, K = x_sorted.shape
     N = w.shape[1]
     if x_sorted.is_cuda:
         c = _require_ext("moe_grouped_gemm")
         y = torch.empty(Mpad, N, dtype=x_sorted.dtype, device=x_sorted.device)
         c.moe_grouped_gemm(y, x_sorted.contiguous(), w.contiguous(),
-                           expert_ids.contiguous())
+                           expert_ids.contiguous(),
+                           None if bias is None else bias.contiguous())
         return y
     y = torch.zeros(Mpad, N, dtype=x_sorted.dtype)
     for t in range(Mpad // MOE_TILE):
@@ -679,7 +681,10 @@ def moe_grouped_gemm(x_sorted: torch.Tensor, w: torch.Tensor,
         if e < 0:
             continue
         rows = slice(t * MOE_TILE, (t + 1) * MOE_TILE)
-        y[rows] = x_sorted[rows] @ w[e].t().to(x_sorted.dtype)
+        r = x_sorted[rows] @ w[e].t().to(x_sorted.dtype)
+        if bias is not None:
+            r = r + bias[e].to(r.dtype)
+        y[rows] = r
     return y
 
 

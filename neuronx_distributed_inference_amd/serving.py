@@ -262,11 +262,6 @@ class ServingEngine:
             pos[i, 0] = r.pos
             seq_ids[i] = slot
         slot_idx = torch.tensor(slots, dtype=torch.long, device=self.device)
-        ids = self._slot_tok.index_select(0, slot_idx).view(-1, 1)
-        if ids.shape[0] < B:
-            ids = torch.cat([ids, torch.zeros(B - ids.shape[0], 1,
-                                              dtype=torch.long,
-                                              device=self.device)], dim=0)
         reqs = [self.active[s] for s in slots]
         self._set_adapters(reqs)
         kw = {}
@@ -280,6 +275,14 @@ class ServingEngine:
         sp = self._params_for(reqs)
 
         def run():
+            # the _slot_tok gather MUST happen on the executor stream:
+            # the previous step's scatter into it ran there, and the two
+            # are ordered only within that stream
+            ids = self._slot_tok.index_select(0, slot_idx).view(-1, 1)
+            if ids.shape[0] < B:
+                ids = torch.cat([ids, torch.zeros(B - ids.shape[0], 1,
+                                                  dtype=torch.long,
+                                                  device=self.device)], dim=0)
             if self._capture_decode and self.paged:
                 toks = self._run_decode_graph(ids, pos, seq_ids, sp,
                                               kw["block_table"])
