@@ -138,7 +138,11 @@ def sdpa(
     scores = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale
     if mask is not None:
         if mask.dtype == torch.bool:
-            scores = scores.masked_fill(~mask, torch.finfo(torch.float32).min)
+            # -inf (not finfo.min): a FULLY-masked row (pad query) must
+            # produce zero output, not a uniform average (softmax over a
+            # constant-min row is uniform — the kernel/ref divergence the
+            # ragged ring test caught)
+            scores = scores.masked_fill(~mask, float("-inf"))
         else:
             scores = scores + mask.float()
     if sinks is not None:
@@ -147,6 +151,7 @@ def sdpa(
         probs = torch.softmax(scores, dim=-1)[..., :-1]
     else:
         probs = torch.softmax(scores, dim=-1)
+    probs = torch.nan_to_num(probs)  # fully-masked rows -> 0
     out = torch.matmul(probs, v.float())
     return out.to(q.dtype)
 

@@ -57,7 +57,8 @@ class ServingEngine:
                  eos_token_id: Optional[int] = None,
                  capture_decode: Optional[bool] = None,
                  on_token=None, lora_state=None, bucket_chunks: bool = True,
-                 async_depth: Optional[int] = None):
+                 async_depth: Optional[int] = None,
+                 mixed_chunking: bool = False):
         self.model = causal_lm
         ec = causal_lm.config.neuron_config
         self.batch = ec.batch_size
@@ -115,6 +116,13 @@ class ServingEngine:
         self._slot_tok = torch.zeros(self.max_slots, dtype=torch.long,
                                      device=self.device)
         self._inflight: Deque = collections.deque()  # (slots, AsyncResult)
+        # mixed chunked prefill (ref chunked-prefill tile buckets,
+        # autobucketing.py:101-146 / kvcache/utils.py:209-506): one
+        # multi-token TKG step carries a prefill CHUNK row alongside the
+        # decoding rows (their single token + -1 pads), so decode latency
+        # never stalls behind a prefill chunk. Greedy engines only (the
+        # mixed pass samples per position greedily).
+        self.mixed_chunking = bool(mixed_chunking) and chunk_size is not None
 
     # ------------------------------------------------------------------
     def add_request(self, prompt_ids: torch.Tensor,
@@ -365,7 +373,9 @@ class ServingEngine:
 
     # ------------------------------------------------------------------
     def step(self):
-        """One scheduling step: admit + prefill one chunk, or decode."""
+        """One scheduling step: admit + prefill one chunk, or decode.
+        With `mixed_chunking`, a prefill chunk rides INSIDE the decode
+        step (one batch: chunk row + decode rows) instead of stalling it."""
         # admit (deferred decode results must land first: a finished slot
         # may be freed and re-assigned, and pending tokens reference it)
         if self.queue:
@@ -380,10 +390,76 @@ class ServingEngine:
             r = self.active[slot]
             if r.prefill_done < r.prompt.shape[0]:
                 self._drain_inflight(0)
-                self._prefill_chunk(r)
+                decoding = [s for s in sorted(self.active)
+                            if self.active[s] is not r
+                            and self.active[s].prefill_done
+                            >= self.active[s].prompt.shape[0]]
+                if self.mixed_chunking and decoding and not self.paged:
+                    self._mixed_step(r, decoding)
+                else:
+                    self._prefill_chunk(r)
                 return
         if self.active:
             self._decode_step()
+
+    def _mixed_step(self, pre: Request, decode_slots):
+        """One multi-token TKG pass: row 0 carries `pre`'s next prefill
+        chunk, the other rows carry one decode token + -1 pads (the
+        pad-skip invariant keeps pads out of the KV cache and attention).
+        Decode rows emit their token from column 0; the chunk row emits
+        only when it completes the prompt."""
+        S = pre.prompt.shape[0]
+        lo = pre.prefill_done
+        hi = S if self.chunk_size is None else min(S, lo + self.chunk_size)
+        L = hi - lo
+        B = self.batch
+        ids = torch.zeros(B, L, dtype=torch.long, device=self.device)
+        pos = torch.full((B, L), -1, dtype=torch.long, device=self.device)
+        seq_ids = torch.full((B,), -1, dtype=torch.long, device=self.device)
+        ids[0, :L] = pre.prompt[lo:hi].to(self.device)
+        pos[0, :L] = torch.arange(lo, hi, device=self.device)
+        seq_ids[0] = pre.slot
+        reqs = [pre] + [self.active[s] for s in decode_slots]
+        for i, slot in enumerate(decode_slots, start=1):
+            if i >= B:
+                decode_slots = decode_slots[: i - 1]
+                reqs = reqs[: i]
+                break
+            r = self.active[slot]
+            ids[i, 0] = r.last_token
+            pos[i, 0] = r.pos
+            seq_ids[i] = slot
+        self._set_adapters(reqs)
+        out = self.model(ids, None, pos, seq_ids, None,
+                         phase="token_generation")
+        toks = out.logits if out.logits is not None else out.tokens
+        if toks.dim() > 2:
+            toks = toks.argmax(-1)
+        toks = toks.view(B, -1)
+        pre.prefill_done = hi
+        host = toks.cpu()
+        if hi == S:  # chunk completes the prompt: emit its first token
+            tok = int(host[0, L - 1])
+            pre.last_token = tok
+            self._slot_tok[pre.slot] = tok
+            pre.pos = S
+            pre.output.append(tok)
+            pre.t_first_token = time.monotonic()
+            if self.on_token is not None:
+                self.on_token(pre.req_id, tok)
+            self._maybe_finish(pre, tok)
+        for i, slot in enumerate(decode_slots, start=1):
+            r = self.active.get(slot)
+            if r is None or r.done:
+                continue
+            tok = int(host[i, 0])
+            r.last_token = tok
+            self._slot_tok[slot] = tok
+            r.pos += 1
+            r.output.append(tok)
+            if self.on_token is not None:
+                self.on_token(r.req_id, tok)
+            self._maybe_finish(r, tok)
 
     def run(self, max_steps: int = 100000) -> Dict[int, List[int]]:
         steps = 0

@@ -333,3 +333,31 @@ def test_serving_rejects_sampling_on_static_greedy():
     # greedy requests still fine
     rid = eng.add_request(torch.randint(0, 128, (6,)), max_new_tokens=2)
     assert len(eng.run()[rid]) == 2
+
+
+@torch.inference_mode()
+def test_serving_mixed_chunking_matches_reference():
+    """Mixed chunked prefill (prefill chunk + decode rows in ONE
+    multi-token TKG batch — ref chunked-prefill tile buckets,
+    autobucketing.py:101-146) must reproduce per-request reference
+    generations, and decode steps must not stall behind chunks."""
+    m = _model(batch=3)
+    prompts = [torch.randint(0, 128, (24,), generator=torch.Generator().manual_seed(i))
+               for i in range(3)]
+    want = {}
+    for i, p in enumerate(prompts):
+        ref = HuggingFaceGenerationAdapter(m)
+        out = ref.generate(p.view(1, -1),
+                           generation_config=GenerationConfig(max_new_tokens=6))
+        want[i] = out[0, 24:].tolist()
+        m.reset()
+    eng = ServingEngine(m, chunk_size=8, mixed_chunking=True)
+    rids = {}
+    # stagger arrivals so chunks overlap decodes
+    rids[0] = eng.add_request(prompts[0], max_new_tokens=6)
+    eng.step(); eng.step(); eng.step()  # prefill req 0 fully
+    rids[1] = eng.add_request(prompts[1], max_new_tokens=6)
+    rids[2] = eng.add_request(prompts[2], max_new_tokens=6)
+    res = eng.run()
+    for i in range(3):
+        assert res[rids[i]] == want[i], (i, res[rids[i]], want[i])
