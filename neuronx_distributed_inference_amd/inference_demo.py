@@ -118,6 +118,12 @@ def setup_run_parser(run: argparse.ArgumentParser):
     run.add_argument("--check-accuracy-mode", default="skip-accuracy-check",
                      choices=CHECK_ACCURACY_MODES)
     run.add_argument("--divergence-difference-tol", type=float, default=0.001)
+    run.add_argument("--capture-inputs-on-divergence", action="store_true",
+                     help="on logit divergence, re-run with tensor capture "
+                          "on every decoder layer and dump the tensors at "
+                          "the divergence step (ref inference_demo.py:616-649)")
+    run.add_argument("--capture-modules", nargs="+", default=None,
+                     help="module names to capture (default: every layer)")
     run.add_argument("--benchmark", action="store_true")
     run.add_argument("--num-benchmark-runs", type=int, default=5)
     run.add_argument("--skip-compile", action="store_true")
@@ -321,18 +327,76 @@ def run_inference(args, model_type: str) -> int:
 
 def run_accuracy_check(args, model, config, input_ids, attention_mask, gc) -> None:
     """Token/logit matching vs the HF CPU implementation of the same
-    checkpoint (ref run_accuracy_check, inference_demo.py:710)."""
+    checkpoint, with optional tensor capture at the divergence step
+    (ref run_accuracy_check :710 + capture-on-divergence :616-649)."""
     from transformers import AutoModelForCausalLM
 
-    from neuronx_distributed_inference_amd.utils.accuracy import check_accuracy
+    from neuronx_distributed_inference_amd.utils.accuracy import (
+        LogitMatchingValidationError,
+        check_accuracy,
+        check_accuracy_logits,
+    )
 
     hf = AutoModelForCausalLM.from_pretrained(args.model_path, torch_dtype=torch.float32)
     hf.eval()
-    check_accuracy(
-        model, input_ids, attention_mask=attention_mask, hf_model=hf,
-        generation_config=gc,
-    )
-    print("accuracy check passed (token matching)")
+    if args.check_accuracy_mode == "token-matching":
+        check_accuracy(
+            model, input_ids, attention_mask=attention_mask, hf_model=hf,
+            generation_config=gc,
+        )
+        print("accuracy check passed (token matching)")
+        return
+    # logit matching: HF greedy loop produces per-position golden logits
+    golden = compute_hf_golden_logits(hf, input_ids, attention_mask,
+                                      gc.max_new_tokens)
+    tol = args.divergence_difference_tol
+    try:
+        check_accuracy_logits(model, input_ids, golden,
+                              attention_mask=attention_mask,
+                              atol=tol, rtol=tol)
+        print("accuracy check passed (logit matching)")
+    except LogitMatchingValidationError as e:
+        if args.capture_inputs_on_divergence:
+            from neuronx_distributed_inference_amd.utils.tensor_capture import (
+                TensorCapture,
+            )
+
+            names = args.capture_modules or [
+                n for n, _ in model.model.named_modules()
+                if n.startswith("layers.") and n.count(".") == 1
+            ]
+            model.reset()
+            with TensorCapture(model.model, names,
+                               max_steps=e.divergence_index + 1) as cap:
+                try:
+                    check_accuracy_logits(model, input_ids, golden,
+                                          attention_mask=attention_mask,
+                                          atol=tol, rtol=tol)
+                except LogitMatchingValidationError:
+                    pass
+            dump = os.path.join(args.compiled_model_path,
+                                f"divergence_capture_idx{e.divergence_index}.pt")
+            cap.save(dump)
+            logger.error("logit divergence at %d; captured %d modules -> %s",
+                         e.divergence_index, len(names), dump)
+        raise
+
+
+@torch.inference_mode()
+def compute_hf_golden_logits(hf, input_ids, attention_mask, n_positions: int):
+    """HF greedy loop collecting the last-token logits at every position
+    (the golden tensor for logit matching)."""
+    out = []
+    ids = input_ids.clone()
+    mask = (attention_mask.clone() if attention_mask is not None
+            else torch.ones_like(ids))
+    for _ in range(n_positions):
+        logits = hf(input_ids=ids, attention_mask=mask).logits[:, -1, :].float()
+        out.append(logits)
+        tok = logits.argmax(-1, keepdim=True)
+        ids = torch.cat([ids, tok], dim=-1)
+        mask = torch.cat([mask, torch.ones_like(tok)], dim=-1)
+    return torch.stack(out)
 
 
 def main(argv=None) -> int:

@@ -645,7 +645,10 @@ def moe_grouped_tables(idx_flat: torch.Tensor, num_experts: int):
     Mpad = ((Ttk + E * (MT - 1)) + MT - 1) // MT * MT
     order = torch.argsort(idx_flat, stable=True)
     sorted_e = idx_flat[order]
-    counts = torch.bincount(idx_flat, minlength=E)
+    # scatter_add, NOT bincount: bincount syncs the host on CUDA (output
+    # sizing), which kills hipGraph capture
+    counts = torch.zeros(E, dtype=torch.long, device=dev).scatter_add_(
+        0, idx_flat.long(), torch.ones_like(idx_flat, dtype=torch.long))
     padded = (counts + MT - 1) // MT * MT
     ends_pad = padded.cumsum(0)                 # per-expert padded end
     starts_pad = ends_pad - padded
