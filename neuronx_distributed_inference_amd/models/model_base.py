@@ -277,6 +277,11 @@ class DecoderModel(nn.Module):
 
         # kept for medusa/eagle heads (graph-safe: a static tensor each capture)
         self.last_hidden = h_last
+        # EAGLE draft prefill needs the FULL prompt hidden states (ref
+        # EAGLE CTE, model_base.py:1931-2093); opt-in to avoid keeping
+        # (B, S, H) alive on every CTE
+        if getattr(self, "eagle_capture_hidden", False):
+            self.full_hidden = h
         logits_shard = self.lm_head(h_last)  # (B, 1|S, V/tp)
 
         # rejection-sampling speculation reads the per-position top-Kg

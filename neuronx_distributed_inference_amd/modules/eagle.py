@@ -72,6 +72,29 @@ class EagleDraftModel(nn.Module):
         return tokens, h_final
 
 
+
+
+def eagle_draft_prefill(target, draft: "EagleDraftModel", input_ids,
+                        attention_mask, position_ids, seq_ids, next_tok):
+    """EAGLE draft CONTEXT ENCODING (ref EAGLE CTE, model_base.py:
+    1931-2093): feed (token_{i+1}, target_hidden_i) for the whole prompt
+    so the draft's KV cache holds a real prefix before the first
+    speculation round. Requires target.model.eagle_capture_hidden=True
+    during the target prefill (full_hidden kept)."""
+    from neuronx_distributed_inference_amd.modules.attention.attention_base import (
+        PHASE_CTE,
+    )
+
+    h_full = getattr(target.model, "full_hidden", None)
+    if h_full is None:
+        return  # capture flag was off: draft starts cold (still correct)
+    B, S = input_ids.shape
+    # shifted tokens: t_1..t_{S-1}, next_tok; hiddens: h_0..h_{S-1}
+    shifted = torch.cat([input_ids[:, 1:], next_tok.view(B, 1)], dim=1)
+    draft(shifted, h_full, position_ids, seq_ids, PHASE_CTE,
+          attention_mask=attention_mask)
+
+
 class EagleGenerationAdapter:
     """EAGLE chain decoding: draft k tokens conditioned on target hiddens,
     verify with one multi-token target forward, accept the matching prefix."""
@@ -98,17 +121,19 @@ class EagleGenerationAdapter:
         seq_ids = torch.arange(B, device=device)
         cum = attention_mask.long().cumsum(-1) - 1
         position_ids = torch.where(attention_mask.bool(), cum, torch.full_like(cum, -1))
+        self.target.model.eagle_capture_hidden = True
         t_out = self.target(input_ids=input_ids, attention_mask=attention_mask,
                             position_ids=position_ids, seq_ids=seq_ids)
+        self.target.model.eagle_capture_hidden = False
         lengths = attention_mask.sum(-1).long()
         next_tok = t_out.tokens.view(B)
         # target hidden of the LAST prompt token conditions the draft
         h_tgt = self.target.model.last_hidden[:, -1, :]     # (B, H)
 
-        # draft prefill: feed (token_{i+1}, target_hidden_i) for the prompt.
-        # For simplicity (and because only the last position's draft state
-        # matters for chaining) we warm the draft cache with the prompt
-        # tokens shifted by one against their own embeddings.
+        # draft prefill: (token_{i+1}, target_hidden_i) over the prompt
+        # fills the draft's KV cache (ref EAGLE CTE)
+        eagle_draft_prefill(self.target, self.draft, input_ids,
+                            attention_mask, position_ids, seq_ids, next_tok)
         out_rows = [input_ids[i][attention_mask[i].bool()].tolist() for i in range(B)]
         for i in range(B):
             out_rows[i].append(int(next_tok[i]))
@@ -301,11 +326,15 @@ class EagleTreeAdapter:
         seq_ids = torch.arange(B, device=device)
         cum = attention_mask.long().cumsum(-1) - 1
         position_ids = torch.where(attention_mask.bool(), cum, torch.full_like(cum, -1))
+        self.target.model.eagle_capture_hidden = True
         t_out = self.target(input_ids=input_ids, attention_mask=attention_mask,
                             position_ids=position_ids, seq_ids=seq_ids)
+        self.target.model.eagle_capture_hidden = False
         lengths = attention_mask.sum(-1).long()
         next_tok = t_out.tokens.view(B)
         h_tgt = self.target.model.last_hidden[:, -1, :]
+        eagle_draft_prefill(self.target, self.draft, input_ids,
+                            attention_mask, position_ids, seq_ids, next_tok)
 
         out_rows = [input_ids[i][attention_mask[i].bool()].tolist() for i in range(B)]
         for i in range(B):
