@@ -54,6 +54,9 @@ def main():
     torch.manual_seed(0)
     col = ColumnParallelLinear(256, 512, dtype=torch.bfloat16).to(dev)
     row = RowParallelLinear(512, 256, dtype=torch.bfloat16).to(dev)
+    with torch.no_grad():  # parallel layers allocate torch.empty weights
+        col.weight.normal_(0, 0.05)
+        row.weight.normal_(0, 0.05)
     xin = torch.randn(8, 256, dtype=torch.bfloat16, device=dev)
     want = row(col(xin))
     torch.cuda.synchronize()

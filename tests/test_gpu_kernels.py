@@ -265,6 +265,7 @@ def test_decode_batch_bucket_ladder_gpu():
 
     ids = torch.randint(0, 128, (3, 8), generator=torch.Generator().manual_seed(5))
     ladder = build([2, 4, 8])
+    ladder.warmup()  # creates the submodel wrappers + captures the ladder
     got = HuggingFaceGenerationAdapter(ladder).generate(
         ids.to(_dev()), generation_config=GenerationConfig(max_new_tokens=8))
     # B=3 must have used the 4-rung (not the full batch 8)
@@ -272,6 +273,7 @@ def test_decode_batch_bucket_ladder_gpu():
     assert tkg._graphs is not None and (4, 1) in tkg._graphs.runners
     single = build(None)
     single.config.neuron_config.enable_batch_bucketing = False
+    single.warmup()
     want = HuggingFaceGenerationAdapter(single).generate(
         ids.to(_dev()), generation_config=GenerationConfig(max_new_tokens=8))
     assert torch.equal(got.cpu(), want.cpu())
