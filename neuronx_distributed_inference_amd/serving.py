@@ -262,7 +262,11 @@ class ServingEngine:
         the next launch (async double-buffering — ref
         modules/async_execution.py, model_base.py:3737)."""
         slots = sorted(self.active)
-        B = self.batch
+        # linear path: the batch stays RAGGED (len(slots) rows) so the
+        # ModelWrapper pads only to the smallest captured batch rung
+        # (partial occupancy pays sub-bucket cost). The paged captured
+        # graph has a fixed batch: pad to it there.
+        B = self.batch if (self._capture_decode and self.paged) else len(slots)
         pos = torch.full((B, 1), -1, dtype=torch.long, device=self.device)
         seq_ids = torch.full((B,), -1, dtype=torch.long, device=self.device)
         for i, slot in enumerate(slots):
