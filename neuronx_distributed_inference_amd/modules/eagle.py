@@ -209,27 +209,20 @@ class EagleTreeSpeculationModel(nn.Module):
         self.target = target_model  # DecoderModel
         self.draft = draft
         self.tree = tree
-        N = tree.num_nodes
-        anc = torch.zeros(N, N, dtype=torch.bool)
-        for n in range(N):
-            for a in tree.ancestors(n):
-                anc[n, a] = True
-        self.register_buffer("anc", anc, persistent=False)
+        from neuronx_distributed_inference_amd.modules.token_tree import (
+            ancestor_matrix,
+        )
+
+        self.register_buffer("anc", ancestor_matrix(tree), persistent=False)
         self.register_buffer("depth_t", torch.tensor(tree.depth), persistent=False)
 
     def _mask(self, nodes, lengths, Smax, device):
-        """(B, 1, len(nodes), Smax) bool: prior + ancestors of each node."""
-        B = lengths.shape[0]
-        N = self.tree.num_nodes
-        cols = torch.arange(Smax, device=device)
-        rel = cols.view(1, -1) - lengths.view(B, 1)            # (B, Smax)
-        in_tree = (rel >= 0) & (rel < N)
-        relc = rel.clamp(0, N - 1)
-        anc_sel = self.anc[nodes]                              # (w, N)
-        tree_m = anc_sel[:, relc].permute(1, 0, 2)             # (B, w, Smax)
-        tree_m = tree_m & in_tree.unsqueeze(1)
-        prior_m = (rel < 0).unsqueeze(1).expand(-1, len(nodes), -1)
-        return (prior_m | tree_m).unsqueeze(1)
+        from neuronx_distributed_inference_amd.modules.token_tree import (
+            tree_attention_mask_batch,
+        )
+
+        return tree_attention_mask_batch(self.tree, self.anc, nodes, lengths,
+                                         Smax, device)
 
     @torch.inference_mode()
     def forward(self, input_ids, prev_hidden, lengths, seq_ids):

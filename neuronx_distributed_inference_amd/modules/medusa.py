@@ -128,3 +128,159 @@ class MedusaGenerationAdapter:
         for i, r in enumerate(out_rows):
             out[i, : len(r)] = torch.tensor(r, device=device)
         return out
+
+
+class MedusaTreeAdapter:
+    """Medusa TREE decoding (ref medusa tree attention: medusa_mask /
+    accepted-index inputs, modeling_llama.py:1059-1187 + KV compaction
+    kv_cache_manager.py:265-321): each head proposes top-k candidates for
+    its position; the static tree's depth-d nodes draw from head d-1's
+    candidate list by child rank; the target verifies ALL nodes in one
+    masked pass (the same tree-verify machinery as EAGLE tree TKG) and
+    acceptance walks the deepest matching path."""
+
+    def __init__(self, model, medusa_heads: MedusaHeads, tree):
+        from neuronx_distributed_inference_amd.modules.token_tree import (
+            ancestor_matrix,
+        )
+
+        self.model = model
+        self.heads = medusa_heads
+        self.tree = tree
+        assert len(tree.branching) <= medusa_heads.num_heads, \
+            "tree depth exceeds medusa head count"
+        self.anc = ancestor_matrix(tree)
+        # child rank within its parent (which top-k slot feeds the node)
+        self.child_rank = [0] * tree.num_nodes
+        seen = {}
+        for n in range(1, tree.num_nodes):
+            p = tree.parent[n]
+            self.child_rank[n] = seen.get(p, 0)
+            seen[p] = self.child_rank[n] + 1
+
+    def _head_topk(self, hidden: torch.Tensor) -> torch.Tensor:
+        """(B, depth, kmax) per-head top-k candidate tokens."""
+        from neuronx_distributed_inference_amd.parallel.ops import distributed_topk
+
+        kmax = max(self.tree.branching)
+        outs = []
+        for d in range(len(self.tree.branching)):
+            blk, head = self.heads.blocks[d], self.heads.heads[d]
+            logits = head(blk(hidden))
+            _, ids = distributed_topk(logits.unsqueeze(1), kmax, dim=-1)
+            outs.append(ids.view(-1, 1, kmax))
+        return torch.cat(outs, dim=1)
+
+    @torch.inference_mode()
+    def _tree_verify(self, node_tokens, lengths, seq_ids):
+        """Target verifies all tree nodes in one masked multi-token pass
+        (rows = L + node index, rope = L + depth)."""
+        from neuronx_distributed_inference_amd.modules.attention.attention_base import (
+            PHASE_TKG,
+        )
+        from neuronx_distributed_inference_amd.modules.token_tree import (
+            tree_attention_mask_batch,
+        )
+        from neuronx_distributed_inference_amd.parallel.ops import distributed_argmax
+
+        t = self.model.model
+        tree = self.tree
+        B = node_tokens.shape[0]
+        N = tree.num_nodes
+        device = node_tokens.device
+        Smax = t.kv_mgr.max_len
+        rows = lengths.view(B, 1) + torch.arange(N, device=device).view(1, N)
+        rope = lengths.view(B, 1) + torch.tensor(tree.depth, device=device).view(1, N)
+        mask = tree_attention_mask_batch(tree, self.anc, list(range(N)),
+                                         lengths, Smax, device)
+        emb = t.embed_tokens(node_tokens)
+        cos, sin = t.rotary.get_cos_sin(rope, device=device)
+        h = emb
+        residual = None
+        for layer in t.layers:
+            h, residual = layer(h, residual, cos, sin, t.kv_mgr, rows,
+                                PHASE_TKG, mask, seq_ids)
+        h_final = t.norm(h, residual)[0] if residual is not None else t.norm(h)
+        logits = t.lm_head(h_final)
+        return distributed_argmax(logits, dim=-1), h_final
+
+    @torch.inference_mode()
+    def generate(self, input_ids, attention_mask=None, generation_config=None, **kw):
+        gc = generation_config or GenerationConfig(**{
+            k: v for k, v in kw.items() if k in GenerationConfig.__dataclass_fields__})
+        B, S0 = input_ids.shape
+        device = input_ids.device
+        tree = self.tree
+        N = tree.num_nodes
+        if attention_mask is None:
+            attention_mask = torch.ones_like(input_ids)
+        max_len = gc.max_length or min(self.model.ec.max_length, S0 + gc.max_new_tokens)
+        max_len = min(max_len, self.model.ec.max_length - N - 2)
+
+        self.model.reset()
+        seq_ids = torch.arange(B, device=device)
+        cum = attention_mask.long().cumsum(-1) - 1
+        position_ids = torch.where(attention_mask.bool(), cum, torch.full_like(cum, -1))
+        t_out = self.model(input_ids=input_ids, attention_mask=attention_mask,
+                           position_ids=position_ids, seq_ids=seq_ids)
+        lengths = attention_mask.sum(-1).long()
+        next_tok = t_out.tokens.view(B)
+        hidden = self.model.model.last_hidden[:, -1, :]
+
+        out_rows = [input_ids[i][attention_mask[i].bool()].tolist() for i in range(B)]
+        for i in range(B):
+            out_rows[i].append(int(next_tok[i]))
+        finished = torch.zeros(B, dtype=torch.bool, device=device)
+
+        while True:
+            if gc.eos_token_id is not None:
+                finished |= next_tok == gc.eos_token_id
+            if bool(finished.all()) or max(len(r) for r in out_rows) >= max_len:
+                break
+            cand = self._head_topk(hidden)              # (B, depth, kmax)
+            node_tokens = torch.zeros(B, N, dtype=torch.long, device=device)
+            node_tokens[:, 0] = next_tok
+            for n in range(1, N):
+                d = tree.depth[n] - 1                   # head index
+                node_tokens[:, n] = cand[:, d, self.child_rank[n]]
+            tgt_toks, h_all = self._tree_verify(node_tokens, lengths, seq_ids)
+            nt, tt = node_tokens.cpu(), tgt_toks.cpu()
+            max_m = len(tree.branching)
+            src = torch.zeros(B, max_m, dtype=torch.long)
+            dst = torch.zeros(B, max_m, dtype=torch.long)
+            bonus_nodes = torch.zeros(B, dtype=torch.long)
+            n_acc = torch.zeros(B, dtype=torch.long)
+            for b in range(B):
+                path, bonus = tree.accept(nt[b], tt[b])
+                L = int(lengths[b])
+                for j in range(max_m):
+                    src[b, j] = L + path[j] if j < len(path) else L
+                    dst[b, j] = L + j + 1 if j < len(path) else L
+                bonus_nodes[b] = bonus
+                n_acc[b] = len(path)
+                if finished[b]:
+                    n_acc[b] = 0
+                    continue
+                row = out_rows[b]
+                emitted = [int(nt[b, p]) for p in path] + [int(tt[b, bonus])]
+                for tok in emitted:
+                    if len(row) >= max_len:
+                        break
+                    row.append(tok)
+                    if gc.eos_token_id is not None and tok == gc.eos_token_id:
+                        finished[b] = True
+                        break
+            self.model.model.kv_mgr.compact_rows(
+                seq_ids, src.to(device), dst.to(device))
+            hidden = h_all.gather(
+                1, bonus_nodes.to(device).view(B, 1, 1).expand(
+                    -1, 1, h_all.shape[-1])).squeeze(1)
+            lengths = lengths + n_acc.to(device) + 1
+            next_tok = torch.tensor([r[-1] for r in out_rows], device=device,
+                                    dtype=torch.long)
+
+        longest = max(len(r) for r in out_rows)
+        out = torch.full((B, longest), gc.pad_token_id, dtype=torch.long, device=device)
+        for i, r in enumerate(out_rows):
+            out[i, : len(r)] = torch.tensor(r, device=device)
+        return out

@@ -110,3 +110,33 @@ class TokenTree:
                 best = acc
                 best_node = node
         return best, best_node
+
+
+def ancestor_matrix(tree: "TokenTree") -> torch.Tensor:
+    """(N, N) bool: anc[n, a] = a is on n's root path (incl. n)."""
+    N = tree.num_nodes
+    anc = torch.zeros(N, N, dtype=torch.bool)
+    for n in range(N):
+        for a in tree.ancestors(n):
+            anc[n, a] = True
+    return anc
+
+
+def tree_attention_mask_batch(tree: "TokenTree", anc: torch.Tensor,
+                              nodes, lengths: torch.Tensor, Smax: int,
+                              device) -> torch.Tensor:
+    """(B, 1, len(nodes), Smax) bool tree mask with PER-ROW prior
+    lengths: node n attends cache cols [0, lengths_b) plus its ancestor
+    rows at lengths_b + ancestor_idx (the batched form of
+    TokenTree.attention_mask; used by EAGLE tree TKG and Medusa tree
+    verification)."""
+    B = lengths.shape[0]
+    N = tree.num_nodes
+    cols = torch.arange(Smax, device=device)
+    rel = cols.view(1, -1) - lengths.view(B, 1)
+    in_tree = (rel >= 0) & (rel < N)
+    relc = rel.clamp(0, N - 1)
+    anc_sel = anc.to(device)[list(nodes)]
+    tree_m = anc_sel[:, relc].permute(1, 0, 2) & in_tree.unsqueeze(1)
+    prior_m = (rel < 0).unsqueeze(1).expand(-1, len(nodes), -1)
+    return (prior_m | tree_m).unsqueeze(1)

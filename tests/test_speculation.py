@@ -407,3 +407,33 @@ def test_eagle_tree_identical_draft_accepts_deep_paths():
         assert all(tree.parent[p] in ([0] + path) or tree.parent[p] == 0
                    for p in path[:1])
         assert 0 <= bonus < tree.num_nodes
+
+
+def test_medusa_tree_matches_target_only():
+    """Medusa TREE decoding (per-head top-k tree + masked tree verify +
+    KV compaction) must equal target-only greedy output."""
+    from neuronx_distributed_inference_amd.modules.medusa import (
+        MedusaHeads,
+        MedusaTreeAdapter,
+    )
+    from neuronx_distributed_inference_amd.modules.token_tree import TokenTree
+
+    target = _build(64, 2, seed=1, spec_len=3)
+    torch.manual_seed(9)
+    heads = MedusaHeads(hidden_size=64, vocab_size=64, num_heads=2)
+    with torch.no_grad():
+        for p in heads.parameters():
+            p.normal_(0, 0.05)
+    tree = TokenTree([3, 2])
+    ids = torch.randint(0, 64, (2, 8))
+    want = HuggingFaceGenerationAdapter(_build(64, 2, seed=1)).generate(
+        ids, generation_config=GenerationConfig(max_new_tokens=18))
+    got = MedusaTreeAdapter(target, heads, tree).generate(
+        ids, generation_config=GenerationConfig(max_new_tokens=18))
+    for i in range(2):
+        row = got[i]
+        n = int((row != 0).float().flip(0).argmax())
+        n = row.shape[0] - n
+        n = min(n, want.shape[1])
+        assert n >= ids.shape[1] + 8
+        assert torch.equal(row[:n], want[i, :n]), (row.tolist(), want[i].tolist())
