@@ -108,8 +108,18 @@ class KVCacheManager(nn.Module):
         return self.ring_layers[layer_idx]
 
     def wrap_positions(self, layer_idx: int, positions: torch.Tensor) -> torch.Tensor:
+        """Ring (compact SWA) write positions: slot = pos % size, and any
+        position whose slot is ALSO claimed by a newer position in the
+        SAME batch is dropped (-1). Without the drop, a prefill longer
+        than the window issues two racing writes to one slot in a single
+        scatter — the GPU writer's order is nondeterministic (caught by
+        the mistral ring e2e: ~0.9 token agreement, serialized runs
+        clean), while the CPU reference writes sequentially."""
         sz = self.layer_cache_sizes[layer_idx]
-        return torch.where(positions >= 0, positions % sz,
+        pmax = positions.amax(dim=-1, keepdim=True)
+        newest = positions > pmax - sz
+        valid = (positions >= 0) & newest
+        return torch.where(valid, positions % sz,
                            torch.full_like(positions, -1))
 
     def compact_rows(self, seq_ids: torch.Tensor, src: torch.Tensor,

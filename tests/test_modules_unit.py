@@ -495,3 +495,28 @@ def test_gpt_oss_mxfp4_experts_match_dequantized():
     # and the grouped CPU path agrees with the per-expert loop
     got_g = mx._forward_grouped(x, wts, idx)
     torch.testing.assert_close(got_g, want, atol=1e-4, rtol=1e-4)
+
+
+def test_ring_wrap_positions_drop_superseded():
+    """Ring cache: positions whose slot is reclaimed by a NEWER position
+    in the same write batch must be dropped (-1) — two racing writes to
+    one slot in a single scatter are order-undefined on the GPU."""
+    import torch
+
+    from neuronx_distributed_inference_amd.modules.kvcache.kv_cache_manager import (
+        KVCacheManager,
+    )
+
+    m = KVCacheManager(num_layers=1, batch_size=2, num_kv_heads=1,
+                       head_dim=8, max_len=16, dtype=torch.float32,
+                       sliding_window=16)
+    pos = torch.tensor([[i for i in range(24)],
+                        [-1] * 4 + [i for i in range(20)]])
+    w = m.wrap_positions(0, pos)
+    # row 0: positions 0..7 collide with 16..23 -> dropped
+    assert (w[0, :8] == -1).all()
+    assert (w[0, 8:] == torch.arange(8, 24) % 16).all()
+    # row 1: pads stay -1; positions 0..3 collide with 16..19
+    assert (w[1, :4] == -1).all()
+    assert (w[1, 4:8] == -1).all()
+    assert (w[1, 8:] == torch.arange(4, 20) % 16).all()
