@@ -114,6 +114,40 @@ def setup_run_parser(run: argparse.ArgumentParser):
     run.add_argument("--pa-block-size", type=int, default=128)
     run.add_argument("--skip-warmup", action="store_true")
     run.add_argument("--on-cpu", action="store_true")
+    # extended parity flags (ref inference_demo.py:73-435 surface)
+    run.add_argument("--async-mode", action="store_true",
+                     help="async double-buffered serving decode")
+    run.add_argument("--output-logits", action="store_true")
+    run.add_argument("--flash-decoding", dest="flash_decoding_enabled",
+                     action="store_true")
+    run.add_argument("--num-cores-per-group", type=int, default=None)
+    run.add_argument("--mlp-cp-degree", type=int, default=1)
+    run.add_argument("--sampling-dp-degree", type=int, default=1)
+    run.add_argument("--moe-tp-degree", type=int, default=None)
+    run.add_argument("--moe-ep-degree", type=int, default=None)
+    run.add_argument("--vocab-parallel", action="store_true")
+    run.add_argument("--compact-swa-kv", action="store_true",
+                     help="ring (compact) KV cache for sliding-window layers")
+    run.add_argument("--is-prefix-caching", action="store_true")
+    run.add_argument("--is-chunked-prefill", action="store_true")
+    run.add_argument("--enable-eagle-speculation", action="store_true")
+    run.add_argument("--num-medusa-heads", type=int, default=0)
+    run.add_argument("--quantization-type", default=None,
+                     choices=[None, "per_tensor_symmetric", "per_channel_symmetric"])
+    run.add_argument("--quantization-dtype", default=None,
+                     choices=[None, "f8e4m3", "int8"])
+    run.add_argument("--rpl-reduce-dtype", default=None)
+    run.add_argument("--attention-dtype", default=None)
+    run.add_argument("--cast-type", default="config", choices=["config", "as-declared"])
+    run.add_argument("--no-batch-bucketing", dest="enable_batch_bucketing",
+                     action="store_false", default=True)
+    run.add_argument("--token-generation-batch-buckets", nargs="+", type=int,
+                     default=None)
+    run.add_argument("--padding-side", default="right", choices=["right", "left"])
+    run.add_argument("--lora-paths", nargs="+", default=None,
+                     help="name=path LoRA adapter dirs to preload")
+    run.add_argument("--max-loras", type=int, default=4)
+    run.add_argument("--max-lora-rank", type=int, default=16)
     # pipeline control
     run.add_argument("--check-accuracy-mode", default="skip-accuracy-check",
                      choices=CHECK_ACCURACY_MODES)
@@ -177,6 +211,33 @@ def create_engine_config(args, model_type: str) -> EngineConfig:
         is_block_kv_layout=args.is_block_kv_layout,
         pa_num_blocks=args.pa_num_blocks,
         pa_block_size=args.pa_block_size,
+        async_mode=getattr(args, "async_mode", False),
+        output_logits=getattr(args, "output_logits", False),
+        flash_decoding_enabled=getattr(args, "flash_decoding_enabled", False),
+        num_cores_per_group=getattr(args, "num_cores_per_group", None),
+        mlp_cp_degree=getattr(args, "mlp_cp_degree", 1),
+        sampling_dp_degree=getattr(args, "sampling_dp_degree", 1),
+        moe_tp_degree=getattr(args, "moe_tp_degree", None),
+        moe_ep_degree=getattr(args, "moe_ep_degree", None),
+        vocab_parallel=getattr(args, "vocab_parallel", False),
+        compact_swa_kv=getattr(args, "compact_swa_kv", False),
+        is_prefix_caching=getattr(args, "is_prefix_caching", False),
+        is_chunked_prefill=getattr(args, "is_chunked_prefill", False),
+        enable_eagle_speculation=getattr(args, "enable_eagle_speculation", False),
+        num_medusa_heads=getattr(args, "num_medusa_heads", 0),
+        is_medusa=getattr(args, "num_medusa_heads", 0) > 0,
+        medusa_speculation_length=getattr(args, "medusa_speculation_length", 0),
+        quantization_type=getattr(args, "quantization_type", None),
+        quantization_dtype=getattr(args, "quantization_dtype", None),
+        cast_type=getattr(args, "cast_type", "config"),
+        enable_batch_bucketing=getattr(args, "enable_batch_bucketing", True),
+        token_generation_batch_buckets=getattr(
+            args, "token_generation_batch_buckets", None),
+        padding_side=getattr(args, "padding_side", "right"),
+        **({"rpl_reduce_dtype": args.rpl_reduce_dtype}
+           if getattr(args, "rpl_reduce_dtype", None) else {}),
+        **({"attention_dtype": args.attention_dtype}
+           if getattr(args, "attention_dtype", None) else {}),
     )
 
 
