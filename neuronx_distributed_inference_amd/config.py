@@ -172,6 +172,9 @@ class EngineConfig:
             "token_generation_batch_buckets", None
         )
         self.enable_batch_bucketing: bool = kwargs.pop("enable_batch_bucketing", True)
+        # bucket selection strategy (ref pad_inputs strategies
+        # model_wrapper.py:730-829): first_fit | second_fit | max
+        self.bucket_strategy: str = kwargs.pop("bucket_strategy", "first_fit")
         self.bucket_n_active_tokens: bool = kwargs.pop("bucket_n_active_tokens", False)
 
         # --- parallelism degrees (ref config.py:360-390) ---

@@ -159,7 +159,9 @@ class ModelWrapper:
         B, S = input_ids.shape
 
         if self.is_cte:
-            bucket = ab.get_target_bucket(S, self.buckets)
+            bucket = ab.get_target_bucket(
+                S, self.buckets,
+                strategy=getattr(self.ec, "bucket_strategy", "first_fit"))
             input_ids = pad_dim(input_ids, 1, bucket, 0)
             attention_mask = pad_dim(attention_mask, 1, bucket, 0)
             position_ids = pad_dim(position_ids, 1, bucket, -1)

@@ -87,8 +87,19 @@ def generate_batch_buckets(
 
 
 def get_target_bucket(length: int, buckets: List[int], strategy: str = "first_fit") -> int:
-    """Smallest bucket >= length (ref model_wrapper.get_target_bucket :831)."""
-    for b in sorted(buckets):
+    """Bucket selection (ref model_wrapper.get_target_bucket :831 +
+    pad_inputs strategies :730-829):
+      first_fit  — smallest bucket >= length (default)
+      second_fit — one rung above first fit when it exists (headroom for
+                   in-place growth before the next re-bucket)
+      max        — always the largest bucket
+    """
+    bs = sorted(buckets)
+    if strategy == "max":
+        return bs[-1]
+    for i, b in enumerate(bs):
         if b >= length:
+            if strategy == "second_fit" and i + 1 < len(bs):
+                return bs[i + 1]
             return b
     raise ValueError(f"length {length} exceeds largest bucket {max(buckets)}")

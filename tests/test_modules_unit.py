@@ -520,3 +520,15 @@ def test_ring_wrap_positions_drop_superseded():
     assert (w[1, :4] == -1).all()
     assert (w[1, 4:8] == -1).all()
     assert (w[1, 8:] == torch.arange(4, 20) % 16).all()
+
+
+def test_bucket_strategies():
+    from neuronx_distributed_inference_amd.modules.autobucketing import (
+        get_target_bucket,
+    )
+
+    bs = [128, 256, 512, 1024]
+    assert get_target_bucket(200, bs) == 256
+    assert get_target_bucket(200, bs, "second_fit") == 512
+    assert get_target_bucket(1000, bs, "second_fit") == 1024
+    assert get_target_bucket(5, bs, "max") == 1024
