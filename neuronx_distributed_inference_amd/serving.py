@@ -163,7 +163,8 @@ class ServingEngine:
     def _params_for(self, reqs) -> Optional[torch.Tensor]:
         if self.model.config.neuron_config.on_device_sampling_config is None:
             return None
-        B = self.batch
+        # sized to the live requests: the wrapper pads to its batch rung
+        B = max(len(reqs), 1)
         p = torch.ones(B, 3, dtype=torch.float32, device=self.device)
         for i, r in enumerate(reqs):
             p[i, 0] = r.top_k
@@ -285,6 +286,11 @@ class ServingEngine:
             full[: len(reqs)] = self._block_table(reqs)
             kw["block_table"] = full
         sp = self._params_for(reqs)
+        if sp is not None and sp.shape[0] < B:
+            # paged captured graph has a fixed batch: pad the live rows
+            sp = torch.cat([sp, torch.ones(B - sp.shape[0], 3,
+                                           dtype=sp.dtype,
+                                           device=sp.device)], dim=0)
 
         def run():
             # the _slot_tok gather MUST happen on the executor stream:
