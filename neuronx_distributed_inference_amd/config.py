@@ -123,6 +123,41 @@ class FusedSpecConfig:
         )
 
 
+@dataclass
+class TensorCaptureConfig:
+    """Capture chosen modules' outputs during forward for divergence
+    debugging (ref TensorCaptureConfig, config.py:1121-1169; collection
+    model_base.py:1026-1149). Wired by CausalLMBase.attach_tensor_capture
+    and the CLI's --capture-inputs-on-divergence flow."""
+
+    module_names: List[str] = field(default_factory=list)
+    max_steps: int = 16
+
+    def to_dict(self):
+        return dict(self.__dict__)
+
+    @classmethod
+    def from_dict(cls, d):
+        return cls(**{k: v for k, v in d.items() if k in cls.__dataclass_fields__})
+
+
+@dataclass
+class TensorReplacementConfig:
+    """Feed golden tensors into chosen modules to isolate the diverging
+    submodule (ref TensorReplacementConfig, config.py:1172-1202;
+    utils/tensor_replacement/registry.py). `replacements` maps module
+    name -> path of a torch-saved tensor fed as the module's input."""
+
+    replacements: Dict[str, str] = field(default_factory=dict)
+
+    def to_dict(self):
+        return dict(self.__dict__)
+
+    @classmethod
+    def from_dict(cls, d):
+        return cls(**{k: v for k, v in d.items() if k in cls.__dataclass_fields__})
+
+
 class EngineConfig:
     """The single engine config (analog of the reference's NeuronConfig god
     object, config.py:92-643). Unknown kwargs warn rather than raise, like
@@ -214,6 +249,14 @@ class EngineConfig:
                 self.chunked_prefill_config
             )
         self.kv_cache_quant: bool = kwargs.pop("kv_cache_quant", False)
+        tc = kwargs.pop("tensor_capture_config", None)
+        if isinstance(tc, dict):
+            tc = TensorCaptureConfig.from_dict(tc)
+        self.tensor_capture_config: Optional[TensorCaptureConfig] = tc
+        tr = kwargs.pop("tensor_replacement_config", None)
+        if isinstance(tr, dict):
+            tr = TensorReplacementConfig.from_dict(tr)
+        self.tensor_replacement_config: Optional[TensorReplacementConfig] = tr
         # GPT-OSS MXFP4 MoE weights kept packed (e2m1 + e8m0), dequantized
         # in-kernel (ref models/gpt_oss/mx_layout_transform.py)
         self.moe_mxfp4: bool = kwargs.pop("moe_mxfp4", False)

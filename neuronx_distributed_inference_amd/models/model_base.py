@@ -542,6 +542,34 @@ class CausalLMBase(nn.Module):
         self.kv_cache_populated = True
         return ModelOutput(tokens=tokens, logits=None)
 
+    def attach_tensor_capture(self):
+        """Context manager capturing the configured modules' outputs per
+        step (ref TensorCaptureConfig config.py:1121-1169 + collection
+        model_base.py:1026-1149). Uses the engine config's
+        tensor_capture_config; defaults to every decoder layer."""
+        from neuronx_distributed_inference_amd.utils.tensor_capture import (
+            TensorCapture,
+        )
+
+        cfg = getattr(self.ec, "tensor_capture_config", None)
+        names = (cfg.module_names if cfg and cfg.module_names else
+                 [n for n, _ in self.model.named_modules()
+                  if n.startswith("layers.") and n.count(".") == 1])
+        return TensorCapture(self.model, names,
+                             max_steps=cfg.max_steps if cfg else 16)
+
+    def apply_tensor_replacements(self):
+        """Context manager substituting golden tensors into the
+        configured modules (ref TensorReplacementConfig
+        config.py:1172-1202): isolates which submodule diverges."""
+        from neuronx_distributed_inference_amd.utils.tensor_capture import (
+            TensorReplacement,
+        )
+
+        cfg = self.ec.tensor_replacement_config
+        repl = {name: torch.load(path) for name, path in cfg.replacements.items()}
+        return TensorReplacement(self.model, repl)
+
     def reset(self):
         """Clear KV state (ref model_base.py:3926)."""
         self.kv_cache_populated = False

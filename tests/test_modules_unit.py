@@ -532,3 +532,60 @@ def test_bucket_strategies():
     assert get_target_bucket(200, bs, "second_fit") == 512
     assert get_target_bucket(1000, bs, "second_fit") == 1024
     assert get_target_bucket(5, bs, "max") == 1024
+
+
+def test_tensor_capture_replacement_configs():
+    """TensorCapture/TensorReplacement sub-configs (ref config.py:
+    1121-1202): JSON round trip + attach_tensor_capture records module
+    outputs + replacements feed goldens."""
+    import tempfile
+
+    import torch
+
+    from neuronx_distributed_inference_amd.config import (
+        EngineConfig,
+        TensorCaptureConfig,
+        TensorReplacementConfig,
+    )
+    from neuronx_distributed_inference_amd.models.llama.modeling_llama import (
+        LlamaForCausalLM,
+        LlamaInferenceConfig,
+    )
+
+    ec = EngineConfig(
+        batch_size=1, seq_len=32, max_length=32, torch_dtype="float32",
+        tp_degree=1,
+        tensor_capture_config=TensorCaptureConfig(
+            module_names=["layers.0"], max_steps=4),
+    )
+    d = ec.to_dict()
+    ec2 = EngineConfig.from_dict(d)
+    assert ec2.tensor_capture_config.module_names == ["layers.0"]
+    assert ec2.tensor_capture_config.max_steps == 4
+
+    cfg = LlamaInferenceConfig(
+        neuron_config=ec, vocab_size=64, hidden_size=32, intermediate_size=64,
+        num_hidden_layers=2, num_attention_heads=2, num_key_value_heads=1,
+        max_position_embeddings=64, rms_norm_eps=1e-5, rope_theta=10000.0,
+        tie_word_embeddings=False)
+    torch.manual_seed(0)
+    m = LlamaForCausalLM(cfg).eval()
+    ids = torch.randint(0, 64, (1, 8))
+    with m.attach_tensor_capture() as cap:
+        m(input_ids=ids)
+    assert len(cap.captured["layers.0"]) == 1
+    assert cap.captured["layers.0"][0].shape[:2] == (1, 8)
+
+    # replacement: feed a golden input into layer 1 and observe it used
+    golden = torch.zeros(1, 8, 32)
+    with tempfile.NamedTemporaryFile(suffix=".pt", delete=False) as f:
+        torch.save(golden, f.name)
+        path = f.name
+    m.config.neuron_config.tensor_replacement_config = TensorReplacementConfig(
+        replacements={"layers.1": path})
+    m.reset()
+    with m.attach_tensor_capture() as cap2, m.apply_tensor_replacements():
+        m(input_ids=ids)
+    # layer 1 received zeros -> its output differs from the unreplaced run
+    assert not torch.allclose(cap2.captured["layers.0"][0],
+                              golden)
