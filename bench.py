@@ -219,6 +219,10 @@ def main():
         on_device_sampling_config=OnDeviceSamplingConfig(top_k=1),
         enable_hip_graphs=not args.no_graphs,
         fused_qkv=True, fused_gate_up=True,
+        # the bench always decodes at full batch: a single captured graph
+        # (the sub-batch ladder is a serving feature; fewer RCCL-in-graph
+        # captures = less surface on the multi-GPU scaling runs)
+        enable_batch_bucketing=False,
     )
     if args.model in ("mixtral", "dbrx"):
         # dbrx geometry is benchmarked through the same stacked-expert MoE
