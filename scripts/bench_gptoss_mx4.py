@@ -77,7 +77,11 @@ def main():
         # numerics sanity
         got = mx(x, w, idx)
         want = bf(x, w, idx)
-        err = (got.float() - want.float()).abs().max().item()
+        # RELATIVE error: outputs are O(100+) at this scale (clamped-GLU
+        # boundary effects amplify bf16 rounding absolutely, not
+        # relatively)
+        err = ((got.float() - want.float()).abs().max()
+               / want.float().abs().max().clamp(min=1)).item()
         t_mx = t(lambda: mx(x, w, idx))
         os.environ["NXDI_MOE_GROUPED"] = "1"
         t_bf_grouped = t(lambda: bf(x, w, idx))
@@ -85,7 +89,7 @@ def main():
         t_bf_loop = t(lambda: bf(x, w, idx))
         print(f"T={T:4d}: mxfp4 {t_mx:7.1f} us | bf16 grouped {t_bf_grouped:7.1f}"
               f" | bf16 loop {t_bf_loop:7.1f} | speedup vs best bf16 "
-              f"{min(t_bf_grouped, t_bf_loop)/t_mx:.2f}x | maxerr {err:.3f}",
+              f"{min(t_bf_grouped, t_bf_loop)/t_mx:.2f}x | relerr {err:.4f}",
               flush=True)
 
 
