@@ -226,15 +226,9 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_mx4_kernel(
     using MF = MgMfma<T>;
     using vec8 = typename MF::vec8;
     constexpr int MT = 64;
-    constexpr int XBYTES = MT * 128;     // x bf16 image per chunk
-    constexpr int WBYTES = MG_BN * 128;  // dequantized W image per chunk
-    constexpr int PBYTES = MG_BN * 32;   // packed e2m1 per chunk (2 KB)
-    constexpr int SBYTES = MG_BN * 2;    // e8m0 scales per chunk (128 B)
-    constexpr int CHUNK_B = XBYTES + WBYTES + PBYTES + SBYTES;
-    // per-wave glds per chunk: every wave stages 1 KB of x; wave 0 ALSO
-    // issues the packed-W (2 x b16-per-lane... 2 x 1KB) + scales (1)
-    constexpr int GX = 1;
-    constexpr int GW = GX + 3;
+    constexpr int G = 1;                 // glds per wave per chunk (x only)
+    constexpr int XBYTES = MT * 128;
+    constexpr int CHUNK_B = XBYTES + MG_BN * 128;
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
 
@@ -255,68 +249,36 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_mx4_kernel(
 
     const char* xsrc = reinterpret_cast<const char*>(x) +
                        ((long)(m0 + wid * 8) * K) * sizeof(T);
-    // packed W: wave 0 stages 2 KB/chunk as 2 glds of 1 KB (each lane
-    // 16 B): instr j covers rows [j*32, j*32+32) with lane l -> row
-    // j*32 + l/2, 16 B half (l&1) of the row's 32-B chunk piece
-    const long wq_rowbytes = (long)K / 2;
-    const char* wqbase = reinterpret_cast<const char*>(wq) +
-                         ((long)e * N + n0) * wq_rowbytes;
-    const long ws_rowbytes = (long)K / 32;
-    const char* wsbase = reinterpret_cast<const char*>(ws) +
-                         ((long)e * N + n0) * ws_rowbytes;
+    // this lane stages W row (n0 + wid*8 + lane/8), k-slot lane%8
+    const int wrow_l = wid * 8 + (lane >> 3);
+    const int wslot = lane & 7;          // 8 bf16 = 4 packed bytes
+    const unsigned char* wq_row = wq + ((long)e * N + n0 + wrow_l) * (K / 2);
+    const unsigned char* ws_row = ws + ((long)e * N + n0 + wrow_l) * (K / 32);
 
     char* xdst[MG_NBUF];
-    char* wimg[MG_NBUF];
-    char* wpk[MG_NBUF];
-    char* wsc[MG_NBUF];
+    char* wdst[MG_NBUF];
 #pragma unroll
     for (int b = 0; b < MG_NBUF; ++b) {
-        char* base = smem + b * CHUNK_B;
-        xdst[b] = base + wid * 1024;
-        wimg[b] = base + XBYTES;
-        wpk[b] = base + XBYTES + WBYTES;
-        wsc[b] = base + XBYTES + WBYTES + PBYTES;
+        xdst[b] = smem + b * CHUNK_B + wid * 1024;
+        wdst[b] = smem + b * CHUNK_B + XBYTES;
     }
 
-#define MX_STAGE(t, b)                                                       \
+    // x: glds with counted vmcnt; W: ordinary loads + in-register
+    // dequant + swizzled ds_write (the ds_write slot XOR matches the
+    // read-side XOR; note hipcc waits vmcnt(0) for the W loads while a
+    // glds is in flight — the known mixed-kind cost, acceptable v1)
+#define MX_STAGE_X(t, b)                                                     \
+    mg_glds_rows128<0>(xsrc + (long)(t) * 128, (long)K * sizeof(T), xdst[b])
+#define MX_STAGE_W(t, b)                                                     \
     do {                                                                     \
-        const long koff = (long)(t) * 128;                                   \
-        mg_glds_rows128<0>(xsrc + koff, (long)K * sizeof(T), xdst[b]);       \
-        if (wid == 0) {                                                      \
-            const long pkoff = (long)(t) * 32;                               \
-            _Pragma("unroll") for (int j = 0; j < 2; ++j) {                  \
-                const char* src = wqbase + (long)(j * 32 + (lane >> 1)) *    \
-                                      wq_rowbytes + pkoff + (lane & 1) * 16; \
-                __builtin_amdgcn_global_load_lds(                            \
-                    (const __attribute__((address_space(1))) unsigned int*)src, \
-                    (__attribute__((address_space(3))) unsigned int*)         \
-                        (wpk[b] + j * 1024), 16, 0, 2);                      \
-            }                                                                \
-            const char* ssrc = wsbase + (long)lane * ws_rowbytes + (t) * 2;  \
-            __builtin_amdgcn_global_load_lds(                                \
-                (const __attribute__((address_space(1))) unsigned int*)ssrc, \
-                (__attribute__((address_space(3))) unsigned int*)wsc[b],     \
-                2, 0, 2);                                                    \
-        }                                                                    \
-    } while (0)
-
-    // dequant pass: LDS packed -> LDS bf16 image (ds ops only; each of
-    // the 512 threads expands one u32 = 8 e2m1 into one swizzled 16-B
-    // image slot)
-#define MX_DEQUANT(b)                                                        \
-    do {                                                                     \
-        const int tid = (int)threadIdx.x;                                    \
-        const int row = tid >> 3;                                            \
-        const int slot = tid & 7;                                            \
-        const unsigned pkv = *reinterpret_cast<const unsigned*>(             \
-            wpk[b] + row * 32 + slot * 4);                                   \
-        const float sc = exp2f(                                              \
-            (int)*reinterpret_cast<const unsigned char*>(                    \
-                wsc[b] + row * 2 + (slot >> 2)) - 127);                      \
+        const int kc = (t) * MG_BK;                                          \
+        const unsigned pk = *reinterpret_cast<const unsigned*>(              \
+            wq_row + (kc + wslot * 8) / 2);                                  \
+        const float sc = exp2f((int)ws_row[(kc + wslot * 8) / 32] - 127);    \
         T tmp[8];                                                            \
-        mx4_dequant8<T>(pkv, 0.5f * sc, tmp);                                \
-        const int sl = slot ^ (row & 7);                                     \
-        *reinterpret_cast<int4*>(wimg[b] + row * 128 + sl * 16) =            \
+        mx4_dequant8<T>(pk, 0.5f * sc, tmp);                                 \
+        const int sl = wslot ^ (lane >> 3);                                  \
+        *reinterpret_cast<int4*>(wdst[b] + wrow_l * 128 + sl * 16) =         \
             *reinterpret_cast<int4*>(tmp);                                   \
     } while (0)
 
@@ -350,26 +312,22 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_mx4_kernel(
         }                                                                    \
     } while (0)
 
-    // two raw barriers per chunk: packed(t) landed -> dequant(t) ->
-    // image visible -> fragments + MFMA; glds for t+1/t+2 stay in
-    // flight across both barriers (counted vmcnt, never 0 mid-loop)
 #define MX_BODY(t, b, b2)                                                    \
     do {                                                                     \
-        if ((t) + 1 < nchunks) {                                             \
-            if (wid == 0) mg_vm_wait<GW>();                                  \
-            else mg_vm_wait<GX>();                                           \
-        } else mg_vm_wait<0>();                                              \
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                   \
-        __builtin_amdgcn_s_barrier();                                        \
-        MX_DEQUANT(b);                                                       \
+        if ((t) + 1 < nchunks) mg_vm_wait<G>();                              \
+        else mg_vm_wait<0>();                                                \
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                   \
         __builtin_amdgcn_s_barrier();                                        \
         MX_FRAGS(b);                                                         \
-        if ((t) + 2 < nchunks) MX_STAGE((t) + 2, b2);                        \
+        if ((t) + 2 < nchunks) {                                             \
+            MX_STAGE_X((t) + 2, b2);                                         \
+            MX_STAGE_W((t) + 2, b2);                                         \
+        }                                                                    \
     } while (0)
 
-    MX_STAGE(0, 0);
-    if (nchunks > 1) MX_STAGE(1, 1);
+    MX_STAGE_X(0, 0);
+    MX_STAGE_W(0, 0);
+    if (nchunks > 1) { MX_STAGE_X(1, 1); MX_STAGE_W(1, 1); }
     int t = 0;
     while (t + 3 <= nchunks) {
         MX_BODY(t + 0, 0, 2);
@@ -381,8 +339,8 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_mx4_kernel(
     if (t + 1 < nchunks) MX_BODY(t + 1, 1, 0);
 #undef MX_BODY
 #undef MX_FRAGS
-#undef MX_DEQUANT
-#undef MX_STAGE
+#undef MX_STAGE_W
+#undef MX_STAGE_X
     mg_vm_wait<0>();
 
     const int n_out = n0 + wn0 + lrow;
@@ -461,7 +419,7 @@ void moe_grouped_gemm_mx4(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
     TORCH_CHECK(wq.scalar_type() == at::ScalarType::Byte &&
                 ws.scalar_type() == at::ScalarType::Byte, "u8 packed");
 
-    constexpr int LDS = MG_NBUF * (64 * 128 + MG_BN * 128 + MG_BN * 32 + MG_BN * 2);
+    constexpr int LDS = MG_NBUF * (64 * 128 + MG_BN * 128);
     auto stream = at::hip::getCurrentHIPStream();
     dim3 grid(N / MG_BN, Mpad / 64);
 
