@@ -236,6 +236,13 @@ class DecoderModel(nn.Module):
             # (h + deferred residual)
             if post_fns and li in post_fns:
                 h = post_fns[li](h, input_ids, phase)
+            # EAGLE-3 conditions the draft on SEVERAL target layers'
+            # hiddens (ref model_base.py:1402-1437: 3 layers concatenated)
+            cap = getattr(self, "eagle3_capture_layers", None)
+            if cap and li in cap:
+                if li == cap[0]:
+                    self.eagle3_hiddens = {}
+                self.eagle3_hiddens[li] = (h + residual) if residual is not None else h
         if residual is not None:
             h, _ = self.norm(h, residual)
         else:

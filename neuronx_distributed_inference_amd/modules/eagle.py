@@ -385,3 +385,142 @@ class EagleTreeAdapter:
         for i, r in enumerate(out_rows):
             out[i, : len(r)] = torch.tensor(r, device=device)
         return out
+
+
+class Eagle3DraftModel(nn.Module):
+    """EAGLE-3 draft (ref model_base.py:1402-1437): the combiner takes
+    the token embedding plus THREE target layers' hidden states
+    concatenated — fc: (emb, h_low, h_mid, h_high) -> H. The target
+    exposes them via `eagle3_capture_layers`/`eagle3_hiddens`."""
+
+    def __init__(self, base_model, hidden_size: int, dtype: torch.dtype,
+                 capture_layers):
+        super().__init__()
+        self.base = base_model          # DecoderModel (draft-sized)
+        self.capture_layers = list(capture_layers)
+        n = 1 + len(self.capture_layers)
+        self.fc = nn.Linear(n * hidden_size, hidden_size, bias=False, dtype=dtype)
+
+    @torch.inference_mode()
+    def forward(self, input_ids, prev_hiddens, position_ids, seq_ids, phase,
+                attention_mask=None, rope_position_ids=None):
+        """prev_hiddens: (B, S, 3H) — the concatenated capture-layer
+        hiddens at the conditioning positions."""
+        b = self.base
+        emb = b.embed_tokens(input_ids)
+        h = self.fc(torch.cat([emb, prev_hiddens.to(emb.dtype)], dim=-1))
+        rp = rope_position_ids if rope_position_ids is not None else position_ids
+        cos, sin = b.rotary.get_cos_sin(rp, device=h.device)
+        residual = None
+        for layer in b.layers:
+            h, residual = layer(h, residual, cos, sin, b.kv_mgr, position_ids,
+                                phase, attention_mask, seq_ids)
+        h_final = b.norm(h, residual)[0] if residual is not None else b.norm(h)
+        logits = b.lm_head(h_final)
+        tokens = distributed_argmax(logits, dim=-1)
+        return tokens, h_final
+
+
+class Eagle3GenerationAdapter:
+    """EAGLE-3 chain decoding: like EAGLE-1 but the draft's conditioning
+    vector is cat of 3 target layers' hiddens; after the first step the
+    draft chains on its OWN hidden replicated across the 3 slots (the
+    reference's draft-feedback convention)."""
+
+    def __init__(self, target, draft: Eagle3DraftModel, speculation_length: int):
+        self.target = target
+        self.draft = draft
+        self.k = speculation_length
+        target.model.eagle3_capture_layers = draft.capture_layers
+
+    def _cat_capture(self, B):
+        hs = self.target.model.eagle3_hiddens
+        return torch.cat([hs[li] for li in self.draft.capture_layers], dim=-1)
+
+    @torch.inference_mode()
+    def generate(self, input_ids, attention_mask=None, generation_config=None, **kw):
+        gc = generation_config or GenerationConfig(**{
+            k: v for k, v in kw.items() if k in GenerationConfig.__dataclass_fields__})
+        B, S0 = input_ids.shape
+        device = input_ids.device
+        if attention_mask is None:
+            attention_mask = torch.ones_like(input_ids)
+        max_len = gc.max_length or min(self.target.ec.max_length, S0 + gc.max_new_tokens)
+        max_len = min(max_len, self.target.ec.max_length - self.k - 2)
+        k = self.k
+        nh = len(self.draft.capture_layers)
+
+        self.target.reset()
+        self.draft.base.reset_cache()
+        seq_ids = torch.arange(B, device=device)
+        cum = attention_mask.long().cumsum(-1) - 1
+        position_ids = torch.where(attention_mask.bool(), cum, torch.full_like(cum, -1))
+        t_out = self.target(input_ids=input_ids, attention_mask=attention_mask,
+                            position_ids=position_ids, seq_ids=seq_ids)
+        lengths = attention_mask.sum(-1).long()
+        next_tok = t_out.tokens.view(B)
+        cap = self._cat_capture(B)                    # (B, S, 3H)
+        # draft context encoding over the prompt (shifted tokens vs the
+        # captured hiddens)
+        from neuronx_distributed_inference_amd.modules.attention.attention_base import (
+            PHASE_CTE,
+        )
+
+        shifted = torch.cat([input_ids[:, 1:], next_tok.view(B, 1)], dim=1)
+        self.draft(shifted, cap, position_ids, seq_ids, PHASE_CTE,
+                   attention_mask=attention_mask)
+        h_cond = cap[:, -1:, :]                       # (B, 1, 3H)
+
+        out_rows = [input_ids[i][attention_mask[i].bool()].tolist() for i in range(B)]
+        for i in range(B):
+            out_rows[i].append(int(next_tok[i]))
+        finished = torch.zeros(B, dtype=torch.bool, device=device)
+
+        while True:
+            if gc.eos_token_id is not None:
+                finished |= next_tok == gc.eos_token_id
+            if bool(finished.all()) or max(len(r) for r in out_rows) >= max_len:
+                break
+            cand = []
+            d_tok = next_tok.view(B, 1)
+            d_hid = h_cond
+            d_pos = lengths.view(B, 1)
+            for j in range(k):
+                t_j, h_j = self.draft(d_tok, d_hid, d_pos + j, seq_ids, PHASE_TKG)
+                d_tok = t_j.view(B, 1)
+                d_hid = h_j[:, -1:, :].repeat(1, 1, nh)   # draft feedback
+                cand.append(d_tok)
+            cand = torch.cat(cand, dim=1)
+
+            verify_in = torch.cat([next_tok.view(B, 1), cand], dim=1)
+            verify_pos = lengths.view(B, 1) + torch.arange(k + 1, device=device).view(1, -1)
+            v_out = self.target(input_ids=verify_in, position_ids=verify_pos,
+                                seq_ids=seq_ids)
+            target_toks = v_out.tokens.view(B, k + 1)
+            cap_all = self._cat_capture(B)            # (B, k+1, 3H)
+
+            match = (cand == target_toks[:, :k])
+            n_accept = match.cumprod(dim=1).sum(dim=1)
+            for i in range(B):
+                if finished[i]:
+                    continue
+                m = int(n_accept[i])
+                row = out_rows[i]
+                for tok in cand[i, :m].tolist() + [int(target_toks[i, m])]:
+                    if len(row) >= max_len:
+                        break
+                    row.append(tok)
+                    if gc.eos_token_id is not None and tok == gc.eos_token_id:
+                        finished[i] = True
+                        break
+            h_cond = cap_all.gather(
+                1, n_accept.view(B, 1, 1).expand(-1, 1, cap_all.shape[-1]))
+            lengths = lengths + n_accept + 1
+            next_tok = torch.tensor([r[-1] for r in out_rows], device=device,
+                                    dtype=torch.long)
+
+        longest = max(len(r) for r in out_rows)
+        out = torch.full((B, longest), gc.pad_token_id, dtype=torch.long, device=device)
+        for i, r in enumerate(out_rows):
+            out[i, : len(r)] = torch.tensor(r, device=device)
+        return out

@@ -437,3 +437,33 @@ def test_medusa_tree_matches_target_only():
         n = min(n, want.shape[1])
         assert n >= ids.shape[1] + 8
         assert torch.equal(row[:n], want[i, :n]), (row.tolist(), want[i].tolist())
+
+
+def test_eagle3_matches_target_only():
+    """EAGLE-3 (draft conditioned on 3 target layers' concatenated
+    hiddens — ref model_base.py:1402-1437) must equal target-only greedy
+    output."""
+    from neuronx_distributed_inference_amd.modules.eagle import (
+        Eagle3DraftModel,
+        Eagle3GenerationAdapter,
+    )
+
+    target = _build(64, 2, seed=1, spec_len=3)
+    draft_lm = _build(64, 1, seed=7)
+    torch.manual_seed(8)
+    draft = Eagle3DraftModel(draft_lm.model, hidden_size=64,
+                             dtype=torch.float32, capture_layers=[0, 1, 1])
+    with torch.no_grad():
+        draft.fc.weight.normal_(0, 0.05)
+    ids = torch.randint(0, 64, (2, 8))
+    want = HuggingFaceGenerationAdapter(_build(64, 2, seed=1)).generate(
+        ids, generation_config=GenerationConfig(max_new_tokens=20))
+    got = Eagle3GenerationAdapter(target, draft, speculation_length=3).generate(
+        ids, generation_config=GenerationConfig(max_new_tokens=20))
+    for i in range(2):
+        row = got[i]
+        n = int((row != 0).float().flip(0).argmax())
+        n = row.shape[0] - n
+        n = min(n, want.shape[1])
+        assert n >= ids.shape[1] + 8
+        assert torch.equal(row[:n], want[i, :n]), (row.tolist(), want[i].tolist())
