@@ -279,6 +279,9 @@ class EngineConfig:
         self.num_medusa_heads: int = kwargs.pop("num_medusa_heads", 0)
         self.is_medusa: bool = kwargs.pop("is_medusa", False)
         self.token_tree_config = kwargs.pop("token_tree_config", None)
+        # EAGLE-3: target layer indices whose hiddens the draft consumes;
+        # set before load() so hipGraph capture records the hidden writes
+        self.eagle3_capture_layers = kwargs.pop("eagle3_capture_layers", None)
 
         # --- quantization (ref config.py:215-241) ---
         self.quantized: bool = kwargs.pop("quantized", False)

@@ -131,6 +131,14 @@ def setup_run_parser(run: argparse.ArgumentParser):
     run.add_argument("--is-prefix-caching", action="store_true")
     run.add_argument("--is-chunked-prefill", action="store_true")
     run.add_argument("--enable-eagle-speculation", action="store_true")
+    run.add_argument("--enable-token-tree", action="store_true",
+                     help="EAGLE TREE speculation (needs --enable-eagle-speculation)")
+    run.add_argument("--token-tree-branching", default="2,2",
+                     help="per-level branching factors of the static draft tree")
+    run.add_argument("--enable-eagle3", action="store_true",
+                     help="EAGLE-3: draft conditioned on 3 target layers' hiddens")
+    run.add_argument("--eagle3-capture-layers", default=None,
+                     help="csv of target layer indices (default low/mid/high)")
     run.add_argument("--num-medusa-heads", type=int, default=0)
     run.add_argument("--quantization-type", default=None,
                      choices=[None, "per_tensor_symmetric", "per_channel_symmetric"])
@@ -173,6 +181,10 @@ def create_engine_config(args, model_type: str) -> EngineConfig:
             top_k=args.top_k, top_p=args.top_p, temperature=args.temperature,
             do_sample=args.do_sample, global_topk=args.global_topk,
         )
+    elif getattr(args, "enable_eagle_speculation", False):
+        # EAGLE verification consumes on-device greedy tokens (the ref
+        # likewise forces on-device sampling under speculation).
+        odsc = OnDeviceSamplingConfig(top_k=1)
     cls = MoEEngineConfig if model_type in ("mixtral", "dbrx", "qwen3_moe") else EngineConfig
     return cls(
         batch_size=args.batch_size,
@@ -298,6 +310,14 @@ def run_inference(args, model_type: str) -> int:
 
     engine_config = create_engine_config(args, model_type)
     config = config_cls.from_hf_config(args.model_path, neuron_config=engine_config)
+    if getattr(args, "enable_eagle3", False):
+        # resolve capture layers BEFORE load() so graph capture records them
+        if args.eagle3_capture_layers:
+            caps = [int(x) for x in args.eagle3_capture_layers.split(",")]
+        else:
+            n = config.num_hidden_layers
+            caps = [1, n // 2, n - 2] if n >= 4 else [0, 0, 0]
+        engine_config.eagle3_capture_layers = caps
     app = InferenceApplication(model_cls, config, model_path=args.model_path)
 
     rank0 = ps.get_tensor_model_parallel_rank() == 0
@@ -346,7 +366,54 @@ def run_inference(args, model_type: str) -> int:
         if torch.distributed.is_initialized():
             torch.distributed.barrier()
         draft = draft_app.load(draft_dir, device=device)
-        if args.enable_fused_speculation:
+        if args.enable_eagle_speculation:
+            # EAGLE drafts wrap the draft DecoderModel with the fc combiner;
+            # fc weights come from the draft checkpoint when present
+            # (ref eagle draft fc, modeling_llama.py:1161-1167), else
+            # random-init (synthetic demos).
+            from neuronx_distributed_inference_amd.modules.checkpoint import (
+                load_state_dict,
+            )
+            from neuronx_distributed_inference_amd.modules.eagle import (
+                Eagle3DraftModel,
+                Eagle3GenerationAdapter,
+                EagleDraftModel,
+                EagleGenerationAdapter,
+                EagleTreeAdapter,
+            )
+
+            hid = config.hidden_size
+            dt = draft_ec.torch_dtype
+            if isinstance(dt, str):
+                dt = getattr(torch, dt)
+            if args.enable_eagle3:
+                edraft = Eagle3DraftModel(
+                    draft.model, hid, dt, engine_config.eagle3_capture_layers)
+            else:
+                edraft = EagleDraftModel(draft.model, hid, dt)
+            try:
+                sd = load_state_dict(args.draft_model_path)
+                fc_w = sd.get("fc.weight", sd.get("model.fc.weight"))
+                if fc_w is not None and fc_w.shape == edraft.fc.weight.shape:
+                    with torch.no_grad():
+                        edraft.fc.weight.copy_(fc_w)
+            except (OSError, ValueError):
+                pass
+            edraft = edraft.to(device)
+            if args.enable_eagle3:
+                adapter = Eagle3GenerationAdapter(
+                    model, edraft, args.speculation_length)
+            elif args.enable_token_tree:
+                from neuronx_distributed_inference_amd.modules.token_tree import (
+                    TokenTree,
+                )
+
+                branching = [int(x) for x in args.token_tree_branching.split(",")]
+                adapter = EagleTreeAdapter(model, edraft, TokenTree(branching))
+            else:
+                adapter = EagleGenerationAdapter(
+                    model, edraft, args.speculation_length)
+        elif args.enable_fused_speculation:
             rej = None
             if args.spec_rejection_sampling:
                 rej = engine_config.on_device_sampling_config or \

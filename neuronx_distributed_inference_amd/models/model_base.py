@@ -69,6 +69,14 @@ class DecoderModel(nn.Module):
         self.hidden_size = config.hidden_size
         self.num_layers = config.num_hidden_layers
         self.dtype = ec.torch_dtype
+        # EAGLE hidden capture must be armed BEFORE hipGraph capture so the
+        # CTE graph records the (static) full_hidden write; the demo sets
+        # enable_eagle_speculation on the config, adapters may also toggle
+        # the attribute directly in eager runs
+        self.eagle_capture_hidden = bool(
+            getattr(ec, "enable_eagle_speculation", False))
+        if getattr(ec, "eagle3_capture_layers", None):
+            self.eagle3_capture_layers = list(ec.eagle3_capture_layers)
 
         self.embed_tokens = ParallelEmbedding(
             config.vocab_size, config.hidden_size, dtype=self.dtype,

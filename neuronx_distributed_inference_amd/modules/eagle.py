@@ -89,6 +89,10 @@ def eagle_draft_prefill(target, draft: "EagleDraftModel", input_ids,
     if h_full is None:
         return  # capture flag was off: draft starts cold (still correct)
     B, S = input_ids.shape
+    # the app may have padded CTE to a context bucket: captured hiddens
+    # then carry the padded S — the prompt occupies the first S positions
+    # (right padding)
+    h_full = h_full[:, :S]
     # shifted tokens: t_1..t_{S-1}, next_tok; hiddens: h_0..h_{S-1}
     shifted = torch.cat([input_ids[:, 1:], next_tok.view(B, 1)], dim=1)
     draft(shifted, h_full, position_ids, seq_ids, PHASE_CTE,
@@ -128,6 +132,8 @@ class EagleGenerationAdapter:
         lengths = attention_mask.sum(-1).long()
         next_tok = t_out.tokens.view(B)
         # target hidden of the LAST prompt token conditions the draft
+        # (model_base already gathered CTE last_hidden at the true last
+        # prompt position, mask-aware)
         h_tgt = self.target.model.last_hidden[:, -1, :]     # (B, H)
 
         # draft prefill: (token_{i+1}, target_hidden_i) over the prompt
@@ -162,7 +168,7 @@ class EagleGenerationAdapter:
             verify_pos = lengths.view(B, 1) + torch.arange(k + 1, device=device).view(1, -1)
             v_out = self.target(input_ids=verify_in, position_ids=verify_pos, seq_ids=seq_ids)
             target_toks = v_out.tokens.view(B, k + 1)
-            hidden_all = self.target.model.last_hidden      # (B, k+1, H)
+            hidden_all = self.target.model.last_hidden[:, : k + 1]  # (B, k+1, H)
 
             match = (cand == target_toks[:, :k])
             n_accept = match.cumprod(dim=1).sum(dim=1)
@@ -459,7 +465,7 @@ class Eagle3GenerationAdapter:
                             position_ids=position_ids, seq_ids=seq_ids)
         lengths = attention_mask.sum(-1).long()
         next_tok = t_out.tokens.view(B)
-        cap = self._cat_capture(B)                    # (B, S, 3H)
+        cap = self._cat_capture(B)[:, :S0]            # (B, S, 3H)
         # draft context encoding over the prompt (shifted tokens vs the
         # captured hiddens)
         from neuronx_distributed_inference_amd.modules.attention.attention_base import (
@@ -469,7 +475,9 @@ class Eagle3GenerationAdapter:
         shifted = torch.cat([input_ids[:, 1:], next_tok.view(B, 1)], dim=1)
         self.draft(shifted, cap, position_ids, seq_ids, PHASE_CTE,
                    attention_mask=attention_mask)
-        h_cond = cap[:, -1:, :]                       # (B, 1, 3H)
+        h_cond = cap.gather(
+            1, (lengths - 1).clamp(min=0).view(B, 1, 1).expand(-1, 1, cap.shape[-1])
+        )                                             # (B, 1, 3H)
 
         out_rows = [input_ids[i][attention_mask[i].bool()].tolist() for i in range(B)]
         for i in range(B):
@@ -497,7 +505,7 @@ class Eagle3GenerationAdapter:
             v_out = self.target(input_ids=verify_in, position_ids=verify_pos,
                                 seq_ids=seq_ids)
             target_toks = v_out.tokens.view(B, k + 1)
-            cap_all = self._cat_capture(B)            # (B, k+1, 3H)
+            cap_all = self._cat_capture(B)[:, : k + 1]  # (B, k+1, 3H)
 
             match = (cand == target_toks[:, :k])
             n_accept = match.cumprod(dim=1).sum(dim=1)

@@ -67,3 +67,30 @@ def test_inference_demo_on_device_sampling_flags(tiny_hf_checkpoint, tmp_path):
         "--batch-size", "1", "--max-context-length", "16", "--seq-len", "32",
     ])
     assert rc == 0
+
+
+@pytest.mark.parametrize("extra", [
+    [],                                             # EAGLE-1 chain
+    ["--enable-token-tree", "--token-tree-branching", "2,2"],
+    ["--enable-eagle3"],
+])
+def test_inference_demo_eagle_flags(tiny_hf_checkpoint, tmp_path, extra):
+    """CLI wiring of the EAGLE family (chain / tree / EAGLE-3): draft =
+    same tiny checkpoint; fc is random-init (no fc key in the ckpt), so
+    only the pipeline contract is checked, not acceptance quality."""
+    from neuronx_distributed_inference_amd.inference_demo import main
+
+    compiled = str(tmp_path / ("compiled_eagle" + str(len(extra))))
+    rc = main([
+        "--model-type", "llama", "run",
+        "--model-path", tiny_hf_checkpoint,
+        "--compiled-model-path", compiled,
+        "--prompt-ids", json.dumps([[5, 9, 13, 21, 7]]),
+        "--max-new-tokens", "6",
+        "--on-cpu", "--torch-dtype", "float32",
+        "--batch-size", "1", "--max-context-length", "32", "--seq-len", "64",
+        "--draft-model-path", tiny_hf_checkpoint,
+        "--speculation-length", "2",
+        "--enable-eagle-speculation",
+    ] + extra)
+    assert rc == 0
