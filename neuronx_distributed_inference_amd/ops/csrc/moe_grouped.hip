@@ -66,8 +66,12 @@ DEV_INLINE void mg_vm_wait() {
     asm volatile("s_waitcnt vmcnt(%c0)" ::"i"(N) : "memory");
 }
 
-// MT=64 row tiles (8 waves as 2(M) x 4(N): wave = 32 rows x 16 cols)
-template <typename T>
+// MT=64 row tiles, BNT-wide column tiles (8 waves as 2(M) x 4(N): each
+// wave = 32 rows x BNT/4 cols). Wider BNT cuts the x re-read traffic
+// (x is re-read once per column tile) and raises MFMA work per staged
+// byte; BNT=256 costs occupancy 1 (120 KB LDS) but each block then
+// streams 4x the W per x fetch.
+template <typename T, int BNT>
 __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
     T* __restrict__ y,              // (Mpad, N) sorted-space output
     const T* __restrict__ x,        // (Mpad, K) sorted-space input
@@ -78,10 +82,11 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
     using MF = MgMfma<T>;
     using vec8 = typename MF::vec8;
     constexpr int MT = 64;
-    constexpr int NXG = 1;          // x glds per wave per chunk (8 rows)
-    constexpr int G = NXG + 1;
+    constexpr int NWG = BNT / 64;   // W glds per wave per chunk (8 rows each)
+    constexpr int G = 1 + NWG;      // total glds per wave per chunk
+    constexpr int NFRAG = BNT / 64; // 16-col fragments per wave
     constexpr int XBYTES = MT * 128;
-    constexpr int CHUNK_B = XBYTES + MG_BN * 128;
+    constexpr int CHUNK_B = XBYTES + BNT * 128;
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
 
@@ -89,47 +94,54 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
     const int rt = blockIdx.y;      // row tile
     const int e = expert_ids[rt];
     if (e < 0) return;              // beyond this step's padded rows
-    const int n0 = tile * MG_BN;
+    const int n0 = tile * BNT;
     const int m0 = rt * MT;
     const int nchunks = K / MG_BK;
 
     const int lane = threadIdx.x & 63;
     const int wid = threadIdx.x >> 6;
     const int wm0 = (wid >> 2) * 32;        // 2 waves along M
-    const int wn0 = (wid & 3) * 16;         // 4 waves along N
+    const int wn0 = (wid & 3) * (BNT / 4);  // 4 waves along N
     const int lrow = lane & 15;
     const int lslot = lane >> 4;
 
     const char* xsrc = reinterpret_cast<const char*>(x) +
                        ((long)(m0 + wid * 8) * K) * sizeof(T);
-    const char* wsrc = reinterpret_cast<const char*>(w) +
-                       (((long)e * N + n0 + wid * 8) * K) * sizeof(T);
+    const char* wsrc[NWG];
+#pragma unroll
+    for (int i = 0; i < NWG; ++i)
+        wsrc[i] = reinterpret_cast<const char*>(w) +
+                  (((long)e * N + n0 + wid * (8 * NWG) + i * 8) * K) * sizeof(T);
 
     char* xdst[MG_NBUF];
     char* wdst[MG_NBUF];
 #pragma unroll
     for (int b = 0; b < MG_NBUF; ++b) {
         xdst[b] = smem + b * CHUNK_B + wid * 1024;
-        wdst[b] = smem + b * CHUNK_B + XBYTES + wid * 1024;
+        wdst[b] = smem + b * CHUNK_B + XBYTES + wid * (NWG * 1024);
     }
 
 #define MG_STAGE(t, b)                                                       \
     do {                                                                     \
         const long koff = (long)(t) * 128;                                   \
         mg_glds_rows128<0>(xsrc + koff, (long)K * sizeof(T), xdst[b]);       \
-        mg_glds_rows128<2>(wsrc + koff, (long)K * sizeof(T), wdst[b]);       \
+        _Pragma("unroll") for (int i = 0; i < NWG; ++i)                      \
+            mg_glds_rows128<2>(wsrc[i] + koff, (long)K * sizeof(T),          \
+                               wdst[b] + i * 1024);                          \
     } while (0)
 
-    mg_f32x4 acc[2];
-    acc[0] = {0.f, 0.f, 0.f, 0.f};
-    acc[1] = {0.f, 0.f, 0.f, 0.f};
+    mg_f32x4 acc[2][NFRAG];
+#pragma unroll
+    for (int qi = 0; qi < 2; ++qi)
+#pragma unroll
+        for (int nj = 0; nj < NFRAG; ++nj) acc[qi][nj] = {0.f, 0.f, 0.f, 0.f};
 
 #define MG_FRAGS(b)                                                          \
     do {                                                                     \
         const char* xb = smem + (b) * CHUNK_B;                               \
         const char* wb = xb + XBYTES;                                        \
         _Pragma("unroll") for (int ks = 0; ks < 2; ++ks) {                   \
-            vec8 a0, a1, bb;                                                 \
+            vec8 a0, a1, bb[NFRAG];                                          \
             {                                                                \
                 const int m = wm0 + lrow;                                    \
                 const int s = (ks * 4 + lslot) ^ (m & 7);                    \
@@ -140,13 +152,15 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
                 const int s = (ks * 4 + lslot) ^ (m & 7);                    \
                 a1 = *reinterpret_cast<const vec8*>(xb + m * 128 + s * 16);  \
             }                                                                \
-            {                                                                \
-                const int n = wn0 + lrow;                                    \
+            _Pragma("unroll") for (int nj = 0; nj < NFRAG; ++nj) {           \
+                const int n = wn0 + nj * 16 + lrow;                          \
                 const int s = (ks * 4 + lslot) ^ (n & 7);                    \
-                bb = *reinterpret_cast<const vec8*>(wb + n * 128 + s * 16);  \
+                bb[nj] = *reinterpret_cast<const vec8*>(wb + n * 128 + s * 16); \
             }                                                                \
-            acc[0] = MF::mfma(a0, bb, acc[0]);                               \
-            acc[1] = MF::mfma(a1, bb, acc[1]);                               \
+            _Pragma("unroll") for (int nj = 0; nj < NFRAG; ++nj) {           \
+                acc[0][nj] = MF::mfma(a0, bb[nj], acc[0][nj]);               \
+                acc[1][nj] = MF::mfma(a1, bb[nj], acc[1][nj]);               \
+            }                                                                \
         }                                                                    \
     } while (0)
 
@@ -176,15 +190,20 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
 #undef MG_STAGE
     mg_vm_wait<0>();
 
-    // epilogue: y[m0 + wm0 + qi*16 + lslot*4 + r][n0 + wn0 + lrow]
-    const int n_out = n0 + wn0 + lrow;
-    const float bv = bias ? VecTraits<T>::to_float(bias[(long)e * N + n_out]) : 0.f;
+    // epilogue: y[m0 + wm0 + qi*16 + lslot*4 + r][n0 + wn0 + nj*16 + lrow]
 #pragma unroll
-    for (int qi = 0; qi < 2; ++qi) {
+    for (int nj = 0; nj < NFRAG; ++nj) {
+        const int n_out = n0 + wn0 + nj * 16 + lrow;
+        const float bv =
+            bias ? VecTraits<T>::to_float(bias[(long)e * N + n_out]) : 0.f;
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            const int m = m0 + wm0 + qi * 16 + lslot * 4 + r;
-            y[(long)m * N + n_out] = VecTraits<T>::from_float(acc[qi][r] + bv);
+        for (int qi = 0; qi < 2; ++qi) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int m = m0 + wm0 + qi * 16 + lslot * 4 + r;
+                y[(long)m * N + n_out] =
+                    VecTraits<T>::from_float(acc[qi][nj][r] + bv);
+            }
         }
     }
 }
@@ -355,41 +374,143 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_mx4_kernel(
     }
 }
 
+// ---------------------------------------------------------------------------
+// Fused segment-table builder: ONE launch replaces the ~15-kernel torch
+// chain (argsort/scatter_add/cumsum/searchsorted — measured 142 us of
+// pure launch latency at decode sizes vs ~10 us here). Single workgroup:
+// counts via LDS atomics, serial exclusive scans over E (E <= 1024),
+// then each assignment claims its within-expert slot with an atomic
+// cursor. Order within an expert is therefore NOT the stable argsort
+// order — downstream (gather / grouped GEMM / combine) only needs
+// order/dest to be a consistent permutation, which any atomic order is
+// (fp combine order on GPU is nondeterministic anyway via index_add_).
+constexpr int MT_TABLES = 1024;  // threads
+
+__global__ __launch_bounds__(MT_TABLES, 1) void moe_tables_kernel(
+    long* __restrict__ order,        // (Ttk,) sorted pos -> assignment idx
+    long* __restrict__ dest,         // (Ttk,) sorted pos -> padded row
+    int* __restrict__ expert_ids,    // (ntiles,) expert per 64-row tile
+    const int* __restrict__ idx,     // (Ttk,) expert per assignment
+    const int Ttk, const int E, const int ntiles) {
+    extern __shared__ int tsm[];
+    int* counts = tsm;               // (E,)
+    int* starts_raw = tsm + E;       // (E,) exclusive scan of counts
+    int* starts_pad = tsm + 2 * E;   // (E,) exclusive scan of padded
+    int* cursor = tsm + 3 * E;       // (E,) atomic within-expert rank
+    __shared__ int total_pad;
+
+    for (int e = threadIdx.x; e < E; e += blockDim.x) {
+        counts[e] = 0;
+        cursor[e] = 0;
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < Ttk; i += blockDim.x)
+        atomicAdd(&counts[idx[i]], 1);
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        int raw = 0, pad = 0;
+        for (int e = 0; e < E; ++e) {
+            starts_raw[e] = raw;
+            starts_pad[e] = pad;
+            raw += counts[e];
+            pad += (counts[e] + 63) / 64 * 64;
+        }
+        total_pad = pad;
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < Ttk; i += blockDim.x) {
+        const int e = idx[i];
+        const int r = atomicAdd(&cursor[e], 1);   // within-expert rank
+        const int p = starts_raw[e] + r;          // sorted position
+        order[p] = i;
+        dest[p] = starts_pad[e] + r;
+    }
+    // expert per row tile: linear search over E (ntiles and E are small)
+    for (int t = threadIdx.x; t < ntiles; t += blockDim.x) {
+        const int row = t * 64;
+        int eid = -1;
+        if (row < total_pad) {
+            for (int e = 0; e < E; ++e) {
+                const int end =
+                    starts_pad[e] + (counts[e] + 63) / 64 * 64;
+                if (row >= starts_pad[e] && row < end && counts[e] > 0) {
+                    eid = e;
+                    break;
+                }
+            }
+        }
+        expert_ids[t] = eid;
+    }
+}
+
 }  // namespace
+
+void moe_grouped_tables_gpu(torch::Tensor order, torch::Tensor dest,
+                            torch::Tensor expert_ids, torch::Tensor idx_flat,
+                            long num_experts) {
+    const int Ttk = idx_flat.numel();
+    const int E = (int)num_experts;
+    const int ntiles = expert_ids.numel();
+    TORCH_CHECK(idx_flat.scalar_type() == at::ScalarType::Int &&
+                idx_flat.is_contiguous(), "idx_flat int32 contiguous");
+    TORCH_CHECK(order.scalar_type() == at::ScalarType::Long &&
+                dest.scalar_type() == at::ScalarType::Long &&
+                expert_ids.scalar_type() == at::ScalarType::Int, "out dtypes");
+    TORCH_CHECK(order.numel() == Ttk && dest.numel() == Ttk, "out sizes");
+    TORCH_CHECK(E <= 4096, "num_experts too large for LDS tables");
+    const int lds = 4 * E * (int)sizeof(int);
+    auto stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(moe_tables_kernel, dim3(1), dim3(MT_TABLES), lds,
+                       stream, order.data_ptr<long>(), dest.data_ptr<long>(),
+                       expert_ids.data_ptr<int>(), idx_flat.data_ptr<int>(),
+                       Ttk, E, ntiles);
+    HIP_CHECK_KERNEL();
+}
 
 void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                       torch::Tensor expert_ids,
-                      c10::optional<torch::Tensor> bias) {
+                      c10::optional<torch::Tensor> bias, long bn) {
     TORCH_CHECK(y.dim() == 2 && x.dim() == 2 && w.dim() == 3, "shapes");
     TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && y.is_contiguous() &&
                 expert_ids.is_contiguous(), "contiguous tensors required");
     const int Mpad = x.size(0), K = x.size(1);
     const int N = w.size(1);
     TORCH_CHECK(w.size(2) == K && y.size(0) == Mpad && y.size(1) == N, "shape");
-    TORCH_CHECK(N % MG_BN == 0, "N must be a multiple of 64");
     TORCH_CHECK(K % MG_BK == 0, "K must be a multiple of 64");
     TORCH_CHECK(Mpad % 64 == 0, "padded rows must be a multiple of 64");
     TORCH_CHECK(expert_ids.scalar_type() == at::ScalarType::Int &&
                 expert_ids.numel() == Mpad / 64, "expert_ids must be int (Mpad/64,)");
+    // bn: column-tile width. 0 = auto (widest tile that divides N).
+    int BNT = (int)bn;
+    if (BNT == 0) BNT = (N % 256 == 0) ? 256 : (N % 128 == 0) ? 128 : 64;
+    TORCH_CHECK(BNT == 64 || BNT == 128 || BNT == 256, "bn must be 64/128/256");
+    TORCH_CHECK(N % BNT == 0, "N must be a multiple of bn");
 
-    constexpr int LDS = MG_NBUF * (64 * 128 + MG_BN * 128);
     auto stream = at::hip::getCurrentHIPStream();
-    dim3 grid(N / MG_BN, Mpad / 64);
+    dim3 grid(N / BNT, Mpad / 64);
 
-#define MG_T(T)                                                               \
+#define MG_LAUNCH(T, BNTV)                                                    \
     do {                                                                      \
+        constexpr int LDS = MG_NBUF * (64 * 128 + BNTV * 128);                \
         static bool attr_set = false;                                         \
         if (!attr_set) {                                                      \
-            (void)hipFuncSetAttribute((const void*)moe_grouped_kernel<T>,     \
-                                      hipFuncAttributeMaxDynamicSharedMemorySize, \
-                                      LDS);                                   \
+            (void)hipFuncSetAttribute(                                        \
+                (const void*)moe_grouped_kernel<T, BNTV>,                     \
+                hipFuncAttributeMaxDynamicSharedMemorySize, LDS);             \
             attr_set = true;                                                  \
         }                                                                     \
-        hipLaunchKernelGGL((moe_grouped_kernel<T>), grid, dim3(512), LDS,     \
-                           stream, (T*)y.data_ptr(), (const T*)x.data_ptr(),  \
+        hipLaunchKernelGGL((moe_grouped_kernel<T, BNTV>), grid, dim3(512),    \
+                           LDS, stream, (T*)y.data_ptr(),                     \
+                           (const T*)x.data_ptr(),                            \
                            (const T*)w.data_ptr(), expert_ids.data_ptr<int>(),\
                            bias ? (const T*)bias->data_ptr() : nullptr,       \
                            N, K);                                             \
+    } while (0)
+#define MG_T(T)                                                               \
+    do {                                                                      \
+        if (BNT == 64) MG_LAUNCH(T, 64);                                      \
+        else if (BNT == 128) MG_LAUNCH(T, 128);                               \
+        else MG_LAUNCH(T, 256);                                               \
     } while (0)
 
     if (x.scalar_type() == at::ScalarType::BFloat16) {
@@ -400,6 +521,7 @@ void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
         TORCH_CHECK(false, "moe_grouped_gemm: bf16/f16 only");
     }
 #undef MG_T
+#undef MG_LAUNCH
     HIP_CHECK_KERNEL();
 }
 

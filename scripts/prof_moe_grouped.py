@@ -40,7 +40,7 @@ def main():
     print(f"full dense  : {t(lambda: m._forward_dense(x, w, idx)):8.1f} us")
 
     flat = idx.view(-1)
-    print(f"tables      : {t(lambda: ops.moe_grouped_tables(flat, E)):8.1f} us")
+    print(f"tables(fused): {t(lambda: ops.moe_grouped_tables(flat, E)):8.1f} us")
     order, dest, expert_ids, Mpad = ops.moe_grouped_tables(flat, E)
     tok = order // topk
     x_sorted = torch.zeros(Mpad, H, dtype=torch.bfloat16, device=dev)
@@ -50,12 +50,24 @@ def main():
     print(f"gather      : {t(gather):8.1f} us")
     gather()
     wmat = m.w_gate
-    print(f"gemm gate (N={I}): {t(lambda: ops.moe_grouped_gemm(x_sorted, wmat, expert_ids)):8.1f} us  "
-          f"[{Mpad} rows; weight {E*I*H*2/1e6:.0f} MB]")
+    # numerics + perf for each column-tile width
+    ref_g = ops.moe_grouped_gemm(x_sorted, wmat, expert_ids, bn=64)
+    for bn in (64, 128, 256):
+        gg = ops.moe_grouped_gemm(x_sorted, wmat, expert_ids, bn=bn)
+        err = (gg.float() - ref_g.float()).abs().max().item()
+        print(f"gemm gate (N={I}) bn={bn}: "
+              f"{t(lambda: ops.moe_grouped_gemm(x_sorted, wmat, expert_ids, bn=bn)):8.1f} us"
+              f"  maxdiff-vs-bn64 {err:.4f}  [{Mpad} rows; weight {E*I*H*2/1e6:.0f} MB]")
     g = ops.moe_grouped_gemm(x_sorted, wmat, expert_ids)
     u = ops.moe_grouped_gemm(x_sorted, m.w_up, expert_ids)
     h = ops.silu_mul(g, u)
-    print(f"gemm down (N={H}): {t(lambda: ops.moe_grouped_gemm(h, m.w_down, expert_ids)):8.1f} us")
+    ref_d = ops.moe_grouped_gemm(h, m.w_down, expert_ids, bn=64)
+    for bn in (64, 128, 256):
+        dd = ops.moe_grouped_gemm(h, m.w_down, expert_ids, bn=bn)
+        err = (dd.float() - ref_d.float()).abs().max().item()
+        print(f"gemm down (N={H}) bn={bn}: "
+              f"{t(lambda: ops.moe_grouped_gemm(h, m.w_down, expert_ids, bn=bn)):8.1f} us"
+              f"  maxdiff-vs-bn64 {err:.4f}")
     y = ops.moe_grouped_gemm(h, m.w_down, expert_ids)
 
     def combine():
