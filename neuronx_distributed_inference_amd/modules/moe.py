@@ -193,14 +193,13 @@ class ExpertMLPs(nn.Module):
         """x: (T, H); weights/idx: (T, top_k) over GLOBAL expert ids.
         Returns (T, H) partial (summed by the caller's TP all-reduce)."""
         T, H = x.shape
-        # Grouped expert GEMM is OPT-IN (NXDI_MOE_GROUPED=1): measured on
-        # Mixtral-8x7B TP=1, the dense bmm path WINS at decode batches
-        # (b64: 23.7 vs 25.8 ms/step; b256: 32.0 vs 37.6) because
-        # rocBLAS's batched GEMM runs ~1.4 PF/s effective while the
-        # 64x64-tile grouped kernel reaches ~0.3 PF/s — the 4x FLOP
-        # redundancy is cheaper than the tile-efficiency gap. The
-        # grouped kernel stays (correct, tested, graph-safe) as the
-        # substrate for a future 256^2-tile schedule.
+        # Grouped expert GEMM is the GPU DEFAULT (opt out with
+        # NXDI_MOE_GROUPED=0). With wide column tiles (bn=256: 4x fewer
+        # x re-reads) and the fused one-launch segment-table kernel
+        # (142 -> 13 us of launch latency), grouped beats the dense bmm
+        # path on Mixtral decode T=256 top-2 (826 vs 877 us/layer;
+        # profiles/r02_bench_summary.md) on top of removing the top_k/E
+        # redundant FLOPs.
         import os as _os
 
         grouped_env = _os.environ.get("NXDI_MOE_GROUPED")
@@ -209,7 +208,7 @@ class ExpertMLPs(nn.Module):
             and x.dtype in (torch.bfloat16, torch.float16)
             and self.w_gate.shape[2] % 64 == 0 and self.w_gate.shape[1] % 64 == 0
             and H % 64 == 0
-            and grouped_env == "1"
+            and grouped_env != "0"
         )
         if use_grouped:
             return self._forward_grouped(x, weights, idx)

@@ -482,8 +482,12 @@ void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                 expert_ids.numel() == Mpad / 64, "expert_ids must be int (Mpad/64,)");
     // bn: column-tile width. 0 = auto (widest tile that divides N).
     int BNT = (int)bn;
-    if (BNT == 0) BNT = (N % 256 == 0) ? 256 : (N % 128 == 0) ? 128 : 64;
-    TORCH_CHECK(BNT == 64 || BNT == 128 || BNT == 256, "bn must be 64/128/256");
+    if (BNT == 0)
+        BNT = (N % 256 == 0) ? 256
+              : (N % 192 == 0) ? 192
+              : (N % 128 == 0) ? 128 : 64;
+    TORCH_CHECK(BNT == 64 || BNT == 128 || BNT == 192 || BNT == 256,
+                "bn must be 64/128/192/256");
     TORCH_CHECK(N % BNT == 0, "N must be a multiple of bn");
 
     auto stream = at::hip::getCurrentHIPStream();
@@ -510,6 +514,7 @@ void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
     do {                                                                      \
         if (BNT == 64) MG_LAUNCH(T, 64);                                      \
         else if (BNT == 128) MG_LAUNCH(T, 128);                               \
+        else if (BNT == 192) MG_LAUNCH(T, 192);                               \
         else MG_LAUNCH(T, 256);                                               \
     } while (0)
 

@@ -85,8 +85,9 @@ def main():
         t_mx = t(lambda: mx(x, w, idx))
         os.environ["NXDI_MOE_GROUPED"] = "1"
         t_bf_grouped = t(lambda: bf(x, w, idx))
-        os.environ.pop("NXDI_MOE_GROUPED")
+        os.environ["NXDI_MOE_GROUPED"] = "0"
         t_bf_loop = t(lambda: bf(x, w, idx))
+        os.environ.pop("NXDI_MOE_GROUPED")
         print(f"T={T:4d}: mxfp4 {t_mx:7.1f} us | bf16 grouped {t_bf_grouped:7.1f}"
               f" | bf16 loop {t_bf_loop:7.1f} | speedup vs best bf16 "
               f"{min(t_bf_grouped, t_bf_loop)/t_mx:.2f}x | relerr {err:.4f}",
