@@ -203,12 +203,17 @@ class ExpertMLPs(nn.Module):
         import os as _os
 
         grouped_env = _os.environ.get("NXDI_MOE_GROUPED")
+        # small batches leave too few 64-row tiles to fill 256 CUs at the
+        # wide-tile occupancy (measured: Mixtral T=64 grouped 27.7 vs
+        # dense 23.7 ms/step; T=256 grouped 31.2 vs dense 32.0) — dense
+        # bmm keeps those
+        big_enough = x.shape[0] * idx.shape[1] >= 384
         use_grouped = (
             x.is_cuda and ops.extension_available()
             and x.dtype in (torch.bfloat16, torch.float16)
             and self.w_gate.shape[2] % 64 == 0 and self.w_gate.shape[1] % 64 == 0
             and H % 64 == 0
-            and grouped_env != "0"
+            and (grouped_env == "1" or (grouped_env != "0" and big_enough))
         )
         if use_grouped:
             return self._forward_grouped(x, weights, idx)

@@ -751,7 +751,7 @@ def mx4_quantize(w: torch.Tensor):
 
 def moe_grouped_gemm_mx4(x_sorted: torch.Tensor, wq: torch.Tensor,
                          ws: torch.Tensor, expert_ids: torch.Tensor,
-                         bias=None) -> torch.Tensor:
+                         bias=None, bn: int = 0) -> torch.Tensor:
     """Grouped expert GEMM with MXFP4-packed weights (E, N, K/2)+(E, N,
     K/32), dequantized in-kernel during LDS staging (~3.8x less weight
     traffic than bf16). CPU: dequant reference."""
@@ -762,7 +762,7 @@ def moe_grouped_gemm_mx4(x_sorted: torch.Tensor, wq: torch.Tensor,
         y = torch.empty(Mpad, N, dtype=x_sorted.dtype, device=x_sorted.device)
         c.moe_grouped_gemm_mx4(y, x_sorted.contiguous(), wq.contiguous(),
                                ws.contiguous(), expert_ids.contiguous(),
-                               None if bias is None else bias.contiguous())
+                               None if bias is None else bias.contiguous(), bn)
         return y
     y = torch.zeros(Mpad, N, dtype=x_sorted.dtype)
     for t in range(Mpad // MOE_TILE):

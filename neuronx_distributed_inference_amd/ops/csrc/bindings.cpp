@@ -55,7 +55,7 @@ void moe_grouped_tables_gpu(torch::Tensor order, torch::Tensor dest,
                             long num_experts);
 void moe_grouped_gemm_mx4(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
                           torch::Tensor ws, torch::Tensor expert_ids,
-                          c10::optional<torch::Tensor> bias);
+                          c10::optional<torch::Tensor> bias, long bn);
 bool lt_available();
 void lt_linear(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                c10::optional<torch::Tensor> bias);
@@ -120,7 +120,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("moe_grouped_gemm_mx4", &moe_grouped_gemm_mx4,
           "grouped expert GEMM, MXFP4 packed weights dequantized in-kernel (CDNA4)",
           py::arg("y"), py::arg("x"), py::arg("wq"), py::arg("ws"),
-          py::arg("expert_ids"), py::arg("bias") = py::none());
+          py::arg("expert_ids"), py::arg("bias") = py::none(),
+          py::arg("bn") = 0);
     m.def("lt_available", &lt_available, "hipblaslt usable on this device");
     m.def("lt_linear", &lt_linear, "autotuned hipBLASLt linear (graph-safe)",
           py::arg("y"), py::arg("x"), py::arg("w"), py::arg("bias") = py::none());

@@ -233,7 +233,7 @@ DEV_INLINE void mx4_dequant8(unsigned p, float half_scale, T* out) {
     }
 }
 
-template <typename T>
+template <typename T, int BNT>
 __global__ __launch_bounds__(512, 2) void moe_grouped_mx4_kernel(
     T* __restrict__ y,                   // (Mpad, N)
     const T* __restrict__ x,             // (Mpad, K)
@@ -246,8 +246,10 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_mx4_kernel(
     using vec8 = typename MF::vec8;
     constexpr int MT = 64;
     constexpr int G = 1;                 // glds per wave per chunk (x only)
+    constexpr int NWG = BNT / 64;        // W row groups per lane per chunk
+    constexpr int NFRAG = BNT / 64;
     constexpr int XBYTES = MT * 128;
-    constexpr int CHUNK_B = XBYTES + MG_BN * 128;
+    constexpr int CHUNK_B = XBYTES + BNT * 128;
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
 
@@ -255,24 +257,29 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_mx4_kernel(
     const int rt = blockIdx.y;
     const int e = expert_ids[rt];
     if (e < 0) return;
-    const int n0 = tile * MG_BN;
+    const int n0 = tile * BNT;
     const int m0 = rt * MT;
     const int nchunks = K / MG_BK;
 
     const int lane = threadIdx.x & 63;
     const int wid = threadIdx.x >> 6;
     const int wm0 = (wid >> 2) * 32;
-    const int wn0 = (wid & 3) * 16;
+    const int wn0 = (wid & 3) * (BNT / 4);
     const int lrow = lane & 15;
     const int lslot = lane >> 4;
 
     const char* xsrc = reinterpret_cast<const char*>(x) +
                        ((long)(m0 + wid * 8) * K) * sizeof(T);
-    // this lane stages W row (n0 + wid*8 + lane/8), k-slot lane%8
+    // this lane stages W rows (n0 + g*64 + wid*8 + lane/8), k-slot lane%8
     const int wrow_l = wid * 8 + (lane >> 3);
     const int wslot = lane & 7;          // 8 bf16 = 4 packed bytes
-    const unsigned char* wq_row = wq + ((long)e * N + n0 + wrow_l) * (K / 2);
-    const unsigned char* ws_row = ws + ((long)e * N + n0 + wrow_l) * (K / 32);
+    const unsigned char* wq_row[NWG];
+    const unsigned char* ws_row[NWG];
+#pragma unroll
+    for (int g = 0; g < NWG; ++g) {
+        wq_row[g] = wq + ((long)e * N + n0 + g * 64 + wrow_l) * (K / 2);
+        ws_row[g] = ws + ((long)e * N + n0 + g * 64 + wrow_l) * (K / 32);
+    }
 
     char* xdst[MG_NBUF];
     char* wdst[MG_NBUF];
@@ -291,26 +298,31 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_mx4_kernel(
 #define MX_STAGE_W(t, b)                                                     \
     do {                                                                     \
         const int kc = (t) * MG_BK;                                          \
-        const unsigned pk = *reinterpret_cast<const unsigned*>(              \
-            wq_row + (kc + wslot * 8) / 2);                                  \
-        const float sc = exp2f((int)ws_row[(kc + wslot * 8) / 32] - 127);    \
-        T tmp[8];                                                            \
-        mx4_dequant8<T>(pk, 0.5f * sc, tmp);                                 \
-        const int sl = wslot ^ (lane >> 3);                                  \
-        *reinterpret_cast<int4*>(wdst[b] + wrow_l * 128 + sl * 16) =         \
-            *reinterpret_cast<int4*>(tmp);                                   \
+        _Pragma("unroll") for (int g = 0; g < NWG; ++g) {                    \
+            const unsigned pk = *reinterpret_cast<const unsigned*>(          \
+                wq_row[g] + (kc + wslot * 8) / 2);                           \
+            const float sc =                                                 \
+                exp2f((int)ws_row[g][(kc + wslot * 8) / 32] - 127);          \
+            T tmp[8];                                                        \
+            mx4_dequant8<T>(pk, 0.5f * sc, tmp);                             \
+            const int sl = wslot ^ (lane >> 3);                              \
+            *reinterpret_cast<int4*>(wdst[b] + (g * 64 + wrow_l) * 128 +     \
+                                     sl * 16) = *reinterpret_cast<int4*>(tmp); \
+        }                                                                    \
     } while (0)
 
-    mg_f32x4 acc[2];
-    acc[0] = {0.f, 0.f, 0.f, 0.f};
-    acc[1] = {0.f, 0.f, 0.f, 0.f};
+    mg_f32x4 acc[2][NFRAG];
+#pragma unroll
+    for (int qi = 0; qi < 2; ++qi)
+#pragma unroll
+        for (int nj = 0; nj < NFRAG; ++nj) acc[qi][nj] = {0.f, 0.f, 0.f, 0.f};
 
 #define MX_FRAGS(b)                                                          \
     do {                                                                     \
         const char* xb = smem + (b) * CHUNK_B;                               \
         const char* wb = xb + XBYTES;                                        \
         _Pragma("unroll") for (int ks = 0; ks < 2; ++ks) {                   \
-            vec8 a0, a1, bb;                                                 \
+            vec8 a0, a1, bb[NFRAG];                                          \
             {                                                                \
                 const int m = wm0 + lrow;                                    \
                 const int sfrag = (ks * 4 + lslot) ^ (m & 7);                \
@@ -321,13 +333,15 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_mx4_kernel(
                 const int sfrag = (ks * 4 + lslot) ^ (m & 7);                \
                 a1 = *reinterpret_cast<const vec8*>(xb + m * 128 + sfrag * 16); \
             }                                                                \
-            {                                                                \
-                const int n = wn0 + lrow;                                    \
+            _Pragma("unroll") for (int nj = 0; nj < NFRAG; ++nj) {           \
+                const int n = wn0 + nj * 16 + lrow;                          \
                 const int sfrag = (ks * 4 + lslot) ^ (n & 7);                \
-                bb = *reinterpret_cast<const vec8*>(wb + n * 128 + sfrag * 16); \
+                bb[nj] = *reinterpret_cast<const vec8*>(wb + n * 128 + sfrag * 16); \
             }                                                                \
-            acc[0] = MF::mfma(a0, bb, acc[0]);                               \
-            acc[1] = MF::mfma(a1, bb, acc[1]);                               \
+            _Pragma("unroll") for (int nj = 0; nj < NFRAG; ++nj) {           \
+                acc[0][nj] = MF::mfma(a0, bb[nj], acc[0][nj]);               \
+                acc[1][nj] = MF::mfma(a1, bb[nj], acc[1][nj]);               \
+            }                                                                \
         }                                                                    \
     } while (0)
 
@@ -362,14 +376,19 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_mx4_kernel(
 #undef MX_STAGE_X
     mg_vm_wait<0>();
 
-    const int n_out = n0 + wn0 + lrow;
-    const float bv = bias ? VecTraits<T>::to_float(bias[(long)e * N + n_out]) : 0.f;
 #pragma unroll
-    for (int qi = 0; qi < 2; ++qi) {
+    for (int nj = 0; nj < NFRAG; ++nj) {
+        const int n_out = n0 + wn0 + nj * 16 + lrow;
+        const float bv =
+            bias ? VecTraits<T>::to_float(bias[(long)e * N + n_out]) : 0.f;
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            const int m = m0 + wm0 + qi * 16 + lslot * 4 + r;
-            y[(long)m * N + n_out] = VecTraits<T>::from_float(acc[qi][r] + bv);
+        for (int qi = 0; qi < 2; ++qi) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int m = m0 + wm0 + qi * 16 + lslot * 4 + r;
+                y[(long)m * N + n_out] =
+                    VecTraits<T>::from_float(acc[qi][nj][r] + bv);
+            }
         }
     }
 }
@@ -532,7 +551,7 @@ void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
 
 void moe_grouped_gemm_mx4(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
                           torch::Tensor ws, torch::Tensor expert_ids,
-                          c10::optional<torch::Tensor> bias) {
+                          c10::optional<torch::Tensor> bias, long bn) {
     TORCH_CHECK(y.dim() == 2 && x.dim() == 2 && wq.dim() == 3 && ws.dim() == 3,
                 "shapes");
     TORCH_CHECK(x.is_contiguous() && wq.is_contiguous() && ws.is_contiguous() &&
@@ -542,30 +561,46 @@ void moe_grouped_gemm_mx4(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
     const int N = wq.size(1);
     TORCH_CHECK(wq.size(2) == K / 2 && ws.size(2) == K / 32, "packed shape");
     TORCH_CHECK(y.size(0) == Mpad && y.size(1) == N, "shape");
-    TORCH_CHECK(N % MG_BN == 0 && K % MG_BK == 0 && Mpad % 64 == 0, "align");
+    TORCH_CHECK(K % MG_BK == 0 && Mpad % 64 == 0, "align");
     TORCH_CHECK(wq.scalar_type() == at::ScalarType::Byte &&
                 ws.scalar_type() == at::ScalarType::Byte, "u8 packed");
+    int BNT = (int)bn;
+    if (BNT == 0)
+        BNT = (N % 256 == 0) ? 256
+              : (N % 192 == 0) ? 192
+              : (N % 128 == 0) ? 128 : 64;
+    TORCH_CHECK(BNT == 64 || BNT == 128 || BNT == 192 || BNT == 256,
+                "bn must be 64/128/192/256");
+    TORCH_CHECK(N % BNT == 0, "N must be a multiple of bn");
 
-    constexpr int LDS = MG_NBUF * (64 * 128 + MG_BN * 128);
     auto stream = at::hip::getCurrentHIPStream();
-    dim3 grid(N / MG_BN, Mpad / 64);
+    dim3 grid(N / BNT, Mpad / 64);
 
-#define MGX_T(T)                                                              \
+#define MGX_LAUNCH(T, BNTV)                                                   \
     do {                                                                      \
+        constexpr int LDS = MG_NBUF * (64 * 128 + BNTV * 128);                \
         static bool attr_set = false;                                         \
         if (!attr_set) {                                                      \
-            (void)hipFuncSetAttribute((const void*)moe_grouped_mx4_kernel<T>, \
-                                      hipFuncAttributeMaxDynamicSharedMemorySize, \
-                                      LDS);                                   \
+            (void)hipFuncSetAttribute(                                        \
+                (const void*)moe_grouped_mx4_kernel<T, BNTV>,                 \
+                hipFuncAttributeMaxDynamicSharedMemorySize, LDS);             \
             attr_set = true;                                                  \
         }                                                                     \
-        hipLaunchKernelGGL((moe_grouped_mx4_kernel<T>), grid, dim3(512), LDS, \
-                           stream, (T*)y.data_ptr(), (const T*)x.data_ptr(),  \
+        hipLaunchKernelGGL((moe_grouped_mx4_kernel<T, BNTV>), grid,           \
+                           dim3(512), LDS, stream, (T*)y.data_ptr(),          \
+                           (const T*)x.data_ptr(),                            \
                            (const unsigned char*)wq.data_ptr(),               \
                            (const unsigned char*)ws.data_ptr(),               \
                            expert_ids.data_ptr<int>(),                        \
                            bias.has_value() ? (const T*)bias->data_ptr() : nullptr, \
                            N, K);                                             \
+    } while (0)
+#define MGX_T(T)                                                              \
+    do {                                                                      \
+        if (BNT == 64) MGX_LAUNCH(T, 64);                                     \
+        else if (BNT == 128) MGX_LAUNCH(T, 128);                              \
+        else if (BNT == 192) MGX_LAUNCH(T, 192);                              \
+        else MGX_LAUNCH(T, 256);                                              \
     } while (0)
 
     if (x.scalar_type() == at::ScalarType::BFloat16) {
@@ -576,5 +611,6 @@ void moe_grouped_gemm_mx4(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
         TORCH_CHECK(false, "moe_grouped_gemm_mx4: bf16/f16 only");
     }
 #undef MGX_T
+#undef MGX_LAUNCH
     HIP_CHECK_KERNEL();
 }
