@@ -175,10 +175,24 @@ class ExpertMLPs(nn.Module):
             x_rows = x_rows * gatew.unsqueeze(-1)
         x_sorted = torch.zeros(Mpad, H, dtype=x.dtype, device=x.device)
         x_sorted.index_copy_(0, dest, x_rows)
-        g = ops.moe_grouped_gemm(x_sorted, self.w_gate, expert_ids)
-        u = ops.moe_grouped_gemm(x_sorted, self.w_up, expert_ids)
+
+        # widest column tile that still yields >= ~2 blocks/CU — wide
+        # tiles cut x re-reads 4x but run at occupancy 1, so a small
+        # grid (TP-sharded N or small batch) must fall back to narrower
+        # tiles to fill the chip
+        def _bn(N, ntiles=Mpad // 64):
+            # occupancy is 1 block/CU at bn>=192 (120 KB LDS), 2 below
+            for cand in (256, 192, 128):
+                occ = 1 if cand >= 192 else 2
+                if N % cand == 0 and (N // cand) * ntiles >= 256 * occ:
+                    return cand
+            return 64
+
+        I, Hn = self.w_gate.shape[1], self.w_down.shape[1]
+        g = ops.moe_grouped_gemm(x_sorted, self.w_gate, expert_ids, bn=_bn(I))
+        u = ops.moe_grouped_gemm(x_sorted, self.w_up, expert_ids, bn=_bn(I))
         h = self.act_mul(g, u)
-        y_sorted = ops.moe_grouped_gemm(h, self.w_down, expert_ids)
+        y_sorted = ops.moe_grouped_gemm(h, self.w_down, expert_ids, bn=_bn(Hn))
         contrib = y_sorted.index_select(0, dest)
         if not self.scale_input:
             contrib = contrib * gatew.unsqueeze(-1)

@@ -565,10 +565,11 @@ void moe_grouped_gemm_mx4(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
     TORCH_CHECK(wq.scalar_type() == at::ScalarType::Byte &&
                 ws.scalar_type() == at::ScalarType::Byte, "u8 packed");
     int BNT = (int)bn;
-    if (BNT == 0)
-        BNT = (N % 256 == 0) ? 256
-              : (N % 192 == 0) ? 192
-              : (N % 128 == 0) ? 128 : 64;
+    // auto = 64 for mx4: the register-staged packed-W path pays hipcc's
+    // mixed-load vmcnt drains once per W row-group, so wide tiles LOSE
+    // (measured 531 -> 679 us at gpt-oss T=64 with bn=192); the wide
+    // instantiations stay for explicit use
+    if (BNT == 0) BNT = 64;
     TORCH_CHECK(BNT == 64 || BNT == 128 || BNT == 192 || BNT == 256,
                 "bn must be 64/128/192/256");
     TORCH_CHECK(N % BNT == 0, "N must be a multiple of bn");
