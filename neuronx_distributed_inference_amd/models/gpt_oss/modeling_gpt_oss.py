@@ -156,23 +156,26 @@ class GptOssExperts(nn.Module):
         glu = gate * torch.sigmoid(gate * self.alpha)
         return (up + 1) * glu
 
-    def _gemm(self, x_sorted, which, expert_ids):
+    def _gemm(self, x_in, which, expert_ids, x_gather=None, combine=None):
         from neuronx_distributed_inference_amd import ops as _ops
 
         if which == "gate_up":
             if self.mxfp4:
                 return _ops.moe_grouped_gemm_mx4(
-                    x_sorted, self.gate_up_q, self.gate_up_s, expert_ids,
+                    x_in, self.gate_up_q, self.gate_up_s, expert_ids,
                     bias=self.gate_up_proj_bias)
             return _ops.moe_grouped_gemm(
-                x_sorted, self.gate_up_proj, expert_ids,
-                bias=self.gate_up_proj_bias)
+                x_in, self.gate_up_proj, expert_ids,
+                bias=self.gate_up_proj_bias, x_gather=x_gather)
+        kw = {} if combine is None else dict(
+            combine_out=combine[0], combine_scale=combine[1],
+            combine_gather=combine[2])
         if self.mxfp4:
             return _ops.moe_grouped_gemm_mx4(
-                x_sorted, self.down_q, self.down_s, expert_ids,
+                x_in, self.down_q, self.down_s, expert_ids,
                 bias=self.down_proj_bias)
         return _ops.moe_grouped_gemm(
-            x_sorted, self.down_proj, expert_ids, bias=self.down_proj_bias)
+            x_in, self.down_proj, expert_ids, bias=self.down_proj_bias, **kw)
 
     def _forward_grouped(self, x, weights, idx):
         from neuronx_distributed_inference_amd import ops as _ops
@@ -183,23 +186,36 @@ class GptOssExperts(nn.Module):
         flat = (idx - self._local_offset()).view(-1)
         local = (flat >= 0) & (flat < El)
         flat_routed = torch.where(local, flat, torch.full_like(flat, El))
-        order, dest, expert_ids, Mpad = _ops.moe_grouped_tables(flat_routed, El + 1)
+        order, dest, expert_ids, gtab, Mpad = _ops.moe_grouped_tables(
+            flat_routed, El + 1, top_k)
         expert_ids = torch.where(expert_ids >= El,
                                  torch.full_like(expert_ids, -1), expert_ids)
-        tok = order // top_k
-        gatew = weights.view(-1)[order].to(x.dtype)
+        gatew = weights.view(-1)[order]
         lmask = local[order]
-        x_sorted = torch.zeros(Mpad, H, dtype=x.dtype, device=x.device)
-        x_sorted.index_copy_(0, dest, x.index_select(0, tok))
-        gu = self._gemm(x_sorted, "gate_up", expert_ids)
+        if self.mxfp4:
+            # mx4 kernel has no fused gather: materialize sorted x
+            tok = order // top_k
+            x_in = torch.zeros(Mpad, H, dtype=x.dtype, device=x.device)
+            x_in.index_copy_(0, dest, x.index_select(0, tok))
+            gu = self._gemm(x_in, "gate_up", expert_ids)
+        else:
+            gu = self._gemm(x, "gate_up", expert_ids, x_gather=gtab)
         h = self._glu(gu)
-        y_sorted = self._gemm(h.contiguous(), "down", expert_ids)
-        contrib = y_sorted.index_select(0, dest) * gatew.unsqueeze(-1)
-        contrib = torch.where(lmask.unsqueeze(-1), contrib,
-                              torch.zeros_like(contrib))
-        out = torch.zeros(T, H, dtype=x.dtype, device=x.device)
-        out.index_add_(0, tok, contrib)
-        return out
+        gw_tab = torch.zeros(Mpad, dtype=torch.float32, device=x.device)
+        gw_tab[dest] = torch.where(lmask, gatew.float(),
+                                   torch.zeros_like(gatew, dtype=torch.float32))
+        if self.mxfp4:
+            y_sorted = self._gemm(h.contiguous(), "down", expert_ids)
+            contrib = y_sorted.index_select(0, dest) * gatew.to(x.dtype).unsqueeze(-1)
+            contrib = torch.where(lmask.unsqueeze(-1), contrib,
+                                  torch.zeros_like(contrib))
+            out = torch.zeros(T, H, dtype=x.dtype, device=x.device)
+            out.index_add_(0, order // top_k, contrib)
+            return out
+        out32 = torch.zeros(T, H, dtype=torch.float32, device=x.device)
+        self._gemm(h.contiguous(), "down", expert_ids,
+                   combine=(out32, gw_tab, gtab))
+        return out32.to(x.dtype)
 
     def _dense_w(self, e_local: int, which: str) -> torch.Tensor:
         """[N, K] weight of one local expert (dequantized if mxfp4)."""

@@ -162,19 +162,13 @@ class ExpertMLPs(nn.Module):
         flat = (idx - offset).view(-1)
         local = (flat >= 0) & (flat < El)
         flat_routed = torch.where(local, flat, torch.full_like(flat, El))
-        order, dest, expert_ids, Mpad = ops.moe_grouped_tables(
-            flat_routed, El + 1)
+        order, dest, expert_ids, gtab, Mpad = ops.moe_grouped_tables(
+            flat_routed, El + 1, top_k)
         # sentinel segment (id El) never computes
         expert_ids = torch.where(expert_ids >= El,
                                  torch.full_like(expert_ids, -1), expert_ids)
-        tok = order // top_k
-        gatew = weights.view(-1)[order].to(x.dtype)
+        gatew = weights.view(-1)[order]
         lmask = local[order]
-        x_rows = x.index_select(0, tok)
-        if self.scale_input:
-            x_rows = x_rows * gatew.unsqueeze(-1)
-        x_sorted = torch.zeros(Mpad, H, dtype=x.dtype, device=x.device)
-        x_sorted.index_copy_(0, dest, x_rows)
 
         # widest column tile that still yields >= ~2 blocks/CU — wide
         # tiles cut x re-reads 4x but run at occupancy 1, so a small
@@ -189,19 +183,34 @@ class ExpertMLPs(nn.Module):
             return 64
 
         I, Hn = self.w_gate.shape[1], self.w_down.shape[1]
-        g = ops.moe_grouped_gemm(x_sorted, self.w_gate, expert_ids, bn=_bn(I))
-        u = ops.moe_grouped_gemm(x_sorted, self.w_up, expert_ids, bn=_bn(I))
+        if self.scale_input:
+            # input scaling is nonlinear through the GLU: keep the
+            # materialized gather for this (rare) variant
+            tok = order // top_k
+            x_rows = x.index_select(0, tok) * gatew.to(x.dtype).unsqueeze(-1)
+            x_in = torch.zeros(Mpad, H, dtype=x.dtype, device=x.device)
+            x_in.index_copy_(0, dest, x_rows)
+            xg = None
+        else:
+            x_in = x                      # token gather fuses into staging
+            xg = gtab
+        g = ops.moe_grouped_gemm(x_in, self.w_gate, expert_ids, bn=_bn(I),
+                                 x_gather=xg)
+        u = ops.moe_grouped_gemm(x_in, self.w_up, expert_ids, bn=_bn(I),
+                                 x_gather=xg)
         h = self.act_mul(g, u)
-        y_sorted = ops.moe_grouped_gemm(h, self.w_down, expert_ids, bn=_bn(Hn))
-        contrib = y_sorted.index_select(0, dest)
-        if not self.scale_input:
-            contrib = contrib * gatew.unsqueeze(-1)
-        # skipped tiles hold uninitialized memory: mask, don't multiply
-        contrib = torch.where(lmask.unsqueeze(-1), contrib,
-                              torch.zeros_like(contrib))
-        out = torch.zeros(T, H, dtype=x.dtype, device=x.device)
-        out.index_add_(0, tok, contrib)
-        return out
+        # routing-weighted combine fuses into the down epilogue (fp32
+        # atomics, one rmw per element); non-local assignments are
+        # zero-scaled
+        gw_tab = torch.zeros(Mpad, dtype=torch.float32, device=x.device)
+        scale = gatew.float() if not self.scale_input \
+            else torch.ones_like(gatew, dtype=torch.float32)
+        gw_tab[dest] = torch.where(lmask, scale, torch.zeros_like(scale))
+        out32 = torch.zeros(T, H, dtype=torch.float32, device=x.device)
+        ops.moe_grouped_gemm(h, self.w_down, expert_ids, bn=_bn(Hn),
+                             combine_out=out32, combine_scale=gw_tab,
+                             combine_gather=gtab)
+        return out32.to(x.dtype)
 
     def forward(self, x: torch.Tensor, weights: torch.Tensor, idx: torch.Tensor):
         """x: (T, H); weights/idx: (T, top_k) over GLOBAL expert ids.

@@ -629,13 +629,18 @@ def attention_decode_paged(q, k_cache, v_cache, seq_lens, block_table,
 MOE_TILE = 64
 
 
-def moe_grouped_tables(idx_flat: torch.Tensor, num_experts: int):
+def moe_grouped_tables(idx_flat: torch.Tensor, num_experts: int,
+                       top_k: int = 1):
     """idx_flat: (T*top_k,) local expert id per assignment. Returns
-    (order, dest, expert_ids, Mpad):
-      order      (Ttk,)  argsort by expert (stable)
+    (order, dest, expert_ids, gtab, Mpad):
+      order      (Ttk,)  argsort by expert (stable on CPU; atomic —
+                         arbitrary within-expert permutation — on GPU)
       dest       (Ttk,)  destination row of each sorted assignment in the
                          padded sorted space
       expert_ids (Mpad/64,) int32 expert per row tile, -1 past the end
+      gtab       (Mpad,) int32 padded row -> TOKEN row (order[p]//top_k),
+                         -1 on padding; feeds the fused gather/combine
+                         epilogues of moe_grouped_gemm
       Mpad       static worst-case padded rows (multiple of 64)
     """
     E = num_experts
@@ -651,9 +656,10 @@ def moe_grouped_tables(idx_flat: torch.Tensor, num_experts: int):
         order = torch.empty(Ttk, dtype=torch.long, device=dev)
         dest = torch.empty(Ttk, dtype=torch.long, device=dev)
         expert_ids = torch.empty(Mpad // MT, dtype=torch.int32, device=dev)
-        c.moe_grouped_tables(order, dest, expert_ids,
-                             idx_flat.to(torch.int32).contiguous(), E)
-        return order, dest, expert_ids, Mpad
+        gtab = torch.empty(Mpad, dtype=torch.int32, device=dev)
+        c.moe_grouped_tables(order, dest, expert_ids, gtab,
+                             idx_flat.to(torch.int32).contiguous(), E, top_k)
+        return order, dest, expert_ids, gtab, Mpad
     order = torch.argsort(idx_flat, stable=True)
     sorted_e = idx_flat[order]
     # scatter_add, NOT bincount: bincount syncs the host on CUDA (output
@@ -671,36 +677,71 @@ def moe_grouped_tables(idx_flat: torch.Tensor, num_experts: int):
     total_pad = ends_pad[-1] if E > 0 else torch.zeros((), device=dev)
     expert_ids = torch.where(tile_start < total_pad, eid,
                              torch.full_like(eid, -1)).to(torch.int32)
-    return order, dest, expert_ids, Mpad
+    gtab = torch.full((Mpad,), -1, dtype=torch.int32, device=dev)
+    gtab[dest] = (order // top_k).to(torch.int32)
+    return order, dest, expert_ids, gtab, Mpad
 
 
 def moe_grouped_gemm(x_sorted: torch.Tensor, w: torch.Tensor,
                      expert_ids: torch.Tensor, bias=None,
-                     bn: int = 0) -> torch.Tensor:
+                     bn: int = 0, x_gather=None, combine_out=None,
+                     combine_scale=None, combine_gather=None) -> torch.Tensor:
     """y_sorted[m] = x_sorted[m] @ w[expert_of(m)]^T (+ bias[expert]).
     x_sorted (Mpad, K), w (E, N, K), expert_ids (Mpad/64,) int32, bias
     (E, N) or None. GPU: one grouped HIP kernel; CPU reference:
-    per-tile loop."""
-    Mpad, K = x_sorted.shape
+    per-tile loop.
+
+    Fusions (GPU + CPU ref):
+      x_gather (Mpad,) int32: x_sorted is instead the UNSORTED (T, K)
+        activations; each padded row stages x[x_gather[m]] (the token
+        gather fuses into the glds staging).
+      combine_out (T, N) fp32 + combine_gather (Mpad,) [+ combine_scale
+        (Mpad,) fp32]: instead of writing y, atomically accumulate
+        (acc + bias) * scale into combine_out[token] — the routing-
+        weighted combine fuses into the epilogue. Returns combine_out.
+    """
+    K = x_sorted.shape[1]
+    Mpad = expert_ids.numel() * MOE_TILE if x_gather is not None \
+        else x_sorted.shape[0]
     N = w.shape[1]
     if x_sorted.is_cuda:
         c = _require_ext("moe_grouped_gemm")
+        # y is the kernel's nominal output; untouched when combining
         y = torch.empty(Mpad, N, dtype=x_sorted.dtype, device=x_sorted.device)
         c.moe_grouped_gemm(y, x_sorted.contiguous(), w.contiguous(),
                            expert_ids.contiguous(),
-                           None if bias is None else bias.contiguous(), bn)
-        return y
+                           None if bias is None else bias.contiguous(), bn,
+                           None if x_gather is None else x_gather.contiguous(),
+                           combine_out,
+                           combine_scale,
+                           None if combine_gather is None
+                           else combine_gather.contiguous())
+        return combine_out if combine_out is not None else y
+    # ---- CPU reference ----
+    out = combine_out
     y = torch.zeros(Mpad, N, dtype=x_sorted.dtype)
     for t in range(Mpad // MOE_TILE):
         e = int(expert_ids[t])
         if e < 0:
             continue
         rows = slice(t * MOE_TILE, (t + 1) * MOE_TILE)
-        r = x_sorted[rows] @ w[e].t().to(x_sorted.dtype)
+        if x_gather is not None:
+            xg = x_sorted[x_gather[rows].clamp(min=0).long()]
+        else:
+            xg = x_sorted[rows]
+        r = xg @ w[e].t().to(x_sorted.dtype)
         if bias is not None:
             r = r + bias[e].to(r.dtype)
-        y[rows] = r
-    return y
+        if out is not None:
+            cg = combine_gather[rows].long()
+            sc = (combine_scale[rows] if combine_scale is not None
+                  else torch.ones(MOE_TILE))
+            for j in range(MOE_TILE):
+                if cg[j] >= 0:
+                    out[cg[j]] += r[j].float() * float(sc[j])
+        else:
+            y[rows] = r
+    return out if out is not None else y
 
 
 # ----------------------------------------------------------------------

@@ -49,10 +49,15 @@ void fp8_epilogue(torch::Tensor out, torch::Tensor y32, torch::Tensor w_scale,
                   c10::optional<torch::Tensor> bias);
 void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                       torch::Tensor expert_ids,
-                      c10::optional<torch::Tensor> bias, long bn);
+                      c10::optional<torch::Tensor> bias, long bn,
+                      c10::optional<torch::Tensor> x_gather,
+                      c10::optional<torch::Tensor> combine_out,
+                      c10::optional<torch::Tensor> combine_scale,
+                      c10::optional<torch::Tensor> combine_gather);
 void moe_grouped_tables_gpu(torch::Tensor order, torch::Tensor dest,
-                            torch::Tensor expert_ids, torch::Tensor idx_flat,
-                            long num_experts);
+                            torch::Tensor expert_ids, torch::Tensor gtab,
+                            torch::Tensor idx_flat, long num_experts,
+                            long topk);
 void moe_grouped_gemm_mx4(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
                           torch::Tensor ws, torch::Tensor expert_ids,
                           c10::optional<torch::Tensor> bias, long bn);
@@ -112,11 +117,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("moe_grouped_tables", &moe_grouped_tables_gpu,
           "fused MoE segment tables: one launch vs the torch op chain",
           py::arg("order"), py::arg("dest"), py::arg("expert_ids"),
-          py::arg("idx_flat"), py::arg("num_experts"));
+          py::arg("gtab"), py::arg("idx_flat"), py::arg("num_experts"),
+          py::arg("topk"));
     m.def("moe_grouped_gemm", &moe_grouped_gemm,
           "grouped expert GEMM over sorted padded token segments (CDNA4)",
           py::arg("y"), py::arg("x"), py::arg("w"), py::arg("expert_ids"),
-          py::arg("bias") = py::none(), py::arg("bn") = 0);
+          py::arg("bias") = py::none(), py::arg("bn") = 0,
+          py::arg("x_gather") = py::none(),
+          py::arg("combine_out") = py::none(),
+          py::arg("combine_scale") = py::none(),
+          py::arg("combine_gather") = py::none());
     m.def("moe_grouped_gemm_mx4", &moe_grouped_gemm_mx4,
           "grouped expert GEMM, MXFP4 packed weights dequantized in-kernel (CDNA4)",
           py::arg("y"), py::arg("x"), py::arg("wq"), py::arg("ws"),

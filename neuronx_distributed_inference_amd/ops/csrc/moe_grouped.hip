@@ -66,6 +66,15 @@ DEV_INLINE void mg_vm_wait() {
     asm volatile("s_waitcnt vmcnt(%c0)" ::"i"(N) : "memory");
 }
 
+// per-lane-addressed glds (row gather): the caller precomputes each
+// lane's full source address (row via an index table + XOR slot); the
+// LDS dest stays the wave-uniform lane-linear base.
+DEV_INLINE void mg_glds_lane128(const char* src_lane, char* lds_dst_base) {
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src_lane,
+        (__attribute__((address_space(3))) unsigned int*)lds_dst_base, 16, 0, 0);
+}
+
 // MT=64 row tiles, BNT-wide column tiles (8 waves as 2(M) x 4(N): each
 // wave = 32 rows x BNT/4 cols). Wider BNT cuts the x re-read traffic
 // (x is re-read once per column tile) and raises MFMA work per staged
@@ -74,10 +83,14 @@ DEV_INLINE void mg_vm_wait() {
 template <typename T, int BNT>
 __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
     T* __restrict__ y,              // (Mpad, N) sorted-space output
-    const T* __restrict__ x,        // (Mpad, K) sorted-space input
+    const T* __restrict__ x,        // (Mpad, K) sorted — or (T, K) with x_gather
     const T* __restrict__ w,        // (E, N, K) stacked expert weights
     const int* __restrict__ expert_ids,  // (n_row_tiles,) expert or -1
     const T* __restrict__ bias,     // (E, N) per-expert bias or null
+    const int* __restrict__ x_gather,    // (Mpad,) padded row -> x row, or null
+    float* __restrict__ combine_out,     // (T, N) fp32 accum, or null
+    const float* __restrict__ combine_scale,  // (Mpad,) routing weight
+    const int* __restrict__ combine_gather,   // (Mpad,) padded row -> token
     const int N, const int K) {
     using MF = MgMfma<T>;
     using vec8 = typename MF::vec8;
@@ -105,8 +118,13 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
     const int lrow = lane & 15;
     const int lslot = lane >> 4;
 
-    const char* xsrc = reinterpret_cast<const char*>(x) +
-                       ((long)(m0 + wid * 8) * K) * sizeof(T);
+    // per-lane x source: row via gather table (fused token gather) or
+    // linear sorted row; XOR source-swizzle baked into the address
+    const int xr_lin = m0 + wid * 8 + (lane >> 3);
+    const int xr = x_gather ? max(x_gather[xr_lin], 0) : xr_lin;
+    const int xslot = (lane & 7) ^ (lane >> 3);
+    const char* xsrc_lane = reinterpret_cast<const char*>(x) +
+                            (long)xr * K * sizeof(T) + xslot * 16;
     const char* wsrc[NWG];
 #pragma unroll
     for (int i = 0; i < NWG; ++i)
@@ -124,7 +142,7 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
 #define MG_STAGE(t, b)                                                       \
     do {                                                                     \
         const long koff = (long)(t) * 128;                                   \
-        mg_glds_rows128<0>(xsrc + koff, (long)K * sizeof(T), xdst[b]);       \
+        mg_glds_lane128(xsrc_lane + koff, xdst[b]);                          \
         _Pragma("unroll") for (int i = 0; i < NWG; ++i)                      \
             mg_glds_rows128<2>(wsrc[i] + koff, (long)K * sizeof(T),          \
                                wdst[b] + i * 1024);                          \
@@ -191,6 +209,9 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
     mg_vm_wait<0>();
 
     // epilogue: y[m0 + wm0 + qi*16 + lslot*4 + r][n0 + wn0 + nj*16 + lrow]
+    // — or, fused combine: atomically accumulate the routing-weighted
+    // contribution straight into the (T, N) fp32 output (one rmw per
+    // element, NOT per k-chunk — the v2 atomics lesson)
 #pragma unroll
     for (int nj = 0; nj < NFRAG; ++nj) {
         const int n_out = n0 + wn0 + nj * 16 + lrow;
@@ -201,8 +222,18 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 const int m = m0 + wm0 + qi * 16 + lslot * 4 + r;
-                y[(long)m * N + n_out] =
-                    VecTraits<T>::from_float(acc[qi][nj][r] + bv);
+                if (combine_out) {
+                    const int tk = combine_gather[m];
+                    if (tk >= 0) {
+                        const float sc =
+                            combine_scale ? combine_scale[m] : 1.f;
+                        atomicAdd(&combine_out[(long)tk * N + n_out],
+                                  (acc[qi][nj][r] + bv) * sc);
+                    }
+                } else {
+                    y[(long)m * N + n_out] =
+                        VecTraits<T>::from_float(acc[qi][nj][r] + bv);
+                }
             }
         }
     }
@@ -409,8 +440,9 @@ __global__ __launch_bounds__(MT_TABLES, 1) void moe_tables_kernel(
     long* __restrict__ order,        // (Ttk,) sorted pos -> assignment idx
     long* __restrict__ dest,         // (Ttk,) sorted pos -> padded row
     int* __restrict__ expert_ids,    // (ntiles,) expert per 64-row tile
+    int* __restrict__ gtab,          // (ntiles*64,) padded row -> token, -1 pad
     const int* __restrict__ idx,     // (Ttk,) expert per assignment
-    const int Ttk, const int E, const int ntiles) {
+    const int Ttk, const int E, const int ntiles, const int topk) {
     extern __shared__ int tsm[];
     int* counts = tsm;               // (E,)
     int* starts_raw = tsm + E;       // (E,) exclusive scan of counts
@@ -436,13 +468,17 @@ __global__ __launch_bounds__(MT_TABLES, 1) void moe_tables_kernel(
         }
         total_pad = pad;
     }
+    for (int m = threadIdx.x; m < ntiles * 64; m += blockDim.x)
+        gtab[m] = -1;                             // padding rows
     __syncthreads();
     for (int i = threadIdx.x; i < Ttk; i += blockDim.x) {
         const int e = idx[i];
         const int r = atomicAdd(&cursor[e], 1);   // within-expert rank
         const int p = starts_raw[e] + r;          // sorted position
         order[p] = i;
-        dest[p] = starts_pad[e] + r;
+        const int d = starts_pad[e] + r;
+        dest[p] = d;
+        gtab[d] = i / topk;                       // padded row -> token row
     }
     // expert per row tile: linear search over E (ntiles and E are small)
     for (int t = threadIdx.x; t < ntiles; t += blockDim.x) {
@@ -465,8 +501,9 @@ __global__ __launch_bounds__(MT_TABLES, 1) void moe_tables_kernel(
 }  // namespace
 
 void moe_grouped_tables_gpu(torch::Tensor order, torch::Tensor dest,
-                            torch::Tensor expert_ids, torch::Tensor idx_flat,
-                            long num_experts) {
+                            torch::Tensor expert_ids, torch::Tensor gtab,
+                            torch::Tensor idx_flat, long num_experts,
+                            long topk) {
     const int Ttk = idx_flat.numel();
     const int E = (int)num_experts;
     const int ntiles = expert_ids.numel();
@@ -474,31 +511,57 @@ void moe_grouped_tables_gpu(torch::Tensor order, torch::Tensor dest,
                 idx_flat.is_contiguous(), "idx_flat int32 contiguous");
     TORCH_CHECK(order.scalar_type() == at::ScalarType::Long &&
                 dest.scalar_type() == at::ScalarType::Long &&
-                expert_ids.scalar_type() == at::ScalarType::Int, "out dtypes");
-    TORCH_CHECK(order.numel() == Ttk && dest.numel() == Ttk, "out sizes");
+                expert_ids.scalar_type() == at::ScalarType::Int &&
+                gtab.scalar_type() == at::ScalarType::Int, "out dtypes");
+    TORCH_CHECK(order.numel() == Ttk && dest.numel() == Ttk &&
+                gtab.numel() == ntiles * 64, "out sizes");
     TORCH_CHECK(E <= 4096, "num_experts too large for LDS tables");
     const int lds = 4 * E * (int)sizeof(int);
     auto stream = at::hip::getCurrentHIPStream();
     hipLaunchKernelGGL(moe_tables_kernel, dim3(1), dim3(MT_TABLES), lds,
                        stream, order.data_ptr<long>(), dest.data_ptr<long>(),
-                       expert_ids.data_ptr<int>(), idx_flat.data_ptr<int>(),
-                       Ttk, E, ntiles);
+                       expert_ids.data_ptr<int>(), gtab.data_ptr<int>(),
+                       idx_flat.data_ptr<int>(), Ttk, E, ntiles, (int)topk);
     HIP_CHECK_KERNEL();
 }
 
 void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                       torch::Tensor expert_ids,
-                      c10::optional<torch::Tensor> bias, long bn) {
+                      c10::optional<torch::Tensor> bias, long bn,
+                      c10::optional<torch::Tensor> x_gather,
+                      c10::optional<torch::Tensor> combine_out,
+                      c10::optional<torch::Tensor> combine_scale,
+                      c10::optional<torch::Tensor> combine_gather) {
     TORCH_CHECK(y.dim() == 2 && x.dim() == 2 && w.dim() == 3, "shapes");
     TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && y.is_contiguous() &&
                 expert_ids.is_contiguous(), "contiguous tensors required");
-    const int Mpad = x.size(0), K = x.size(1);
+    const int K = x.size(1);
     const int N = w.size(1);
+    // with a gather table, x holds the UNSORTED tokens; Mpad comes from
+    // the tile table
+    const int Mpad = x_gather.has_value() ? (int)expert_ids.numel() * 64
+                                          : (int)x.size(0);
     TORCH_CHECK(w.size(2) == K && y.size(0) == Mpad && y.size(1) == N, "shape");
     TORCH_CHECK(K % MG_BK == 0, "K must be a multiple of 64");
     TORCH_CHECK(Mpad % 64 == 0, "padded rows must be a multiple of 64");
     TORCH_CHECK(expert_ids.scalar_type() == at::ScalarType::Int &&
                 expert_ids.numel() == Mpad / 64, "expert_ids must be int (Mpad/64,)");
+    if (x_gather.has_value())
+        TORCH_CHECK(x_gather->scalar_type() == at::ScalarType::Int &&
+                    x_gather->numel() == Mpad && x_gather->is_contiguous(),
+                    "x_gather must be int32 (Mpad,)");
+    if (combine_out.has_value()) {
+        TORCH_CHECK(combine_gather.has_value(),
+                    "combine_out needs combine_gather");
+        TORCH_CHECK(combine_out->scalar_type() == at::ScalarType::Float &&
+                    combine_out->size(1) == N && combine_out->is_contiguous(),
+                    "combine_out must be fp32 (T, N)");
+        TORCH_CHECK(combine_gather->scalar_type() == at::ScalarType::Int &&
+                    combine_gather->numel() == Mpad, "combine_gather (Mpad,)");
+        if (combine_scale.has_value())
+            TORCH_CHECK(combine_scale->scalar_type() == at::ScalarType::Float &&
+                        combine_scale->numel() == Mpad, "combine_scale (Mpad,)");
+    }
     // bn: column-tile width. 0 = auto (widest tile that divides N).
     int BNT = (int)bn;
     if (BNT == 0)
@@ -527,6 +590,13 @@ void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                            (const T*)x.data_ptr(),                            \
                            (const T*)w.data_ptr(), expert_ids.data_ptr<int>(),\
                            bias ? (const T*)bias->data_ptr() : nullptr,       \
+                           x_gather ? x_gather->data_ptr<int>() : nullptr,    \
+                           combine_out ? combine_out->data_ptr<float>()       \
+                                       : nullptr,                             \
+                           combine_scale ? combine_scale->data_ptr<float>()   \
+                                         : nullptr,                           \
+                           combine_gather ? combine_gather->data_ptr<int>()   \
+                                          : nullptr,                          \
                            N, K);                                             \
     } while (0)
 #define MG_T(T)                                                               \

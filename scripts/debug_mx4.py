@@ -16,7 +16,7 @@ def check(E, N, K, Ttk, bias=False):
     wq, ws = ops.mx4_quantize(w)
     b = (torch.randn(E, N, dtype=torch.bfloat16) * 0.1) if bias else None
     idx = torch.randint(0, E, (Ttk,))
-    order, dest, expert_ids, Mpad = ops.moe_grouped_tables(idx, E)
+    order, dest, expert_ids, gtab, Mpad = ops.moe_grouped_tables(idx, E)
     x_rows = torch.randn(Ttk, K, dtype=torch.bfloat16) * 0.1
     x_sorted = torch.zeros(Mpad, K, dtype=torch.bfloat16)
     x_sorted.index_copy_(0, dest, x_rows)

@@ -41,7 +41,7 @@ def main():
 
     flat = idx.view(-1)
     print(f"tables(fused): {t(lambda: ops.moe_grouped_tables(flat, E)):8.1f} us")
-    order, dest, expert_ids, Mpad = ops.moe_grouped_tables(flat, E)
+    order, dest, expert_ids, gtab, Mpad = ops.moe_grouped_tables(flat, E)
     tok = order // topk
     x_sorted = torch.zeros(Mpad, H, dtype=torch.bfloat16, device=dev)
 

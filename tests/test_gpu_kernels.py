@@ -834,7 +834,7 @@ def test_moe_grouped_gemm_kernel():
     dev = _dev()
     E, N, K = 4, 256, 512
     idx = torch.randint(0, E, (100,), device=dev)
-    order, dest, expert_ids, Mpad = ops.moe_grouped_tables(idx, E)
+    order, dest, expert_ids, gtab, Mpad = ops.moe_grouped_tables(idx, E)
     x_rows = torch.randn(100, K, dtype=torch.bfloat16, device=dev) * 0.1
     w = torch.randn(E, N, K, dtype=torch.bfloat16, device=dev) * 0.1
     x_sorted = torch.zeros(Mpad, K, dtype=torch.bfloat16, device=dev)
@@ -896,7 +896,7 @@ def test_moe_grouped_mx4_kernel_gpu():
     wq, ws = ops.mx4_quantize(w)
     bias = torch.randn(E, N, dtype=torch.bfloat16) * 0.1
     idx = torch.randint(0, E, (90,))
-    order, dest, expert_ids, Mpad = ops.moe_grouped_tables(idx, E)
+    order, dest, expert_ids, gtab, Mpad = ops.moe_grouped_tables(idx, E)
     x_rows = torch.randn(90, K, dtype=torch.bfloat16) * 0.1
     x_sorted = torch.zeros(Mpad, K, dtype=torch.bfloat16)
     x_sorted.index_copy_(0, dest, x_rows)
@@ -1030,8 +1030,8 @@ def test_moe_tables_fused_matches_torch_chain():
     dev = _dev()
     for E, Ttk in ((8, 512), (32, 100), (4, 64), (128, 4096)):
         idx = torch.randint(0, E, (Ttk,), device=dev)
-        order_g, dest_g, eids_g, Mpad_g = ops.moe_grouped_tables(idx, E)
-        order_c, dest_c, eids_c, Mpad_c = ops.moe_grouped_tables(idx.cpu(), E)
+        order_g, dest_g, eids_g, gtab_g, Mpad_g = ops.moe_grouped_tables(idx, E)
+        order_c, dest_c, eids_c, gtab_c, Mpad_c = ops.moe_grouped_tables(idx.cpu(), E)
         assert Mpad_g == Mpad_c
         assert torch.equal(eids_g.cpu(), eids_c)
         og, dg = order_g.cpu(), dest_g.cpu()
@@ -1058,7 +1058,7 @@ def test_moe_grouped_gemm_bn_tiles(bn):
     dev = _dev()
     E, N, K = 4, 768, 512
     idx = torch.randint(0, E, (200,), device=dev)
-    order, dest, expert_ids, Mpad = ops.moe_grouped_tables(idx, E)
+    order, dest, expert_ids, gtab, Mpad = ops.moe_grouped_tables(idx, E)
     x_rows = torch.randn(200, K, dtype=torch.bfloat16, device=dev) * 0.1
     w = torch.randn(E, N, K, dtype=torch.bfloat16, device=dev) * 0.1
     x_sorted = torch.zeros(Mpad, K, dtype=torch.bfloat16, device=dev)
@@ -1071,3 +1071,36 @@ def test_moe_grouped_gemm_bn_tiles(bn):
         m = se == e
         want[m] = x_sorted[dest[m]] @ w[e].t()
     torch.testing.assert_close(got.float(), want.float(), atol=2e-2, rtol=2e-2)
+
+
+def test_moe_grouped_gemm_fusions_gpu():
+    """Fused x-gather + fused combine epilogue vs the materialized
+    pipeline on GPU."""
+    torch.manual_seed(5)
+    dev = _dev()
+    E, T, top_k, K, N = 8, 200, 2, 512, 768
+    idx = torch.randint(0, E, (T * top_k,), device=dev)
+    order, dest, eids, gtab, Mpad = ops.moe_grouped_tables(idx, E, top_k)
+    x = torch.randn(T, K, dtype=torch.bfloat16, device=dev) * 0.1
+    w = torch.randn(E, N, K, dtype=torch.bfloat16, device=dev) * 0.1
+    gatew = torch.rand(T * top_k, device=dev)[order]
+
+    tok = order // top_k
+    xs = torch.zeros(Mpad, K, dtype=torch.bfloat16, device=dev)
+    xs.index_copy_(0, dest, x.index_select(0, tok))
+    y = ops.moe_grouped_gemm(xs, w, eids)
+    contrib = y.index_select(0, dest) * gatew.to(y.dtype).unsqueeze(-1)
+    want = torch.zeros(T, N, dtype=torch.bfloat16, device=dev)
+    want.index_add_(0, tok, contrib)
+
+    y2 = ops.moe_grouped_gemm(x, w, eids, x_gather=gtab)
+    torch.testing.assert_close(y2.index_select(0, dest).float(),
+                               y.index_select(0, dest).float(),
+                               atol=2e-2, rtol=2e-2)
+    gw_tab = torch.zeros(Mpad, dtype=torch.float32, device=dev)
+    gw_tab[dest] = gatew.float()
+    out32 = torch.zeros(T, N, dtype=torch.float32, device=dev)
+    got = ops.moe_grouped_gemm(xs, w, eids, combine_out=out32,
+                               combine_scale=gw_tab, combine_gather=gtab)
+    torch.testing.assert_close(got.to(torch.bfloat16).float(), want.float(),
+                               atol=5e-2, rtol=5e-2)

@@ -373,7 +373,7 @@ def test_moe_grouped_tables():
     torch.manual_seed(0)
     E, Ttk = 5, 37
     idx = torch.randint(0, E, (Ttk,))
-    order, dest, expert_ids, Mpad = ops.moe_grouped_tables(idx, E)
+    order, dest, expert_ids, gtab, Mpad = ops.moe_grouped_tables(idx, E)
     assert Mpad % 64 == 0 and dest.max() < Mpad
     # each sorted assignment's destination tile belongs to its expert
     sorted_e = idx[order]
@@ -589,3 +589,39 @@ def test_tensor_capture_replacement_configs():
     # layer 1 received zeros -> its output differs from the unreplaced run
     assert not torch.allclose(cap2.captured["layers.0"][0],
                               golden)
+
+
+def test_moe_grouped_gemm_fusions_cpu_ref():
+    """Fused x-gather staging + fused routing-weighted combine epilogue
+    must equal the materialized gather + index_add combine (CPU ref)."""
+    import torch
+
+    from neuronx_distributed_inference_amd import ops
+
+    torch.manual_seed(4)
+    E, T, top_k, K, N = 4, 50, 2, 64, 128
+    idx = torch.randint(0, E, (T * top_k,))
+    order, dest, eids, gtab, Mpad = ops.moe_grouped_tables(idx, E, top_k)
+    x = torch.randn(T, K)
+    w = torch.randn(E, N, K) * 0.1
+    gatew = torch.rand(T * top_k)[order]
+
+    # reference: materialized pipeline
+    tok = order // top_k
+    xs = torch.zeros(Mpad, K)
+    xs.index_copy_(0, dest, x.index_select(0, tok))
+    y = ops.moe_grouped_gemm(xs, w, eids)
+    contrib = y.index_select(0, dest) * gatew.unsqueeze(-1)
+    want = torch.zeros(T, N)
+    want.index_add_(0, tok, contrib)
+
+    # fused: gather in, combine out
+    y2 = ops.moe_grouped_gemm(x, w, eids, x_gather=gtab)
+    torch.testing.assert_close(y2.index_select(0, dest),
+                               y.index_select(0, dest))
+    gw_tab = torch.zeros(Mpad)
+    gw_tab[dest] = gatew
+    out32 = torch.zeros(T, N)
+    got = ops.moe_grouped_gemm(xs, w, eids, combine_out=out32,
+                               combine_scale=gw_tab, combine_gather=gtab)
+    torch.testing.assert_close(got, want, atol=1e-4, rtol=1e-4)
