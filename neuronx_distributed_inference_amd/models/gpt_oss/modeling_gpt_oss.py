@@ -159,21 +159,21 @@ class GptOssExperts(nn.Module):
     def _gemm(self, x_in, which, expert_ids, x_gather=None, combine=None):
         from neuronx_distributed_inference_amd import ops as _ops
 
+        kw = {} if combine is None else dict(
+            combine_out=combine[0], combine_scale=combine[1],
+            combine_gather=combine[2])
         if which == "gate_up":
             if self.mxfp4:
                 return _ops.moe_grouped_gemm_mx4(
                     x_in, self.gate_up_q, self.gate_up_s, expert_ids,
-                    bias=self.gate_up_proj_bias)
+                    bias=self.gate_up_proj_bias, x_gather=x_gather)
             return _ops.moe_grouped_gemm(
                 x_in, self.gate_up_proj, expert_ids,
                 bias=self.gate_up_proj_bias, x_gather=x_gather)
-        kw = {} if combine is None else dict(
-            combine_out=combine[0], combine_scale=combine[1],
-            combine_gather=combine[2])
         if self.mxfp4:
             return _ops.moe_grouped_gemm_mx4(
                 x_in, self.down_q, self.down_s, expert_ids,
-                bias=self.down_proj_bias)
+                bias=self.down_proj_bias, **kw)
         return _ops.moe_grouped_gemm(
             x_in, self.down_proj, expert_ids, bias=self.down_proj_bias, **kw)
 
@@ -192,26 +192,11 @@ class GptOssExperts(nn.Module):
                                  torch.full_like(expert_ids, -1), expert_ids)
         gatew = weights.view(-1)[order]
         lmask = local[order]
-        if self.mxfp4:
-            # mx4 kernel has no fused gather: materialize sorted x
-            tok = order // top_k
-            x_in = torch.zeros(Mpad, H, dtype=x.dtype, device=x.device)
-            x_in.index_copy_(0, dest, x.index_select(0, tok))
-            gu = self._gemm(x_in, "gate_up", expert_ids)
-        else:
-            gu = self._gemm(x, "gate_up", expert_ids, x_gather=gtab)
+        gu = self._gemm(x, "gate_up", expert_ids, x_gather=gtab)
         h = self._glu(gu)
         gw_tab = torch.zeros(Mpad, dtype=torch.float32, device=x.device)
         gw_tab[dest] = torch.where(lmask, gatew.float(),
                                    torch.zeros_like(gatew, dtype=torch.float32))
-        if self.mxfp4:
-            y_sorted = self._gemm(h.contiguous(), "down", expert_ids)
-            contrib = y_sorted.index_select(0, dest) * gatew.to(x.dtype).unsqueeze(-1)
-            contrib = torch.where(lmask.unsqueeze(-1), contrib,
-                                  torch.zeros_like(contrib))
-            out = torch.zeros(T, H, dtype=x.dtype, device=x.device)
-            out.index_add_(0, order // top_k, contrib)
-            return out
         out32 = torch.zeros(T, H, dtype=torch.float32, device=x.device)
         self._gemm(h.contiguous(), "down", expert_ids,
                    combine=(out32, gw_tab, gtab))

@@ -792,28 +792,49 @@ def mx4_quantize(w: torch.Tensor):
 
 def moe_grouped_gemm_mx4(x_sorted: torch.Tensor, wq: torch.Tensor,
                          ws: torch.Tensor, expert_ids: torch.Tensor,
-                         bias=None, bn: int = 0) -> torch.Tensor:
+                         bias=None, bn: int = 0, x_gather=None,
+                         combine_out=None, combine_scale=None,
+                         combine_gather=None) -> torch.Tensor:
     """Grouped expert GEMM with MXFP4-packed weights (E, N, K/2)+(E, N,
     K/32), dequantized in-kernel during LDS staging (~3.8x less weight
     traffic than bf16). CPU: dequant reference."""
-    Mpad, K = x_sorted.shape
+    K = x_sorted.shape[1]
+    Mpad = expert_ids.numel() * MOE_TILE if x_gather is not None \
+        else x_sorted.shape[0]
     N = wq.shape[1]
     if x_sorted.is_cuda:
         c = _require_ext("moe_grouped_gemm_mx4")
         y = torch.empty(Mpad, N, dtype=x_sorted.dtype, device=x_sorted.device)
         c.moe_grouped_gemm_mx4(y, x_sorted.contiguous(), wq.contiguous(),
                                ws.contiguous(), expert_ids.contiguous(),
-                               None if bias is None else bias.contiguous(), bn)
-        return y
+                               None if bias is None else bias.contiguous(), bn,
+                               None if x_gather is None else x_gather.contiguous(),
+                               combine_out, combine_scale,
+                               None if combine_gather is None
+                               else combine_gather.contiguous())
+        return combine_out if combine_out is not None else y
+    out = combine_out
     y = torch.zeros(Mpad, N, dtype=x_sorted.dtype)
     for t in range(Mpad // MOE_TILE):
         e = int(expert_ids[t])
         if e < 0:
             continue
         rows = slice(t * MOE_TILE, (t + 1) * MOE_TILE)
+        if x_gather is not None:
+            xg = x_sorted[x_gather[rows].clamp(min=0).long()]
+        else:
+            xg = x_sorted[rows]
         wd = mx4_dequant(wq[e], ws[e]).to(x_sorted.dtype)
-        r = x_sorted[rows] @ wd.t()
+        r = xg @ wd.t()
         if bias is not None:
             r = r + bias[e].to(r.dtype)
-        y[rows] = r
-    return y
+        if out is not None:
+            cg = combine_gather[rows].long()
+            sc = (combine_scale[rows] if combine_scale is not None
+                  else torch.ones(MOE_TILE))
+            for j in range(MOE_TILE):
+                if cg[j] >= 0:
+                    out[cg[j]] += r[j].float() * float(sc[j])
+        else:
+            y[rows] = r
+    return out if out is not None else y

@@ -60,7 +60,11 @@ void moe_grouped_tables_gpu(torch::Tensor order, torch::Tensor dest,
                             long topk);
 void moe_grouped_gemm_mx4(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
                           torch::Tensor ws, torch::Tensor expert_ids,
-                          c10::optional<torch::Tensor> bias, long bn);
+                          c10::optional<torch::Tensor> bias, long bn,
+                          c10::optional<torch::Tensor> x_gather,
+                          c10::optional<torch::Tensor> combine_out,
+                          c10::optional<torch::Tensor> combine_scale,
+                          c10::optional<torch::Tensor> combine_gather);
 bool lt_available();
 void lt_linear(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                c10::optional<torch::Tensor> bias);
@@ -131,7 +135,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "grouped expert GEMM, MXFP4 packed weights dequantized in-kernel (CDNA4)",
           py::arg("y"), py::arg("x"), py::arg("wq"), py::arg("ws"),
           py::arg("expert_ids"), py::arg("bias") = py::none(),
-          py::arg("bn") = 0);
+          py::arg("bn") = 0, py::arg("x_gather") = py::none(),
+          py::arg("combine_out") = py::none(),
+          py::arg("combine_scale") = py::none(),
+          py::arg("combine_gather") = py::none());
     m.def("lt_available", &lt_available, "hipblaslt usable on this device");
     m.def("lt_linear", &lt_linear, "autotuned hipBLASLt linear (graph-safe)",
           py::arg("y"), py::arg("x"), py::arg("w"), py::arg("bias") = py::none());

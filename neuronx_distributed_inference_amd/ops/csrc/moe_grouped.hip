@@ -267,11 +267,15 @@ DEV_INLINE void mx4_dequant8(unsigned p, float half_scale, T* out) {
 template <typename T, int BNT>
 __global__ __launch_bounds__(512, 2) void moe_grouped_mx4_kernel(
     T* __restrict__ y,                   // (Mpad, N)
-    const T* __restrict__ x,             // (Mpad, K)
+    const T* __restrict__ x,             // (Mpad, K) — or (T, K) w/ x_gather
     const unsigned char* __restrict__ wq,  // (E, N, K/2) packed e2m1
     const unsigned char* __restrict__ ws,  // (E, N, K/32) e8m0 scales
     const int* __restrict__ expert_ids,  // (n_row_tiles,)
     const T* __restrict__ bias,          // (E, N) or null
+    const int* __restrict__ x_gather,    // (Mpad,) or null
+    float* __restrict__ combine_out,     // (T, N) fp32 or null
+    const float* __restrict__ combine_scale,  // (Mpad,)
+    const int* __restrict__ combine_gather,   // (Mpad,)
     const int N, const int K) {
     using MF = MgMfma<T>;
     using vec8 = typename MF::vec8;
@@ -299,8 +303,11 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_mx4_kernel(
     const int lrow = lane & 15;
     const int lslot = lane >> 4;
 
-    const char* xsrc = reinterpret_cast<const char*>(x) +
-                       ((long)(m0 + wid * 8) * K) * sizeof(T);
+    const int xr_lin = m0 + wid * 8 + (lane >> 3);
+    const int xr = x_gather ? max(x_gather[xr_lin], 0) : xr_lin;
+    const int xslot = (lane & 7) ^ (lane >> 3);
+    const char* xsrc_lane = reinterpret_cast<const char*>(x) +
+                            (long)xr * K * sizeof(T) + xslot * 16;
     // this lane stages W rows (n0 + g*64 + wid*8 + lane/8), k-slot lane%8
     const int wrow_l = wid * 8 + (lane >> 3);
     const int wslot = lane & 7;          // 8 bf16 = 4 packed bytes
@@ -325,7 +332,7 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_mx4_kernel(
     // read-side XOR; note hipcc waits vmcnt(0) for the W loads while a
     // glds is in flight — the known mixed-kind cost, acceptable v1)
 #define MX_STAGE_X(t, b)                                                     \
-    mg_glds_rows128<0>(xsrc + (long)(t) * 128, (long)K * sizeof(T), xdst[b])
+    mg_glds_lane128(xsrc_lane + (long)(t) * 128, xdst[b])
 #define MX_STAGE_W(t, b)                                                     \
     do {                                                                     \
         const int kc = (t) * MG_BK;                                          \
@@ -621,14 +628,29 @@ void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
 
 void moe_grouped_gemm_mx4(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
                           torch::Tensor ws, torch::Tensor expert_ids,
-                          c10::optional<torch::Tensor> bias, long bn) {
+                          c10::optional<torch::Tensor> bias, long bn,
+                          c10::optional<torch::Tensor> x_gather,
+                          c10::optional<torch::Tensor> combine_out,
+                          c10::optional<torch::Tensor> combine_scale,
+                          c10::optional<torch::Tensor> combine_gather) {
     TORCH_CHECK(y.dim() == 2 && x.dim() == 2 && wq.dim() == 3 && ws.dim() == 3,
                 "shapes");
     TORCH_CHECK(x.is_contiguous() && wq.is_contiguous() && ws.is_contiguous() &&
                 y.is_contiguous() && expert_ids.is_contiguous(),
                 "contiguous tensors required");
-    const int Mpad = x.size(0), K = x.size(1);
+    const int K = x.size(1);
+    const int Mpad = x_gather.has_value() ? (int)expert_ids.numel() * 64
+                                          : (int)x.size(0);
     const int N = wq.size(1);
+    if (x_gather.has_value())
+        TORCH_CHECK(x_gather->scalar_type() == at::ScalarType::Int &&
+                    x_gather->numel() == Mpad, "x_gather (Mpad,) int32");
+    if (combine_out.has_value()) {
+        TORCH_CHECK(combine_gather.has_value() &&
+                    combine_out->scalar_type() == at::ScalarType::Float &&
+                    combine_out->size(1) == N &&
+                    combine_gather->numel() == Mpad, "combine args");
+    }
     TORCH_CHECK(wq.size(2) == K / 2 && ws.size(2) == K / 32, "packed shape");
     TORCH_CHECK(y.size(0) == Mpad && y.size(1) == N, "shape");
     TORCH_CHECK(K % MG_BK == 0 && Mpad % 64 == 0, "align");
@@ -664,6 +686,13 @@ void moe_grouped_gemm_mx4(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
                            (const unsigned char*)ws.data_ptr(),               \
                            expert_ids.data_ptr<int>(),                        \
                            bias.has_value() ? (const T*)bias->data_ptr() : nullptr, \
+                           x_gather ? x_gather->data_ptr<int>() : nullptr,    \
+                           combine_out ? combine_out->data_ptr<float>()       \
+                                       : nullptr,                             \
+                           combine_scale ? combine_scale->data_ptr<float>()   \
+                                         : nullptr,                           \
+                           combine_gather ? combine_gather->data_ptr<int>()   \
+                                          : nullptr,                          \
                            N, K);                                             \
     } while (0)
 #define MGX_T(T)                                                              \
