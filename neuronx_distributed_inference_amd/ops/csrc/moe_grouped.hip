@@ -424,8 +424,18 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_mx4_kernel(
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 const int m = m0 + wm0 + qi * 16 + lslot * 4 + r;
-                y[(long)m * N + n_out] =
-                    VecTraits<T>::from_float(acc[qi][nj][r] + bv);
+                if (combine_out) {
+                    const int tk = combine_gather[m];
+                    if (tk >= 0) {
+                        const float sc =
+                            combine_scale ? combine_scale[m] : 1.f;
+                        atomicAdd(&combine_out[(long)tk * N + n_out],
+                                  (acc[qi][nj][r] + bv) * sc);
+                    }
+                } else {
+                    y[(long)m * N + n_out] =
+                        VecTraits<T>::from_float(acc[qi][nj][r] + bv);
+                }
             }
         }
     }
