@@ -172,27 +172,15 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
         const int rpg = (span + NG - 1) / NG;
         const int g_lo = split_lo + grp * rpg;
         const int g_hi = min(g_lo + rpg, split_hi);
-        bool valid_next;
-        // clamp the initial prefetch row: late lane-groups with an empty
-        // range (g_lo >= g_hi) would otherwise load past the valid span
-        // (linear: past the tensor end on the last row/head; paged: index
-        // block_table at l/block_size >= max_blocks)
-        long off = row_addr(min(g_lo, split_hi - 1), valid_next);
-        Vec8<CT> kv = load_vec8(kbase + off);
-        Vec8<CT> vv = load_vec8(vbase + off);
-        for (int l = g_lo; l < g_hi; ++l) {
-            const bool valid_cur = valid_next;
-            Vec8<CT> kv_cur = kv, vv_cur = vv;
-            const int lnext = (l + 1 < g_hi) ? l + 1 : l;
-            off = row_addr(lnext, valid_next);
-            kv = load_vec8(kbase + off);
-            vv = load_vec8(vbase + off);
-            if (PAGED && !valid_cur) continue;
+        // row-compute body (shared by the 2-deep software pipeline legs)
+        auto attn_row = [&](int l, bool valid_cur, const Vec8<CT>& kv_cur,
+                            const Vec8<CT>& vv_cur) {
+            if (PAGED && !valid_cur) return;
             if (ring) {
                 const int p_s = pmax - ((pmax - l) % Smax);  // slot's true pos
                 if (p_s >= eff_len ||
                     (window > 0 && p_s <= eff_len - 1 - window))
-                    continue;
+                    return;
             }
             const Vec8<CT>& vv_use = vv_cur;
 
@@ -233,6 +221,41 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
 #pragma unroll
                 for (int j = 0; j < 8; ++j)
                     o_acc[g][j] = o_acc[g][j] * corr + p * VecTraits<CT>::to_float(vv_use.v[j]);
+            }
+        };
+
+        // 2-deep pipeline: two row buffers alternate, so at each wait
+        // point the OTHER buffer's load is still in flight and the
+        // waited-on load has had ~2 iterations to land (depth-1 exposed
+        // most of the HBM latency per row: the register copy forces the
+        // vmcnt wait before the next load can issue).
+        // Clamps: empty ranges and tails re-load a safe row.
+        bool v0, v1;
+        long off0 = row_addr(min(g_lo, split_hi - 1), v0);
+        Vec8<CT> k0 = load_vec8(kbase + off0);
+        Vec8<CT> w0 = load_vec8(vbase + off0);
+        long off1 = row_addr((g_lo + 1 < g_hi) ? g_lo + 1
+                                               : min(g_lo, split_hi - 1), v1);
+        Vec8<CT> k1 = load_vec8(kbase + off1);
+        Vec8<CT> w1 = load_vec8(vbase + off1);
+        for (int l = g_lo; l < g_hi; l += 2) {
+            {
+                const bool vc = v0;
+                const Vec8<CT> kc = k0, wc = w0;
+                const int ln = (l + 2 < g_hi) ? l + 2 : l;
+                off0 = row_addr(ln, v0);
+                k0 = load_vec8(kbase + off0);
+                w0 = load_vec8(vbase + off0);
+                attn_row(l, vc, kc, wc);
+            }
+            if (l + 1 < g_hi) {
+                const bool vc = v1;
+                const Vec8<CT> kc = k1, wc = w1;
+                const int ln = (l + 3 < g_hi) ? l + 3 : l + 1;
+                off1 = row_addr(ln, v1);
+                k1 = load_vec8(kbase + off1);
+                w1 = load_vec8(vbase + off1);
+                attn_row(l + 1, vc, kc, wc);
             }
         }
     }

@@ -25,3 +25,14 @@ for Smax in (2048, 8192, 32768):
     t = bench(lambda: ops.attention_decode(q, kc, vc, pos))
     ts = bench(lambda: kc.sum(dtype=torch.float32) + vc.sum(dtype=torch.float32))
     print(f"S={Smax:6d} decode {t*1e6:8.1f}us {gb/t/1e3:5.2f} TB/s | raw-sum {ts*1e6:8.1f}us {gb/ts/1e3:5.2f} TB/s")
+
+# headline decode shape: B=256, short ctx (the bench config)
+for B2, S2, ctx in ((256, 176, 128), (256, 2048, 2000)):
+    kc = torch.randn(B2, 8, S2, 128, dtype=torch.bfloat16, device=dev)
+    vc = torch.randn(B2, 8, S2, 128, dtype=torch.bfloat16, device=dev)
+    q = torch.randn(B2, 32, 1, 128, dtype=torch.bfloat16, device=dev)
+    pos = torch.full((B2, 1), ctx, dtype=torch.int32, device=dev)
+    gb = B2 * 8 * (ctx + 1) * 128 * 2 * 2 / 1e9
+    t = bench(lambda: ops.attention_decode(q, kc, vc, pos))
+    print(f"B={B2} ctx={ctx:5d} decode {t*1e6:8.1f}us "
+          f"{gb/t/1e3:5.2f} TB/s effective")
