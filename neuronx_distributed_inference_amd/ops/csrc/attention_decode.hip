@@ -59,11 +59,7 @@ template <> struct Dot2<__half> {
     }
 };
 
-// WPH (wave-per-head): each WAVE owns one (b, hkv, sq) — 4x the cache
-// rows per lane-group and ZERO cross-wave LDS combine / syncthreads.
-// Used when B*Hkv*Sq is large (short-ctx big-batch decode, where the
-// per-block fixed costs dominated: measured 2.4 TB/s at B=256 ctx=128).
-template <typename T, typename CT, int D, int GQ, bool PAGED, bool WPH = false>
+template <typename T, typename CT, int D, int GQ, bool PAGED>
 __global__ __launch_bounds__(256) void decode_attn_pass1(
     float* __restrict__ o_part,        // (B, Hq, Sq, nsplit, D)
     float* __restrict__ ml_part,       // (B, Hq, Sq, nsplit, 2)
@@ -76,31 +72,20 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
     const float scale, const int window, const int ring,
     const int Hkv, const int Sq, const int gq,
     const int Smax, const int chunk, const int nsplit,
-    const int block_size, const int max_blocks, const int Btot) {
+    const int block_size, const int max_blocks) {
     using VT = VecTraits<T>;
     constexpr int GS = D / 8;          // lanes per group
-    constexpr int NG = WPH ? (WAVE_SIZE / GS) : (256 / GS);  // groups/owner
+    constexpr int NG = 256 / GS;       // groups per workgroup
 
     const int split = blockIdx.x;
-    const int lane = threadIdx.x & (WAVE_SIZE - 1);
-    const int wid = threadIdx.x / WAVE_SIZE;
-
-    int hkv, sq, b;
-    if constexpr (WPH) {
-        const int fid = blockIdx.y * 4 + wid;     // flat (b, hkv, sq)
-        if (fid >= Btot * Hkv * Sq) return;       // tail waves (no syncs)
-        b = fid / (Hkv * Sq);
-        hkv = (fid / Sq) % Hkv;
-        sq = fid % Sq;
-    } else {
-        hkv = blockIdx.y / Sq;
-        sq = blockIdx.y % Sq;
-        b = blockIdx.z;
-    }
+    const int hkv = blockIdx.y / Sq;
+    const int sq = blockIdx.y % Sq;
+    const int b = blockIdx.z;
     const int Hq = Hkv * gq;
 
-    // groups: whole-block in the cooperative layout, within-wave for WPH
-    const int grp = WPH ? (lane / GS) : (int)(threadIdx.x / GS);
+    const int lane = threadIdx.x & (WAVE_SIZE - 1);
+    const int wid = threadIdx.x / WAVE_SIZE;
+    const int grp = threadIdx.x / GS;         // 0..NG-1
     const int c = threadIdx.x % GS;           // lane-in-group
 
     const int row = seq_ids[b];
@@ -296,30 +281,7 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
         }
     }
 
-    if constexpr (WPH) {
-        // the butterfly already combined this wave's NG groups; every
-        // lane of group 0 holds the full (b, hkv, sq) result for its c
-        // slice — write partials directly, no LDS, no syncthreads
-        if (lane < GS) {
-#pragma unroll
-            for (int g = 0; g < GQ; ++g) {
-                if (g >= gq) continue;
-                const long base =
-                    ((((long)b * Hq + hkv * gq + g) * Sq + sq) * nsplit + split);
-#pragma unroll
-                for (int j = 0; j < 8; ++j)
-                    o_part[base * D + c * 8 + j] = o_acc[g][j];
-                if (c == 0) {
-                    ml_part[base * 2 + 0] = m[g];
-                    ml_part[base * 2 + 1] = lsum[g];
-                }
-            }
-        }
-        return;
-    }
-
     // ---- combine the 4 waves through LDS ----
-    if constexpr (!WPH) {
     __shared__ float o_lds[4][GQ > 4 ? GQ : 4][D];
     __shared__ float ml_lds[4][GQ > 4 ? GQ : 4][2];
     if (lane < GS) {
@@ -362,7 +324,6 @@ __global__ __launch_bounds__(256) void decode_attn_pass1(
             }
         }
     }
-    }  // !WPH
 }
 
 template <typename T, int D>
@@ -434,31 +395,23 @@ static void launch_decode(torch::Tensor& out, const torch::Tensor& q,
     auto ml_part = torch::empty({(long)B * Hq * Sq * nsplit * 2}, opts);
     auto stream = at::hip::getCurrentHIPStream();
 
-    // wave-per-head when (b, hkv, sq) pairs alone fill the chip (big-
-    // batch short-ctx decode): 4x the rows per lane-group and no LDS
-    // combine — the cooperative 4-wave layout stays for small B*Hkv
-    // where per-pair parallelism is scarce
-    const int pairs = B * Hkv * Sq;
-    const bool wph = pairs >= 1024 && nsplit == 1;
-    dim3 grid1(nsplit, wph ? (pairs + 3) / 4 : Hkv * Sq, wph ? 1 : B);
-#define P1_LAUNCH(PAGEDV, WPHV, BT)                                           \
-    hipLaunchKernelGGL((decode_attn_pass1<T, CT, D, GQ, PAGEDV, WPHV>),       \
-                       grid1, dim3(256), 0, stream,                           \
-                       o_part.data_ptr<float>(), ml_part.data_ptr<float>(),   \
-                       (const T*)q.data_ptr(), (const CT*)k_cache.data_ptr(), \
-                       (const CT*)v_cache.data_ptr(), seq_lens.data_ptr<int>(), \
-                       seq_ids.data_ptr<int>(), BT,                           \
-                       scale, window, PAGEDV ? 0 : ring, Hkv, Sq, gq, Smax,   \
-                       chunk, nsplit, PAGEDV ? block_size : 0,                \
-                       PAGEDV ? max_blocks : 0, B)
+    dim3 grid1(nsplit, Hkv * Sq, B);
     if (paged) {
-        if (wph) P1_LAUNCH(true, true, block_table->data_ptr<int>());
-        else P1_LAUNCH(true, false, block_table->data_ptr<int>());
+        hipLaunchKernelGGL((decode_attn_pass1<T, CT, D, GQ, true>), grid1, dim3(256), 0, stream,
+                           o_part.data_ptr<float>(), ml_part.data_ptr<float>(),
+                           (const T*)q.data_ptr(), (const CT*)k_cache.data_ptr(),
+                           (const CT*)v_cache.data_ptr(), seq_lens.data_ptr<int>(),
+                           seq_ids.data_ptr<int>(), block_table->data_ptr<int>(),
+                           scale, window, 0, Hkv, Sq, gq, Smax, chunk, nsplit,
+                           block_size, max_blocks);
     } else {
-        if (wph) P1_LAUNCH(false, true, (const int*)nullptr);
-        else P1_LAUNCH(false, false, (const int*)nullptr);
+        hipLaunchKernelGGL((decode_attn_pass1<T, CT, D, GQ, false>), grid1, dim3(256), 0, stream,
+                           o_part.data_ptr<float>(), ml_part.data_ptr<float>(),
+                           (const T*)q.data_ptr(), (const CT*)k_cache.data_ptr(),
+                           (const CT*)v_cache.data_ptr(), seq_lens.data_ptr<int>(),
+                           seq_ids.data_ptr<int>(), nullptr,
+                           scale, window, ring, Hkv, Sq, gq, Smax, chunk, nsplit, 0, 0);
     }
-#undef P1_LAUNCH
     HIP_CHECK_KERNEL();
     dim3 grid2(B * Hq * Sq);
     hipLaunchKernelGGL((decode_attn_pass2<T, D>), grid2, dim3(WAVE_SIZE), 0, stream,
