@@ -94,3 +94,23 @@ def test_inference_demo_eagle_flags(tiny_hf_checkpoint, tmp_path, extra):
         "--enable-eagle-speculation",
     ] + extra)
     assert rc == 0
+
+
+def test_eagle_forces_on_device_greedy_sampling():
+    """EAGLE verification consumes on-device tokens: the config builder
+    must arm greedy on-device sampling when --enable-eagle-speculation is
+    set without --on-device-sampling."""
+    from neuronx_distributed_inference_amd.inference_demo import (
+        create_engine_config,
+        parse_args,
+    )
+
+    args = parse_args([
+        "--model-type", "llama", "run",
+        "--model-path", "x", "--compiled-model-path", "y",
+        "--enable-eagle-speculation", "--speculation-length", "3",
+    ])
+    ec = create_engine_config(args, "llama")
+    assert ec.on_device_sampling_config is not None
+    assert ec.on_device_sampling_config.top_k == 1
+    assert ec.enable_eagle_speculation
