@@ -89,6 +89,9 @@ def setup_run_parser(run: argparse.ArgumentParser):
     run.add_argument("--draft-model-path", default=None)
     run.add_argument("--draft-model-type", default=None,
                      help="model family of the draft (defaults to the target's)")
+    run.add_argument("--draft-tp-degree", type=int, default=None,
+                     help="run the draft at a smaller TP (sub-communicator; "
+                          "must divide --tp-degree)")
     run.add_argument("--speculation-length", type=int, default=0)
     run.add_argument("--enable-fused-speculation", action="store_true")
     run.add_argument("--spec-rejection-sampling", action="store_true",
@@ -303,6 +306,7 @@ def run_inference(args, model_type: str) -> int:
         expert_model_parallel_size=args.ep_degree,
         context_parallel_size=args.cp_degree,
         attention_dp_size=args.attention_dp_degree,
+        draft_tp_size=getattr(args, "draft_tp_degree", None),
     )
     device = torch.device(f"cuda:{local_rank}") if (
         torch.cuda.is_available() and not args.on_cpu
@@ -356,6 +360,9 @@ def run_inference(args, model_type: str) -> int:
             args.draft_model_type or model_type)
         draft_ec = create_engine_config(args, args.draft_model_type or model_type)
         draft_ec.speculation_length = 0
+        draft_tp = getattr(args, "draft_tp_degree", None)
+        if draft_tp and draft_tp < args.tp_degree:
+            draft_ec.tp_degree = draft_tp
         draft_cfg = d_cfg_cls.from_hf_config(
             args.draft_model_path, neuron_config=draft_ec)
         draft_app = InferenceApplication(
@@ -365,7 +372,18 @@ def run_inference(args, model_type: str) -> int:
             draft_app.compile(draft_dir)
         if torch.distributed.is_initialized():
             torch.distributed.barrier()
-        draft = draft_app.load(draft_dir, device=device)
+        if draft_tp and draft_tp < args.tp_degree:
+            # build/shard/run the draft over the reduced-TP subgroup
+            # (sharding math is config-driven; only the collectives need
+            # the group override)
+            from neuronx_distributed_inference_amd.modules.speculation import (
+                DraftTPWrapper,
+            )
+
+            with ps.draft_tp_context():
+                draft = DraftTPWrapper(draft_app.load(draft_dir, device=device))
+        else:
+            draft = draft_app.load(draft_dir, device=device)
         if args.enable_eagle_speculation:
             # EAGLE drafts wrap the draft DecoderModel with the fc combiner;
             # fc weights come from the draft checkpoint when present

@@ -19,6 +19,39 @@ import torch
 from neuronx_distributed_inference_amd.utils.hf_adapter import GenerationConfig
 
 
+class DraftTPWrapper:
+    """Proxy that runs every draft call inside the reduced-TP group
+    override (ref speculative-draft group, application_base.py:61-66):
+    the draft's parallel layers and distributed argmax/topk then
+    collect over draft_tp ranks instead of the full TP group. Weights
+    are replicated across subgroups, so every rank computes the draft
+    locally — no token broadcast."""
+
+    def __init__(self, inner):
+        object.__setattr__(self, "_inner", inner)
+
+    def __call__(self, *a, **kw):
+        from neuronx_distributed_inference_amd.parallel import state as ps
+
+        with ps.draft_tp_context():
+            return self._inner(*a, **kw)
+
+    def generate(self, *a, **kw):
+        from neuronx_distributed_inference_amd.parallel import state as ps
+
+        with ps.draft_tp_context():
+            return self._inner.generate(*a, **kw)
+
+    def reset(self, *a, **kw):
+        from neuronx_distributed_inference_amd.parallel import state as ps
+
+        with ps.draft_tp_context():
+            return self._inner.reset(*a, **kw)
+
+    def __getattr__(self, name):
+        return getattr(object.__getattribute__(self, "_inner"), name)
+
+
 class SpeculativeGenerationAdapter:
     def __init__(self, model, draft_model, speculation_length: int):
         self.model = model

@@ -25,6 +25,8 @@ from dataclasses import dataclass, field
 from typing import List, Optional
 
 import torch
+from contextlib import contextmanager
+
 import torch.distributed as dist
 
 logger = logging.getLogger("nxdi_amd")
@@ -167,12 +169,19 @@ def initialize_model_parallel(
 
     # Speculative-draft group: draft model at smaller TP
     # (ref application_base.py:61-66, utils/distributed.py:30-36).
+    # EVERY rank joins one subgroup (the draft's weights are replicated
+    # across subgroups), so all ranks compute the draft locally at the
+    # reduced degree — no broadcast of draft tokens, and the draft's
+    # all-reduces run over draft_tp ranks instead of tp.
     if draft_tp_size and draft_tp_size < tp:
+        if tp % draft_tp_size:
+            raise ValueError("draft_tp_size must divide tp_degree")
         for mesh in st.tp_mesh:
-            sub = mesh[:draft_tp_size]
-            g = dist.new_group(sub)
-            if rank in sub:
-                st.draft_tp_group = g
+            for i in range(tp // draft_tp_size):
+                sub = mesh[i * draft_tp_size : (i + 1) * draft_tp_size]
+                g = dist.new_group(sub)
+                if rank in sub:
+                    st.draft_tp_group = g
 
     _STATE = st
 
@@ -200,16 +209,22 @@ def get_world_group():
 
 
 def get_tensor_model_parallel_group():
+    if _TP_OVERRIDE:
+        return _TP_OVERRIDE[-1]
     return _STATE.tp_group
 
 
 def get_tensor_model_parallel_size() -> int:
     if not dist.is_initialized():
         return 1
+    if _TP_OVERRIDE:
+        return dist.get_world_size(group=_TP_OVERRIDE[-1])
     return min(_STATE.tp_degree, _dist_world()) if _STATE.initialized else 1
 
 
 def get_tensor_model_parallel_rank() -> int:
+    if _TP_OVERRIDE:
+        return _group_rank(_TP_OVERRIDE[-1])
     return _group_rank(_STATE.tp_group)
 
 
@@ -267,6 +282,26 @@ def get_moe_tp_group():
 
 def get_draft_tp_group():
     return _STATE.draft_tp_group
+
+
+_TP_OVERRIDE: list = []
+
+
+@contextmanager
+def draft_tp_context():
+    """Route the default TP group/size/rank to the speculative-draft
+    subgroup for the duration (ref get_tp_group override,
+    utils/distributed.py:30-36): the draft model's parallel layers and
+    distributed argmax/topk then operate over the reduced-TP group.
+    No-op when no draft group was initialized."""
+    if _STATE.draft_tp_group is None:
+        yield
+        return
+    _TP_OVERRIDE.append(_STATE.draft_tp_group)
+    try:
+        yield
+    finally:
+        _TP_OVERRIDE.pop()
 
 
 def init_distributed_from_env(backend: Optional[str] = None) -> int:

@@ -637,3 +637,94 @@ def _worker_cp_flash_decoding_e2e(rank):
 
 def test_cp_plus_flash_decoding_e2e():
     spawn("_worker_cp_flash_decoding_e2e")
+
+
+def _worker_draft_tp_speculation(rank):
+    """Two-model speculation with the TARGET at TP=2 and the DRAFT at
+    TP=1 via the draft sub-communicator (ref speculative-draft group):
+    output must equal the target-only TP=2 greedy stream."""
+    from neuronx_distributed_inference_amd.config import EngineConfig
+    from neuronx_distributed_inference_amd.models.llama.modeling_llama import (
+        LlamaForCausalLM,
+        LlamaInferenceConfig,
+    )
+    from neuronx_distributed_inference_amd.modules.speculation import (
+        DraftTPWrapper,
+        SpeculativeGenerationAdapter,
+    )
+    from neuronx_distributed_inference_amd.parallel import state as ps
+    from neuronx_distributed_inference_amd.utils.checkpoint_sharding import (
+        shard_state_dict_for_rank,
+    )
+    from neuronx_distributed_inference_amd.utils.hf_adapter import (
+        GenerationConfig,
+        HuggingFaceGenerationAdapter,
+    )
+
+    def make_cfg(tp, spec_len=0, layers=2, hidden=64):
+        from neuronx_distributed_inference_amd.config import (
+            OnDeviceSamplingConfig,
+        )
+
+        ec = EngineConfig(batch_size=2, max_batch_size=2, seq_len=96,
+                          max_length=96, torch_dtype="float32", tp_degree=tp,
+                          speculation_length=spec_len,
+                          on_device_sampling_config=OnDeviceSamplingConfig(
+                              top_k=1))
+        return LlamaInferenceConfig(
+            neuron_config=ec, vocab_size=128, hidden_size=hidden,
+            intermediate_size=hidden * 2, num_hidden_layers=layers,
+            num_attention_heads=4, num_key_value_heads=2,
+            max_position_embeddings=128, rms_norm_eps=1e-5,
+            rope_theta=10000.0, hidden_act="silu", tie_word_embeddings=False)
+
+    # reference: target-only greedy at TP=1
+    ps.destroy_model_parallel()
+    ps.initialize_model_parallel(tensor_model_parallel_size=1)
+    torch.manual_seed(21)
+    full = LlamaForCausalLM(make_cfg(1)).eval()
+    with torch.no_grad():
+        for p in full.parameters():
+            if p.dim() >= 2:
+                p.normal_(0, 0.1)
+    ids = torch.randint(0, 128, (2, 8))
+    want = HuggingFaceGenerationAdapter(full).generate(
+        ids, generation_config=GenerationConfig(max_new_tokens=12))
+
+    # target TP=2 + draft TP=1 (replicated across the two subgroups)
+    ps.destroy_model_parallel()
+    ps.initialize_model_parallel(tensor_model_parallel_size=WORLD,
+                                 draft_tp_size=1)
+    target = LlamaForCausalLM(make_cfg(WORLD, spec_len=4)).eval()
+    full_sd = {k: v for k, v in full.model.state_dict().items()
+               if not k.startswith("kv_mgr.")}
+    shard = shard_state_dict_for_rank(full_sd, target.model, rank, WORLD)
+    target.model.load_state_dict(shard, strict=False)
+
+    torch.manual_seed(33)  # identical draft weights on every rank
+    with ps.draft_tp_context():
+        draft_lm = LlamaForCausalLM(make_cfg(1, layers=1, hidden=32)).eval()
+        with torch.no_grad():
+            for p in draft_lm.parameters():
+                if p.dim() >= 2:
+                    p.normal_(0, 0.1)
+    draft = DraftTPWrapper(draft_lm)
+
+    got = SpeculativeGenerationAdapter(target, draft, 4).generate(
+        ids, generation_config=GenerationConfig(max_new_tokens=12))
+    for i in range(2):
+        row = got[i]
+        n = row.shape[0] - int((row != 0).flip(0).float().argmax())
+        n = min(n, want.shape[1])
+        assert n >= ids.shape[1] + 8
+        assert torch.equal(row[:n], want[i, :n]), (row.tolist(),
+                                                   want[i].tolist())
+    # sanity: the draft group really is size 1 inside the context
+    with ps.draft_tp_context():
+        assert ps.get_tensor_model_parallel_size() == 1
+    assert ps.get_tensor_model_parallel_size() == WORLD
+    return True
+
+
+def test_draft_tp_speculation():
+    spawn("_worker_draft_tp_speculation")
