@@ -196,10 +196,15 @@ class FusedSpeculationModel(torch.nn.Module):
                 input_ids, position_ids, seq_ids, sampling_params)
         d_in = input_ids
         cand = []
+        from neuronx_distributed_inference_amd.parallel import state as _ps
+
         for i in range(k):
-            d_tok, _ = self.draft(
-                d_in, None, position_ids + i, seq_ids, None, phase=PHASE_TKG
-            )
+            # draft collectives run over the draft sub-communicator when
+            # one was initialized (no-op context otherwise)
+            with _ps.draft_tp_context():
+                d_tok, _ = self.draft(
+                    d_in, None, position_ids + i, seq_ids, None,
+                    phase=PHASE_TKG)
             d_in = d_tok.view(B, 1)
             cand.append(d_in)
         cand = torch.cat(cand, dim=1)                       # (B, k)
@@ -251,8 +256,13 @@ class FusedSpeculationModel(torch.nn.Module):
             # the draft model's own sampler output is ignored: the round
             # samples d_j itself from the SAME filtered distribution it
             # uses for q(d_j), so accept ratios are exact by construction
-            self.draft(d_in, None, position_ids + i, seq_ids,
-                       sampling_params, phase=PHASE_TKG)
+            from neuronx_distributed_inference_amd.parallel import (
+                state as _ps,
+            )
+
+            with _ps.draft_tp_context():
+                self.draft(d_in, None, position_ids + i, seq_ids,
+                           sampling_params, phase=PHASE_TKG)
             vals, idxs = self.draft.spec_topk           # (B,1,Kg)
             from neuronx_distributed_inference_amd.modules.generation.sampling import (
                 filtered_probs,
