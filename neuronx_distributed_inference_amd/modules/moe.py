@@ -128,6 +128,23 @@ class ExpertMLPs(nn.Module):
         moe_tp = ps.get_tensor_model_parallel_size() // self.ep_degree
         return (tp_rank // max(moe_tp, 1)) * self.num_local_experts
 
+    def quantize_fp8(self) -> bool:
+        """Convert the stacked expert weights to fp8-e4m3 (1 byte/elem,
+        per-output-channel fp32 scale): HALVES the decode weight stream.
+        Returns False (and leaves bf16) when the shard shapes don't meet
+        the fp8 kernel's column-tile constraint (N % 128 == 0)."""
+        I, H = self.w_gate.shape[1], self.w_down.shape[1]
+        if I % 128 or H % 128:
+            return False
+        for name in ("w_gate", "w_up", "w_down"):
+            w = getattr(self, name)
+            q, sc = ops.fp8_moe_quantize(w.data)
+            del self._parameters[name]
+            self.register_buffer(name + "_q", q)
+            self.register_buffer(name + "_s", sc)
+        self.fp8 = True
+        return True
+
     def _forward_dense(self, x: torch.Tensor, weights: torch.Tensor,
                        idx: torch.Tensor) -> torch.Tensor:
         T, H = x.shape
@@ -182,7 +199,11 @@ class ExpertMLPs(nn.Module):
                     return cand
             return 64
 
-        I, Hn = self.w_gate.shape[1], self.w_down.shape[1]
+        fp8 = getattr(self, "fp8", False)
+        if fp8:
+            I, Hn = self.w_gate_q.shape[1], self.w_down_q.shape[1]
+        else:
+            I, Hn = self.w_gate.shape[1], self.w_down.shape[1]
         if self.scale_input:
             # input scaling is nonlinear through the GLU: keep the
             # materialized gather for this (rare) variant
@@ -194,10 +215,22 @@ class ExpertMLPs(nn.Module):
         else:
             x_in = x                      # token gather fuses into staging
             xg = gtab
-        g = ops.moe_grouped_gemm(x_in, self.w_gate, expert_ids, bn=_bn(I),
-                                 x_gather=xg)
-        u = ops.moe_grouped_gemm(x_in, self.w_up, expert_ids, bn=_bn(I),
-                                 x_gather=xg)
+        if fp8:
+            # fp8 column tiles are 128/256 only
+            def _bn8(N, ntiles=Mpad // 64):
+                if N % 256 == 0 and (N // 256) * ntiles >= 256:
+                    return 256
+                return 128
+
+            g = ops.moe_grouped_gemm_fp8(x_in, self.w_gate_q, self.w_gate_s,
+                                         expert_ids, bn=_bn8(I), x_gather=xg)
+            u = ops.moe_grouped_gemm_fp8(x_in, self.w_up_q, self.w_up_s,
+                                         expert_ids, bn=_bn8(I), x_gather=xg)
+        else:
+            g = ops.moe_grouped_gemm(x_in, self.w_gate, expert_ids, bn=_bn(I),
+                                     x_gather=xg)
+            u = ops.moe_grouped_gemm(x_in, self.w_up, expert_ids, bn=_bn(I),
+                                     x_gather=xg)
         h = self.act_mul(g, u)
         # routing-weighted combine fuses into the down epilogue (fp32
         # atomics, one rmw per element); non-local assignments are
@@ -207,9 +240,15 @@ class ExpertMLPs(nn.Module):
             else torch.ones_like(gatew, dtype=torch.float32)
         gw_tab[dest] = torch.where(lmask, scale, torch.zeros_like(scale))
         out32 = torch.zeros(T, H, dtype=torch.float32, device=x.device)
-        ops.moe_grouped_gemm(h, self.w_down, expert_ids, bn=_bn(Hn),
-                             combine_out=out32, combine_scale=gw_tab,
-                             combine_gather=gtab)
+        if fp8:
+            ops.moe_grouped_gemm_fp8(h, self.w_down_q, self.w_down_s,
+                                     expert_ids, bn=_bn8(Hn),
+                                     combine_out=out32, combine_scale=gw_tab,
+                                     combine_gather=gtab)
+        else:
+            ops.moe_grouped_gemm(h, self.w_down, expert_ids, bn=_bn(Hn),
+                                 combine_out=out32, combine_scale=gw_tab,
+                                 combine_gather=gtab)
         return out32.to(x.dtype)
 
     def forward(self, x: torch.Tensor, weights: torch.Tensor, idx: torch.Tensor):
@@ -225,6 +264,8 @@ class ExpertMLPs(nn.Module):
         # redundant FLOPs.
         import os as _os
 
+        if getattr(self, "fp8", False):
+            return self._forward_grouped(x, weights, idx)
         grouped_env = _os.environ.get("NXDI_MOE_GROUPED")
         # small batches leave too few 64-row tiles to fill 256 CUs at the
         # wide-tile occupancy (measured: Mixtral T=64 grouped 27.7 vs

@@ -744,6 +744,73 @@ def moe_grouped_gemm(x_sorted: torch.Tensor, w: torch.Tensor,
     return out if out is not None else y
 
 
+def fp8_moe_quantize(w: torch.Tensor):
+    """Per-output-channel e4m3fn quantization of stacked expert weights
+    (E, N, K) -> (wq (E, N, K) uint8, scale (E, N) fp32). e4m3fn max
+    normal = 448."""
+    amax = w.float().abs().amax(dim=-1).clamp(min=1e-12)   # (E, N)
+    scale = amax / 448.0
+    q = (w.float() / scale.unsqueeze(-1)).clamp(-448, 448)
+    wq = q.to(torch.float8_e4m3fn).view(torch.uint8)
+    return wq.contiguous(), scale.to(torch.float32).contiguous()
+
+
+def fp8_moe_dequant(wq: torch.Tensor, scale: torch.Tensor) -> torch.Tensor:
+    return wq.view(torch.float8_e4m3fn).float() * scale.unsqueeze(-1).float()
+
+
+def moe_grouped_gemm_fp8(x_sorted: torch.Tensor, wq: torch.Tensor,
+                         wscale: torch.Tensor, expert_ids: torch.Tensor,
+                         bias=None, bn: int = 0, x_gather=None,
+                         combine_out=None, combine_scale=None,
+                         combine_gather=None) -> torch.Tensor:
+    """Grouped expert GEMM with fp8-e4m3 weights (E, N, K) u8 + per-
+    channel fp32 scales (E, N): HALF the weight stream of bf16 through
+    the same all-glds pipeline; dequant happens at the LDS fragment read
+    and the scale folds into the epilogue. Same gather/combine fusions
+    as the bf16 kernel. CPU: dequant reference."""
+    K = x_sorted.shape[1]
+    Mpad = expert_ids.numel() * MOE_TILE if x_gather is not None \
+        else x_sorted.shape[0]
+    N = wq.shape[1]
+    if x_sorted.is_cuda:
+        c = _require_ext("moe_grouped_gemm_fp8")
+        y = torch.empty(Mpad, N, dtype=x_sorted.dtype, device=x_sorted.device)
+        c.moe_grouped_gemm_fp8(y, x_sorted.contiguous(), wq.contiguous(),
+                               wscale.contiguous(), expert_ids.contiguous(),
+                               None if bias is None else bias.contiguous(), bn,
+                               None if x_gather is None else x_gather.contiguous(),
+                               combine_out, combine_scale,
+                               None if combine_gather is None
+                               else combine_gather.contiguous())
+        return combine_out if combine_out is not None else y
+    out = combine_out
+    y = torch.zeros(Mpad, N, dtype=x_sorted.dtype)
+    for t in range(Mpad // MOE_TILE):
+        e = int(expert_ids[t])
+        if e < 0:
+            continue
+        rows = slice(t * MOE_TILE, (t + 1) * MOE_TILE)
+        if x_gather is not None:
+            xg = x_sorted[x_gather[rows].clamp(min=0).long()]
+        else:
+            xg = x_sorted[rows]
+        wd = fp8_moe_dequant(wq[e], wscale[e]).to(x_sorted.dtype)
+        r = xg @ wd.t()
+        if bias is not None:
+            r = r + bias[e].to(r.dtype)
+        if out is not None:
+            cg = combine_gather[rows].long()
+            sc = (combine_scale[rows] if combine_scale is not None
+                  else torch.ones(MOE_TILE))
+            for j in range(MOE_TILE):
+                if cg[j] >= 0:
+                    out[cg[j]] += r[j].float() * float(sc[j])
+        else:
+            y[rows] = r
+    return out if out is not None else y
+
+
 # ----------------------------------------------------------------------
 # OCP MXFP4 (SURVEY K18; ref models/gpt_oss/mx_layout_transform.py):
 # e2m1 nibbles packed 2/byte + one e8m0 scale per 32 elements.

@@ -58,6 +58,13 @@ void moe_grouped_tables_gpu(torch::Tensor order, torch::Tensor dest,
                             torch::Tensor expert_ids, torch::Tensor gtab,
                             torch::Tensor idx_flat, long num_experts,
                             long topk);
+void moe_grouped_gemm_fp8(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
+                          torch::Tensor wscale, torch::Tensor expert_ids,
+                          c10::optional<torch::Tensor> bias, long bn,
+                          c10::optional<torch::Tensor> x_gather,
+                          c10::optional<torch::Tensor> combine_out,
+                          c10::optional<torch::Tensor> combine_scale,
+                          c10::optional<torch::Tensor> combine_gather);
 void moe_grouped_gemm_mx4(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
                           torch::Tensor ws, torch::Tensor expert_ids,
                           c10::optional<torch::Tensor> bias, long bn,
@@ -128,6 +135,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("y"), py::arg("x"), py::arg("w"), py::arg("expert_ids"),
           py::arg("bias") = py::none(), py::arg("bn") = 0,
           py::arg("x_gather") = py::none(),
+          py::arg("combine_out") = py::none(),
+          py::arg("combine_scale") = py::none(),
+          py::arg("combine_gather") = py::none());
+    m.def("moe_grouped_gemm_fp8", &moe_grouped_gemm_fp8,
+          "grouped expert GEMM, fp8-e4m3 weights dequantized at the LDS "
+          "fragment read, per-channel scale in the epilogue (CDNA4)",
+          py::arg("y"), py::arg("x"), py::arg("wq"), py::arg("wscale"),
+          py::arg("expert_ids"), py::arg("bias") = py::none(),
+          py::arg("bn") = 0, py::arg("x_gather") = py::none(),
           py::arg("combine_out") = py::none(),
           py::arg("combine_scale") = py::none(),
           py::arg("combine_gather") = py::none());

@@ -442,6 +442,196 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_mx4_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// FP8 (OCP e4m3fn) grouped expert GEMM: weights stored 1 byte/elem with
+// ONE fp32 scale per output channel (E, N). Unlike the MXFP4 variant
+// (register-staged packed W -> mixed-load vmcnt drains), this keeps the
+// ALL-GLDS pipeline: the raw fp8 bytes stage through LDS exactly like
+// bf16 (half-width [rows][64B] images, 16 rows per 1 KB glds piece),
+// and the B fragment dequantizes on the LDS->register read (4x
+// cvt_pk_f32_fp8 per fragment — trivial next to the chunk's DMA time).
+// The per-channel scale multiplies the fp32 accumulator once in the
+// epilogue (scale is constant along K, so post-scaling is exact).
+// Restriction: BNT in {128, 256} so every wave stages the same number
+// of W pieces (uniform vmcnt counts).
+
+typedef float mg_f32x2 __attribute__((ext_vector_type(2)));
+
+template <typename T>
+DEV_INLINE void fp8x8_to_vec8(const unsigned* p2, T* out) {
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+        const unsigned w = p2[h];
+        mg_f32x2 lo = __builtin_amdgcn_cvt_pk_f32_fp8(w, false);
+        mg_f32x2 hi = __builtin_amdgcn_cvt_pk_f32_fp8(w, true);
+        out[4 * h + 0] = VecTraits<T>::from_float(lo[0]);
+        out[4 * h + 1] = VecTraits<T>::from_float(lo[1]);
+        out[4 * h + 2] = VecTraits<T>::from_float(hi[0]);
+        out[4 * h + 3] = VecTraits<T>::from_float(hi[1]);
+    }
+}
+
+template <typename T, int BNT>
+__global__ __launch_bounds__(512, 2) void moe_grouped_fp8_kernel(
+    T* __restrict__ y,                   // (Mpad, N)
+    const T* __restrict__ x,             // (Mpad, K) — or (T, K) w/ x_gather
+    const unsigned char* __restrict__ wq,   // (E, N, K) e4m3fn bytes
+    const float* __restrict__ wscale,    // (E, N) per-channel scale
+    const int* __restrict__ expert_ids,  // (n_row_tiles,)
+    const T* __restrict__ bias,          // (E, N) or null
+    const int* __restrict__ x_gather,    // (Mpad,) or null
+    float* __restrict__ combine_out,     // (T, N) fp32 or null
+    const float* __restrict__ combine_scale,  // (Mpad,)
+    const int* __restrict__ combine_gather,   // (Mpad,)
+    const int N, const int K) {
+    using MF = MgMfma<T>;
+    using vec8 = typename MF::vec8;
+    constexpr int MT = 64;
+    constexpr int NWG = BNT / 128;  // W glds per wave per chunk (16 rows each)
+    constexpr int G = 1 + NWG;
+    constexpr int NFRAG = BNT / 64;
+    constexpr int XBYTES = MT * 128;
+    constexpr int CHUNK_B = XBYTES + BNT * 64;
+    static_assert(BNT == 128 || BNT == 256, "fp8 grouped: bn 128/256 only");
+
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+
+    const int tile = blockIdx.x;
+    const int rt = blockIdx.y;
+    const int e = expert_ids[rt];
+    if (e < 0) return;
+    const int n0 = tile * BNT;
+    const int m0 = rt * MT;
+    const int nchunks = K / MG_BK;
+
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int wm0 = (wid >> 2) * 32;
+    const int wn0 = (wid & 3) * (BNT / 4);
+    const int lrow = lane & 15;
+    const int lslot = lane >> 4;
+
+    const int xr_lin = m0 + wid * 8 + (lane >> 3);
+    const int xr = x_gather ? max(x_gather[xr_lin], 0) : xr_lin;
+    const int xslot = (lane & 7) ^ (lane >> 3);
+    const char* xsrc_lane = reinterpret_cast<const char*>(x) +
+                            (long)xr * K * sizeof(T) + xslot * 16;
+    // W piece: 16 rows x 64 B; this lane covers row lane/4, 16B-slot
+    // lane%4 (linear slots — the B read is 8 B so 4-way bank overlap,
+    // irrelevant next to the DMA-bound chunk time)
+    const char* wsrc_lane[NWG];
+#pragma unroll
+    for (int i = 0; i < NWG; ++i)
+        wsrc_lane[i] = reinterpret_cast<const char*>(wq) +
+                       ((long)e * N + n0 + wid * (16 * NWG) + i * 16 +
+                        (lane >> 2)) * K + (lane & 3) * 16;
+
+    char* xdst[MG_NBUF];
+    char* wdst[MG_NBUF];
+#pragma unroll
+    for (int b = 0; b < MG_NBUF; ++b) {
+        xdst[b] = smem + b * CHUNK_B + wid * 1024;
+        wdst[b] = smem + b * CHUNK_B + XBYTES + wid * (NWG * 1024);
+    }
+
+#define F8_STAGE(t, b)                                                       \
+    do {                                                                     \
+        mg_glds_lane128(xsrc_lane + (long)(t) * 128, xdst[b]);               \
+        _Pragma("unroll") for (int i = 0; i < NWG; ++i)                      \
+            mg_glds_lane128(wsrc_lane[i] + (long)(t) * 64,                   \
+                            wdst[b] + i * 1024);                             \
+    } while (0)
+
+    mg_f32x4 acc[2][NFRAG];
+#pragma unroll
+    for (int qi = 0; qi < 2; ++qi)
+#pragma unroll
+        for (int nj = 0; nj < NFRAG; ++nj) acc[qi][nj] = {0.f, 0.f, 0.f, 0.f};
+
+#define F8_FRAGS(b)                                                          \
+    do {                                                                     \
+        const char* xb = smem + (b) * CHUNK_B;                               \
+        const char* wb = xb + XBYTES;                                        \
+        _Pragma("unroll") for (int ks = 0; ks < 2; ++ks) {                   \
+            vec8 a0, a1, bb[NFRAG];                                          \
+            {                                                                \
+                const int m = wm0 + lrow;                                    \
+                const int s = (ks * 4 + lslot) ^ (m & 7);                    \
+                a0 = *reinterpret_cast<const vec8*>(xb + m * 128 + s * 16);  \
+            }                                                                \
+            {                                                                \
+                const int m = wm0 + 16 + lrow;                               \
+                const int s = (ks * 4 + lslot) ^ (m & 7);                    \
+                a1 = *reinterpret_cast<const vec8*>(xb + m * 128 + s * 16);  \
+            }                                                                \
+            _Pragma("unroll") for (int nj = 0; nj < NFRAG; ++nj) {           \
+                const int n = wn0 + nj * 16 + lrow;                          \
+                const unsigned* p2 = reinterpret_cast<const unsigned*>(      \
+                    wb + n * 64 + (ks * 4 + lslot) * 8);                     \
+                T tmp[8];                                                    \
+                fp8x8_to_vec8<T>(p2, tmp);                                   \
+                bb[nj] = *reinterpret_cast<const vec8*>(tmp);                \
+            }                                                                \
+            _Pragma("unroll") for (int nj = 0; nj < NFRAG; ++nj) {           \
+                acc[0][nj] = MF::mfma(a0, bb[nj], acc[0][nj]);               \
+                acc[1][nj] = MF::mfma(a1, bb[nj], acc[1][nj]);               \
+            }                                                                \
+        }                                                                    \
+    } while (0)
+
+#define F8_BODY(t, b, b2)                                                    \
+    do {                                                                     \
+        if ((t) + 1 < nchunks) mg_vm_wait<G>();                              \
+        else mg_vm_wait<0>();                                                \
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                   \
+        __builtin_amdgcn_s_barrier();                                        \
+        F8_FRAGS(b);                                                         \
+        if ((t) + 2 < nchunks) F8_STAGE((t) + 2, b2);                        \
+    } while (0)
+
+    F8_STAGE(0, 0);
+    if (nchunks > 1) F8_STAGE(1, 1);
+    int t = 0;
+    while (t + 3 <= nchunks) {
+        F8_BODY(t + 0, 0, 2);
+        F8_BODY(t + 1, 1, 0);
+        F8_BODY(t + 2, 2, 1);
+        t += 3;
+    }
+    if (t < nchunks) F8_BODY(t, 0, 2);
+    if (t + 1 < nchunks) F8_BODY(t + 1, 1, 0);
+#undef F8_BODY
+#undef F8_FRAGS
+#undef F8_STAGE
+    mg_vm_wait<0>();
+
+#pragma unroll
+    for (int nj = 0; nj < NFRAG; ++nj) {
+        const int n_out = n0 + wn0 + nj * 16 + lrow;
+        const float wsc = wscale[(long)e * N + n_out];
+        const float bv =
+            bias ? VecTraits<T>::to_float(bias[(long)e * N + n_out]) : 0.f;
+#pragma unroll
+        for (int qi = 0; qi < 2; ++qi) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int m = m0 + wm0 + qi * 16 + lslot * 4 + r;
+                const float v = acc[qi][nj][r] * wsc + bv;
+                if (combine_out) {
+                    const int tk = combine_gather[m];
+                    if (tk >= 0) {
+                        const float sc =
+                            combine_scale ? combine_scale[m] : 1.f;
+                        atomicAdd(&combine_out[(long)tk * N + n_out], v * sc);
+                    }
+                } else {
+                    y[(long)m * N + n_out] = VecTraits<T>::from_float(v);
+                }
+            }
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
 // Fused segment-table builder: ONE launch replaces the ~15-kernel torch
 // chain (argsort/scatter_add/cumsum/searchsorted — measured 142 us of
 // pure launch latency at decode sizes vs ~10 us here). Single workgroup:
@@ -633,6 +823,83 @@ void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
     }
 #undef MG_T
 #undef MG_LAUNCH
+    HIP_CHECK_KERNEL();
+}
+
+void moe_grouped_gemm_fp8(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
+                          torch::Tensor wscale, torch::Tensor expert_ids,
+                          c10::optional<torch::Tensor> bias, long bn,
+                          c10::optional<torch::Tensor> x_gather,
+                          c10::optional<torch::Tensor> combine_out,
+                          c10::optional<torch::Tensor> combine_scale,
+                          c10::optional<torch::Tensor> combine_gather) {
+    TORCH_CHECK(y.dim() == 2 && x.dim() == 2 && wq.dim() == 3 &&
+                wscale.dim() == 2, "shapes");
+    TORCH_CHECK(x.is_contiguous() && wq.is_contiguous() &&
+                wscale.is_contiguous() && y.is_contiguous() &&
+                expert_ids.is_contiguous(), "contiguous tensors required");
+    const int K = x.size(1);
+    const int Mpad = x_gather.has_value() ? (int)expert_ids.numel() * 64
+                                          : (int)x.size(0);
+    const int N = wq.size(1);
+    TORCH_CHECK(wq.size(2) == K && wscale.size(1) == N, "packed shape");
+    TORCH_CHECK(y.size(0) == Mpad && y.size(1) == N, "shape");
+    TORCH_CHECK(K % MG_BK == 0 && Mpad % 64 == 0, "align");
+    TORCH_CHECK(wq.scalar_type() == at::ScalarType::Byte &&
+                wscale.scalar_type() == at::ScalarType::Float, "wq u8 + f32");
+    int BNT = (int)bn;
+    if (BNT == 0) BNT = (N % 256 == 0) ? 256 : 128;
+    TORCH_CHECK((BNT == 128 || BNT == 256) && N % BNT == 0,
+                "fp8 grouped: bn must be 128/256 and divide N");
+    if (x_gather.has_value())
+        TORCH_CHECK(x_gather->scalar_type() == at::ScalarType::Int &&
+                    x_gather->numel() == Mpad, "x_gather (Mpad,) int32");
+    if (combine_out.has_value())
+        TORCH_CHECK(combine_gather.has_value() &&
+                    combine_out->scalar_type() == at::ScalarType::Float &&
+                    combine_out->size(1) == N &&
+                    combine_gather->numel() == Mpad, "combine args");
+
+    auto stream = at::hip::getCurrentHIPStream();
+    dim3 grid(N / BNT, Mpad / 64);
+
+#define MGF_LAUNCH(T, BNTV)                                                   \
+    do {                                                                      \
+        constexpr int LDS = MG_NBUF * (64 * 128 + BNTV * 64);                 \
+        static bool attr_set = false;                                         \
+        if (!attr_set) {                                                      \
+            (void)hipFuncSetAttribute(                                        \
+                (const void*)moe_grouped_fp8_kernel<T, BNTV>,                 \
+                hipFuncAttributeMaxDynamicSharedMemorySize, LDS);             \
+            attr_set = true;                                                  \
+        }                                                                     \
+        hipLaunchKernelGGL((moe_grouped_fp8_kernel<T, BNTV>), grid,           \
+                           dim3(512), LDS, stream, (T*)y.data_ptr(),          \
+                           (const T*)x.data_ptr(),                            \
+                           (const unsigned char*)wq.data_ptr(),               \
+                           wscale.data_ptr<float>(),                          \
+                           expert_ids.data_ptr<int>(),                        \
+                           bias.has_value() ? (const T*)bias->data_ptr() : nullptr, \
+                           x_gather ? x_gather->data_ptr<int>() : nullptr,    \
+                           combine_out ? combine_out->data_ptr<float>()       \
+                                       : nullptr,                             \
+                           combine_scale ? combine_scale->data_ptr<float>()   \
+                                         : nullptr,                           \
+                           combine_gather ? combine_gather->data_ptr<int>()   \
+                                          : nullptr,                          \
+                           N, K);                                             \
+    } while (0)
+
+    if (x.scalar_type() == at::ScalarType::BFloat16) {
+        if (BNT == 128) MGF_LAUNCH(__hip_bfloat16, 128);
+        else MGF_LAUNCH(__hip_bfloat16, 256);
+    } else if (x.scalar_type() == at::ScalarType::Half) {
+        if (BNT == 128) MGF_LAUNCH(__half, 128);
+        else MGF_LAUNCH(__half, 256);
+    } else {
+        TORCH_CHECK(false, "moe_grouped_gemm_fp8: bf16/f16 only");
+    }
+#undef MGF_LAUNCH
     HIP_CHECK_KERNEL();
 }
 

@@ -115,6 +115,10 @@ def quantize_model_fp8(model: nn.Module, skip_names=("lm_head",)):
                 continue
             mod.quantize_fp8()
             n += 1
+        elif hasattr(mod, "quantize_fp8") and type(mod).__name__ == "ExpertMLPs":
+            # stacked MoE expert weights: fp8 grouped kernel path
+            if mod.quantize_fp8():
+                n += 1
     return n
 
 

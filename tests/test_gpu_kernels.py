@@ -1104,3 +1104,69 @@ def test_moe_grouped_gemm_fusions_gpu():
                                combine_scale=gw_tab, combine_gather=gtab)
     torch.testing.assert_close(got.to(torch.bfloat16).float(), want.float(),
                                atol=5e-2, rtol=5e-2)
+
+
+def test_moe_grouped_fp8_kernel_gpu():
+    """fp8-e4m3 grouped GEMM (all-glds staging, LDS-read dequant,
+    epilogue per-channel scale) vs the CPU dequant reference, incl. the
+    gather/combine fusions."""
+    torch.manual_seed(7)
+    dev = _dev()
+    E, T, top_k, K, N = 4, 150, 2, 512, 768
+    idx = torch.randint(0, E, (T * top_k,), device=dev)
+    order, dest, eids, gtab, Mpad = ops.moe_grouped_tables(idx, E, top_k)
+    x = torch.randn(T, K, dtype=torch.bfloat16, device=dev) * 0.3
+    w = torch.randn(E, N, K) * 0.1
+    wq, ws = ops.fp8_moe_quantize(w)
+    wq, ws = wq.to(dev), ws.to(dev)
+
+    # CPU dequant reference over the same tables
+    y_cpu = ops.moe_grouped_gemm_fp8(
+        torch.zeros(Mpad, K, dtype=torch.bfloat16).index_copy_(
+            0, dest.cpu(), x.cpu().index_select(0, (order // top_k).cpu())),
+        wq.cpu(), ws.cpu(), eids.cpu())
+
+    for bn in (128, 256):
+        y = ops.moe_grouped_gemm_fp8(x, wq, ws, eids, bn=bn, x_gather=gtab)
+        d = (y.cpu().index_select(0, dest.cpu()).float()
+             - y_cpu.index_select(0, dest.cpu()).float())
+        rel = d.abs().max() / y_cpu.float().abs().max().clamp(min=1e-6)
+        assert rel < 3e-2, (bn, float(rel))
+
+    # fused combine
+    gatew = torch.rand(T * top_k, device=dev)[order]
+    gw = torch.zeros(Mpad, dtype=torch.float32, device=dev)
+    gw[dest] = gatew.float()
+    out32 = torch.zeros(T, N, dtype=torch.float32, device=dev)
+    got = ops.moe_grouped_gemm_fp8(x, wq, ws, eids, x_gather=gtab,
+                                   combine_out=out32, combine_scale=gw,
+                                   combine_gather=gtab)
+    want = torch.zeros(T, N)
+    contrib = y_cpu.index_select(0, dest.cpu()).float() * \
+        gatew.cpu().float().unsqueeze(-1)
+    want.index_add_(0, (order // top_k).cpu(), contrib)
+    rel = (got.cpu() - want).abs().max() / want.abs().max().clamp(min=1e-6)
+    assert rel < 3e-2, float(rel)
+
+
+def test_moe_fp8_module_gpu():
+    """ExpertMLPs.quantize_fp8 end-to-end on GPU vs the bf16 dense path."""
+    from neuronx_distributed_inference_amd.modules.moe import ExpertMLPs
+
+    torch.manual_seed(8)
+    dev = _dev()
+    E, H, I, T, top_k = 8, 512, 896, 64, 2
+    m = ExpertMLPs(num_experts=E, hidden_size=H, intermediate_size=I,
+                   tp_degree=1, dtype=torch.bfloat16)
+    with torch.no_grad():
+        for p in m.parameters():
+            p.normal_(0, 0.05)
+    m = m.to(dev)
+    x = torch.randn(T, H, dtype=torch.bfloat16, device=dev) * 0.3
+    w, idx = torch.softmax(torch.randn(T, E, device=dev), -1).topk(top_k, -1)
+    want = m._forward_dense(x, w, idx)
+    assert m.quantize_fp8()
+    got = m(x, w, idx)
+    rel = ((got.float() - want.float()).abs().max()
+           / want.float().abs().max().clamp(min=1e-3))
+    assert rel < 0.08, float(rel)

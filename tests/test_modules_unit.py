@@ -625,3 +625,30 @@ def test_moe_grouped_gemm_fusions_cpu_ref():
     got = ops.moe_grouped_gemm(xs, w, eids, combine_out=out32,
                                combine_scale=gw_tab, combine_gather=gtab)
     torch.testing.assert_close(got, want, atol=1e-4, rtol=1e-4)
+
+
+def test_moe_fp8_grouped_cpu_matches_bf16():
+    """ExpertMLPs fp8 mode (e4m3 weights + per-channel scales through the
+    grouped path) vs the bf16 dense path on CPU."""
+    import torch
+
+    from neuronx_distributed_inference_amd.modules.moe import ExpertMLPs
+
+    torch.manual_seed(6)
+    E, H, I, T, top_k = 4, 256, 512, 20, 2
+    m = ExpertMLPs(num_experts=E, hidden_size=H, intermediate_size=I,
+                   tp_degree=1, dtype=torch.float32)
+    with torch.no_grad():
+        for p in m.parameters():
+            p.normal_(0, 0.05)
+    x = torch.randn(T, H) * 0.3
+    w, idx = torch.softmax(torch.randn(T, E), -1).topk(top_k, dim=-1)
+    want = m._forward_dense(x, w, idx)
+    assert m.quantize_fp8()
+    assert not any(n in dict(m.named_parameters())
+                   for n in ("w_gate", "w_up", "w_down"))
+    got = m._forward_grouped(x, w, idx)
+    # e4m3 weight round-off: ~2^-3 relative per weight, averaged down by
+    # the K-dim fan-in
+    err = (got - want).abs().max() / want.abs().max().clamp(min=1e-6)
+    assert err < 0.06, float(err)
