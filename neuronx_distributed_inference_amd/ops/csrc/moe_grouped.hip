@@ -445,14 +445,17 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_mx4_kernel(
 // FP8 (OCP e4m3fn) grouped expert GEMM: weights stored 1 byte/elem with
 // ONE fp32 scale per output channel (E, N). Unlike the MXFP4 variant
 // (register-staged packed W -> mixed-load vmcnt drains), this keeps the
-// ALL-GLDS pipeline: the raw fp8 bytes stage through LDS exactly like
-// bf16 (half-width [rows][64B] images, 16 rows per 1 KB glds piece),
-// and the B fragment dequantizes on the LDS->register read (4x
-// cvt_pk_f32_fp8 per fragment — trivial next to the chunk's DMA time).
-// The per-channel scale multiplies the fp32 accumulator once in the
-// epilogue (scale is constant along K, so post-scaling is exact).
-// Restriction: BNT in {128, 256} so every wave stages the same number
-// of W pieces (uniform vmcnt counts).
+// ALL-GLDS pipeline — and it uses a DOUBLE-DEPTH K chunk (BK=128): the
+// grouped pipeline is chunk-count-bound, not byte-bound (a 64-deep fp8
+// chunk moves half the bytes in the SAME time — measured: no speedup),
+// so the fp8 kernel stages 128 k-elems per ring step: W is then the
+// exact [rows][128B] XOR-swizzled image of the bf16 kernel, x stages as
+// TWO 128B half-images, and each step runs 4 MFMA k-slices. Chunk count
+// halves; the 2x weight-stream saving becomes wall-clock.
+// The B fragment dequantizes on the LDS->register read (4x
+// cvt_pk_f32_fp8) and the per-channel scale multiplies the fp32
+// accumulator once in the epilogue (scale is constant along K).
+// Restrictions: BNT in {128, 256}; K % 128 == 0.
 
 typedef float mg_f32x2 __attribute__((ext_vector_type(2)));
 
@@ -486,11 +489,12 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_fp8_kernel(
     using MF = MgMfma<T>;
     using vec8 = typename MF::vec8;
     constexpr int MT = 64;
-    constexpr int NWG = BNT / 128;  // W glds per wave per chunk (16 rows each)
-    constexpr int G = 1 + NWG;
+    constexpr int BKF = 128;        // k-elems per ring step (double depth)
+    constexpr int NWG = BNT / 64;   // W glds per wave per step (8 rows x 128B)
+    constexpr int G = 2 + NWG;      // + TWO x half-image glds
     constexpr int NFRAG = BNT / 64;
-    constexpr int XBYTES = MT * 128;
-    constexpr int CHUNK_B = XBYTES + BNT * 64;
+    constexpr int XBYTES = 2 * MT * 128;   // two 128B half-images
+    constexpr int CHUNK_B = XBYTES + BNT * 128;
     static_assert(BNT == 128 || BNT == 256, "fp8 grouped: bn 128/256 only");
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -501,7 +505,7 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_fp8_kernel(
     if (e < 0) return;
     const int n0 = tile * BNT;
     const int m0 = rt * MT;
-    const int nchunks = K / MG_BK;
+    const int nchunks = K / BKF;
 
     const int lane = threadIdx.x & 63;
     const int wid = threadIdx.x >> 6;
@@ -515,15 +519,13 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_fp8_kernel(
     const int xslot = (lane & 7) ^ (lane >> 3);
     const char* xsrc_lane = reinterpret_cast<const char*>(x) +
                             (long)xr * K * sizeof(T) + xslot * 16;
-    // W piece: 16 rows x 64 B; this lane covers row lane/4, 16B-slot
-    // lane%4 (linear slots — the B read is 8 B so 4-way bank overlap,
-    // irrelevant next to the DMA-bound chunk time)
-    const char* wsrc_lane[NWG];
+    // W rows: 1 byte/elem so a 128-deep chunk is the bf16 kernel's
+    // [rows][128B] XOR-swizzled image verbatim
+    const char* wsrc[NWG];
 #pragma unroll
     for (int i = 0; i < NWG; ++i)
-        wsrc_lane[i] = reinterpret_cast<const char*>(wq) +
-                       ((long)e * N + n0 + wid * (16 * NWG) + i * 16 +
-                        (lane >> 2)) * K + (lane & 3) * 16;
+        wsrc[i] = reinterpret_cast<const char*>(wq) +
+                  ((long)e * N + n0 + wid * (8 * NWG) + i * 8) * K;
 
     char* xdst[MG_NBUF];
     char* wdst[MG_NBUF];
@@ -533,12 +535,15 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_fp8_kernel(
         wdst[b] = smem + b * CHUNK_B + XBYTES + wid * (NWG * 1024);
     }
 
+    // x half-images: k [0,64) at offset 0, k [64,128) at +MT*128
 #define F8_STAGE(t, b)                                                       \
     do {                                                                     \
-        mg_glds_lane128(xsrc_lane + (long)(t) * 128, xdst[b]);               \
+        const long xoff = (long)(t) * 256;                                   \
+        mg_glds_lane128(xsrc_lane + xoff, xdst[b]);                          \
+        mg_glds_lane128(xsrc_lane + xoff + 128, xdst[b] + MT * 128);         \
         _Pragma("unroll") for (int i = 0; i < NWG; ++i)                      \
-            mg_glds_lane128(wsrc_lane[i] + (long)(t) * 64,                   \
-                            wdst[b] + i * 1024);                             \
+            mg_glds_rows128<2>(wsrc[i] + (long)(t) * 128, (long)K,           \
+                               wdst[b] + i * 1024);                          \
     } while (0)
 
     mg_f32x4 acc[2][NFRAG];
@@ -547,26 +552,32 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_fp8_kernel(
 #pragma unroll
         for (int nj = 0; nj < NFRAG; ++nj) acc[qi][nj] = {0.f, 0.f, 0.f, 0.f};
 
+    // 4 MFMA k-slices per step: slice kq reads x half-image kq/2 with
+    // local ks = kq%2; the W byte group is 16B-slot (kq*2 + lslot/2)
+    // XOR-swizzled, byte-half lslot&1
 #define F8_FRAGS(b)                                                          \
     do {                                                                     \
         const char* xb = smem + (b) * CHUNK_B;                               \
         const char* wb = xb + XBYTES;                                        \
-        _Pragma("unroll") for (int ks = 0; ks < 2; ++ks) {                   \
+        _Pragma("unroll") for (int kq = 0; kq < 4; ++kq) {                   \
+            const char* xh = xb + (kq >> 1) * (MT * 128);                    \
+            const int ks = kq & 1;                                           \
             vec8 a0, a1, bb[NFRAG];                                          \
             {                                                                \
                 const int m = wm0 + lrow;                                    \
                 const int s = (ks * 4 + lslot) ^ (m & 7);                    \
-                a0 = *reinterpret_cast<const vec8*>(xb + m * 128 + s * 16);  \
+                a0 = *reinterpret_cast<const vec8*>(xh + m * 128 + s * 16);  \
             }                                                                \
             {                                                                \
                 const int m = wm0 + 16 + lrow;                               \
                 const int s = (ks * 4 + lslot) ^ (m & 7);                    \
-                a1 = *reinterpret_cast<const vec8*>(xb + m * 128 + s * 16);  \
+                a1 = *reinterpret_cast<const vec8*>(xh + m * 128 + s * 16);  \
             }                                                                \
             _Pragma("unroll") for (int nj = 0; nj < NFRAG; ++nj) {           \
                 const int n = wn0 + nj * 16 + lrow;                          \
+                const int s16 = (kq * 2 + (lslot >> 1)) ^ (n & 7);           \
                 const unsigned* p2 = reinterpret_cast<const unsigned*>(      \
-                    wb + n * 64 + (ks * 4 + lslot) * 8);                     \
+                    wb + n * 128 + s16 * 16 + (lslot & 1) * 8);              \
                 T tmp[8];                                                    \
                 fp8x8_to_vec8<T>(p2, tmp);                                   \
                 bb[nj] = *reinterpret_cast<const vec8*>(tmp);                \
@@ -844,7 +855,8 @@ void moe_grouped_gemm_fp8(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
     const int N = wq.size(1);
     TORCH_CHECK(wq.size(2) == K && wscale.size(1) == N, "packed shape");
     TORCH_CHECK(y.size(0) == Mpad && y.size(1) == N, "shape");
-    TORCH_CHECK(K % MG_BK == 0 && Mpad % 64 == 0, "align");
+    TORCH_CHECK(K % 128 == 0 && Mpad % 64 == 0,
+                "fp8 grouped: K must be a multiple of 128");
     TORCH_CHECK(wq.scalar_type() == at::ScalarType::Byte &&
                 wscale.scalar_type() == at::ScalarType::Float, "wq u8 + f32");
     int BNT = (int)bn;
@@ -865,7 +877,7 @@ void moe_grouped_gemm_fp8(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
 
 #define MGF_LAUNCH(T, BNTV)                                                   \
     do {                                                                      \
-        constexpr int LDS = MG_NBUF * (64 * 128 + BNTV * 64);                 \
+        constexpr int LDS = MG_NBUF * (2 * 64 * 128 + BNTV * 128);                 \
         static bool attr_set = false;                                         \
         if (!attr_set) {                                                      \
             (void)hipFuncSetAttribute(                                        \
