@@ -685,7 +685,8 @@ def moe_grouped_tables(idx_flat: torch.Tensor, num_experts: int,
 def moe_grouped_gemm(x_sorted: torch.Tensor, w: torch.Tensor,
                      expert_ids: torch.Tensor, bias=None,
                      bn: int = 0, x_gather=None, combine_out=None,
-                     combine_scale=None, combine_gather=None) -> torch.Tensor:
+                     combine_scale=None, combine_gather=None,
+                     bk: int = 0) -> torch.Tensor:
     """y_sorted[m] = x_sorted[m] @ w[expert_of(m)]^T (+ bias[expert]).
     x_sorted (Mpad, K), w (E, N, K), expert_ids (Mpad/64,) int32, bias
     (E, N) or None. GPU: one grouped HIP kernel; CPU reference:
@@ -715,7 +716,7 @@ def moe_grouped_gemm(x_sorted: torch.Tensor, w: torch.Tensor,
                            combine_out,
                            combine_scale,
                            None if combine_gather is None
-                           else combine_gather.contiguous())
+                           else combine_gather.contiguous(), bk)
         return combine_out if combine_out is not None else y
     # ---- CPU reference ----
     out = combine_out

@@ -53,7 +53,7 @@ void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                       c10::optional<torch::Tensor> x_gather,
                       c10::optional<torch::Tensor> combine_out,
                       c10::optional<torch::Tensor> combine_scale,
-                      c10::optional<torch::Tensor> combine_gather);
+                      c10::optional<torch::Tensor> combine_gather, long bk);
 void moe_grouped_tables_gpu(torch::Tensor order, torch::Tensor dest,
                             torch::Tensor expert_ids, torch::Tensor gtab,
                             torch::Tensor idx_flat, long num_experts,
@@ -137,7 +137,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("x_gather") = py::none(),
           py::arg("combine_out") = py::none(),
           py::arg("combine_scale") = py::none(),
-          py::arg("combine_gather") = py::none());
+          py::arg("combine_gather") = py::none(), py::arg("bk") = 0);
     m.def("moe_grouped_gemm_fp8", &moe_grouped_gemm_fp8,
           "grouped expert GEMM, fp8-e4m3 weights dequantized at the LDS "
           "fragment read, per-channel scale in the epilogue (CDNA4)",

@@ -80,7 +80,11 @@ DEV_INLINE void mg_glds_lane128(const char* src_lane, char* lds_dst_base) {
 // (x is re-read once per column tile) and raises MFMA work per staged
 // byte; BNT=256 costs occupancy 1 (120 KB LDS) but each block then
 // streams 4x the W per x fetch.
-template <typename T, int BNT>
+// BKT: k-elems per ring step (64 or 128). The pipeline is chunk-COUNT
+// bound at these tile sizes (fixed wait/barrier cost per step), so the
+// deeper step — staged as (BKT/64) 128B half-images per operand —
+// halves the step count where LDS permits (bn<=128 at BKT=128).
+template <typename T, int BNT, int BKT = 64>
 __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
     T* __restrict__ y,              // (Mpad, N) sorted-space output
     const T* __restrict__ x,        // (Mpad, K) sorted — or (T, K) with x_gather
@@ -95,11 +99,12 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
     using MF = MgMfma<T>;
     using vec8 = typename MF::vec8;
     constexpr int MT = 64;
-    constexpr int NWG = BNT / 64;   // W glds per wave per chunk (8 rows each)
-    constexpr int G = 1 + NWG;      // total glds per wave per chunk
+    constexpr int NH = BKT / 64;    // 128B half-images per operand per step
+    constexpr int NWG = BNT / 64;   // W glds per wave per half-image
+    constexpr int G = (1 + NWG) * NH;  // total glds per wave per step
     constexpr int NFRAG = BNT / 64; // 16-col fragments per wave
-    constexpr int XBYTES = MT * 128;
-    constexpr int CHUNK_B = XBYTES + BNT * 128;
+    constexpr int XBYTES = NH * MT * 128;
+    constexpr int CHUNK_B = XBYTES + NH * BNT * 128;
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
 
@@ -109,7 +114,7 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
     if (e < 0) return;              // beyond this step's padded rows
     const int n0 = tile * BNT;
     const int m0 = rt * MT;
-    const int nchunks = K / MG_BK;
+    const int nchunks = K / BKT;
 
     const int lane = threadIdx.x & 63;
     const int wid = threadIdx.x >> 6;
@@ -141,11 +146,15 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
 
 #define MG_STAGE(t, b)                                                       \
     do {                                                                     \
-        const long koff = (long)(t) * 128;                                   \
-        mg_glds_lane128(xsrc_lane + koff, xdst[b]);                          \
-        _Pragma("unroll") for (int i = 0; i < NWG; ++i)                      \
-            mg_glds_rows128<2>(wsrc[i] + koff, (long)K * sizeof(T),          \
-                               wdst[b] + i * 1024);                          \
+        const long koff = (long)(t) * (BKT * 2);                             \
+        _Pragma("unroll") for (int h = 0; h < NH; ++h) {                     \
+            mg_glds_lane128(xsrc_lane + koff + h * 128,                      \
+                            xdst[b] + h * (MT * 128));                       \
+            _Pragma("unroll") for (int i = 0; i < NWG; ++i)                  \
+                mg_glds_rows128<2>(wsrc[i] + koff + h * 128,                 \
+                                   (long)K * sizeof(T),                      \
+                                   wdst[b] + h * (BNT * 128) + i * 1024);    \
+        }                                                                    \
     } while (0)
 
     mg_f32x4 acc[2][NFRAG];
@@ -156,9 +165,12 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
 
 #define MG_FRAGS(b)                                                          \
     do {                                                                     \
-        const char* xb = smem + (b) * CHUNK_B;                               \
-        const char* wb = xb + XBYTES;                                        \
-        _Pragma("unroll") for (int ks = 0; ks < 2; ++ks) {                   \
+        _Pragma("unroll") for (int kq = 0; kq < 2 * NH; ++kq) {              \
+            const char* xb =                                                 \
+                smem + (b) * CHUNK_B + (kq >> 1) * (MT * 128);               \
+            const char* wb =                                                 \
+                smem + (b) * CHUNK_B + XBYTES + (kq >> 1) * (BNT * 128);     \
+            const int ks = kq & 1;                                           \
             vec8 a0, a1, bb[NFRAG];                                          \
             {                                                                \
                 const int m = wm0 + lrow;                                    \
@@ -749,7 +761,7 @@ void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                       c10::optional<torch::Tensor> x_gather,
                       c10::optional<torch::Tensor> combine_out,
                       c10::optional<torch::Tensor> combine_scale,
-                      c10::optional<torch::Tensor> combine_gather) {
+                      c10::optional<torch::Tensor> combine_gather, long bk) {
     TORCH_CHECK(y.dim() == 2 && x.dim() == 2 && w.dim() == 3, "shapes");
     TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && y.is_contiguous() &&
                 expert_ids.is_contiguous(), "contiguous tensors required");
@@ -789,21 +801,30 @@ void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
     TORCH_CHECK(BNT == 64 || BNT == 128 || BNT == 192 || BNT == 256,
                 "bn must be 64/128/192/256");
     TORCH_CHECK(N % BNT == 0, "N must be a multiple of bn");
+    // bk: k-depth per ring step. 128 halves the step count (the pipeline
+    // is step-count-bound) but its LDS only fits at bn <= 128.
+    int BKT = (int)bk;
+    if (BKT == 0) BKT = 64;
+    TORCH_CHECK(BKT == 64 || BKT == 128, "bk must be 64/128");
+    TORCH_CHECK(BKT == 64 || BNT <= 128, "bk=128 needs bn<=128 (LDS)");
+    TORCH_CHECK(K % (MG_BK * (BKT / 64)) == 0, "K must be a multiple of bk");
 
     auto stream = at::hip::getCurrentHIPStream();
     dim3 grid(N / BNT, Mpad / 64);
 
-#define MG_LAUNCH(T, BNTV)                                                    \
+#define MG_LAUNCH(T, BNTV, BKTV)                                              \
     do {                                                                      \
-        constexpr int LDS = MG_NBUF * (64 * 128 + BNTV * 128);                \
+        constexpr int LDS =                                                   \
+            MG_NBUF * (BKTV / 64) * (64 * 128 + BNTV * 128);                  \
         static bool attr_set = false;                                         \
         if (!attr_set) {                                                      \
             (void)hipFuncSetAttribute(                                        \
-                (const void*)moe_grouped_kernel<T, BNTV>,                     \
+                (const void*)moe_grouped_kernel<T, BNTV, BKTV>,               \
                 hipFuncAttributeMaxDynamicSharedMemorySize, LDS);             \
             attr_set = true;                                                  \
         }                                                                     \
-        hipLaunchKernelGGL((moe_grouped_kernel<T, BNTV>), grid, dim3(512),    \
+        hipLaunchKernelGGL((moe_grouped_kernel<T, BNTV, BKTV>), grid,         \
+                           dim3(512),                                         \
                            LDS, stream, (T*)y.data_ptr(),                     \
                            (const T*)x.data_ptr(),                            \
                            (const T*)w.data_ptr(), expert_ids.data_ptr<int>(),\
@@ -819,10 +840,13 @@ void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
     } while (0)
 #define MG_T(T)                                                               \
     do {                                                                      \
-        if (BNT == 64) MG_LAUNCH(T, 64);                                      \
-        else if (BNT == 128) MG_LAUNCH(T, 128);                               \
-        else if (BNT == 192) MG_LAUNCH(T, 192);                               \
-        else MG_LAUNCH(T, 256);                                               \
+        if (BKT == 128) {                                                     \
+            if (BNT == 64) MG_LAUNCH(T, 64, 128);                             \
+            else MG_LAUNCH(T, 128, 128);                                      \
+        } else if (BNT == 64) MG_LAUNCH(T, 64, 64);                           \
+        else if (BNT == 128) MG_LAUNCH(T, 128, 64);                           \
+        else if (BNT == 192) MG_LAUNCH(T, 192, 64);                           \
+        else MG_LAUNCH(T, 256, 64);                                           \
     } while (0)
 
     if (x.scalar_type() == at::ScalarType::BFloat16) {

@@ -52,22 +52,24 @@ def main():
     wmat = m.w_gate
     # numerics + perf for each column-tile width
     ref_g = ops.moe_grouped_gemm(x_sorted, wmat, expert_ids, bn=64)
-    for bn in (64, 128, 256):
-        gg = ops.moe_grouped_gemm(x_sorted, wmat, expert_ids, bn=bn)
-        err = (gg.float() - ref_g.float()).abs().max().item()
-        print(f"gemm gate (N={I}) bn={bn}: "
-              f"{t(lambda: ops.moe_grouped_gemm(x_sorted, wmat, expert_ids, bn=bn)):8.1f} us"
-              f"  maxdiff-vs-bn64 {err:.4f}  [{Mpad} rows; weight {E*I*H*2/1e6:.0f} MB]")
+    for bn, bk in ((64, 64), (128, 64), (256, 64), (64, 128), (128, 128)):
+        gg = ops.moe_grouped_gemm(x_sorted, wmat, expert_ids, bn=bn, bk=bk)
+        err = (gg.index_select(0, dest).float()
+               - ref_g.index_select(0, dest).float()).abs().max().item()
+        print(f"gemm gate (N={I}) bn={bn} bk={bk}: "
+              f"{t(lambda: ops.moe_grouped_gemm(x_sorted, wmat, expert_ids, bn=bn, bk=bk)):8.1f} us"
+              f"  maxdiff {err:.4f}  [{Mpad} rows; weight {E*I*H*2/1e6:.0f} MB]")
     g = ops.moe_grouped_gemm(x_sorted, wmat, expert_ids)
     u = ops.moe_grouped_gemm(x_sorted, m.w_up, expert_ids)
     h = ops.silu_mul(g, u)
     ref_d = ops.moe_grouped_gemm(h, m.w_down, expert_ids, bn=64)
-    for bn in (64, 128, 256):
-        dd = ops.moe_grouped_gemm(h, m.w_down, expert_ids, bn=bn)
-        err = (dd.float() - ref_d.float()).abs().max().item()
-        print(f"gemm down (N={H}) bn={bn}: "
-              f"{t(lambda: ops.moe_grouped_gemm(h, m.w_down, expert_ids, bn=bn)):8.1f} us"
-              f"  maxdiff-vs-bn64 {err:.4f}")
+    for bn, bk in ((64, 64), (128, 64), (256, 64), (64, 128), (128, 128)):
+        dd = ops.moe_grouped_gemm(h, m.w_down, expert_ids, bn=bn, bk=bk)
+        err = (dd.index_select(0, dest).float()
+               - ref_d.index_select(0, dest).float()).abs().max().item()
+        print(f"gemm down (N={H}) bn={bn} bk={bk}: "
+              f"{t(lambda: ops.moe_grouped_gemm(h, m.w_down, expert_ids, bn=bn, bk=bk)):8.1f} us"
+              f"  maxdiff {err:.4f}")
     y = ops.moe_grouped_gemm(h, m.w_down, expert_ids)
 
     def combine():

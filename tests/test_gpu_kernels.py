@@ -1170,3 +1170,26 @@ def test_moe_fp8_module_gpu():
     rel = ((got.float() - want.float()).abs().max()
            / want.float().abs().max().clamp(min=1e-3))
     assert rel < 0.08, float(rel)
+
+
+@pytest.mark.parametrize("bnbk", [(64, 128), (128, 128)])
+def test_moe_grouped_gemm_bk128(bnbk):
+    """Double-depth (BK=128) chunk variant vs the per-expert reference."""
+    bn, bk = bnbk
+    torch.manual_seed(2)
+    dev = _dev()
+    E, N, K = 4, 768, 640
+    idx = torch.randint(0, E, (200,), device=dev)
+    order, dest, expert_ids, gtab, Mpad = ops.moe_grouped_tables(idx, E)
+    x_rows = torch.randn(200, K, dtype=torch.bfloat16, device=dev) * 0.1
+    w = torch.randn(E, N, K, dtype=torch.bfloat16, device=dev) * 0.1
+    x_sorted = torch.zeros(Mpad, K, dtype=torch.bfloat16, device=dev)
+    x_sorted.index_copy_(0, dest, x_rows.index_select(0, order))
+    y = ops.moe_grouped_gemm(x_sorted, w, expert_ids, bn=bn, bk=bk)
+    got = y.index_select(0, dest)
+    se = idx[order]
+    want = torch.empty_like(got)
+    for e in range(E):
+        m = se == e
+        want[m] = x_sorted[dest[m]] @ w[e].t()
+    torch.testing.assert_close(got.float(), want.float(), atol=2e-2, rtol=2e-2)
