@@ -243,6 +243,35 @@ def skinny_gemm5(x2: torch.Tensor, weight: torch.Tensor, bias=None, rs=None,
     return y
 
 
+_SK5F8_WS = {}
+
+
+def skinny_gemm5_fp8(x2: torch.Tensor, wq_u8: torch.Tensor,
+                     wscale: torch.Tensor, bias=None,
+                     splitk: int = 1) -> torch.Tensor:
+    """y = x2 @ dequant(wq)^T (+bias): fp8-e4m3 weights (N, K) u8 with a
+    per-channel fp32 scale, decode batches (M <= 256). Double-depth
+    (BK=128) all-glds pipeline — see sk5_fp8_kernel."""
+    c = _require_ext("skinny_gemm5_fp8")
+    M, K = x2.shape
+    N = wq_u8.shape[0]
+    y = torch.empty(M, N, dtype=x2.dtype, device=x2.device)
+    part = None
+    if splitk > 1:
+        rows = ((M + 127) // 128) * 128
+        key = (rows, N, splitk, x2.device)
+        part = _SK5F8_WS.get(key)
+        if part is None or part.numel() < splitk * rows * N:
+            part = torch.empty(splitk, rows, N, dtype=torch.float32,
+                               device=x2.device)
+            _SK5F8_WS[key] = part
+    c.skinny_gemm5_fp8(y, x2, wq_u8,
+                       wscale.reshape(-1).contiguous().float(),
+                       None if bias is None else bias.contiguous(),
+                       part, splitk)
+    return y
+
+
 def _sg5_candidates(M: int, N: int, K: int):
     """(mt, splitk) variants: splitk sized so total blocks land near
     0.5-2x the 256 CUs (guide: per-block latency ~constant in N); row

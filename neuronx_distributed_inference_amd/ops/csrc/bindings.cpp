@@ -43,6 +43,9 @@ void skinny_gemm5(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                   c10::optional<torch::Tensor> bias,
                   c10::optional<torch::Tensor> rs,
                   c10::optional<torch::Tensor> part, long splitk, long mode, long mt);
+void skinny_gemm5_fp8(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
+                      torch::Tensor wscale, c10::optional<torch::Tensor> bias,
+                      c10::optional<torch::Tensor> part, long splitk);
 void fp8_quant(torch::Tensor x, torch::Tensor out, torch::Tensor scale,
                torch::Tensor amax_ws);
 void fp8_epilogue(torch::Tensor out, torch::Tensor y32, torch::Tensor w_scale,
@@ -115,6 +118,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("y"), py::arg("x"), py::arg("w"), py::arg("bias") = py::none(),
           py::arg("rs") = py::none(), py::arg("part") = py::none(),
           py::arg("splitk") = 1, py::arg("mode") = 0, py::arg("mt") = 0);
+    m.def("skinny_gemm5_fp8", &skinny_gemm5_fp8,
+          "fp8-weight skinny decode GEMM: double-depth chunks, LDS-read "
+          "dequant, per-channel scale (CDNA4)",
+          py::arg("y"), py::arg("x"), py::arg("wq"), py::arg("wscale"),
+          py::arg("bias") = py::none(), py::arg("part") = py::none(),
+          py::arg("splitk") = 1);
     m.def("skinny_gemm3", &skinny_gemm3,
           "skinny-M GEMM v3: full-K register accum, depth-3 prefetch (CDNA4)",
           py::arg("y"), py::arg("x"), py::arg("w"), py::arg("bias") = py::none());

@@ -338,6 +338,212 @@ __global__ __launch_bounds__(512, 2) void sk5_kernel(
     }
 }
 
+// ---------------------------------------------------------------------------
+// FP8-weight skinny GEMM (sk5f8): y = x(bf16) @ dequant(W fp8)^T with a
+// per-output-channel scale. Same verified pipeline as sk5 but with a
+// DOUBLE-DEPTH chunk (BK=128): 1-byte weights make a 128-deep chunk the
+// same [64 rows][128B] XOR-swizzled image as sk5's bf16 chunk, so the
+// ring moves the SAME bytes per step over HALF the steps (the pipeline
+// is step-count-bound at these sizes — measured on the grouped-MoE fp8
+// kernel: -26% Mixtral step). x stages as two 128B half-images.
+// MT is fixed at 128 (LDS: 3 x (32 KB x + 8 KB W) = 120 KB); M > 128
+// runs 2 row-blocks whose duplicate W stream hits the shared LLC.
+// Dequant = 4x cvt_pk_f32_fp8 at the LDS fragment read; the channel
+// scale multiplies the accumulator once in the epilogue.
+
+typedef float sk5_f32x2 __attribute__((ext_vector_type(2)));
+
+template <typename T>
+DEV_INLINE void sk5_fp8x8(const unsigned* p2, T* out) {
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+        const unsigned w = p2[h];
+        sk5_f32x2 lo = __builtin_amdgcn_cvt_pk_f32_fp8(w, false);
+        sk5_f32x2 hi = __builtin_amdgcn_cvt_pk_f32_fp8(w, true);
+        out[4 * h + 0] = VecTraits<T>::from_float(lo[0]);
+        out[4 * h + 1] = VecTraits<T>::from_float(lo[1]);
+        out[4 * h + 2] = VecTraits<T>::from_float(hi[0]);
+        out[4 * h + 3] = VecTraits<T>::from_float(hi[1]);
+    }
+}
+
+// per-lane-addressed glds (the caller bakes row + slot into the address)
+DEV_INLINE void sk5_glds_lane(const char* src_lane, char* lds_dst_base) {
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src_lane,
+        (__attribute__((address_space(3))) unsigned int*)lds_dst_base, 16, 0, 0);
+}
+
+template <typename T>
+__global__ __launch_bounds__(512, 2) void sk5_fp8_kernel(
+    T* __restrict__ y,             // (M, N) when splitk == 1
+    float* __restrict__ part,      // (splitk, mblocks*128, N) fp32
+    const T* __restrict__ x,       // (M, K) bf16/f16
+    const unsigned char* __restrict__ wq,  // (N, K) e4m3fn
+    const float* __restrict__ wscale,      // (N,)
+    const T* __restrict__ bias,    // (N,) or null (splitk==1)
+    const int M, const int N, const int K, const int splitk) {
+    using MF = Sk5Mfma<T>;
+    using vec8 = typename MF::vec8;
+    constexpr int MT = 128;
+    constexpr int BKF = 128;
+    constexpr int NXG = 2;               // x glds per wave per half-image
+    constexpr int G = 2 * NXG + 1;       // 4 x + 1 W per wave per step
+    constexpr int XBYTES = 2 * MT * 128; // two half-images
+    constexpr int CHUNK_B = XBYTES + BN * 128;
+    constexpr int NFRAG = 2;             // BN/WN/16 with WN=2
+
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+
+    const int tile = blockIdx.x / splitk;
+    const int slice = blockIdx.x % splitk;
+    const int n0 = tile * BN;
+    const int m0blk = blockIdx.y * MT;
+    const int kspan = K / splitk;
+    const int k_lo = slice * kspan;
+    const int nchunks = kspan / BKF;
+
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int wm0 = (wid >> 1) * 32;     // 4 waves along M
+    const int wn0 = (wid & 1) * 32;      // 2 waves along N
+    const int lrow = lane & 15;
+    const int lslot = lane >> 4;
+
+    // x: per-lane source addresses (row clamp for ragged M)
+    const int xslot = (lane & 7) ^ (lane >> 3);
+    const char* xsrc_lane[NXG];
+#pragma unroll
+    for (int i = 0; i < NXG; ++i) {
+        const int r = min(m0blk + wid * 16 + i * 8 + (lane >> 3), M - 1);
+        xsrc_lane[i] = reinterpret_cast<const char*>(x) +
+                       ((long)r * K + k_lo) * sizeof(T) + xslot * 16;
+    }
+    // W: this wave stages rows n0 + wid*8 .. +8 (one 1 KB piece per step)
+    const char* wsrc = reinterpret_cast<const char*>(wq) +
+                       (long)(n0 + wid * 8) * K + k_lo;
+
+    char* xdst[NBUF];
+    char* wdst[NBUF];
+#pragma unroll
+    for (int b = 0; b < NBUF; ++b) {
+        xdst[b] = smem + b * CHUNK_B + wid * (NXG * 1024);
+        wdst[b] = smem + b * CHUNK_B + XBYTES + wid * 1024;
+    }
+
+#define F8S_STAGE(t, b)                                                      \
+    do {                                                                     \
+        const long xoff = (long)(t) * 256;                                   \
+        _Pragma("unroll") for (int h = 0; h < 2; ++h)                        \
+            _Pragma("unroll") for (int i = 0; i < NXG; ++i)                  \
+                sk5_glds_lane(xsrc_lane[i] + xoff + h * 128,                 \
+                              xdst[b] + h * (MT * 128) + i * 1024);          \
+        glds_rows128<2>(wsrc + (long)(t) * 128, (long)K, wdst[b]);           \
+    } while (0)
+
+    sk5_f32x4 acc[2][NFRAG];
+#pragma unroll
+    for (int qi = 0; qi < 2; ++qi)
+#pragma unroll
+        for (int nj = 0; nj < NFRAG; ++nj) acc[qi][nj] = {0.f, 0.f, 0.f, 0.f};
+
+#define F8S_FRAGS(b)                                                         \
+    do {                                                                     \
+        const char* wb = smem + (b) * CHUNK_B + XBYTES;                      \
+        _Pragma("unroll") for (int kq = 0; kq < 4; ++kq) {                   \
+            const char* xh =                                                 \
+                smem + (b) * CHUNK_B + (kq >> 1) * (MT * 128);               \
+            const int ks = kq & 1;                                           \
+            vec8 a[2], bb[NFRAG];                                            \
+            _Pragma("unroll") for (int qi = 0; qi < 2; ++qi) {               \
+                const int m = wm0 + qi * 16 + lrow;                          \
+                const int sx = (ks * 4 + lslot) ^ (m & 7);                   \
+                a[qi] = *reinterpret_cast<const vec8*>(                      \
+                    xh + m * 128 + sx * 16);                                 \
+            }                                                                \
+            _Pragma("unroll") for (int nj = 0; nj < NFRAG; ++nj) {           \
+                const int n = wn0 + nj * 16 + lrow;                          \
+                const int s16 = (kq * 2 + (lslot >> 1)) ^ (n & 7);           \
+                const unsigned* p2 = reinterpret_cast<const unsigned*>(      \
+                    wb + n * 128 + s16 * 16 + (lslot & 1) * 8);              \
+                T tmp[8];                                                    \
+                sk5_fp8x8<T>(p2, tmp);                                       \
+                bb[nj] = *reinterpret_cast<const vec8*>(tmp);                \
+            }                                                                \
+            _Pragma("unroll") for (int qi = 0; qi < 2; ++qi)                 \
+                _Pragma("unroll") for (int nj = 0; nj < NFRAG; ++nj)         \
+                    acc[qi][nj] = MF::mfma(a[qi], bb[nj], acc[qi][nj]);      \
+        }                                                                    \
+    } while (0)
+
+#define F8S_BODY(t, b, b2)                                                   \
+    do {                                                                     \
+        if ((t) + 1 < nchunks) vm_wait<G>();                                 \
+        else vm_wait<0>();                                                   \
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                   \
+        __builtin_amdgcn_s_barrier();                                        \
+        F8S_FRAGS(b);                                                        \
+        if ((t) + 2 < nchunks) F8S_STAGE((t) + 2, b2);                       \
+    } while (0)
+
+    F8S_STAGE(0, 0);
+    if (nchunks > 1) F8S_STAGE(1, 1);
+    int t = 0;
+    while (t + 3 <= nchunks) {
+        F8S_BODY(t + 0, 0, 2);
+        F8S_BODY(t + 1, 1, 0);
+        F8S_BODY(t + 2, 2, 1);
+        t += 3;
+    }
+    if (t < nchunks) F8S_BODY(t, 0, 2);
+    if (t + 1 < nchunks) F8S_BODY(t + 1, 1, 0);
+#undef F8S_BODY
+#undef F8S_FRAGS
+#undef F8S_STAGE
+    vm_wait<0>();
+
+    // epilogue (channel scale folds here; splitk partials already scaled
+    // so sk5_reduce sums them unchanged)
+    if (splitk > 1) {
+        const int mt_tot = gridDim.y * MT;
+        float* p = part + (long)slice * mt_tot * N;
+#pragma unroll
+        for (int qi = 0; qi < 2; ++qi)
+#pragma unroll
+            for (int nj = 0; nj < NFRAG; ++nj) {
+                const int n = n0 + wn0 + nj * 16 + lrow;
+                const float wsc = wscale[n];
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const int m = m0blk + wm0 + qi * 16 + lslot * 4 + r;
+                    p[(long)m * N + n] = acc[qi][nj][r] * wsc;
+                }
+            }
+    } else {
+        float bv[NFRAG], wsc[NFRAG];
+#pragma unroll
+        for (int nj = 0; nj < NFRAG; ++nj) {
+            const int n = n0 + wn0 + nj * 16 + lrow;
+            wsc[nj] = wscale[n];
+            bv[nj] = bias ? VecTraits<T>::to_float(bias[n]) : 0.f;
+        }
+#pragma unroll
+        for (int qi = 0; qi < 2; ++qi)
+#pragma unroll
+            for (int nj = 0; nj < NFRAG; ++nj) {
+                const int n = n0 + wn0 + nj * 16 + lrow;
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const int m = m0blk + wm0 + qi * 16 + lslot * 4 + r;
+                    if (m < M) {
+                        y[(long)m * N + n] = VecTraits<T>::from_float(
+                            acc[qi][nj][r] * wsc[nj] + bv[nj]);
+                    }
+                }
+            }
+    }
+}
+
 // fold the splitk fp32 partials: y[m,n] = rs?[m] * sum_s part[s,m,n] + bias?[n]
 template <typename T>
 __global__ void sk5_reduce_kernel(T* __restrict__ y,
@@ -422,7 +628,70 @@ void launch_sk5(torch::Tensor& y, const torch::Tensor& x, const torch::Tensor& w
     }
 }
 
+template <typename T>
+void launch_sk5f8(torch::Tensor& y, const torch::Tensor& x,
+                  const torch::Tensor& wq, const torch::Tensor& wscale,
+                  const c10::optional<torch::Tensor>& bias,
+                  const c10::optional<torch::Tensor>& part, int splitk) {
+    const int M = x.size(0), K = x.size(1), N = wq.size(0);
+    constexpr int MT = 128;
+    const int mblocks = (M + MT - 1) / MT;
+    constexpr int LDS = NBUF * (2 * MT * 128 + BN * 128);
+    static bool attr_set = false;
+    if (!attr_set) {
+        (void)hipFuncSetAttribute((const void*)sk5_fp8_kernel<T>,
+                                  hipFuncAttributeMaxDynamicSharedMemorySize,
+                                  LDS);
+        attr_set = true;
+    }
+    auto stream = at::hip::getCurrentHIPStream();
+    const T* bias_p = bias.has_value() ? (const T*)bias->data_ptr() : nullptr;
+    float* part_p = part.has_value() ? part->data_ptr<float>() : nullptr;
+    TORCH_CHECK(splitk == 1 || part_p != nullptr,
+                "splitk > 1 needs a (splitk, mblocks*128, N) fp32 workspace");
+    dim3 grid((N / BN) * splitk, mblocks);
+    hipLaunchKernelGGL((sk5_fp8_kernel<T>), grid, dim3(512), LDS, stream,
+                       (T*)y.data_ptr(), part_p, (const T*)x.data_ptr(),
+                       (const unsigned char*)wq.data_ptr(),
+                       wscale.data_ptr<float>(),
+                       splitk == 1 ? bias_p : nullptr, M, N, K, splitk);
+    HIP_CHECK_KERNEL();
+    if (splitk > 1) {
+        const long total = ((long)M * N + 3) / 4;
+        const int threads = 256;
+        const int blocks =
+            (int)std::min<long>((total + threads - 1) / threads, 2048);
+        hipLaunchKernelGGL((sk5_reduce_kernel<T>), dim3(blocks), dim3(threads),
+                           0, stream, (T*)y.data_ptr(), part_p, nullptr,
+                           bias_p, M, N, mblocks * MT, splitk);
+        HIP_CHECK_KERNEL();
+    }
+}
+
 }  // namespace
+
+void skinny_gemm5_fp8(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
+                      torch::Tensor wscale, c10::optional<torch::Tensor> bias,
+                      c10::optional<torch::Tensor> part, long splitk) {
+    TORCH_CHECK(x.dim() == 2 && wq.dim() == 2 && y.dim() == 2, "2-D tensors");
+    TORCH_CHECK(x.is_contiguous() && wq.is_contiguous() && y.is_contiguous() &&
+                wscale.is_contiguous(), "contiguous tensors required");
+    const int M = x.size(0), K = x.size(1), N = wq.size(0);
+    TORCH_CHECK(wq.size(1) == K && y.size(0) == M && y.size(1) == N &&
+                wscale.numel() == N, "shape");
+    TORCH_CHECK(wq.scalar_type() == at::ScalarType::Byte &&
+                wscale.scalar_type() == at::ScalarType::Float, "wq u8 + f32");
+    TORCH_CHECK(N % BN == 0, "N must be a multiple of 64");
+    TORCH_CHECK(K % (128 * splitk) == 0, "K must be a multiple of 128*splitk");
+    TORCH_CHECK(M >= 8 && M <= 256, "decode batches only (8 <= M <= 256)");
+    if (x.scalar_type() == at::ScalarType::BFloat16) {
+        launch_sk5f8<__hip_bfloat16>(y, x, wq, wscale, bias, part, (int)splitk);
+    } else if (x.scalar_type() == at::ScalarType::Half) {
+        launch_sk5f8<__half>(y, x, wq, wscale, bias, part, (int)splitk);
+    } else {
+        TORCH_CHECK(false, "skinny_gemm5_fp8: bf16/f16 x only");
+    }
+}
 
 void skinny_gemm5(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                   c10::optional<torch::Tensor> bias,

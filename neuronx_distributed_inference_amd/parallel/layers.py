@@ -18,6 +18,8 @@ from __future__ import annotations
 
 from typing import Optional
 
+import os
+
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
@@ -75,13 +77,36 @@ def fp8_linear(x: torch.Tensor, w_fp8: torch.Tensor, w_scale: torch.Tensor, bias
     """y = x @ dequant(w)^T (+bias). GPU: torch._scaled_mm with dynamic
     per-tensor activation scale; CPU: dequantized matmul."""
     if x.is_cuda:
+        import neuronx_distributed_inference_amd.ops as _ops
+
+        # decode-size batches: hand-written fp8-weight skinny GEMM
+        # (double-depth all-glds pipeline, bf16 x — no activation quant
+        # kernels, half the weight stream; see sk5_fp8_kernel). The
+        # hipBLASLt fp8 scaled-mm keeps prefill-size M.
+        M = x.numel() // x.shape[-1]
+        N, K = w_fp8.shape
+        if (8 <= M <= 256 and N % 64 == 0 and K % 128 == 0
+                and x.dtype in (torch.bfloat16, torch.float16)
+                and os.environ.get("NXDI_DISABLE_SK5F8") != "1"):
+            # splitk: fill the chip (~2 blocks/CU at 2 row-blocks)
+            mb = (M + 127) // 128
+            sk = 1
+            for cand in (8, 4, 2):
+                if K % (128 * cand) == 0 and                         (N // 64) * cand * mb <= 1024:
+                    sk = cand
+                    break
+            if (N // 64) * sk * mb < 128:
+                sk = 1
+            y = _ops.skinny_gemm5_fp8(
+                x.reshape(-1, K).contiguous(),
+                w_fp8.view(torch.uint8), w_scale, bias=bias, splitk=sk)
+            return y.view(*x.shape[:-1], N)
+
         # hipBLASLt scaled-mm takes per-tensor scales on this stack; the
         # per-row weight scale rides in the HIP epilogue kernel. Dynamic
         # activation quant runs as two HIP kernels (amax atomic + cast) —
         # no fp32 materialization of x, no host sync (K6-analog,
         # ops/csrc/fp8_quant.hip).
-        import neuronx_distributed_inference_amd.ops as _ops
-
         c = _ops._require_ext("fp8_quant")
         shape = x.shape[:-1] + (w_fp8.shape[0],)
         x2 = x.reshape(-1, x.shape[-1]).contiguous()

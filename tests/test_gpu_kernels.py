@@ -1193,3 +1193,41 @@ def test_moe_grouped_gemm_bk128(bnbk):
         m = se == e
         want[m] = x_sorted[dest[m]] @ w[e].t()
     torch.testing.assert_close(got.float(), want.float(), atol=2e-2, rtol=2e-2)
+
+
+@pytest.mark.parametrize("msk", [(64, 1), (128, 2), (256, 4), (100, 1)])
+def test_skinny_gemm5_fp8_gpu(msk):
+    """fp8-weight skinny decode GEMM vs the dequantized torch reference
+    (BK=128 double-depth pipeline, per-channel scale, splitk fold)."""
+    M, sk = msk
+    torch.manual_seed(9)
+    dev = _dev()
+    N, K = 512, 1024
+    x = torch.randn(M, K, dtype=torch.bfloat16, device=dev) * 0.2
+    w = torch.randn(N, K) * 0.1
+    scale = w.abs().amax(dim=1).clamp(min=1e-8) / 448.0
+    wq = (w / scale.unsqueeze(1)).clamp(-448, 448).to(torch.float8_e4m3fn)
+    bias = torch.randn(N, dtype=torch.bfloat16, device=dev) * 0.1
+    wq_d = wq.view(torch.uint8).to(dev)
+    sc_d = scale.float().to(dev)
+    got = ops.skinny_gemm5_fp8(x, wq_d, sc_d, bias=bias, splitk=sk)
+    wd = (wq.float() * scale.unsqueeze(1)).to(dev)
+    want = x.float() @ wd.t().float() + bias.float()
+    torch.testing.assert_close(got.float(), want, atol=5e-2, rtol=5e-2)
+
+
+def test_fp8_linear_decode_path_gpu():
+    """fp8_linear at decode M routes through the skinny fp8 kernel and
+    matches the dequantized reference."""
+    from neuronx_distributed_inference_amd.parallel.layers import fp8_linear
+
+    torch.manual_seed(10)
+    dev = _dev()
+    M, N, K = 256, 576, 1152  # K % 128 = 0, N % 64 = 0
+    x = torch.randn(M, K, dtype=torch.bfloat16, device=dev) * 0.2
+    w = torch.randn(N, K) * 0.1
+    scale = w.abs().amax(dim=1, keepdim=True).clamp(min=1e-8) / 448.0
+    wq = (w / scale).clamp(-448, 448).to(torch.float8_e4m3fn).to(dev)
+    got = fp8_linear(x, wq, scale.float().to(dev), None)
+    want = x.float() @ (wq.float().cpu() * scale).t().to(dev).float()
+    torch.testing.assert_close(got.float(), want, atol=6e-2, rtol=6e-2)
