@@ -454,6 +454,209 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_mx4_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// MXFP4 v3 ("mx4b"): all-glds packed-weight grouped GEMM for shapes with
+// K % 256 == 0 (Mixtral/Llama-style MoE dims; GPT-OSS K=2880 keeps v1).
+// A 256-deep K chunk packs to EXACTLY the [rows][128B] XOR-swizzled glds
+// image of the bf16 kernel (256 e2m1 nibbles = 128 B), so the packed
+// weights stream through the verified pipeline at full step depth —
+// 3.8x less weight traffic. x stages as FOUR 128B half-images.
+// The e8m0 block scales stage as a glds side-image too (16 B per row =
+// the scales of chunks t and t+1; ~4% redundant, but it keeps EVERY
+// load in the counted-glds pipeline — per-fragment global scale reads
+// would re-trigger the mixed-load vmcnt drain that sank mx4 v1).
+// Scale pieces are staged by waves 0..NS-1 only; s_waitcnt is wave-
+// uniform, so those waves just use a count one higher.
+// Dequant (nibble magnitude table * e8m0) happens at the LDS read.
+// LDS at bn=128: 3 x (32K x + 16K W + 2K s) = 150 KB.
+
+template <typename T, int BNT>
+__global__ __launch_bounds__(512, 2) void moe_grouped_mx4b_kernel(
+    T* __restrict__ y,                   // (Mpad, N)
+    const T* __restrict__ x,             // (Mpad, K) — or (T, K) w/ x_gather
+    const unsigned char* __restrict__ wq,  // (E, N, K/2) packed e2m1
+    const unsigned char* __restrict__ ws,  // (E, N, K/32) e8m0 scales
+    const int* __restrict__ expert_ids,  // (n_row_tiles,)
+    const T* __restrict__ bias,          // (E, N) or null
+    const int* __restrict__ x_gather,    // (Mpad,) or null
+    float* __restrict__ combine_out,     // (T, N) fp32 or null
+    const float* __restrict__ combine_scale,  // (Mpad,)
+    const int* __restrict__ combine_gather,   // (Mpad,)
+    const int N, const int K) {
+    using MF = MgMfma<T>;
+    using vec8 = typename MF::vec8;
+    constexpr int MT = 64;
+    constexpr int BKB = 256;        // k-elems per ring step
+    constexpr int NH = 4;           // x 128B half-images per step
+    constexpr int NWG = BNT / 64;   // W glds per wave per step
+    constexpr int NS = BNT / 64;    // scale pieces per step (64 rows each)
+    constexpr int G = NH + NWG;     // glds per ORDINARY wave per step
+    constexpr int NFRAG = BNT / 64;
+    constexpr int XBYTES = NH * MT * 128;
+    constexpr int WBYTES = BNT * 128;
+    constexpr int CHUNK_B = XBYTES + WBYTES + BNT * 16;
+    static_assert(BNT == 64 || BNT == 128, "mx4b: bn 64/128 only");
+
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+
+    const int tile = blockIdx.x;
+    const int rt = blockIdx.y;
+    const int e = expert_ids[rt];
+    if (e < 0) return;
+    const int n0 = tile * BNT;
+    const int m0 = rt * MT;
+    const int nchunks = K / BKB;
+    const int srow_bytes = K / 32;  // scale bytes per weight row
+
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int wm0 = (wid >> 2) * 32;
+    const int wn0 = (wid & 3) * (BNT / 4);
+    const int lrow = lane & 15;
+    const int lslot = lane >> 4;
+
+    const int xr_lin = m0 + wid * 8 + (lane >> 3);
+    const int xr = x_gather ? max(x_gather[xr_lin], 0) : xr_lin;
+    const int xslot = (lane & 7) ^ (lane >> 3);
+    const char* xsrc_lane = reinterpret_cast<const char*>(x) +
+                            (long)xr * K * sizeof(T) + xslot * 16;
+    // packed W rows: K/2 bytes; 128 B per row per chunk -> standard
+    // 8-rows-per-1KB pieces with the XOR source swizzle
+    const char* wsrc[NWG];
+#pragma unroll
+    for (int i = 0; i < NWG; ++i)
+        wsrc[i] = reinterpret_cast<const char*>(wq) +
+                  ((long)e * N + n0 + wid * (8 * NWG) + i * 8) * (K / 2);
+    // scale piece (waves 0..NS-1): lane covers ONE row's 16 B
+    const bool sc_wave = wid < NS;
+    const char* ssrc_row = reinterpret_cast<const char*>(ws) +
+                           ((long)e * N + n0 + wid * 64 + lane) * srow_bytes;
+
+    char* xdst[MG_NBUF];
+    char* wdst[MG_NBUF];
+    char* sdst[MG_NBUF];
+#pragma unroll
+    for (int b = 0; b < MG_NBUF; ++b) {
+        xdst[b] = smem + b * CHUNK_B + wid * 1024;
+        wdst[b] = smem + b * CHUNK_B + XBYTES + wid * (NWG * 1024);
+        sdst[b] = smem + b * CHUNK_B + XBYTES + WBYTES + wid * 1024;
+    }
+
+#define B3_STAGE(t, b)                                                       \
+    do {                                                                     \
+        const long xoff = (long)(t) * 512;                                   \
+        _Pragma("unroll") for (int h = 0; h < NH; ++h)                       \
+            mg_glds_lane128(xsrc_lane + xoff + h * 128,                      \
+                            xdst[b] + h * (MT * 128));                       \
+        _Pragma("unroll") for (int i = 0; i < NWG; ++i)                      \
+            mg_glds_rows128<2>(wsrc[i] + (long)(t) * 128, (long)(K / 2),     \
+                               wdst[b] + i * 1024);                          \
+        if (sc_wave)                                                         \
+            mg_glds_lane128(ssrc_row + min((t) * 8, srow_bytes - 16),        \
+                            sdst[b]);                                        \
+    } while (0)
+
+    mg_f32x4 acc[2][NFRAG];
+#pragma unroll
+    for (int qi = 0; qi < 2; ++qi)
+#pragma unroll
+        for (int nj = 0; nj < NFRAG; ++nj) acc[qi][nj] = {0.f, 0.f, 0.f, 0.f};
+
+#define B3_FRAGS(t, b)                                                       \
+    do {                                                                     \
+        const char* wb = smem + (b) * CHUNK_B + XBYTES;                      \
+        const char* sb = smem + (b) * CHUNK_B + XBYTES + WBYTES;             \
+        const int sdelta = (t) * 8 - min((t) * 8, srow_bytes - 16);          \
+        _Pragma("unroll") for (int kq = 0; kq < 8; ++kq) {                   \
+            const char* xh =                                                 \
+                smem + (b) * CHUNK_B + (kq >> 1) * (MT * 128);               \
+            const int ks = kq & 1;                                           \
+            vec8 a0, a1, bb[NFRAG];                                          \
+            {                                                                \
+                const int m = wm0 + lrow;                                    \
+                const int sx = (ks * 4 + lslot) ^ (m & 7);                   \
+                a0 = *reinterpret_cast<const vec8*>(xh + m * 128 + sx * 16); \
+            }                                                                \
+            {                                                                \
+                const int m = wm0 + 16 + lrow;                               \
+                const int sx = (ks * 4 + lslot) ^ (m & 7);                   \
+                a1 = *reinterpret_cast<const vec8*>(xh + m * 128 + sx * 16); \
+            }                                                                \
+            _Pragma("unroll") for (int nj = 0; nj < NFRAG; ++nj) {           \
+                const int n = wn0 + nj * 16 + lrow;                          \
+                const int s16 = kq ^ (n & 7);                                \
+                const unsigned pk = *reinterpret_cast<const unsigned*>(      \
+                    wb + n * 128 + s16 * 16 + lslot * 4);                    \
+                const unsigned char se = *reinterpret_cast<                  \
+                    const unsigned char*>(sb + n * 16 + sdelta + kq);        \
+                const float sc = exp2f((int)se - 127) * 0.5f;                \
+                T tmp[8];                                                    \
+                mx4_dequant8<T>(pk, sc, tmp);                                \
+                bb[nj] = *reinterpret_cast<const vec8*>(tmp);                \
+            }                                                                \
+            _Pragma("unroll") for (int nj = 0; nj < NFRAG; ++nj) {           \
+                acc[0][nj] = MF::mfma(a0, bb[nj], acc[0][nj]);               \
+                acc[1][nj] = MF::mfma(a1, bb[nj], acc[1][nj]);               \
+            }                                                                \
+        }                                                                    \
+    } while (0)
+
+    // s_waitcnt is wave-uniform: scale-staging waves carry one extra
+    // in-flight glds per step
+#define B3_BODY(t, b, b2)                                                    \
+    do {                                                                     \
+        if ((t) + 1 < nchunks) {                                             \
+            if (sc_wave) mg_vm_wait<G + 1>();                                \
+            else mg_vm_wait<G>();                                            \
+        } else mg_vm_wait<0>();                                              \
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                   \
+        __builtin_amdgcn_s_barrier();                                        \
+        B3_FRAGS(t, b);                                                      \
+        if ((t) + 2 < nchunks) B3_STAGE((t) + 2, b2);                        \
+    } while (0)
+
+    B3_STAGE(0, 0);
+    if (nchunks > 1) B3_STAGE(1, 1);
+    int t = 0;
+    while (t + 3 <= nchunks) {
+        B3_BODY(t + 0, 0, 2);
+        B3_BODY(t + 1, 1, 0);
+        B3_BODY(t + 2, 2, 1);
+        t += 3;
+    }
+    if (t < nchunks) B3_BODY(t, 0, 2);
+    if (t + 1 < nchunks) B3_BODY(t + 1, 1, 0);
+#undef B3_BODY
+#undef B3_FRAGS
+#undef B3_STAGE
+    mg_vm_wait<0>();
+
+#pragma unroll
+    for (int nj = 0; nj < NFRAG; ++nj) {
+        const int n_out = n0 + wn0 + nj * 16 + lrow;
+        const float bv =
+            bias ? VecTraits<T>::to_float(bias[(long)e * N + n_out]) : 0.f;
+#pragma unroll
+        for (int qi = 0; qi < 2; ++qi) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int m = m0 + wm0 + qi * 16 + lslot * 4 + r;
+                const float v = acc[qi][nj][r] + bv;
+                if (combine_out) {
+                    const int tk = combine_gather[m];
+                    if (tk >= 0) {
+                        const float sc =
+                            combine_scale ? combine_scale[m] : 1.f;
+                        atomicAdd(&combine_out[(long)tk * N + n_out], v * sc);
+                    }
+                } else {
+                    y[(long)m * N + n_out] = VecTraits<T>::from_float(v);
+                }
+            }
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
 // FP8 (OCP e4m3fn) grouped expert GEMM: weights stored 1 byte/elem with
 // ONE fp32 scale per output channel (E, N). Unlike the MXFP4 variant
 // (register-staged packed W -> mixed-load vmcnt drains), this keeps the
@@ -969,6 +1172,55 @@ void moe_grouped_gemm_mx4(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
     TORCH_CHECK(K % MG_BK == 0 && Mpad % 64 == 0, "align");
     TORCH_CHECK(wq.scalar_type() == at::ScalarType::Byte &&
                 ws.scalar_type() == at::ScalarType::Byte, "u8 packed");
+    // v3 fast path: all-glds packed pipeline with 256-deep chunks
+    // (Mixtral-style dims). Falls through to v1 (register-staged) for
+    // K % 256 != 0 (e.g. GPT-OSS K=2880).
+    if (K % 256 == 0 && bn == 0 &&
+        getenv("NXDI_MX4_V1") == nullptr) {
+        const int BNT3 = (N % 128 == 0) ? 128 : 64;
+        if (N % BNT3 == 0) {
+            auto stream3 = at::hip::getCurrentHIPStream();
+            dim3 grid3(N / BNT3, Mpad / 64);
+#define MX4B_LAUNCH(T, BNTV)                                                  \
+    do {                                                                      \
+        constexpr int LDS =                                                   \
+            MG_NBUF * (4 * 64 * 128 + BNTV * 128 + BNTV * 16);                \
+        static bool attr_set = false;                                         \
+        if (!attr_set) {                                                      \
+            (void)hipFuncSetAttribute(                                        \
+                (const void*)moe_grouped_mx4b_kernel<T, BNTV>,                \
+                hipFuncAttributeMaxDynamicSharedMemorySize, LDS);             \
+            attr_set = true;                                                  \
+        }                                                                     \
+        hipLaunchKernelGGL((moe_grouped_mx4b_kernel<T, BNTV>), grid3,         \
+                           dim3(512), LDS, stream3, (T*)y.data_ptr(),         \
+                           (const T*)x.data_ptr(),                            \
+                           (const unsigned char*)wq.data_ptr(),               \
+                           (const unsigned char*)ws.data_ptr(),               \
+                           expert_ids.data_ptr<int>(),                        \
+                           bias.has_value() ? (const T*)bias->data_ptr() : nullptr, \
+                           x_gather ? x_gather->data_ptr<int>() : nullptr,    \
+                           combine_out ? combine_out->data_ptr<float>()       \
+                                       : nullptr,                             \
+                           combine_scale ? combine_scale->data_ptr<float>()   \
+                                         : nullptr,                           \
+                           combine_gather ? combine_gather->data_ptr<int>()   \
+                                          : nullptr,                          \
+                           N, K);                                             \
+    } while (0)
+            if (x.scalar_type() == at::ScalarType::BFloat16) {
+                if (BNT3 == 128) MX4B_LAUNCH(__hip_bfloat16, 128);
+                else MX4B_LAUNCH(__hip_bfloat16, 64);
+            } else {
+                if (BNT3 == 128) MX4B_LAUNCH(__half, 128);
+                else MX4B_LAUNCH(__half, 64);
+            }
+#undef MX4B_LAUNCH
+            HIP_CHECK_KERNEL();
+            return;
+        }
+    }
+
     int BNT = (int)bn;
     // auto = 64 for mx4: the register-staged packed-W path pays hipcc's
     // mixed-load vmcnt drains once per W row-group, so wide tiles LOSE
