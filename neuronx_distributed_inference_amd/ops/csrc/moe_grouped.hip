@@ -1198,7 +1198,7 @@ void moe_grouped_gemm_mx4(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
     // (Mixtral-style dims). Falls through to v1 (register-staged) for
     // K % 256 != 0 (e.g. GPT-OSS K=2880).
     if (K % 256 == 0 && bn == 0 &&
-        getenv("NXDI_MX4_V3") != nullptr) {
+        getenv("NXDI_MX4_V1") == nullptr) {
         const int BNT3 = (N % 128 == 0) ? 128 : 64;
         if (N % BNT3 == 0) {
             auto stream3 = at::hip::getCurrentHIPStream();

@@ -7,7 +7,6 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch  # noqa: E402
 
-os.environ.setdefault("NXDI_MX4_V3", "1")
 
 from neuronx_distributed_inference_amd import ops  # noqa: E402
 

@@ -57,3 +57,38 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+def perf():
+    dev = torch.device("cuda:0")
+    torch.manual_seed(0)
+    E, H, I, T, topk = 8, 4096, 14336, 256, 2
+    idx = torch.randint(0, E, (T * topk,), device=dev)
+    order, dest, eids, gtab, Mpad = ops.moe_grouped_tables(idx, E, topk)
+    x = torch.randn(T, H, dtype=torch.bfloat16, device=dev) * 0.3
+    w = torch.randn(E, I, H) * 0.05
+    q4, s4 = ops.mx4_quantize(w)
+    q4, s4 = q4.to(dev), s4.to(dev)
+
+    def t(fn, n=30):
+        for _ in range(5):
+            fn()
+        torch.cuda.synchronize()
+        a, b = torch.cuda.Event(True), torch.cuda.Event(True)
+        a.record()
+        for _ in range(n):
+            fn()
+        b.record()
+        torch.cuda.synchronize()
+        return a.elapsed_time(b) / n * 1000
+
+    os.environ.pop("NXDI_MX4_V3", None)
+    t1 = t(lambda: ops.moe_grouped_gemm_mx4(x, q4, s4, eids, x_gather=gtab))
+    os.environ["NXDI_MX4_V3"] = "1"
+    t3 = t(lambda: ops.moe_grouped_gemm_mx4(x, q4, s4, eids, x_gather=gtab))
+    del os.environ["NXDI_MX4_V3"]
+    print(f"mixtral gate mx4: v1 {t1:7.1f} us | v3-hw {t3:7.1f} us")
+
+
+if __name__ == "__main__":
+    perf()
