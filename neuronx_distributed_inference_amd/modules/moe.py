@@ -145,6 +145,24 @@ class ExpertMLPs(nn.Module):
         self.fp8 = True
         return True
 
+    def quantize_mx4(self) -> bool:
+        """Convert the stacked expert weights to OCP MXFP4 (e2m1 nibbles
+        + e8m0 per-32 block scales, 4.25 bits/weight — a 3.8x memory cut
+        AND faster than bf16 on K%256 shapes via the all-glds v3 kernel:
+        160 vs 217 us at the Mixtral gate GEMM)."""
+        I, H = self.w_gate.shape[1], self.w_down.shape[1]
+        K1, K2 = self.w_gate.shape[2], self.w_down.shape[2]
+        if I % 64 or H % 64 or K1 % 256 or K2 % 256:
+            return False
+        for name in ("w_gate", "w_up", "w_down"):
+            w = getattr(self, name)
+            q, sc = ops.mx4_quantize(w.data.float())
+            del self._parameters[name]
+            self.register_buffer(name + "_q", q.to(w.device))
+            self.register_buffer(name + "_s", sc.to(w.device))
+        self.mx4 = True
+        return True
+
     def _forward_dense(self, x: torch.Tensor, weights: torch.Tensor,
                        idx: torch.Tensor) -> torch.Tensor:
         T, H = x.shape
@@ -200,7 +218,8 @@ class ExpertMLPs(nn.Module):
             return 64
 
         fp8 = getattr(self, "fp8", False)
-        if fp8:
+        mx4 = getattr(self, "mx4", False)
+        if fp8 or mx4:
             I, Hn = self.w_gate_q.shape[1], self.w_down_q.shape[1]
         else:
             I, Hn = self.w_gate.shape[1], self.w_down.shape[1]
@@ -215,7 +234,12 @@ class ExpertMLPs(nn.Module):
         else:
             x_in = x                      # token gather fuses into staging
             xg = gtab
-        if fp8:
+        if mx4:
+            g = ops.moe_grouped_gemm_mx4(x_in, self.w_gate_q, self.w_gate_s,
+                                         expert_ids, x_gather=xg)
+            u = ops.moe_grouped_gemm_mx4(x_in, self.w_up_q, self.w_up_s,
+                                         expert_ids, x_gather=xg)
+        elif fp8:
             # fp8 column tiles are 128/256 only
             def _bn8(N, ntiles=Mpad // 64):
                 if N % 256 == 0 and (N // 256) * ntiles >= 256:
@@ -240,7 +264,12 @@ class ExpertMLPs(nn.Module):
             else torch.ones_like(gatew, dtype=torch.float32)
         gw_tab[dest] = torch.where(lmask, scale, torch.zeros_like(scale))
         out32 = torch.zeros(T, H, dtype=torch.float32, device=x.device)
-        if fp8:
+        if mx4:
+            ops.moe_grouped_gemm_mx4(h, self.w_down_q, self.w_down_s,
+                                     expert_ids, combine_out=out32,
+                                     combine_scale=gw_tab,
+                                     combine_gather=gtab)
+        elif fp8:
             ops.moe_grouped_gemm_fp8(h, self.w_down_q, self.w_down_s,
                                      expert_ids, bn=_bn8(Hn),
                                      combine_out=out32, combine_scale=gw_tab,
@@ -264,7 +293,7 @@ class ExpertMLPs(nn.Module):
         # redundant FLOPs.
         import os as _os
 
-        if getattr(self, "fp8", False):
+        if getattr(self, "fp8", False) or getattr(self, "mx4", False):
             return self._forward_grouped(x, weights, idx)
         grouped_env = _os.environ.get("NXDI_MOE_GROUPED")
         # small batches leave too few 64-row tiles to fill 256 CUs at the

@@ -652,3 +652,26 @@ def test_moe_fp8_grouped_cpu_matches_bf16():
     # the K-dim fan-in
     err = (got - want).abs().max() / want.abs().max().clamp(min=1e-6)
     assert err < 0.06, float(err)
+
+
+def test_moe_mx4_grouped_cpu_matches_bf16():
+    """ExpertMLPs MXFP4 mode vs the dense path on CPU (4-bit noise is
+    ~12% relative on random weights — tolerance reflects that)."""
+    import torch
+
+    from neuronx_distributed_inference_amd.modules.moe import ExpertMLPs
+
+    torch.manual_seed(11)
+    E, H, I, T, top_k = 4, 256, 512, 16, 2
+    m = ExpertMLPs(num_experts=E, hidden_size=H, intermediate_size=I,
+                   tp_degree=1, dtype=torch.float32)
+    with torch.no_grad():
+        for p in m.parameters():
+            p.normal_(0, 0.05)
+    x = torch.randn(T, H) * 0.3
+    w, idx = torch.softmax(torch.randn(T, E), -1).topk(top_k, dim=-1)
+    want = m._forward_dense(x, w, idx)
+    assert m.quantize_mx4()
+    got = m._forward_grouped(x, w, idx)
+    err = (got - want).abs().max() / want.abs().max().clamp(min=1e-6)
+    assert err < 0.35, float(err)  # 4-bit weights: coarse but bounded
