@@ -56,6 +56,8 @@ def parse_args():
     p.add_argument("--no-graphs", action="store_true")
     p.add_argument("--e2e", action="store_true",
                    help="also run the generate()-level benchmark (p50/p99 + throughput)")
+    p.add_argument("--moe-mx4", action="store_true",
+                   help="MXFP4 expert weights (4.25 bits) via the grouped v3 kernel")
     p.add_argument("--fp8", action="store_true",
                    help="fp8 (e4m3) weight quantization via hipBLASLt scaled-mm")
     p.add_argument("--spec", action="store_true",
@@ -250,6 +252,13 @@ def main():
             if p.dim() >= 2:
                 p.normal_(0, 0.02)
     model.reset()
+    if args.moe_mx4:
+        from neuronx_distributed_inference_amd.modules.moe import ExpertMLPs
+
+        nm = sum(1 for m in model.model.modules()
+                 if isinstance(m, ExpertMLPs) and m.quantize_mx4())
+        if rank == 0:
+            print(f"# quantized {nm} expert stacks to MXFP4", flush=True)
     if args.fp8:
         from neuronx_distributed_inference_amd.parallel.layers import quantize_model_fp8
 
@@ -319,7 +328,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": ("fp8-weights+bf16" if args.fp8 else dtype),
+            "dtype": ("fp8-weights+bf16" if args.fp8 else ("mxfp4-experts+bf16" if args.moe_mx4 else dtype)),
             "data": "synthetic",
             "config": {
                 "model": {"8b": "llama-3.1-8b", "70b": "llama-3.1-70b",

@@ -1231,3 +1231,27 @@ def test_fp8_linear_decode_path_gpu():
     got = fp8_linear(x, wq, scale.float().to(dev), None)
     want = x.float() @ (wq.float().cpu() * scale).t().to(dev).float()
     torch.testing.assert_close(got.float(), want, atol=6e-2, rtol=6e-2)
+
+
+def test_moe_mx4_module_gpu():
+    """ExpertMLPs MXFP4 mode end-to-end on GPU (v3 kernel on K%256
+    shapes) vs the bf16 dense path."""
+    from neuronx_distributed_inference_amd.modules.moe import ExpertMLPs
+
+    torch.manual_seed(12)
+    dev = _dev()
+    E, H, I, T, top_k = 8, 512, 1024, 64, 2
+    m = ExpertMLPs(num_experts=E, hidden_size=H, intermediate_size=I,
+                   tp_degree=1, dtype=torch.bfloat16)
+    with torch.no_grad():
+        for p in m.parameters():
+            p.normal_(0, 0.05)
+    m = m.to(dev)
+    x = torch.randn(T, H, dtype=torch.bfloat16, device=dev) * 0.3
+    w, idx = torch.softmax(torch.randn(T, E, device=dev), -1).topk(top_k, -1)
+    want = m._forward_dense(x, w, idx)
+    assert m.quantize_mx4()
+    got = m(x, w, idx)
+    rel = ((got.float() - want.float()).abs().max()
+           / want.float().abs().max().clamp(min=1e-3))
+    assert rel < 0.35, float(rel)
