@@ -156,7 +156,8 @@ class ExpertMLPs(nn.Module):
             return False
         for name in ("w_gate", "w_up", "w_down"):
             w = getattr(self, name)
-            q, sc = ops.mx4_quantize(w.data.float())
+            # the quantizer's e2m1 grid lives on CPU
+            q, sc = ops.mx4_quantize(w.data.float().cpu())
             del self._parameters[name]
             self.register_buffer(name + "_q", q.to(w.device))
             self.register_buffer(name + "_s", sc.to(w.device))
