@@ -156,8 +156,7 @@ class ExpertMLPs(nn.Module):
             return False
         for name in ("w_gate", "w_up", "w_down"):
             w = getattr(self, name)
-            # the quantizer's e2m1 grid lives on CPU
-            q, sc = ops.mx4_quantize(w.data.float().cpu())
+            q, sc = ops.mx4_quantize(w.data.float())
             del self._parameters[name]
             self.register_buffer(name + "_q", q.to(w.device))
             self.register_buffer(name + "_s", sc.to(w.device))
