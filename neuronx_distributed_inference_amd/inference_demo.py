@@ -133,6 +133,9 @@ def setup_run_parser(run: argparse.ArgumentParser):
                      help="ring (compact) KV cache for sliding-window layers")
     run.add_argument("--is-prefix-caching", action="store_true")
     run.add_argument("--is-chunked-prefill", action="store_true")
+    run.add_argument("--quantize-moe", choices=["fp8", "mx4"], default=None,
+                     help="runtime-quantize MoE expert stacks (fp8-e4m3 "
+                          "per-channel, or OCP MXFP4 at 4.25 bits)")
     run.add_argument("--enable-eagle-speculation", action="store_true")
     run.add_argument("--enable-token-tree", action="store_true",
                      help="EAGLE TREE speculation (needs --enable-eagle-speculation)")
@@ -334,6 +337,18 @@ def run_inference(args, model_type: str) -> int:
         return 0
 
     model = app.load(args.compiled_model_path, device=device)
+    if getattr(args, "quantize_moe", None):
+        from neuronx_distributed_inference_amd.modules.moe import ExpertMLPs
+
+        nq = 0
+        for mod in model.model.modules():
+            if isinstance(mod, ExpertMLPs):
+                ok = (mod.quantize_fp8() if args.quantize_moe == "fp8"
+                      else mod.quantize_mx4())
+                nq += int(ok)
+        if rank0:
+            logger.info("quantized %d expert stacks to %s", nq,
+                        args.quantize_moe)
     tokenizer = load_tokenizer(args.compiled_model_path) or load_tokenizer(args.model_path)
     input_ids, attention_mask = get_input_ids(args, tokenizer, device)
 

@@ -114,3 +114,31 @@ def test_eagle_forces_on_device_greedy_sampling():
     assert ec.on_device_sampling_config is not None
     assert ec.on_device_sampling_config.top_k == 1
     assert ec.enable_eagle_speculation
+
+
+def test_inference_demo_quantize_moe_flag(tmp_path):
+    """--quantize-moe runtime-quantizes expert stacks (tiny mixtral)."""
+    from transformers import MixtralConfig, MixtralForCausalLM
+
+    from neuronx_distributed_inference_amd.inference_demo import main
+
+    torch.manual_seed(0)
+    cfg = MixtralConfig(
+        vocab_size=128, hidden_size=64, intermediate_size=512,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        num_local_experts=4, num_experts_per_tok=2,
+        max_position_embeddings=256, tie_word_embeddings=False)
+    m = MixtralForCausalLM(cfg)
+    path = tmp_path / "mx"
+    m.save_pretrained(str(path), safe_serialization=True)
+    rc = main([
+        "--model-type", "mixtral", "run",
+        "--model-path", str(path),
+        "--compiled-model-path", str(tmp_path / "cmx"),
+        "--prompt-ids", json.dumps([[1, 2, 3, 4]]),
+        "--max-new-tokens", "4",
+        "--on-cpu", "--torch-dtype", "float32",
+        "--batch-size", "1", "--max-context-length", "16", "--seq-len", "32",
+        "--quantize-moe", "fp8",
+    ])
+    assert rc == 0
