@@ -44,6 +44,15 @@ class MoERouter(nn.Module):
 
     def forward(self, x: torch.Tensor):
         logits = F.linear(x.float(), self.weight.float())  # (T, E)
+        if (logits.is_cuda and ops.extension_available()
+                and self.top_k <= 8):
+            # fused top-k + activation + normalize: one launch vs the
+            # ~6-kernel torch chain (launch-bound at decode sizes)
+            sig = self.act == "sigmoid"
+            return ops.router_topk(
+                logits, self.top_k, sig,
+                (not sig) and self.pre_softmax,
+                (not sig) and self.normalize)
         if self.act == "sigmoid":
             vals, idx = logits.topk(self.top_k, dim=-1)
             return torch.sigmoid(vals), idx

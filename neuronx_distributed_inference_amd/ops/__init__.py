@@ -655,6 +655,20 @@ def attention_decode_paged(q, k_cache, v_cache, seq_lens, block_table,
 # segment. All table construction is tensor ops with STATIC shapes, so
 # the whole MoE layer is hipGraph-capturable.
 
+def router_topk(logits: torch.Tensor, top_k: int, act_sigmoid: bool,
+                pre_softmax: bool, normalize: bool):
+    """Fused MoE router selection: top-k + activation + normalization in
+    ONE launch (the torch chain costs ~6 kernels/layer). GPU only;
+    logits (T, E) fp32. Returns (vals (T, k) f32, idx (T, k) i64)."""
+    c = _require_ext("router_topk")
+    T = logits.shape[0]
+    vals = torch.empty(T, top_k, dtype=torch.float32, device=logits.device)
+    idx = torch.empty(T, top_k, dtype=torch.long, device=logits.device)
+    c.router_topk(vals, idx, logits.contiguous(), top_k,
+                  int(act_sigmoid), int(pre_softmax), int(normalize))
+    return vals, idx
+
+
 MOE_TILE = 64
 
 

@@ -75,6 +75,9 @@ void moe_grouped_gemm_mx4(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
                           c10::optional<torch::Tensor> combine_out,
                           c10::optional<torch::Tensor> combine_scale,
                           c10::optional<torch::Tensor> combine_gather);
+void router_topk(torch::Tensor vals, torch::Tensor idx, torch::Tensor logits,
+                 long top_k, long act_sigmoid, long pre_softmax,
+                 long normalize);
 bool lt_available();
 void lt_linear(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                c10::optional<torch::Tensor> bias);
@@ -164,6 +167,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("combine_out") = py::none(),
           py::arg("combine_scale") = py::none(),
           py::arg("combine_gather") = py::none());
+    m.def("router_topk", &router_topk,
+          "fused MoE router top-k + activation + normalize (CDNA4)",
+          py::arg("vals"), py::arg("idx"), py::arg("logits"),
+          py::arg("top_k"), py::arg("act_sigmoid"), py::arg("pre_softmax"),
+          py::arg("normalize"));
     m.def("lt_available", &lt_available, "hipblaslt usable on this device");
     m.def("lt_linear", &lt_linear, "autotuned hipBLASLt linear (graph-safe)",
           py::arg("y"), py::arg("x"), py::arg("w"), py::arg("bias") = py::none());

@@ -32,6 +32,7 @@ sources = [
     os.path.join(CSRC, "skinny_gemm4.hip"),
     os.path.join(CSRC, "skinny_gemm5.hip"),
     os.path.join(CSRC, "moe_grouped.hip"),
+    os.path.join(CSRC, "router_topk.hip"),
     os.path.join(CSRC, "lt_gemm.cpp"),
 ]
 

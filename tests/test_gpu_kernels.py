@@ -1255,3 +1255,25 @@ def test_moe_mx4_module_gpu():
     rel = ((got.float() - want.float()).abs().max()
            / want.float().abs().max().clamp(min=1e-3))
     assert rel < 0.35, float(rel)
+
+
+@pytest.mark.parametrize("mode", ["post", "pre", "sigmoid", "postnorm"])
+def test_router_topk_fused_gpu(mode):
+    """Fused router vs the torch chain for every activation variant."""
+    from neuronx_distributed_inference_amd.modules.moe import MoERouter
+
+    torch.manual_seed(13)
+    dev = _dev()
+    E, H, T, k = 16, 64, 200, 4
+    r = MoERouter(H, E, k,
+                  normalize=(mode == "postnorm"),
+                  pre_softmax=(mode == "pre"),
+                  act=("sigmoid" if mode == "sigmoid" else "softmax"))
+    with torch.no_grad():
+        r.weight.normal_(0, 0.5)
+    x = torch.randn(T, H)
+    want_v, want_i = r(x)                 # CPU torch chain
+    r = r.to(dev)
+    got_v, got_i = r(x.to(dev))
+    assert torch.equal(got_i.cpu(), want_i), mode
+    torch.testing.assert_close(got_v.cpu(), want_v, atol=1e-5, rtol=1e-5)
