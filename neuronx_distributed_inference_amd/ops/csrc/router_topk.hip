@@ -62,12 +62,14 @@ __global__ void router_topk_kernel(
         for (int j = 0; j < KMAX; ++j)
             if (j < k) bv[j] = 1.f / (1.f + __expf(-bv[j]));
     } else {
-        // softmax over the selected k (bv[0] is the max)
+        // softmax over the selected k (bv[0] is the max — snapshot it
+        // BEFORE the in-place exp overwrites it)
+        const float m0 = bv[0];
         float s = 0.f;
 #pragma unroll
         for (int j = 0; j < KMAX; ++j)
             if (j < k) {
-                bv[j] = __expf(bv[j] - bv[0]);
+                bv[j] = __expf(bv[j] - m0);
                 s += bv[j];
             }
 #pragma unroll
