@@ -45,7 +45,7 @@ class MoERouter(nn.Module):
     def forward(self, x: torch.Tensor):
         logits = F.linear(x.float(), self.weight.float())  # (T, E)
         if (logits.is_cuda and ops.extension_available()
-                and self.top_k <= 8):
+                and self.top_k <= 8 and self.num_experts >= self.top_k):
             # fused top-k + activation + normalize: one launch vs the
             # ~6-kernel torch chain (launch-bound at decode sizes)
             sig = self.act == "sigmoid"
