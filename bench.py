@@ -37,6 +37,11 @@ MODELS = {
                     num_hidden_layers=32, num_attention_heads=32,
                     num_key_value_heads=8, vocab_size=32000,
                     num_local_experts=8, num_experts_per_tok=2),
+    # Qwen3-30B-A3B geometry: 128 experts top-8 (fine-grained MoE)
+    "qwen3moe": dict(hidden_size=2048, intermediate_size=768,
+                     num_hidden_layers=48, num_attention_heads=32,
+                     num_key_value_heads=4, head_dim=128, vocab_size=151936,
+                     num_local_experts=128, num_experts_per_tok=8),
     # DBRX-132B geometry (BASELINE config 4): 16 experts top-4
     "dbrx": dict(hidden_size=6144, intermediate_size=10752,
                  num_hidden_layers=40, num_attention_heads=48,
@@ -226,7 +231,7 @@ def main():
         # captures = less surface on the multi-GPU scaling runs)
         enable_batch_bucketing=False,
     )
-    if args.model in ("mixtral", "dbrx"):
+    if args.model in ("mixtral", "dbrx", "qwen3moe"):
         # dbrx geometry is benchmarked through the same stacked-expert MoE
         # module (the DBRX model class differs in norms/clip, same compute
         # shape); random-init synthetic benchmarking per BASELINE
@@ -334,6 +339,7 @@ def main():
                 "model": {"8b": "llama-3.1-8b", "70b": "llama-3.1-70b",
                           "tiny": "tiny-llama-2l", "1b": "llama-3.2-1b",
                           "mixtral": "mixtral-8x7b",
+                          "qwen3moe": "qwen3-30b-a3b-geometry",
                           "dbrx": "dbrx-132b-geometry"}[args.model],
                 "global_batch": global_batch,
                 "seq_len": total_len,
