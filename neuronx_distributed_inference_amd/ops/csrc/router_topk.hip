@@ -15,82 +15,105 @@ namespace {
 
 constexpr int KMAX = 8;
 
-__global__ void router_topk_kernel(
+// ONE WAVE per token row (the first cut ran one THREAD per row and
+// serialized the E-loop — 193 us/launch at E=128, 24% of the Qwen3-MoE
+// step). Selection = k rounds of wave-argmax: each lane scans its
+// strided slice of the E logits (excluding already-chosen indices) and
+// a 6-step shuffle reduction finds the global (value, lowest-index)
+// max, broadcast to all lanes.
+__global__ __launch_bounds__(256) void router_topk_kernel(
     float* __restrict__ vals,        // (T, k)
     long* __restrict__ idx,          // (T, k)
     const float* __restrict__ logits,  // (T, E)
     const int T, const int E, const int k,
     const int act_sigmoid, const int pre_softmax, const int normalize) {
-    const int row = blockIdx.x * blockDim.x + threadIdx.x;
+    const int wid = threadIdx.x >> 6;
+    const int lane = threadIdx.x & 63;
+    const int row = blockIdx.x * 4 + wid;
     if (row >= T) return;
     const float* lr = logits + (long)row * E;
 
-    float bv[KMAX];
-    int bi[KMAX];
+    float cv[KMAX];
+    int ci[KMAX];
 #pragma unroll
     for (int j = 0; j < KMAX; ++j) {
-        bv[j] = -1e30f;
-        bi[j] = -1;
+        cv[j] = -1e30f;
+        ci[j] = -1;
     }
-    float fmax = -1e30f, fsum = 0.f;
-    for (int e = 0; e < E; ++e) {
-        const float v = lr[e];
-        fmax = fmaxf(fmax, v);
-        // descending insertion (stable: strict > keeps the earlier index
-        // on ties, matching torch.topk's lower-index preference)
+    for (int round = 0; round < k; ++round) {
+        float lv = -1e30f;
+        int li = E;  // sentinel larger than any index
+        for (int e = lane; e < E; e += 64) {
+            bool taken = false;
 #pragma unroll
-        for (int j = 0; j < KMAX; ++j) {
-            if (j >= k) break;
-            if (v > bv[j]) {
-                for (int m = k - 1; m > j; --m) {
-                    bv[m] = bv[m - 1];
-                    bi[m] = bi[m - 1];
-                }
-                bv[j] = v;
-                bi[j] = e;
-                break;
+            for (int j = 0; j < KMAX; ++j)
+                if (j < round) taken = taken || (e == ci[j]);
+            const float v = lr[e];
+            if (!taken && (v > lv || (v == lv && e < li))) {
+                lv = v;
+                li = e;
             }
         }
+#pragma unroll
+        for (int off = 32; off >= 1; off >>= 1) {
+            const float ov = __shfl_down(lv, off, 64);
+            const int oi = __shfl_down(li, off, 64);
+            if (ov > lv || (ov == lv && oi < li)) {
+                lv = ov;
+                li = oi;
+            }
+        }
+        lv = __shfl(lv, 0, 64);
+        li = __shfl(li, 0, 64);
+        cv[round] = lv;
+        ci[round] = li;
     }
+
     if (pre_softmax) {
-        for (int e = 0; e < E; ++e) fsum += __expf(lr[e] - fmax);
+        // full-E softmax: wave-reduce max (cv[0] IS the max) and sum
+        float ls = 0.f;
+        for (int e = lane; e < E; e += 64) ls += __expf(lr[e] - cv[0]);
+#pragma unroll
+        for (int off = 32; off >= 1; off >>= 1)
+            ls += __shfl_down(ls, off, 64);
+        ls = __shfl(ls, 0, 64);
 #pragma unroll
         for (int j = 0; j < KMAX; ++j)
-            if (j < k) bv[j] = __expf(bv[j] - fmax) / fsum;
+            if (j < k) cv[j] = __expf(cv[j] - cv[0]) / ls;
     } else if (act_sigmoid) {
 #pragma unroll
         for (int j = 0; j < KMAX; ++j)
-            if (j < k) bv[j] = 1.f / (1.f + __expf(-bv[j]));
+            if (j < k) cv[j] = 1.f / (1.f + __expf(-cv[j]));
     } else {
-        // softmax over the selected k (bv[0] is the max — snapshot it
-        // BEFORE the in-place exp overwrites it)
-        const float m0 = bv[0];
+        const float m0 = cv[0];
         float s = 0.f;
 #pragma unroll
         for (int j = 0; j < KMAX; ++j)
             if (j < k) {
-                bv[j] = __expf(bv[j] - m0);
-                s += bv[j];
+                cv[j] = __expf(cv[j] - m0);
+                s += cv[j];
             }
 #pragma unroll
         for (int j = 0; j < KMAX; ++j)
-            if (j < k) bv[j] /= s;
+            if (j < k) cv[j] /= s;
     }
     if (normalize) {
         float s = 0.f;
 #pragma unroll
         for (int j = 0; j < KMAX; ++j)
-            if (j < k) s += bv[j];
+            if (j < k) s += cv[j];
 #pragma unroll
         for (int j = 0; j < KMAX; ++j)
-            if (j < k) bv[j] /= fmaxf(s, 1e-20f);
+            if (j < k) cv[j] /= fmaxf(s, 1e-20f);
     }
+    if (lane == 0) {
 #pragma unroll
-    for (int j = 0; j < KMAX; ++j)
-        if (j < k) {
-            vals[(long)row * k + j] = bv[j];
-            idx[(long)row * k + j] = bi[j];
-        }
+        for (int j = 0; j < KMAX; ++j)
+            if (j < k) {
+                vals[(long)row * k + j] = cv[j];
+                idx[(long)row * k + j] = ci[j];
+            }
+    }
 }
 
 }  // namespace
@@ -106,8 +129,8 @@ void router_topk(torch::Tensor vals, torch::Tensor idx, torch::Tensor logits,
                 vals.scalar_type() == at::ScalarType::Float &&
                 idx.scalar_type() == at::ScalarType::Long, "out shapes");
     auto stream = at::hip::getCurrentHIPStream();
-    const int threads = 256;
-    const int blocks = (T + threads - 1) / threads;
+    const int threads = 256;          // 4 waves; one wave per row
+    const int blocks = (T + 3) / 4;
     hipLaunchKernelGGL(router_topk_kernel, dim3(blocks), dim3(threads), 0,
                        stream, vals.data_ptr<float>(), idx.data_ptr<long>(),
                        logits.data_ptr<float>(), T, E, (int)top_k,
