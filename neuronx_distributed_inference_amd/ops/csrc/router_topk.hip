@@ -70,16 +70,18 @@ __global__ __launch_bounds__(256) void router_topk_kernel(
     }
 
     if (pre_softmax) {
-        // full-E softmax: wave-reduce max (cv[0] IS the max) and sum
+        // full-E softmax: cv[0] IS the max — snapshot before the
+        // in-place exp overwrites it
+        const float fm = cv[0];
         float ls = 0.f;
-        for (int e = lane; e < E; e += 64) ls += __expf(lr[e] - cv[0]);
+        for (int e = lane; e < E; e += 64) ls += __expf(lr[e] - fm);
 #pragma unroll
         for (int off = 32; off >= 1; off >>= 1)
             ls += __shfl_down(ls, off, 64);
         ls = __shfl(ls, 0, 64);
 #pragma unroll
         for (int j = 0; j < KMAX; ++j)
-            if (j < k) cv[j] = __expf(cv[j] - cv[0]) / ls;
+            if (j < k) cv[j] = __expf(cv[j] - fm) / ls;
     } else if (act_sigmoid) {
 #pragma unroll
         for (int j = 0; j < KMAX; ++j)
