@@ -206,8 +206,25 @@ class ExpertMLPs(nn.Module):
         flat = (idx - offset).view(-1)
         local = (flat >= 0) & (flat < El)
         flat_routed = torch.where(local, flat, torch.full_like(flat, El))
+        fp8 = getattr(self, "fp8", False)
+        mx4 = getattr(self, "mx4", False)
+        # fine-grained MoE (few tokens per expert): a 32-row tile halves
+        # the padded-row waste. bf16 kernel only (quantized kernels are
+        # 64-row); opt-in via NXDI_MOE_TILE32=1 while it soaks.
+        import os as _os
+
+        if fp8 or mx4:
+            I0 = self.w_gate_q.shape[1]
+        else:
+            I0 = self.w_gate.shape[1]
+        tile = 32 if (
+            not fp8 and not mx4
+            and _os.environ.get("NXDI_MOE_TILE32") == "1"
+            and T * top_k < 48 * El
+            and I0 % 128 == 0 and H % 128 == 0
+        ) else 64
         order, dest, expert_ids, gtab, Mpad = ops.moe_grouped_tables(
-            flat_routed, El + 1, top_k)
+            flat_routed, El + 1, top_k, tile=tile)
         # sentinel segment (id El) never computes
         expert_ids = torch.where(expert_ids >= El,
                                  torch.full_like(expert_ids, -1), expert_ids)
@@ -218,7 +235,10 @@ class ExpertMLPs(nn.Module):
         # tiles cut x re-reads 4x but run at occupancy 1, so a small
         # grid (TP-sharded N or small batch) must fall back to narrower
         # tiles to fill the chip
-        def _bn(N, ntiles=Mpad // 64):
+        def _bn(N, ntiles=Mpad // tile):
+            if tile == 32:
+                return 256 if (N % 256 == 0
+                               and (N // 256) * ntiles >= 256) else 128
             # occupancy is 1 block/CU at bn>=192 (120 KB LDS), 2 below
             for cand in (256, 192, 128):
                 occ = 1 if cand >= 192 else 2
@@ -226,8 +246,6 @@ class ExpertMLPs(nn.Module):
                     return cand
             return 64
 
-        fp8 = getattr(self, "fp8", False)
-        mx4 = getattr(self, "mx4", False)
         if fp8 or mx4:
             I, Hn = self.w_gate_q.shape[1], self.w_down_q.shape[1]
         else:
@@ -261,9 +279,9 @@ class ExpertMLPs(nn.Module):
                                          expert_ids, bn=_bn8(I), x_gather=xg)
         else:
             g = ops.moe_grouped_gemm(x_in, self.w_gate, expert_ids, bn=_bn(I),
-                                     x_gather=xg)
+                                     x_gather=xg, mt=tile)
             u = ops.moe_grouped_gemm(x_in, self.w_up, expert_ids, bn=_bn(I),
-                                     x_gather=xg)
+                                     x_gather=xg, mt=tile)
         h = self.act_mul(g, u)
         # routing-weighted combine fuses into the down epilogue (fp32
         # atomics, one rmw per element); non-local assignments are
@@ -286,7 +304,7 @@ class ExpertMLPs(nn.Module):
         else:
             ops.moe_grouped_gemm(h, self.w_down, expert_ids, bn=_bn(Hn),
                                  combine_out=out32, combine_scale=gw_tab,
-                                 combine_gather=gtab)
+                                 combine_gather=gtab, mt=tile)
         return out32.to(x.dtype)
 
     def forward(self, x: torch.Tensor, weights: torch.Tensor, idx: torch.Tensor):

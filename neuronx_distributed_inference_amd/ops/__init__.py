@@ -673,7 +673,7 @@ MOE_TILE = 64
 
 
 def moe_grouped_tables(idx_flat: torch.Tensor, num_experts: int,
-                       top_k: int = 1):
+                       top_k: int = 1, tile: int = MOE_TILE):
     """idx_flat: (T*top_k,) local expert id per assignment. Returns
     (order, dest, expert_ids, gtab, Mpad):
       order      (Ttk,)  argsort by expert (stable on CPU; atomic —
@@ -689,7 +689,7 @@ def moe_grouped_tables(idx_flat: torch.Tensor, num_experts: int,
     E = num_experts
     Ttk = idx_flat.numel()
     dev = idx_flat.device
-    MT = MOE_TILE
+    MT = tile
     Mpad = ((Ttk + E * (MT - 1)) + MT - 1) // MT * MT
     if idx_flat.is_cuda:
         # ONE fused kernel (vs ~15 tiny torch launches = 140 us at decode
@@ -701,7 +701,8 @@ def moe_grouped_tables(idx_flat: torch.Tensor, num_experts: int,
         expert_ids = torch.empty(Mpad // MT, dtype=torch.int32, device=dev)
         gtab = torch.empty(Mpad, dtype=torch.int32, device=dev)
         c.moe_grouped_tables(order, dest, expert_ids, gtab,
-                             idx_flat.to(torch.int32).contiguous(), E, top_k)
+                             idx_flat.to(torch.int32).contiguous(), E, top_k,
+                             MT)
         return order, dest, expert_ids, gtab, Mpad
     order = torch.argsort(idx_flat, stable=True)
     sorted_e = idx_flat[order]
@@ -729,7 +730,7 @@ def moe_grouped_gemm(x_sorted: torch.Tensor, w: torch.Tensor,
                      expert_ids: torch.Tensor, bias=None,
                      bn: int = 0, x_gather=None, combine_out=None,
                      combine_scale=None, combine_gather=None,
-                     bk: int = 0) -> torch.Tensor:
+                     bk: int = 0, mt: int = MOE_TILE) -> torch.Tensor:
     """y_sorted[m] = x_sorted[m] @ w[expert_of(m)]^T (+ bias[expert]).
     x_sorted (Mpad, K), w (E, N, K), expert_ids (Mpad/64,) int32, bias
     (E, N) or None. GPU: one grouped HIP kernel; CPU reference:
@@ -745,7 +746,7 @@ def moe_grouped_gemm(x_sorted: torch.Tensor, w: torch.Tensor,
         weighted combine fuses into the epilogue. Returns combine_out.
     """
     K = x_sorted.shape[1]
-    Mpad = expert_ids.numel() * MOE_TILE if x_gather is not None \
+    Mpad = expert_ids.numel() * mt if x_gather is not None \
         else x_sorted.shape[0]
     N = w.shape[1]
     if x_sorted.is_cuda:
@@ -759,16 +760,16 @@ def moe_grouped_gemm(x_sorted: torch.Tensor, w: torch.Tensor,
                            combine_out,
                            combine_scale,
                            None if combine_gather is None
-                           else combine_gather.contiguous(), bk)
+                           else combine_gather.contiguous(), bk, mt)
         return combine_out if combine_out is not None else y
     # ---- CPU reference ----
     out = combine_out
     y = torch.zeros(Mpad, N, dtype=x_sorted.dtype)
-    for t in range(Mpad // MOE_TILE):
+    for t in range(Mpad // mt):
         e = int(expert_ids[t])
         if e < 0:
             continue
-        rows = slice(t * MOE_TILE, (t + 1) * MOE_TILE)
+        rows = slice(t * mt, (t + 1) * mt)
         if x_gather is not None:
             xg = x_sorted[x_gather[rows].clamp(min=0).long()]
         else:
@@ -779,8 +780,8 @@ def moe_grouped_gemm(x_sorted: torch.Tensor, w: torch.Tensor,
         if out is not None:
             cg = combine_gather[rows].long()
             sc = (combine_scale[rows] if combine_scale is not None
-                  else torch.ones(MOE_TILE))
-            for j in range(MOE_TILE):
+                  else torch.ones(mt))
+            for j in range(mt):
                 if cg[j] >= 0:
                     out[cg[j]] += r[j].float() * float(sc[j])
         else:

@@ -56,11 +56,12 @@ void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                       c10::optional<torch::Tensor> x_gather,
                       c10::optional<torch::Tensor> combine_out,
                       c10::optional<torch::Tensor> combine_scale,
-                      c10::optional<torch::Tensor> combine_gather, long bk);
+                      c10::optional<torch::Tensor> combine_gather, long bk,
+                      long mt);
 void moe_grouped_tables_gpu(torch::Tensor order, torch::Tensor dest,
                             torch::Tensor expert_ids, torch::Tensor gtab,
                             torch::Tensor idx_flat, long num_experts,
-                            long topk);
+                            long topk, long tile);
 void moe_grouped_gemm_fp8(torch::Tensor y, torch::Tensor x, torch::Tensor wq,
                           torch::Tensor wscale, torch::Tensor expert_ids,
                           c10::optional<torch::Tensor> bias, long bn,
@@ -141,7 +142,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused MoE segment tables: one launch vs the torch op chain",
           py::arg("order"), py::arg("dest"), py::arg("expert_ids"),
           py::arg("gtab"), py::arg("idx_flat"), py::arg("num_experts"),
-          py::arg("topk"));
+          py::arg("topk"), py::arg("tile") = 64);
     m.def("moe_grouped_gemm", &moe_grouped_gemm,
           "grouped expert GEMM over sorted padded token segments (CDNA4)",
           py::arg("y"), py::arg("x"), py::arg("w"), py::arg("expert_ids"),
@@ -149,7 +150,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("x_gather") = py::none(),
           py::arg("combine_out") = py::none(),
           py::arg("combine_scale") = py::none(),
-          py::arg("combine_gather") = py::none(), py::arg("bk") = 0);
+          py::arg("combine_gather") = py::none(), py::arg("bk") = 0,
+          py::arg("mt") = 64);
     m.def("moe_grouped_gemm_fp8", &moe_grouped_gemm_fp8,
           "grouped expert GEMM, fp8-e4m3 weights dequantized at the LDS "
           "fragment read, per-channel scale in the epilogue (CDNA4)",

@@ -84,7 +84,12 @@ DEV_INLINE void mg_glds_lane128(const char* src_lane, char* lds_dst_base) {
 // bound at these tile sizes (fixed wait/barrier cost per step), so the
 // deeper step — staged as (BKT/64) 128B half-images per operand —
 // halves the step count where LDS permits (bn<=128 at BKT=128).
-template <typename T, int BNT, int BKT = 64>
+// MTT: rows per tile. 32 halves the padded-row waste of fine-grained
+// MoE (128 experts top-8 decode runs ~16 tokens/expert — a 64-row tile
+// pads ~4x). At MTT=32 all 8 waves span the SAME 32 rows and split the
+// columns 8 ways (BNT >= 128), and only waves 0..3 stage the four 1 KB
+// x pieces (their vmcnt count is one higher — wave-uniform branch).
+template <typename T, int BNT, int BKT = 64, int MTT = 64>
 __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
     T* __restrict__ y,              // (Mpad, N) sorted-space output
     const T* __restrict__ x,        // (Mpad, K) sorted — or (T, K) with x_gather
@@ -98,11 +103,15 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
     const int N, const int K) {
     using MF = MgMfma<T>;
     using vec8 = typename MF::vec8;
-    constexpr int MT = 64;
+    constexpr int MT = MTT;
+    static_assert(MTT == 64 || (MTT == 32 && BKT == 64 && BNT >= 128),
+                  "MTT=32 supports BKT=64 and BNT>=128");
     constexpr int NH = BKT / 64;    // 128B half-images per operand per step
     constexpr int NWG = BNT / 64;   // W glds per wave per half-image
-    constexpr int G = (1 + NWG) * NH;  // total glds per wave per step
-    constexpr int NFRAG = BNT / 64; // 16-col fragments per wave
+    constexpr int NXW = MTT / 8;    // waves that stage an x piece (4 or 8)
+    constexpr int G = (MTT == 64 ? (1 + NWG) * NH : NWG);  // per ordinary wave
+    constexpr int NFRAG = (MTT == 64 ? BNT / 64 : BNT / 128);
+    constexpr int WNW = (MTT == 64 ? 4 : 8);   // waves along N
     constexpr int XBYTES = NH * MT * 128;
     constexpr int CHUNK_B = XBYTES + NH * BNT * 128;
 
@@ -118,14 +127,16 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
 
     const int lane = threadIdx.x & 63;
     const int wid = threadIdx.x >> 6;
-    const int wm0 = (wid >> 2) * 32;        // 2 waves along M
-    const int wn0 = (wid & 3) * (BNT / 4);  // 4 waves along N
+    const int wm0 = (MTT == 64 ? (wid >> 2) * 32 : 0);
+    const int wn0 = (wid % WNW) * (BNT / WNW);
     const int lrow = lane & 15;
     const int lslot = lane >> 4;
 
     // per-lane x source: row via gather table (fused token gather) or
-    // linear sorted row; XOR source-swizzle baked into the address
-    const int xr_lin = m0 + wid * 8 + (lane >> 3);
+    // linear sorted row; XOR source-swizzle baked into the address.
+    // MTT=32: only waves 0..3 stage x (one 8-row piece each)
+    const bool x_wave = (MTT == 64) || wid < NXW;
+    const int xr_lin = m0 + (x_wave ? wid * 8 : 0) + (lane >> 3);
     const int xr = x_gather ? max(x_gather[xr_lin], 0) : xr_lin;
     const int xslot = (lane & 7) ^ (lane >> 3);
     const char* xsrc_lane = reinterpret_cast<const char*>(x) +
@@ -140,7 +151,7 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
     char* wdst[MG_NBUF];
 #pragma unroll
     for (int b = 0; b < MG_NBUF; ++b) {
-        xdst[b] = smem + b * CHUNK_B + wid * 1024;
+        xdst[b] = smem + b * CHUNK_B + (x_wave ? wid : 0) * 1024;
         wdst[b] = smem + b * CHUNK_B + XBYTES + wid * (NWG * 1024);
     }
 
@@ -148,8 +159,9 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
     do {                                                                     \
         const long koff = (long)(t) * (BKT * 2);                             \
         _Pragma("unroll") for (int h = 0; h < NH; ++h) {                     \
-            mg_glds_lane128(xsrc_lane + koff + h * 128,                      \
-                            xdst[b] + h * (MT * 128));                       \
+            if (x_wave)                                                      \
+                mg_glds_lane128(xsrc_lane + koff + h * 128,                  \
+                                xdst[b] + h * (MT * 128));                   \
             _Pragma("unroll") for (int i = 0; i < NWG; ++i)                  \
                 mg_glds_rows128<2>(wsrc[i] + koff + h * 128,                 \
                                    (long)K * sizeof(T),                      \
@@ -196,8 +208,10 @@ __global__ __launch_bounds__(512, 2) void moe_grouped_kernel(
 
 #define MG_BODY(t, b, b2)                                                    \
     do {                                                                     \
-        if ((t) + 1 < nchunks) mg_vm_wait<G>();                              \
-        else mg_vm_wait<0>();                                                \
+        if ((t) + 1 < nchunks) {                                             \
+            if (MTT == 32 && x_wave) mg_vm_wait<G + 1>();                    \
+            else mg_vm_wait<G>();                                            \
+        } else mg_vm_wait<0>();                                              \
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                   \
         __builtin_amdgcn_s_barrier();                                        \
         MG_FRAGS(b);                                                         \
@@ -894,10 +908,11 @@ constexpr int MT_TABLES = 1024;  // threads
 __global__ __launch_bounds__(MT_TABLES, 1) void moe_tables_kernel(
     long* __restrict__ order,        // (Ttk,) sorted pos -> assignment idx
     long* __restrict__ dest,         // (Ttk,) sorted pos -> padded row
-    int* __restrict__ expert_ids,    // (ntiles,) expert per 64-row tile
-    int* __restrict__ gtab,          // (ntiles*64,) padded row -> token, -1 pad
+    int* __restrict__ expert_ids,    // (ntiles,) expert per row tile
+    int* __restrict__ gtab,          // (ntiles*tile,) padded row -> token, -1 pad
     const int* __restrict__ idx,     // (Ttk,) expert per assignment
-    const int Ttk, const int E, const int ntiles, const int topk) {
+    const int Ttk, const int E, const int ntiles, const int topk,
+    const int tile) {
     extern __shared__ int tsm[];
     int* counts = tsm;               // (E,)
     int* starts_raw = tsm + E;       // (E,) exclusive scan of counts
@@ -919,11 +934,11 @@ __global__ __launch_bounds__(MT_TABLES, 1) void moe_tables_kernel(
             starts_raw[e] = raw;
             starts_pad[e] = pad;
             raw += counts[e];
-            pad += (counts[e] + 63) / 64 * 64;
+            pad += (counts[e] + tile - 1) / tile * tile;
         }
         total_pad = pad;
     }
-    for (int m = threadIdx.x; m < ntiles * 64; m += blockDim.x)
+    for (int m = threadIdx.x; m < ntiles * tile; m += blockDim.x)
         gtab[m] = -1;                             // padding rows
     __syncthreads();
     for (int i = threadIdx.x; i < Ttk; i += blockDim.x) {
@@ -937,12 +952,12 @@ __global__ __launch_bounds__(MT_TABLES, 1) void moe_tables_kernel(
     }
     // expert per row tile: linear search over E (ntiles and E are small)
     for (int t = threadIdx.x; t < ntiles; t += blockDim.x) {
-        const int row = t * 64;
+        const int row = t * tile;
         int eid = -1;
         if (row < total_pad) {
             for (int e = 0; e < E; ++e) {
                 const int end =
-                    starts_pad[e] + (counts[e] + 63) / 64 * 64;
+                    starts_pad[e] + (counts[e] + tile - 1) / tile * tile;
                 if (row >= starts_pad[e] && row < end && counts[e] > 0) {
                     eid = e;
                     break;
@@ -958,7 +973,7 @@ __global__ __launch_bounds__(MT_TABLES, 1) void moe_tables_kernel(
 void moe_grouped_tables_gpu(torch::Tensor order, torch::Tensor dest,
                             torch::Tensor expert_ids, torch::Tensor gtab,
                             torch::Tensor idx_flat, long num_experts,
-                            long topk) {
+                            long topk, long tile) {
     const int Ttk = idx_flat.numel();
     const int E = (int)num_experts;
     const int ntiles = expert_ids.numel();
@@ -969,14 +984,16 @@ void moe_grouped_tables_gpu(torch::Tensor order, torch::Tensor dest,
                 expert_ids.scalar_type() == at::ScalarType::Int &&
                 gtab.scalar_type() == at::ScalarType::Int, "out dtypes");
     TORCH_CHECK(order.numel() == Ttk && dest.numel() == Ttk &&
-                gtab.numel() == ntiles * 64, "out sizes");
+                gtab.numel() == ntiles * tile, "out sizes");
+    TORCH_CHECK(tile == 64 || tile == 32, "tile must be 32 or 64");
     TORCH_CHECK(E <= 4096, "num_experts too large for LDS tables");
     const int lds = 4 * E * (int)sizeof(int);
     auto stream = at::hip::getCurrentHIPStream();
     hipLaunchKernelGGL(moe_tables_kernel, dim3(1), dim3(MT_TABLES), lds,
                        stream, order.data_ptr<long>(), dest.data_ptr<long>(),
                        expert_ids.data_ptr<int>(), gtab.data_ptr<int>(),
-                       idx_flat.data_ptr<int>(), Ttk, E, ntiles, (int)topk);
+                       idx_flat.data_ptr<int>(), Ttk, E, ntiles, (int)topk,
+                       (int)tile);
     HIP_CHECK_KERNEL();
 }
 
@@ -986,21 +1003,25 @@ void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                       c10::optional<torch::Tensor> x_gather,
                       c10::optional<torch::Tensor> combine_out,
                       c10::optional<torch::Tensor> combine_scale,
-                      c10::optional<torch::Tensor> combine_gather, long bk) {
+                      c10::optional<torch::Tensor> combine_gather, long bk,
+                      long mt) {
     TORCH_CHECK(y.dim() == 2 && x.dim() == 2 && w.dim() == 3, "shapes");
     TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && y.is_contiguous() &&
                 expert_ids.is_contiguous(), "contiguous tensors required");
     const int K = x.size(1);
     const int N = w.size(1);
+    const int MTT = mt ? (int)mt : 64;
+    TORCH_CHECK(MTT == 64 || MTT == 32, "mt must be 32 or 64");
     // with a gather table, x holds the UNSORTED tokens; Mpad comes from
     // the tile table
-    const int Mpad = x_gather.has_value() ? (int)expert_ids.numel() * 64
+    const int Mpad = x_gather.has_value() ? (int)expert_ids.numel() * MTT
                                           : (int)x.size(0);
     TORCH_CHECK(w.size(2) == K && y.size(0) == Mpad && y.size(1) == N, "shape");
     TORCH_CHECK(K % MG_BK == 0, "K must be a multiple of 64");
-    TORCH_CHECK(Mpad % 64 == 0, "padded rows must be a multiple of 64");
+    TORCH_CHECK(Mpad % MTT == 0, "padded rows must be a multiple of mt");
     TORCH_CHECK(expert_ids.scalar_type() == at::ScalarType::Int &&
-                expert_ids.numel() == Mpad / 64, "expert_ids must be int (Mpad/64,)");
+                expert_ids.numel() == Mpad / MTT,
+                "expert_ids must be int (Mpad/mt,)");
     if (x_gather.has_value())
         TORCH_CHECK(x_gather->scalar_type() == at::ScalarType::Int &&
                     x_gather->numel() == Mpad && x_gather->is_contiguous(),
@@ -1034,8 +1055,10 @@ void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
     TORCH_CHECK(BKT == 64 || BNT <= 128, "bk=128 needs bn<=128 (LDS)");
     TORCH_CHECK(K % (MG_BK * (BKT / 64)) == 0, "K must be a multiple of bk");
 
+    TORCH_CHECK(MTT == 64 || (BKT == 64 && (BNT == 128 || BNT == 256)),
+                "mt=32 needs bk=64 and bn in {128, 256}");
     auto stream = at::hip::getCurrentHIPStream();
-    dim3 grid(N / BNT, Mpad / 64);
+    dim3 grid(N / BNT, Mpad / MTT);
 
 #define MG_LAUNCH(T, BNTV, BKTV)                                              \
     do {                                                                      \
@@ -1063,9 +1086,37 @@ void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
                                           : nullptr,                          \
                            N, K);                                             \
     } while (0)
+#define MG_LAUNCH32(T, BNTV)                                                  \
+    do {                                                                      \
+        constexpr int LDS = MG_NBUF * (32 * 128 + BNTV * 128);                \
+        static bool attr_set = false;                                         \
+        if (!attr_set) {                                                      \
+            (void)hipFuncSetAttribute(                                        \
+                (const void*)moe_grouped_kernel<T, BNTV, 64, 32>,             \
+                hipFuncAttributeMaxDynamicSharedMemorySize, LDS);             \
+            attr_set = true;                                                  \
+        }                                                                     \
+        hipLaunchKernelGGL((moe_grouped_kernel<T, BNTV, 64, 32>), grid,       \
+                           dim3(512),                                         \
+                           LDS, stream, (T*)y.data_ptr(),                     \
+                           (const T*)x.data_ptr(),                            \
+                           (const T*)w.data_ptr(), expert_ids.data_ptr<int>(),\
+                           bias ? (const T*)bias->data_ptr() : nullptr,       \
+                           x_gather ? x_gather->data_ptr<int>() : nullptr,    \
+                           combine_out ? combine_out->data_ptr<float>()       \
+                                       : nullptr,                             \
+                           combine_scale ? combine_scale->data_ptr<float>()   \
+                                         : nullptr,                           \
+                           combine_gather ? combine_gather->data_ptr<int>()   \
+                                          : nullptr,                          \
+                           N, K);                                             \
+    } while (0)
 #define MG_T(T)                                                               \
     do {                                                                      \
-        if (BKT == 128) {                                                     \
+        if (MTT == 32) {                                                      \
+            if (BNT == 128) MG_LAUNCH32(T, 128);                              \
+            else MG_LAUNCH32(T, 256);                                         \
+        } else if (BKT == 128) {                                              \
             if (BNT == 64) MG_LAUNCH(T, 64, 128);                             \
             else MG_LAUNCH(T, 128, 128);                                      \
         } else if (BNT == 64) MG_LAUNCH(T, 64, 64);                           \
@@ -1082,6 +1133,7 @@ void moe_grouped_gemm(torch::Tensor y, torch::Tensor x, torch::Tensor w,
         TORCH_CHECK(false, "moe_grouped_gemm: bf16/f16 only");
     }
 #undef MG_T
+#undef MG_LAUNCH32
 #undef MG_LAUNCH
     HIP_CHECK_KERNEL();
 }

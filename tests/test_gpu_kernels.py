@@ -1277,3 +1277,26 @@ def test_router_topk_fused_gpu(mode):
     got_v, got_i = r(x.to(dev))
     assert torch.equal(got_i.cpu(), want_i), mode
     torch.testing.assert_close(got_v.cpu(), want_v, atol=1e-5, rtol=1e-5)
+
+
+def test_moe_grouped_tile32_gpu(monkeypatch):
+    """mt=32 grouped kernel (fine-grained tile) vs the dense bmm path on
+    GPU at a 128-expert top-8 shape."""
+    from neuronx_distributed_inference_amd.modules.moe import ExpertMLPs
+
+    monkeypatch.setenv("NXDI_MOE_TILE32", "1")
+    torch.manual_seed(15)
+    dev = _dev()
+    E, H, I, T, top_k = 32, 256, 384, 48, 8
+    m = ExpertMLPs(num_experts=E, hidden_size=H, intermediate_size=I,
+                   tp_degree=1, dtype=torch.bfloat16)
+    with torch.no_grad():
+        for p in m.parameters():
+            p.normal_(0, 0.05)
+    m = m.to(dev)
+    x = torch.randn(T, H, dtype=torch.bfloat16, device=dev) * 0.3
+    w, idx = torch.softmax(torch.randn(T, E, device=dev), -1).topk(top_k, -1)
+    want = m._forward_dense(x, w, idx)
+    got = m._forward_grouped(x, w, idx)
+    torch.testing.assert_close(got.float(), want.float(),
+                               atol=5e-2, rtol=5e-2)

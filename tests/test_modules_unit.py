@@ -675,3 +675,24 @@ def test_moe_mx4_grouped_cpu_matches_bf16():
     got = m._forward_grouped(x, w, idx)
     err = (got - want).abs().max() / want.abs().max().clamp(min=1e-6)
     assert err < 0.35, float(err)  # 4-bit weights: coarse but bounded
+
+
+def test_moe_grouped_tile32_cpu_matches_dense(monkeypatch):
+    """32-row tile path (fine-grained MoE) vs the dense path on CPU."""
+    import torch
+
+    from neuronx_distributed_inference_amd.modules.moe import ExpertMLPs
+
+    monkeypatch.setenv("NXDI_MOE_TILE32", "1")
+    torch.manual_seed(14)
+    E, H, I, T, top_k = 16, 128, 256, 24, 4   # 6 tokens/expert avg
+    m = ExpertMLPs(num_experts=E, hidden_size=H, intermediate_size=I,
+                   tp_degree=1, dtype=torch.float32)
+    with torch.no_grad():
+        for p in m.parameters():
+            p.normal_(0, 0.05)
+    x = torch.randn(T, H) * 0.3
+    w, idx = torch.softmax(torch.randn(T, E), -1).topk(top_k, dim=-1)
+    want = m._forward_dense(x, w, idx)
+    got = m._forward_grouped(x, w, idx)
+    torch.testing.assert_close(got, want, atol=1e-4, rtol=1e-4)
