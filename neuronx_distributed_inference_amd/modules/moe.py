@@ -219,7 +219,7 @@ class ExpertMLPs(nn.Module):
             I0 = self.w_gate.shape[1]
         tile = 32 if (
             not fp8 and not mx4
-            and _os.environ.get("NXDI_MOE_TILE32") == "1"
+            and _os.environ.get("NXDI_MOE_TILE32") != "0"
             and T * top_k < 48 * El
             and I0 % 128 == 0 and H % 128 == 0
         ) else 64
